@@ -1,0 +1,579 @@
+/*
+ * oracle.c — CPU restatement of the Cloudberry hot path (TEST INFRASTRUCTURE).
+ *
+ * Each function cites the reference file:line it follows (paths relative to
+ * /root/reference).  See oracle.h for the usage contract: this library is
+ * the parity checker and reported CPU baseline only — never the product path.
+ */
+#include "oracle.h"
+#include <stdlib.h>
+#include <string.h>
+#include <math.h>
+
+/* ======================================================================
+ * Hashing — bit-exact restatements
+ * ====================================================================== */
+
+static inline uint32_t rot32(uint32_t x, int k) { return (x << k) | (x >> (32 - k)); }
+
+/*
+ * Jenkins final() — src/common/hashfn.c:133-142.
+ */
+#define ORC_FINAL(a,b,c) \
+{ \
+  c ^= b; c -= rot32(b,14); \
+  a ^= c; a -= rot32(c,11); \
+  b ^= a; b -= rot32(a,25); \
+  c ^= b; c -= rot32(b,16); \
+  a ^= c; a -= rot32(c, 4); \
+  b ^= a; b -= rot32(a,14); \
+  c ^= b; c -= rot32(b,24); \
+}
+
+/* hash_bytes_uint32 — src/common/hashfn.c:620-637 */
+uint32_t orc_hash_bytes_uint32(uint32_t k)
+{
+    uint32_t a, b, c;
+    a = b = c = 0x9e3779b9 + (uint32_t) sizeof(uint32_t) + 3923095;
+    a += k;
+    ORC_FINAL(a, b, c);
+    return c;
+}
+
+/* hashint8 — src/backend/access/hash/hashfunc.c:85-101 */
+uint32_t orc_hashint8(int64_t val)
+{
+    uint32_t lohalf = (uint32_t) val;
+    uint32_t hihalf = (uint32_t) ((uint64_t) val >> 32);
+    lohalf ^= (val >= 0) ? hihalf : ~hihalf;
+    return orc_hash_bytes_uint32(lohalf);
+}
+
+/*
+ * cdbhash for a single not-null int8 key —
+ * src/backend/cdb/cdbhash.c:171-247: cdbhashinit (hash=0, non-legacy) then
+ * per attribute: rotate-left-1 then XOR the type hash (hashint8 here).
+ */
+uint32_t orc_cdbhash_i64(int64_t v)
+{
+    uint32_t hashkey = 0;                                  /* cdbhashinit */
+    hashkey = (hashkey << 1) | ((hashkey & 0x80000000u) ? 1 : 0);
+    hashkey ^= orc_hashint8(v);
+    return hashkey;
+}
+
+/* jump_consistent_hash — src/backend/cdb/cdbhash.c:530-541 (Lamping-Veach) */
+int32_t orc_jump_consistent_hash(uint64_t key, int32_t num_segments)
+{
+    int64_t b = -1;
+    int64_t j = 0;
+    while (j < num_segments)
+    {
+        b = j;
+        key = key * 2862933555777941757ULL + 1;
+        j = (int64_t) ((double) (b + 1) *
+                       ((double) (1LL << 31) / (double) ((key >> 33) + 1)));
+    }
+    return (int32_t) b;
+}
+
+/*
+ * Motion / DISTRIBUTED BY routing for an int8 key —
+ * nodeMotion.c:1088 evalHashKey → cdbhashreduce (cdbhash.c:253-285),
+ * REDUCE_JUMP_HASH branch (the non-legacy default).
+ */
+int32_t orc_route_i64(int64_t key, int32_t nsegs)
+{
+    return orc_jump_consistent_hash((uint64_t) orc_cdbhash_i64(key), nsegs);
+}
+
+void orc_route_i64_batch(const int64_t *keys, int64_t n, int32_t nsegs, int32_t *out)
+{
+    for (int64_t i = 0; i < n; i++)
+        out[i] = orc_route_i64(keys[i], nsegs);
+}
+
+/*
+ * CRC-32C in PostgreSQL COMP_CRC32C semantics (reflected Castagnoli,
+ * init 0xFFFFFFFF, NO final inversion — "by historical accident" the AO
+ * block checksums are not inverted, cdbappendonlystorageformat.c:41-47).
+ */
+static uint32_t crc32c_table[256];
+static int crc32c_ready = 0;
+static void crc32c_init(void)
+{
+    for (uint32_t i = 0; i < 256; i++)
+    {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++)
+            c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+        crc32c_table[i] = c;
+    }
+    crc32c_ready = 1;
+}
+uint32_t orc_crc32c(uint32_t crc, const void *buf, size_t len)
+{
+    const uint8_t *p = (const uint8_t *) buf;
+    if (!crc32c_ready) crc32c_init();
+    while (len--)
+        crc = crc32c_table[(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+    return crc;
+}
+
+/* ======================================================================
+ * Dates — DateADT is int32 days since 2000-01-01 (src/include/utils/date.h:23,
+ * POSTGRES_EPOCH_JDATE datatype/timestamp.h:209).
+ * ====================================================================== */
+
+/* days_from_civil (Howard Hinnant's algorithm), shifted to the 2000-01-01 epoch */
+int32_t orc_date_adt(int y, int m, int d)
+{
+    int64_t yy = y - (m <= 2);
+    int64_t era = (yy >= 0 ? yy : yy - 399) / 400;
+    unsigned yoe = (unsigned) (yy - era * 400);
+    unsigned doy = (153u * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
+    unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
+    int64_t days_unix = era * 146097 + (int64_t) doe - 719468;   /* days since 1970-01-01 */
+    return (int32_t) (days_unix - 10957);                        /* 2000-01-01 is unix day 10957 */
+}
+
+/* ======================================================================
+ * Deterministic synthetic TPC-H-shaped data (SURVEY §8d distributions).
+ * THE CONTRACT: identical formulas in cloudberry_amd/csrc/gx_kernels.hip.
+ * ====================================================================== */
+
+uint64_t orc_splitmix64(uint64_t x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+uint64_t orc_mix(uint64_t seed, uint64_t stream, uint64_t idx)
+{
+    return orc_splitmix64(orc_splitmix64(seed ^ (stream * 0xA24BAED4963EE407ULL)) + idx);
+}
+
+/* stream ids (shared constants) */
+enum {
+    ST_CUST_SEG = 1,
+    ST_ORD_CUST = 2,
+    ST_ORD_DATE = 3,
+    ST_ORD_PRIO = 4,
+    ST_LI_COUNT = 5,
+    ST_LI_SHIP  = 6,
+    ST_LI_PRICE = 7,
+    ST_LI_DISC  = 8,
+};
+
+int64_t orc_ncustomer(double sf) { return (int64_t) (150000.0 * sf + 0.5); }
+int64_t orc_norders(double sf)   { return (int64_t) (1500000.0 * sf + 0.5); }
+
+/* date ranges (TPC-H 4.3.2/4.3.3): orders 1992-01-01..1998-08-02,
+ * lineitem shipdate 1992-01-02..1998-12-01 */
+#define ORD_DATE_LO() orc_date_adt(1992, 1, 1)
+#define ORD_DATE_SPAN 2405   /* inclusive span-1; lo + mix%(span+1) */
+#define LI_DATE_LO()  orc_date_adt(1992, 1, 2)
+#define LI_DATE_SPAN  2525
+
+static inline uint8_t gen_mktsegment(uint64_t seed, int64_t i)
+{ return (uint8_t) (orc_mix(seed, ST_CUST_SEG, (uint64_t) i) % 5); }
+
+static inline int64_t gen_ocustkey(uint64_t seed, int64_t i, int64_t ncust)
+{
+    int64_t pool = (ncust * 2) / 3;            /* 1/3 of customers orderless */
+    if (pool < 1) pool = 1;
+    return 1 + (int64_t) (orc_mix(seed, ST_ORD_CUST, (uint64_t) i) % (uint64_t) pool);
+}
+static inline int32_t gen_odate(uint64_t seed, int64_t i)
+{ return ORD_DATE_LO() + (int32_t) (orc_mix(seed, ST_ORD_DATE, (uint64_t) i) % (ORD_DATE_SPAN + 1)); }
+static inline int32_t gen_oprio(uint64_t seed, int64_t i)
+{ return (int32_t) (orc_mix(seed, ST_ORD_PRIO, (uint64_t) i) % 5); }
+
+static inline int32_t gen_nlines(uint64_t seed, int64_t okey)
+{ return 1 + (int32_t) (orc_mix(seed, ST_LI_COUNT, (uint64_t) okey) % 7); }
+static inline int32_t gen_shipdate(uint64_t seed, int64_t okey, int32_t line)
+{ return LI_DATE_LO() + (int32_t) (orc_mix(seed, ST_LI_SHIP, (uint64_t) okey * 8 + line) % (LI_DATE_SPAN + 1)); }
+static inline double gen_price(uint64_t seed, int64_t okey, int32_t line)
+{ return (double) (90000 + orc_mix(seed, ST_LI_PRICE, (uint64_t) okey * 8 + line) % 10410001ULL) / 100.0; }
+static inline double gen_discount(uint64_t seed, int64_t okey, int32_t line)
+{ return (double) (orc_mix(seed, ST_LI_DISC, (uint64_t) okey * 8 + line) % 11) / 100.0; }
+
+int orc_gen_customer(double sf, uint64_t seed, int seg, int nsegs, orc_customer *out)
+{
+    int64_t n = orc_ncustomer(sf), kept = 0;
+    int64_t cap = (nsegs == 1) ? n : (n / nsegs + (int64_t) (4.0 * sqrt((double) n / nsegs)) + 64);
+    out->c_custkey = malloc(sizeof(int64_t) * cap);
+    out->c_mktsegment = malloc(cap);
+    if (!out->c_custkey || !out->c_mktsegment) return -1;
+    for (int64_t i = 0; i < n; i++)
+    {
+        int64_t key = i + 1;
+        if (nsegs > 1 && orc_route_i64(key, nsegs) != seg) continue;
+        if (kept == cap)
+        {
+            cap = cap * 2;
+            out->c_custkey = realloc(out->c_custkey, sizeof(int64_t) * cap);
+            out->c_mktsegment = realloc(out->c_mktsegment, cap);
+        }
+        out->c_custkey[kept] = key;
+        out->c_mktsegment[kept] = gen_mktsegment(seed, i);
+        kept++;
+    }
+    out->n = kept;
+    return 0;
+}
+
+int orc_gen_orders(double sf, uint64_t seed, int seg, int nsegs, orc_orders *out)
+{
+    int64_t n = orc_norders(sf), ncust = orc_ncustomer(sf), kept = 0;
+    int64_t cap = (nsegs == 1) ? n : (n / nsegs + (int64_t) (4.0 * sqrt((double) n / nsegs)) + 64);
+    out->o_orderkey = malloc(sizeof(int64_t) * cap);
+    out->o_custkey = malloc(sizeof(int64_t) * cap);
+    out->o_orderdate = malloc(sizeof(int32_t) * cap);
+    out->o_shippriority = malloc(sizeof(int32_t) * cap);
+    for (int64_t i = 0; i < n; i++)
+    {
+        int64_t key = i + 1;
+        if (nsegs > 1 && orc_route_i64(key, nsegs) != seg) continue;
+        if (kept == cap)
+        {
+            cap *= 2;
+            out->o_orderkey = realloc(out->o_orderkey, sizeof(int64_t) * cap);
+            out->o_custkey = realloc(out->o_custkey, sizeof(int64_t) * cap);
+            out->o_orderdate = realloc(out->o_orderdate, sizeof(int32_t) * cap);
+            out->o_shippriority = realloc(out->o_shippriority, sizeof(int32_t) * cap);
+        }
+        out->o_orderkey[kept] = key;
+        out->o_custkey[kept] = gen_ocustkey(seed, i, ncust);
+        out->o_orderdate[kept] = gen_odate(seed, i);
+        out->o_shippriority[kept] = gen_oprio(seed, i);
+        kept++;
+    }
+    out->n = kept;
+    return 0;
+}
+
+int orc_gen_lineitem(double sf, uint64_t seed, int seg, int nsegs, orc_lineitem *out)
+{
+    int64_t nord = orc_norders(sf), kept = 0;
+    int64_t cap = (nsegs == 1) ? nord * 4 + 64 : (nord * 4 / nsegs + (int64_t) (8.0 * sqrt((double) nord * 4 / nsegs)) + 64);
+    out->l_orderkey = malloc(sizeof(int64_t) * cap);
+    out->l_extendedprice = malloc(sizeof(double) * cap);
+    out->l_discount = malloc(sizeof(double) * cap);
+    out->l_shipdate = malloc(sizeof(int32_t) * cap);
+    for (int64_t o = 1; o <= nord; o++)
+    {
+        if (nsegs > 1 && orc_route_i64(o, nsegs) != seg) continue;
+        int32_t nl = gen_nlines(seed, o);
+        if (kept + nl > cap)
+        {
+            cap = cap * 2 + nl;
+            out->l_orderkey = realloc(out->l_orderkey, sizeof(int64_t) * cap);
+            out->l_extendedprice = realloc(out->l_extendedprice, sizeof(double) * cap);
+            out->l_discount = realloc(out->l_discount, sizeof(double) * cap);
+            out->l_shipdate = realloc(out->l_shipdate, sizeof(int32_t) * cap);
+        }
+        for (int32_t j = 0; j < nl; j++)
+        {
+            out->l_orderkey[kept] = o;
+            out->l_extendedprice[kept] = gen_price(seed, o, j);
+            out->l_discount[kept] = gen_discount(seed, o, j);
+            out->l_shipdate[kept] = gen_shipdate(seed, o, j);
+            kept++;
+        }
+    }
+    out->n = kept;
+    return 0;
+}
+
+void orc_free_customer(orc_customer *c)
+{ free(c->c_custkey); free(c->c_mktsegment); c->c_custkey = NULL; c->c_mktsegment = NULL; }
+void orc_free_orders(orc_orders *o)
+{ free(o->o_orderkey); free(o->o_custkey); free(o->o_orderdate); free(o->o_shippriority);
+  o->o_orderkey = o->o_custkey = NULL; o->o_orderdate = o->o_shippriority = NULL; }
+void orc_free_lineitem(orc_lineitem *l)
+{ free(l->l_orderkey); free(l->l_extendedprice); free(l->l_discount); free(l->l_shipdate);
+  l->l_orderkey = NULL; l->l_extendedprice = l->l_discount = NULL; l->l_shipdate = NULL; }
+void orc_free(void *p) { free(p); }
+
+/* ======================================================================
+ * AOCS column-store codec — byte-exact vs the reference writer for
+ * appendonly=column, compresstype=none, checksum=true, NOT NULL fixed-width.
+ *
+ * Stream = AO SmallContent blocks (cdbappendonlystorage_int.h:64-147):
+ *   [0..8)   header bitfields  (headerKind=1, hasFirstRowNum=1,
+ *            executorBlockKind=1 = AOCSBK_BLOCK datumstream.c:39-44)
+ *   [8..12)  block CRC32C over [16, blockLen)   (cdbappendonlystorageformat.c:50-78)
+ *   [12..16) header CRC32C over [0,12)          (cdbappendonlystorageformat.c:25-48)
+ *   [16..24) firstRowNum int64                  (cdbappendonlystorageformat.c:89-123)
+ *   [24..)   content: DatumStreamBlock_Orig 16B (datumstreamblock.h:73-84)
+ *            {version=0,flags=0,ndatum,encrypted=0,nullsz=0,sz} then datums
+ *            at MAXALIGN(16)=16 (datumstreamblock.c:273-279)
+ *   whole block rounded up to 8 B (AOStorage_RoundUp8, cdbappendonlystorage.h:39)
+ *
+ * Capacity per block (DatumStreamBlockWrite_OrigHasSpace,
+ * datumstreamblock.c:1508-1560 with maxDataBlockSize = blocksize −
+ * AoHeader_Size(false,true,true)=24, datumstream.c:588-631):
+ *   accept while 16 + n*width + width < blocksize-24  and  n+1 < 16383.
+ * ====================================================================== */
+
+int32_t orc_aocs_rows_per_block(int width, int32_t blocksize)
+{
+    int32_t maxdata = blocksize - 24;
+    int32_t n = 0;
+    while ((n + 1 < 16383) && (16 + n * width + width < maxdata))
+        n++;
+    return n;                    /* rows in every full block */
+}
+
+int64_t orc_aocs_encoded_size(int width, int64_t nrows, int32_t blocksize)
+{
+    int32_t rpb = orc_aocs_rows_per_block(width, blocksize);
+    int64_t nblocks = (nrows + rpb - 1) / rpb;
+    int64_t sz = 0;
+    for (int64_t b = 0; b < nblocks; b++)
+    {
+        int64_t rows = (b == nblocks - 1) ? (nrows - b * (int64_t) rpb) : rpb;
+        int64_t content = 16 + rows * width;
+        sz += (24 + content + 7) & ~7LL;
+    }
+    return sz;
+}
+
+static void put_u32le(uint8_t *p, uint32_t v) { memcpy(p, &v, 4); }
+
+int64_t orc_aocs_encode(const void *vals, int width, int64_t nrows,
+                        int64_t first_rownum, int32_t blocksize,
+                        uint8_t *out, int64_t outcap)
+{
+    int32_t rpb = orc_aocs_rows_per_block(width, blocksize);
+    const uint8_t *src = (const uint8_t *) vals;
+    int64_t off = 0, row = 0;
+    while (row < nrows)
+    {
+        int32_t rows = (int32_t) ((nrows - row < rpb) ? (nrows - row) : rpb);
+        int32_t sz = rows * width;
+        int32_t content = 16 + sz;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        if (off + blocklen > outcap) return -1;
+        uint8_t *blk = out + off;
+        memset(blk, 0, blocklen);
+
+        /* AOSmallContentHeader Init macros, cdbappendonlystorage_int.h:150-170 */
+        uint32_t b03 = 0, b47 = 0;
+        b03 |= (1u << 28);                       /* headerKind = SmallContent */
+        b03 |= (1u << 27);                       /* hasFirstRowNum */
+        b03 |= (1u << 24);                       /* executorBlockKind = AOCSBK_BLOCK */
+        b03 |= (0x00FFFC00u & ((uint32_t) rows << 10));      /* rowCount 14b */
+        b03 |= (((uint32_t) content >> 11) & 0x3FFu);        /* dataLength hi 10b */
+        b47 |= (((uint32_t) content & 0x7FFu) << 21);        /* dataLength lo 11b */
+        /* compressedLength = 0 */
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+
+        /* DatumStreamBlock_Orig, datumstreamblock.h:73-84 */
+        uint8_t *content_p = blk + 24;
+        int16_t v16;
+        v16 = 0;              memcpy(content_p + 0, &v16, 2);   /* version = Original */
+        v16 = 0;              memcpy(content_p + 2, &v16, 2);   /* flags (no null bitmap) */
+        v16 = (int16_t) rows; memcpy(content_p + 4, &v16, 2);   /* ndatum */
+        v16 = 0;              memcpy(content_p + 6, &v16, 2);   /* encrypted */
+        int32_t v32 = 0;      memcpy(content_p + 8, &v32, 4);   /* nullsz */
+        v32 = sz;             memcpy(content_p + 12, &v32, 4);  /* sz */
+        memcpy(content_p + 16, src + row * (int64_t) width, sz);
+
+        /* checksums: block CRC over [16, blockLen) incl. trailing zero pad,
+         * then header CRC over [0,12)  (cdbappendonlystorageformat.c:160-190) */
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+
+        off += blocklen;
+        row += rows;
+    }
+    return off;
+}
+
+int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
+                        void *out_vals, int64_t cap, int verify_checksums)
+{
+    int64_t off = 0, row = 0;
+    uint8_t *dst = (uint8_t *) out_vals;
+    while (off + 24 <= nbytes)
+    {
+        uint32_t b03, b47;
+        memcpy(&b03, stream + off, 4);
+        memcpy(&b47, stream + off + 4, 4);
+        if (b03 == 0 && b47 == 0) break;         /* zero padding tail */
+        uint32_t kind = (b03 >> 28) & 7;
+        uint32_t hasfrn = (b03 >> 27) & 1;
+        uint32_t rows = (b03 & 0x00FFFC00u) >> 10;
+        uint32_t datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+        uint32_t complen = b47 & 0x1FFFFFu;
+        if (kind != 1 || !hasfrn || complen != 0) return -1;
+        int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
+        if (off + blocklen > nbytes) return -1;
+        if (verify_checksums)
+        {
+            uint32_t bc, hc;
+            memcpy(&bc, stream + off + 8, 4);
+            memcpy(&hc, stream + off + 12, 4);
+            if (hc != orc_crc32c(0xFFFFFFFFu, stream + off, 12)) return -2;
+            if (bc != orc_crc32c(0xFFFFFFFFu, stream + off + 16, blocklen - 16)) return -2;
+        }
+        const uint8_t *content = stream + off + 24;
+        int16_t version, ndatum;
+        int32_t sz;
+        memcpy(&version, content, 2);
+        memcpy(&ndatum, content + 4, 2);
+        memcpy(&sz, content + 12, 4);
+        if (version != 0 || (uint32_t) ndatum != rows || sz != (int32_t) rows * width)
+            return -1;
+        if (row + rows > cap) return -1;
+        memcpy(dst + row * (int64_t) width, content + 16, sz);
+        row += rows;
+        off += blocklen;
+    }
+    return row;
+}
+
+/* ======================================================================
+ * Q3 pipeline — reference executor semantics:
+ *   scan+filter  execScan.c:161-263 (strict qual; data is NOT NULL)
+ *   hash join    nodeHashjoin.c:252-834 (inner, build then probe;
+ *                build-before-outer also satisfies the prefetch_inner
+ *                Motion-deadlock rule nodeHashjoin.c:332-337)
+ *   hash agg     nodeAgg.c:2743,2288 with SUM(float8)=float8pl float.c:769;
+ *                group key (l_orderkey,o_orderdate,o_shippriority) is
+ *                functionally determined by l_orderkey here.
+ * The internal hash-table layout/hash is NOT the reference's (simplehash):
+ * parity-irrelevant per SURVEY §8a (result set identical).
+ * ====================================================================== */
+
+typedef struct {
+    int64_t *keys;               /* EMPTY = INT64_MIN */
+    uint64_t mask;
+} set64;
+
+static uint64_t hmix64(uint64_t x)   /* internal table hash (not parity-relevant) */
+{
+    x ^= x >> 33; x *= 0xFF51AFD7ED558CCDULL;
+    x ^= x >> 33; x *= 0xC4CEB9FE1A85EC53ULL;
+    x ^= x >> 33; return x;
+}
+
+#define EMPTY_KEY INT64_MIN
+
+static void set_init(set64 *s, int64_t want)
+{
+    uint64_t sz = 16;
+    while (sz < (uint64_t) want * 2) sz <<= 1;
+    s->keys = malloc(sizeof(int64_t) * sz);
+    s->mask = sz - 1;
+    for (uint64_t i = 0; i < sz; i++) s->keys[i] = EMPTY_KEY;
+}
+static void set_insert(set64 *s, int64_t k)
+{
+    uint64_t i = hmix64((uint64_t) k) & s->mask;
+    while (s->keys[i] != EMPTY_KEY) { if (s->keys[i] == k) return; i = (i + 1) & s->mask; }
+    s->keys[i] = k;
+}
+static int set_contains(const set64 *s, int64_t k)
+{
+    uint64_t i = hmix64((uint64_t) k) & s->mask;
+    while (s->keys[i] != EMPTY_KEY) { if (s->keys[i] == k) return 1; i = (i + 1) & s->mask; }
+    return 0;
+}
+
+typedef struct {
+    int64_t *key;                /* o_orderkey, EMPTY_KEY empty */
+    int32_t *odate;
+    int32_t *oprio;
+    double  *rev;
+    int64_t *cnt;
+    uint64_t mask;
+} q3tab;
+
+static void tab_init(q3tab *t, int64_t want)
+{
+    uint64_t sz = 16;
+    while (sz < (uint64_t) want * 2) sz <<= 1;
+    t->key = malloc(sizeof(int64_t) * sz);
+    t->odate = malloc(sizeof(int32_t) * sz);
+    t->oprio = malloc(sizeof(int32_t) * sz);
+    t->rev = calloc(sz, sizeof(double));
+    t->cnt = calloc(sz, sizeof(int64_t));
+    t->mask = sz - 1;
+    for (uint64_t i = 0; i < sz; i++) t->key[i] = EMPTY_KEY;
+}
+
+static int q3_group_cmp(const void *a, const void *b)
+{
+    const orc_q3_group *x = a, *y = b;
+    return (x->l_orderkey > y->l_orderkey) - (x->l_orderkey < y->l_orderkey);
+}
+
+int64_t orc_q3(const orc_customer *c, const orc_orders *o,
+               const orc_lineitem *l, int32_t cutoff, orc_q3_group **out)
+{
+    /* 1. customer: filter c_mktsegment='BUILDING' → key set */
+    set64 cust;
+    set_init(&cust, c->n + 16);   /* worst case: every customer qualifies */
+    for (int64_t i = 0; i < c->n; i++)
+        if (c->c_mktsegment[i] == 0)
+            set_insert(&cust, c->c_custkey[i]);
+
+    /* 2. orders: filter o_orderdate < cutoff, semijoin customer, build table */
+    q3tab t;
+    tab_init(&t, o->n + 16);      /* worst case: every order qualifies */
+    for (int64_t i = 0; i < o->n; i++)
+    {
+        if (!(o->o_orderdate[i] < cutoff)) continue;
+        if (!set_contains(&cust, o->o_custkey[i])) continue;
+        int64_t k = o->o_orderkey[i];
+        uint64_t j = hmix64((uint64_t) k) & t.mask;
+        while (t.key[j] != EMPTY_KEY && t.key[j] != k) j = (j + 1) & t.mask;
+        t.key[j] = k;            /* o_orderkey unique → no duplicate entries */
+        t.odate[j] = o->o_orderdate[i];
+        t.oprio[j] = o->o_shippriority[i];
+    }
+
+    /* 3. lineitem: filter l_shipdate > cutoff, probe, aggregate */
+    for (int64_t i = 0; i < l->n; i++)
+    {
+        if (!(l->l_shipdate[i] > cutoff)) continue;
+        int64_t k = l->l_orderkey[i];
+        uint64_t j = hmix64((uint64_t) k) & t.mask;
+        while (t.key[j] != EMPTY_KEY && t.key[j] != k) j = (j + 1) & t.mask;
+        if (t.key[j] == EMPTY_KEY) continue;
+        t.rev[j] += l->l_extendedprice[i] * (1.0 - l->l_discount[i]);
+        t.cnt[j]++;
+    }
+
+    /* 4. extract groups with ≥1 joined lineitem, sorted by l_orderkey */
+    int64_t ng = 0;
+    for (uint64_t j = 0; j <= t.mask; j++)
+        if (t.key[j] != EMPTY_KEY && t.cnt[j] > 0) ng++;
+    orc_q3_group *g = malloc(sizeof(orc_q3_group) * (ng ? ng : 1));
+    int64_t w = 0;
+    for (uint64_t j = 0; j <= t.mask; j++)
+        if (t.key[j] != EMPTY_KEY && t.cnt[j] > 0)
+        {
+            g[w].l_orderkey = t.key[j];
+            g[w].o_orderdate = t.odate[j];
+            g[w].o_shippriority = t.oprio[j];
+            g[w].revenue = t.rev[j];
+            g[w].nitems = t.cnt[j];
+            w++;
+        }
+    qsort(g, ng, sizeof(orc_q3_group), q3_group_cmp);
+
+    free(cust.keys);
+    free(t.key); free(t.odate); free(t.oprio); free(t.rev); free(t.cnt);
+    *out = g;
+    return ng;
+}

@@ -1,0 +1,200 @@
+"""ctypes binding for liboracle.so (TEST INFRASTRUCTURE — see oracle/oracle.h:
+only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg may
+import this module; it is never the product path)."""
+import ctypes
+import os
+
+import numpy as np
+
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+_SO = os.path.join(_ROOT, "oracle", "liboracle.so")
+
+
+class _Customer(ctypes.Structure):
+    _fields_ = [("c_custkey", ctypes.POINTER(ctypes.c_int64)),
+                ("c_mktsegment", ctypes.POINTER(ctypes.c_uint8)),
+                ("n", ctypes.c_int64)]
+
+
+class _Orders(ctypes.Structure):
+    _fields_ = [("o_orderkey", ctypes.POINTER(ctypes.c_int64)),
+                ("o_custkey", ctypes.POINTER(ctypes.c_int64)),
+                ("o_orderdate", ctypes.POINTER(ctypes.c_int32)),
+                ("o_shippriority", ctypes.POINTER(ctypes.c_int32)),
+                ("n", ctypes.c_int64)]
+
+
+class _Lineitem(ctypes.Structure):
+    _fields_ = [("l_orderkey", ctypes.POINTER(ctypes.c_int64)),
+                ("l_extendedprice", ctypes.POINTER(ctypes.c_double)),
+                ("l_discount", ctypes.POINTER(ctypes.c_double)),
+                ("l_shipdate", ctypes.POINTER(ctypes.c_int32)),
+                ("n", ctypes.c_int64)]
+
+
+class _Group(ctypes.Structure):
+    _fields_ = [("l_orderkey", ctypes.c_int64),
+                ("o_orderdate", ctypes.c_int32),
+                ("o_shippriority", ctypes.c_int32),
+                ("revenue", ctypes.c_double),
+                ("nitems", ctypes.c_int64)]
+
+
+def _load():
+    if not os.path.exists(_SO):
+        raise RuntimeError(f"{_SO} missing — run `make -C oracle` first")
+    lib = ctypes.CDLL(_SO)
+    lib.orc_hash_bytes_uint32.restype = ctypes.c_uint32
+    lib.orc_hash_bytes_uint32.argtypes = [ctypes.c_uint32]
+    lib.orc_hashint8.restype = ctypes.c_uint32
+    lib.orc_hashint8.argtypes = [ctypes.c_int64]
+    lib.orc_cdbhash_i64.restype = ctypes.c_uint32
+    lib.orc_cdbhash_i64.argtypes = [ctypes.c_int64]
+    lib.orc_jump_consistent_hash.restype = ctypes.c_int32
+    lib.orc_jump_consistent_hash.argtypes = [ctypes.c_uint64, ctypes.c_int32]
+    lib.orc_route_i64.restype = ctypes.c_int32
+    lib.orc_route_i64.argtypes = [ctypes.c_int64, ctypes.c_int32]
+    lib.orc_route_i64_batch.restype = None
+    lib.orc_route_i64_batch.argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                        ctypes.c_int32, ctypes.c_void_p]
+    lib.orc_crc32c.restype = ctypes.c_uint32
+    lib.orc_crc32c.argtypes = [ctypes.c_uint32, ctypes.c_void_p, ctypes.c_size_t]
+    lib.orc_date_adt.restype = ctypes.c_int32
+    lib.orc_date_adt.argtypes = [ctypes.c_int] * 3
+    lib.orc_splitmix64.restype = ctypes.c_uint64
+    lib.orc_splitmix64.argtypes = [ctypes.c_uint64]
+    lib.orc_mix.restype = ctypes.c_uint64
+    lib.orc_mix.argtypes = [ctypes.c_uint64] * 3
+    lib.orc_ncustomer.restype = ctypes.c_int64
+    lib.orc_ncustomer.argtypes = [ctypes.c_double]
+    lib.orc_norders.restype = ctypes.c_int64
+    lib.orc_norders.argtypes = [ctypes.c_double]
+    for name, st in (("customer", _Customer), ("orders", _Orders), ("lineitem", _Lineitem)):
+        fn = getattr(lib, f"orc_gen_{name}")
+        fn.restype = ctypes.c_int
+        fn.argtypes = [ctypes.c_double, ctypes.c_uint64, ctypes.c_int, ctypes.c_int,
+                       ctypes.POINTER(st)]
+    lib.orc_aocs_rows_per_block.restype = ctypes.c_int32
+    lib.orc_aocs_rows_per_block.argtypes = [ctypes.c_int, ctypes.c_int32]
+    lib.orc_aocs_encoded_size.restype = ctypes.c_int64
+    lib.orc_aocs_encoded_size.argtypes = [ctypes.c_int, ctypes.c_int64, ctypes.c_int32]
+    lib.orc_aocs_encode.restype = ctypes.c_int64
+    lib.orc_aocs_encode.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+                                    ctypes.c_int64, ctypes.c_int32,
+                                    ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_decode.restype = ctypes.c_int64
+    lib.orc_aocs_decode.argtypes = [ctypes.c_void_p, ctypes.c_int64, ctypes.c_int,
+                                    ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.orc_q3.restype = ctypes.c_int64
+    lib.orc_q3.argtypes = [ctypes.POINTER(_Customer), ctypes.POINTER(_Orders),
+                           ctypes.POINTER(_Lineitem), ctypes.c_int32,
+                           ctypes.POINTER(ctypes.POINTER(_Group))]
+    return lib
+
+
+lib = _load()
+
+CUTOFF_19950315 = lib.orc_date_adt(1995, 3, 15)
+
+
+def _np(ptr, n, dtype):
+    return np.ctypeslib.as_array(ptr, shape=(n,)).astype(dtype, copy=True)
+
+
+def gen_customer(sf, seed=42, seg=0, nsegs=1):
+    t = _Customer()
+    assert lib.orc_gen_customer(sf, seed, seg, nsegs, ctypes.byref(t)) == 0
+    out = {"c_custkey": _np(t.c_custkey, t.n, np.int64),
+           "c_mktsegment": _np(t.c_mktsegment, t.n, np.uint8)}
+    lib.orc_free(t.c_custkey); lib.orc_free(t.c_mktsegment)
+    return out
+
+
+def gen_orders(sf, seed=42, seg=0, nsegs=1):
+    t = _Orders()
+    assert lib.orc_gen_orders(sf, seed, seg, nsegs, ctypes.byref(t)) == 0
+    out = {"o_orderkey": _np(t.o_orderkey, t.n, np.int64),
+           "o_custkey": _np(t.o_custkey, t.n, np.int64),
+           "o_orderdate": _np(t.o_orderdate, t.n, np.int32),
+           "o_shippriority": _np(t.o_shippriority, t.n, np.int32)}
+    for f in ("o_orderkey", "o_custkey", "o_orderdate", "o_shippriority"):
+        lib.orc_free(getattr(t, f))
+    return out
+
+
+def gen_lineitem(sf, seed=42, seg=0, nsegs=1):
+    t = _Lineitem()
+    assert lib.orc_gen_lineitem(sf, seed, seg, nsegs, ctypes.byref(t)) == 0
+    out = {"l_orderkey": _np(t.l_orderkey, t.n, np.int64),
+           "l_extendedprice": _np(t.l_extendedprice, t.n, np.float64),
+           "l_discount": _np(t.l_discount, t.n, np.float64),
+           "l_shipdate": _np(t.l_shipdate, t.n, np.int32)}
+    for f in ("l_orderkey", "l_extendedprice", "l_discount", "l_shipdate"):
+        lib.orc_free(getattr(t, f))
+    return out
+
+
+def _as_struct(table, st, fields):
+    t = st()
+    arrs = []
+    for f, ct in fields:
+        a = np.ascontiguousarray(table[f])
+        arrs.append(a)
+        setattr(t, f, a.ctypes.data_as(ctypes.POINTER(ct)))
+    t.n = len(arrs[0])
+    t._keepalive = arrs
+    return t
+
+
+def q3(cust, orders, lineitem, cutoff=None):
+    """Run the oracle Q3 pipeline on numpy column dicts; returns a dict of arrays."""
+    c = _as_struct(cust, _Customer, [("c_custkey", ctypes.c_int64),
+                                     ("c_mktsegment", ctypes.c_uint8)])
+    o = _as_struct(orders, _Orders, [("o_orderkey", ctypes.c_int64),
+                                     ("o_custkey", ctypes.c_int64),
+                                     ("o_orderdate", ctypes.c_int32),
+                                     ("o_shippriority", ctypes.c_int32)])
+    li = _as_struct(lineitem, _Lineitem, [("l_orderkey", ctypes.c_int64),
+                                          ("l_extendedprice", ctypes.c_double),
+                                          ("l_discount", ctypes.c_double),
+                                          ("l_shipdate", ctypes.c_int32)])
+    gp = ctypes.POINTER(_Group)()
+    ng = lib.orc_q3(ctypes.byref(c), ctypes.byref(o), ctypes.byref(li),
+                    CUTOFF_19950315 if cutoff is None else cutoff, ctypes.byref(gp))
+    assert ng >= 0
+    res = {"l_orderkey": np.array([gp[i].l_orderkey for i in range(ng)], np.int64),
+           "o_orderdate": np.array([gp[i].o_orderdate for i in range(ng)], np.int32),
+           "o_shippriority": np.array([gp[i].o_shippriority for i in range(ng)], np.int32),
+           "revenue": np.array([gp[i].revenue for i in range(ng)], np.float64),
+           "nitems": np.array([gp[i].nitems for i in range(ng)], np.int64)}
+    lib.orc_free(gp)
+    return res
+
+
+def route(keys, nsegs):
+    """Vectorised bit-exact Motion routing: jump_consistent_hash(cdbhash(k))."""
+    keys = np.ascontiguousarray(keys, np.int64)
+    out = np.zeros(len(keys), np.int32)
+    lib.orc_route_i64_batch(keys.ctypes.data, len(keys), nsegs, out.ctypes.data)
+    return out
+
+
+def aocs_encode(vals):
+    """Encode a fixed-width NOT NULL numpy column into an AOCS stream (bytes)."""
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    n = len(vals)
+    cap = lib.orc_aocs_encoded_size(width, n, 32768)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode(vals.ctypes.data, width, n, 1, 32768, buf.ctypes.data, cap)
+    assert got == cap, (got, cap)
+    return buf.tobytes()
+
+
+def aocs_decode(stream, width, nrows, dtype, verify=True):
+    buf = np.frombuffer(stream, np.uint8)
+    out = np.zeros(nrows, dtype)
+    got = lib.orc_aocs_decode(buf.ctypes.data, len(buf), width,
+                              out.ctypes.data, nrows, 1 if verify else 0)
+    assert got == nrows, got
+    return out
