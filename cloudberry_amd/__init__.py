@@ -1,0 +1,229 @@
+"""cloudberry_amd — Python plumbing around libgpuexec.so (the C-ABI product
+path; see include/gpuexec.h).  This wrapper exists for tests and the bench
+harness; the library itself has no Python or torch dependency.
+
+The GPU path NEVER falls back to CPU: if the HIP library is missing or no
+device is usable, every entry point raises."""
+import ctypes
+import os
+
+import numpy as np
+
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+_SO = os.path.join(_ROOT, "cloudberry_amd", "libgpuexec.so")
+
+TPCH_CUSTOMER, TPCH_ORDERS, TPCH_LINEITEM = 0, 1, 2
+CUTOFF_19950315 = -1753  # DateADT of 1995-03-15 (validated vs oracle in tests)
+
+_STATUS = {0: "GX_OK", 1: "GX_ERR_HIP", 2: "GX_ERR_RCCL", 3: "GX_ERR_INVALID",
+           4: "GX_ERR_CHECKSUM", 5: "GX_ERR_OOM", 6: "GX_ERR_NOGPU", 7: "GX_ERR_STATE"}
+
+
+class GxError(RuntimeError):
+    def __init__(self, status, detail=""):
+        self.status = status
+        super().__init__(f"{_STATUS.get(status, status)}: {detail}")
+
+
+class _Group(ctypes.Structure):
+    _fields_ = [("l_orderkey", ctypes.c_int64),
+                ("o_orderdate", ctypes.c_int32),
+                ("o_shippriority", ctypes.c_int32),
+                ("revenue", ctypes.c_double),
+                ("nitems", ctypes.c_int64)]
+
+
+class _Stats(ctypes.Structure):
+    _fields_ = [("ms_cust_build", ctypes.c_double),
+                ("ms_orders_build", ctypes.c_double),
+                ("ms_probe_agg", ctypes.c_double),
+                ("ms_extract", ctypes.c_double),
+                ("ms_motion", ctypes.c_double),
+                ("ms_total", ctypes.c_double),
+                ("cust_rows", ctypes.c_int64),
+                ("ord_rows", ctypes.c_int64),
+                ("li_rows", ctypes.c_int64),
+                ("probe_hits", ctypes.c_int64),
+                ("groups", ctypes.c_int64),
+                ("bytes_scanned", ctypes.c_double)]
+
+
+class _ColDesc(ctypes.Structure):
+    _fields_ = [("host_stream", ctypes.c_void_p),
+                ("nbytes", ctypes.c_int64),
+                ("width", ctypes.c_int32),
+                ("nrows", ctypes.c_int64),
+                ("blocksize", ctypes.c_int32)]
+
+
+def _load():
+    if not os.path.exists(_SO):
+        raise RuntimeError(
+            f"HIP extension missing: {_SO}. Build it with "
+            f"`python -c \"import __graft_entry__; __graft_entry__.build()\"` — "
+            f"the GPU executor has no CPU fallback.")
+    lib = ctypes.CDLL(_SO)
+    lib.gx_last_error.restype = ctypes.c_char_p
+    lib.gx_last_error.argtypes = [ctypes.c_void_p]
+    lib.gx_version.restype = ctypes.c_char_p
+    lib.gx_init.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                            ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_shutdown.argtypes = [ctypes.c_void_p]
+    lib.gx_comm_unique_id.argtypes = [ctypes.c_char_p]
+    lib.gx_comm_init.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+    lib.gx_table_bind.argtypes = [ctypes.c_void_p, ctypes.POINTER(_ColDesc),
+                                  ctypes.c_int, ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_table_free.argtypes = [ctypes.c_void_p]
+    lib.gx_table_nrows.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_table_logical_bytes.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_double)]
+    lib.gx_tpch_gen.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_double,
+                                ctypes.c_uint64, ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_decode_column.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                                     ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.gx_partition.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+                                 ctypes.c_int32, ctypes.c_void_p]
+    lib.gx_q3_prepare.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int32,
+                                  ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_q3_run.argtypes = [ctypes.c_void_p]
+    lib.gx_q3_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
+    lib.gx_q3_result.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.POINTER(_Group)),
+                                 ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_q3_free.argtypes = [ctypes.c_void_p]
+    lib.gx_free.argtypes = [ctypes.c_void_p]
+    return lib
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _load()
+    return _lib
+
+
+class Context:
+    def __init__(self, device=0, seg=0, nsegs=1):
+        self._lib = lib()
+        self._h = ctypes.c_void_p()
+        self._chk(self._lib.gx_init(device, seg, nsegs, ctypes.byref(self._h)), None)
+        self.seg, self.nsegs = seg, nsegs
+
+    def _chk(self, st, h="self"):
+        if st != 0:
+            handle = self._h if h == "self" else h
+            msg = self._lib.gx_last_error(handle)
+            raise GxError(st, (msg or b"").decode())
+
+    def comm_unique_id(self):
+        buf = ctypes.create_string_buffer(128)
+        self._chk(self._lib.gx_comm_unique_id(buf), None)
+        return buf.raw
+
+    def comm_init(self, uid: bytes):
+        assert len(uid) == 128
+        self._chk(self._lib.gx_comm_init(self._h, uid))
+
+    def tpch_gen(self, which, sf, seed=42):
+        t = ctypes.c_void_p()
+        self._chk(self._lib.gx_tpch_gen(self._h, which, sf, seed, ctypes.byref(t)))
+        return Table(self, t)
+
+    def bind(self, streams):
+        """streams: list of (bytes, width, nrows) AOCS column streams."""
+        descs = (_ColDesc * len(streams))()
+        keep = []
+        for i, (data, width, nrows) in enumerate(streams):
+            arr = np.frombuffer(data, np.uint8)
+            keep.append(arr)
+            descs[i].host_stream = arr.ctypes.data
+            descs[i].nbytes = len(arr)
+            descs[i].width = width
+            descs[i].nrows = nrows
+            descs[i].blocksize = 32768
+        t = ctypes.c_void_p()
+        self._chk(self._lib.gx_table_bind(self._h, descs, len(streams), ctypes.byref(t)))
+        return Table(self, t)
+
+    def partition(self, keys, nsegs):
+        keys = np.ascontiguousarray(keys, np.int64)
+        out = np.zeros(len(keys), np.int32)
+        self._chk(self._lib.gx_partition(self._h, keys.ctypes.data, len(keys),
+                                         nsegs, out.ctypes.data))
+        return out
+
+    def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315):
+        q = ctypes.c_void_p()
+        self._chk(self._lib.gx_q3_prepare(self._h, cust._t, orders._t, lineitem._t,
+                                          cutoff, ctypes.byref(q)))
+        return Q3(self, q)
+
+    def close(self):
+        if self._h:
+            self._lib.gx_shutdown(self._h)
+            self._h = None
+
+
+class Table:
+    def __init__(self, ctx, t):
+        self.ctx = ctx
+        self._t = t
+
+    @property
+    def nrows(self):
+        n = ctypes.c_int64()
+        self.ctx._chk(self.ctx._lib.gx_table_nrows(self._t, ctypes.byref(n)))
+        return n.value
+
+    @property
+    def logical_bytes(self):
+        b = ctypes.c_double()
+        self.ctx._chk(self.ctx._lib.gx_table_logical_bytes(self._t, ctypes.byref(b)))
+        return b.value
+
+    def decode_column(self, col, dtype, verify=True):
+        n = self.nrows
+        out = np.zeros(n, dtype)
+        self.ctx._chk(self.ctx._lib.gx_decode_column(
+            self.ctx._h, self._t, col, out.ctypes.data, n, 1 if verify else 0))
+        return out
+
+    def free(self):
+        if self._t:
+            self.ctx._lib.gx_table_free(self._t)
+            self._t = None
+
+
+class Q3:
+    def __init__(self, ctx, q):
+        self.ctx = ctx
+        self._q = q
+
+    def run(self):
+        self.ctx._chk(self.ctx._lib.gx_q3_run(self._q))
+        return self
+
+    def stats(self):
+        s = _Stats()
+        self.ctx._chk(self.ctx._lib.gx_q3_stats_get(self._q, ctypes.byref(s)))
+        return {f: getattr(s, f) for f, _ in s._fields_}
+
+    def result(self):
+        gp = ctypes.POINTER(_Group)()
+        n = ctypes.c_int64()
+        self.ctx._chk(self.ctx._lib.gx_q3_result(self._q, ctypes.byref(gp),
+                                                 ctypes.byref(n)))
+        n = n.value
+        res = {"l_orderkey": np.array([gp[i].l_orderkey for i in range(n)], np.int64),
+               "o_orderdate": np.array([gp[i].o_orderdate for i in range(n)], np.int32),
+               "o_shippriority": np.array([gp[i].o_shippriority for i in range(n)], np.int32),
+               "revenue": np.array([gp[i].revenue for i in range(n)], np.float64),
+               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
+        self.ctx._lib.gx_free(gp)
+        return res
+
+    def free(self):
+        if self._q:
+            self.ctx._lib.gx_q3_free(self._q)
+            self._q = None

@@ -1,0 +1,1266 @@
+/*
+ * gx_gpuexec.hip — MI355X-native (gfx950) executor for Cloudberry's
+ * segment-local scan→hash-join→hash-agg pipeline + hash-redistribute Motion.
+ *
+ * Implements include/gpuexec.h.  This is the PRODUCT path: hand-written HIP
+ * kernels over AOCS column streams resident in HBM, RCCL over xGMI for the
+ * Motion exchange.  The oracle (oracle/) is never referenced here.
+ *
+ * Reference semantics being replaced, per stage (paths under /root/reference):
+ *   scan+filter   executor/nodeSeqscan.c:57, execScan.c:161-263,
+ *                 access/aocs/aocsam.c:1131-1259 (per-row datum fetch → here:
+ *                 O(1) block addressing + coalesced loads, headers skipped)
+ *   hash join     executor/nodeHash.c:1886,2098-2251, nodeHashjoin.c:252-834
+ *                 (chain buckets → here: SoA open addressing, build-then-probe;
+ *                 layout/internal hash parity-irrelevant)
+ *   hash agg      executor/nodeAgg.c:836,2288,2743 + float.c:769 float8pl
+ *                 (per-row transition → here: f64 atomic add per group slot)
+ *   Motion        executor/nodeMotion.c:1088,1181 + cdb/cdbhash.c:189-285,
+ *                 530-541 (bit-exact routing), cdb/motion/* + contrib/
+ *                 interconnect UDP (→ RCCL grouped send/recv, columnar batches)
+ */
+#include "gx_internal.h"
+#include "../../include/gpuexec.h"
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+/* ================= error plumbing ================= */
+
+static char g_global_err[512] = "";
+
+struct gx_ctx {
+    int device = -1;
+    int seg = 0;
+    int nsegs = 1;
+    hipStream_t stream = nullptr;
+    ncclComm_t comm = nullptr;
+    char err[512] = "";
+};
+
+static void set_err(gx_ctx *ctx, const char *fmt, const char *detail)
+{
+    char *dst = ctx ? ctx->err : g_global_err;
+    snprintf(dst, 512, fmt, detail);
+}
+
+#define HIP_CHK(ctx, call)                                                    \
+    do {                                                                      \
+        hipError_t _e = (call);                                               \
+        if (_e != hipSuccess) {                                               \
+            set_err(ctx, "HIP error: %s (" #call ")", hipGetErrorString(_e)); \
+            return (_e == hipErrorOutOfMemory) ? GX_ERR_OOM : GX_ERR_HIP;     \
+        }                                                                     \
+    } while (0)
+
+#define RCCL_CHK(ctx, call)                                                   \
+    do {                                                                      \
+        ncclResult_t _r = (call);                                             \
+        if (_r != ncclSuccess) {                                              \
+            set_err(ctx, "RCCL error: %s (" #call ")", ncclGetErrorString(_r)); \
+            return GX_ERR_RCCL;                                               \
+        }                                                                     \
+    } while (0)
+
+extern "C" const char *gx_last_error(const gx_ctx *ctx)
+{
+    return ctx ? ctx->err : g_global_err;
+}
+
+extern "C" const char *gx_version(void) { return "gpuexec 0.1 (gfx950)"; }
+
+/* ================= small device helpers ================= */
+
+static constexpr int TPB = 256;          /* threads per block (4 waves) */
+static constexpr int GRID = 2048;        /* grid-stride grid (fills 256 CUs) */
+
+static inline int64_t pow2_at_least(int64_t want)
+{
+    int64_t sz = 1024;
+    while (sz < want) sz <<= 1;
+    return sz;
+}
+
+/* ================= generation: count / scan / emit ================= */
+/* Deterministic sharded generation: each thread owns a contiguous chunk of
+ * GLOBAL units (rows for customer/orders, orders for lineitem), counts the
+ * rows it keeps for this segment, host exclusive-scans the per-thread
+ * counts (stable order = global order, matching the oracle), then emit. */
+
+static constexpr int64_t GEN_CHUNK = 512;
+
+__global__ void k_count_cust(uint64_t seed, int64_t n, int seg, int nsegs, uint32_t *counts)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, n);
+    if (lo >= n) { counts[t] = 0; return; }
+    uint32_t c = 0;
+    for (int64_t i = lo; i < hi; i++)
+        if (nsegs == 1 || gx_route_i64(i + 1, nsegs) == seg) c++;
+    counts[t] = c;
+}
+
+__global__ void k_emit_cust(uint64_t seed, int64_t n, int seg, int nsegs,
+                            const uint64_t *offs, int64_t *custkey, uint8_t *mkt)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, n);
+    if (lo >= n) return;
+    uint64_t w = offs[t];
+    for (int64_t i = lo; i < hi; i++)
+        if (nsegs == 1 || gx_route_i64(i + 1, nsegs) == seg)
+        {
+            custkey[w] = i + 1;
+            mkt[w] = gx_gen_mktsegment(seed, i);
+            w++;
+        }
+}
+
+__global__ void k_count_ord(uint64_t seed, int64_t n, int seg, int nsegs, uint32_t *counts)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, n);
+    if (lo >= n) { counts[t] = 0; return; }
+    uint32_t c = 0;
+    for (int64_t i = lo; i < hi; i++)
+        if (nsegs == 1 || gx_route_i64(i + 1, nsegs) == seg) c++;
+    counts[t] = c;
+}
+
+__global__ void k_emit_ord(uint64_t seed, int64_t n, int64_t ncust, int seg, int nsegs,
+                           const uint64_t *offs, int64_t *okey, int64_t *ocust,
+                           int32_t *odate, int32_t *oprio)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, n);
+    if (lo >= n) return;
+    uint64_t w = offs[t];
+    for (int64_t i = lo; i < hi; i++)
+        if (nsegs == 1 || gx_route_i64(i + 1, nsegs) == seg)
+        {
+            okey[w] = i + 1;
+            ocust[w] = gx_gen_ocustkey(seed, i, ncust);
+            odate[w] = gx_gen_odate(seed, i);
+            oprio[w] = gx_gen_oprio(seed, i);
+            w++;
+        }
+}
+
+__global__ void k_count_li(uint64_t seed, int64_t nord, int seg, int nsegs, uint32_t *counts)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, nord);
+    if (lo >= nord) { counts[t] = 0; return; }
+    uint32_t c = 0;
+    for (int64_t o = lo + 1; o <= hi; o++)
+        if (nsegs == 1 || gx_route_i64(o, nsegs) == seg)
+            c += gx_gen_nlines(seed, o);
+    counts[t] = c;
+}
+
+__global__ void k_emit_li(uint64_t seed, int64_t nord, int seg, int nsegs,
+                          const uint64_t *offs, int64_t *lkey, double *price,
+                          double *disc, int32_t *ship)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, nord);
+    if (lo >= nord) return;
+    uint64_t w = offs[t];
+    for (int64_t o = lo + 1; o <= hi; o++)
+    {
+        if (!(nsegs == 1 || gx_route_i64(o, nsegs) == seg)) continue;
+        int32_t nl = gx_gen_nlines(seed, o);
+        for (int32_t j = 0; j < nl; j++)
+        {
+            lkey[w] = o;
+            price[w] = gx_gen_price(seed, o, j);
+            disc[w] = gx_gen_discount(seed, o, j);
+            ship[w] = gx_gen_shipdate(seed, o, j);
+            w++;
+        }
+    }
+}
+
+/* ================= AOCS encode / decode ================= */
+
+/* Writes headers + datums for one AO block per workgroup (CRCs in a second
+ * pass).  The stream buffer must be zeroed first (pad bytes stay 0). */
+template <typename T>
+__global__ void k_encode(const T *vals, int64_t nrows, int32_t rpb,
+                         int64_t full_len, int64_t nblocks, uint8_t *stream)
+{
+    for (int64_t b = blockIdx.x; b < nblocks; b += gridDim.x)
+    {
+        int64_t base = b * (int64_t) rpb;
+        int32_t rows = (int32_t) min((int64_t) rpb, nrows - base);
+        uint8_t *blk = stream + b * full_len;
+        T *datum = (T *) (blk + GX_AOCS_DATUM_OFF);
+        for (int32_t i = threadIdx.x; i < rows; i += blockDim.x)
+            datum[i] = vals[base + i];
+        if (threadIdx.x == 0)
+        {
+            int32_t sz = rows * (int32_t) sizeof(T);
+            int32_t content = 16 + sz;
+            /* AOSmallContentHeader (cdbappendonlystorage_int.h:150-170) */
+            uint32_t b03 = (1u << 28) | (1u << 27) | (1u << 24) |
+                           (0x00FFFC00u & ((uint32_t) rows << 10)) |
+                           (((uint32_t) content >> 11) & 0x3FFu);
+            uint32_t b47 = ((uint32_t) content & 0x7FFu) << 21;
+            ((uint32_t *) blk)[0] = b03;
+            ((uint32_t *) blk)[1] = b47;
+            ((int64_t *) blk)[2] = base + 1;     /* firstRowNum */
+            /* DatumStreamBlock_Orig (datumstreamblock.h:73-84) */
+            uint8_t *c = blk + 24;
+            ((int16_t *) c)[0] = 0;               /* version Original */
+            ((int16_t *) c)[1] = 0;               /* flags */
+            ((int16_t *) c)[2] = (int16_t) rows;  /* ndatum */
+            ((int16_t *) c)[3] = 0;               /* encrypted */
+            ((int32_t *) c)[2] = 0;               /* nullsz */
+            ((int32_t *) c)[3] = sz;              /* sz */
+        }
+    }
+}
+
+/* CRC32C (pg COMP_CRC32C state, no final xor — cdbappendonlystorageformat.c:
+ * 41-47).  One thread per AO block; table built in LDS per workgroup. */
+__device__ __forceinline__ uint32_t d_crc32c(const uint32_t *tab, uint32_t crc,
+                                             const uint8_t *p, int64_t len)
+{
+    /* bulk 8 bytes per iteration from a register (per-thread stride-1 stays
+     * in L1 across iterations) */
+    while (len >= 8 && ((uintptr_t) p & 7)) { crc = tab[(crc ^ *p++) & 0xFF] ^ (crc >> 8); len--; }
+    while (len >= 8)
+    {
+        uint64_t v = *(const uint64_t *) p;
+        for (int k = 0; k < 8; k++)
+        {
+            crc = tab[(crc ^ (uint32_t) v) & 0xFF] ^ (crc >> 8);
+            v >>= 8;
+        }
+        p += 8; len -= 8;
+    }
+    while (len--) crc = tab[(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+    return crc;
+}
+
+__device__ void d_crc_table_init(uint32_t *tab)
+{
+    for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++)
+            c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+        tab[i] = c;
+    }
+    __syncthreads();
+}
+
+__global__ void k_crc_fill(uint8_t *stream, int64_t nblocks, int64_t full_len,
+                           int64_t nrows, int32_t rpb, int32_t width)
+{
+    __shared__ uint32_t tab[256];
+    d_crc_table_init(tab);
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        int64_t base = b * (int64_t) rpb;
+        int64_t rows = min((int64_t) rpb, nrows - base);
+        int64_t blen = gx_aocs_block_len(width, rows);
+        uint8_t *blk = stream + b * full_len;
+        ((uint32_t *) blk)[2] = d_crc32c(tab, 0xFFFFFFFFu, blk + 16, blen - 16);
+        ((uint32_t *) blk)[3] = d_crc32c(tab, 0xFFFFFFFFu, blk, 12);
+    }
+}
+
+/* decode: one workgroup per AO block → flat values; header sanity-checked */
+template <typename T>
+__global__ void k_decode(const uint8_t *stream, int64_t nblocks, int64_t full_len,
+                         int64_t nrows, int32_t rpb, T *out, int *err)
+{
+    for (int64_t b = blockIdx.x; b < nblocks; b += gridDim.x)
+    {
+        const uint8_t *blk = stream + b * full_len;
+        uint32_t b03 = ((const uint32_t *) blk)[0];
+        uint32_t rows = (b03 & 0x00FFFC00u) >> 10;
+        if (threadIdx.x == 0)
+        {
+            uint32_t b47 = ((const uint32_t *) blk)[1];
+            uint32_t datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+            int64_t frn = ((const int64_t *) blk)[2];
+            int16_t version = ((const int16_t *) (blk + 24))[0];
+            int16_t ndatum = ((const int16_t *) (blk + 24))[2];
+            int32_t sz = ((const int32_t *) (blk + 24))[3];
+            if (((b03 >> 28) & 7) != 1 || !((b03 >> 27) & 1) ||
+                (b47 & 0x1FFFFFu) != 0 || version != 0 ||
+                (uint32_t) ndatum != rows || sz != (int32_t) rows * (int32_t) sizeof(T) ||
+                datalen != 16 + (uint32_t) sz || frn != b * (int64_t) rpb + 1)
+                atomicOr(err, 1);
+        }
+        const T *datum = (const T *) (blk + GX_AOCS_DATUM_OFF);
+        int64_t base = b * (int64_t) rpb;
+        int64_t rows_i = min((int64_t) rows, nrows - base);
+        for (int64_t i = threadIdx.x; i < rows_i; i += blockDim.x)
+            out[base + i] = datum[i];
+    }
+}
+
+__global__ void k_verify_crc(const uint8_t *stream, int64_t nblocks, int64_t full_len,
+                             int64_t nrows, int32_t rpb, int32_t width, int *err)
+{
+    __shared__ uint32_t tab[256];
+    d_crc_table_init(tab);
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        int64_t base = b * (int64_t) rpb;
+        int64_t rows = min((int64_t) rpb, nrows - base);
+        int64_t blen = gx_aocs_block_len(width, rows);
+        const uint8_t *blk = stream + b * full_len;
+        if (((const uint32_t *) blk)[3] != d_crc32c(tab, 0xFFFFFFFFu, blk, 12) ||
+            ((const uint32_t *) blk)[2] != d_crc32c(tab, 0xFFFFFFFFu, blk + 16, blen - 16))
+            atomicOr(err, 2);
+    }
+}
+
+/* ================= Motion routing ================= */
+
+__global__ void k_route(const int64_t *keys, int64_t n, int32_t nsegs, int32_t *out)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
+        out[i] = gx_route_i64(keys[i], nsegs);
+}
+
+/* ================= Q3 kernels ================= */
+
+/* customer: count BUILDING rows (for set sizing) */
+__global__ void k_cust_count(const uint8_t *mkt_s, gx_colmeta mkt_m,
+                             unsigned long long *count)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long local = 0;
+    for (; i < mkt_m.nrows; i += stride)
+        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) == 0) local++;
+    if (local) atomicAdd(count, local);
+}
+
+/* customer: filter mktsegment=BUILDING, insert c_custkey into open set.
+ * Replaces the build side of the cust⋈orders join (nodeHash.c:1886). */
+__global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
+                             const uint8_t *mkt_s, gx_colmeta mkt_m,
+                             unsigned long long *set, uint64_t mask)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < key_m.nrows; i += stride)
+    {
+        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) != 0) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
+        uint64_t slot = gx_hmix64(k) & mask;
+        while (true)
+        {
+            unsigned long long prev = atomicCAS(&set[slot], 0ULL, (unsigned long long) k);
+            if (prev == 0ULL || prev == (unsigned long long) k) break;
+            slot = (slot + 1) & mask;
+        }
+    }
+}
+
+__device__ __forceinline__ bool d_set_contains(const unsigned long long *set,
+                                               uint64_t mask, uint64_t k)
+{
+    uint64_t slot = gx_hmix64(k) & mask;
+    while (true)
+    {
+        unsigned long long v = set[slot];
+        if (v == 0ULL) return false;
+        if (v == (unsigned long long) k) return true;
+        slot = (slot + 1) & mask;
+    }
+}
+
+/* orders local path: count qualifying rows (date filter + customer semijoin) */
+__global__ void k_orders_count(const uint8_t *od_s, gx_colmeta od_m,
+                               const uint8_t *oc_s, gx_colmeta oc_m,
+                               int32_t cutoff,
+                               const unsigned long long *cset, uint64_t cmask,
+                               unsigned long long *count)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long local = 0;
+    for (; i < od_m.nrows; i += stride)
+    {
+        if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
+        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        local++;
+    }
+    if (local) atomicAdd(count, local);
+}
+
+/* orders local path: build the join/agg table keyed by o_orderkey.
+ * (ExecHashTableInsert nodeHash.c:1886; o_orderkey unique → 1 entry/key;
+ *  payload doubles as the agg group state, nodeAgg.c group = join row) */
+__global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
+                               const uint8_t *oc_s, gx_colmeta oc_m,
+                               const uint8_t *od_s, gx_colmeta od_m,
+                               const uint8_t *op_s, gx_colmeta op_m,
+                               int32_t cutoff,
+                               const unsigned long long *cset, uint64_t cmask,
+                               unsigned long long *tkey, int32_t *tdate,
+                               int32_t *tprio, uint64_t tmask)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < ok_m.nrows; i += stride)
+    {
+        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
+        if (!(od < cutoff)) continue;
+        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
+        uint64_t slot = gx_hmix64(k) & tmask;
+        while (true)
+        {
+            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
+            if (prev == 0ULL)
+            {
+                tdate[slot] = od;
+                tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
+                break;
+            }
+            if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
+            slot = (slot + 1) & tmask;
+        }
+    }
+}
+
+/* lineitem probe + aggregate — THE dominant kernel.
+ * scan (aocsam.c:1131 semantics) + probe (nodeHashjoin.c:553-652) +
+ * SUM transition (nodeAgg.c:836 + float.c:769) fused; build slots ARE the
+ * agg groups (group key functionally determined by l_orderkey). */
+__global__ void k_li_probe_agg(const uint8_t *lk_s, gx_colmeta lk_m,
+                               const uint8_t *pr_s, gx_colmeta pr_m,
+                               const uint8_t *di_s, gx_colmeta di_m,
+                               const uint8_t *sh_s, gx_colmeta sh_m,
+                               int32_t cutoff,
+                               const unsigned long long *tkey, int32_t, /*unused*/
+                               double *trev, unsigned long long *tcnt,
+                               uint64_t tmask,
+                               unsigned long long *hits)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long local_hits = 0;
+    for (; i < lk_m.nrows; i += stride)
+    {
+        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+        uint64_t slot = gx_hmix64(k) & tmask;
+        bool found = false;
+        while (true)
+        {
+            unsigned long long v = tkey[slot];
+            if (v == 0ULL) break;
+            if (v == (unsigned long long) k) { found = true; break; }
+            slot = (slot + 1) & tmask;
+        }
+        if (!found) continue;
+        double price = gx_col_get<double>(pr_s, pr_m, i);
+        double disc = gx_col_get<double>(di_s, di_m, i);
+        atomicAdd(&trev[slot], price * (1.0 - disc));
+        atomicAdd(&tcnt[slot], 1ULL);
+        local_hits++;
+    }
+    if (local_hits) atomicAdd(hits, local_hits);
+}
+
+/* extract groups with ≥1 matched lineitem into SoA result arrays */
+__global__ void k_extract(const unsigned long long *tkey, const int32_t *tdate,
+                          const int32_t *tprio, const double *trev,
+                          const unsigned long long *tcnt, uint64_t tslots,
+                          int64_t *okey, int32_t *odate, int32_t *oprio,
+                          double *rev, int64_t *cnt, unsigned long long *cursor)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < (int64_t) tslots; i += stride)
+    {
+        if (tkey[i] == 0ULL || tcnt[i] == 0ULL) continue;
+        unsigned long long w = atomicAdd(cursor, 1ULL);
+        okey[w] = (int64_t) tkey[i];
+        odate[w] = tdate[i];
+        oprio[w] = tprio[i];
+        rev[w] = trev[i];
+        cnt[w] = (int64_t) tcnt[i];
+    }
+}
+
+/* ----- Motion path (nsegs>1) ----- */
+
+/* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
+__global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
+                              const uint8_t *oc_s, gx_colmeta oc_m,
+                              int32_t cutoff, int nsegs,
+                              unsigned long long *hist)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < od_m.nrows; i += stride)
+    {
+        if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
+        int32_t d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
+        atomicAdd(&hist[d], 1ULL);
+    }
+}
+
+__global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
+                              const uint8_t *oc_s, gx_colmeta oc_m,
+                              const uint8_t *od_s, gx_colmeta od_m,
+                              const uint8_t *op_s, gx_colmeta op_m,
+                              int32_t cutoff, int nsegs,
+                              unsigned long long *cursors, /* pre-set to region starts */
+                              gx_ord_row *out)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < od_m.nrows; i += stride)
+    {
+        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
+        if (!(od < cutoff)) continue;
+        int64_t oc = gx_col_get<int64_t>(oc_s, oc_m, i);
+        int32_t d = gx_route_i64(oc, nsegs);
+        unsigned long long w = atomicAdd(&cursors[d], 1ULL);
+        out[w].okey = gx_col_get<int64_t>(ok_s, ok_m, i);
+        out[w].ocust = oc;
+        out[w].odate = od;
+        out[w].oprio = gx_col_get<int32_t>(op_s, op_m, i);
+    }
+}
+
+/* received orders rows: probe local customer set, histogram by route(okey) */
+__global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
+                            const unsigned long long *cset, uint64_t cmask,
+                            int nsegs, unsigned long long *hist)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
+    {
+        if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
+        atomicAdd(&hist[gx_route_i64(rows[i].okey, nsegs)], 1ULL);
+    }
+}
+
+__global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
+                            const unsigned long long *cset, uint64_t cmask,
+                            int nsegs, unsigned long long *cursors,
+                            gx_qual_row *out)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
+    {
+        if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
+        int32_t d = gx_route_i64(rows[i].okey, nsegs);
+        unsigned long long w = atomicAdd(&cursors[d], 1ULL);
+        out[w].okey = rows[i].okey;
+        out[w].odate = rows[i].odate;
+        out[w].oprio = rows[i].oprio;
+    }
+}
+
+/* received qualifying orders → build the join/agg table */
+__global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
+                                  unsigned long long *tkey, int32_t *tdate,
+                                  int32_t *tprio, uint64_t tmask)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
+    {
+        uint64_t k = (uint64_t) rows[i].okey;
+        uint64_t slot = gx_hmix64(k) & tmask;
+        while (true)
+        {
+            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
+            if (prev == 0ULL)
+            {
+                tdate[slot] = rows[i].odate;
+                tprio[slot] = rows[i].oprio;
+                break;
+            }
+            if (prev == (unsigned long long) k) break;
+            slot = (slot + 1) & tmask;
+        }
+    }
+}
+
+/* ================= host-side structures ================= */
+
+struct gx_col {
+    uint8_t *dstream = nullptr;  /* device AOCS stream */
+    gx_colmeta m{};
+};
+
+struct gx_table {
+    gx_ctx *ctx = nullptr;
+    std::vector<gx_col> cols;
+    int64_t nrows = 0;
+};
+
+struct gx_q3 {
+    gx_ctx *ctx = nullptr;
+    gx_table *cust = nullptr, *ord = nullptr, *li = nullptr;
+    int32_t cutoff = 0;
+    /* run state (device) */
+    unsigned long long *cset = nullptr;
+    uint64_t cmask = 0;
+    unsigned long long *tkey = nullptr;
+    int32_t *tdate = nullptr, *tprio = nullptr;
+    double *trev = nullptr;
+    unsigned long long *tcnt = nullptr;
+    uint64_t tmask = 0;
+    /* result (device SoA) */
+    int64_t *r_okey = nullptr;
+    int32_t *r_odate = nullptr, *r_oprio = nullptr;
+    double *r_rev = nullptr;
+    int64_t *r_cnt = nullptr;
+    int64_t ngroups = 0;
+    int64_t qual_orders = 0;
+    gx_q3_stats stats{};
+    bool ran = false;
+};
+
+/* ================= lifecycle ================= */
+
+extern "C" gx_status gx_init(int device_id, int seg_id, int nsegs, gx_ctx **out)
+{
+    int ndev = 0;
+    hipError_t e = hipGetDeviceCount(&ndev);
+    if (e != hipSuccess || ndev == 0)
+    {
+        snprintf(g_global_err, sizeof g_global_err,
+                 "no usable HIP device (%s) — the GPU executor does not fall back to CPU",
+                 hipGetErrorString(e));
+        return GX_ERR_NOGPU;
+    }
+    gx_ctx *ctx = new gx_ctx();
+    ctx->device = device_id;
+    ctx->seg = seg_id;
+    ctx->nsegs = nsegs;
+    hipError_t e2 = hipSetDevice(device_id);
+    if (e2 != hipSuccess) { set_err(nullptr, "hipSetDevice: %s", hipGetErrorString(e2)); delete ctx; return GX_ERR_HIP; }
+    e2 = hipStreamCreate(&ctx->stream);
+    if (e2 != hipSuccess) { set_err(nullptr, "hipStreamCreate: %s", hipGetErrorString(e2)); delete ctx; return GX_ERR_HIP; }
+    *out = ctx;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_shutdown(gx_ctx *ctx)
+{
+    if (!ctx) return GX_OK;
+    if (ctx->comm) ncclCommDestroy(ctx->comm);
+    if (ctx->stream) hipStreamDestroy(ctx->stream);
+    delete ctx;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_comm_unique_id(unsigned char uid[GX_UNIQUE_ID_BYTES])
+{
+    static_assert(sizeof(ncclUniqueId) == GX_UNIQUE_ID_BYTES, "uid size");
+    ncclUniqueId id;
+    if (ncclGetUniqueId(&id) != ncclSuccess) return GX_ERR_RCCL;
+    memcpy(uid, &id, GX_UNIQUE_ID_BYTES);
+    return GX_OK;
+}
+
+extern "C" gx_status gx_comm_init(gx_ctx *ctx, const unsigned char uid[GX_UNIQUE_ID_BYTES])
+{
+    if (!ctx) return GX_ERR_INVALID;
+    ncclUniqueId id;
+    memcpy(&id, uid, GX_UNIQUE_ID_BYTES);
+    RCCL_CHK(ctx, ncclCommInitRank(&ctx->comm, ctx->nsegs, id, ctx->seg));
+    return GX_OK;
+}
+
+/* ================= tables ================= */
+
+static gx_status encode_column_device(gx_ctx *ctx, const void *dvals, int width,
+                                      int64_t nrows, gx_col *col)
+{
+    int32_t rpb = gx_aocs_rows_per_block(width, 32768);
+    int64_t nblocks = (nrows + rpb - 1) / rpb;
+    int64_t full_len = gx_aocs_block_len(width, rpb);
+    int64_t last_rows = nrows - (nblocks - 1) * (int64_t) rpb;
+    int64_t bytes = (nblocks - 1) * full_len + gx_aocs_block_len(width, last_rows);
+    uint8_t *stream = nullptr;
+    HIP_CHK(ctx, hipMalloc(&stream, bytes));
+    HIP_CHK(ctx, hipMemsetAsync(stream, 0, bytes, ctx->stream));
+    int grid = (int) std::min<int64_t>(nblocks, 65535);
+    if (width == 8)
+        hipLaunchKernelGGL(k_encode<int64_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           (const int64_t *) dvals, nrows, rpb, full_len, nblocks, stream);
+    else if (width == 4)
+        hipLaunchKernelGGL(k_encode<int32_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           (const int32_t *) dvals, nrows, rpb, full_len, nblocks, stream);
+    else
+        hipLaunchKernelGGL(k_encode<int8_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           (const int8_t *) dvals, nrows, rpb, full_len, nblocks, stream);
+    hipLaunchKernelGGL(k_crc_fill, dim3(GRID), dim3(64), 0, ctx->stream,
+                       stream, nblocks, full_len, nrows, rpb, width);
+    HIP_CHK(ctx, hipGetLastError());
+    col->dstream = stream;
+    col->m.width = width;
+    col->m.rpb = rpb;
+    col->m.nrows = nrows;
+    col->m.full_block_len = full_len;
+    col->m.nbytes = bytes;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncols,
+                                   gx_table **out)
+{
+    if (!ctx || !cols || ncols <= 0) return GX_ERR_INVALID;
+    gx_table *t = new gx_table();
+    t->ctx = ctx;
+    t->nrows = cols[0].nrows;
+    for (int c = 0; c < ncols; c++)
+    {
+        gx_col col;
+        col.m.width = cols[c].width;
+        col.m.rpb = gx_aocs_rows_per_block(cols[c].width, cols[c].blocksize);
+        col.m.nrows = cols[c].nrows;
+        col.m.full_block_len = gx_aocs_block_len(cols[c].width, col.m.rpb);
+        col.m.nbytes = cols[c].nbytes;
+        hipError_t e = hipMalloc(&col.dstream, cols[c].nbytes);
+        if (e != hipSuccess) { set_err(ctx, "hipMalloc: %s", hipGetErrorString(e)); delete t; return GX_ERR_OOM; }
+        e = hipMemcpyAsync(col.dstream, cols[c].host_stream, cols[c].nbytes,
+                           hipMemcpyHostToDevice, ctx->stream);
+        if (e != hipSuccess) { set_err(ctx, "hipMemcpy: %s", hipGetErrorString(e)); delete t; return GX_ERR_HIP; }
+        t->cols.push_back(col);
+    }
+    hipStreamSynchronize(ctx->stream);
+    *out = t;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_table_free(gx_table *t)
+{
+    if (!t) return GX_OK;
+    for (auto &c : t->cols)
+        if (c.dstream) hipFree(c.dstream);
+    delete t;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_table_nrows(const gx_table *t, int64_t *out)
+{
+    if (!t) return GX_ERR_INVALID;
+    *out = t->nrows;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_table_logical_bytes(const gx_table *t, double *out)
+{
+    if (!t) return GX_ERR_INVALID;
+    double b = 0;
+    for (auto &c : t->cols)
+        b += (double) c.m.nrows * c.m.width;
+    *out = b;
+    return GX_OK;
+}
+
+/* count/scan/emit helper */
+static gx_status scan_counts(gx_ctx *ctx, uint32_t *dcounts, int64_t nthreads,
+                             uint64_t **doffs_out, int64_t *total_out)
+{
+    std::vector<uint32_t> h(nthreads);
+    HIP_CHK(ctx, hipMemcpyAsync(h.data(), dcounts, nthreads * 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    std::vector<uint64_t> offs(nthreads);
+    uint64_t acc = 0;
+    for (int64_t i = 0; i < nthreads; i++) { offs[i] = acc; acc += h[i]; }
+    uint64_t *doffs = nullptr;
+    HIP_CHK(ctx, hipMalloc(&doffs, nthreads * 8));
+    HIP_CHK(ctx, hipMemcpyAsync(doffs, offs.data(), nthreads * 8, hipMemcpyHostToDevice, ctx->stream));
+    *doffs_out = doffs;
+    *total_out = (int64_t) acc;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
+                                 uint64_t seed, gx_table **out)
+{
+    if (!ctx) return GX_ERR_INVALID;
+    int64_t ncust = (int64_t) (150000.0 * sf + 0.5);
+    int64_t nord = (int64_t) (1500000.0 * sf + 0.5);
+    int64_t nglobal = (which == GX_TPCH_CUSTOMER) ? ncust : nord;
+    int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
+    int64_t blocks = (nthreads + TPB - 1) / TPB;
+    uint32_t *dcounts = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dcounts, std::max<int64_t>(nthreads, 1) * 4));
+
+    gx_table *t = new gx_table();
+    t->ctx = ctx;
+    gx_status st = GX_OK;
+
+    if (which == GX_TPCH_CUSTOMER)
+    {
+        hipLaunchKernelGGL(k_count_cust, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
+        uint64_t *doffs; int64_t n;
+        st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
+        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        int64_t *dkey; uint8_t *dmkt;
+        HIP_CHK(ctx, hipMalloc(&dkey, n * 8));
+        HIP_CHK(ctx, hipMalloc(&dmkt, std::max<int64_t>(n, 1)));
+        hipLaunchKernelGGL(k_emit_cust, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ctx->seg, ctx->nsegs, doffs, dkey, dmkt);
+        gx_col c0, c1;
+        st = encode_column_device(ctx, dkey, 8, n, &c0);
+        if (st == GX_OK) st = encode_column_device(ctx, dmkt, 1, n, &c1);
+        HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+        hipFree(dkey); hipFree(dmkt); hipFree(doffs);
+        t->cols = {c0, c1};
+        t->nrows = n;
+    }
+    else if (which == GX_TPCH_ORDERS)
+    {
+        hipLaunchKernelGGL(k_count_ord, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
+        uint64_t *doffs; int64_t n;
+        st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
+        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        int64_t *dok, *doc; int32_t *dod, *dop;
+        HIP_CHK(ctx, hipMalloc(&dok, n * 8));
+        HIP_CHK(ctx, hipMalloc(&doc, n * 8));
+        HIP_CHK(ctx, hipMalloc(&dod, n * 4));
+        HIP_CHK(ctx, hipMalloc(&dop, n * 4));
+        hipLaunchKernelGGL(k_emit_ord, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ncust, ctx->seg, ctx->nsegs, doffs, dok, doc, dod, dop);
+        gx_col c0, c1, c2, c3;
+        st = encode_column_device(ctx, dok, 8, n, &c0);
+        if (st == GX_OK) st = encode_column_device(ctx, doc, 8, n, &c1);
+        if (st == GX_OK) st = encode_column_device(ctx, dod, 4, n, &c2);
+        if (st == GX_OK) st = encode_column_device(ctx, dop, 4, n, &c3);
+        HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+        hipFree(dok); hipFree(doc); hipFree(dod); hipFree(dop); hipFree(doffs);
+        t->cols = {c0, c1, c2, c3};
+        t->nrows = n;
+    }
+    else
+    {
+        hipLaunchKernelGGL(k_count_li, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
+        uint64_t *doffs; int64_t n;
+        st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
+        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        int64_t *dlk; double *dpr, *ddi; int32_t *dsh;
+        HIP_CHK(ctx, hipMalloc(&dlk, n * 8));
+        HIP_CHK(ctx, hipMalloc(&dpr, n * 8));
+        HIP_CHK(ctx, hipMalloc(&ddi, n * 8));
+        HIP_CHK(ctx, hipMalloc(&dsh, n * 4));
+        hipLaunchKernelGGL(k_emit_li, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                           seed, nglobal, ctx->seg, ctx->nsegs, doffs, dlk, dpr, ddi, dsh);
+        gx_col c0, c1, c2, c3;
+        st = encode_column_device(ctx, dlk, 8, n, &c0);
+        if (st == GX_OK) st = encode_column_device(ctx, dpr, 8, n, &c1);
+        if (st == GX_OK) st = encode_column_device(ctx, ddi, 8, n, &c2);
+        if (st == GX_OK) st = encode_column_device(ctx, dsh, 4, n, &c3);
+        HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+        hipFree(dlk); hipFree(dpr); hipFree(ddi); hipFree(dsh); hipFree(doffs);
+        t->cols = {c0, c1, c2, c3};
+        t->nrows = n;
+    }
+    hipFree(dcounts);
+    HIP_CHK(ctx, hipGetLastError());
+    if (st != GX_OK) { gx_table_free(t); return st; }
+    *out = t;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx,
+                                      void *host_out, int64_t cap_rows,
+                                      int verify_checksums)
+{
+    if (!ctx || !t || colidx < 0 || colidx >= (int) t->cols.size()) return GX_ERR_INVALID;
+    const gx_col &c = t->cols[colidx];
+    if (c.m.nrows > cap_rows) return GX_ERR_INVALID;
+    int64_t nblocks = (c.m.nrows + c.m.rpb - 1) / c.m.rpb;
+    void *dout = nullptr;
+    int *derr = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dout, c.m.nrows * (int64_t) c.m.width));
+    HIP_CHK(ctx, hipMalloc(&derr, 4));
+    HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, ctx->stream));
+    int grid = (int) std::min<int64_t>(nblocks, 65535);
+    if (c.m.width == 8)
+        hipLaunchKernelGGL(k_decode<int64_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
+                           (int64_t *) dout, derr);
+    else if (c.m.width == 4)
+        hipLaunchKernelGGL(k_decode<int32_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
+                           (int32_t *) dout, derr);
+    else
+        hipLaunchKernelGGL(k_decode<int8_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
+                           c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
+                           (int8_t *) dout, derr);
+    if (verify_checksums)
+        hipLaunchKernelGGL(k_verify_crc, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
+                           c.m.width, derr);
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, dout, c.m.nrows * (int64_t) c.m.width,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    hipFree(dout); hipFree(derr);
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
+    if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    return GX_OK;
+}
+
+extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t n,
+                                  int32_t nsegs, int32_t *host_out)
+{
+    if (!ctx || n < 0) return GX_ERR_INVALID;
+    int64_t *dk = nullptr; int32_t *dr = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dk, n * 8));
+    HIP_CHK(ctx, hipMalloc(&dr, n * 4));
+    HIP_CHK(ctx, hipMemcpyAsync(dk, host_keys, n * 8, hipMemcpyHostToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_route, dim3(GRID), dim3(TPB), 0, ctx->stream, dk, n, nsegs, dr);
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, dr, n * 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    hipFree(dk); hipFree(dr);
+    HIP_CHK(ctx, hipGetLastError());
+    return GX_OK;
+}
+
+/* ================= Q3 ================= */
+
+extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
+                                   gx_table *lineitem, int32_t cutoff, gx_q3 **out)
+{
+    if (!ctx || !customer || !orders || !lineitem) return GX_ERR_INVALID;
+    if (customer->cols.size() != 2 || orders->cols.size() != 4 ||
+        lineitem->cols.size() != 4) return GX_ERR_INVALID;
+    gx_q3 *q = new gx_q3();
+    q->ctx = ctx;
+    q->cust = customer;
+    q->ord = orders;
+    q->li = lineitem;
+    q->cutoff = cutoff;
+    *out = q;
+    return GX_OK;
+}
+
+static void q3_free_runstate(gx_q3 *q)
+{
+    auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
+    fr(q->cset); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+    fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
+}
+
+extern "C" gx_status gx_q3_run(gx_q3 *q)
+{
+    if (!q) return GX_ERR_INVALID;
+    gx_ctx *ctx = q->ctx;
+    hipStream_t s = ctx->stream;
+    q3_free_runstate(q);
+    memset(&q->stats, 0, sizeof q->stats);
+
+    hipEvent_t ev[8];
+    for (auto &e : ev) HIP_CHK(ctx, hipEventCreate(&e));
+
+    const gx_col &ck = q->cust->cols[0], &cm = q->cust->cols[1];
+    const gx_col &ok = q->ord->cols[0], &oc = q->ord->cols[1],
+                 &od = q->ord->cols[2], &op = q->ord->cols[3];
+    const gx_col &lk = q->li->cols[0], &lp = q->li->cols[1],
+                 &ld = q->li->cols[2], &ls = q->li->cols[3];
+
+    unsigned long long *dcount = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dcount, 8));
+
+    /* ---- stage 1: customer BUILDING set ---- */
+    HIP_CHK(ctx, hipEventRecord(ev[0], s));
+    HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
+    hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
+                       cm.dstream, cm.m, dcount);
+    unsigned long long n_building = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&n_building, dcount, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
+    HIP_CHK(ctx, hipMalloc(&q->cset, cslots * 8));
+    HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * 8, s));
+    q->cmask = cslots - 1;
+    hipLaunchKernelGGL(k_cust_build, dim3(GRID), dim3(TPB), 0, s,
+                       ck.dstream, ck.m, cm.dstream, cm.m, q->cset, q->cmask);
+    HIP_CHK(ctx, hipEventRecord(ev[1], s));
+
+    /* ---- stage 2: orders build (local or via Motions) ---- */
+    int64_t qual = 0;
+    double ms_motion = 0.0;
+    if (ctx->nsegs == 1)
+    {
+        HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
+        hipLaunchKernelGGL(k_orders_count, dim3(GRID), dim3(TPB), 0, s,
+                           od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
+                           q->cset, q->cmask, dcount);
+        unsigned long long nq = 0;
+        HIP_CHK(ctx, hipMemcpyAsync(&nq, dcount, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        qual = (int64_t) nq;
+        uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
+        q->tmask = tslots - 1;
+        HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
+        HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
+        HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
+        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
+        hipLaunchKernelGGL(k_orders_build, dim3(GRID), dim3(TPB), 0, s,
+                           ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                           op.dstream, op.m, q->cutoff, q->cset, q->cmask,
+                           q->tkey, q->tdate, q->tprio, q->tmask);
+    }
+    else
+    {
+        if (!ctx->comm) { set_err(ctx, "nsegs>1 but gx_comm_init not called%s", ""); return GX_ERR_STATE; }
+        int n = ctx->nsegs;
+        hipEvent_t mev0, mev1;
+        hipEventCreate(&mev0); hipEventCreate(&mev1);
+        HIP_CHK(ctx, hipEventRecord(mev0, s));
+
+        /* Motion 1: filtered orders by route(o_custkey) */
+        unsigned long long *dhist = nullptr;
+        HIP_CHK(ctx, hipMalloc(&dhist, n * 8));
+        HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
+        hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
+                           od.dstream, od.m, oc.dstream, oc.m, q->cutoff, n, dhist);
+        std::vector<unsigned long long> h1(n);
+        HIP_CHK(ctx, hipMemcpyAsync(h1.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        std::vector<unsigned long long> off1(n + 1, 0);
+        for (int i = 0; i < n; i++) off1[i + 1] = off1[i] + h1[i];
+        unsigned long long send1_n = off1[n];
+        gx_ord_row *send1 = nullptr;
+        HIP_CHK(ctx, hipMalloc(&send1, std::max<uint64_t>(send1_n, 1) * sizeof(gx_ord_row)));
+        unsigned long long *dcur = nullptr;
+        HIP_CHK(ctx, hipMalloc(&dcur, n * 8));
+        HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
+        hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
+                           ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                           op.dstream, op.m, q->cutoff, n, dcur, send1);
+
+        /* exchange counts (all-gather of per-dest counts) */
+        unsigned long long *dcnts_all = nullptr, *dcnts_mine = nullptr;
+        HIP_CHK(ctx, hipMalloc(&dcnts_mine, n * 8));
+        HIP_CHK(ctx, hipMalloc(&dcnts_all, (int64_t) n * n * 8));
+        HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h1.data(), n * 8, hipMemcpyHostToDevice, s));
+        RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
+        std::vector<unsigned long long> cnts_all(n * n);
+        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
+                                    hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        std::vector<unsigned long long> rcv1(n), roff1(n + 1, 0);
+        for (int r = 0; r < n; r++) rcv1[r] = cnts_all[(int64_t) r * n + ctx->seg];
+        for (int r = 0; r < n; r++) roff1[r + 1] = roff1[r] + rcv1[r];
+        unsigned long long recv1_n = roff1[n];
+        gx_ord_row *recv1 = nullptr;
+        HIP_CHK(ctx, hipMalloc(&recv1, std::max<uint64_t>(recv1_n, 1) * sizeof(gx_ord_row)));
+        RCCL_CHK(ctx, ncclGroupStart());
+        for (int r = 0; r < n; r++)
+        {
+            if (h1[r])
+                RCCL_CHK(ctx, ncclSend(send1 + off1[r], h1[r] * sizeof(gx_ord_row),
+                                       ncclInt8, r, ctx->comm, s));
+            if (rcv1[r])
+                RCCL_CHK(ctx, ncclRecv(recv1 + roff1[r], rcv1[r] * sizeof(gx_ord_row),
+                                       ncclInt8, r, ctx->comm, s));
+        }
+        RCCL_CHK(ctx, ncclGroupEnd());
+
+        /* Motion 2: probe local customer set, route qualifying by o_orderkey */
+        HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
+        hipLaunchKernelGGL(k_qual_hist, dim3(GRID), dim3(TPB), 0, s,
+                           recv1, (int64_t) recv1_n, q->cset, q->cmask, n, dhist);
+        std::vector<unsigned long long> h2(n);
+        HIP_CHK(ctx, hipMemcpyAsync(h2.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        std::vector<unsigned long long> off2(n + 1, 0);
+        for (int i = 0; i < n; i++) off2[i + 1] = off2[i] + h2[i];
+        gx_qual_row *send2 = nullptr;
+        HIP_CHK(ctx, hipMalloc(&send2, std::max<uint64_t>(off2[n], 1) * sizeof(gx_qual_row)));
+        HIP_CHK(ctx, hipMemcpyAsync(dcur, off2.data(), n * 8, hipMemcpyHostToDevice, s));
+        hipLaunchKernelGGL(k_qual_emit, dim3(GRID), dim3(TPB), 0, s,
+                           recv1, (int64_t) recv1_n, q->cset, q->cmask, n, dcur, send2);
+        HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h2.data(), n * 8, hipMemcpyHostToDevice, s));
+        RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
+        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
+                                    hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        std::vector<unsigned long long> rcv2(n), roff2(n + 1, 0);
+        for (int r = 0; r < n; r++) rcv2[r] = cnts_all[(int64_t) r * n + ctx->seg];
+        for (int r = 0; r < n; r++) roff2[r + 1] = roff2[r] + rcv2[r];
+        unsigned long long recv2_n = roff2[n];
+        gx_qual_row *recv2 = nullptr;
+        HIP_CHK(ctx, hipMalloc(&recv2, std::max<uint64_t>(recv2_n, 1) * sizeof(gx_qual_row)));
+        RCCL_CHK(ctx, ncclGroupStart());
+        for (int r = 0; r < n; r++)
+        {
+            if (h2[r])
+                RCCL_CHK(ctx, ncclSend(send2 + off2[r], h2[r] * sizeof(gx_qual_row),
+                                       ncclInt8, r, ctx->comm, s));
+            if (rcv2[r])
+                RCCL_CHK(ctx, ncclRecv(recv2 + roff2[r], rcv2[r] * sizeof(gx_qual_row),
+                                       ncclInt8, r, ctx->comm, s));
+        }
+        RCCL_CHK(ctx, ncclGroupEnd());
+
+        qual = (int64_t) recv2_n;
+        uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
+        q->tmask = tslots - 1;
+        HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
+        HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
+        HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
+        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
+        hipLaunchKernelGGL(k_build_from_rows, dim3(GRID), dim3(TPB), 0, s,
+                           recv2, qual, q->tkey, q->tdate, q->tprio, q->tmask);
+        HIP_CHK(ctx, hipEventRecord(mev1, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        float mms = 0;
+        hipEventElapsedTime(&mms, mev0, mev1);
+        ms_motion = mms;
+        hipEventDestroy(mev0); hipEventDestroy(mev1);
+        hipFree(dhist); hipFree(dcur); hipFree(send1); hipFree(recv1);
+        hipFree(send2); hipFree(recv2); hipFree(dcnts_mine); hipFree(dcnts_all);
+    }
+    q->qual_orders = qual;
+    HIP_CHK(ctx, hipEventRecord(ev[2], s));
+
+    /* ---- stage 3: lineitem scan+probe+agg (dominant kernel) ---- */
+    unsigned long long *dhits = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dhits, 8));
+    HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
+    hipLaunchKernelGGL(k_li_probe_agg, dim3(GRID), dim3(TPB), 0, s,
+                       lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                       ls.dstream, ls.m, q->cutoff, q->tkey, 0,
+                       q->trev, q->tcnt, q->tmask, dhits);
+    HIP_CHK(ctx, hipEventRecord(ev[3], s));
+
+    /* ---- stage 4: extract ---- */
+    int64_t rescap = std::max<int64_t>(qual, 1);
+    HIP_CHK(ctx, hipMalloc(&q->r_okey, rescap * 8));
+    HIP_CHK(ctx, hipMalloc(&q->r_odate, rescap * 4));
+    HIP_CHK(ctx, hipMalloc(&q->r_oprio, rescap * 4));
+    HIP_CHK(ctx, hipMalloc(&q->r_rev, rescap * 8));
+    HIP_CHK(ctx, hipMalloc(&q->r_cnt, rescap * 8));
+    HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
+    hipLaunchKernelGGL(k_extract, dim3(GRID), dim3(TPB), 0, s,
+                       q->tkey, q->tdate, q->tprio, q->trev, q->tcnt, q->tmask + 1,
+                       q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
+    HIP_CHK(ctx, hipEventRecord(ev[4], s));
+
+    unsigned long long ngroups = 0, hits = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&ngroups, dcount, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&hits, dhits, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    q->ngroups = (int64_t) ngroups;
+
+    float t01 = 0, t12 = 0, t23 = 0, t34 = 0, t04 = 0;
+    hipEventElapsedTime(&t01, ev[0], ev[1]);
+    hipEventElapsedTime(&t12, ev[1], ev[2]);
+    hipEventElapsedTime(&t23, ev[2], ev[3]);
+    hipEventElapsedTime(&t34, ev[3], ev[4]);
+    hipEventElapsedTime(&t04, ev[0], ev[4]);
+    q->stats.ms_cust_build = t01;
+    q->stats.ms_orders_build = t12 - ms_motion;
+    q->stats.ms_motion = ms_motion;
+    q->stats.ms_probe_agg = t23;
+    q->stats.ms_extract = t34;
+    q->stats.ms_total = t04;
+    q->stats.cust_rows = q->cust->nrows;
+    q->stats.ord_rows = q->ord->nrows;
+    q->stats.li_rows = q->li->nrows;
+    q->stats.probe_hits = (int64_t) hits;
+    q->stats.groups = q->ngroups;
+    double b = 0, bb = 0;
+    gx_table_logical_bytes(q->cust, &b); bb += b;
+    gx_table_logical_bytes(q->ord, &b); bb += b;
+    gx_table_logical_bytes(q->li, &b); bb += b;
+    q->stats.bytes_scanned = bb;
+
+    hipFree(dcount); hipFree(dhits);
+    for (auto &e : ev) hipEventDestroy(e);
+    q->ran = true;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_q3_stats_get(const gx_q3 *q, gx_q3_stats *out)
+{
+    if (!q || !q->ran) return GX_ERR_STATE;
+    *out = q->stats;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
+{
+    if (!q || !q->ran) return GX_ERR_STATE;
+    gx_ctx *ctx = q->ctx;
+    int64_t n = q->ngroups;
+    std::vector<int64_t> okey(n), cnt(n);
+    std::vector<int32_t> odate(n), oprio(n);
+    std::vector<double> rev(n);
+    if (n)
+    {
+        HIP_CHK(ctx, hipMemcpy(okey.data(), q->r_okey, n * 8, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(odate.data(), q->r_odate, n * 4, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(oprio.data(), q->r_oprio, n * 4, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(rev.data(), q->r_rev, n * 8, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(cnt.data(), q->r_cnt, n * 8, hipMemcpyDeviceToHost));
+    }
+    std::vector<int64_t> idx(n);
+    for (int64_t i = 0; i < n; i++) idx[i] = i;
+    std::sort(idx.begin(), idx.end(),
+              [&](int64_t a, int64_t b) { return okey[a] < okey[b]; });
+    gx_q3_group *g = (gx_q3_group *) malloc(sizeof(gx_q3_group) * std::max<int64_t>(n, 1));
+    for (int64_t i = 0; i < n; i++)
+    {
+        g[i].l_orderkey = okey[idx[i]];
+        g[i].o_orderdate = odate[idx[i]];
+        g[i].o_shippriority = oprio[idx[i]];
+        g[i].revenue = rev[idx[i]];
+        g[i].nitems = cnt[idx[i]];
+    }
+    *out = g;
+    *ngroups = n;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_q3_free(gx_q3 *q)
+{
+    if (!q) return GX_OK;
+    q3_free_runstate(q);
+    delete q;
+    return GX_OK;
+}
+
+extern "C" void gx_free(void *p) { free(p); }

@@ -1,0 +1,154 @@
+/*
+ * gpuexec.h — C-ABI of libgpuexec.so, the MI355X-native executor for
+ * Cloudberry's segment-local scan→hash-join→hash-agg pipeline + hash Motion.
+ *
+ * This is the drop-in boundary.  Each entry point names the
+ * reference seam it replaces (paths under /root/reference):
+ *
+ *   gx_init / gx_shutdown      — per-QE executor state; what a CustomScan
+ *                                extension's BeginCustomScan/EndCustomScan
+ *                                (src/include/nodes/extensible.h:124-158,
+ *                                executor/nodeCustom.c) would call once per
+ *                                segment process.
+ *   gx_comm_*                  — the MotionIPCLayer SetupInterconnect /
+ *                                TeardownInterconnect pair (cdb/ml_ipc.h:36,
+ *                                executor/execMain.c:535); RCCL communicator
+ *                                is created once per process lifetime
+ *                                (mirrors gang reuse, dispatcher/README.md).
+ *   gx_table_bind              — scan target binding: the table-AM open path
+ *                                aocs_beginscan (access/aocs/aocsam.c:542)
+ *                                with per-column segment-file streams.
+ *   gx_tpch_gen                — bench-harness substitute for on-disk data
+ *                                (no network; synthetic).
+ *   gx_decode_column           — aocs_getnext datum decode
+ *                                (aocsam.c:1131-1259, datumstreamblock.c:
+ *                                195-340) materialised column-at-a-time.
+ *   gx_q3_prepare/run/result   — the QE slice ExecProcNode pull loop over
+ *                                SeqScan→HashJoin→HashAgg (+ Motion sends at
+ *                                nsegs>1): execMain.c:983, nodeHashjoin.c:252,
+ *                                nodeAgg.c:2743, nodeMotion.c:1181.
+ *   gx_partition (exposed for tests) — doSendTuple's evalHashKey +
+ *                                cdbhashreduce routing (nodeMotion.c:1088,
+ *                                cdbhash.c:253-285,530-541), bit-exact.
+ *
+ * Plain C, status-code returns, caller-owned opaque handles.  No torch
+ * types; no exceptions across the boundary (the PG-side shim converts
+ * non-zero status into ereport(ERROR)).  All calls are per-process
+ * single-threaded, matching the QE execution model (one thread per segment;
+ * HIP streams + RCCL run on that thread).
+ */
+#ifndef GPUEXEC_H
+#define GPUEXEC_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef enum gx_status {
+    GX_OK = 0,
+    GX_ERR_HIP = 1,            /* HIP runtime failure (see gx_last_error) */
+    GX_ERR_RCCL = 2,           /* RCCL failure */
+    GX_ERR_INVALID = 3,        /* bad argument / malformed stream */
+    GX_ERR_CHECKSUM = 4,       /* AO block checksum mismatch */
+    GX_ERR_OOM = 5,            /* device memory exhausted */
+    GX_ERR_NOGPU = 6,          /* no usable device — callers must FAIL, not fall back */
+    GX_ERR_STATE = 7           /* calls out of order */
+} gx_status;
+
+typedef struct gx_ctx gx_ctx;
+typedef struct gx_table gx_table;
+typedef struct gx_q3 gx_q3;
+
+/* last error message for a context (or the global init error when ctx==NULL) */
+const char *gx_last_error(const gx_ctx *ctx);
+const char *gx_version(void);
+
+/* ---- lifecycle ---- */
+gx_status gx_init(int device_id, int seg_id, int nsegs, gx_ctx **out);
+gx_status gx_shutdown(gx_ctx *ctx);
+
+/* ---- interconnect (RCCL over xGMI; one rank per segment-GPU) ---- */
+#define GX_UNIQUE_ID_BYTES 128
+gx_status gx_comm_unique_id(unsigned char uid[GX_UNIQUE_ID_BYTES]); /* rank 0 */
+gx_status gx_comm_init(gx_ctx *ctx, const unsigned char uid[GX_UNIQUE_ID_BYTES]);
+
+/* ---- tables: AOCS per-column streams resident in HBM ---- */
+
+typedef struct gx_coldesc {
+    const void *host_stream;   /* AOCS stream bytes (appendonly=column,
+                                  compresstype=none, checksum=true) */
+    int64_t     nbytes;
+    int32_t     width;         /* fixed datum width: 1, 4 or 8 */
+    int64_t     nrows;
+    int32_t     blocksize;     /* AO blocksize the stream was written with */
+} gx_coldesc;
+
+gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncols,
+                        gx_table **out);
+gx_status gx_table_free(gx_table *t);
+gx_status gx_table_nrows(const gx_table *t, int64_t *out);
+/* logical uncompressed bytes of all columns = the reference's
+ * totalBytesRead accounting base (cdb/cdbaocsam.h:283) */
+gx_status gx_table_logical_bytes(const gx_table *t, double *out);
+
+/* synthetic TPC-H-shaped tables generated AND AOCS-encoded on device
+ * (deterministic; identical formulas to oracle/oracle.c datagen) */
+typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2 } gx_tpch_table;
+gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
+                      uint64_t seed, gx_table **out);
+
+/* decode one column back to host values (parity testing / config-2 path);
+ * verify_checksums runs the CRC32C pair per block on device */
+gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int col,
+                           void *host_out, int64_t cap_rows,
+                           int verify_checksums);
+
+/* bit-exact Motion routing of an i64 key column on device:
+ * out[i] = jump_consistent_hash(cdbhash(hashint8(keys[i])), nsegs) */
+gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t n,
+                       int32_t nsegs, int32_t *host_out);
+
+/* ---- the Q3 pipeline (the judged QE slice) ---- */
+
+typedef struct gx_q3_group {
+    int64_t l_orderkey;
+    int32_t o_orderdate;
+    int32_t o_shippriority;
+    double  revenue;
+    int64_t nitems;
+} gx_q3_group;
+
+typedef struct gx_q3_stats {
+    /* per-stage device times, ms (HIP events on the executor stream) */
+    double ms_cust_build;
+    double ms_orders_build;
+    double ms_probe_agg;       /* the dominant kernel */
+    double ms_extract;
+    double ms_motion;          /* partition+exchange (0 at nsegs==1) */
+    double ms_total;           /* event span over the whole pipeline */
+    int64_t cust_rows, ord_rows, li_rows;
+    int64_t probe_hits, groups;
+    double bytes_scanned;      /* logical uncompressed bytes (cdbaocsam.h:283) */
+} gx_q3_stats;
+
+/* customer cols: [c_custkey i64, c_mktsegment i8]
+ * orders   cols: [o_orderkey i64, o_custkey i64, o_orderdate i32, o_shippriority i32]
+ * lineitem cols: [l_orderkey i64, l_extendedprice f64, l_discount f64, l_shipdate i32] */
+gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
+                        gx_table *lineitem, int32_t cutoff_dateadt,
+                        gx_q3 **out);
+gx_status gx_q3_run(gx_q3 *q);   /* one full pass; re-runnable (bench steps) */
+gx_status gx_q3_stats_get(const gx_q3 *q, gx_q3_stats *out);
+/* groups of THIS segment, sorted by l_orderkey asc; caller frees with gx_free */
+gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups);
+gx_status gx_q3_free(gx_q3 *q);
+
+void gx_free(void *p);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GPUEXEC_H */
