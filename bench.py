@@ -1,0 +1,190 @@
+#!/usr/bin/env python3
+"""bench.py — TPC-H Q3 @ SF100 on the MI355X executor (BASELINE.json metric:
+rows/sec + GB/s scanned).
+
+One step = one full Q3 pass (customer set build → orders build [+ RCCL
+Motions at N>1] → lineitem scan+probe+agg → extract) over AOCS streams
+already resident in HBM.  Synthetic TPC-H-shaped tables (no network), hash-
+distributed across ranks by the reference's own DISTRIBUTED BY keys; N>1 is
+STRONG scaling (fixed SF split across segments, the reference's MPP model).
+
+Single: python bench.py --gpus 1 --steps K --warmup W
+Multi:  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+            --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+HBM_PEAK_GBPS = 8000.0   # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def log(rank, *a):
+    if rank == 0:
+        print(*a, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(sample_sf=8.0):
+    """Oracle (CPU restatement, 'port') timed on this host: the reported CPU
+    baseline (BASELINE.md — the reference publishes no numbers).  Bounded
+    sample; single thread; pipeline only (inputs pre-materialized)."""
+    from oracle import pyapi as orc   # checker/baseline use only
+    c = orc.gen_customer(sample_sf)
+    o = orc.gen_orders(sample_sf)
+    li = orc.gen_lineitem(sample_sf)
+    reps, t = 0, 0.0
+    t_end = time.time() + 12.0
+    while time.time() < t_end or reps < 2:
+        t0 = time.perf_counter()
+        orc.q3(c, o, li)
+        t += time.perf_counter() - t0
+        reps += 1
+        if reps >= 8:
+            break
+    rows = len(li["l_orderkey"]) * reps
+    return {"value": rows / t, "unit": "rows/s", "cores": 1, "kind": "port",
+            "sample": f"tpch_q3_sf{sample_sf:g} oracle pipeline x{reps} "
+                      f"({len(li['l_orderkey'])} lineitem rows/pass, single thread)"}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--sf", type=float, default=100.0)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n = max(world, 1)
+    assert n == args.gpus or world == 1, \
+        f"WORLD_SIZE={world} but --gpus={args.gpus}"
+    n = args.gpus
+
+    import numpy as np
+    import torch
+    import cloudberry_amd as gx
+
+    dist = None
+    if n > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group(backend="gloo")  # bootstrap/barriers only;
+        dist = tdist                              # data path is our own RCCL comm
+
+    ctx = gx.Context(device=local_rank, seg=rank, nsegs=n)
+    torch.cuda.set_device(local_rank)
+
+    if n > 1:
+        # broadcast RCCL unique id over gloo, then build the communicator
+        if rank == 0:
+            uid = ctx.comm_unique_id()
+            t = torch.tensor(list(uid), dtype=torch.uint8)
+        else:
+            t = torch.zeros(128, dtype=torch.uint8)
+        dist.broadcast(t, src=0)
+        ctx.comm_init(bytes(t.tolist()))
+
+    # ---- setup (untimed): generate + AOCS-encode this segment's shard ----
+    t0 = time.time()
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, args.sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, args.sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, args.sf)
+    q = ctx.q3(cust, ordr, li)
+    log(rank, f"setup: sf={args.sf} rank {rank}/{n}: "
+              f"{cust.nrows} cust, {ordr.nrows} ord, {li.nrows} li rows "
+              f"({time.time()-t0:.1f}s)")
+
+    li_rows_local = li.nrows
+    bytes_local = cust.logical_bytes + ordr.logical_bytes + li.logical_bytes
+    if n > 1:
+        t = torch.tensor([li_rows_local, int(bytes_local)], dtype=torch.int64)
+        dist.all_reduce(t)
+        li_rows_total, bytes_total = int(t[0]), float(t[1])
+    else:
+        li_rows_total, bytes_total = li_rows_local, bytes_local
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        q.run()
+
+    # ---- timed region ----
+    if dist: dist.barrier()
+    torch.cuda.synchronize()
+    t_start = time.perf_counter()
+    probe_ms = 0.0
+    for _ in range(args.steps):
+        q.run()
+        probe_ms += q.stats()["ms_probe_agg"]
+    torch.cuda.synchronize()
+    if dist: dist.barrier()
+    elapsed = time.perf_counter() - t_start
+
+    # MAX over ranks
+    if n > 1:
+        t = torch.tensor([elapsed])
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    st = q.stats()
+    ms_step = elapsed * 1000.0 / args.steps
+    rows_per_sec = li_rows_total / (elapsed / args.steps)
+    gbps = bytes_total / 1e9 / (elapsed / args.steps)
+
+    # roofline: dominant kernel = fused lineitem scan+probe+agg.
+    # algorithmic bytes/launch = lineitem 4 cols (28 B/row) of
+    # THIS rank; duration from HIP events on the executor stream.
+    probe_ms_avg = probe_ms / args.steps
+    li_bytes_local = li_rows_local * 28.0
+    achieved = li_bytes_local / 1e9 / (probe_ms_avg / 1000.0)
+    roofline = {"bound": "hbm", "achieved": round(achieved, 1),
+                "peak": HBM_PEAK_GBPS, "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK_GBPS, 4), "traffic": None}
+
+    cpu = None
+    if rank == 0 and n == 1 and not args.no_cpu_baseline:
+        log(rank, "timing CPU baseline (oracle, bounded sample)...")
+        cpu = cpu_baseline_leg()
+
+    if rank == 0:
+        out = {
+            "metric": "tpch_q3_rows_per_sec",
+            "value": round(rows_per_sec, 1),
+            "unit": "rows/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {"workload": f"tpch_q3_sf{args.sf:g}", "sf": args.sf,
+                       "lineitem_rows": li_rows_total,
+                       "scan_bytes": bytes_total,
+                       "gbps_scanned": round(gbps, 1),
+                       "parallelism": f"mpp{n}",
+                       "groups": st["groups"],
+                       "stage_ms": {k: round(st[k], 3) for k in
+                                    ("ms_cust_build", "ms_orders_build",
+                                     "ms_motion", "ms_probe_agg", "ms_extract")}},
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(out), flush=True)
+
+    q.free(); li.free(); ordr.free(); cust.free(); ctx.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
