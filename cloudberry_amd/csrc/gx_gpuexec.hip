@@ -807,8 +807,10 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     int64_t nglobal = (which == GX_TPCH_CUSTOMER) ? ncust : nord;
     int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
     int64_t blocks = (nthreads + TPB - 1) / TPB;
+    /* the count kernels write counts[t] for EVERY launched thread */
+    int64_t nthreads_alloc = blocks * TPB;
     uint32_t *dcounts = nullptr;
-    HIP_CHK(ctx, hipMalloc(&dcounts, std::max<int64_t>(nthreads, 1) * 4));
+    HIP_CHK(ctx, hipMalloc(&dcounts, std::max<int64_t>(nthreads_alloc, 1) * 4));
 
     gx_table *t = new gx_table();
     t->ctx = ctx;
