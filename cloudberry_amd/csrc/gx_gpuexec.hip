@@ -358,12 +358,8 @@ __global__ void k_cust_count(const uint8_t *mkt_s, gx_colmeta mkt_m,
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0;
-    if (i < mkt_m.nrows)
-    {
-        gx_colcur mk; mk.init(mkt_s, mkt_m, i, stride);
-        for (; i < mkt_m.nrows; i += stride, mk.advance())
-            if (mk.load<uint8_t>() == 0) local++;
-    }
+    for (; i < mkt_m.nrows; i += stride)
+        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) == 0) local++;
     gx_wave_count_add(count, local);
 }
 
@@ -375,14 +371,10 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    if (i >= key_m.nrows) return;
-    gx_colcur kc, mk;
-    kc.init(key_s, key_m, i, stride);
-    mk.init(mkt_s, mkt_m, i, stride);
-    for (; i < key_m.nrows; i += stride, kc.advance(), mk.advance())
+    for (; i < key_m.nrows; i += stride)
     {
-        if (mk.load<uint8_t>() != 0) continue;
-        uint64_t k = (uint64_t) kc.load<int64_t>();
+        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) != 0) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
         uint64_t slot = gx_hmix64(k) & mask;
         while (true)
         {
@@ -416,17 +408,11 @@ __global__ void k_orders_count(const uint8_t *od_s, gx_colmeta od_m,
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0;
-    if (i < od_m.nrows)
+    for (; i < od_m.nrows; i += stride)
     {
-        gx_colcur dc, cc;
-        dc.init(od_s, od_m, i, stride);
-        cc.init(oc_s, oc_m, i, stride);
-        for (; i < od_m.nrows; i += stride, dc.advance(), cc.advance())
-        {
-            if (!(dc.load<int32_t>() < cutoff)) continue;
-            if (!d_set_contains(cset, cmask, (uint64_t) cc.load<int64_t>())) continue;
-            local++;
-        }
+        if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
+        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        local++;
     }
     gx_wave_count_add(count, local);
 }
@@ -445,19 +431,12 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    if (i >= ok_m.nrows) return;
-    gx_colcur kc, cc, dc, pc;
-    kc.init(ok_s, ok_m, i, stride);
-    cc.init(oc_s, oc_m, i, stride);
-    dc.init(od_s, od_m, i, stride);
-    pc.init(op_s, op_m, i, stride);
-    for (; i < ok_m.nrows;
-         i += stride, kc.advance(), cc.advance(), dc.advance(), pc.advance())
+    for (; i < ok_m.nrows; i += stride)
     {
-        int32_t od = dc.load<int32_t>();
+        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!(od < cutoff)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) cc.load<int64_t>())) continue;
-        uint64_t k = (uint64_t) kc.load<int64_t>();
+        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
         uint64_t slot = gx_hmix64(k) & tmask;
         while (true)
         {
@@ -465,7 +444,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
             if (prev == 0ULL)
             {
                 tdate[slot] = od;
-                tprio[slot] = pc.load<int32_t>();
+                tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
                 break;
             }
             if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
@@ -491,34 +470,25 @@ __global__ void k_li_probe_agg(const uint8_t *lk_s, gx_colmeta lk_m,
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local_hits = 0;
-    if (i < lk_m.nrows)
+    for (; i < lk_m.nrows; i += stride)
     {
-        gx_colcur kc, pc, dc, sc;
-        kc.init(lk_s, lk_m, i, stride);
-        pc.init(pr_s, pr_m, i, stride);
-        dc.init(di_s, di_m, i, stride);
-        sc.init(sh_s, sh_m, i, stride);
-        for (; i < lk_m.nrows;
-             i += stride, kc.advance(), pc.advance(), dc.advance(), sc.advance())
+        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+        uint64_t slot = gx_hmix64(k) & tmask;
+        bool found = false;
+        while (true)
         {
-            if (!(sc.load<int32_t>() > cutoff)) continue;
-            uint64_t k = (uint64_t) kc.load<int64_t>();
-            uint64_t slot = gx_hmix64(k) & tmask;
-            bool found = false;
-            while (true)
-            {
-                unsigned long long v = tkey[slot];
-                if (v == 0ULL) break;
-                if (v == (unsigned long long) k) { found = true; break; }
-                slot = (slot + 1) & tmask;
-            }
-            if (!found) continue;
-            double price = pc.load<double>();
-            double disc = dc.load<double>();
-            atomicAdd(&trev[slot], price * (1.0 - disc));
-            atomicAdd(&tcnt[slot], 1ULL);
-            local_hits++;
+            unsigned long long v = tkey[slot];
+            if (v == 0ULL) break;
+            if (v == (unsigned long long) k) { found = true; break; }
+            slot = (slot + 1) & tmask;
         }
+        if (!found) continue;
+        double price = gx_col_get<double>(pr_s, pr_m, i);
+        double disc = gx_col_get<double>(di_s, di_m, i);
+        atomicAdd(&trev[slot], price * (1.0 - disc));
+        atomicAdd(&tcnt[slot], 1ULL);
+        local_hits++;
     }
     gx_wave_count_add(hits, local_hits);
 }
@@ -797,6 +767,7 @@ static gx_status encode_column_device(gx_ctx *ctx, const void *dvals, int width,
     col->m.nrows = nrows;
     col->m.full_block_len = full_len;
     col->m.nbytes = bytes;
+    gx_colmeta_finish(&col->m);
     return GX_OK;
 }
 
@@ -815,6 +786,7 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
         col.m.nrows = cols[c].nrows;
         col.m.full_block_len = gx_aocs_block_len(cols[c].width, col.m.rpb);
         col.m.nbytes = cols[c].nbytes;
+        gx_colmeta_finish(&col.m);
         hipError_t e = hipMalloc(&col.dstream, cols[c].nbytes);
         if (e != hipSuccess) { set_err(ctx, "hipMalloc: %s", hipGetErrorString(e)); delete t; return GX_ERR_OOM; }
         e = hipMemcpyAsync(col.dstream, cols[c].host_stream, cols[c].nbytes,
@@ -1388,3 +1360,29 @@ extern "C" gx_status gx_q3_free(gx_q3 *q)
 }
 
 extern "C" void gx_free(void *p) { free(p); }
+
+/* host-side self-test of the division-free row→block addressing (callable
+ * without a GPU; exercised by tests/test_abi_cpu.py) */
+extern "C" int gx_selftest_addressing(void)
+{
+    for (int width : {1, 4, 8})
+    {
+        gx_colmeta m{};
+        m.width = width;
+        m.rpb = gx_aocs_rows_per_block(width, 32768);
+        m.full_block_len = gx_aocs_block_len(width, m.rpb);
+        gx_colmeta_finish(&m);
+        /* boundaries of the first 4M blocks plus large sampled rows */
+        for (int64_t b = 0; b < 4 * 1000 * 1000; b += 997)
+            for (int64_t r : {(int64_t) 0, (int64_t) (m.rpb - 1)})
+            {
+                int64_t row = b * (int64_t) m.rpb + r;
+                int64_t bb = (int64_t) gx_mulhi64((uint64_t) row, m.magic);
+                if (bb != b) return 1;
+            }
+        for (int64_t row = 1; row < (int64_t) 1e15; row = row * 3 + 7)
+            if ((int64_t) gx_mulhi64((uint64_t) row, m.magic) != row / m.rpb)
+                return 2;
+    }
+    return 0;
+}

@@ -56,6 +56,15 @@ def test_init_without_gpu_fails_loudly():
     assert st == 6  # GX_ERR_NOGPU
 
 
+def test_division_free_addressing_exact():
+    """gx_mulhi64(row, magic) must equal row // rpb for every width's rpb
+    (the GPU scan kernels rely on it for O(1) AOCS addressing)."""
+    _build_if_needed()
+    lib = ctypes.CDLL(SO)
+    lib.gx_selftest_addressing.restype = ctypes.c_int
+    assert lib.gx_selftest_addressing() == 0
+
+
 def test_gfx950_code_object_embedded():
     _build_if_needed()
     with open(SO, "rb") as f:
