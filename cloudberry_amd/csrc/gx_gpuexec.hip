@@ -417,38 +417,63 @@ __global__ void k_orders_count(const uint8_t *od_s, gx_colmeta od_m,
     gx_wave_count_add(count, local);
 }
 
+/* 16-bit slot tags: the probe's miss path (the common case) walks a small
+ * L3-resident array instead of the 8-B key array.  tag 0 = empty slot; real
+ * tags have the high bit forced so they are never 0.  Tags are a FILTER:
+ * a tag match still verifies the full key.  Written by the single winning
+ * inserter after its CAS; probe kernels run only after build completes. */
+__device__ __forceinline__ uint16_t gx_tag(uint64_t h)
+{ return (uint16_t) (h >> 48) | 0x8000; }
+
 /* orders local path: build the join/agg table keyed by o_orderkey.
  * (ExecHashTableInsert nodeHash.c:1886; o_orderkey unique → 1 entry/key;
  *  payload doubles as the agg group state, nodeAgg.c group = join row) */
+static constexpr int OB = 4;     /* rows per thread per iteration (ILP batch) */
+
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
                                int32_t cutoff,
                                const unsigned long long *cset, uint64_t cmask,
-                               unsigned long long *tkey, int32_t *tdate,
-                               int32_t *tprio, uint64_t tmask)
+                               unsigned long long *tkey, uint16_t *ttag,
+                               int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < ok_m.nrows; i += stride)
+    int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * OB;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x * OB;
+    for (; base < ok_m.nrows; base += stride)
     {
-        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
-        if (!(od < cutoff)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
-        uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
-        uint64_t slot = gx_hmix64(k) & tmask;
-        while (true)
-        {
-            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
-            if (prev == 0ULL)
+        int nb = (int) min((int64_t) OB, ok_m.nrows - base);
+        int32_t od[OB];
+        int64_t ocust[OB];
+#pragma unroll
+        for (int b = 0; b < OB; b++)
+            if (b < nb)
             {
-                tdate[slot] = od;
-                tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
-                break;
+                od[b] = gx_col_get<int32_t>(od_s, od_m, base + b);
+                ocust[b] = gx_col_get<int64_t>(oc_s, oc_m, base + b);
             }
-            if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
-            slot = (slot + 1) & tmask;
+#pragma unroll
+        for (int b = 0; b < OB; b++)
+        {
+            if (b >= nb || !(od[b] < cutoff)) continue;
+            if (!d_set_contains(cset, cmask, (uint64_t) ocust[b])) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, base + b);
+            uint64_t h = gx_hmix64(k);
+            uint64_t slot = h & tmask;
+            while (true)
+            {
+                unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
+                if (prev == 0ULL)
+                {
+                    tdate[slot] = od[b];
+                    tprio[slot] = gx_col_get<int32_t>(op_s, op_m, base + b);
+                    ttag[slot] = gx_tag(h);
+                    break;
+                }
+                if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
+                slot = (slot + 1) & tmask;
+            }
         }
     }
 }
@@ -457,38 +482,61 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * scan (aocsam.c:1131 semantics) + probe (nodeHashjoin.c:553-652) +
  * SUM transition (nodeAgg.c:836 + float.c:769) fused; build slots ARE the
  * agg groups (group key functionally determined by l_orderkey). */
+static constexpr int LB = 4;     /* lineitem rows per thread per iteration */
+
 __global__ void k_li_probe_agg(const uint8_t *lk_s, gx_colmeta lk_m,
                                const uint8_t *pr_s, gx_colmeta pr_m,
                                const uint8_t *di_s, gx_colmeta di_m,
                                const uint8_t *sh_s, gx_colmeta sh_m,
                                int32_t cutoff,
-                               const unsigned long long *tkey, int32_t, /*unused*/
+                               const unsigned long long *tkey,
+                               const uint16_t *ttag,
                                double *trev, unsigned long long *tcnt,
                                uint64_t tmask,
                                unsigned long long *hits)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * LB;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x * LB;
     unsigned long long local_hits = 0;
-    for (; i < lk_m.nrows; i += stride)
+    for (; base < lk_m.nrows; base += stride)
     {
-        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
-        uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
-        uint64_t slot = gx_hmix64(k) & tmask;
-        bool found = false;
-        while (true)
+        int nb = (int) min((int64_t) LB, lk_m.nrows - base);
+        /* batch the sequential loads — independent, so they overlap (G7);
+         * keys are loaded unconditionally: the passing rows touch nearly
+         * every key cache line anyway at 54% selectivity */
+        int32_t ship[LB];
+        int64_t key[LB];
+#pragma unroll
+        for (int b = 0; b < LB; b++)
+            if (b < nb)
+            {
+                ship[b] = gx_col_get<int32_t>(sh_s, sh_m, base + b);
+                key[b] = gx_col_get<int64_t>(lk_s, lk_m, base + b);
+            }
+#pragma unroll
+        for (int b = 0; b < LB; b++)
         {
-            unsigned long long v = tkey[slot];
-            if (v == 0ULL) break;
-            if (v == (unsigned long long) k) { found = true; break; }
-            slot = (slot + 1) & tmask;
+            if (b >= nb || !(ship[b] > cutoff)) continue;
+            uint64_t k = (uint64_t) key[b];
+            uint64_t h = gx_hmix64(k);
+            uint64_t slot = h & tmask;
+            uint16_t tg = gx_tag(h);
+            bool found = false;
+            while (true)
+            {
+                uint16_t tt = ttag[slot];
+                if (tt == 0) break;
+                if (tt == tg && tkey[slot] == (unsigned long long) k)
+                { found = true; break; }
+                slot = (slot + 1) & tmask;
+            }
+            if (!found) continue;
+            double price = gx_col_get<double>(pr_s, pr_m, base + b);
+            double disc = gx_col_get<double>(di_s, di_m, base + b);
+            atomicAdd(&trev[slot], price * (1.0 - disc));
+            atomicAdd(&tcnt[slot], 1ULL);
+            local_hits++;
         }
-        if (!found) continue;
-        double price = gx_col_get<double>(pr_s, pr_m, i);
-        double disc = gx_col_get<double>(di_s, di_m, i);
-        atomicAdd(&trev[slot], price * (1.0 - disc));
-        atomicAdd(&tcnt[slot], 1ULL);
-        local_hits++;
     }
     gx_wave_count_add(hits, local_hits);
 }
@@ -619,15 +667,16 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
 
 /* received qualifying orders → build the join/agg table */
 __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
-                                  unsigned long long *tkey, int32_t *tdate,
-                                  int32_t *tprio, uint64_t tmask)
+                                  unsigned long long *tkey, uint16_t *ttag,
+                                  int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < n; i += stride)
     {
         uint64_t k = (uint64_t) rows[i].okey;
-        uint64_t slot = gx_hmix64(k) & tmask;
+        uint64_t h = gx_hmix64(k);
+        uint64_t slot = h & tmask;
         while (true)
         {
             unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
@@ -635,6 +684,7 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
             {
                 tdate[slot] = rows[i].odate;
                 tprio[slot] = rows[i].oprio;
+                ttag[slot] = gx_tag(h);
                 break;
             }
             if (prev == (unsigned long long) k) break;
@@ -666,6 +716,7 @@ struct gx_q3 {
     unsigned long long *cset = nullptr;
     uint64_t cmask = 0;
     unsigned long long *tkey = nullptr;
+    uint16_t *ttag = nullptr;
     int32_t *tdate = nullptr, *tprio = nullptr;
     double *trev = nullptr;
     unsigned long long *tcnt = nullptr;
@@ -1015,7 +1066,7 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
 static void q3_free_runstate(gx_q3 *q)
 {
     auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
-    fr(q->cset); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+    fr(q->cset); fr(q->tkey); fr(q->ttag); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits);
     q->sized = false;
@@ -1066,6 +1117,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
         q->tmask = tslots - 1;
         HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->ttag, tslots * 2));
         HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
@@ -1118,12 +1170,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         uint64_t tslots = q->tmask + 1;
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->ttag, 0, tslots * 2, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
         hipLaunchKernelGGL(k_orders_build, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                            op.dstream, op.m, q->cutoff, q->cset, q->cmask,
-                           q->tkey, q->tdate, q->tprio, q->tmask);
+                           q->tkey, q->ttag, q->tdate, q->tprio, q->tmask);
         qual = q->rescap;
     }
     else
@@ -1226,9 +1279,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->tkey == nullptr || tslots > q->tmask + 1)
         {
             auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
-            fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+            fr(q->tkey); fr(q->ttag); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
             HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
+            HIP_CHK(ctx, hipMalloc(&q->ttag, tslots * 2));
             HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
@@ -1242,10 +1296,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             q->tmask = tslots - 1;
         }
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->ttag, 0, (q->tmask + 1) * 2, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, (q->tmask + 1) * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, (q->tmask + 1) * 8, s));
         hipLaunchKernelGGL(k_build_from_rows, dim3(GRID), dim3(TPB), 0, s,
-                           recv2, qual, q->tkey, q->tdate, q->tprio, q->tmask);
+                           recv2, qual, q->tkey, q->ttag, q->tdate, q->tprio, q->tmask);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0;
@@ -1263,7 +1318,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
     hipLaunchKernelGGL(k_li_probe_agg, dim3(GRID), dim3(TPB), 0, s,
                        lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                       ls.dstream, ls.m, q->cutoff, q->tkey, 0,
+                       ls.dstream, ls.m, q->cutoff, q->tkey, q->ttag,
                        q->trev, q->tcnt, q->tmask, dhits);
     HIP_CHK(ctx, hipEventRecord(ev[3], s));
 
