@@ -428,8 +428,6 @@ __device__ __forceinline__ uint16_t gx_tag(uint64_t h)
 /* orders local path: build the join/agg table keyed by o_orderkey.
  * (ExecHashTableInsert nodeHash.c:1886; o_orderkey unique → 1 entry/key;
  *  payload doubles as the agg group state, nodeAgg.c group = join row) */
-static constexpr int OB = 4;     /* rows per thread per iteration (ILP batch) */
-
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
@@ -439,41 +437,28 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                unsigned long long *tkey, uint16_t *ttag,
                                int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
-    int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * OB;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x * OB;
-    for (; base < ok_m.nrows; base += stride)
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < ok_m.nrows; i += stride)
     {
-        int nb = (int) min((int64_t) OB, ok_m.nrows - base);
-        int32_t od[OB];
-        int64_t ocust[OB];
-#pragma unroll
-        for (int b = 0; b < OB; b++)
-            if (b < nb)
-            {
-                od[b] = gx_col_get<int32_t>(od_s, od_m, base + b);
-                ocust[b] = gx_col_get<int64_t>(oc_s, oc_m, base + b);
-            }
-#pragma unroll
-        for (int b = 0; b < OB; b++)
+        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
+        if (!(od < cutoff)) continue;
+        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
+        uint64_t h = gx_hmix64(k);
+        uint64_t slot = h & tmask;
+        while (true)
         {
-            if (b >= nb || !(od[b] < cutoff)) continue;
-            if (!d_set_contains(cset, cmask, (uint64_t) ocust[b])) continue;
-            uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, base + b);
-            uint64_t h = gx_hmix64(k);
-            uint64_t slot = h & tmask;
-            while (true)
+            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
+            if (prev == 0ULL)
             {
-                unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
-                if (prev == 0ULL)
-                {
-                    tdate[slot] = od[b];
-                    tprio[slot] = gx_col_get<int32_t>(op_s, op_m, base + b);
-                    ttag[slot] = gx_tag(h);
-                    break;
-                }
-                if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
-                slot = (slot + 1) & tmask;
+                tdate[slot] = od;
+                tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
+                ttag[slot] = gx_tag(h);
+                break;
             }
+            if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
+            slot = (slot + 1) & tmask;
         }
     }
 }
@@ -482,60 +467,88 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * scan (aocsam.c:1131 semantics) + probe (nodeHashjoin.c:553-652) +
  * SUM transition (nodeAgg.c:836 + float.c:769) fused; build slots ARE the
  * agg groups (group key functionally determined by l_orderkey). */
-static constexpr int LB = 4;     /* lineitem rows per thread per iteration */
-
-__global__ void k_li_probe_agg(const uint8_t *lk_s, gx_colmeta lk_m,
-                               const uint8_t *pr_s, gx_colmeta pr_m,
-                               const uint8_t *di_s, gx_colmeta di_m,
-                               const uint8_t *sh_s, gx_colmeta sh_m,
-                               int32_t cutoff,
-                               const unsigned long long *tkey,
-                               const uint16_t *ttag,
-                               double *trev, unsigned long long *tcnt,
-                               uint64_t tmask,
-                               unsigned long long *hits)
+/* lineitem probe+agg variants (A/B-able via GX_PROBE_VARIANT):
+ *   B  = rows per thread per iteration (1 = plain grid-stride)
+ *   TAGS = probe the 16-bit tag array before the 8-B key array */
+template <int B, bool TAGS>
+__global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
+                                 const uint8_t *pr_s, gx_colmeta pr_m,
+                                 const uint8_t *di_s, gx_colmeta di_m,
+                                 const uint8_t *sh_s, gx_colmeta sh_m,
+                                 int32_t cutoff,
+                                 const unsigned long long *tkey,
+                                 const uint16_t *ttag,
+                                 double *trev, unsigned long long *tcnt,
+                                 uint64_t tmask,
+                                 unsigned long long *hits)
 {
-    int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * LB;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x * LB;
     unsigned long long local_hits = 0;
-    for (; base < lk_m.nrows; base += stride)
-    {
-        int nb = (int) min((int64_t) LB, lk_m.nrows - base);
-        /* batch the sequential loads — independent, so they overlap (G7);
-         * keys are loaded unconditionally: the passing rows touch nearly
-         * every key cache line anyway at 54% selectivity */
-        int32_t ship[LB];
-        int64_t key[LB];
-#pragma unroll
-        for (int b = 0; b < LB; b++)
-            if (b < nb)
-            {
-                ship[b] = gx_col_get<int32_t>(sh_s, sh_m, base + b);
-                key[b] = gx_col_get<int64_t>(lk_s, lk_m, base + b);
-            }
-#pragma unroll
-        for (int b = 0; b < LB; b++)
+    auto probe = [&](uint64_t k) -> uint64_t {
+        uint64_t h = gx_hmix64(k);
+        uint64_t slot = h & tmask;
+        if constexpr (TAGS)
         {
-            if (b >= nb || !(ship[b] > cutoff)) continue;
-            uint64_t k = (uint64_t) key[b];
-            uint64_t h = gx_hmix64(k);
-            uint64_t slot = h & tmask;
             uint16_t tg = gx_tag(h);
-            bool found = false;
             while (true)
             {
                 uint16_t tt = ttag[slot];
-                if (tt == 0) break;
-                if (tt == tg && tkey[slot] == (unsigned long long) k)
-                { found = true; break; }
+                if (tt == 0) return ~0ULL;
+                if (tt == tg && tkey[slot] == (unsigned long long) k) return slot;
                 slot = (slot + 1) & tmask;
             }
-            if (!found) continue;
-            double price = gx_col_get<double>(pr_s, pr_m, base + b);
-            double disc = gx_col_get<double>(di_s, di_m, base + b);
-            atomicAdd(&trev[slot], price * (1.0 - disc));
-            atomicAdd(&tcnt[slot], 1ULL);
-            local_hits++;
+        }
+        else
+        {
+            while (true)
+            {
+                unsigned long long v = tkey[slot];
+                if (v == 0ULL) return ~0ULL;
+                if (v == (unsigned long long) k) return slot;
+                slot = (slot + 1) & tmask;
+            }
+        }
+    };
+    auto hit = [&](uint64_t slot, int64_t i) {
+        double price = gx_col_get<double>(pr_s, pr_m, i);
+        double disc = gx_col_get<double>(di_s, di_m, i);
+        atomicAdd(&trev[slot], price * (1.0 - disc));
+        atomicAdd(&tcnt[slot], 1ULL);
+        local_hits++;
+    };
+    if constexpr (B == 1)
+    {
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        for (; i < lk_m.nrows; i += stride)
+        {
+            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            uint64_t slot = probe((uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i));
+            if (slot != ~0ULL) hit(slot, i);
+        }
+    }
+    else
+    {
+        int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * B;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x * B;
+        for (; base < lk_m.nrows; base += stride)
+        {
+            int nb = (int) min((int64_t) B, lk_m.nrows - base);
+            int32_t ship[B];
+            int64_t key[B];
+#pragma unroll
+            for (int b = 0; b < B; b++)
+                if (b < nb)
+                {
+                    ship[b] = gx_col_get<int32_t>(sh_s, sh_m, base + b);
+                    key[b] = gx_col_get<int64_t>(lk_s, lk_m, base + b);
+                }
+#pragma unroll
+            for (int b = 0; b < B; b++)
+            {
+                if (b >= nb || !(ship[b] > cutoff)) continue;
+                uint64_t slot = probe((uint64_t) key[b]);
+                if (slot != ~0ULL) hit(slot, base + b);
+            }
         }
     }
     gx_wave_count_add(hits, local_hits);
@@ -1316,10 +1329,26 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     /* ---- stage 3: lineitem scan+probe+agg (dominant kernel) ---- */
     unsigned long long *dhits = q->dhits;
     HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
-    hipLaunchKernelGGL(k_li_probe_agg, dim3(GRID), dim3(TPB), 0, s,
-                       lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                       ls.dstream, ls.m, q->cutoff, q->tkey, q->ttag,
-                       q->trev, q->tcnt, q->tmask, dhits);
+    {
+        const char *pv = getenv("GX_PROBE_VARIANT");
+        int variant = pv ? atoi(pv) : 0;
+        auto launch = [&](auto kern) {
+            hipLaunchKernelGGL(kern, dim3(GRID), dim3(TPB), 0, s,
+                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                               ls.dstream, ls.m, q->cutoff, q->tkey, q->ttag,
+                               q->trev, q->tcnt, q->tmask, dhits);
+        };
+        switch (variant)
+        {
+            default:
+            case 0: launch(k_li_probe_agg_t<1, false>); break;
+            case 1: launch(k_li_probe_agg_t<1, true>); break;
+            case 2: launch(k_li_probe_agg_t<4, true>); break;
+            case 3: launch(k_li_probe_agg_t<4, false>); break;
+            case 4: launch(k_li_probe_agg_t<2, false>); break;
+            case 5: launch(k_li_probe_agg_t<8, false>); break;
+        }
+    }
     HIP_CHK(ctx, hipEventRecord(ev[3], s));
 
     /* ---- stage 4: extract ---- */
