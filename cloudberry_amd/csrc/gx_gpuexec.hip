@@ -399,42 +399,50 @@ __device__ __forceinline__ bool d_set_contains(const unsigned long long *set,
 }
 
 /* orders local path: count qualifying rows (date filter + customer semijoin) */
-__global__ void k_orders_count(const uint8_t *od_s, gx_colmeta od_m,
+__global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
+                               const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                int32_t cutoff,
                                const unsigned long long *cset, uint64_t cmask,
-                               unsigned long long *count)
+                               unsigned long long *count,
+                               unsigned long long *maxkey)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    unsigned long long local = 0;
+    unsigned long long local = 0, kmax = 0;
     for (; i < od_m.nrows; i += stride)
     {
         if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
         if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
         local++;
+        unsigned long long k = (unsigned long long) gx_col_get<int64_t>(ok_s, ok_m, i);
+        if (k > kmax) kmax = k;
     }
     gx_wave_count_add(count, local);
+    for (int o = 32; o; o >>= 1)
+    {
+        unsigned long long v = __shfl_down(kmax, o, 64);
+        if (v > kmax) kmax = v;
+    }
+    if ((threadIdx.x & 63) == 0 && kmax)
+        atomicMax(maxkey, kmax);
 }
-
-/* 16-bit slot tags: the probe's miss path (the common case) walks a small
- * L3-resident array instead of the 8-B key array.  tag 0 = empty slot; real
- * tags have the high bit forced so they are never 0.  Tags are a FILTER:
- * a tag match still verifies the full key.  Written by the single winning
- * inserter after its CAS; probe kernels run only after build completes. */
-__device__ __forceinline__ uint16_t gx_tag(uint64_t h)
-{ return (uint16_t) (h >> 48) | 0x8000; }
 
 /* orders local path: build the join/agg table keyed by o_orderkey.
  * (ExecHashTableInsert nodeHash.c:1886; o_orderkey unique → 1 entry/key;
  *  payload doubles as the agg group state, nodeAgg.c group = join row) */
+/* KT = u32 when every qualifying o_orderkey < 2^32 (detected at sizing; the
+ * key array then sits comfortably in the 256 MiB Infinity Cache), u64
+ * otherwise.  Key compares stay exact either way (PG narrow-int hashing
+ * spirit; sentinel 0 is safe — orderkeys start at 1). */
+template <typename KT>
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
                                int32_t cutoff,
                                const unsigned long long *cset, uint64_t cmask,
-                               unsigned long long *tkey, uint16_t *ttag,
+                               KT *tkey,
                                int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
@@ -445,19 +453,17 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
         if (!(od < cutoff)) continue;
         if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
-        uint64_t h = gx_hmix64(k);
-        uint64_t slot = h & tmask;
+        uint64_t slot = gx_hmix64(k) & tmask;
         while (true)
         {
-            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
-            if (prev == 0ULL)
+            KT prev = atomicCAS(&tkey[slot], (KT) 0, (KT) k);
+            if (prev == (KT) 0)
             {
                 tdate[slot] = od;
                 tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
-                ttag[slot] = gx_tag(h);
                 break;
             }
-            if (prev == (unsigned long long) k) break;   /* unique keys: no-op */
+            if (prev == (KT) k) break;   /* unique keys: no-op */
             slot = (slot + 1) & tmask;
         }
     }
@@ -467,45 +473,29 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * scan (aocsam.c:1131 semantics) + probe (nodeHashjoin.c:553-652) +
  * SUM transition (nodeAgg.c:836 + float.c:769) fused; build slots ARE the
  * agg groups (group key functionally determined by l_orderkey). */
-/* lineitem probe+agg variants (A/B-able via GX_PROBE_VARIANT):
- *   B  = rows per thread per iteration (1 = plain grid-stride)
- *   TAGS = probe the 16-bit tag array before the 8-B key array */
-template <int B, bool TAGS>
+/* lineitem probe+agg (the dominant kernel).  B = rows per thread per
+ * iteration (A/B via GX_PROBE_VARIANT; measured best B=1 on gfx950), KT =
+ * table key type (u32 when the build side's keys fit — see k_orders_build). */
+template <int B, typename KT>
 __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
                                  const uint8_t *di_s, gx_colmeta di_m,
                                  const uint8_t *sh_s, gx_colmeta sh_m,
                                  int32_t cutoff,
-                                 const unsigned long long *tkey,
-                                 const uint16_t *ttag,
+                                 const KT *tkey,
                                  double *trev, unsigned long long *tcnt,
                                  uint64_t tmask,
                                  unsigned long long *hits)
 {
     unsigned long long local_hits = 0;
     auto probe = [&](uint64_t k) -> uint64_t {
-        uint64_t h = gx_hmix64(k);
-        uint64_t slot = h & tmask;
-        if constexpr (TAGS)
+        uint64_t slot = gx_hmix64(k) & tmask;
+        while (true)
         {
-            uint16_t tg = gx_tag(h);
-            while (true)
-            {
-                uint16_t tt = ttag[slot];
-                if (tt == 0) return ~0ULL;
-                if (tt == tg && tkey[slot] == (unsigned long long) k) return slot;
-                slot = (slot + 1) & tmask;
-            }
-        }
-        else
-        {
-            while (true)
-            {
-                unsigned long long v = tkey[slot];
-                if (v == 0ULL) return ~0ULL;
-                if (v == (unsigned long long) k) return slot;
-                slot = (slot + 1) & tmask;
-            }
+            KT v = tkey[slot];
+            if (v == (KT) 0) return ~0ULL;
+            if (v == (KT) k) return slot;
+            slot = (slot + 1) & tmask;
         }
     };
     auto hit = [&](uint64_t slot, int64_t i) {
@@ -559,7 +549,8 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
  * range, counts its keeps, claims an output region with ONE atomic, then
  * writes (cdna_hip_programming.md G12 — a single shared cursor serializes;
  * the first version lost 6 ms to ~500k same-address atomics). */
-__global__ void k_extract(const unsigned long long *tkey, const int32_t *tdate,
+template <typename KT>
+__global__ void k_extract(const KT *tkey, const int32_t *tdate,
                           const int32_t *tprio, const double *trev,
                           const unsigned long long *tcnt, uint64_t tslots,
                           int64_t *okey, int32_t *odate, int32_t *oprio,
@@ -575,7 +566,7 @@ __global__ void k_extract(const unsigned long long *tkey, const int32_t *tdate,
     /* pass 1: count my keeps (thread-strided over the block's range) */
     unsigned int mine = 0;
     for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
-        if (tkey[i] != 0ULL && tcnt[i] != 0ULL) mine++;
+        if (tkey[i] != (KT) 0 && tcnt[i] != 0ULL) mine++;
     scan[threadIdx.x] = mine;
     __syncthreads();
     /* exclusive scan of 256 per-thread counts (Hillis-Steele in LDS) */
@@ -594,7 +585,7 @@ __global__ void k_extract(const unsigned long long *tkey, const int32_t *tdate,
     /* pass 2: write at my claimed positions */
     for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
     {
-        if (tkey[i] == 0ULL || tcnt[i] == 0ULL) continue;
+        if (tkey[i] == (KT) 0 || tcnt[i] == 0ULL) continue;
         okey[w] = (int64_t) tkey[i];
         odate[w] = tdate[i];
         oprio[w] = tprio[i];
@@ -679,8 +670,9 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
 }
 
 /* received qualifying orders → build the join/agg table */
+template <typename KT>
 __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
-                                  unsigned long long *tkey, uint16_t *ttag,
+                                  KT *tkey,
                                   int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
@@ -688,19 +680,17 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
     for (; i < n; i += stride)
     {
         uint64_t k = (uint64_t) rows[i].okey;
-        uint64_t h = gx_hmix64(k);
-        uint64_t slot = h & tmask;
+        uint64_t slot = gx_hmix64(k) & tmask;
         while (true)
         {
-            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, (unsigned long long) k);
-            if (prev == 0ULL)
+            KT prev = atomicCAS(&tkey[slot], (KT) 0, (KT) k);
+            if (prev == (KT) 0)
             {
                 tdate[slot] = rows[i].odate;
                 tprio[slot] = rows[i].oprio;
-                ttag[slot] = gx_tag(h);
                 break;
             }
-            if (prev == (unsigned long long) k) break;
+            if (prev == (KT) k) break;
             slot = (slot + 1) & tmask;
         }
     }
@@ -728,8 +718,8 @@ struct gx_q3 {
     bool sized = false;
     unsigned long long *cset = nullptr;
     uint64_t cmask = 0;
-    unsigned long long *tkey = nullptr;
-    uint16_t *ttag = nullptr;
+    void *tkey = nullptr;            /* u32 or u64 slots, see key_width */
+    int key_width = 8;
     int32_t *tdate = nullptr, *tprio = nullptr;
     double *trev = nullptr;
     unsigned long long *tcnt = nullptr;
@@ -1079,7 +1069,7 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
 static void q3_free_runstate(gx_q3 *q)
 {
     auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
-    fr(q->cset); fr(q->tkey); fr(q->ttag); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+    fr(q->cset); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits);
     q->sized = false;
@@ -1120,17 +1110,20 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     if (ctx->nsegs == 1)
     {
         HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for maxkey */
         hipLaunchKernelGGL(k_orders_count, dim3(GRID), dim3(TPB), 0, s,
+                           q->ord->cols[0].dstream, q->ord->cols[0].m,
                            od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
-                           q->cset, q->cmask, q->dcount);
-        unsigned long long nq = 0;
+                           q->cset, q->cmask, q->dcount, q->dhits);
+        unsigned long long nq = 0, kmax = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&nq, q->dcount, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipMemcpyAsync(&kmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         qual = (int64_t) nq;
+        q->key_width = (kmax < (1ULL << 32)) ? 4 : 8;
         uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
         q->tmask = tslots - 1;
-        HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
-        HIP_CHK(ctx, hipMalloc(&q->ttag, tslots * 2));
+        HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * q->key_width));
         HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
@@ -1182,14 +1175,19 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     if (ctx->nsegs == 1)
     {
         uint64_t tslots = q->tmask + 1;
-        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * 8, s));
-        HIP_CHK(ctx, hipMemsetAsync(q->ttag, 0, tslots * 2, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
-        hipLaunchKernelGGL(k_orders_build, dim3(GRID), dim3(TPB), 0, s,
-                           ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                           op.dstream, op.m, q->cutoff, q->cset, q->cmask,
-                           q->tkey, q->ttag, q->tdate, q->tprio, q->tmask);
+        if (q->key_width == 4)
+            hipLaunchKernelGGL(k_orders_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                               op.dstream, op.m, q->cutoff, q->cset, q->cmask,
+                               (unsigned int *) q->tkey, q->tdate, q->tprio, q->tmask);
+        else
+            hipLaunchKernelGGL(k_orders_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                               op.dstream, op.m, q->cutoff, q->cset, q->cmask,
+                               (unsigned long long *) q->tkey, q->tdate, q->tprio, q->tmask);
         qual = q->rescap;
     }
     else
@@ -1292,10 +1290,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->tkey == nullptr || tslots > q->tmask + 1)
         {
             auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
-            fr(q->tkey); fr(q->ttag); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+            fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
+            q->key_width = 8;
             HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
-            HIP_CHK(ctx, hipMalloc(&q->ttag, tslots * 2));
             HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
@@ -1309,11 +1307,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             q->tmask = tslots - 1;
         }
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * 8, s));
-        HIP_CHK(ctx, hipMemsetAsync(q->ttag, 0, (q->tmask + 1) * 2, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, (q->tmask + 1) * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, (q->tmask + 1) * 8, s));
-        hipLaunchKernelGGL(k_build_from_rows, dim3(GRID), dim3(TPB), 0, s,
-                           recv2, qual, q->tkey, q->ttag, q->tdate, q->tprio, q->tmask);
+        hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                           recv2, qual, (unsigned long long *) q->tkey,
+                           q->tdate, q->tprio, q->tmask);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0;
@@ -1332,30 +1330,49 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
-        auto launch = [&](auto kern) {
+        auto launch = [&](auto kern, auto *keys) {
             hipLaunchKernelGGL(kern, dim3(GRID), dim3(TPB), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, q->cutoff, q->tkey, q->ttag,
+                               ls.dstream, ls.m, q->cutoff, keys,
                                q->trev, q->tcnt, q->tmask, dhits);
         };
-        switch (variant)
+        if (q->key_width == 4)
         {
-            default:
-            case 0: launch(k_li_probe_agg_t<1, false>); break;
-            case 1: launch(k_li_probe_agg_t<1, true>); break;
-            case 2: launch(k_li_probe_agg_t<4, true>); break;
-            case 3: launch(k_li_probe_agg_t<4, false>); break;
-            case 4: launch(k_li_probe_agg_t<2, false>); break;
-            case 5: launch(k_li_probe_agg_t<8, false>); break;
+            auto *keys = (const unsigned int *) q->tkey;
+            switch (variant)
+            {
+                default:
+                case 0: launch(k_li_probe_agg_t<1, unsigned int>, keys); break;
+                case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
+                case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
+            }
+        }
+        else
+        {
+            auto *keys = (const unsigned long long *) q->tkey;
+            switch (variant)
+            {
+                default:
+                case 0: launch(k_li_probe_agg_t<1, unsigned long long>, keys); break;
+                case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
+                case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
+            }
         }
     }
     HIP_CHK(ctx, hipEventRecord(ev[3], s));
 
     /* ---- stage 4: extract ---- */
     HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
-    hipLaunchKernelGGL(k_extract, dim3(GRID), dim3(TPB), 0, s,
-                       q->tkey, q->tdate, q->tprio, q->trev, q->tcnt, q->tmask + 1,
-                       q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
+    if (q->key_width == 4)
+        hipLaunchKernelGGL(k_extract<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                           (const unsigned int *) q->tkey, q->tdate, q->tprio,
+                           q->trev, q->tcnt, q->tmask + 1,
+                           q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
+    else
+        hipLaunchKernelGGL(k_extract<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                           (const unsigned long long *) q->tkey, q->tdate, q->tprio,
+                           q->trev, q->tcnt, q->tmask + 1,
+                           q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
     HIP_CHK(ctx, hipEventRecord(ev[4], s));
 
     unsigned long long ngroups = 0, hits = 0;
