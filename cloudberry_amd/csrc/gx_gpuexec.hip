@@ -474,8 +474,11 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * SUM transition (nodeAgg.c:836 + float.c:769) fused; build slots ARE the
  * agg groups (group key functionally determined by l_orderkey). */
 /* lineitem probe+agg (the dominant kernel).  B = rows per thread per
- * iteration (A/B via GX_PROBE_VARIANT; measured best B=1 on gfx950), KT =
- * table key type (u32 when the build side's keys fit — see k_orders_build). */
+ * iteration; the batched form keeps B independent load chains in flight
+ * (the kernel is latency-bound at full occupancy: PMC shows 85% WAIT_ANY,
+ * ~300M L3 probe round-trips).  Guard-free main region — the tail is a
+ * separate scalar loop — and filtered lanes probe slot 0 (L1-resident)
+ * instead of branching, so each load batch stays in one basic block. */
 template <int B, typename KT>
 __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
@@ -488,14 +491,14 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  unsigned long long *hits)
 {
     unsigned long long local_hits = 0;
-    auto probe = [&](uint64_t k) -> uint64_t {
-        uint64_t slot = gx_hmix64(k) & tmask;
+    auto resolve = [&](uint64_t k, uint64_t slot, KT v) -> uint64_t {
+        /* first slot already loaded as v; walk on collision */
         while (true)
         {
-            KT v = tkey[slot];
             if (v == (KT) 0) return ~0ULL;
             if (v == (KT) k) return slot;
             slot = (slot + 1) & tmask;
+            v = tkey[slot];
         }
     };
     auto hit = [&](uint64_t slot, int64_t i) {
@@ -512,33 +515,54 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         for (; i < lk_m.nrows; i += stride)
         {
             if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
-            uint64_t slot = probe((uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i));
-            if (slot != ~0ULL) hit(slot, i);
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = gx_hmix64(k) & tmask;
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r != ~0ULL) hit(r, i);
         }
     }
     else
     {
         int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * B;
         int64_t stride = gridDim.x * (int64_t) blockDim.x * B;
-        for (; base < lk_m.nrows; base += stride)
+        for (; base + B <= lk_m.nrows; base += stride)
         {
-            int nb = (int) min((int64_t) B, lk_m.nrows - base);
             int32_t ship[B];
             int64_t key[B];
 #pragma unroll
             for (int b = 0; b < B; b++)
-                if (b < nb)
-                {
-                    ship[b] = gx_col_get<int32_t>(sh_s, sh_m, base + b);
-                    key[b] = gx_col_get<int64_t>(lk_s, lk_m, base + b);
-                }
+                ship[b] = gx_col_get<int32_t>(sh_s, sh_m, base + b);
+#pragma unroll
+            for (int b = 0; b < B; b++)
+                key[b] = gx_col_get<int64_t>(lk_s, lk_m, base + b);
+            bool pass[B];
+            uint64_t slot[B];
+            KT v[B];
 #pragma unroll
             for (int b = 0; b < B; b++)
             {
-                if (b >= nb || !(ship[b] > cutoff)) continue;
-                uint64_t slot = probe((uint64_t) key[b]);
-                if (slot != ~0ULL) hit(slot, base + b);
+                pass[b] = ship[b] > cutoff;
+                slot[b] = pass[b] ? (gx_hmix64((uint64_t) key[b]) & tmask) : 0;
             }
+#pragma unroll
+            for (int b = 0; b < B; b++)
+                v[b] = tkey[slot[b]];        /* filtered lanes hit slot 0 in L1 */
+#pragma unroll
+            for (int b = 0; b < B; b++)
+            {
+                if (!pass[b]) continue;
+                uint64_t r = resolve((uint64_t) key[b], slot[b], v[b]);
+                if (r != ~0ULL) hit(r, base + b);
+            }
+        }
+        /* tail rows (at most B-1 per thread, only near nrows) */
+        for (int64_t i = base; i < lk_m.nrows && i < base + B; i++)
+        {
+            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot0 = gx_hmix64(k) & tmask;
+            uint64_t r = resolve(k, slot0, tkey[slot0]);
+            if (r != ~0ULL) hit(r, i);
         }
     }
     gx_wave_count_add(hits, local_hits);
@@ -1345,6 +1369,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 0: launch(k_li_probe_agg_t<1, unsigned int>, keys); break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
+                case 5: launch(k_li_probe_agg_t<8, unsigned int>, keys); break;
             }
         }
         else
@@ -1356,6 +1381,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 0: launch(k_li_probe_agg_t<1, unsigned long long>, keys); break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
+                case 5: launch(k_li_probe_agg_t<8, unsigned long long>, keys); break;
             }
         }
     }
