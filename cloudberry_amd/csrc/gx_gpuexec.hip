@@ -352,22 +352,37 @@ __device__ __forceinline__ void gx_wave_count_add(unsigned long long *dst,
 }
 
 /* customer: count BUILDING rows (for set sizing) */
-__global__ void k_cust_count(const uint8_t *mkt_s, gx_colmeta mkt_m,
-                             unsigned long long *count)
+__global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
+                             const uint8_t *mkt_s, gx_colmeta mkt_m,
+                             unsigned long long *count,
+                             unsigned long long *maxkey)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    unsigned long long local = 0;
+    unsigned long long local = 0, kmax = 0;
     for (; i < mkt_m.nrows; i += stride)
-        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) == 0) local++;
+        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) == 0)
+        {
+            local++;
+            unsigned long long k = (unsigned long long) gx_col_get<int64_t>(key_s, key_m, i);
+            if (k > kmax) kmax = k;
+        }
     gx_wave_count_add(count, local);
+    for (int o = 32; o; o >>= 1)
+    {
+        unsigned long long v = __shfl_down(kmax, o, 64);
+        if (v > kmax) kmax = v;
+    }
+    if ((threadIdx.x & 63) == 0 && kmax)
+        atomicMax(maxkey, kmax);
 }
 
 /* customer: filter mktsegment=BUILDING, insert c_custkey into open set.
  * Replaces the build side of the cust⋈orders join (nodeHash.c:1886). */
+template <typename KS>
 __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
-                             unsigned long long *set, uint64_t mask)
+                             KS *set, uint64_t mask)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
@@ -378,32 +393,34 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
         uint64_t slot = gx_hmix64(k) & mask;
         while (true)
         {
-            unsigned long long prev = atomicCAS(&set[slot], 0ULL, (unsigned long long) k);
-            if (prev == 0ULL || prev == (unsigned long long) k) break;
+            KS prev = atomicCAS(&set[slot], (KS) 0, (KS) k);
+            if (prev == (KS) 0 || prev == (KS) k) break;
             slot = (slot + 1) & mask;
         }
     }
 }
 
-__device__ __forceinline__ bool d_set_contains(const unsigned long long *set,
+template <typename KS>
+__device__ __forceinline__ bool d_set_contains(const KS *set,
                                                uint64_t mask, uint64_t k)
 {
     uint64_t slot = gx_hmix64(k) & mask;
     while (true)
     {
-        unsigned long long v = set[slot];
-        if (v == 0ULL) return false;
-        if (v == (unsigned long long) k) return true;
+        KS v = set[slot];
+        if (v == (KS) 0) return false;
+        if (v == (KS) k) return true;
         slot = (slot + 1) & mask;
     }
 }
 
 /* orders local path: count qualifying rows (date filter + customer semijoin) */
+template <typename KS>
 __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                int32_t cutoff,
-                               const unsigned long long *cset, uint64_t cmask,
+                               const KS *cset, uint64_t cmask,
                                unsigned long long *count,
                                unsigned long long *maxkey)
 {
@@ -435,13 +452,13 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
  * key array then sits comfortably in the 256 MiB Infinity Cache), u64
  * otherwise.  Key compares stay exact either way (PG narrow-int hashing
  * spirit; sentinel 0 is safe — orderkeys start at 1). */
-template <typename KT>
+template <typename KT, typename KS>
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
                                int32_t cutoff,
-                               const unsigned long long *cset, uint64_t cmask,
+                               const KS *cset, uint64_t cmask,
                                KT *tkey,
                                int32_t *tdate, int32_t *tprio, uint64_t tmask)
 {
@@ -662,8 +679,9 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
 }
 
 /* received orders rows: probe local customer set, histogram by route(okey) */
+template <typename KS>
 __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
-                            const unsigned long long *cset, uint64_t cmask,
+                            const KS *cset, uint64_t cmask,
                             int nsegs, unsigned long long *hist)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
@@ -675,8 +693,9 @@ __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
     }
 }
 
+template <typename KS>
 __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
-                            const unsigned long long *cset, uint64_t cmask,
+                            const KS *cset, uint64_t cmask,
                             int nsegs, unsigned long long *cursors,
                             gx_qual_row *out)
 {
@@ -740,7 +759,8 @@ struct gx_q3 {
     /* run state (device) — allocated on first run, reused across steps
      * (a re-run rebuilds every table; only the ALLOCATIONS persist) */
     bool sized = false;
-    unsigned long long *cset = nullptr;
+    void *cset = nullptr;            /* u32 or u64 slots, see cset_width */
+    int cset_width = 8;
     uint64_t cmask = 0;
     void *tkey = nullptr;            /* u32 or u64 slots, see key_width */
     int key_width = 8;
@@ -1112,18 +1132,27 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMalloc(&q->dhits, 8));
 
     HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for max custkey */
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                       cm.dstream, cm.m, q->dcount);
-    unsigned long long n_building = 0;
-    HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
-    HIP_CHK(ctx, hipStreamSynchronize(s));
-    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
-    HIP_CHK(ctx, hipMalloc(&q->cset, cslots * 8));
-    HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * 8, s));
-    q->cmask = cslots - 1;
-    hipLaunchKernelGGL(k_cust_build, dim3(GRID), dim3(TPB), 0, s,
                        q->cust->cols[0].dstream, q->cust->cols[0].m,
-                       cm.dstream, cm.m, q->cset, q->cmask);
+                       cm.dstream, cm.m, q->dcount, q->dhits);
+    unsigned long long n_building = 0, cmax = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
+    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
+    HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
+    HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * q->cset_width, s));
+    q->cmask = cslots - 1;
+    if (q->cset_width == 4)
+        hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                           q->cust->cols[0].dstream, q->cust->cols[0].m,
+                           cm.dstream, cm.m, (unsigned int *) q->cset, q->cmask);
+    else
+        hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                           q->cust->cols[0].dstream, q->cust->cols[0].m,
+                           cm.dstream, cm.m, (unsigned long long *) q->cset, q->cmask);
 
     /* local qualifying-order count bounds the table for BOTH paths: at
      * nsegs>1 the table holds rows received for THIS segment; the global
@@ -1135,10 +1164,16 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     {
         HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for maxkey */
-        hipLaunchKernelGGL(k_orders_count, dim3(GRID), dim3(TPB), 0, s,
-                           q->ord->cols[0].dstream, q->ord->cols[0].m,
-                           od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
-                           q->cset, q->cmask, q->dcount, q->dhits);
+        if (q->cset_width == 4)
+            hipLaunchKernelGGL(k_orders_count<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               q->ord->cols[0].dstream, q->ord->cols[0].m,
+                               od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
+                               (const unsigned int *) q->cset, q->cmask, q->dcount, q->dhits);
+        else
+            hipLaunchKernelGGL(k_orders_count<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               q->ord->cols[0].dstream, q->ord->cols[0].m,
+                               od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
+                               (const unsigned long long *) q->cset, q->cmask, q->dcount, q->dhits);
         unsigned long long nq = 0, kmax = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&nq, q->dcount, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipMemcpyAsync(&kmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
@@ -1188,9 +1223,15 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
     /* ---- stage 1: customer BUILDING set (rebuilt every run) ---- */
     HIP_CHK(ctx, hipEventRecord(ev[0], s));
-    HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, (q->cmask + 1) * 8, s));
-    hipLaunchKernelGGL(k_cust_build, dim3(GRID), dim3(TPB), 0, s,
-                       ck.dstream, ck.m, cm.dstream, cm.m, q->cset, q->cmask);
+    HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, (q->cmask + 1) * q->cset_width, s));
+    if (q->cset_width == 4)
+        hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                           ck.dstream, ck.m, cm.dstream, cm.m,
+                           (unsigned int *) q->cset, q->cmask);
+    else
+        hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                           ck.dstream, ck.m, cm.dstream, cm.m,
+                           (unsigned long long *) q->cset, q->cmask);
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
 
     /* ---- stage 2: orders build (local or via Motions) ---- */
@@ -1202,16 +1243,22 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
-        if (q->key_width == 4)
-            hipLaunchKernelGGL(k_orders_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+        auto launch_build = [&](auto *tk, auto *cs) {
+            hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
+                                               std::decay_t<decltype(*cs)>>),
+                               dim3(GRID), dim3(TPB), 0, s,
                                ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                               op.dstream, op.m, q->cutoff, q->cset, q->cmask,
-                               (unsigned int *) q->tkey, q->tdate, q->tprio, q->tmask);
+                               op.dstream, op.m, q->cutoff, cs, q->cmask,
+                               tk, q->tdate, q->tprio, q->tmask);
+        };
+        if (q->key_width == 4 && q->cset_width == 4)
+            launch_build((unsigned int *) q->tkey, (const unsigned int *) q->cset);
+        else if (q->key_width == 4)
+            launch_build((unsigned int *) q->tkey, (const unsigned long long *) q->cset);
+        else if (q->cset_width == 4)
+            launch_build((unsigned long long *) q->tkey, (const unsigned int *) q->cset);
         else
-            hipLaunchKernelGGL(k_orders_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                               op.dstream, op.m, q->cutoff, q->cset, q->cmask,
-                               (unsigned long long *) q->tkey, q->tdate, q->tprio, q->tmask);
+            launch_build((unsigned long long *) q->tkey, (const unsigned long long *) q->cset);
         qual = q->rescap;
     }
     else
@@ -1273,8 +1320,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
         /* Motion 2: probe local customer set, route qualifying by o_orderkey */
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
-        hipLaunchKernelGGL(k_qual_hist, dim3(GRID), dim3(TPB), 0, s,
-                           recv1, (int64_t) recv1_n, q->cset, q->cmask, n, dhist);
+        if (q->cset_width == 4)
+            hipLaunchKernelGGL(k_qual_hist<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
+                               q->cmask, n, dhist);
+        else
+            hipLaunchKernelGGL(k_qual_hist<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
+                               q->cmask, n, dhist);
         std::vector<unsigned long long> h2(n);
         HIP_CHK(ctx, hipMemcpyAsync(h2.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -1283,8 +1336,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         gx_qual_row *send2 = nullptr;
         HIP_CHK(ctx, hipMalloc(&send2, std::max<uint64_t>(off2[n], 1) * sizeof(gx_qual_row)));
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off2.data(), n * 8, hipMemcpyHostToDevice, s));
-        hipLaunchKernelGGL(k_qual_emit, dim3(GRID), dim3(TPB), 0, s,
-                           recv1, (int64_t) recv1_n, q->cset, q->cmask, n, dcur, send2);
+        if (q->cset_width == 4)
+            hipLaunchKernelGGL(k_qual_emit<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
+                               q->cmask, n, dcur, send2);
+        else
+            hipLaunchKernelGGL(k_qual_emit<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
+                               q->cmask, n, dcur, send2);
         HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h2.data(), n * 8, hipMemcpyHostToDevice, s));
         RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
         HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
@@ -1354,8 +1413,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
+        const char *pg = getenv("GX_PROBE_GRID");
+        int pgrid = pg ? atoi(pg) : GRID;
+        const char *pt = getenv("GX_PROBE_TPB");
+        int ptpb = pt ? atoi(pt) : TPB;
         auto launch = [&](auto kern, auto *keys) {
-            hipLaunchKernelGGL(kern, dim3(GRID), dim3(TPB), 0, s,
+            hipLaunchKernelGGL(kern, dim3(pgrid), dim3(ptpb), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
                                ls.dstream, ls.m, q->cutoff, keys,
                                q->trev, q->tcnt, q->tmask, dhits);
