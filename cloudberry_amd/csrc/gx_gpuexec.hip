@@ -362,6 +362,33 @@ struct gx_slotmap {
     }
 };
 
+/* Blocked bloom filter over the customer build keys — the GPU counterpart
+ * of the reference's runtime filter pushdown (nodeRuntimeFilter.c,
+ * CreateRuntimeFilter nodeHashjoin.c:2297, consumed nodeSeqscan.c:83-116):
+ * one 64-bit word per key carries two bits, so a negative costs a single
+ * L2-resident load instead of an open-addressing walk. */
+__device__ __forceinline__ unsigned long long d_bloom_mask_of(uint64_t h)
+{
+    int b1 = (int) ((h >> 32) & 63);
+    int b2 = (int) ((h >> 38) & 63);
+    return (1ULL << b1) | (1ULL << b2);
+}
+
+__device__ __forceinline__ void d_bloom_set(unsigned long long *bloom,
+                                            uint64_t wmask, uint64_t k)
+{
+    uint64_t h = gx_hmix64(k);
+    atomicOr(&bloom[h & wmask], d_bloom_mask_of(h));
+}
+
+__device__ __forceinline__ bool d_bloom_test(const unsigned long long *bloom,
+                                             uint64_t wmask, uint64_t k)
+{
+    uint64_t h = gx_hmix64(k);
+    unsigned long long m = d_bloom_mask_of(h);
+    return (bloom[h & wmask] & m) == m;
+}
+
 /* wave-aggregated counter add: ONE atomic per 64-lane wave (G12) */
 __device__ __forceinline__ void gx_wave_count_add(unsigned long long *dst,
                                                   unsigned long long v)
@@ -403,7 +430,8 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
 template <typename KS>
 __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
-                             KS *set, uint64_t mask)
+                             KS *set, uint64_t mask,
+                             unsigned long long *bloom, uint64_t bwmask)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
@@ -411,6 +439,7 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
     {
         if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) != 0) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
+        d_bloom_set(bloom, bwmask, k);
         uint64_t slot = gx_hmix64(k) & mask;
         while (true)
         {
@@ -442,6 +471,7 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                int32_t cutoff,
                                const KS *cset, uint64_t cmask,
+                               const unsigned long long *bloom, uint64_t bwmask,
                                unsigned long long *count,
                                unsigned long long *maxkey,
                                unsigned long long *minkey)
@@ -452,7 +482,9 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
     for (; i < od_m.nrows; i += stride)
     {
         if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
+        if (!d_bloom_test(bloom, bwmask, ck)) continue;
+        if (!d_set_contains(cset, cmask, ck)) continue;
         local++;
         unsigned long long k = (unsigned long long) gx_col_get<int64_t>(ok_s, ok_m, i);
         if (k > kmax) kmax = k;
@@ -487,6 +519,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *op_s, gx_colmeta op_m,
                                int32_t cutoff,
                                const KS *cset, uint64_t cmask,
+                               const unsigned long long *bloom, uint64_t bwmask,
                                KT *tkey,
                                int32_t *tdate, int32_t *tprio, gx_slotmap smap)
 {
@@ -497,7 +530,9 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
     {
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!(od < cutoff)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i))) continue;
+        uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
+        if (!d_bloom_test(bloom, bwmask, ck)) continue;
+        if (!d_set_contains(cset, cmask, ck)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
         uint64_t slot = smap.slot0(k);
         while (true)
@@ -712,12 +747,14 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
 template <typename KS>
 __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
                             const KS *cset, uint64_t cmask,
+                            const unsigned long long *bloom, uint64_t bwmask,
                             int nsegs, unsigned long long *hist)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < n; i += stride)
     {
+        if (!d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust)) continue;
         if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
         atomicAdd(&hist[gx_route_i64(rows[i].okey, nsegs)], 1ULL);
     }
@@ -726,6 +763,7 @@ __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
 template <typename KS>
 __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
                             const KS *cset, uint64_t cmask,
+                            const unsigned long long *bloom, uint64_t bwmask,
                             int nsegs, unsigned long long *cursors,
                             gx_qual_row *out)
 {
@@ -733,6 +771,7 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < n; i += stride)
     {
+        if (!d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust)) continue;
         if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
         int32_t d = gx_route_i64(rows[i].okey, nsegs);
         unsigned long long w = atomicAdd(&cursors[d], 1ULL);
@@ -793,6 +832,8 @@ struct gx_q3 {
     void *cset = nullptr;            /* u32 or u64 slots, see cset_width */
     int cset_width = 8;
     uint64_t cmask = 0;
+    unsigned long long *bloom = nullptr;   /* blocked bloom over cset keys */
+    uint64_t bwmask = 0;                   /* bloom word-index mask */
     void *tkey = nullptr;            /* u32 or u64 slots, see key_width */
     int key_width = 8;
     gx_slotmap smap{};               /* slot mapping (interpolation or hash) */
@@ -1145,7 +1186,7 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
 static void q3_free_runstate(gx_q3 *q)
 {
     auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
-    fr(q->cset); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
+    fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
     q->sized = false;
@@ -1178,14 +1219,21 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * q->cset_width, s));
     q->cmask = cslots - 1;
+    uint64_t bwords = (uint64_t) pow2_at_least(
+        std::max<int64_t>((int64_t) n_building * 16 / 64, 4096));
+    HIP_CHK(ctx, hipMalloc(&q->bloom, bwords * 8));
+    HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, bwords * 8, s));
+    q->bwmask = bwords - 1;
     if (q->cset_width == 4)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[0].dstream, q->cust->cols[0].m,
-                           cm.dstream, cm.m, (unsigned int *) q->cset, q->cmask);
+                           cm.dstream, cm.m, (unsigned int *) q->cset, q->cmask,
+                           q->bloom, q->bwmask);
     else
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[0].dstream, q->cust->cols[0].m,
-                           cm.dstream, cm.m, (unsigned long long *) q->cset, q->cmask);
+                           cm.dstream, cm.m, (unsigned long long *) q->cset, q->cmask,
+                           q->bloom, q->bwmask);
 
     /* local qualifying-order count bounds the table for BOTH paths: at
      * nsegs>1 the table holds rows received for THIS segment; the global
@@ -1202,13 +1250,15 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
             hipLaunchKernelGGL(k_orders_count<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                                q->ord->cols[0].dstream, q->ord->cols[0].m,
                                od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
-                               (const unsigned int *) q->cset, q->cmask, q->dcount,
+                               (const unsigned int *) q->cset, q->cmask,
+                               q->bloom, q->bwmask, q->dcount,
                                q->dhits, q->dmin);
         else
             hipLaunchKernelGGL(k_orders_count<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                q->ord->cols[0].dstream, q->ord->cols[0].m,
                                od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
-                               (const unsigned long long *) q->cset, q->cmask, q->dcount,
+                               (const unsigned long long *) q->cset, q->cmask,
+                               q->bloom, q->bwmask, q->dcount,
                                q->dhits, q->dmin);
         unsigned long long nq = 0, kmax = 0, kmin = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&nq, q->dcount, 8, hipMemcpyDeviceToHost, s));
@@ -1273,14 +1323,15 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     /* ---- stage 1: customer BUILDING set (rebuilt every run) ---- */
     HIP_CHK(ctx, hipEventRecord(ev[0], s));
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, (q->cmask + 1) * q->cset_width, s));
+    HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, (q->bwmask + 1) * 8, s));
     if (q->cset_width == 4)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                            ck.dstream, ck.m, cm.dstream, cm.m,
-                           (unsigned int *) q->cset, q->cmask);
+                           (unsigned int *) q->cset, q->cmask, q->bloom, q->bwmask);
     else
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                            ck.dstream, ck.m, cm.dstream, cm.m,
-                           (unsigned long long *) q->cset, q->cmask);
+                           (unsigned long long *) q->cset, q->cmask, q->bloom, q->bwmask);
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
 
     /* ---- stage 2: orders build (local or via Motions) ---- */
@@ -1298,6 +1349,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                dim3(GRID), dim3(TPB), 0, s,
                                ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                                op.dstream, op.m, q->cutoff, cs, q->cmask,
+                               q->bloom, q->bwmask,
                                tk, q->tdate, q->tprio, q->smap);
         };
         if (q->key_width == 4 && q->cset_width == 4)
@@ -1372,11 +1424,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_qual_hist<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
-                               q->cmask, n, dhist);
+                               q->cmask, q->bloom, q->bwmask, n, dhist);
         else
             hipLaunchKernelGGL(k_qual_hist<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
-                               q->cmask, n, dhist);
+                               q->cmask, q->bloom, q->bwmask, n, dhist);
         std::vector<unsigned long long> h2(n);
         HIP_CHK(ctx, hipMemcpyAsync(h2.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -1388,11 +1440,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_qual_emit<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
-                               q->cmask, n, dcur, send2);
+                               q->cmask, q->bloom, q->bwmask, n, dcur, send2);
         else
             hipLaunchKernelGGL(k_qual_emit<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
-                               q->cmask, n, dcur, send2);
+                               q->cmask, q->bloom, q->bwmask, n, dcur, send2);
         HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h2.data(), n * 8, hipMemcpyHostToDevice, s));
         RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
         HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
