@@ -90,6 +90,9 @@ def _load():
                                  ctypes.POINTER(ctypes.c_int64)]
     lib.gx_q3_free.argtypes = [ctypes.c_void_p]
     lib.gx_free.argtypes = [ctypes.c_void_p]
+    lib.gx_test_motion1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
+                                    ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_int64, ctypes.POINTER(ctypes.c_int64)]
     return lib
 
 
@@ -152,6 +155,18 @@ class Context:
         self._chk(self._lib.gx_partition(self._h, keys.ctypes.data, len(keys),
                                          nsegs, out.ctypes.data))
         return out
+
+    def test_motion1(self, orders, nsegs, cutoff=CUTOFF_19950315):
+        """Run the Motion-1 partition kernels; returns (counts, rows dict)."""
+        cap = orders.nrows
+        counts = np.zeros(nsegs, np.int64)
+        rows = np.zeros(cap, dtype=[("okey", np.int64), ("ocust", np.int64),
+                                    ("odate", np.int32), ("oprio", np.int32)])
+        total = ctypes.c_int64()
+        self._chk(self._lib.gx_test_motion1(self._h, orders._t, cutoff, nsegs,
+                                            counts.ctypes.data, rows.ctypes.data,
+                                            cap, ctypes.byref(total)))
+        return counts, rows[:total.value]
 
     def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315):
         q = ctypes.c_void_p()

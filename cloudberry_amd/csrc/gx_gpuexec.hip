@@ -1655,6 +1655,50 @@ extern "C" gx_status gx_q3_free(gx_q3 *q)
 
 extern "C" void gx_free(void *p) { free(p); }
 
+/* Test ABI: run the Motion-1 partition kernels (filter + route + emit) on a
+ * bound/generated orders table for a given nsegs, returning the packed rows
+ * and per-destination counts to the host.  Lets the Motion path's kernels be
+ * parity-tested on ONE GPU (the RCCL exchange itself is a thin, separately
+ * covered layer).  Mirrors doSendTuple routing (nodeMotion.c:1181). */
+extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
+                                     int32_t cutoff, int nsegs,
+                                     int64_t *out_counts /* nsegs */,
+                                     gx_ord_row *out_rows /* cap */,
+                                     int64_t cap, int64_t *out_total)
+{
+    if (!ctx || !orders || orders->cols.size() != 4) return GX_ERR_INVALID;
+    hipStream_t s = ctx->stream;
+    const gx_col &ok = orders->cols[0], &oc = orders->cols[1],
+                 &od = orders->cols[2], &op = orders->cols[3];
+    unsigned long long *dhist = nullptr, *dcur = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dhist, nsegs * 8));
+    HIP_CHK(ctx, hipMalloc(&dcur, nsegs * 8));
+    HIP_CHK(ctx, hipMemsetAsync(dhist, 0, nsegs * 8, s));
+    hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
+                       od.dstream, od.m, oc.dstream, oc.m, cutoff, nsegs, dhist);
+    std::vector<unsigned long long> h(nsegs);
+    HIP_CHK(ctx, hipMemcpyAsync(h.data(), dhist, nsegs * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    std::vector<unsigned long long> off(nsegs + 1, 0);
+    for (int i = 0; i < nsegs; i++) off[i + 1] = off[i] + h[i];
+    int64_t total = (int64_t) off[nsegs];
+    if (total > cap) { hipFree(dhist); hipFree(dcur); return GX_ERR_INVALID; }
+    gx_ord_row *dsend = nullptr;
+    HIP_CHK(ctx, hipMalloc(&dsend, std::max<int64_t>(total, 1) * sizeof(gx_ord_row)));
+    HIP_CHK(ctx, hipMemcpyAsync(dcur, off.data(), nsegs * 8, hipMemcpyHostToDevice, s));
+    hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
+                       ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                       op.dstream, op.m, cutoff, nsegs, dcur, dsend);
+    HIP_CHK(ctx, hipMemcpyAsync(out_rows, dsend, total * sizeof(gx_ord_row),
+                                hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    for (int i = 0; i < nsegs; i++) out_counts[i] = (int64_t) h[i];
+    *out_total = total;
+    hipFree(dhist); hipFree(dcur); hipFree(dsend);
+    return GX_OK;
+}
+
 /* host-side self-test of the division-free row→block addressing (callable
  * without a GPU; exercised by tests/test_abi_cpu.py) */
 extern "C" int gx_selftest_addressing(void)

@@ -219,12 +219,8 @@ struct gx_colcur {
 
 /* ---------- packed Motion rows ---------- */
 
-struct gx_ord_row {              /* Motion-1 payload: filtered orders */
-    int64_t okey;
-    int64_t ocust;
-    int32_t odate;
-    int32_t oprio;
-};
+/* gx_ord_row (Motion-1 payload) is declared in include/gpuexec.h — it is
+ * also a test-ABI type. */
 
 struct gx_qual_row {             /* Motion-2 payload: qualifying orders */
     int64_t okey;

@@ -148,6 +148,15 @@ gx_status gx_q3_free(gx_q3 *q);
 
 void gx_free(void *p);
 
+/* ---- test-only entry points (parity harness; not part of the drop-in) ---- */
+typedef struct gx_ord_row { int64_t okey, ocust; int32_t odate, oprio; } gx_ord_row;
+/* Motion-1 partition kernels on one GPU: filter orders by cutoff, route by
+ * cdbhash(o_custkey), emit packed rows grouped by destination segment. */
+gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders, int32_t cutoff,
+                          int nsegs, int64_t *out_counts, gx_ord_row *out_rows,
+                          int64_t cap, int64_t *out_total);
+int gx_selftest_addressing(void);
+
 #ifdef __cplusplus
 }
 #endif

@@ -115,6 +115,34 @@ def test_q3_parity_other_seed(ctx, orc):
     _q3_parity_at(ctx, orc, 0.05, seed=7)
 
 
+# ---------------- Motion partition kernels (one GPU) ----------------
+
+def test_motion1_partition_kernels(ctx, orc):
+    """k_ord_m1_hist/emit: every emitted row routed bit-exactly per
+    jump_consistent_hash(cdbhash(o_custkey)); no qualifying row lost."""
+    sf = 0.05
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    o = orc.gen_orders(sf)
+    for nsegs in (2, 8):
+        counts, rows = ctx.test_motion1(ordr, nsegs)
+        qual = o["o_orderdate"] < orc.CUTOFF_19950315
+        assert len(rows) == int(qual.sum())
+        assert counts.sum() == len(rows)
+        # per-destination regions contain exactly the rows routing there
+        want_dest = orc.route(o["o_custkey"][qual], nsegs)
+        got_dest = orc.route(rows["ocust"], nsegs)
+        off = 0
+        for d in range(nsegs):
+            seg_rows = rows[off:off + counts[d]]
+            assert (orc.route(seg_rows["ocust"], nsegs) == d).all()
+            off += counts[d]
+        # same multiset of rows overall
+        assert (np.sort(rows["okey"]) == np.sort(o["o_orderkey"][qual])).all()
+        np.testing.assert_array_equal(np.bincount(got_dest, minlength=nsegs),
+                                      np.bincount(want_dest, minlength=nsegs))
+    ordr.free()
+
+
 # ---------------- edge cases ----------------
 
 def test_q3_empty_result(ctx, orc):
