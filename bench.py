@@ -79,8 +79,10 @@ def main():
         tdist.init_process_group(backend="gloo")  # bootstrap/barriers only;
         dist = tdist                              # data path is our own RCCL comm
 
-    ctx = gx.Context(device=local_rank, seg=rank, nsegs=n)
-    torch.cuda.set_device(local_rank)
+    ndev = max(torch.cuda.device_count(), 1)
+    device = local_rank % ndev       # oversubscription only for dev testing
+    ctx = gx.Context(device=device, seg=rank, nsegs=n)
+    torch.cuda.set_device(device)
 
     if n > 1:
         # broadcast RCCL unique id over gloo, then build the communicator
