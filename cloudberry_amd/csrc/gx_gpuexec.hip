@@ -603,6 +603,23 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -1)
+    {
+        /* block-chunked: each workgroup owns a contiguous row range, so with
+         * the interpolation slot layout its probes (and rev/cnt atomics)
+         * stay inside a tiny contiguous table window — L2-resident. */
+        int64_t chunk = (lk_m.nrows + gridDim.x - 1) / gridDim.x;
+        int64_t lo = blockIdx.x * chunk;
+        int64_t hi = min(lo + chunk, lk_m.nrows);
+        for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        {
+            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = smap.slot0(k);
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r != ~0ULL) hit(r, i);
+        }
+    }
     else
     {
         int64_t base = (blockIdx.x * (int64_t) blockDim.x + threadIdx.x) * B;
@@ -1537,6 +1554,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned int>, keys); break;
+                case 6: launch(k_li_probe_agg_t<-1, unsigned int>, keys); break;
             }
         }
         else
@@ -1549,6 +1567,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned long long>, keys); break;
+                case 6: launch(k_li_probe_agg_t<-1, unsigned long long>, keys); break;
             }
         }
     }

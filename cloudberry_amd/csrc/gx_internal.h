@@ -182,43 +182,6 @@ GX_HD T gx_col_get(const uint8_t *stream, const gx_colmeta m, int64_t row)
                          r * (int64_t) sizeof(T));
 }
 
-/* Incremental stream cursor: a grid-stride loop advances row by a constant
- * stride, so block/offset bookkeeping needs no per-row 64-bit division
- * (one division at init; adds + one compare per step afterwards). */
-struct gx_colcur {
-    const uint8_t *p;            /* address of current datum */
-    int32_t off;                 /* row offset within current block */
-    int32_t rpb;
-    int32_t width;
-    int64_t skip;                /* full_block_len - rpb*width (header+pad hop) */
-    int64_t adv_bytes;           /* stride%rpb datums + stride/rpb blocks, in bytes */
-    int32_t adv_off;             /* stride % rpb */
-
-    __device__ __forceinline__ void init(const uint8_t *stream, const gx_colmeta m,
-                                         int64_t row, int64_t stride)
-    {
-        int64_t b = row / m.rpb;
-        off = (int32_t) (row - b * m.rpb);
-        rpb = m.rpb;
-        width = m.width;
-        skip = m.full_block_len - (int64_t) m.rpb * m.width;
-        int64_t sdiv = stride / m.rpb;
-        adv_off = (int32_t) (stride - sdiv * m.rpb);
-        adv_bytes = sdiv * m.full_block_len + (int64_t) adv_off * m.width;
-        p = stream + b * m.full_block_len + GX_AOCS_DATUM_OFF + (int64_t) off * m.width;
-    }
-    template <typename T>
-    __device__ __forceinline__ T load() const { return *(const T *) p; }
-    __device__ __forceinline__ void advance()
-    {
-        p += adv_bytes;
-        off += adv_off;
-        if (off >= rpb) { off -= rpb; p += skip; }
-    }
-};
-
-/* ---------- packed Motion rows ---------- */
-
 /* gx_ord_row (Motion-1 payload) is declared in include/gpuexec.h — it is
  * also a test-ABI type. */
 

@@ -115,6 +115,26 @@ def test_q3_parity_other_seed(ctx, orc):
     _q3_parity_at(ctx, orc, 0.05, seed=7)
 
 
+def _top10(res):
+    """Q3's final ORDER BY revenue DESC, o_orderdate LIMIT 10 (runs on the
+    QD in the reference — trivial rows, host-side here)."""
+    order = np.lexsort((res["o_orderdate"], -res["revenue"]))[:10]
+    return {k: v[order] for k, v in res.items()}
+
+
+def test_q3_top10_matches_oracle(ctx, orc):
+    sf = 0.5
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    got = _top10(ctx.q3(cust, ordr, li).run().result())
+    want = _top10(orc.q3(orc.gen_customer(sf), orc.gen_orders(sf),
+                         orc.gen_lineitem(sf)))
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-6)
+    li.free(); ordr.free(); cust.free()
+
+
 # ---------------- Motion partition kernels (one GPU) ----------------
 
 def test_motion1_partition_kernels(ctx, orc):
