@@ -29,10 +29,24 @@ def log(rank, *a):
         print(*a, file=sys.stderr, flush=True)
 
 
+def _usable_cores():
+    """Honest thread count: cgroup cpu quota if set, else affinity."""
+    try:
+        with open("/sys/fs/cgroup/cpu.max") as f:
+            quota, period = f.read().split()
+            if quota != "max":
+                return max(1, int(int(quota) / int(period)))
+    except OSError:
+        pass
+    return len(os.sched_getaffinity(0))
+
+
 def cpu_baseline_leg(sample_sf=8.0):
-    """Oracle (CPU restatement, 'port') timed on this host: the reported CPU
-    baseline (BASELINE.md — the reference publishes no numbers).  Bounded
-    sample; single thread; pipeline only (inputs pre-materialized)."""
+    """Oracle (CPU restatement, 'port', OpenMP) timed on this host: the
+    reported CPU baseline (BASELINE.md — the reference publishes no
+    numbers).  Bounded sample; pipeline only (inputs pre-materialized)."""
+    cores = int(os.environ.get("GX_CPU_THREADS", _usable_cores()))
+    os.environ.setdefault("OMP_NUM_THREADS", str(cores))
     from oracle import pyapi as orc   # checker/baseline use only
     c = orc.gen_customer(sample_sf)
     o = orc.gen_orders(sample_sf)
@@ -47,9 +61,10 @@ def cpu_baseline_leg(sample_sf=8.0):
         if reps >= 8:
             break
     rows = len(li["l_orderkey"]) * reps
-    return {"value": rows / t, "unit": "rows/s", "cores": 1, "kind": "port",
+    return {"value": rows / t, "unit": "rows/s", "cores": cores, "kind": "port",
             "sample": f"tpch_q3_sf{sample_sf:g} oracle pipeline x{reps} "
-                      f"({len(li['l_orderkey'])} lineitem rows/pass, single thread)"}
+                      f"({len(li['l_orderkey'])} lineitem rows/pass, "
+                      f"{cores} OpenMP threads)"}
 
 
 def main():
