@@ -46,11 +46,22 @@ def cpu_baseline_leg(sample_sf=8.0):
     reported CPU baseline (BASELINE.md — the reference publishes no
     numbers).  Bounded sample; pipeline only (inputs pre-materialized)."""
     cores = int(os.environ.get("GX_CPU_THREADS", _usable_cores()))
-    os.environ.setdefault("OMP_NUM_THREADS", str(cores))
     from oracle import pyapi as orc   # checker/baseline use only
     c = orc.gen_customer(sample_sf)
     o = orc.gen_orders(sample_sf)
     li = orc.gen_lineitem(sample_sf)
+    # calibrate: cgroup quotas can make "nproc" threads slower than fewer —
+    # pick the faster of {1, cores/2, cores} on one pass and report that count
+    best, cores_eff = None, 1
+    for t_try in sorted({1, max(1, cores // 2), cores}):
+        orc.set_threads(t_try)
+        t0 = time.perf_counter()
+        orc.q3(c, o, li)
+        dt = time.perf_counter() - t0
+        if best is None or dt < best:
+            best, cores_eff = dt, t_try
+    orc.set_threads(cores_eff)
+    cores = cores_eff
     reps, t = 0, 0.0
     t_end = time.time() + 12.0
     while time.time() < t_end or reps < 2:

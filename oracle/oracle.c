@@ -13,11 +13,13 @@
 #include <omp.h>
 #endif
 
-int orc_q3_threads(void)
+int orc_set_threads(int n)
 {
 #ifdef _OPENMP
+    if (n > 0) omp_set_num_threads(n);
     return omp_get_max_threads();
 #else
+    (void) n;
     return 1;
 #endif
 }

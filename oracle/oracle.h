@@ -34,6 +34,7 @@ uint32_t orc_crc32c(uint32_t crc, const void *buf, size_t len); /* pg COMP_CRC32
 int32_t orc_date_adt(int year, int month, int day);
 
 /* ---- deterministic synthetic data (contract shared with HIP kernels) ---- */
+int orc_set_threads(int n);   /* OpenMP thread count; returns effective */
 uint64_t orc_splitmix64(uint64_t x);
 uint64_t orc_mix(uint64_t seed, uint64_t stream, uint64_t idx);
 

@@ -61,6 +61,8 @@ def _load():
     lib.orc_crc32c.argtypes = [ctypes.c_uint32, ctypes.c_void_p, ctypes.c_size_t]
     lib.orc_date_adt.restype = ctypes.c_int32
     lib.orc_date_adt.argtypes = [ctypes.c_int] * 3
+    lib.orc_set_threads.restype = ctypes.c_int
+    lib.orc_set_threads.argtypes = [ctypes.c_int]
     lib.orc_splitmix64.restype = ctypes.c_uint64
     lib.orc_splitmix64.argtypes = [ctypes.c_uint64]
     lib.orc_mix.restype = ctypes.c_uint64
@@ -95,6 +97,10 @@ def _load():
 lib = _load()
 
 CUTOFF_19950315 = lib.orc_date_adt(1995, 3, 15)
+
+
+def set_threads(n):
+    return lib.orc_set_threads(n)
 
 
 def _np(ptr, n, dtype):
