@@ -53,7 +53,8 @@ class _ColDesc(ctypes.Structure):
                 ("nbytes", ctypes.c_int64),
                 ("width", ctypes.c_int32),
                 ("nrows", ctypes.c_int64),
-                ("blocksize", ctypes.c_int32)]
+                ("blocksize", ctypes.c_int32),
+                ("format", ctypes.c_int32)]
 
 
 def _load():
@@ -134,10 +135,12 @@ class Context:
         return Table(self, t)
 
     def bind(self, streams):
-        """streams: list of (bytes, width, nrows) AOCS column streams."""
+        """streams: list of (bytes, width, nrows[, format]) AOCS column
+        streams; format 0 = Orig fixed, 1 = Dense/RLE."""
         descs = (_ColDesc * len(streams))()
         keep = []
-        for i, (data, width, nrows) in enumerate(streams):
+        for i, spec in enumerate(streams):
+            data, width, nrows = spec[:3]
             arr = np.frombuffer(data, np.uint8)
             keep.append(arr)
             descs[i].host_stream = arr.ctypes.data
@@ -145,6 +148,7 @@ class Context:
             descs[i].width = width
             descs[i].nrows = nrows
             descs[i].blocksize = 32768
+            descs[i].format = spec[3] if len(spec) > 3 else 0
         t = ctypes.c_void_p()
         self._chk(self._lib.gx_table_bind(self._h, descs, len(streams), ctypes.byref(t)))
         return Table(self, t)

@@ -329,6 +329,99 @@ __global__ void k_verify_crc(const uint8_t *stream, int64_t nblocks, int64_t ful
     }
 }
 
+__global__ void k_verify_crc_dir(const uint8_t *stream, const gx_blockref *dir,
+                                 int64_t nblocks, int *err)
+{
+    __shared__ uint32_t tab[256];
+    d_crc_table_init(tab);
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        const uint8_t *blk = stream + dir[b].offset;
+        uint32_t b03 = ((const uint32_t *) blk)[0];
+        uint32_t b47 = ((const uint32_t *) blk)[1];
+        uint32_t kind = (b03 >> 28) & 7;
+        uint32_t datalen = (kind == 1)
+            ? (((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21))
+            : (b03 & 0x1FFFFFu);
+        int64_t blen = (24 + (int64_t) datalen + 7) & ~7LL;
+        if (((const uint32_t *) blk)[3] != d_crc32c(tab, 0xFFFFFFFFu, blk, 12) ||
+            ((const uint32_t *) blk)[2] != d_crc32c(tab, 0xFFFFFFFFu, blk + 16, blen - 16))
+            atomicOr(err, 2);
+    }
+}
+
+/* Dense(±RLE) block decode — one THREAD per AO block (blocks decode in
+ * parallel across the grid; within a block the varint walk is inherently
+ * serial).  Format: oracle/oracle.c RLE codec comments; reader semantics
+ * datumstreamblock.h:1724-1912. */
+template <typename T>
+__global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
+                               int64_t nblocks, int64_t nrows, T *out, int *err)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        const uint8_t *blk = stream + dir[b].offset;
+        const uint8_t *c = blk + 24;
+        int16_t version = ((const int16_t *) c)[0];
+        int16_t flags = ((const int16_t *) c)[1];
+        int32_t logical = ((const int32_t *) c)[1];
+        int32_t phys = ((const int32_t *) c)[2];
+        int32_t psize = ((const int32_t *) c)[3];
+        T *dst = out + dir[b].first_row;
+        if (dir[b].first_row + logical > nrows || logical != dir[b].rows ||
+            psize != phys * (int32_t) sizeof(T))
+        { atomicOr(err, 1); continue; }
+        if (version == 0)
+        {
+            /* Orig: ndatum at content+4 (int16), datums at +16 */
+            int16_t nd = ((const int16_t *) c)[2];
+            if (nd != logical) { atomicOr(err, 1); continue; }
+            const T *d = (const T *) (c + 16);
+            for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
+            continue;
+        }
+        if (version != 1 && version != 2) { atomicOr(err, 1); continue; }
+        if (flags & 0x1) { atomicOr(err, 1); continue; }   /* nulls: not yet */
+        if (!(flags & 0x2))
+        {
+            if (logical != phys) { atomicOr(err, 1); continue; }
+            const T *d = (const T *) (c + 16);
+            for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
+            continue;
+        }
+        if (flags & 0x4) { atomicOr(err, 2); continue; }   /* delta: round 2 */
+        int32_t bmbits = ((const int32_t *) c)[5];
+        int32_t csize = ((const int32_t *) c)[7];
+        if (((const int32_t *) c)[4] != 0 || bmbits != phys)
+        { atomicOr(err, 1); continue; }
+        const uint8_t *bmp = c + 32;
+        const uint8_t *cnts = bmp + ((bmbits + 7) >> 3);
+        int32_t hdr = 32 + ((bmbits + 7) >> 3) + csize;
+        const T *datum = (const T *) (c + ((hdr + 7) & ~7));
+        int64_t w = 0;
+        int32_t coff = 0;
+        for (int32_t p = 0; p < phys; p++)
+        {
+            int64_t reps = 1;
+            if (bmp[p >> 3] & (1u << (p & 7)))
+            {
+                int32_t n = (cnts[coff] >> 6) + 1;
+                int32_t v = cnts[coff] & 0x3F;
+                for (int32_t i = 1; i < n; i++) v = (v << 8) | cnts[coff + i];
+                coff += n;
+                reps += v;
+            }
+            if (w + reps > logical) { atomicOr(err, 1); break; }
+            T val = datum[p];
+            for (int64_t r = 0; r < reps; r++) dst[w + r] = val;
+            w += reps;
+        }
+        if (w != logical || coff != csize) atomicOr(err, 1);
+    }
+}
+
 /* ================= Motion routing ================= */
 
 __global__ void k_route(const int64_t *keys, int64_t n, int32_t nsegs, int32_t *out)
@@ -831,6 +924,10 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
 struct gx_col {
     uint8_t *dstream = nullptr;  /* device AOCS stream */
     gx_colmeta m{};
+    int format = 0;              /* 0 = fixed Orig blocks (O(1) addressing),
+                                    1 = Dense/RLE (directory-based) */
+    gx_blockref *ddir = nullptr; /* device block directory (format 1) */
+    int64_t nblocks = 0;
 };
 
 struct gx_table {
@@ -959,6 +1056,43 @@ static gx_status encode_column_device(gx_ctx *ctx, const void *dvals, int width,
     return GX_OK;
 }
 
+/* walk a stream's AO envelope headers on the HOST, building the per-block
+ * directory a variable-geometry (Dense/RLE) stream needs */
+static gx_status parse_block_dir(const uint8_t *s, int64_t nbytes,
+                                 std::vector<gx_blockref> &dir, int64_t *rows_out)
+{
+    int64_t off = 0, row = 0;
+    while (off + 24 <= nbytes)
+    {
+        uint32_t b03, b47;
+        memcpy(&b03, s + off, 4);
+        memcpy(&b47, s + off + 4, 4);
+        if (b03 == 0 && b47 == 0) break;
+        uint32_t kind = (b03 >> 28) & 7;
+        uint32_t rows, datalen;
+        if (kind == 1)
+        {
+            rows = (b03 & 0x00FFFC00u) >> 10;
+            datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+            if ((b47 & 0x1FFFFFu) != 0) return GX_ERR_INVALID;
+        }
+        else if (kind == 3)
+        {
+            rows = b47 & 0x3FFFFFFFu;
+            datalen = b03 & 0x1FFFFFu;
+        }
+        else
+            return GX_ERR_INVALID;
+        int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
+        if (off + blocklen > nbytes) return GX_ERR_INVALID;
+        dir.push_back({off, row, (int32_t) rows, 0});
+        row += rows;
+        off += blocklen;
+    }
+    *rows_out = row;
+    return GX_OK;
+}
+
 extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncols,
                                    gx_table **out)
 {
@@ -969,12 +1103,32 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
     for (int c = 0; c < ncols; c++)
     {
         gx_col col;
+        col.format = cols[c].format;
         col.m.width = cols[c].width;
         col.m.rpb = gx_aocs_rows_per_block(cols[c].width, cols[c].blocksize);
         col.m.nrows = cols[c].nrows;
         col.m.full_block_len = gx_aocs_block_len(cols[c].width, col.m.rpb);
         col.m.nbytes = cols[c].nbytes;
         gx_colmeta_finish(&col.m);
+        if (col.format == 1)
+        {
+            std::vector<gx_blockref> dir;
+            int64_t rows = 0;
+            gx_status st = parse_block_dir((const uint8_t *) cols[c].host_stream,
+                                           cols[c].nbytes, dir, &rows);
+            if (st != GX_OK || rows != cols[c].nrows)
+            {
+                set_err(ctx, "bad Dense/RLE stream%s", "");
+                delete t;
+                return GX_ERR_INVALID;
+            }
+            col.nblocks = (int64_t) dir.size();
+            hipError_t e = hipMalloc(&col.ddir, dir.size() * sizeof(gx_blockref));
+            if (e != hipSuccess) { delete t; return GX_ERR_OOM; }
+            hipMemcpyAsync(col.ddir, dir.data(), dir.size() * sizeof(gx_blockref),
+                           hipMemcpyHostToDevice, ctx->stream);
+            hipStreamSynchronize(ctx->stream);
+        }
         hipError_t e = hipMalloc(&col.dstream, cols[c].nbytes);
         if (e != hipSuccess) { set_err(ctx, "hipMalloc: %s", hipGetErrorString(e)); delete t; return GX_ERR_OOM; }
         e = hipMemcpyAsync(col.dstream, cols[c].host_stream, cols[c].nbytes,
@@ -991,7 +1145,10 @@ extern "C" gx_status gx_table_free(gx_table *t)
 {
     if (!t) return GX_OK;
     for (auto &c : t->cols)
+    {
         if (c.dstream) hipFree(c.dstream);
+        if (c.ddir) hipFree(c.ddir);
+    }
     delete t;
     return GX_OK;
 }
@@ -1138,7 +1295,22 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     HIP_CHK(ctx, hipMalloc(&derr, 4));
     HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, ctx->stream));
     int grid = (int) std::min<int64_t>(nblocks, 65535);
-    if (c.m.width == 8)
+    if (c.format == 1)
+    {
+        if (c.m.width == 8)
+            hipLaunchKernelGGL(k_decode_dense<int64_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int64_t *) dout, derr);
+        else if (c.m.width == 4)
+            hipLaunchKernelGGL(k_decode_dense<int32_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int32_t *) dout, derr);
+        else
+            hipLaunchKernelGGL(k_decode_dense<int8_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int8_t *) dout, derr);
+        if (verify_checksums)
+            hipLaunchKernelGGL(k_verify_crc_dir, dim3(GRID), dim3(64), 0, ctx->stream,
+                               c.dstream, c.ddir, c.nblocks, derr);
+    }
+    else if (c.m.width == 8)
         hipLaunchKernelGGL(k_decode<int64_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
                            c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
                            (int64_t *) dout, derr);
@@ -1150,7 +1322,7 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
         hipLaunchKernelGGL(k_decode<int8_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
                            c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
                            (int8_t *) dout, derr);
-    if (verify_checksums)
+    if (verify_checksums && c.format == 0)
         hipLaunchKernelGGL(k_verify_crc, dim3(GRID), dim3(64), 0, ctx->stream,
                            c.dstream, nblocks, c.m.full_block_len, c.m.nrows, c.m.rpb,
                            c.m.width, derr);
@@ -1190,6 +1362,14 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
     if (!ctx || !customer || !orders || !lineitem) return GX_ERR_INVALID;
     if (customer->cols.size() != 2 || orders->cols.size() != 4 ||
         lineitem->cols.size() != 4) return GX_ERR_INVALID;
+    for (auto *t : {customer, orders, lineitem})
+        for (auto &c : t->cols)
+            if (c.format != 0)
+            {
+                set_err(ctx, "Q3 pipeline requires fixed-format (Orig) streams; "
+                             "decode RLE columns first%s", "");
+                return GX_ERR_INVALID;
+            }
     gx_q3 *q = new gx_q3();
     q->ctx = ctx;
     q->cust = customer;

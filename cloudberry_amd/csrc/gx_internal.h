@@ -182,6 +182,14 @@ GX_HD T gx_col_get(const uint8_t *stream, const gx_colmeta m, int64_t row)
                          r * (int64_t) sizeof(T));
 }
 
+/* per-block directory entry for variable-geometry (RLE/Dense) streams */
+struct gx_blockref {
+    int64_t offset;              /* byte offset of the AO block */
+    int64_t first_row;           /* 0-based logical first row */
+    int32_t rows;                /* logical rows in the block */
+    int32_t pad;
+};
+
 /* gx_ord_row (Motion-1 payload) is declared in include/gpuexec.h — it is
  * also a test-ABI type. */
 

@@ -79,11 +79,14 @@ gx_status gx_comm_init(gx_ctx *ctx, const unsigned char uid[GX_UNIQUE_ID_BYTES])
 
 typedef struct gx_coldesc {
     const void *host_stream;   /* AOCS stream bytes (appendonly=column,
-                                  compresstype=none, checksum=true) */
+                                  checksum=true) */
     int64_t     nbytes;
     int32_t     width;         /* fixed datum width: 1, 4 or 8 */
     int64_t     nrows;
     int32_t     blocksize;     /* AO blocksize the stream was written with */
+    int32_t     format;        /* 0 = Orig (compresstype=none), 1 =
+                                  Dense/Dense_Enhanced incl. RLE_TYPE
+                                  (no-null, no-delta subset) */
 } gx_coldesc;
 
 gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncols,

@@ -333,6 +333,9 @@ void orc_free(void *p) { free(p); }
  *   accept while 16 + n*width + width < blocksize-24  and  n+1 < 16383.
  * ====================================================================== */
 
+static int64_t decode_dense_content(const uint8_t *c, int width,
+                                    uint8_t *dst, int64_t cap_rows);
+
 int32_t orc_aocs_rows_per_block(int width, int32_t blocksize)
 {
     int32_t maxdata = blocksize - 24;
@@ -425,10 +428,21 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
         if (b03 == 0 && b47 == 0) break;         /* zero padding tail */
         uint32_t kind = (b03 >> 28) & 7;
         uint32_t hasfrn = (b03 >> 27) & 1;
-        uint32_t rows = (b03 & 0x00FFFC00u) >> 10;
-        uint32_t datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
-        uint32_t complen = b47 & 0x1FFFFFu;
-        if (kind != 1 || !hasfrn || complen != 0) return -1;
+        uint32_t rows, datalen;
+        if (kind == 1)                          /* SmallContent */
+        {
+            rows = (b03 & 0x00FFFC00u) >> 10;
+            datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+            if ((b47 & 0x1FFFFFu) != 0) return -1;   /* bulk-compressed: no */
+        }
+        else if (kind == 3)                     /* NonBulkDenseContent */
+        {
+            rows = b47 & 0x3FFFFFFFu;
+            datalen = b03 & 0x1FFFFFu;
+        }
+        else
+            return -1;
+        if (!hasfrn) return -1;
         int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
         if (off + blocklen > nbytes) return -1;
         if (verify_checksums)
@@ -440,19 +454,249 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
             if (bc != orc_crc32c(0xFFFFFFFFu, stream + off + 16, blocklen - 16)) return -2;
         }
         const uint8_t *content = stream + off + 24;
-        int16_t version, ndatum;
-        int32_t sz;
+        int16_t version;
         memcpy(&version, content, 2);
-        memcpy(&ndatum, content + 4, 2);
-        memcpy(&sz, content + 12, 4);
-        if (version != 0 || (uint32_t) ndatum != rows || sz != (int32_t) rows * width)
+        if (version == 0)                       /* Original */
+        {
+            int16_t ndatum;
+            int32_t sz;
+            memcpy(&ndatum, content + 4, 2);
+            memcpy(&sz, content + 12, 4);
+            if ((uint32_t) ndatum != rows || sz != (int32_t) rows * width)
+                return -1;
+            if (row + rows > cap) return -1;
+            memcpy(dst + row * (int64_t) width, content + 16, sz);
+            row += rows;
+        }
+        else if (version == 1 || version == 2)  /* Dense / Dense_Enhanced */
+        {
+            int64_t got = decode_dense_content(content, width,
+                                               dst + row * (int64_t) width,
+                                               cap - row);
+            if (got < 0 || (uint32_t) got != rows) return -1;
+            row += got;
+        }
+        else
             return -1;
-        if (row + rows > cap) return -1;
-        memcpy(dst + row * (int64_t) width, content + 16, sz);
-        row += rows;
         off += blocklen;
     }
     return row;
+}
+
+
+/* ======================================================================
+ * RLE_TYPE codec — DatumStreamVersion_Dense_Enhanced, no-null no-delta
+ * subset (compresstype=rle_type, compresslevel=1: no bulk compression,
+ * datumstream.c:397-418).  Formats followed:
+ *   Dense header      datumstreamblock.h:108-136 (version=2, flags=RLE)
+ *   Rle_Extension     datumstreamblock.h:142-172
+ *   compress bitmap   LSB-first per byte, one bit per PHYSICAL datum
+ *                     (ON = datum carries a repeat count), reader walk
+ *                     datumstreamblock.h:1754-1912
+ *   repeat counts     Int32Compress varint (1-4 B, top 2 bits = len-1),
+ *                     datumstreamblock.h:604-700; count = EXTRA repeats
+ *   AO envelope       SmallContent when logical rows ≤ 16383 else
+ *                     NonBulkDenseContent (both regular 8-B headers,
+ *                     cdbappendonlystorage.h:76-96, _int.h:253-347)
+ * Capacity mirrors the writer's reservation: content < blocksize − 32
+ * (maxAoHeaderSize for Dense streams, datumstream.c:597-605).
+ * NOTE: conforming-subset writer, not byte-exact with the reference's
+ * rle_type writer (which also delta-compresses int/date columns); the
+ * decoder is what the GPU path needs — DESIGN.md "next rows".
+ * ====================================================================== */
+
+static int32_t varint_size(int32_t v)
+{
+    if (v <= 0x3F) return 1;
+    if (v <= 0x3FFF) return 2;
+    if (v <= 0x3FFFFF) return 3;
+    return 4;
+}
+
+static int32_t varint_encode(uint8_t *b, int32_t v)
+{
+    if (v <= 0x3F) { b[0] = (uint8_t) v; return 1; }
+    if (v <= 0x3FFF) { b[0] = (1 << 6) | (uint8_t) (v >> 8); b[1] = (uint8_t) v; return 2; }
+    if (v <= 0x3FFFFF) { b[0] = (2 << 6) | (uint8_t) (v >> 16); b[1] = (uint8_t) (v >> 8); b[2] = (uint8_t) v; return 3; }
+    b[0] = (3 << 6) | (uint8_t) (v >> 24); b[1] = (uint8_t) (v >> 16);
+    b[2] = (uint8_t) (v >> 8); b[3] = (uint8_t) v;
+    return 4;
+}
+
+static int32_t varint_decode(const uint8_t *b, int32_t *len)
+{
+    int32_t n = (b[0] >> 6) + 1;
+    int32_t v = b[0] & 0x3F;
+    for (int32_t i = 1; i < n; i++) v = (v << 8) | b[i];
+    *len = n;
+    return v;
+}
+
+int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
+                            int64_t first_rownum, int32_t blocksize,
+                            uint8_t *out, int64_t outcap)
+{
+    const uint8_t *src = (const uint8_t *) vals;
+    int32_t maxdata = blocksize - 32;
+    int64_t off = 0, row = 0;
+    /* per-block scratch */
+    int32_t cap_phys = blocksize;              /* physical ≤ content bytes */
+    uint8_t *pvals = malloc((size_t) cap_phys * width);
+    int32_t *extra = malloc(sizeof(int32_t) * cap_phys);
+
+    while (row < nrows)
+    {
+        int32_t phys = 0, non = 0;             /* physical datums, ON bits */
+        int32_t csize = 0;                     /* repeat-counts bytes */
+        int64_t logical = 0;
+        /* greedy fill */
+        while (row + logical < nrows && logical < 0x3FFFFFFE)
+        {
+            const uint8_t *d = src + (row + logical) * width;
+            int same = phys > 0 &&
+                       memcmp(pvals + (size_t) (phys - 1) * width, d, width) == 0 &&
+                       extra[phys - 1] < 0x3FFFFFFF;
+            int32_t new_phys = phys, new_non = non, new_csize = csize;
+            if (same)
+            {
+                int32_t e = extra[phys - 1];
+                if (e == 0) { new_non++; new_csize += 1; }
+                else new_csize += varint_size(e + 1) - varint_size(e);
+            }
+            else
+                new_phys++;
+            int32_t bm = (new_phys + 7) >> 3;
+            int32_t hdr = 16 + (new_non ? 16 + bm + new_csize : 0);
+            int64_t tot = ((hdr + 7) & ~7) + (int64_t) new_phys * width;
+            if (tot >= maxdata && logical > 0)
+                break;
+            if (same)
+            {
+                extra[phys - 1]++;
+                non = new_non; csize = new_csize;
+            }
+            else
+            {
+                memcpy(pvals + (size_t) phys * width, d, width);
+                extra[phys] = 0;
+                phys = new_phys;
+            }
+            logical++;
+        }
+
+        /* emit block */
+        int has_rle = non > 0;
+        int32_t bm = (phys + 7) >> 3;
+        int32_t hdr = 16 + (has_rle ? 16 + bm + csize : 0);
+        int32_t datum_off = (hdr + 7) & ~7;
+        int32_t content = datum_off + phys * width;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        if (off + blocklen > outcap) { free(pvals); free(extra); return -1; }
+        uint8_t *blk = out + off;
+        memset(blk, 0, blocklen);
+
+        uint32_t kind = (logical <= 16383) ? 1u : 3u;   /* Small / NonBulkDense */
+        uint32_t b03 = (kind << 28) | (1u << 27) | (1u << 24);
+        uint32_t b47 = 0;
+        if (kind == 1)
+        {
+            b03 |= (0x00FFFC00u & ((uint32_t) logical << 10)) |
+                   (((uint32_t) content >> 11) & 0x3FFu);
+            b47 = (((uint32_t) content & 0x7FFu) << 21);
+        }
+        else
+        {
+            b03 |= ((uint32_t) content & 0x1FFFFFu);
+            b47 = (uint32_t) logical & 0x3FFFFFFFu;
+        }
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+
+        uint8_t *c = blk + 24;
+        int16_t v16 = 2; memcpy(c, &v16, 2);                 /* Dense_Enhanced */
+        v16 = has_rle ? 2 : 0; memcpy(c + 2, &v16, 2);       /* flags */
+        int32_t v32 = (int32_t) logical; memcpy(c + 4, &v32, 4);
+        v32 = phys; memcpy(c + 8, &v32, 4);
+        v32 = phys * width; memcpy(c + 12, &v32, 4);
+        if (has_rle)
+        {
+            v32 = 0; memcpy(c + 16, &v32, 4);                /* null bitmap bits */
+            v32 = phys; memcpy(c + 20, &v32, 4);             /* compress bitmap bits */
+            v32 = non; memcpy(c + 24, &v32, 4);
+            v32 = csize; memcpy(c + 28, &v32, 4);
+            uint8_t *bmp = c + 32;
+            uint8_t *cnts = bmp + bm;
+            int32_t w = 0;
+            for (int32_t p = 0; p < phys; p++)
+                if (extra[p] > 0)
+                {
+                    bmp[p >> 3] |= (uint8_t) (1u << (p & 7));
+                    w += varint_encode(cnts + w, extra[p]);
+                }
+        }
+        memcpy(c + datum_off, pvals, (size_t) phys * width);
+
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+
+        off += blocklen;
+        row += logical;
+    }
+    free(pvals); free(extra);
+    return off;
+}
+
+/* decode one Dense(±RLE) content area; returns rows written or -1 */
+static int64_t decode_dense_content(const uint8_t *c, int width,
+                                    uint8_t *dst, int64_t cap_rows)
+{
+    int16_t version, flags;
+    int32_t logical, phys, psize;
+    memcpy(&version, c, 2);
+    memcpy(&flags, c + 2, 2);
+    memcpy(&logical, c + 4, 4);
+    memcpy(&phys, c + 8, 4);
+    memcpy(&psize, c + 12, 4);
+    if (psize != phys * width || logical > cap_rows) return -1;
+    if (flags & 0x1) return -1;                /* null bitmap unsupported here */
+    if (!(flags & 0x2))
+    {
+        if (logical != phys) return -1;
+        memcpy(dst, c + 16, (size_t) psize);
+        return logical;
+    }
+    if (flags & 0x4) return -1;                /* delta not yet (round 2) */
+    int32_t nullbits, bmbits, ncnt, csize;
+    memcpy(&nullbits, c + 16, 4);
+    memcpy(&bmbits, c + 20, 4);
+    memcpy(&ncnt, c + 24, 4);
+    memcpy(&csize, c + 28, 4);
+    if (nullbits != 0 || bmbits != phys) return -1;
+    const uint8_t *bmp = c + 32;
+    const uint8_t *cnts = bmp + ((bmbits + 7) >> 3);
+    int32_t hdr = 32 + ((bmbits + 7) >> 3) + csize;
+    const uint8_t *datum = c + ((hdr + 7) & ~7);
+    int64_t w = 0;
+    int32_t coff = 0, seen = 0;
+    for (int32_t p = 0; p < phys; p++)
+    {
+        int64_t reps = 1;
+        if (bmp[p >> 3] & (1u << (p & 7)))
+        {
+            int32_t len, v = varint_decode(cnts + coff, &len);
+            coff += len;
+            seen++;
+            reps += v;
+        }
+        if (w + reps > cap_rows) return -1;
+        for (int64_t r = 0; r < reps; r++)
+            memcpy(dst + (w + r) * width, datum + (size_t) p * width, width);
+        w += reps;
+    }
+    if (seen != ncnt || coff != csize || w != logical) return -1;
+    return w;
 }
 
 /* ======================================================================

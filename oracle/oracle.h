@@ -84,6 +84,13 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
 /* rows per full block for a fixed-width NOT NULL column */
 int32_t orc_aocs_rows_per_block(int width, int32_t blocksize);
 
+/* RLE_TYPE (Dense_Enhanced, no-null no-delta subset): encoder emits
+ * conforming blocks (run-length compress bitmap + varint repeat counts);
+ * orc_aocs_decode handles Orig AND Dense(±RLE) blocks transparently. */
+int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
+                            int64_t first_rownum, int32_t blocksize,
+                            uint8_t *out, int64_t outcap);
+
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {
     int64_t l_orderkey;

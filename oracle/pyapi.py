@@ -80,6 +80,10 @@ def _load():
     lib.orc_aocs_rows_per_block.argtypes = [ctypes.c_int, ctypes.c_int32]
     lib.orc_aocs_encoded_size.restype = ctypes.c_int64
     lib.orc_aocs_encoded_size.argtypes = [ctypes.c_int, ctypes.c_int64, ctypes.c_int32]
+    lib.orc_aocs_encode_rle.restype = ctypes.c_int64
+    lib.orc_aocs_encode_rle.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+                                        ctypes.c_int64, ctypes.c_int32,
+                                        ctypes.c_void_p, ctypes.c_int64]
     lib.orc_aocs_encode.restype = ctypes.c_int64
     lib.orc_aocs_encode.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
                                     ctypes.c_int64, ctypes.c_int32,
@@ -195,6 +199,18 @@ def aocs_encode(vals):
     got = lib.orc_aocs_encode(vals.ctypes.data, width, n, 1, 32768, buf.ctypes.data, cap)
     assert got == cap, (got, cap)
     return buf.tobytes()
+
+
+def aocs_encode_rle(vals):
+    """RLE_TYPE (Dense_Enhanced) encode of a fixed-width NOT NULL column."""
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_rle(vals.ctypes.data, width, len(vals), 1, 32768,
+                                  buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
 
 
 def aocs_decode(stream, width, nrows, dtype, verify=True):

@@ -62,6 +62,33 @@ def test_bind_oracle_streams_roundtrip(ctx, orc):
     t.free()
 
 
+def test_rle_stream_gpu_decode(ctx, orc):
+    """Dense_Enhanced RLE streams bound through the C-ABI decode on device to
+    the original values (clustered keys like l_orderkey), checksums verified."""
+    rng = np.random.default_rng(12)
+    keys = np.repeat(np.arange(1, 60000, dtype=np.int64),
+                     rng.integers(1, 8, 59999))
+    dates = rng.integers(-3000, 0, 70000).astype(np.int32)
+    t = ctx.bind([(orc.aocs_encode_rle(keys), 8, len(keys), 1),
+                  (orc.aocs_encode_rle(dates), 4, len(dates), 1)])
+    np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True), keys)
+    np.testing.assert_array_equal(t.decode_column(1, np.int32, verify=True), dates)
+    t.free()
+
+
+def test_rle_q3_input_rejected(ctx, orc):
+    """The fused Q3 kernels require fixed-format streams (for now) — an RLE
+    table must be rejected loudly, never silently mis-scanned."""
+    keys = np.repeat(np.arange(1, 1000, dtype=np.int64), 3)
+    t = ctx.bind([(orc.aocs_encode_rle(keys), 8, len(keys), 1),
+                  (orc.aocs_encode(np.zeros(len(keys), np.int8)), 1, len(keys))])
+    c2 = ctx.tpch_gen(gx.TPCH_ORDERS, 0.01)
+    c3 = ctx.tpch_gen(gx.TPCH_LINEITEM, 0.01)
+    with pytest.raises(gx.GxError):
+        ctx.q3(t, c2, c3)
+    t.free(); c2.free(); c3.free()
+
+
 def test_corrupted_stream_rejected(ctx, orc):
     vals = np.arange(5000, dtype=np.int64)
     s = bytearray(orc.aocs_encode(vals))
