@@ -81,6 +81,15 @@ extern "C" const char *gx_version(void) { return "gpuexec 0.1 (gfx950)"; }
 static constexpr int TPB = 256;          /* threads per block (4 waves) */
 static constexpr int GRID = 2048;        /* grid-stride grid (fills 256 CUs) */
 
+/* RAII for TEMPORARY device buffers: early returns (HIP_CHK/RCCL_CHK) free
+ * them automatically.  Long-lived state stays owned by gx_table / gx_q3. */
+struct devbuf {
+    void *p = nullptr;
+    ~devbuf() { if (p) (void) hipFree(p); }
+    hipError_t alloc(size_t bytes) { return hipMalloc(&p, bytes ? bytes : 1); }
+    template <typename T> T *as() const { return (T *) p; }
+};
+
 static inline int64_t pow2_at_least(int64_t want)
 {
     int64_t sz = 1024;
@@ -1289,10 +1298,11 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     const gx_col &c = t->cols[colidx];
     if (c.m.nrows > cap_rows) return GX_ERR_INVALID;
     int64_t nblocks = (c.m.nrows + c.m.rpb - 1) / c.m.rpb;
-    void *dout = nullptr;
-    int *derr = nullptr;
-    HIP_CHK(ctx, hipMalloc(&dout, c.m.nrows * (int64_t) c.m.width));
-    HIP_CHK(ctx, hipMalloc(&derr, 4));
+    devbuf dout_b, derr_b;
+    HIP_CHK(ctx, dout_b.alloc(c.m.nrows * (int64_t) c.m.width));
+    HIP_CHK(ctx, derr_b.alloc(4));
+    void *dout = dout_b.p;
+    int *derr = derr_b.as<int>();
     HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, ctx->stream));
     int grid = (int) std::min<int64_t>(nblocks, 65535);
     if (c.format == 1)
@@ -1331,7 +1341,6 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
                                 hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-    hipFree(dout); hipFree(derr);
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
     if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
@@ -1342,14 +1351,14 @@ extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t
                                   int32_t nsegs, int32_t *host_out)
 {
     if (!ctx || n < 0) return GX_ERR_INVALID;
-    int64_t *dk = nullptr; int32_t *dr = nullptr;
-    HIP_CHK(ctx, hipMalloc(&dk, n * 8));
-    HIP_CHK(ctx, hipMalloc(&dr, n * 4));
-    HIP_CHK(ctx, hipMemcpyAsync(dk, host_keys, n * 8, hipMemcpyHostToDevice, ctx->stream));
-    hipLaunchKernelGGL(k_route, dim3(GRID), dim3(TPB), 0, ctx->stream, dk, n, nsegs, dr);
-    HIP_CHK(ctx, hipMemcpyAsync(host_out, dr, n * 4, hipMemcpyDeviceToHost, ctx->stream));
+    devbuf dk, dr;
+    HIP_CHK(ctx, dk.alloc(n * 8));
+    HIP_CHK(ctx, dr.alloc(n * 4));
+    HIP_CHK(ctx, hipMemcpyAsync(dk.p, host_keys, n * 8, hipMemcpyHostToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_route, dim3(GRID), dim3(TPB), 0, ctx->stream,
+                       dk.as<int64_t>(), n, nsegs, dr.as<int32_t>());
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, dr.p, n * 4, hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-    hipFree(dk); hipFree(dr);
     HIP_CHK(ctx, hipGetLastError());
     return GX_OK;
 }
@@ -1417,7 +1426,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * q->cset_width, s));
     q->cmask = cslots - 1;
     uint64_t bwords = (uint64_t) pow2_at_least(
-        std::max<int64_t>((int64_t) n_building * 16 / 64, 4096));
+        std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
     HIP_CHK(ctx, hipMalloc(&q->bloom, bwords * 8));
     HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, bwords * 8, s));
     q->bwmask = bwords - 1;
@@ -1781,7 +1790,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     hipEventElapsedTime(&t34, ev[3], ev[4]);
     hipEventElapsedTime(&t04, ev[0], ev[4]);
     q->stats.ms_cust_build = t01;
-    q->stats.ms_orders_build = t12 - ms_motion;
+    q->stats.ms_orders_build = std::max(0.0, (double) t12 - ms_motion);
     q->stats.ms_motion = ms_motion;
     q->stats.ms_probe_agg = t23;
     q->stats.ms_extract = t34;
@@ -1869,9 +1878,11 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     hipStream_t s = ctx->stream;
     const gx_col &ok = orders->cols[0], &oc = orders->cols[1],
                  &od = orders->cols[2], &op = orders->cols[3];
-    unsigned long long *dhist = nullptr, *dcur = nullptr;
-    HIP_CHK(ctx, hipMalloc(&dhist, nsegs * 8));
-    HIP_CHK(ctx, hipMalloc(&dcur, nsegs * 8));
+    devbuf dhist_b, dcur_b, dsend_b;
+    HIP_CHK(ctx, dhist_b.alloc(nsegs * 8));
+    HIP_CHK(ctx, dcur_b.alloc(nsegs * 8));
+    unsigned long long *dhist = dhist_b.as<unsigned long long>();
+    unsigned long long *dcur = dcur_b.as<unsigned long long>();
     HIP_CHK(ctx, hipMemsetAsync(dhist, 0, nsegs * 8, s));
     hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                        od.dstream, od.m, oc.dstream, oc.m, cutoff, nsegs, dhist);
@@ -1881,9 +1892,9 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     std::vector<unsigned long long> off(nsegs + 1, 0);
     for (int i = 0; i < nsegs; i++) off[i + 1] = off[i] + h[i];
     int64_t total = (int64_t) off[nsegs];
-    if (total > cap) { hipFree(dhist); hipFree(dcur); return GX_ERR_INVALID; }
-    gx_ord_row *dsend = nullptr;
-    HIP_CHK(ctx, hipMalloc(&dsend, std::max<int64_t>(total, 1) * sizeof(gx_ord_row)));
+    if (total > cap) return GX_ERR_INVALID;
+    HIP_CHK(ctx, dsend_b.alloc(std::max<int64_t>(total, 1) * sizeof(gx_ord_row)));
+    gx_ord_row *dsend = dsend_b.as<gx_ord_row>();
     HIP_CHK(ctx, hipMemcpyAsync(dcur, off.data(), nsegs * 8, hipMemcpyHostToDevice, s));
     hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                        ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
@@ -1894,7 +1905,6 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     HIP_CHK(ctx, hipGetLastError());
     for (int i = 0; i < nsegs; i++) out_counts[i] = (int64_t) h[i];
     *out_total = total;
-    hipFree(dhist); hipFree(dcur); hipFree(dsend);
     return GX_OK;
 }
 
