@@ -1577,8 +1577,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipEventRecord(mev0, s));
 
         /* Motion 1: filtered orders by route(o_custkey) */
-        unsigned long long *dhist = nullptr;
-        HIP_CHK(ctx, hipMalloc(&dhist, n * 8));
+        devbuf dhist_b, dcur_b, send1_b, recv1_b, send2_b, recv2_b,
+               cnts_mine_b, cnts_all_b;
+        HIP_CHK(ctx, dhist_b.alloc(n * 8));
+        unsigned long long *dhist = dhist_b.as<unsigned long long>();
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                            od.dstream, od.m, oc.dstream, oc.m, q->cutoff, n, dhist);
@@ -1588,19 +1590,21 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         std::vector<unsigned long long> off1(n + 1, 0);
         for (int i = 0; i < n; i++) off1[i + 1] = off1[i] + h1[i];
         unsigned long long send1_n = off1[n];
-        gx_ord_row *send1 = nullptr;
-        HIP_CHK(ctx, hipMalloc(&send1, std::max<uint64_t>(send1_n, 1) * sizeof(gx_ord_row)));
-        unsigned long long *dcur = nullptr;
-        HIP_CHK(ctx, hipMalloc(&dcur, n * 8));
+        (void) send1_n;
+        HIP_CHK(ctx, send1_b.alloc(std::max<uint64_t>(send1_n, 1) * sizeof(gx_ord_row)));
+        gx_ord_row *send1 = send1_b.as<gx_ord_row>();
+        HIP_CHK(ctx, dcur_b.alloc(n * 8));
+        unsigned long long *dcur = dcur_b.as<unsigned long long>();
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                            op.dstream, op.m, q->cutoff, n, dcur, send1);
 
         /* exchange counts (all-gather of per-dest counts) */
-        unsigned long long *dcnts_all = nullptr, *dcnts_mine = nullptr;
-        HIP_CHK(ctx, hipMalloc(&dcnts_mine, n * 8));
-        HIP_CHK(ctx, hipMalloc(&dcnts_all, (int64_t) n * n * 8));
+        HIP_CHK(ctx, cnts_mine_b.alloc(n * 8));
+        HIP_CHK(ctx, cnts_all_b.alloc((int64_t) n * n * 8));
+        unsigned long long *dcnts_mine = cnts_mine_b.as<unsigned long long>();
+        unsigned long long *dcnts_all = cnts_all_b.as<unsigned long long>();
         HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h1.data(), n * 8, hipMemcpyHostToDevice, s));
         RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
         std::vector<unsigned long long> cnts_all(n * n);
@@ -1611,8 +1615,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         for (int r = 0; r < n; r++) rcv1[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff1[r + 1] = roff1[r] + rcv1[r];
         unsigned long long recv1_n = roff1[n];
-        gx_ord_row *recv1 = nullptr;
-        HIP_CHK(ctx, hipMalloc(&recv1, std::max<uint64_t>(recv1_n, 1) * sizeof(gx_ord_row)));
+        HIP_CHK(ctx, recv1_b.alloc(std::max<uint64_t>(recv1_n, 1) * sizeof(gx_ord_row)));
+        gx_ord_row *recv1 = recv1_b.as<gx_ord_row>();
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
@@ -1640,8 +1644,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipStreamSynchronize(s));
         std::vector<unsigned long long> off2(n + 1, 0);
         for (int i = 0; i < n; i++) off2[i + 1] = off2[i] + h2[i];
-        gx_qual_row *send2 = nullptr;
-        HIP_CHK(ctx, hipMalloc(&send2, std::max<uint64_t>(off2[n], 1) * sizeof(gx_qual_row)));
+        HIP_CHK(ctx, send2_b.alloc(std::max<uint64_t>(off2[n], 1) * sizeof(gx_qual_row)));
+        gx_qual_row *send2 = send2_b.as<gx_qual_row>();
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off2.data(), n * 8, hipMemcpyHostToDevice, s));
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_qual_emit<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
@@ -1660,8 +1664,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         for (int r = 0; r < n; r++) rcv2[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff2[r + 1] = roff2[r] + rcv2[r];
         unsigned long long recv2_n = roff2[n];
-        gx_qual_row *recv2 = nullptr;
-        HIP_CHK(ctx, hipMalloc(&recv2, std::max<uint64_t>(recv2_n, 1) * sizeof(gx_qual_row)));
+        HIP_CHK(ctx, recv2_b.alloc(std::max<uint64_t>(recv2_n, 1) * sizeof(gx_qual_row)));
+        gx_qual_row *recv2 = recv2_b.as<gx_qual_row>();
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
@@ -1711,8 +1715,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         hipEventElapsedTime(&mms, mev0, mev1);
         ms_motion = mms;
         hipEventDestroy(mev0); hipEventDestroy(mev1);
-        hipFree(dhist); hipFree(dcur); hipFree(send1); hipFree(recv1);
-        hipFree(send2); hipFree(recv2); hipFree(dcnts_mine); hipFree(dcnts_all);
     }
     q->qual_orders = qual;
     HIP_CHK(ctx, hipEventRecord(ev[2], s));
