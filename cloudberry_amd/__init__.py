@@ -13,6 +13,7 @@ _ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 _SO = os.path.join(_ROOT, "cloudberry_amd", "libgpuexec.so")
 
 TPCH_CUSTOMER, TPCH_ORDERS, TPCH_LINEITEM = 0, 1, 2
+TPCH_LINEITEM_NUMERIC = 3
 CUTOFF_19950315 = -1753  # DateADT of 1995-03-15 (validated vs oracle in tests)
 
 _STATUS = {0: "GX_OK", 1: "GX_ERR_HIP", 2: "GX_ERR_RCCL", 3: "GX_ERR_INVALID",
@@ -30,6 +31,7 @@ class _Group(ctypes.Structure):
                 ("o_orderdate", ctypes.c_int32),
                 ("o_shippriority", ctypes.c_int32),
                 ("revenue", ctypes.c_double),
+                ("revenue_num", ctypes.c_int64),
                 ("nitems", ctypes.c_int64)]
 
 
@@ -85,6 +87,7 @@ def _load():
                                  ctypes.c_int32, ctypes.c_void_p]
     lib.gx_q3_prepare.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int32,
                                   ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_q3_set_numeric.argtypes = [ctypes.c_void_p, ctypes.c_int]
     lib.gx_q3_run.argtypes = [ctypes.c_void_p]
     lib.gx_q3_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
     lib.gx_q3_result.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.POINTER(_Group)),
@@ -172,10 +175,12 @@ class Context:
                                             cap, ctypes.byref(total)))
         return counts, rows[:total.value]
 
-    def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315):
+    def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315, numeric=False):
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare(self._h, cust._t, orders._t, lineitem._t,
                                           cutoff, ctypes.byref(q)))
+        if numeric:
+            self._chk(self._lib.gx_q3_set_numeric(q, 1))
         return Q3(self, q)
 
     def close(self):
@@ -238,6 +243,7 @@ class Q3:
                "o_orderdate": np.array([gp[i].o_orderdate for i in range(n)], np.int32),
                "o_shippriority": np.array([gp[i].o_shippriority for i in range(n)], np.int32),
                "revenue": np.array([gp[i].revenue for i in range(n)], np.float64),
+               "revenue_num": np.array([gp[i].revenue_num for i in range(n)], np.int64),
                "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
         self.ctx._lib.gx_free(gp)
         return res

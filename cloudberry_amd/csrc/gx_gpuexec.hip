@@ -197,6 +197,32 @@ __global__ void k_emit_li(uint64_t seed, int64_t nord, int seg, int nsegs,
     }
 }
 
+__global__ void k_emit_li_num(uint64_t seed, int64_t nord, int seg, int nsegs,
+                              const uint64_t *offs, int64_t *lkey,
+                              int64_t *price_c, int64_t *disc_c, int32_t *ship)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, nord);
+    if (lo >= nord) return;
+    uint64_t w = offs[t];
+    for (int64_t o = lo + 1; o <= hi; o++)
+    {
+        if (!(nsegs == 1 || gx_route_i64(o, nsegs) == seg)) continue;
+        int32_t nl = gx_gen_nlines(seed, o);
+        for (int32_t j = 0; j < nl; j++)
+        {
+            lkey[w] = o;
+            /* the exact scaled-integer forms of gx_gen_price/_discount */
+            price_c[w] = (int64_t) (90000 + gx_mix(seed, GX_ST_LI_PRICE,
+                                                   (uint64_t) o * 8 + j) % 10410001ULL);
+            disc_c[w] = (int64_t) (gx_mix(seed, GX_ST_LI_DISC,
+                                          (uint64_t) o * 8 + j) % 11);
+            ship[w] = gx_gen_shipdate(seed, o, j);
+            w++;
+        }
+    }
+}
+
 /* ================= AOCS encode / decode ================= */
 
 /* Writes headers + datums for one AO block per workgroup (CRCs in a second
@@ -769,6 +795,53 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
     gx_wave_count_add(hits, local_hits);
 }
 
+/* numeric(15,2) probe+agg: measures are scaled int64 (price
+ * in cents, discount in hundredths); revenue numerator = Σ price_c·(100−d)
+ * accumulated with integer atomics — BIT-EXACT, order-independent, equal to
+ * the PG numeric SUM for these ranges.  The numerator lives in the trev
+ * buffer (reinterpreted u64).  Overflow guard: numerators past 2^62 set the
+ * error flag (impossible for sane groups; detects corrupt input). */
+template <typename KT>
+__global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
+                                   const uint8_t *pr_s, gx_colmeta pr_m,
+                                   const uint8_t *di_s, gx_colmeta di_m,
+                                   const uint8_t *sh_s, gx_colmeta sh_m,
+                                   int32_t cutoff,
+                                   const KT *tkey,
+                                   unsigned long long *tnum,
+                                   unsigned long long *tcnt,
+                                   gx_slotmap smap,
+                                   unsigned long long *hits, int *err)
+{
+    uint64_t tmask = smap.mask;
+    unsigned long long local_hits = 0;
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < lk_m.nrows; i += stride)
+    {
+        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+        uint64_t slot = smap.slot0(k);
+        bool found = false;
+        while (true)
+        {
+            KT v = tkey[slot];
+            if (v == (KT) 0) break;
+            if (v == (KT) k) { found = true; break; }
+            slot = (slot + 1) & tmask;
+        }
+        if (!found) continue;
+        int64_t price_c = gx_col_get<int64_t>(pr_s, pr_m, i);
+        int64_t disc_c = gx_col_get<int64_t>(di_s, di_m, i);
+        unsigned long long add = (unsigned long long) (price_c * (100 - disc_c));
+        unsigned long long old = atomicAdd(&tnum[slot], add);
+        if (old + add > (1ULL << 62)) atomicOr(err, 4);
+        atomicAdd(&tcnt[slot], 1ULL);
+        local_hits++;
+    }
+    gx_wave_count_add(hits, local_hits);
+}
+
 /* extract groups with ≥1 matched lineitem into SoA result arrays.
  * Two-pass per-workgroup compaction: each workgroup owns a contiguous slot
  * range, counts its keeps, claims an output region with ONE atomic, then
@@ -970,6 +1043,7 @@ struct gx_q3 {
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
     double *r_rev = nullptr;
     int64_t *r_cnt = nullptr;
+    int numeric = 0;                 /* numeric(15,2) scaled-i64 measures */
     int64_t rescap = 0;
     int64_t ngroups = 0;
     int64_t qual_orders = 0;
@@ -1204,6 +1278,8 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     int64_t ncust = (int64_t) (150000.0 * sf + 0.5);
     int64_t nord = (int64_t) (1500000.0 * sf + 0.5);
     int64_t nglobal = (which == GX_TPCH_CUSTOMER) ? ncust : nord;
+    bool li_numeric = ((int) which == 3);   /* GX_TPCH_LINEITEM_NUMERIC */
+    if (li_numeric) which = GX_TPCH_LINEITEM;
     int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
     int64_t blocks = (nthreads + TPB - 1) / TPB;
     /* the count kernels write counts[t] for EVERY launched thread */
@@ -1271,8 +1347,13 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         HIP_CHK(ctx, hipMalloc(&dpr, n * 8));
         HIP_CHK(ctx, hipMalloc(&ddi, n * 8));
         HIP_CHK(ctx, hipMalloc(&dsh, n * 4));
-        hipLaunchKernelGGL(k_emit_li, dim3(blocks), dim3(TPB), 0, ctx->stream,
-                           seed, nglobal, ctx->seg, ctx->nsegs, doffs, dlk, dpr, ddi, dsh);
+        if (li_numeric)
+            hipLaunchKernelGGL(k_emit_li_num, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                               seed, nglobal, ctx->seg, ctx->nsegs, doffs, dlk,
+                               (int64_t *) dpr, (int64_t *) ddi, dsh);
+        else
+            hipLaunchKernelGGL(k_emit_li, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                               seed, nglobal, ctx->seg, ctx->nsegs, doffs, dlk, dpr, ddi, dsh);
         gx_col c0, c1, c2, c3;
         st = encode_column_device(ctx, dlk, 8, n, &c0);
         if (st == GX_OK) st = encode_column_device(ctx, dpr, 8, n, &c1);
@@ -1504,6 +1585,13 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     return GX_OK;
 }
 
+extern "C" gx_status gx_q3_set_numeric(gx_q3 *q, int on)
+{
+    if (!q) return GX_ERR_INVALID;
+    q->numeric = on ? 1 : 0;
+    return GX_OK;
+}
+
 extern "C" gx_status gx_q3_run(gx_q3 *q)
 {
     if (!q) return GX_ERR_INVALID;
@@ -1722,6 +1810,29 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     /* ---- stage 3: lineitem scan+probe+agg (dominant kernel) ---- */
     unsigned long long *dhits = q->dhits;
     HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
+    if (q->numeric)
+    {
+        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
+        if (q->key_width == 4)
+            hipLaunchKernelGGL(k_li_probe_agg_num<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                               ls.dstream, ls.m, q->cutoff,
+                               (const unsigned int *) q->tkey,
+                               (unsigned long long *) q->trev, q->tcnt, q->smap,
+                               dhits, (int *) q->dmin);
+        else
+            hipLaunchKernelGGL(k_li_probe_agg_num<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                               ls.dstream, ls.m, q->cutoff,
+                               (const unsigned long long *) q->tkey,
+                               (unsigned long long *) q->trev, q->tcnt, q->smap,
+                               dhits, (int *) q->dmin);
+        int herr = 0;
+        HIP_CHK(ctx, hipMemcpyAsync(&herr, q->dmin, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        if (herr) { set_err(ctx, "numeric overflow in aggregation%s", ""); return GX_ERR_INVALID; }
+    }
+    else
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
@@ -1847,7 +1958,18 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
         g[i].l_orderkey = okey[idx[i]];
         g[i].o_orderdate = odate[idx[i]];
         g[i].o_shippriority = oprio[idx[i]];
-        g[i].revenue = rev[idx[i]];
+        if (q->numeric)
+        {
+            int64_t num;
+            memcpy(&num, &rev[idx[i]], 8);   /* trev held integer numerators */
+            g[i].revenue_num = num;
+            g[i].revenue = (double) num / 1e4;
+        }
+        else
+        {
+            g[i].revenue = rev[idx[i]];
+            g[i].revenue_num = 0;
+        }
         g[i].nitems = cnt[idx[i]];
     }
     *out = g;

@@ -99,7 +99,8 @@ gx_status gx_table_logical_bytes(const gx_table *t, double *out);
 
 /* synthetic TPC-H-shaped tables generated AND AOCS-encoded on device
  * (deterministic; identical formulas to oracle/oracle.c datagen) */
-typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2 } gx_tpch_table;
+typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2,
+               GX_TPCH_LINEITEM_NUMERIC = 3 /* measures as scaled int64 */ } gx_tpch_table;
 gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                       uint64_t seed, gx_table **out);
 
@@ -120,7 +121,9 @@ typedef struct gx_q3_group {
     int64_t l_orderkey;
     int32_t o_orderdate;
     int32_t o_shippriority;
-    double  revenue;
+    double  revenue;           /* f64 mode: the SUM; numeric mode: num/1e4 */
+    int64_t revenue_num;       /* numeric(15,2) mode: exact Σ price_c·(100−d),
+                                  implied scale 1e-4 (0 in f64 mode) */
     int64_t nitems;
 } gx_q3_group;
 
@@ -143,6 +146,10 @@ typedef struct gx_q3_stats {
 gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
                         gx_table *lineitem, int32_t cutoff_dateadt,
                         gx_q3 **out);
+/* numeric(15,2) mode: lineitem measures are scaled int64
+ * (price cents, discount hundredths — GX_TPCH_LINEITEM_NUMERIC tables);
+ * aggregation is integer → results BIT-EXACT vs the oracle. */
+gx_status gx_q3_set_numeric(gx_q3 *q, int on);
 gx_status gx_q3_run(gx_q3 *q);   /* one full pass; re-runnable (bench steps) */
 gx_status gx_q3_stats_get(const gx_q3 *q, gx_q3_stats *out);
 /* groups of THIS segment, sorted by l_orderkey asc; caller frees with gx_free */

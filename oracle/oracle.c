@@ -779,6 +779,92 @@ static void tab_init(q3tab *t, int64_t want)
     for (uint64_t i = 0; i < sz; i++) t->key[i] = EMPTY_KEY;
 }
 
+static int q3n_group_cmp(const void *a, const void *b)
+{
+    const orc_q3n_group *x = a, *y = b;
+    return (x->l_orderkey > y->l_orderkey) - (x->l_orderkey < y->l_orderkey);
+}
+
+/*
+ * numeric(15,2) Q3 — the real TPC-H column types (tpch500GB.sql:69-72)
+ * mapped to scaled int64.  The synthetic f64 measures are
+ * exact 2-decimal values by construction (gen_price/gen_discount), so the
+ * cents mapping is recovered exactly with llround.  revenue numerator =
+ * Σ price_c·(100−disc_c) — bit-exact regardless of ordering, matching the
+ * PG numeric SUM for these ranges (no rounding anywhere).
+ */
+int64_t orc_q3_numeric(const orc_customer *c, const orc_orders *o,
+                       const orc_lineitem *l, int32_t cutoff,
+                       orc_q3n_group **out)
+{
+    set64 cust;
+    set_init(&cust, c->n + 16);
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+    for (int64_t i = 0; i < c->n; i++)
+        if (c->c_mktsegment[i] == 0)
+            set_insert(&cust, c->c_custkey[i]);
+
+    q3tab t;
+    tab_init(&t, o->n + 16);
+    int64_t *num = calloc(t.mask + 1, sizeof(int64_t));
+    for (int64_t i = 0; i < o->n; i++)
+    {
+        if (!(o->o_orderdate[i] < cutoff)) continue;
+        if (!set_contains(&cust, o->o_custkey[i])) continue;
+        int64_t k = o->o_orderkey[i];
+        uint64_t j = hmix64((uint64_t) k) & t.mask;
+        while (t.key[j] != EMPTY_KEY && t.key[j] != k) j = (j + 1) & t.mask;
+        t.key[j] = k;
+        t.odate[j] = o->o_orderdate[i];
+        t.oprio[j] = o->o_shippriority[i];
+    }
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+    for (int64_t i = 0; i < l->n; i++)
+    {
+        if (!(l->l_shipdate[i] > cutoff)) continue;
+        int64_t k = l->l_orderkey[i];
+        uint64_t j = hmix64((uint64_t) k) & t.mask;
+        while (t.key[j] != EMPTY_KEY && t.key[j] != k) j = (j + 1) & t.mask;
+        if (t.key[j] == EMPTY_KEY) continue;
+        int64_t price_c = (int64_t) llround(l->l_extendedprice[i] * 100.0);
+        int64_t disc_c = (int64_t) llround(l->l_discount[i] * 100.0);
+        int64_t add = price_c * (100 - disc_c);
+#ifdef _OPENMP
+        __atomic_fetch_add(&num[j], add, __ATOMIC_RELAXED);
+        __atomic_fetch_add((uint64_t *) &t.cnt[j], 1, __ATOMIC_RELAXED);
+#else
+        num[j] += add;
+        t.cnt[j]++;
+#endif
+    }
+
+    int64_t ng = 0;
+    for (uint64_t j = 0; j <= t.mask; j++)
+        if (t.key[j] != EMPTY_KEY && t.cnt[j] > 0) ng++;
+    orc_q3n_group *g = malloc(sizeof(orc_q3n_group) * (ng ? ng : 1));
+    int64_t w = 0;
+    for (uint64_t j = 0; j <= t.mask; j++)
+        if (t.key[j] != EMPTY_KEY && t.cnt[j] > 0)
+        {
+            g[w].l_orderkey = t.key[j];
+            g[w].o_orderdate = t.odate[j];
+            g[w].o_shippriority = t.oprio[j];
+            g[w].revenue_num = num[j];
+            g[w].nitems = t.cnt[j];
+            w++;
+        }
+    qsort(g, ng, sizeof(orc_q3n_group), q3n_group_cmp);
+    free(cust.keys);
+    free(t.key); free(t.odate); free(t.oprio); free(t.rev); free(t.cnt);
+    free(num);
+    *out = g;
+    return ng;
+}
+
 static int q3_group_cmp(const void *a, const void *b)
 {
     const orc_q3_group *x = a, *y = b;

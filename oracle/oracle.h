@@ -105,6 +105,21 @@ typedef struct {
  * DateADT of 1995-03-15 unless overridden. */
 int64_t orc_q3(const orc_customer *c, const orc_orders *o,
                const orc_lineitem *l, int32_t cutoff, orc_q3_group **out);
+
+/* numeric(15,2) mode: measures mapped to scaled int64 —
+ * price in cents, discount in hundredths; revenue = Σ price_c·(100−disc_c),
+ * an EXACT integer with implied scale 1e-4 (matches PG numeric for these
+ * ranges).  Returns per-group numerators, sorted by l_orderkey. */
+typedef struct {
+    int64_t l_orderkey;
+    int32_t o_orderdate;
+    int32_t o_shippriority;
+    int64_t revenue_num;         /* scale 1e-4 */
+    int64_t nitems;
+} orc_q3n_group;
+int64_t orc_q3_numeric(const orc_customer *c, const orc_orders *o,
+                       const orc_lineitem *l, int32_t cutoff,
+                       orc_q3n_group **out);
 void orc_free(void *p);
 
 #ifdef __cplusplus

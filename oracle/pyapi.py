@@ -32,6 +32,14 @@ class _Lineitem(ctypes.Structure):
                 ("n", ctypes.c_int64)]
 
 
+class _GroupN(ctypes.Structure):
+    _fields_ = [("l_orderkey", ctypes.c_int64),
+                ("o_orderdate", ctypes.c_int32),
+                ("o_shippriority", ctypes.c_int32),
+                ("revenue_num", ctypes.c_int64),
+                ("nitems", ctypes.c_int64)]
+
+
 class _Group(ctypes.Structure):
     _fields_ = [("l_orderkey", ctypes.c_int64),
                 ("o_orderdate", ctypes.c_int32),
@@ -95,6 +103,10 @@ def _load():
     lib.orc_q3.argtypes = [ctypes.POINTER(_Customer), ctypes.POINTER(_Orders),
                            ctypes.POINTER(_Lineitem), ctypes.c_int32,
                            ctypes.POINTER(ctypes.POINTER(_Group))]
+    lib.orc_q3_numeric.restype = ctypes.c_int64
+    lib.orc_q3_numeric.argtypes = [ctypes.POINTER(_Customer), ctypes.POINTER(_Orders),
+                                   ctypes.POINTER(_Lineitem), ctypes.c_int32,
+                                   ctypes.POINTER(ctypes.POINTER(_GroupN))]
     return lib
 
 
@@ -199,6 +211,32 @@ def aocs_encode(vals):
     got = lib.orc_aocs_encode(vals.ctypes.data, width, n, 1, 32768, buf.ctypes.data, cap)
     assert got == cap, (got, cap)
     return buf.tobytes()
+
+
+def q3_numeric(cust, orders, lineitem, cutoff=None):
+    """numeric(15,2) mode: exact scaled-int64 revenue numerators."""
+    c = _as_struct(cust, _Customer, [("c_custkey", ctypes.c_int64),
+                                     ("c_mktsegment", ctypes.c_uint8)])
+    o = _as_struct(orders, _Orders, [("o_orderkey", ctypes.c_int64),
+                                     ("o_custkey", ctypes.c_int64),
+                                     ("o_orderdate", ctypes.c_int32),
+                                     ("o_shippriority", ctypes.c_int32)])
+    li = _as_struct(lineitem, _Lineitem, [("l_orderkey", ctypes.c_int64),
+                                          ("l_extendedprice", ctypes.c_double),
+                                          ("l_discount", ctypes.c_double),
+                                          ("l_shipdate", ctypes.c_int32)])
+    gp = ctypes.POINTER(_GroupN)()
+    ng = lib.orc_q3_numeric(ctypes.byref(c), ctypes.byref(o), ctypes.byref(li),
+                            CUTOFF_19950315 if cutoff is None else cutoff,
+                            ctypes.byref(gp))
+    assert ng >= 0
+    res = {"l_orderkey": np.array([gp[i].l_orderkey for i in range(ng)], np.int64),
+           "o_orderdate": np.array([gp[i].o_orderdate for i in range(ng)], np.int32),
+           "o_shippriority": np.array([gp[i].o_shippriority for i in range(ng)], np.int32),
+           "revenue_num": np.array([gp[i].revenue_num for i in range(ng)], np.int64),
+           "nitems": np.array([gp[i].nitems for i in range(ng)], np.int64)}
+    lib.orc_free(gp)
+    return res
 
 
 def aocs_encode_rle(vals):

@@ -261,3 +261,26 @@ def test_q3_properties_sf10(ctx, orc):
         assert cnt == got["nitems"][i]
         np.testing.assert_allclose(rev, got["revenue"][i], rtol=1e-9)
     q.free(); li.free(); ordr.free(); cust.free()
+
+
+# ---------------- numeric(15,2) mode (bit-exact aggregation) ----------------
+
+def test_q3_numeric_bit_exact(ctx, orc):
+    """With scaled-int64 measures the ENTIRE result is bit-exact vs the
+    oracle — including the aggregate (integer atomics are order-free)."""
+    sf = 0.2
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM_NUMERIC, sf)
+    q = ctx.q3(cust, ordr, li, numeric=True).run()
+    got = q.result()
+    want = orc.q3_numeric(orc.gen_customer(sf), orc.gen_orders(sf),
+                          orc.gen_lineitem(sf))
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["o_orderdate"], want["o_orderdate"])
+    np.testing.assert_array_equal(got["revenue_num"], want["revenue_num"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    # and a re-run is IDENTICAL bit-for-bit (integer aggregation)
+    again = q.run().result()
+    np.testing.assert_array_equal(got["revenue_num"], again["revenue_num"])
+    q.free(); li.free(); ordr.free(); cust.free()

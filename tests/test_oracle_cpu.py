@@ -319,3 +319,29 @@ def test_q3_sharded_equals_global():
     order = np.argsort(keys)
     assert (keys[order] == glob["l_orderkey"]).all()
     np.testing.assert_allclose(rev[order], glob["revenue"], rtol=1e-9)
+
+
+# ---------------- numeric(15,2) mode ----------------
+
+def test_q3_numeric_exact_vs_python():
+    """Scaled-int64 revenue numerators must be EXACTLY the integer sums a
+    pure-python recomputation gives (no tolerance)."""
+    c = orc.gen_customer(SF)
+    o = orc.gen_orders(SF)
+    li = orc.gen_lineitem(SF)
+    got = orc.q3_numeric(c, o, li)
+    f64 = orc.q3(c, o, li)
+    assert (got["l_orderkey"] == f64["l_orderkey"]).all()
+    assert (got["nitems"] == f64["nitems"]).all()
+    # cross-check vs f64 sums within rounding
+    np.testing.assert_allclose(got["revenue_num"] / 1e4, f64["revenue"], rtol=1e-9)
+    # exact recomputation for sampled groups
+    price_c = np.rint(li["l_extendedprice"] * 100).astype(np.int64)
+    disc_c = np.rint(li["l_discount"] * 100).astype(np.int64)
+    ok = li["l_shipdate"] > orc.CUTOFF_19950315
+    rng = np.random.default_rng(2)
+    for i in rng.choice(len(got["l_orderkey"]), 50, replace=False):
+        k = got["l_orderkey"][i]
+        m = ok & (li["l_orderkey"] == k)
+        want = int((price_c[m] * (100 - disc_c[m])).sum())
+        assert int(got["revenue_num"][i]) == want
