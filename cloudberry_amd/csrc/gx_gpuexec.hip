@@ -90,6 +90,12 @@ struct devbuf {
     template <typename T> T *as() const { return (T *) p; }
 };
 
+static inline int env_int(const char *name, int dflt)
+{
+    const char *v = getenv(name);
+    return v ? atoi(v) : dflt;
+}
+
 static inline int64_t pow2_at_least(int64_t want)
 {
     int64_t sz = 1024;
@@ -640,7 +646,7 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
  * key array then sits comfortably in the 256 MiB Infinity Cache), u64
  * otherwise.  Key compares stay exact either way (PG narrow-int hashing
  * spirit; sentinel 0 is safe — orderkeys start at 1). */
-template <typename KT, typename KS>
+template <typename KT, typename KS, bool CHUNKED = false>
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
@@ -651,10 +657,22 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                KT *tkey,
                                int32_t *tdate, int32_t *tprio, gx_slotmap smap)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    int64_t i, stride, iend;
+    if constexpr (CHUNKED)
+    {
+        int64_t chunk = (ok_m.nrows + gridDim.x - 1) / gridDim.x;
+        i = blockIdx.x * chunk + threadIdx.x;
+        iend = min((int64_t) blockIdx.x * chunk + chunk, ok_m.nrows);
+        stride = blockDim.x;
+    }
+    else
+    {
+        i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        iend = ok_m.nrows;
+        stride = gridDim.x * (int64_t) blockDim.x;
+    }
     uint64_t tmask = smap.mask;
-    for (; i < ok_m.nrows; i += stride)
+    for (; i < iend; i += stride)
     {
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!(od < cutoff)) continue;
@@ -1637,14 +1655,25 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
+        int ogrid = env_int("GX_ORDERS_GRID", GRID);
+        bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
         auto launch_build = [&](auto *tk, auto *cs) {
-            hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
-                                               std::decay_t<decltype(*cs)>>),
-                               dim3(GRID), dim3(TPB), 0, s,
-                               ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                               op.dstream, op.m, q->cutoff, cs, q->cmask,
-                               q->bloom, q->bwmask,
-                               tk, q->tdate, q->tprio, q->smap);
+            if (ochunk)
+                hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
+                                                   std::decay_t<decltype(*cs)>, true>),
+                                   dim3(ogrid), dim3(TPB), 0, s,
+                                   ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                                   op.dstream, op.m, q->cutoff, cs, q->cmask,
+                                   q->bloom, q->bwmask,
+                                   tk, q->tdate, q->tprio, q->smap);
+            else
+                hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
+                                                   std::decay_t<decltype(*cs)>, false>),
+                                   dim3(ogrid), dim3(TPB), 0, s,
+                                   ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                                   op.dstream, op.m, q->cutoff, cs, q->cmask,
+                                   q->bloom, q->bwmask,
+                                   tk, q->tdate, q->tprio, q->smap);
         };
         if (q->key_width == 4 && q->cset_width == 4)
             launch_build((unsigned int *) q->tkey, (const unsigned int *) q->cset);
@@ -1877,13 +1906,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
     /* ---- stage 4: extract ---- */
     HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
+    int egrid = env_int("GX_EXTRACT_GRID", GRID);
     if (q->key_width == 4)
-        hipLaunchKernelGGL(k_extract<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+        hipLaunchKernelGGL(k_extract<unsigned int>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned int *) q->tkey, q->tdate, q->tprio,
                            q->trev, q->tcnt, q->tmask + 1,
                            q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
     else
-        hipLaunchKernelGGL(k_extract<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+        hipLaunchKernelGGL(k_extract<unsigned long long>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned long long *) q->tkey, q->tdate, q->tprio,
                            q->trev, q->tcnt, q->tmask + 1,
                            q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
