@@ -1655,7 +1655,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
-        int ogrid = env_int("GX_ORDERS_GRID", GRID);
+        int ogrid = env_int("GX_ORDERS_GRID", 32768);  /* measured optimum */
         bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
         auto launch_build = [&](auto *tk, auto *cs) {
             if (ochunk)
@@ -1906,7 +1906,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
     /* ---- stage 4: extract ---- */
     HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
-    int egrid = env_int("GX_EXTRACT_GRID", GRID);
+    int egrid = env_int("GX_EXTRACT_GRID", 32768);  /* measured optimum */
     if (q->key_width == 4)
         hipLaunchKernelGGL(k_extract<unsigned int>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned int *) q->tkey, q->tdate, q->tprio,
