@@ -14,6 +14,7 @@ _SO = os.path.join(_ROOT, "cloudberry_amd", "libgpuexec.so")
 
 TPCH_CUSTOMER, TPCH_ORDERS, TPCH_LINEITEM = 0, 1, 2
 TPCH_LINEITEM_NUMERIC = 3
+TPCH_LINEITEM_RLEKEY = 4
 CUTOFF_19950315 = -1753  # DateADT of 1995-03-15 (validated vs oracle in tests)
 
 _STATUS = {0: "GX_OK", 1: "GX_ERR_HIP", 2: "GX_ERR_RCCL", 3: "GX_ERR_INVALID",

@@ -813,6 +813,212 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
     gx_wave_count_add(hits, local_hits);
 }
 
+
+/* Fused Q3 probe+agg over an RLE-compressed l_orderkey column (the real
+ * §8f-2 payoff): one workgroup per Dense/RLE block, ONE probe per RUN
+ * (lineitem is clustered by l_orderkey — the reference's own table order —
+ * so a run is one order's lineitems), and the ship/price/discount columns
+ * are touched only for runs whose key is in the join table.
+ * Fast parallel count decode relies on this writer's fixed 2-byte repeat
+ * encodings (csize == 2·ncnt detects it); other conforming streams take the
+ * serial per-block walk. */
+static constexpr int RLE_MAX_PHYS = 4096;
+
+template <typename KT>
+__global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
+                                   int64_t nblocks,
+                                   const uint8_t *pr_s, gx_colmeta pr_m,
+                                   const uint8_t *di_s, gx_colmeta di_m,
+                                   const uint8_t *sh_s, gx_colmeta sh_m,
+                                   int32_t cutoff,
+                                   const KT *tkey,
+                                   double *trev, unsigned long long *tcnt,
+                                   gx_slotmap smap,
+                                   unsigned long long *hits, int *err)
+{
+    __shared__ uint32_t s_starts[RLE_MAX_PHYS + 1];   /* exclusive row starts */
+    __shared__ uint32_t s_part[256];
+    __shared__ uint32_t s_bytepop[RLE_MAX_PHYS / 8];  /* bitmap byte popcount prefix */
+    uint64_t tmask = smap.mask;
+    unsigned long long local_hits = 0;
+
+    for (int64_t b = blockIdx.x; b < nblocks; b += gridDim.x)
+    {
+        const uint8_t *blk = lk_s + dir[b].offset;
+        const uint8_t *c = blk + 24;
+        int16_t version = ((const int16_t *) c)[0];
+        int16_t flags = ((const int16_t *) c)[1];
+        int32_t logical = ((const int32_t *) c)[1];
+        int32_t phys = ((const int32_t *) c)[2];
+        if ((version != 1 && version != 2) || (flags & 0x5) || phys > RLE_MAX_PHYS ||
+            logical != dir[b].rows)
+        {
+            if (threadIdx.x == 0) atomicOr(err, 1);
+            __syncthreads();
+            continue;
+        }
+        const int64_t *datum;
+        const uint8_t *bmp = nullptr, *cnts = nullptr;
+        int32_t ncnt = 0, csize = 0;
+        if (flags & 0x2)
+        {
+            int32_t bmbits = ((const int32_t *) c)[5];
+            ncnt = ((const int32_t *) c)[6];
+            csize = ((const int32_t *) c)[7];
+            bmp = c + 32;
+            cnts = bmp + ((bmbits + 7) >> 3);
+            int32_t hdr = 32 + ((bmbits + 7) >> 3) + csize;
+            datum = (const int64_t *) (c + ((hdr + 7) & ~7));
+            if (bmbits != phys)
+            {
+                if (threadIdx.x == 0) atomicOr(err, 1);
+                __syncthreads();
+                continue;
+            }
+        }
+        else
+            datum = (const int64_t *) (c + 16);
+
+        /* ---- phase 1: per-physical repeat counts into s_starts[p] ---- */
+        if (bmp && csize == 2 * ncnt)
+        {
+            /* parallel: byte-popcount prefix of the bitmap, then fixed-stride
+             * count lookup per ON bit */
+            int nbytes = (phys + 7) >> 3;
+            int C = (nbytes + blockDim.x - 1) / blockDim.x;
+            int lo = threadIdx.x * C, hi = min(lo + C, nbytes);
+            uint32_t acc = 0;
+            for (int i = lo; i < hi; i++)
+            {
+                s_bytepop[i] = acc;
+                acc += __popc((unsigned) bmp[i]);
+            }
+            s_part[threadIdx.x] = acc;
+            __syncthreads();
+            if (threadIdx.x == 0)
+            {
+                uint32_t run = 0;
+                for (int t = 0; t < (int) blockDim.x; t++)
+                {
+                    uint32_t v = s_part[t];
+                    s_part[t] = run;
+                    run += v;
+                }
+            }
+            __syncthreads();
+            uint32_t base = s_part[threadIdx.x];
+            for (int i = lo; i < hi; i++)
+                s_bytepop[i] += base;
+            __syncthreads();
+            for (int p = threadIdx.x; p < phys; p += blockDim.x)
+            {
+                uint32_t reps = 1;
+                uint8_t byte = bmp[p >> 3];
+                if (byte & (1u << (p & 7)))
+                {
+                    uint32_t rank = s_bytepop[p >> 3] +
+                                    __popc((unsigned) (byte & ((1u << (p & 7)) - 1)));
+                    reps += (((uint32_t) cnts[2 * rank] & 0x3Fu) << 8) |
+                            cnts[2 * rank + 1];
+                }
+                s_starts[p] = reps;
+            }
+            __syncthreads();
+        }
+        else if (bmp)
+        {
+            if (threadIdx.x == 0)   /* conforming varint stream: serial walk */
+            {
+                int32_t coff = 0;
+                for (int p = 0; p < phys; p++)
+                {
+                    uint32_t reps = 1;
+                    if (bmp[p >> 3] & (1u << (p & 7)))
+                    {
+                        int n = (cnts[coff] >> 6) + 1;
+                        uint32_t v = cnts[coff] & 0x3F;
+                        for (int i = 1; i < n; i++) v = (v << 8) | cnts[coff + i];
+                        coff += n;
+                        reps += v;
+                    }
+                    s_starts[p] = reps;
+                }
+                if (coff != csize) atomicOr(err, 1);
+            }
+            __syncthreads();
+        }
+        else
+        {
+            for (int p = threadIdx.x; p < phys; p += blockDim.x)
+                s_starts[p] = 1;
+            __syncthreads();
+        }
+
+        /* ---- phase 2: exclusive scan of repeats → row starts ---- */
+        {
+            int C = (phys + blockDim.x - 1) / blockDim.x;
+            int lo = threadIdx.x * C, hi = min(lo + C, phys);
+            uint32_t acc = 0;
+            for (int i = lo; i < hi; i++)
+            {
+                uint32_t v = s_starts[i];
+                s_starts[i] = acc;
+                acc += v;
+            }
+            s_part[threadIdx.x] = acc;
+            __syncthreads();
+            if (threadIdx.x == 0)
+            {
+                uint32_t run = 0;
+                for (int t = 0; t < (int) blockDim.x; t++)
+                {
+                    uint32_t v = s_part[t];
+                    s_part[t] = run;
+                    run += v;
+                }
+                s_starts[phys] = run;   /* == logical, checked below */
+            }
+            __syncthreads();
+            uint32_t base = s_part[threadIdx.x];
+            for (int i = lo; i < hi; i++)
+                s_starts[i] += base;
+            __syncthreads();
+            if (threadIdx.x == 0 && (int32_t) s_starts[phys] != logical)
+                atomicOr(err, 1);
+        }
+
+        /* ---- phase 3: one probe per run; rows only for table hits ---- */
+        int64_t first = dir[b].first_row;
+        for (int p = threadIdx.x; p < phys; p += blockDim.x)
+        {
+            uint64_t k = (uint64_t) datum[p];
+            uint64_t slot = smap.slot0(k);
+            bool found = false;
+            while (true)
+            {
+                KT v = tkey[slot];
+                if (v == (KT) 0) break;
+                if (v == (KT) k) { found = true; break; }
+                slot = (slot + 1) & tmask;
+            }
+            if (!found) continue;
+            uint32_t rs = s_starts[p], re = s_starts[p + 1];
+            for (uint32_t r = rs; r < re; r++)
+            {
+                int64_t g = first + r;
+                if (!(gx_col_get<int32_t>(sh_s, sh_m, g) > cutoff)) continue;
+                double price = gx_col_get<double>(pr_s, pr_m, g);
+                double disc = gx_col_get<double>(di_s, di_m, g);
+                atomicAdd(&trev[slot], price * (1.0 - disc));
+                atomicAdd(&tcnt[slot], 1ULL);
+                local_hits++;
+            }
+        }
+        __syncthreads();
+    }
+    gx_wave_count_add(hits, local_hits);
+}
+
 /* numeric(15,2) probe+agg: measures are scaled int64 (price
  * in cents, discount in hundredths); revenue numerator = Σ price_c·(100−d)
  * accumulated with integer atomics — BIT-EXACT, order-independent, equal to
@@ -1157,6 +1363,132 @@ static gx_status encode_column_device(gx_ctx *ctx, const void *dvals, int width,
     return GX_OK;
 }
 
+/* HOST-side RLE_TYPE writer (Dense_Enhanced, no-null no-delta; format notes
+ * in oracle/oracle.c — independent implementation, parity-tested against the
+ * oracle decoder).  Repeat counts are always emitted as 2-byte Int32Compress
+ * encodings (len bits = 01) — conforming, and fixed stride lets the fused
+ * scan kernel locate counts in parallel.  Runs longer than 0x3FFF extras are
+ * split.  Only needed where the device generator can't write RLE directly. */
+static void host_crc32c_table(uint32_t *tab)
+{
+    for (uint32_t i = 0; i < 256; i++)
+    {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++)
+            c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+        tab[i] = c;
+    }
+}
+static uint32_t host_crc32c(const uint32_t *tab, uint32_t crc,
+                            const uint8_t *p, int64_t len)
+{
+    while (len--) crc = tab[(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+    return crc;
+}
+
+static int64_t host_rle_encode(const uint8_t *src_v, int width, int64_t nrows,
+                               int32_t blocksize, std::vector<uint8_t> &out)
+{
+    uint32_t tab[256];
+    host_crc32c_table(tab);
+    const int32_t maxdata = blocksize - 32;
+    int64_t row = 0;
+    std::vector<uint8_t> pvals;
+    std::vector<int32_t> extra;
+    while (row < nrows)
+    {
+        pvals.clear();
+        extra.clear();
+        int32_t non = 0;
+        int64_t logical = 0;
+        while (row + logical < nrows)
+        {
+            const uint8_t *d = src_v + (row + logical) * width;
+            int phys = (int) extra.size();
+            bool same = phys > 0 &&
+                        memcmp(&pvals[(size_t) (phys - 1) * width], d, width) == 0 &&
+                        extra[phys - 1] < 0x3FFF;
+            int new_phys = phys + (same ? 0 : 1);
+            int new_non = non + (same && extra[phys - 1] == 0 ? 1 : 0);
+            int32_t bm = (new_phys + 7) >> 3;
+            int32_t hdr = 16 + (new_non ? 16 + bm + 2 * new_non : 0);
+            int64_t tot = ((hdr + 7) & ~7) + (int64_t) new_phys * width;
+            if (tot >= maxdata && logical > 0)
+                break;
+            if (same)
+            {
+                if (extra[phys - 1] == 0) non++;
+                extra[phys - 1]++;
+            }
+            else
+            {
+                pvals.insert(pvals.end(), d, d + width);
+                extra.push_back(0);
+            }
+            logical++;
+        }
+        int phys = (int) extra.size();
+        bool has_rle = non > 0;
+        int32_t bm = (phys + 7) >> 3;
+        int32_t hdr = 16 + (has_rle ? 16 + bm + 2 * non : 0);
+        int32_t datum_off = (hdr + 7) & ~7;
+        int32_t content = datum_off + phys * width;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        size_t base = out.size();
+        out.resize(base + blocklen, 0);
+        uint8_t *blk = out.data() + base;
+        uint32_t kind = (logical <= 16383) ? 1u : 3u;
+        uint32_t b03 = (kind << 28) | (1u << 27) | (1u << 24);
+        uint32_t b47 = 0;
+        if (kind == 1)
+        {
+            b03 |= (0x00FFFC00u & ((uint32_t) logical << 10)) |
+                   (((uint32_t) content >> 11) & 0x3FFu);
+            b47 = ((uint32_t) content & 0x7FFu) << 21;
+        }
+        else
+        {
+            b03 |= ((uint32_t) content & 0x1FFFFFu);
+            b47 = (uint32_t) logical & 0x3FFFFFFFu;
+        }
+        memcpy(blk, &b03, 4);
+        memcpy(blk + 4, &b47, 4);
+        int64_t frn = row + 1;
+        memcpy(blk + 16, &frn, 8);
+        uint8_t *c = blk + 24;
+        int16_t v16 = 2; memcpy(c, &v16, 2);
+        v16 = has_rle ? 2 : 0; memcpy(c + 2, &v16, 2);
+        int32_t v32 = (int32_t) logical; memcpy(c + 4, &v32, 4);
+        v32 = phys; memcpy(c + 8, &v32, 4);
+        v32 = phys * width; memcpy(c + 12, &v32, 4);
+        if (has_rle)
+        {
+            v32 = 0; memcpy(c + 16, &v32, 4);
+            v32 = phys; memcpy(c + 20, &v32, 4);
+            v32 = non; memcpy(c + 24, &v32, 4);
+            v32 = 2 * non; memcpy(c + 28, &v32, 4);
+            uint8_t *bmp = c + 32;
+            uint8_t *cnts = bmp + bm;
+            int w = 0;
+            for (int pi = 0; pi < phys; pi++)
+                if (extra[pi] > 0)
+                {
+                    bmp[pi >> 3] |= (uint8_t) (1u << (pi & 7));
+                    cnts[w] = (uint8_t) ((1 << 6) | (extra[pi] >> 8));
+                    cnts[w + 1] = (uint8_t) extra[pi];
+                    w += 2;
+                }
+        }
+        memcpy(c + datum_off, pvals.data(), (size_t) phys * width);
+        uint32_t crc = host_crc32c(tab, 0xFFFFFFFFu, blk + 16, blocklen - 16);
+        memcpy(blk + 8, &crc, 4);
+        crc = host_crc32c(tab, 0xFFFFFFFFu, blk, 12);
+        memcpy(blk + 12, &crc, 4);
+        row += logical;
+    }
+    return (int64_t) out.size();
+}
+
 /* walk a stream's AO envelope headers on the HOST, building the per-block
  * directory a variable-geometry (Dense/RLE) stream needs */
 static gx_status parse_block_dir(const uint8_t *s, int64_t nbytes,
@@ -1297,7 +1629,8 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     int64_t nord = (int64_t) (1500000.0 * sf + 0.5);
     int64_t nglobal = (which == GX_TPCH_CUSTOMER) ? ncust : nord;
     bool li_numeric = ((int) which == 3);   /* GX_TPCH_LINEITEM_NUMERIC */
-    if (li_numeric) which = GX_TPCH_LINEITEM;
+    bool li_rlekey = ((int) which == 4);    /* GX_TPCH_LINEITEM_RLEKEY */
+    if (li_numeric || li_rlekey) which = GX_TPCH_LINEITEM;
     int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
     int64_t blocks = (nthreads + TPB - 1) / TPB;
     /* the count kernels write counts[t] for EVERY launched thread */
@@ -1373,7 +1706,39 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
             hipLaunchKernelGGL(k_emit_li, dim3(blocks), dim3(TPB), 0, ctx->stream,
                                seed, nglobal, ctx->seg, ctx->nsegs, doffs, dlk, dpr, ddi, dsh);
         gx_col c0, c1, c2, c3;
-        st = encode_column_device(ctx, dlk, 8, n, &c0);
+        if (li_rlekey)
+        {
+            /* keys → host, RLE-encode, re-upload with a block directory */
+            std::vector<int64_t> hkeys(n);
+            HIP_CHK(ctx, hipMemcpyAsync(hkeys.data(), dlk, n * 8,
+                                        hipMemcpyDeviceToHost, ctx->stream));
+            HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+            std::vector<uint8_t> stream_bytes;
+            host_rle_encode((const uint8_t *) hkeys.data(), 8, n, 32768, stream_bytes);
+            std::vector<gx_blockref> dir;
+            int64_t rows = 0;
+            if (parse_block_dir(stream_bytes.data(), (int64_t) stream_bytes.size(),
+                                dir, &rows) != GX_OK || rows != n)
+            { set_err(ctx, "rle self-encode mismatch%s", ""); delete t; hipFree(dcounts); return GX_ERR_INVALID; }
+            c0.format = 1;
+            c0.nblocks = (int64_t) dir.size();
+            c0.m.width = 8;
+            c0.m.rpb = gx_aocs_rows_per_block(8, 32768);
+            c0.m.nrows = n;
+            c0.m.full_block_len = gx_aocs_block_len(8, c0.m.rpb);
+            c0.m.nbytes = (int64_t) stream_bytes.size();
+            gx_colmeta_finish(&c0.m);
+            HIP_CHK(ctx, hipMalloc(&c0.dstream, stream_bytes.size()));
+            HIP_CHK(ctx, hipMalloc(&c0.ddir, dir.size() * sizeof(gx_blockref)));
+            HIP_CHK(ctx, hipMemcpyAsync(c0.dstream, stream_bytes.data(),
+                                        stream_bytes.size(), hipMemcpyHostToDevice,
+                                        ctx->stream));
+            HIP_CHK(ctx, hipMemcpyAsync(c0.ddir, dir.data(),
+                                        dir.size() * sizeof(gx_blockref),
+                                        hipMemcpyHostToDevice, ctx->stream));
+        }
+        else
+            st = encode_column_device(ctx, dlk, 8, n, &c0);
         if (st == GX_OK) st = encode_column_device(ctx, dpr, 8, n, &c1);
         if (st == GX_OK) st = encode_column_device(ctx, ddi, 8, n, &c2);
         if (st == GX_OK) st = encode_column_device(ctx, dsh, 4, n, &c3);
@@ -1470,12 +1835,13 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
     if (!ctx || !customer || !orders || !lineitem) return GX_ERR_INVALID;
     if (customer->cols.size() != 2 || orders->cols.size() != 4 ||
         lineitem->cols.size() != 4) return GX_ERR_INVALID;
+    /* fused-RLE scan is supported for lineitem's l_orderkey column only */
     for (auto *t : {customer, orders, lineitem})
-        for (auto &c : t->cols)
-            if (c.format != 0)
+        for (size_t ci = 0; ci < t->cols.size(); ci++)
+            if (t->cols[ci].format != 0 && !(t == lineitem && ci == 0))
             {
-                set_err(ctx, "Q3 pipeline requires fixed-format (Orig) streams; "
-                             "decode RLE columns first%s", "");
+                set_err(ctx, "Q3 pipeline supports RLE only on l_orderkey; "
+                             "decode other RLE columns first%s", "");
                 return GX_ERR_INVALID;
             }
     gx_q3 *q = new gx_q3();
@@ -1839,7 +2205,30 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     /* ---- stage 3: lineitem scan+probe+agg (dominant kernel) ---- */
     unsigned long long *dhits = q->dhits;
     HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
-    if (q->numeric)
+    if (lk.format == 1)
+    {
+        if (q->numeric) { set_err(ctx, "numeric+RLE combo not supported%s", ""); return GX_ERR_INVALID; }
+        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
+        int64_t nb = lk.nblocks;
+        int rgrid = (int) std::min<int64_t>(nb, 16384);
+        if (q->key_width == 4)
+            hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned int>, dim3(rgrid), dim3(TPB), 0, s,
+                               lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
+                               ld.dstream, ld.m, ls.dstream, ls.m, q->cutoff,
+                               (const unsigned int *) q->tkey,
+                               q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
+        else
+            hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned long long>, dim3(rgrid), dim3(TPB), 0, s,
+                               lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
+                               ld.dstream, ld.m, ls.dstream, ls.m, q->cutoff,
+                               (const unsigned long long *) q->tkey,
+                               q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
+        int herr = 0;
+        HIP_CHK(ctx, hipMemcpyAsync(&herr, q->dmin, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        if (herr) { set_err(ctx, "malformed RLE block in fused scan%s", ""); return GX_ERR_INVALID; }
+    }
+    else if (q->numeric)
     {
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         if (q->key_width == 4)

@@ -100,7 +100,8 @@ gx_status gx_table_logical_bytes(const gx_table *t, double *out);
 /* synthetic TPC-H-shaped tables generated AND AOCS-encoded on device
  * (deterministic; identical formulas to oracle/oracle.c datagen) */
 typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2,
-               GX_TPCH_LINEITEM_NUMERIC = 3 /* measures as scaled int64 */ } gx_tpch_table;
+               GX_TPCH_LINEITEM_NUMERIC = 3, /* measures as scaled int64 */
+               GX_TPCH_LINEITEM_RLEKEY = 4   /* l_orderkey RLE-compressed */ } gx_tpch_table;
 gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                       uint64_t seed, gx_table **out);
 

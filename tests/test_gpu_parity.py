@@ -284,3 +284,26 @@ def test_q3_numeric_bit_exact(ctx, orc):
     again = q.run().result()
     np.testing.assert_array_equal(got["revenue_num"], again["revenue_num"])
     q.free(); li.free(); ordr.free(); cust.free()
+
+
+# ---------------- fused Q3 scan over RLE-compressed l_orderkey ----------------
+
+def test_q3_rle_keys_parity(ctx, orc):
+    """The run-iterating probe kernel over an RLE l_orderkey column must give
+    identical results to the flat-scan pipeline and the oracle."""
+    for sf in (0.01, 0.3):
+        cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+        ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+        li = ctx.tpch_gen(gx.TPCH_LINEITEM_RLEKEY, sf)
+        # the RLE stream itself decodes to the oracle's keys
+        want_li = orc.gen_lineitem(sf)
+        np.testing.assert_array_equal(li.decode_column(0, np.int64, verify=True),
+                                      want_li["l_orderkey"])
+        q = ctx.q3(cust, ordr, li).run()
+        got = q.result()
+        want = orc.q3(orc.gen_customer(sf), orc.gen_orders(sf), want_li)
+        np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+        np.testing.assert_array_equal(got["nitems"], want["nitems"])
+        np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-6)
+        assert q.stats()["probe_hits"] == int(want["nitems"].sum())
+        q.free(); li.free(); ordr.free(); cust.free()
