@@ -749,6 +749,62 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -4)
+    {
+        /* WAVE-batched: each BLOCK window covers blockDim.x*4 consecutive
+         * rows; lane handles rows {w + tid + b*blockDim.x} — every load
+         * instruction still spans 64 consecutive rows (coalescing identical
+         * to B=1) while each lane keeps 4 independent chains in flight. */
+        const int WB = 4;
+        int64_t win = (int64_t) blockDim.x * WB;
+        int64_t w0 = blockIdx.x * win;
+        int64_t stride = (int64_t) gridDim.x * win;
+        for (; w0 + win <= lk_m.nrows; w0 += stride)
+        {
+            int32_t ship[WB];
+            int64_t key[WB];
+            int64_t r[WB];
+#pragma unroll
+            for (int b = 0; b < WB; b++)
+            {
+                r[b] = w0 + threadIdx.x + (int64_t) b * blockDim.x;
+                ship[b] = gx_col_get<int32_t>(sh_s, sh_m, r[b]);
+            }
+#pragma unroll
+            for (int b = 0; b < WB; b++)
+                key[b] = gx_col_get<int64_t>(lk_s, lk_m, r[b]);
+            bool pass[WB];
+            uint64_t slot[WB];
+            KT v[WB];
+#pragma unroll
+            for (int b = 0; b < WB; b++)
+            {
+                pass[b] = ship[b] > cutoff;
+                slot[b] = pass[b] ? smap.slot0((uint64_t) key[b]) : 0;
+            }
+#pragma unroll
+            for (int b = 0; b < WB; b++)
+                v[b] = tkey[slot[b]];
+#pragma unroll
+            for (int b = 0; b < WB; b++)
+            {
+                if (!pass[b]) continue;
+                uint64_t res = resolve((uint64_t) key[b], slot[b], v[b]);
+                if (res != ~0ULL) hit(res, r[b]);
+            }
+        }
+        /* tail: rows the full windows missed */
+        int64_t done = (lk_m.nrows / win) * win;
+        for (int64_t i = done + blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+             i < lk_m.nrows; i += gridDim.x * (int64_t) blockDim.x)
+        {
+            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t s0 = smap.slot0(k);
+            uint64_t res = resolve(k, s0, tkey[s0]);
+            if (res != ~0ULL) hit(res, i);
+        }
+    }
     else if constexpr (B == -1)
     {
         /* block-chunked: each workgroup owns a contiguous row range, so with
@@ -2275,6 +2331,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned int>, keys); break;
                 case 6: launch(k_li_probe_agg_t<-1, unsigned int>, keys); break;
+                case 7: launch(k_li_probe_agg_t<-4, unsigned int>, keys); break;
             }
         }
         else
@@ -2288,6 +2345,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned long long>, keys); break;
                 case 6: launch(k_li_probe_agg_t<-1, unsigned long long>, keys); break;
+                case 7: launch(k_li_probe_agg_t<-4, unsigned long long>, keys); break;
             }
         }
     }
