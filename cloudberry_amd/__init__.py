@@ -93,6 +93,8 @@ def _load():
     lib.gx_q3_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
     lib.gx_q3_result.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.POINTER(_Group)),
                                  ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_q3_topn.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+                               ctypes.POINTER(ctypes.c_int64)]
     lib.gx_q3_free.argtypes = [ctypes.c_void_p]
     lib.gx_free.argtypes = [ctypes.c_void_p]
     lib.gx_test_motion1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
@@ -248,6 +250,18 @@ class Q3:
                "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
         self.ctx._lib.gx_free(gp)
         return res
+
+    def topn(self, n=10):
+        out = (_Group * n)()
+        m = ctypes.c_int64()
+        self.ctx._chk(self.ctx._lib.gx_q3_topn(self._q, n, out, ctypes.byref(m)))
+        m = m.value
+        return {"l_orderkey": np.array([out[i].l_orderkey for i in range(m)], np.int64),
+                "o_orderdate": np.array([out[i].o_orderdate for i in range(m)], np.int32),
+                "o_shippriority": np.array([out[i].o_shippriority for i in range(m)], np.int32),
+                "revenue": np.array([out[i].revenue for i in range(m)], np.float64),
+                "revenue_num": np.array([out[i].revenue_num for i in range(m)], np.int64),
+                "nitems": np.array([out[i].nitems for i in range(m)], np.int64)}
 
     def free(self):
         if self._q:

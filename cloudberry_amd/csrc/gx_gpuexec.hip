@@ -1281,6 +1281,93 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
     }
 }
 
+/* Q3's final ORDER BY revenue DESC, o_orderdate ASC LIMIT K (SURVEY §8f-3;
+ * nodeSort.c/nodeLimit.c territory in the reference, run on the QD over
+ * trivial row counts).  Each THREAD keeps an exact top-K of its strided
+ * slice (insertion against the current K-th; almost every element fails the
+ * first compare), each BLOCK then selects top-K of its threads' candidates,
+ * and the host merges blocks×K.  Any global top-K member survives each
+ * stage, so the result is exact. */
+static constexpr int TOPK = 10;
+
+__device__ __forceinline__ bool topn_less(double ra, int32_t da,
+                                          double rb, int32_t db)
+{
+    /* "a ranks after b"? ORDER BY revenue DESC, o_orderdate ASC */
+    if (ra != rb) return ra < rb;
+    return da > db;
+}
+
+__global__ void __launch_bounds__(64)
+k_topn(const int64_t *okey, const int32_t *odate,
+       const int32_t *oprio, const double *rev,
+       const int64_t *cnt, int64_t n,
+       int64_t *c_okey, int32_t *c_odate, int32_t *c_oprio,
+       double *c_rev, int64_t *c_cnt)
+{
+    /* one wave per block: 64 threads × K candidates → 640 in LDS */
+    __shared__ double s_rev[64 * TOPK];
+    __shared__ int32_t s_date[64 * TOPK];
+    __shared__ int64_t s_idx[64 * TOPK];
+    int tid = threadIdx.x;
+    double t_rev[TOPK];
+    int32_t t_date[TOPK];
+    int64_t t_idx[TOPK];
+#pragma unroll
+    for (int k = 0; k < TOPK; k++) { t_rev[k] = -1.0; t_idx[k] = -1; t_date[k] = 0; }
+
+    int64_t start = blockIdx.x * (int64_t) blockDim.x + tid;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (int64_t i = start; i < n; i += stride)
+    {
+        double r = rev[i];
+        int32_t d = odate[i];
+        /* fails here for all but ~K·blocks·log(n) elements */
+        if (t_idx[TOPK - 1] >= 0 && topn_less(r, d, t_rev[TOPK - 1], t_date[TOPK - 1]))
+            continue;
+        int k = TOPK - 1;
+        while (k > 0 && (t_idx[k - 1] < 0 ||
+                         topn_less(t_rev[k - 1], t_date[k - 1], r, d)))
+        {
+            t_rev[k] = t_rev[k - 1]; t_date[k] = t_date[k - 1]; t_idx[k] = t_idx[k - 1];
+            k--;
+        }
+        t_rev[k] = r; t_date[k] = d; t_idx[k] = i;
+    }
+#pragma unroll
+    for (int k = 0; k < TOPK; k++)
+    {
+        s_rev[tid * TOPK + k] = t_rev[k];
+        s_date[tid * TOPK + k] = t_date[k];
+        s_idx[tid * TOPK + k] = t_idx[k];
+    }
+    __syncthreads();
+    if (tid == 0)
+    {
+        for (int k = 0; k < TOPK; k++)
+        {
+            int best = -1;
+            for (int j = 0; j < 64 * TOPK; j++)
+            {
+                if (s_idx[j] < 0) continue;
+                if (best < 0 || topn_less(s_rev[best], s_date[best],
+                                          s_rev[j], s_date[j]))
+                    best = j;
+            }
+            int64_t w = (int64_t) blockIdx.x * TOPK + k;
+            if (best < 0) { c_okey[w] = -1; c_rev[w] = -1.0; c_odate[w] = 0;
+                            c_oprio[w] = 0; c_cnt[w] = 0; continue; }
+            int64_t ib = s_idx[best];
+            c_okey[w] = okey[ib];
+            c_odate[w] = odate[ib];
+            c_oprio[w] = oprio[ib];
+            c_rev[w] = rev[ib];
+            c_cnt[w] = cnt[ib];
+            s_idx[best] = -1;
+        }
+    }
+}
+
 /* ================= host-side structures ================= */
 
 struct gx_col {
@@ -2451,6 +2538,67 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
     }
     *out = g;
     *ngroups = n;
+    return GX_OK;
+}
+
+/* top-N of THIS segment's groups (ORDER BY revenue DESC, o_orderdate ASC);
+ * device-side per-block selection, host merge of blocks×K candidates.
+ * At nsegs>1 each rank returns its local top-N; the coordinator-side merge
+ * of nsegs×N rows is the reference's final Gather/Limit (trivial). */
+extern "C" gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *nout)
+{
+    if (!q || !q->ran || topn <= 0 || topn > 10) return GX_ERR_STATE;
+    gx_ctx *ctx = q->ctx;
+    hipStream_t s = ctx->stream;
+    int64_t n = q->ngroups;
+    int grid = 256;
+    devbuf c_okey, c_odate, c_oprio, c_rev, c_cnt;
+    HIP_CHK(ctx, c_okey.alloc(grid * 10 * 8));
+    HIP_CHK(ctx, c_odate.alloc(grid * 10 * 4));
+    HIP_CHK(ctx, c_oprio.alloc(grid * 10 * 4));
+    HIP_CHK(ctx, c_rev.alloc(grid * 10 * 8));
+    HIP_CHK(ctx, c_cnt.alloc(grid * 10 * 8));
+    hipLaunchKernelGGL(k_topn, dim3(grid), dim3(64), 0, s,
+                       q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, n,
+                       c_okey.as<int64_t>(), c_odate.as<int32_t>(),
+                       c_oprio.as<int32_t>(), c_rev.as<double>(),
+                       c_cnt.as<int64_t>());
+    std::vector<int64_t> hk(grid * 10), hc(grid * 10);
+    std::vector<int32_t> hd(grid * 10), hp(grid * 10);
+    std::vector<double> hr(grid * 10);
+    HIP_CHK(ctx, hipMemcpyAsync(hk.data(), c_okey.p, grid * 10 * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hd.data(), c_odate.p, grid * 10 * 4, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hp.data(), c_oprio.p, grid * 10 * 4, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hr.data(), c_rev.p, grid * 10 * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hc.data(), c_cnt.p, grid * 10 * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    std::vector<int> idx;
+    for (int i = 0; i < grid * 10; i++)
+        if (hk[i] >= 0) idx.push_back(i);
+    std::sort(idx.begin(), idx.end(), [&](int a, int b) {
+        if (hr[a] != hr[b]) return hr[a] > hr[b];
+        return hd[a] < hd[b];
+    });
+    int64_t m = std::min<int64_t>(topn, (int64_t) idx.size());
+    for (int64_t i = 0; i < m; i++)
+    {
+        out[i].l_orderkey = hk[idx[i]];
+        out[i].o_orderdate = hd[idx[i]];
+        out[i].o_shippriority = hp[idx[i]];
+        out[i].revenue = q->numeric ? 0.0 : hr[idx[i]];
+        if (q->numeric)
+        {
+            int64_t num;
+            memcpy(&num, &hr[idx[i]], 8);
+            out[i].revenue_num = num;
+            out[i].revenue = (double) num / 1e4;
+        }
+        else
+            out[i].revenue_num = 0;
+        out[i].nitems = hc[idx[i]];
+    }
+    *nout = m;
     return GX_OK;
 }
 

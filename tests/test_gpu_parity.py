@@ -155,12 +155,19 @@ def test_q3_top10_matches_oracle(ctx, orc):
     cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
     ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
     li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
-    got = _top10(ctx.q3(cust, ordr, li).run().result())
+    q = ctx.q3(cust, ordr, li).run()
+    got = _top10(q.result())
     want = _top10(orc.q3(orc.gen_customer(sf), orc.gen_orders(sf),
                          orc.gen_lineitem(sf)))
     np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
     np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-6)
-    li.free(); ordr.free(); cust.free()
+    # the DEVICE top-N kernel (gx_q3_topn) must agree with the host sort
+    dev = q.topn(10)
+    np.testing.assert_array_equal(dev["l_orderkey"], got["l_orderkey"])
+    np.testing.assert_array_equal(dev["o_orderdate"], got["o_orderdate"])
+    np.testing.assert_array_equal(dev["nitems"], got["nitems"])
+    np.testing.assert_allclose(dev["revenue"], got["revenue"], rtol=0)
+    q.free(); li.free(); ordr.free(); cust.free()
 
 
 # ---------------- Motion partition kernels (one GPU) ----------------
