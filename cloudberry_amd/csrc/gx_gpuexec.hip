@@ -392,10 +392,10 @@ __global__ void k_verify_crc_dir(const uint8_t *stream, const gx_blockref *dir,
     }
 }
 
-/* Dense(±RLE) block decode — one THREAD per AO block (blocks decode in
- * parallel across the grid; within a block the varint walk is inherently
- * serial).  Format: oracle/oracle.c RLE codec comments; reader semantics
- * datumstreamblock.h:1724-1912. */
+/* Dense(±RLE±DELTA) block decode — one THREAD per AO block (blocks decode
+ * in parallel across the grid; the per-block walk is inherently serial).
+ * Walker mirrors DatumStreamBlockRead_AdvanceDense/…DenseDelta
+ * (datumstreamblock.h:1624-1912) and oracle/oracle.c decode_dense_content. */
 template <typename T>
 __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
                                int64_t nblocks, int64_t nrows, T *out, int *err)
@@ -411,8 +411,7 @@ __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
         int32_t phys = ((const int32_t *) c)[2];
         int32_t psize = ((const int32_t *) c)[3];
         T *dst = out + dir[b].first_row;
-        if (dir[b].first_row + logical > nrows || logical != dir[b].rows ||
-            psize != phys * (int32_t) sizeof(T))
+        if (dir[b].first_row + logical > nrows || logical != dir[b].rows)
         { atomicOr(err, 1); continue; }
         if (version == 0)
         {
@@ -423,47 +422,85 @@ __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
             for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
             continue;
         }
-        if (version != 1 && version != 2) { atomicOr(err, 1); continue; }
-        if (flags & 0x1) { atomicOr(err, 1); continue; }   /* nulls: not yet */
-        if (!(flags & 0x2))
+        if ((version != 1 && version != 2) || (flags & 0x1) ||
+            psize != phys * (int32_t) sizeof(T))
+        { atomicOr(err, 1); continue; }
+        bool rle = (flags & 0x2) != 0, delta = (flags & 0x4) != 0;
+        if (!rle && !delta)
         {
             if (logical != phys) { atomicOr(err, 1); continue; }
             const T *d = (const T *) (c + 16);
             for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
             continue;
         }
-        if (flags & 0x4) { atomicOr(err, 2); continue; }   /* delta: round 2 */
-        int32_t bmbits = ((const int32_t *) c)[5];
-        int32_t csize = ((const int32_t *) c)[7];
-        if (((const int32_t *) c)[4] != 0 || bmbits != phys)
-        { atomicOr(err, 1); continue; }
-        const uint8_t *bmp = c + 32;
-        const uint8_t *cnts = bmp + ((bmbits + 7) >> 3);
-        int32_t hdr = 32 + ((bmbits + 7) >> 3) + csize;
-        const T *datum = (const T *) (c + ((hdr + 7) & ~7));
-        int64_t w = 0;
-        int32_t coff = 0;
-        for (int32_t p = 0; p < phys; p++)
+        const uint8_t *p = c + 16;
+        int32_t bmbits = 0, csize = 0, dbmbits = 0, dsize = 0;
+        if (rle)
         {
+            if (((const int32_t *) p)[0] != 0) { atomicOr(err, 1); continue; }
+            bmbits = ((const int32_t *) p)[1];
+            csize = ((const int32_t *) p)[3];
+            p += 16;
+        }
+        if (delta)
+        {
+            dbmbits = *(const int32_t *) p;
+            dsize = ((const int32_t *) p)[2];
+            p += 12;
+        }
+        const uint8_t *bmp = nullptr, *cnts = nullptr, *dbm = nullptr, *dbs = nullptr;
+        if (rle) { bmp = p; p += (bmbits + 7) >> 3; cnts = p; p += csize; }
+        if (delta) { dbm = p; p += (dbmbits + 7) >> 3; dbs = p; p += dsize; }
+        int32_t hdr = (int32_t) (p - c);
+        const T *datum = (const T *) (c + ((hdr + 7) & ~7));
+
+        int64_t w = 0;
+        int32_t item = 0, phys_idx = 0, coff = 0, doff = 0;
+        T cur = (T) 0;
+        bool bad = false;
+        while (w < logical)
+        {
+            if ((rle && item >= bmbits) || (delta && item >= dbmbits))
+            { bad = true; break; }
             int64_t reps = 1;
-            if (bmp[p >> 3] & (1u << (p & 7)))
+            if (rle && (bmp[item >> 3] & (1u << (item & 7))))
             {
                 int32_t n = (cnts[coff] >> 6) + 1;
-                int32_t v = cnts[coff] & 0x3F;
+                uint32_t v = cnts[coff] & 0x3F;
                 for (int32_t i = 1; i < n; i++) v = (v << 8) | cnts[coff + i];
                 coff += n;
                 reps += v;
             }
-            if (w + reps > logical) { atomicOr(err, 1); break; }
-            T val = datum[p];
-            for (int64_t r = 0; r < reps; r++) dst[w + r] = val;
+            if (delta && (dbm[item >> 3] & (1u << (item & 7))))
+            {
+                int32_t n = (dbs[doff] >> 6) + 1;
+                bool pos = (dbs[doff] >> 5) & 1;
+                uint64_t mag = dbs[doff] & 0x1F;
+                for (int32_t i = 1; i < n; i++) mag = (mag << 8) | dbs[doff + i];
+                doff += n;
+                if constexpr (sizeof(T) == 8)
+                    cur = (T) (pos ? (uint64_t) cur + mag : (uint64_t) cur - mag);
+                else
+                    cur = (T) (pos ? (uint32_t) cur + (uint32_t) mag
+                                   : (uint32_t) cur - (uint32_t) mag);
+            }
+            else
+            {
+                if (phys_idx >= phys) { bad = true; break; }
+                cur = datum[phys_idx++];
+            }
+            if (w + reps > logical) { bad = true; break; }
+            for (int64_t r = 0; r < reps; r++) dst[w + r] = cur;
             w += reps;
+            item++;
         }
-        if (w != logical || coff != csize) atomicOr(err, 1);
+        if (bad || w != logical || phys_idx != phys ||
+            (rle && coff != csize) || (delta && doff != dsize))
+            atomicOr(err, 1);
     }
 }
 
-/* ================= Motion routing ================= */
+/* ================= Motion routing ================= *//* ================= Motion routing ================= */
 
 __global__ void k_route(const int64_t *keys, int64_t n, int32_t nsegs, int32_t *out)
 {

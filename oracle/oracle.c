@@ -648,7 +648,214 @@ int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
     return off;
 }
 
-/* decode one Dense(±RLE) content area; returns rows written or -1 */
+
+/* sign+magnitude delta varint (Reserved3): top 2 bits = len-1, bit 5 =
+ * POSITIVE flag — datumstreamblock.h:790-930 */
+static int32_t varint3_size(int64_t mag)
+{
+    if (mag <= 0x1F) return 1;
+    if (mag <= 0x1FFF) return 2;
+    if (mag <= 0x1FFFFF) return 3;
+    return 4;
+}
+static int32_t varint3_encode(uint8_t *b, int64_t mag, int positive)
+{
+    int32_t n;
+    if (mag <= 0x1F) { b[0] = (uint8_t) mag; n = 1; }
+    else if (mag <= 0x1FFF) { b[0] = (1 << 6) | (uint8_t) (mag >> 8); b[1] = (uint8_t) mag; n = 2; }
+    else if (mag <= 0x1FFFFF) { b[0] = (2 << 6) | (uint8_t) (mag >> 16); b[1] = (uint8_t) (mag >> 8); b[2] = (uint8_t) mag; n = 3; }
+    else { b[0] = (3 << 6) | (uint8_t) (mag >> 24); b[1] = (uint8_t) (mag >> 16); b[2] = (uint8_t) (mag >> 8); b[3] = (uint8_t) mag; n = 4; }
+    if (positive) b[0] |= 0x20;
+    return n;
+}
+static int64_t varint3_decode(const uint8_t *b, int32_t *len, int *positive)
+{
+    int32_t n = (b[0] >> 6) + 1;
+    *positive = (b[0] >> 5) & 1;
+    int64_t v = b[0] & 0x1F;
+    for (int32_t i = 1; i < n; i++) v = (v << 8) | b[i];
+    *len = n;
+    return v;
+}
+
+/* load/advance helpers for width-generic item compare */
+static uint64_t item_at(const uint8_t *v, int width, int64_t i)
+{
+    uint64_t x = 0;
+    memcpy(&x, v + i * width, width);
+    return x;
+}
+
+int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
+                                  int64_t first_rownum, int32_t blocksize,
+                                  uint8_t *out, int64_t outcap)
+{
+    const uint8_t *src = (const uint8_t *) vals;
+    int32_t maxdata = blocksize - 32;
+    int64_t off = 0, row = 0;
+    int cap_items = blocksize * 2;
+    uint8_t *pvals = malloc((size_t) cap_items * width);  /* physical datums */
+    int32_t *extra = malloc(sizeof(int32_t) * cap_items); /* repeats per NEW item */
+    uint8_t *isdelta = malloc(cap_items);                 /* per NEW item */
+    uint8_t *dbytes = malloc((size_t) cap_items * 4);     /* delta varints */
+
+    while (row < nrows)
+    {
+        int32_t items = 0, phys = 0, non = 0, ndelta = 0, dsize = 0, csize = 0;
+        int64_t logical = 0;
+        uint64_t prev = 0;
+        while (row + logical < nrows && items < cap_items)
+        {
+            uint64_t v = item_at(src, width, row + logical);
+            int same = items > 0 && v == prev && extra[items - 1] < 0x3FFFFFFF;
+            if (same)
+            {
+                /* capacity: count growth */
+                int32_t grow = (extra[items - 1] == 0) ? 1 :
+                               varint_size(extra[items - 1] + 1) - varint_size(extra[items - 1]);
+                int32_t bm_items = ((items + 7) >> 3) * 2;   /* compress + delta bitmaps */
+                int32_t hdr = 16 + 16 + 12 + bm_items + (csize + grow) + dsize;
+                int64_t tot = ((hdr + 7) & ~7) + (int64_t) phys * width;
+                if (tot >= maxdata && logical > 0) break;
+                if (extra[items - 1] == 0) non++;
+                csize += grow;
+                extra[items - 1]++;
+                logical++;
+                continue;
+            }
+            /* new item: delta or physical */
+            int64_t d = 0;
+            int use_delta = 0;
+            if (items > 0)
+            {
+                if (width == 8)
+                    d = (int64_t) (v - prev);
+                else
+                    d = (int64_t) (int32_t) ((uint32_t) v - (uint32_t) prev);
+                int64_t mag = d < 0 ? -d : d;
+                if (mag <= 0x1FFFFFFF && d != 0) use_delta = 1;
+                if (d == 0) use_delta = 0;   /* handled by RLE above anyway */
+                if (use_delta) d = mag * (d < 0 ? -1 : 1);
+            }
+            int32_t dvn = use_delta ? varint3_size(d < 0 ? -d : d) : 0;
+            int32_t new_phys = phys + (use_delta ? 0 : 1);
+            int32_t bm_items = (((items + 1) + 7) >> 3) * 2;
+            int32_t hdr = 16 + 16 + 12 + bm_items + csize + dsize + dvn;
+            int64_t tot = ((hdr + 7) & ~7) + (int64_t) new_phys * width;
+            if (tot >= maxdata && logical > 0) break;
+            isdelta[items] = (uint8_t) use_delta;
+            extra[items] = 0;
+            if (use_delta)
+            {
+                dsize += varint3_encode(dbytes + dsize, d < 0 ? -d : d, d > 0);
+                ndelta++;
+            }
+            else
+            {
+                memcpy(pvals + (size_t) phys * width, &v, width);
+                phys++;
+            }
+            items++;
+            prev = v;
+            logical++;
+        }
+
+        int has_rle = non > 0;
+        int has_delta = ndelta > 0;
+        int32_t bm = (items + 7) >> 3;
+        int32_t hdr = 16 + (has_rle ? 16 : 0) + (has_delta ? 12 : 0)
+                    + (has_rle ? bm : 0) + csize + (has_delta ? bm : 0) + dsize;
+        /* when only one of rle/delta present the other bitmap is absent;
+         * when neither, plain dense */
+        int32_t datum_off = (hdr + 7) & ~7;
+        int32_t content = datum_off + phys * width;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        if (off + blocklen > outcap)
+        { free(pvals); free(extra); free(isdelta); free(dbytes); return -1; }
+        uint8_t *blk = out + off;
+        memset(blk, 0, blocklen);
+
+        uint32_t kind = (logical <= 16383) ? 1u : 3u;
+        uint32_t b03 = (kind << 28) | (1u << 27) | (1u << 24);
+        uint32_t b47 = 0;
+        if (kind == 1)
+        {
+            b03 |= (0x00FFFC00u & ((uint32_t) logical << 10)) |
+                   (((uint32_t) content >> 11) & 0x3FFu);
+            b47 = (((uint32_t) content & 0x7FFu) << 21);
+        }
+        else
+        {
+            b03 |= ((uint32_t) content & 0x1FFFFFu);
+            b47 = (uint32_t) logical & 0x3FFFFFFFu;
+        }
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+
+        uint8_t *c = blk + 24;
+        int16_t v16 = 2; memcpy(c, &v16, 2);
+        v16 = (int16_t) ((has_rle ? 2 : 0) | (has_delta ? 4 : 0));
+        memcpy(c + 2, &v16, 2);
+        int32_t v32 = (int32_t) logical; memcpy(c + 4, &v32, 4);
+        v32 = phys; memcpy(c + 8, &v32, 4);
+        v32 = phys * width; memcpy(c + 12, &v32, 4);
+        uint8_t *p = c + 16;
+        if (has_rle)
+        {
+            v32 = 0; memcpy(p, &v32, 4);
+            v32 = items; memcpy(p + 4, &v32, 4);
+            v32 = non; memcpy(p + 8, &v32, 4);
+            v32 = csize; memcpy(p + 12, &v32, 4);
+            p += 16;
+        }
+        if (has_delta)
+        {
+            v32 = items; memcpy(p, &v32, 4);
+            v32 = ndelta; memcpy(p + 4, &v32, 4);
+            v32 = dsize; memcpy(p + 8, &v32, 4);
+            p += 12;
+        }
+        if (has_rle)
+        {
+            uint8_t *bmp = p;
+            uint8_t *cnts = bmp + bm;
+            int32_t w = 0;
+            for (int32_t it = 0; it < items; it++)
+                if (extra[it] > 0)
+                {
+                    bmp[it >> 3] |= (uint8_t) (1u << (it & 7));
+                    w += varint_encode(cnts + w, extra[it]);
+                }
+            p = cnts + csize;
+        }
+        if (has_delta)
+        {
+            uint8_t *dbm = p;
+            for (int32_t it = 0; it < items; it++)
+                if (isdelta[it])
+                    dbm[it >> 3] |= (uint8_t) (1u << (it & 7));
+            memcpy(dbm + bm, dbytes, dsize);
+            p = dbm + bm + dsize;
+        }
+        memcpy(c + datum_off, pvals, (size_t) phys * width);
+
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+
+        off += blocklen;
+        row += logical;
+    }
+    free(pvals); free(extra); free(isdelta); free(dbytes);
+    return off;
+}
+
+/* decode one Dense(±RLE±DELTA) content area; returns rows written or -1.
+ * Walker follows DatumStreamBlockRead_AdvanceDense/…DenseDelta
+ * (datumstreamblock.h:1624-1912): per NEW item advance the compress bitmap
+ * (repeat count varint when ON) and the delta bitmap (signed-magnitude
+ * varint applied to the running value when ON; physical datum otherwise). */
 static int64_t decode_dense_content(const uint8_t *c, int width,
                                     uint8_t *dst, int64_t cap_rows)
 {
@@ -661,41 +868,87 @@ static int64_t decode_dense_content(const uint8_t *c, int width,
     memcpy(&psize, c + 12, 4);
     if (psize != phys * width || logical > cap_rows) return -1;
     if (flags & 0x1) return -1;                /* null bitmap unsupported here */
-    if (!(flags & 0x2))
+    int rle = (flags & 0x2) != 0, delta = (flags & 0x4) != 0;
+    if (!rle && !delta)
     {
         if (logical != phys) return -1;
         memcpy(dst, c + 16, (size_t) psize);
         return logical;
     }
-    if (flags & 0x4) return -1;                /* delta not yet (round 2) */
-    int32_t nullbits, bmbits, ncnt, csize;
-    memcpy(&nullbits, c + 16, 4);
-    memcpy(&bmbits, c + 20, 4);
-    memcpy(&ncnt, c + 24, 4);
-    memcpy(&csize, c + 28, 4);
-    if (nullbits != 0 || bmbits != phys) return -1;
-    const uint8_t *bmp = c + 32;
-    const uint8_t *cnts = bmp + ((bmbits + 7) >> 3);
-    int32_t hdr = 32 + ((bmbits + 7) >> 3) + csize;
-    const uint8_t *datum = c + ((hdr + 7) & ~7);
-    int64_t w = 0;
-    int32_t coff = 0, seen = 0;
-    for (int32_t p = 0; p < phys; p++)
+    const uint8_t *p = c + 16;
+    int32_t bmbits = 0, ncnt = 0, csize = 0;
+    int32_t dbmbits = 0, ndelta = 0, dsize = 0;
+    if (rle)
     {
+        int32_t nullbits;
+        memcpy(&nullbits, p, 4);
+        memcpy(&bmbits, p + 4, 4);
+        memcpy(&ncnt, p + 8, 4);
+        memcpy(&csize, p + 12, 4);
+        if (nullbits != 0) return -1;
+        p += 16;
+    }
+    if (delta)
+    {
+        memcpy(&dbmbits, p, 4);
+        memcpy(&ndelta, p + 4, 4);
+        memcpy(&dsize, p + 8, 4);
+        p += 12;
+    }
+    const uint8_t *bmp = NULL, *cnts = NULL, *dbm = NULL, *dbs = NULL;
+    if (rle)
+    {
+        bmp = p; p += (bmbits + 7) >> 3;
+        cnts = p; p += csize;
+    }
+    if (delta)
+    {
+        dbm = p; p += (dbmbits + 7) >> 3;
+        dbs = p; p += dsize;
+    }
+    int32_t hdr = (int32_t) (p - c);
+    const uint8_t *datum = c + ((hdr + 7) & ~7);
+
+    int64_t w = 0;
+    int32_t item = 0, phys_idx = 0, coff = 0, doff = 0, dseen = 0;
+    uint64_t cur = 0;
+    while (w < logical)
+    {
+        if ((rle && item >= bmbits) || (delta && item >= dbmbits)) return -1;
         int64_t reps = 1;
-        if (bmp[p >> 3] & (1u << (p & 7)))
+        if (rle && (bmp[item >> 3] & (1u << (item & 7))))
         {
             int32_t len, v = varint_decode(cnts + coff, &len);
             coff += len;
-            seen++;
             reps += v;
         }
-        if (w + reps > cap_rows) return -1;
+        if (delta && (dbm[item >> 3] & (1u << (item & 7))))
+        {
+            int32_t len, pos;
+            int64_t mag = varint3_decode(dbs + doff, &len, &pos);
+            doff += len;
+            dseen++;
+            if (width == 8)
+                cur = pos ? cur + (uint64_t) mag : cur - (uint64_t) mag;
+            else
+                cur = (uint32_t) (pos ? (uint32_t) cur + (uint32_t) mag
+                                      : (uint32_t) cur - (uint32_t) mag);
+        }
+        else
+        {
+            if (phys_idx >= phys) return -1;
+            cur = item_at(datum, width, phys_idx);
+            phys_idx++;
+        }
+        if (w + reps > logical) return -1;
         for (int64_t r = 0; r < reps; r++)
-            memcpy(dst + (w + r) * width, datum + (size_t) p * width, width);
+            memcpy(dst + (w + r) * width, &cur, width);
         w += reps;
+        item++;
     }
-    if (seen != ncnt || coff != csize || w != logical) return -1;
+    if (phys_idx != phys || (rle && (coff != csize || item != bmbits)) ||
+        (delta && (doff != dsize || dseen != ndelta || item != dbmbits)))
+        return -1;
     return w;
 }
 

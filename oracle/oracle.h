@@ -90,6 +90,13 @@ int32_t orc_aocs_rows_per_block(int width, int32_t blocksize);
 int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
                             int64_t first_rownum, int32_t blocksize,
                             uint8_t *out, int64_t outcap);
+/* RLE + DELTA_RANGE (full Dense_Enhanced feature set for NOT NULL int
+ * columns): values within ±0x1FFFFFFF of the previous item are stored as
+ * sign+magnitude varint deltas (datumstreamblock.c:2986, Reserved3 codec
+ * datumstreamblock.h:790-930); repeats still RLE-compress. */
+int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
+                                  int64_t first_rownum, int32_t blocksize,
+                                  uint8_t *out, int64_t outcap);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

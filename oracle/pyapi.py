@@ -92,6 +92,8 @@ def _load():
     lib.orc_aocs_encode_rle.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
                                         ctypes.c_int64, ctypes.c_int32,
                                         ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_encode_rle_delta.restype = ctypes.c_int64
+    lib.orc_aocs_encode_rle_delta.argtypes = lib.orc_aocs_encode_rle.argtypes
     lib.orc_aocs_encode.restype = ctypes.c_int64
     lib.orc_aocs_encode.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
                                     ctypes.c_int64, ctypes.c_int32,
@@ -247,6 +249,19 @@ def aocs_encode_rle(vals):
     buf = np.zeros(cap, np.uint8)
     got = lib.orc_aocs_encode_rle(vals.ctypes.data, width, len(vals), 1, 32768,
                                   buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
+
+
+def aocs_encode_rle_delta(vals):
+    """Full Dense_Enhanced encode: RLE + DELTA_RANGE (sign-magnitude varint
+    deltas ≤ 0x1FFFFFFF) for NOT NULL int columns."""
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_rle_delta(vals.ctypes.data, width, len(vals), 1,
+                                        32768, buf.ctypes.data, cap)
     assert got > 0
     return buf[:got].tobytes()
 
