@@ -1290,6 +1290,35 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
     }
 }
 
+/* min/max key stats over received qualifying orders (sizes the motion
+ * path's table layout exactly like the local path's counting pass) */
+__global__ void k_rows_minmax(const gx_qual_row *rows, int64_t n,
+                              unsigned long long *maxkey,
+                              unsigned long long *minkey)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long kmax = 0, kmin = ~0ULL;
+    for (; i < n; i += stride)
+    {
+        unsigned long long k = (unsigned long long) rows[i].okey;
+        if (k > kmax) kmax = k;
+        if (k < kmin) kmin = k;
+    }
+    for (int o = 32; o; o >>= 1)
+    {
+        unsigned long long v = __shfl_down(kmax, o, 64);
+        if (v > kmax) kmax = v;
+        unsigned long long w = __shfl_down(kmin, o, 64);
+        if (w < kmin) kmin = w;
+    }
+    if ((threadIdx.x & 63) == 0)
+    {
+        if (kmax) atomicMax(maxkey, kmax);
+        if (kmin != ~0ULL) atomicMin(minkey, kmin);
+    }
+}
+
 /* received qualifying orders → build the join/agg table */
 template <typename KT>
 __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
@@ -2342,15 +2371,26 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         RCCL_CHK(ctx, ncclGroupEnd());
 
         qual = (int64_t) recv2_n;
+        /* key stats over the received rows — the motion path then gets the
+         * same u32/interpolation table layout as the local path */
+        HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s));
+        hipLaunchKernelGGL(k_rows_minmax, dim3(GRID), dim3(TPB), 0, s,
+                           recv2, qual, q->dhits, q->dmin);
+        unsigned long long kmax = 0, kmin = 0;
+        HIP_CHK(ctx, hipMemcpyAsync(&kmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipMemcpyAsync(&kmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        int want_kw = (kmax < (1ULL << 32)) ? 4 : 8;
         uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
         /* motion path sizes from the exchanged counts each run */
-        if (q->tkey == nullptr || tslots > q->tmask + 1)
+        if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width)
         {
             auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
             fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
-            q->key_width = 8;
-            HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * 8));
+            q->key_width = want_kw;
+            HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * q->key_width));
             HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
@@ -2363,15 +2403,26 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             HIP_CHK(ctx, hipMalloc(&q->r_cnt, q->rescap * 8));
             q->tmask = tslots - 1;
         }
-        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, (q->tmask + 1) * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, (q->tmask + 1) * 8, s));
         q->smap.mask = q->tmask;
-        q->smap.kmin = 0;
-        q->smap.scale = -1.0;    /* hash mode (per-rank stats TBD) */
-        hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                           recv2, qual, (unsigned long long *) q->tkey,
-                           q->tdate, q->tprio, q->smap);
+        q->smap.kmin = (int64_t) kmin;
+        q->smap.scale = -1.0;
+        if (qual > 0 && kmax >= kmin)
+        {
+            double range = (double) (kmax - kmin) + 1.0;
+            if ((double) qual >= range / ((double) ctx->nsegs * 64.0))
+                q->smap.scale = (double) (q->tmask + 1) / range;
+        }
+        if (q->key_width == 4)
+            hipLaunchKernelGGL(k_build_from_rows<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+                               recv2, qual, (unsigned int *) q->tkey,
+                               q->tdate, q->tprio, q->smap);
+        else
+            hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                               recv2, qual, (unsigned long long *) q->tkey,
+                               q->tdate, q->tprio, q->smap);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0;
