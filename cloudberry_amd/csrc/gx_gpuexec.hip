@@ -24,6 +24,7 @@
 
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
+#include <zlib.h>
 
 #include <algorithm>
 #include <cstdio>
@@ -1698,6 +1699,91 @@ static int64_t host_rle_encode(const uint8_t *src_v, int width, int64_t nrows,
     return (int64_t) out.size();
 }
 
+/* Bulk-decompressed bind: blocks with compressedLength>0 (compresstype=
+ * zlib) are inflated on the HOST at bind time — the reference decompresses
+ * on the CPU during scan too (datumstream.c:1258-1290); we pay it once and
+ * keep the working set uncompressed in HBM.  Returns false when the stream
+ * has no compressed blocks (no copy made). */
+static bool host_decompress_stream(const uint8_t *s, int64_t nbytes,
+                                   std::vector<uint8_t> &out)
+{
+    /* pass 0: anything compressed at all? (avoid copying plain streams) */
+    {
+        int64_t o = 0;
+        bool found = false;
+        while (o + 24 <= nbytes)
+        {
+            uint32_t b03, b47;
+            memcpy(&b03, s + o, 4);
+            memcpy(&b47, s + o + 4, 4);
+            if (b03 == 0 && b47 == 0) break;
+            uint32_t kind = (b03 >> 28) & 7;
+            uint32_t datalen, complen = 0;
+            if (kind == 1)
+            {
+                datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+                complen = b47 & 0x1FFFFFu;
+            }
+            else if (kind == 3)
+                datalen = b03 & 0x1FFFFFu;
+            else
+                return false;
+            if (complen) { found = true; break; }
+            o += (24 + (int64_t) datalen + 7) & ~7LL;
+        }
+        if (!found) return false;
+    }
+    uint32_t tab[256];
+    host_crc32c_table(tab);
+    bool any = false;
+    int64_t off = 0;
+    while (off + 24 <= nbytes)
+    {
+        uint32_t b03, b47;
+        memcpy(&b03, s + off, 4);
+        memcpy(&b47, s + off + 4, 4);
+        if (b03 == 0 && b47 == 0) break;
+        uint32_t kind = (b03 >> 28) & 7;
+        uint32_t datalen, complen = 0;
+        if (kind == 1)
+        {
+            datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+            complen = b47 & 0x1FFFFFu;
+        }
+        else if (kind == 3)
+            datalen = b03 & 0x1FFFFFu;
+        else
+            return false;
+        int64_t blocklen = (24 + (int64_t) (complen ? complen : datalen) + 7) & ~7LL;
+        if (off + blocklen > nbytes) return false;
+        int64_t newlen = (24 + (int64_t) datalen + 7) & ~7LL;
+        size_t base = out.size();
+        out.resize(base + newlen, 0);
+        uint8_t *blk = out.data() + base;
+        memcpy(blk, s + off, 24);
+        if (complen)
+        {
+            any = true;
+            unsigned long dl = datalen;
+            if (uncompress(blk + 24, &dl, s + off + 24, complen) != Z_OK ||
+                dl != datalen)
+                return false;
+            uint32_t nb47;
+            memcpy(&nb47, blk + 4, 4);
+            nb47 &= ~0x1FFFFFu;                 /* compressedLength = 0 */
+            memcpy(blk + 4, &nb47, 4);
+            uint32_t crc = host_crc32c(tab, 0xFFFFFFFFu, blk + 16, newlen - 16);
+            memcpy(blk + 8, &crc, 4);
+            crc = host_crc32c(tab, 0xFFFFFFFFu, blk, 12);
+            memcpy(blk + 12, &crc, 4);
+        }
+        else
+            memcpy(blk + 24, s + off + 24, blocklen - 24);
+        off += blocklen;
+    }
+    return any;
+}
+
 /* walk a stream's AO envelope headers on the HOST, building the per-block
  * directory a variable-geometry (Dense/RLE) stream needs */
 static gx_status parse_block_dir(const uint8_t *s, int64_t nbytes,
@@ -1746,18 +1832,26 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
     {
         gx_col col;
         col.format = cols[c].format;
+        const void *stream_src = cols[c].host_stream;
+        int64_t stream_len = cols[c].nbytes;
+        std::vector<uint8_t> inflated;
+        if (host_decompress_stream((const uint8_t *) stream_src, stream_len, inflated))
+        {
+            stream_src = inflated.data();
+            stream_len = (int64_t) inflated.size();
+        }
         col.m.width = cols[c].width;
         col.m.rpb = gx_aocs_rows_per_block(cols[c].width, cols[c].blocksize);
         col.m.nrows = cols[c].nrows;
         col.m.full_block_len = gx_aocs_block_len(cols[c].width, col.m.rpb);
-        col.m.nbytes = cols[c].nbytes;
+        col.m.nbytes = stream_len;
         gx_colmeta_finish(&col.m);
         if (col.format == 1)
         {
             std::vector<gx_blockref> dir;
             int64_t rows = 0;
-            gx_status st = parse_block_dir((const uint8_t *) cols[c].host_stream,
-                                           cols[c].nbytes, dir, &rows);
+            gx_status st = parse_block_dir((const uint8_t *) stream_src,
+                                           stream_len, dir, &rows);
             if (st != GX_OK || rows != cols[c].nrows)
             {
                 set_err(ctx, "bad Dense/RLE stream%s", "");
@@ -1771,11 +1865,12 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
                            hipMemcpyHostToDevice, ctx->stream);
             hipStreamSynchronize(ctx->stream);
         }
-        hipError_t e = hipMalloc(&col.dstream, cols[c].nbytes);
+        hipError_t e = hipMalloc(&col.dstream, stream_len);
         if (e != hipSuccess) { set_err(ctx, "hipMalloc: %s", hipGetErrorString(e)); delete t; return GX_ERR_OOM; }
-        e = hipMemcpyAsync(col.dstream, cols[c].host_stream, cols[c].nbytes,
+        e = hipMemcpyAsync(col.dstream, stream_src, stream_len,
                            hipMemcpyHostToDevice, ctx->stream);
         if (e != hipSuccess) { set_err(ctx, "hipMemcpy: %s", hipGetErrorString(e)); delete t; return GX_ERR_HIP; }
+        hipStreamSynchronize(ctx->stream);      /* inflated buffer is stack-local */
         t->cols.push_back(col);
     }
     hipStreamSynchronize(ctx->stream);

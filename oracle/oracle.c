@@ -9,6 +9,7 @@
 #include <stdlib.h>
 #include <string.h>
 #include <math.h>
+#include <zlib.h>
 #ifdef _OPENMP
 #include <omp.h>
 #endif
@@ -428,12 +429,12 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
         if (b03 == 0 && b47 == 0) break;         /* zero padding tail */
         uint32_t kind = (b03 >> 28) & 7;
         uint32_t hasfrn = (b03 >> 27) & 1;
-        uint32_t rows, datalen;
+        uint32_t rows, datalen, complen = 0;
         if (kind == 1)                          /* SmallContent */
         {
             rows = (b03 & 0x00FFFC00u) >> 10;
             datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
-            if ((b47 & 0x1FFFFFu) != 0) return -1;   /* bulk-compressed: no */
+            complen = b47 & 0x1FFFFFu;          /* bulk-compressed if >0 */
         }
         else if (kind == 3)                     /* NonBulkDenseContent */
         {
@@ -443,7 +444,7 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
         else
             return -1;
         if (!hasfrn) return -1;
-        int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
+        int64_t blocklen = (24 + (int64_t) (complen ? complen : datalen) + 7) & ~7LL;
         if (off + blocklen > nbytes) return -1;
         if (verify_checksums)
         {
@@ -454,6 +455,16 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
             if (bc != orc_crc32c(0xFFFFFFFFu, stream + off + 16, blocklen - 16)) return -2;
         }
         const uint8_t *content = stream + off + 24;
+        uint8_t *inflated = NULL;
+        if (complen > 0)
+        {
+            inflated = malloc(datalen);
+            unsigned long dl = datalen;
+            if (uncompress(inflated, &dl, content, complen) != Z_OK ||
+                dl != datalen)
+            { free(inflated); return -1; }
+            content = inflated;
+        }
         int16_t version;
         memcpy(&version, content, 2);
         if (version == 0)                       /* Original */
@@ -477,7 +488,11 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
             row += got;
         }
         else
+        {
+            free(inflated);
             return -1;
+        }
+        free(inflated);
         off += blocklen;
     }
     return row;
@@ -848,6 +863,59 @@ int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
         row += logical;
     }
     free(pvals); free(extra); free(isdelta); free(dbytes);
+    return off;
+}
+
+
+int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
+                             int64_t first_rownum, int32_t blocksize,
+                             int level, uint8_t *out, int64_t outcap)
+{
+    int32_t rpb = orc_aocs_rows_per_block(width, blocksize);
+    const uint8_t *src = (const uint8_t *) vals;
+    uint8_t *content = malloc(blocksize + 16);
+    uint8_t *comp = malloc(compressBound(blocksize) + 16);
+    int64_t off = 0, row = 0;
+    while (row < nrows)
+    {
+        int32_t rows = (int32_t) ((nrows - row < rpb) ? (nrows - row) : rpb);
+        int32_t sz = rows * width;
+        int32_t clen = 16 + sz;                 /* Orig content */
+        int16_t v16;
+        int32_t v32;
+        v16 = 0;              memcpy(content + 0, &v16, 2);
+        v16 = 0;              memcpy(content + 2, &v16, 2);
+        v16 = (int16_t) rows; memcpy(content + 4, &v16, 2);
+        v16 = 0;              memcpy(content + 6, &v16, 2);
+        v32 = 0;              memcpy(content + 8, &v32, 4);
+        v32 = sz;             memcpy(content + 12, &v32, 4);
+        memcpy(content + 16, src + row * (int64_t) width, sz);
+
+        unsigned long dlen = compressBound(clen);
+        int zrc = compress2(comp, &dlen, content, clen, level);
+        int use_comp = (zrc == Z_OK) && ((int64_t) dlen < clen);
+        int32_t stored = use_comp ? (int32_t) dlen : clen;
+        const uint8_t *body = use_comp ? comp : content;
+        int64_t blocklen = (24 + stored + 7) & ~7LL;
+        if (off + blocklen > outcap) { free(content); free(comp); return -1; }
+        uint8_t *blk = out + off;
+        memset(blk, 0, blocklen);
+        uint32_t b03 = (1u << 28) | (1u << 27) | (1u << 24) |
+                       (0x00FFFC00u & ((uint32_t) rows << 10)) |
+                       (((uint32_t) clen >> 11) & 0x3FFu);
+        uint32_t b47 = (((uint32_t) clen & 0x7FFu) << 21) |
+                       (use_comp ? ((uint32_t) stored & 0x1FFFFFu) : 0u);
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+        memcpy(blk + 24, body, stored);
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+        off += blocklen;
+        row += rows;
+    }
+    free(content); free(comp);
     return off;
 }
 

@@ -97,6 +97,13 @@ int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
 int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
                                   int64_t first_rownum, int32_t blocksize,
                                   uint8_t *out, int64_t outcap);
+/* zlib bulk compression (compresstype=zlib): per-block deflate of the Orig
+ * content via compress2 (pg_compression.c:272-298); blocks that don't
+ * shrink are stored uncompressed with compressedLength=0 (the reference's
+ * own fallback).  orc_aocs_decode inflates transparently. */
+int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
+                             int64_t first_rownum, int32_t blocksize,
+                             int level, uint8_t *out, int64_t outcap);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

@@ -94,6 +94,11 @@ def _load():
                                         ctypes.c_void_p, ctypes.c_int64]
     lib.orc_aocs_encode_rle_delta.restype = ctypes.c_int64
     lib.orc_aocs_encode_rle_delta.argtypes = lib.orc_aocs_encode_rle.argtypes
+    lib.orc_aocs_encode_zlib.restype = ctypes.c_int64
+    lib.orc_aocs_encode_zlib.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                         ctypes.c_int64, ctypes.c_int64,
+                                         ctypes.c_int32, ctypes.c_int,
+                                         ctypes.c_void_p, ctypes.c_int64]
     lib.orc_aocs_encode.restype = ctypes.c_int64
     lib.orc_aocs_encode.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
                                     ctypes.c_int64, ctypes.c_int32,
@@ -262,6 +267,18 @@ def aocs_encode_rle_delta(vals):
     buf = np.zeros(cap, np.uint8)
     got = lib.orc_aocs_encode_rle_delta(vals.ctypes.data, width, len(vals), 1,
                                         32768, buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
+
+
+def aocs_encode_zlib(vals, level=6):
+    """zlib bulk-compressed AOCS stream (compresstype=zlib)."""
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_zlib(vals.ctypes.data, width, len(vals), 1,
+                                   32768, level, buf.ctypes.data, cap)
     assert got > 0
     return buf[:got].tobytes()
 

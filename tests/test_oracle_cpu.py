@@ -384,3 +384,35 @@ def test_rle_delta_corruption_detected():
     out = np.zeros(len(keys), np.int64)
     assert orc.lib.orc_aocs_decode(buf.ctypes.data, len(buf), 8,
                                    out.ctypes.data, len(keys), 1) == -2
+
+
+# ---------------- zlib bulk compression ----------------
+
+def test_zlib_roundtrip_and_fallback():
+    import ctypes
+    lib = orc.lib
+    lib.orc_aocs_encode_zlib.restype = ctypes.c_int64
+    lib.orc_aocs_encode_zlib.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                         ctypes.c_int64, ctypes.c_int64,
+                                         ctypes.c_int32, ctypes.c_int,
+                                         ctypes.c_void_p, ctypes.c_int64]
+    rng = np.random.default_rng(10)
+    for vals in (np.repeat(np.arange(1, 9000, dtype=np.int64), 4),
+                 rng.integers(-2**60, 2**60, 15000).astype(np.int64),
+                 (-2921 + rng.integers(0, 2526, 30000)).astype(np.int32)):
+        w = vals.itemsize
+        buf = np.zeros(len(vals) * w + 2**20, np.uint8)
+        got = lib.orc_aocs_encode_zlib(vals.ctypes.data, w, len(vals), 1,
+                                       32768, 6, buf.ctypes.data, len(buf))
+        assert got > 0
+        out = np.zeros(len(vals), vals.dtype)
+        dec = lib.orc_aocs_decode(buf.ctypes.data, got, w,
+                                  out.ctypes.data, len(vals), 1)
+        assert dec == len(vals)
+        np.testing.assert_array_equal(out, vals)
+    # repetitive data must actually shrink
+    rep = np.repeat(np.arange(1, 9000, dtype=np.int64), 4)
+    buf = np.zeros(len(rep) * 8 + 2**20, np.uint8)
+    got = lib.orc_aocs_encode_zlib(rep.ctypes.data, 8, len(rep), 1, 32768, 6,
+                                   buf.ctypes.data, len(buf))
+    assert got < len(rep) * 8 / 3
