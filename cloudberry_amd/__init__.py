@@ -100,6 +100,15 @@ def _load():
     lib.gx_test_motion1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
                                     ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
                                     ctypes.c_int64, ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_test_qual.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                                 ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
+                                 ctypes.c_void_p, ctypes.c_int64,
+                                 ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_test_q3_from_qual.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_int64, ctypes.c_void_p,
+                                         ctypes.c_int32,
+                                         ctypes.POINTER(ctypes.POINTER(_Group)),
+                                         ctypes.POINTER(ctypes.c_int64)]
     return lib
 
 
@@ -177,6 +186,39 @@ class Context:
                                             counts.ctypes.data, rows.ctypes.data,
                                             cap, ctypes.byref(total)))
         return counts, rows[:total.value]
+
+    QUAL_DTYPE = np.dtype([("okey", np.int64), ("odate", np.int32),
+                           ("oprio", np.int32)])
+
+    def test_qual(self, customer, rows, nsegs, ):
+        """Motion stage 2 on one GPU: semijoin received orders rows against
+        this segment's customer set, route qualifiers by o_orderkey."""
+        counts = np.zeros(nsegs, np.int64)
+        out = np.zeros(len(rows), self.QUAL_DTYPE)
+        total = ctypes.c_int64()
+        self._chk(self._lib.gx_test_qual(self._h, customer._t, rows.ctypes.data,
+                                         len(rows), nsegs, counts.ctypes.data,
+                                         out.ctypes.data, len(rows),
+                                         ctypes.byref(total)))
+        return counts, out[:total.value]
+
+    def test_q3_from_qual(self, qual_rows, lineitem, cutoff=CUTOFF_19950315):
+        """Motion stage 3 on one GPU: build table from qual rows, probe the
+        local lineitem, return groups sorted by key."""
+        gp = ctypes.POINTER(_Group)()
+        n = ctypes.c_int64()
+        self._chk(self._lib.gx_test_q3_from_qual(self._h, qual_rows.ctypes.data,
+                                                 len(qual_rows), lineitem._t,
+                                                 cutoff, ctypes.byref(gp),
+                                                 ctypes.byref(n)))
+        n = n.value
+        res = {"l_orderkey": np.array([gp[i].l_orderkey for i in range(n)], np.int64),
+               "o_orderdate": np.array([gp[i].o_orderdate for i in range(n)], np.int32),
+               "o_shippriority": np.array([gp[i].o_shippriority for i in range(n)], np.int32),
+               "revenue": np.array([gp[i].revenue for i in range(n)], np.float64),
+               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
+        self._lib.gx_free(gp)
+        return res
 
     def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315, numeric=False):
         q = ctypes.c_void_p()

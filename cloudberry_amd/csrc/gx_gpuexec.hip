@@ -2840,6 +2840,185 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     return GX_OK;
 }
 
+/* Motion stage-2 kernels on one GPU: build THIS segment's customer
+ * set+bloom from `customer`, probe the received Motion-1 rows, and emit
+ * qualifying rows grouped by route(o_orderkey).  With gx_test_motion1 and
+ * gx_test_q3_from_qual this covers every kernel of the nsegs>1 path
+ * without an exchange (the RCCL layer itself is thin and symmetric). */
+static_assert(sizeof(gx_qual_row_abi) == sizeof(gx_qual_row), "qual row abi");
+extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
+                                  const gx_ord_row *host_rows, int64_t n,
+                                  int nsegs, int64_t *out_counts,
+                                  gx_qual_row_abi *out_rows, int64_t cap,
+                                  int64_t *out_total)
+{
+    if (!ctx || !customer || customer->cols.size() != 2) return GX_ERR_INVALID;
+    hipStream_t s = ctx->stream;
+    const gx_col &ck = customer->cols[0], &cm = customer->cols[1];
+    devbuf cnt_b, set_b, bloom_b, rows_b, hist_b, cur_b, send_b;
+    HIP_CHK(ctx, cnt_b.alloc(16));
+    HIP_CHK(ctx, hipMemsetAsync(cnt_b.p, 0, 16, s));
+    hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
+                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       cnt_b.as<unsigned long long>(),
+                       cnt_b.as<unsigned long long>() + 1);
+    unsigned long long nb[2];
+    HIP_CHK(ctx, hipMemcpyAsync(nb, cnt_b.p, 16, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) nb[0] * 2);
+    uint64_t bwords = (uint64_t) pow2_at_least(
+        std::max<int64_t>((int64_t) nb[0] * 8 / 64, 4096));
+    HIP_CHK(ctx, set_b.alloc(cslots * 8));
+    HIP_CHK(ctx, bloom_b.alloc(bwords * 8));
+    HIP_CHK(ctx, hipMemsetAsync(set_b.p, 0, cslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(bloom_b.p, 0, bwords * 8, s));
+    hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       set_b.as<unsigned long long>(), cslots - 1,
+                       bloom_b.as<unsigned long long>(), bwords - 1);
+    HIP_CHK(ctx, rows_b.alloc(std::max<int64_t>(n, 1) * sizeof(gx_ord_row)));
+    HIP_CHK(ctx, hipMemcpyAsync(rows_b.p, host_rows, n * sizeof(gx_ord_row),
+                                hipMemcpyHostToDevice, s));
+    HIP_CHK(ctx, hist_b.alloc(nsegs * 8));
+    HIP_CHK(ctx, cur_b.alloc(nsegs * 8));
+    HIP_CHK(ctx, hipMemsetAsync(hist_b.p, 0, nsegs * 8, s));
+    hipLaunchKernelGGL(k_qual_hist<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                       rows_b.as<gx_ord_row>(), n,
+                       set_b.as<unsigned long long>(), cslots - 1,
+                       bloom_b.as<unsigned long long>(), bwords - 1, nsegs,
+                       hist_b.as<unsigned long long>());
+    std::vector<unsigned long long> h(nsegs);
+    HIP_CHK(ctx, hipMemcpyAsync(h.data(), hist_b.p, nsegs * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    std::vector<unsigned long long> off(nsegs + 1, 0);
+    for (int i = 0; i < nsegs; i++) off[i + 1] = off[i] + h[i];
+    int64_t total = (int64_t) off[nsegs];
+    if (total > cap) return GX_ERR_INVALID;
+    HIP_CHK(ctx, send_b.alloc(std::max<int64_t>(total, 1) * sizeof(gx_qual_row)));
+    HIP_CHK(ctx, hipMemcpyAsync(cur_b.p, off.data(), nsegs * 8, hipMemcpyHostToDevice, s));
+    hipLaunchKernelGGL(k_qual_emit<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                       rows_b.as<gx_ord_row>(), n,
+                       set_b.as<unsigned long long>(), cslots - 1,
+                       bloom_b.as<unsigned long long>(), bwords - 1, nsegs,
+                       cur_b.as<unsigned long long>(), send_b.as<gx_qual_row>());
+    HIP_CHK(ctx, hipMemcpyAsync(out_rows, send_b.p, total * sizeof(gx_qual_row),
+                                hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    for (int i = 0; i < nsegs; i++) out_counts[i] = (int64_t) h[i];
+    *out_total = total;
+    return GX_OK;
+}
+
+/* Motion stage-3 on one GPU: build the join/agg table from received
+ * qualifying rows (k_rows_minmax + k_build_from_rows — the nsegs>1 build)
+ * and run probe+agg+extract over `lineitem`, returning this segment's
+ * groups sorted by key.  Caller frees *out with gx_free. */
+extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
+                                          const gx_qual_row_abi *host_rows,
+                                          int64_t n, gx_table *lineitem,
+                                          int32_t cutoff, gx_q3_group **out,
+                                          int64_t *ngroups)
+{
+    if (!ctx || !lineitem || lineitem->cols.size() != 4) return GX_ERR_INVALID;
+    hipStream_t s = ctx->stream;
+    devbuf rows_b, stat_b, key_b, date_b, prio_b, rev_b, cnt_b2, cur_b, hit_b;
+    HIP_CHK(ctx, rows_b.alloc(std::max<int64_t>(n, 1) * sizeof(gx_qual_row)));
+    HIP_CHK(ctx, hipMemcpyAsync(rows_b.p, host_rows, n * sizeof(gx_qual_row),
+                                hipMemcpyHostToDevice, s));
+    HIP_CHK(ctx, stat_b.alloc(16));
+    HIP_CHK(ctx, hipMemsetAsync(stat_b.p, 0, 8, s));
+    HIP_CHK(ctx, hipMemsetAsync((uint8_t *) stat_b.p + 8, 0xFF, 8, s));
+    hipLaunchKernelGGL(k_rows_minmax, dim3(GRID), dim3(TPB), 0, s,
+                       rows_b.as<gx_qual_row>(), n,
+                       stat_b.as<unsigned long long>(),
+                       stat_b.as<unsigned long long>() + 1);
+    unsigned long long st2[2];
+    HIP_CHK(ctx, hipMemcpyAsync(st2, stat_b.p, 16, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    uint64_t tslots = (uint64_t) pow2_at_least(n * 2);
+    gx_slotmap smap;
+    smap.mask = tslots - 1;
+    smap.kmin = (int64_t) st2[1];
+    smap.scale = -1.0;
+    if (n > 0 && st2[0] >= st2[1])
+    {
+        double range = (double) (st2[0] - st2[1]) + 1.0;
+        if ((double) n >= range / (64.0 * ctx->nsegs))
+            smap.scale = (double) tslots / range;
+    }
+    HIP_CHK(ctx, key_b.alloc(tslots * 8));
+    HIP_CHK(ctx, date_b.alloc(tslots * 4));
+    HIP_CHK(ctx, prio_b.alloc(tslots * 4));
+    HIP_CHK(ctx, rev_b.alloc(tslots * 8));
+    HIP_CHK(ctx, cnt_b2.alloc(tslots * 8));
+    HIP_CHK(ctx, hipMemsetAsync(key_b.p, 0, tslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(rev_b.p, 0, tslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(cnt_b2.p, 0, tslots * 8, s));
+    hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
+                       rows_b.as<gx_qual_row>(), n,
+                       key_b.as<unsigned long long>(), date_b.as<int32_t>(),
+                       prio_b.as<int32_t>(), smap);
+    const gx_col &lk = lineitem->cols[0], &lp = lineitem->cols[1],
+                 &ld = lineitem->cols[2], &ls = lineitem->cols[3];
+    HIP_CHK(ctx, hit_b.alloc(8));
+    HIP_CHK(ctx, hipMemsetAsync(hit_b.p, 0, 8, s));
+    hipLaunchKernelGGL((k_li_probe_agg_t<1, unsigned long long>), dim3(GRID), dim3(TPB), 0, s,
+                       lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                       ls.dstream, ls.m, cutoff,
+                       key_b.as<unsigned long long>(),
+                       rev_b.as<double>(), cnt_b2.as<unsigned long long>(),
+                       smap, hit_b.as<unsigned long long>());
+    devbuf r_okey, r_odate, r_oprio, r_rev, r_cnt;
+    int64_t cap = std::max<int64_t>(n, 1);
+    HIP_CHK(ctx, r_okey.alloc(cap * 8));
+    HIP_CHK(ctx, r_odate.alloc(cap * 4));
+    HIP_CHK(ctx, r_oprio.alloc(cap * 4));
+    HIP_CHK(ctx, r_rev.alloc(cap * 8));
+    HIP_CHK(ctx, r_cnt.alloc(cap * 8));
+    HIP_CHK(ctx, cur_b.alloc(8));
+    HIP_CHK(ctx, hipMemsetAsync(cur_b.p, 0, 8, s));
+    hipLaunchKernelGGL(k_extract<unsigned long long>, dim3(32768), dim3(TPB), 0, s,
+                       key_b.as<unsigned long long>(), date_b.as<int32_t>(),
+                       prio_b.as<int32_t>(), rev_b.as<double>(),
+                       cnt_b2.as<unsigned long long>(), tslots,
+                       r_okey.as<int64_t>(), r_odate.as<int32_t>(),
+                       r_oprio.as<int32_t>(), r_rev.as<double>(),
+                       r_cnt.as<int64_t>(), cur_b.as<unsigned long long>());
+    unsigned long long ng = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&ng, cur_b.p, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    int64_t m = (int64_t) ng;
+    std::vector<int64_t> okey(m), cnt(m);
+    std::vector<int32_t> odate(m), oprio(m);
+    std::vector<double> rev(m);
+    if (m)
+    {
+        HIP_CHK(ctx, hipMemcpy(okey.data(), r_okey.p, m * 8, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(odate.data(), r_odate.p, m * 4, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(oprio.data(), r_oprio.p, m * 4, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(rev.data(), r_rev.p, m * 8, hipMemcpyDeviceToHost));
+        HIP_CHK(ctx, hipMemcpy(cnt.data(), r_cnt.p, m * 8, hipMemcpyDeviceToHost));
+    }
+    std::vector<int64_t> idx(m);
+    for (int64_t i = 0; i < m; i++) idx[i] = i;
+    std::sort(idx.begin(), idx.end(), [&](int64_t a, int64_t b) { return okey[a] < okey[b]; });
+    gx_q3_group *g = (gx_q3_group *) malloc(sizeof(gx_q3_group) * std::max<int64_t>(m, 1));
+    for (int64_t i = 0; i < m; i++)
+    {
+        g[i].l_orderkey = okey[idx[i]];
+        g[i].o_orderdate = odate[idx[i]];
+        g[i].o_shippriority = oprio[idx[i]];
+        g[i].revenue = rev[idx[i]];
+        g[i].revenue_num = 0;
+        g[i].nitems = cnt[idx[i]];
+    }
+    *out = g;
+    *ngroups = m;
+    return GX_OK;
+}
+
 /* host-side self-test of the division-free row→block addressing (callable
  * without a GPU; exercised by tests/test_abi_cpu.py) */
 extern "C" int gx_selftest_addressing(void)

@@ -169,6 +169,15 @@ typedef struct gx_ord_row { int64_t okey, ocust; int32_t odate, oprio; } gx_ord_
 gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders, int32_t cutoff,
                           int nsegs, int64_t *out_counts, gx_ord_row *out_rows,
                           int64_t cap, int64_t *out_total);
+typedef struct gx_qual_row_abi { int64_t okey; int32_t odate, oprio; } gx_qual_row_abi;
+/* Motion stage 2 (received rows → customer semijoin → route by o_orderkey) */
+gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer, const gx_ord_row *rows,
+                       int64_t n, int nsegs, int64_t *out_counts,
+                       gx_qual_row_abi *out_rows, int64_t cap, int64_t *out_total);
+/* Motion stage 3 (received qual rows → table build → probe local lineitem) */
+gx_status gx_test_q3_from_qual(gx_ctx *ctx, const gx_qual_row_abi *rows,
+                               int64_t n, gx_table *lineitem, int32_t cutoff,
+                               gx_q3_group **out, int64_t *ngroups);
 int gx_selftest_addressing(void);
 
 #ifdef __cplusplus
