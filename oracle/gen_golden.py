@@ -22,6 +22,8 @@ ref.hash_bytes_uint32.restype = ctypes.c_uint32
 ref.hash_bytes_uint32.argtypes = [ctypes.c_uint32]
 ref.hash_bytes.restype = ctypes.c_uint32
 ref.hash_bytes.argtypes = [ctypes.c_char_p, ctypes.c_int]
+ref.pg_comp_crc32c_sb8.restype = ctypes.c_uint32
+ref.pg_comp_crc32c_sb8.argtypes = [ctypes.c_uint32, ctypes.c_char_p, ctypes.c_size_t]
 
 
 def ref_hashint8(v):
@@ -41,7 +43,12 @@ def main():
     u32_cases = [0, 1, 42, 0xDEADBEEF, 0xFFFFFFFF] + [rnd.randrange(2**32) for _ in range(32)]
     bytes_cases = [b"", b"a", b"hello", b"BUILDING", bytes(range(32))]
 
+    crc_cases = [b"", b"1", b"123456789", b"BUILDING", bytes(range(256)),
+                 bytes(rnd.randrange(256) for _ in range(1000))]
     out = {
+        "crc32c_state": [{"data": c.hex(),
+                          "state": ref.pg_comp_crc32c_sb8(0xFFFFFFFF, c, len(c))}
+                         for c in crc_cases],
         "hash_bytes_uint32": [{"k": k, "h": ref.hash_bytes_uint32(k)} for k in u32_cases],
         "hashint8": [{"v": v, "h": ref_hashint8(v)} for v in i64_cases],
         "hash_bytes": [{"k": c.hex(), "h": ref.hash_bytes(c, len(c))} for c in bytes_cases],

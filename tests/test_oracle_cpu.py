@@ -416,3 +416,13 @@ def test_zlib_roundtrip_and_fallback():
     got = lib.orc_aocs_encode_zlib(rep.ctypes.data, 8, len(rep), 1, 32768, 6,
                                    buf.ctypes.data, len(buf))
     assert got < len(rep) * 8 / 3
+
+
+def test_crc32c_golden_from_reference():
+    """The oracle/GPU CRC32C state (unfinalised, the AO block convention)
+    vs the reference's own pg_crc32c_sb8.c compiled standalone."""
+    v = _vectors()
+    assert "crc32c_state" in v, "regenerate tests/golden via oracle/gen_golden.py"
+    for e in v["crc32c_state"]:
+        data = bytes.fromhex(e["data"])
+        assert orc.lib.orc_crc32c(0xFFFFFFFF, data, len(data)) == e["state"]
