@@ -57,7 +57,8 @@ class _ColDesc(ctypes.Structure):
                 ("width", ctypes.c_int32),
                 ("nrows", ctypes.c_int64),
                 ("blocksize", ctypes.c_int32),
-                ("format", ctypes.c_int32)]
+                ("format", ctypes.c_int32),
+                ("codec", ctypes.c_int32)]
 
 
 def _load():
@@ -164,6 +165,7 @@ class Context:
             descs[i].nrows = nrows
             descs[i].blocksize = 32768
             descs[i].format = spec[3] if len(spec) > 3 else 0
+            descs[i].codec = spec[4] if len(spec) > 4 else 0
         t = ctypes.c_void_p()
         self._chk(self._lib.gx_table_bind(self._h, descs, len(streams), ctypes.byref(t)))
         return Table(self, t)

@@ -26,6 +26,11 @@
 #include <rccl/rccl.h>
 #include <zlib.h>
 
+/* libzstd.so.1 ships without headers in this image — single-shot API */
+extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap, const void *src,
+                                  size_t srcSize);
+extern "C" unsigned ZSTD_isError(size_t code);
+
 #include <algorithm>
 #include <cstdio>
 #include <cstdlib>
@@ -1705,7 +1710,7 @@ static int64_t host_rle_encode(const uint8_t *src_v, int width, int64_t nrows,
  * keep the working set uncompressed in HBM.  Returns false when the stream
  * has no compressed blocks (no copy made). */
 static bool host_decompress_stream(const uint8_t *s, int64_t nbytes,
-                                   std::vector<uint8_t> &out)
+                                   int codec, std::vector<uint8_t> &out)
 {
     /* pass 0: anything compressed at all? (avoid copying plain streams) */
     {
@@ -1764,10 +1769,19 @@ static bool host_decompress_stream(const uint8_t *s, int64_t nbytes,
         if (complen)
         {
             any = true;
-            unsigned long dl = datalen;
-            if (uncompress(blk + 24, &dl, s + off + 24, complen) != Z_OK ||
-                dl != datalen)
-                return false;
+            if (codec == 2)
+            {
+                size_t r = ZSTD_decompress(blk + 24, datalen, s + off + 24, complen);
+                if (ZSTD_isError(r) || r != datalen)
+                    return false;
+            }
+            else
+            {
+                unsigned long dl = datalen;
+                if (uncompress(blk + 24, &dl, s + off + 24, complen) != Z_OK ||
+                    dl != datalen)
+                    return false;
+            }
             uint32_t nb47;
             memcpy(&nb47, blk + 4, 4);
             nb47 &= ~0x1FFFFFu;                 /* compressedLength = 0 */
@@ -1835,7 +1849,8 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
         const void *stream_src = cols[c].host_stream;
         int64_t stream_len = cols[c].nbytes;
         std::vector<uint8_t> inflated;
-        if (host_decompress_stream((const uint8_t *) stream_src, stream_len, inflated))
+        if (host_decompress_stream((const uint8_t *) stream_src, stream_len,
+                                   cols[c].codec, inflated))
         {
             stream_src = inflated.data();
             stream_len = (int64_t) inflated.size();

@@ -85,8 +85,11 @@ typedef struct gx_coldesc {
     int64_t     nrows;
     int32_t     blocksize;     /* AO blocksize the stream was written with */
     int32_t     format;        /* 0 = Orig (compresstype=none), 1 =
-                                  Dense/Dense_Enhanced incl. RLE_TYPE
-                                  (no-null, no-delta subset) */
+                                  Dense/Dense_Enhanced incl. RLE_TYPE/DELTA
+                                  (no-null subset) */
+    int32_t     codec;         /* bulk codec for compressedLength>0 blocks:
+                                  0/1 = zlib, 2 = zstd (the pg_appendonly
+                                  compresstype analog) */
 } gx_coldesc;
 
 gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncols,

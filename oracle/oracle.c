@@ -10,6 +10,14 @@
 #include <string.h>
 #include <math.h>
 #include <zlib.h>
+
+/* libzstd.so.1 is present without zstd.h — declare the single-shot API */
+extern size_t ZSTD_compress(void *dst, size_t dstCap, const void *src,
+                            size_t srcSize, int level);
+extern size_t ZSTD_decompress(void *dst, size_t dstCap, const void *src,
+                              size_t srcSize);
+extern unsigned ZSTD_isError(size_t code);
+extern size_t ZSTD_compressBound(size_t srcSize);
 #ifdef _OPENMP
 #include <omp.h>
 #endif
@@ -419,6 +427,14 @@ int64_t orc_aocs_encode(const void *vals, int width, int64_t nrows,
 int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
                         void *out_vals, int64_t cap, int verify_checksums)
 {
+    return orc_aocs_decode_c(stream, nbytes, width, out_vals, cap,
+                             verify_checksums, 1);
+}
+
+int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
+                          void *out_vals, int64_t cap, int verify_checksums,
+                          int codec)
+{
     int64_t off = 0, row = 0;
     uint8_t *dst = (uint8_t *) out_vals;
     while (off + 24 <= nbytes)
@@ -459,10 +475,19 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
         if (complen > 0)
         {
             inflated = malloc(datalen);
-            unsigned long dl = datalen;
-            if (uncompress(inflated, &dl, content, complen) != Z_OK ||
-                dl != datalen)
-            { free(inflated); return -1; }
+            if (codec == 2)
+            {
+                size_t r = ZSTD_decompress(inflated, datalen, content, complen);
+                if (ZSTD_isError(r) || r != datalen)
+                { free(inflated); return -1; }
+            }
+            else
+            {
+                unsigned long dl = datalen;
+                if (uncompress(inflated, &dl, content, complen) != Z_OK ||
+                    dl != datalen)
+                { free(inflated); return -1; }
+            }
             content = inflated;
         }
         int16_t version;
@@ -867,14 +892,14 @@ int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
 }
 
 
-int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
-                             int64_t first_rownum, int32_t blocksize,
-                             int level, uint8_t *out, int64_t outcap)
+static int64_t encode_bulk(const void *vals, int width, int64_t nrows,
+                           int64_t first_rownum, int32_t blocksize,
+                           int level, int codec, uint8_t *out, int64_t outcap)
 {
     int32_t rpb = orc_aocs_rows_per_block(width, blocksize);
     const uint8_t *src = (const uint8_t *) vals;
     uint8_t *content = malloc(blocksize + 16);
-    uint8_t *comp = malloc(compressBound(blocksize) + 16);
+    uint8_t *comp = malloc(compressBound(blocksize) + blocksize + 512);
     int64_t off = 0, row = 0;
     while (row < nrows)
     {
@@ -891,9 +916,21 @@ int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
         v32 = sz;             memcpy(content + 12, &v32, 4);
         memcpy(content + 16, src + row * (int64_t) width, sz);
 
-        unsigned long dlen = compressBound(clen);
-        int zrc = compress2(comp, &dlen, content, clen, level);
-        int use_comp = (zrc == Z_OK) && ((int64_t) dlen < clen);
+        unsigned long dlen;
+        int use_comp;
+        if (codec == 2)
+        {
+            size_t r = ZSTD_compress(comp, ZSTD_compressBound(clen),
+                                     content, clen, level);
+            use_comp = !ZSTD_isError(r) && (int64_t) r < clen;
+            dlen = (unsigned long) r;
+        }
+        else
+        {
+            dlen = compressBound(clen);
+            int zrc = compress2(comp, &dlen, content, clen, level);
+            use_comp = (zrc == Z_OK) && ((int64_t) dlen < clen);
+        }
         int32_t stored = use_comp ? (int32_t) dlen : clen;
         const uint8_t *body = use_comp ? comp : content;
         int64_t blocklen = (24 + stored + 7) & ~7LL;
@@ -917,6 +954,22 @@ int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
     }
     free(content); free(comp);
     return off;
+}
+
+int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
+                             int64_t first_rownum, int32_t blocksize,
+                             int level, uint8_t *out, int64_t outcap)
+{
+    return encode_bulk(vals, width, nrows, first_rownum, blocksize, level, 1,
+                       out, outcap);
+}
+
+int64_t orc_aocs_encode_zstd(const void *vals, int width, int64_t nrows,
+                             int64_t first_rownum, int32_t blocksize,
+                             int level, uint8_t *out, int64_t outcap)
+{
+    return encode_bulk(vals, width, nrows, first_rownum, blocksize, level, 2,
+                       out, outcap);
 }
 
 /* decode one Dense(±RLE±DELTA) content area; returns rows written or -1.

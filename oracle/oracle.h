@@ -104,6 +104,16 @@ int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
 int64_t orc_aocs_encode_zlib(const void *vals, int width, int64_t nrows,
                              int64_t first_rownum, int32_t blocksize,
                              int level, uint8_t *out, int64_t outcap);
+/* zstd bulk compression (compresstype=zstd, gpcontrib/zstd — single-shot
+ * ZSTD frames via ZSTD_compressCCtx/decompressDCtx).  The image ships
+ * libzstd.so.1 without headers; prototypes declared locally. */
+int64_t orc_aocs_encode_zstd(const void *vals, int width, int64_t nrows,
+                             int64_t first_rownum, int32_t blocksize,
+                             int level, uint8_t *out, int64_t outcap);
+/* codec for bulk-compressed blocks: 1 = zlib (default), 2 = zstd */
+int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
+                          void *out_vals, int64_t cap, int verify_checksums,
+                          int codec);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

@@ -94,6 +94,15 @@ def _load():
                                         ctypes.c_void_p, ctypes.c_int64]
     lib.orc_aocs_encode_rle_delta.restype = ctypes.c_int64
     lib.orc_aocs_encode_rle_delta.argtypes = lib.orc_aocs_encode_rle.argtypes
+    lib.orc_aocs_encode_zstd.restype = ctypes.c_int64
+    lib.orc_aocs_encode_zstd.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                         ctypes.c_int64, ctypes.c_int64,
+                                         ctypes.c_int32, ctypes.c_int,
+                                         ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_decode_c.restype = ctypes.c_int64
+    lib.orc_aocs_decode_c.argtypes = [ctypes.c_void_p, ctypes.c_int64, ctypes.c_int,
+                                      ctypes.c_void_p, ctypes.c_int64,
+                                      ctypes.c_int, ctypes.c_int]
     lib.orc_aocs_encode_zlib.restype = ctypes.c_int64
     lib.orc_aocs_encode_zlib.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                          ctypes.c_int64, ctypes.c_int64,
@@ -278,6 +287,18 @@ def aocs_encode_zlib(vals, level=6):
     cap = len(vals) * width + (1 << 20)
     buf = np.zeros(cap, np.uint8)
     got = lib.orc_aocs_encode_zlib(vals.ctypes.data, width, len(vals), 1,
+                                   32768, level, buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
+
+
+def aocs_encode_zstd(vals, level=3):
+    """zstd bulk-compressed AOCS stream (compresstype=zstd, gpcontrib/zstd)."""
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_zstd(vals.ctypes.data, width, len(vals), 1,
                                    32768, level, buf.ctypes.data, cap)
     assert got > 0
     return buf[:got].tobytes()

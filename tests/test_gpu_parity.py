@@ -414,3 +414,11 @@ def test_motion_full_pipeline_one_gpu(ctx, orc):
     np.testing.assert_array_equal(keys[order], glob["l_orderkey"])
     np.testing.assert_array_equal(cnt[order], glob["nitems"])
     np.testing.assert_allclose(rev[order], glob["revenue"], rtol=1e-6)
+
+
+def test_zstd_stream_bind(ctx, orc):
+    """zstd bulk-compressed streams bind transparently (codec=2)."""
+    vals = np.repeat(np.arange(1, 30000, dtype=np.int64), 3)
+    t = ctx.bind([(orc.aocs_encode_zstd(vals), 8, len(vals), 0, 2)])
+    np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True), vals)
+    t.free()

@@ -426,3 +426,16 @@ def test_crc32c_golden_from_reference():
     for e in v["crc32c_state"]:
         data = bytes.fromhex(e["data"])
         assert orc.lib.orc_crc32c(0xFFFFFFFF, data, len(data)) == e["state"]
+
+
+def test_zstd_roundtrip():
+    rng = np.random.default_rng(11)
+    for vals in (np.repeat(np.arange(1, 9000, dtype=np.int64), 4),
+                 rng.integers(-2**60, 2**60, 12000).astype(np.int64)):
+        s = orc.aocs_encode_zstd(vals)
+        buf = np.frombuffer(s, np.uint8)
+        out = np.zeros(len(vals), vals.dtype)
+        got = orc.lib.orc_aocs_decode_c(buf.ctypes.data, len(buf), vals.itemsize,
+                                        out.ctypes.data, len(vals), 1, 2)
+        assert got == len(vals)
+        np.testing.assert_array_equal(out, vals)
