@@ -792,6 +792,25 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -2)
+    {
+        /* non-temporal stream loads: keep L2/L3 for the table */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        for (; i < lk_m.nrows; i += stride)
+        {
+            if (!(gx_col_get_nt<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            uint64_t k = (uint64_t) gx_col_get_nt<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = smap.slot0(k);
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r == ~0ULL) continue;
+            double price = gx_col_get_nt<double>(pr_s, pr_m, i);
+            double disc = gx_col_get_nt<double>(di_s, di_m, i);
+            atomicAdd(&trev[slot], price * (1.0 - disc));
+            atomicAdd(&tcnt[slot], 1ULL);
+            local_hits++;
+        }
+    }
     else if constexpr (B == -4)
     {
         /* WAVE-batched: each BLOCK window covers blockDim.x*4 consecutive
@@ -2257,7 +2276,8 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipStreamSynchronize(s));
         qual = (int64_t) nq;
         q->key_width = (kmax < (1ULL << 32)) ? 4 : 8;
-        uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
+        int tf = env_int("GX_TABLE_FACTOR_PCT", 200);
+        uint64_t tslots = (uint64_t) pow2_at_least(qual * tf / 100 + 1);
         q->tmask = tslots - 1;
         /* order-preserving interpolation layout when the qualifying keys'
          * density over [kmin,kmax] is high enough that runs stay short;
@@ -2617,6 +2637,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 5: launch(k_li_probe_agg_t<8, unsigned int>, keys); break;
                 case 6: launch(k_li_probe_agg_t<-1, unsigned int>, keys); break;
                 case 7: launch(k_li_probe_agg_t<-4, unsigned int>, keys); break;
+                case 8: launch(k_li_probe_agg_t<-2, unsigned int>, keys); break;
             }
         }
         else
@@ -2631,6 +2652,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 5: launch(k_li_probe_agg_t<8, unsigned long long>, keys); break;
                 case 6: launch(k_li_probe_agg_t<-1, unsigned long long>, keys); break;
                 case 7: launch(k_li_probe_agg_t<-4, unsigned long long>, keys); break;
+                case 8: launch(k_li_probe_agg_t<-2, unsigned long long>, keys); break;
             }
         }
     }

@@ -173,6 +173,19 @@ GX_HD uint64_t gx_mulhi64(uint64_t a, uint64_t b)
 #endif
 }
 
+/* non-temporal variant: stream data is read once — bypassing cache keeps
+ * L2/L3 for the join table (A/B experiment; device-only) */
+template <typename T>
+__device__ __forceinline__ T gx_col_get_nt(const uint8_t *stream,
+                                           const gx_colmeta m, int64_t row)
+{
+    int64_t b = (int64_t) gx_mulhi64((uint64_t) row, m.magic);
+    int64_t r = row - b * m.rpb;
+    return __builtin_nontemporal_load(
+        (const T *) (stream + b * m.full_block_len + GX_AOCS_DATUM_OFF +
+                     r * (int64_t) sizeof(T)));
+}
+
 template <typename T>
 GX_HD T gx_col_get(const uint8_t *stream, const gx_colmeta m, int64_t row)
 {
