@@ -181,6 +181,14 @@ def main():
                 "peak": HBM_PEAK_GBPS, "unit": "GB/s",
                 "frac": round(achieved / HBM_PEAK_GBPS, 4), "traffic": None}
 
+    # config-2 evidence: the pure scan+filter kernel's GB/s on the widest
+    # filter column (untimed extra; printed into config below)
+    scan_gbps = None
+    if rank == 0 and n == 1:
+        cnt, sf_ms = li.scan_filter(3, ">", gx.CUTOFF_19950315)
+        if sf_ms > 0:
+            scan_gbps = round(li_rows_local * 4.0 / 1e9 / (sf_ms / 1000.0), 1)
+
     cpu = None
     if rank == 0 and n == 1 and not args.no_cpu_baseline:
         log(rank, "timing CPU baseline (oracle, bounded sample)...")
@@ -206,6 +214,7 @@ def main():
                        "gbps_scanned": round(gbps, 1),
                        "parallelism": f"mpp{n}",
                        "lineitem_key_format": "rle_type" if args.rle_keys else "none",
+                       "scan_filter_kernel_gbps": scan_gbps,
                        "groups": st["groups"],
                        "stage_ms": {k: round(st[k], 3) for k in
                                     ("ms_cust_build", "ms_orders_build",

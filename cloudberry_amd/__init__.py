@@ -85,6 +85,10 @@ def _load():
                                 ctypes.c_uint64, ctypes.POINTER(ctypes.c_void_p)]
     lib.gx_decode_column.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                                      ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.gx_scan_filter.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                                   ctypes.c_int, ctypes.c_int64,
+                                   ctypes.POINTER(ctypes.c_int64),
+                                   ctypes.POINTER(ctypes.c_double)]
     lib.gx_partition.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                                  ctypes.c_int32, ctypes.c_void_p]
     lib.gx_q3_prepare.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int32,
@@ -252,6 +256,16 @@ class Table:
         b = ctypes.c_double()
         self.ctx._chk(self.ctx._lib.gx_table_logical_bytes(self._t, ctypes.byref(b)))
         return b.value
+
+    def scan_filter(self, col, op, literal):
+        """SeqScan+qual count; op in {'<','>','==','!='}; returns (count, ms)."""
+        opc = {"<": 0, ">": 1, "==": 2, "!=": 3}[op]
+        n = ctypes.c_int64()
+        ms = ctypes.c_double()
+        self.ctx._chk(self.ctx._lib.gx_scan_filter(self.ctx._h, self._t, col, opc,
+                                                   literal, ctypes.byref(n),
+                                                   ctypes.byref(ms)))
+        return n.value, ms.value
 
     def decode_column(self, col, dtype, verify=True):
         n = self.nrows

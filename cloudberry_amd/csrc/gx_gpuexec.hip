@@ -576,6 +576,27 @@ __device__ __forceinline__ void gx_wave_count_add(unsigned long long *dst,
         atomicAdd(dst, v);
 }
 
+/* Standalone columnar scan + filter (BASELINE config 2; the SeqScan+qual
+ * slice without a join): count rows of one column passing <op, literal>.
+ * op: 0 '<', 1 '>', 2 '=', 3 '!='; three-valued logic degenerates to
+ * two-valued on NOT NULL columns (execScan.c:241). */
+template <typename T>
+__global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
+                              int op, T lit, unsigned long long *count)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long local = 0;
+    for (; i < m.nrows; i += stride)
+    {
+        T v = gx_col_get<T>(col_s, m, i);
+        bool pass = (op == 0) ? (v < lit) : (op == 1) ? (v > lit)
+                  : (op == 2) ? (v == lit) : (v != lit);
+        if (pass) local++;
+    }
+    gx_wave_count_add(count, local);
+}
+
 /* customer: count BUILDING rows (for set sizing) */
 __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
@@ -2146,6 +2167,47 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
     if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    return GX_OK;
+}
+
+/* scan+filter count over one column; ms_out = kernel time (HIP events) */
+extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
+                                    int op, int64_t literal,
+                                    int64_t *count_out, double *ms_out)
+{
+    if (!ctx || !t || col < 0 || col >= (int) t->cols.size() || op < 0 || op > 3)
+        return GX_ERR_INVALID;
+    const gx_col &c = t->cols[col];
+    if (c.format != 0) { set_err(ctx, "scan_filter requires fixed-format column%s", ""); return GX_ERR_INVALID; }
+    hipStream_t s = ctx->stream;
+    devbuf cnt;
+    HIP_CHK(ctx, cnt.alloc(8));
+    HIP_CHK(ctx, hipMemsetAsync(cnt.p, 0, 8, s));
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    HIP_CHK(ctx, hipEventRecord(e0, s));
+    if (c.m.width == 8)
+        hipLaunchKernelGGL(k_scan_filter<int64_t>, dim3(GRID), dim3(TPB), 0, s,
+                           c.dstream, c.m, op, (int64_t) literal,
+                           cnt.as<unsigned long long>());
+    else if (c.m.width == 4)
+        hipLaunchKernelGGL(k_scan_filter<int32_t>, dim3(GRID), dim3(TPB), 0, s,
+                           c.dstream, c.m, op, (int32_t) literal,
+                           cnt.as<unsigned long long>());
+    else
+        hipLaunchKernelGGL(k_scan_filter<int8_t>, dim3(GRID), dim3(TPB), 0, s,
+                           c.dstream, c.m, op, (int8_t) literal,
+                           cnt.as<unsigned long long>());
+    HIP_CHK(ctx, hipEventRecord(e1, s));
+    unsigned long long n = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&n, cnt.p, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    float ms = 0;
+    hipEventElapsedTime(&ms, e0, e1);
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    *count_out = (int64_t) n;
+    if (ms_out) *ms_out = (double) ms;
     return GX_OK;
 }
 

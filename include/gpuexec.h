@@ -114,6 +114,12 @@ gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int col,
                            void *host_out, int64_t cap_rows,
                            int verify_checksums);
 
+/* standalone columnar scan + filter count (the SeqScan+qual slice;
+ * BASELINE config 2): op 0 '<', 1 '>', 2 '=', 3 '!=' vs an integer/date
+ * literal; ms_out = kernel time from HIP events */
+gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col, int op,
+                         int64_t literal, int64_t *count_out, double *ms_out);
+
 /* bit-exact Motion routing of an i64 key column on device:
  * out[i] = jump_consistent_hash(cdbhash(hashint8(keys[i])), nsegs) */
 gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t n,

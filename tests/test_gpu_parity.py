@@ -422,3 +422,26 @@ def test_zstd_stream_bind(ctx, orc):
     t = ctx.bind([(orc.aocs_encode_zstd(vals), 8, len(vals), 0, 2)])
     np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True), vals)
     t.free()
+
+
+def test_scan_filter_counts(ctx, orc):
+    """Standalone SeqScan+qual (BASELINE config 2 semantics): counts on every
+    op match numpy over the oracle's columns."""
+    sf = 0.2
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    want = orc.gen_lineitem(sf)
+    cut = orc.CUTOFF_19950315
+    n, _ = li.scan_filter(3, ">", cut)            # l_shipdate > cutoff
+    assert n == int((want["l_shipdate"] > cut).sum())
+    n, _ = li.scan_filter(3, "<", cut)
+    assert n == int((want["l_shipdate"] < cut).sum())
+    n, _ = li.scan_filter(0, "==", 12345)         # l_orderkey = const
+    assert n == int((want["l_orderkey"] == 12345).sum())
+    n, _ = li.scan_filter(0, "!=", 12345)
+    assert n == int((want["l_orderkey"] != 12345).sum())
+    li.free()
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    wc = orc.gen_customer(sf)
+    n, _ = cust.scan_filter(1, "==", 0)           # c_mktsegment = BUILDING
+    assert n == int((wc["c_mktsegment"] == 0).sum())
+    cust.free()
