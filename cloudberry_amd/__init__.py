@@ -15,6 +15,7 @@ _SO = os.path.join(_ROOT, "cloudberry_amd", "libgpuexec.so")
 TPCH_CUSTOMER, TPCH_ORDERS, TPCH_LINEITEM = 0, 1, 2
 TPCH_LINEITEM_NUMERIC = 3
 TPCH_LINEITEM_RLEKEY = 4
+TPCH_LINEITEM_Q1 = 5
 CUTOFF_19950315 = -1753  # DateADT of 1995-03-15 (validated vs oracle in tests)
 
 _STATUS = {0: "GX_OK", 1: "GX_ERR_HIP", 2: "GX_ERR_RCCL", 3: "GX_ERR_INVALID",
@@ -85,6 +86,9 @@ def _load():
                                 ctypes.c_uint64, ctypes.POINTER(ctypes.c_void_p)]
     lib.gx_decode_column.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                                      ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.gx_q1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
+                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                          ctypes.POINTER(ctypes.c_double)]
     lib.gx_scan_filter.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                                    ctypes.c_int, ctypes.c_int64,
                                    ctypes.POINTER(ctypes.c_int64),
@@ -225,6 +229,20 @@ class Context:
                "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
         self._lib.gx_free(gp)
         return res
+
+    def q1(self, lineitem_q1, cutoff):
+        """TPC-H Q1 core: returns dict of 6-group arrays + kernel ms."""
+        counts = np.zeros(6, np.int64)
+        sp = np.zeros(6, np.float64)
+        sr = np.zeros(6, np.float64)
+        ms = ctypes.c_double()
+        self._chk(self._lib.gx_q1(self._h, lineitem_q1._t, cutoff,
+                                  counts.ctypes.data, sp.ctypes.data,
+                                  sr.ctypes.data, ctypes.byref(ms)))
+        return {"count": counts, "sum_price": sp, "sum_revenue": sr,
+                "avg_price": np.divide(sp, counts, out=np.zeros(6),
+                                       where=counts > 0),
+                "ms": ms.value}
 
     def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315, numeric=False):
         q = ctypes.c_void_p()

@@ -235,6 +235,78 @@ __global__ void k_emit_li_num(uint64_t seed, int64_t nord, int seg, int nsegs,
     }
 }
 
+__global__ void k_emit_li_q1(uint64_t seed, int64_t nord, int seg, int nsegs,
+                             const uint64_t *offs, int8_t *flag, int8_t *status,
+                             double *price, double *disc, int32_t *ship)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t lo = t * GEN_CHUNK, hi = min(lo + GEN_CHUNK, nord);
+    if (lo >= nord) return;
+    uint64_t w = offs[t];
+    for (int64_t o = lo + 1; o <= hi; o++)
+    {
+        if (!(nsegs == 1 || gx_route_i64(o, nsegs) == seg)) continue;
+        int32_t nl = gx_gen_nlines(seed, o);
+        for (int32_t j = 0; j < nl; j++)
+        {
+            flag[w] = (int8_t) gx_gen_returnflag(seed, o, j);
+            status[w] = (int8_t) gx_gen_linestatus(seed, o, j);
+            price[w] = gx_gen_price(seed, o, j);
+            disc[w] = gx_gen_discount(seed, o, j);
+            ship[w] = gx_gen_shipdate(seed, o, j);
+            w++;
+        }
+    }
+}
+
+/* TPC-H Q1 core (BASELINE config 4): GROUP BY l_returnflag,l_linestatus —
+ * 6 fixed groups, so each THREAD accumulates privately in registers, each
+ * wave shuffle-reduces, and one lane per wave issues 18 atomics
+ * (cdna_hip_programming.md Appendix B Reduction pattern; replaces the
+ * per-row simplehash transition of nodeAgg.c:2288,836). */
+__global__ void k_q1_agg(const uint8_t *fl_s, gx_colmeta fl_m,
+                         const uint8_t *st_s, gx_colmeta st_m,
+                         const uint8_t *pr_s, gx_colmeta pr_m,
+                         const uint8_t *di_s, gx_colmeta di_m,
+                         const uint8_t *sh_s, gx_colmeta sh_m,
+                         int32_t cutoff,
+                         unsigned long long *g_count, double *g_price,
+                         double *g_rev)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long cnt[6] = {0, 0, 0, 0, 0, 0};
+    double sp[6] = {0, 0, 0, 0, 0, 0};
+    double sr[6] = {0, 0, 0, 0, 0, 0};
+    for (; i < fl_m.nrows; i += stride)
+    {
+        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) <= cutoff)) continue;
+        int g = gx_col_get<int8_t>(fl_s, fl_m, i) * 2 +
+                gx_col_get<int8_t>(st_s, st_m, i);
+        double p = gx_col_get<double>(pr_s, pr_m, i);
+        double d = gx_col_get<double>(di_s, di_m, i);
+        cnt[g]++;
+        sp[g] += p;
+        sr[g] += p * (1.0 - d);
+    }
+#pragma unroll
+    for (int g = 0; g < 6; g++)
+    {
+        for (int o = 32; o; o >>= 1)
+        {
+            cnt[g] += __shfl_down(cnt[g], o, 64);
+            sp[g] += __shfl_down(sp[g], o, 64);
+            sr[g] += __shfl_down(sr[g], o, 64);
+        }
+        if ((threadIdx.x & 63) == 0 && cnt[g])
+        {
+            atomicAdd(&g_count[g], cnt[g]);
+            atomicAdd(&g_price[g], sp[g]);
+            atomicAdd(&g_rev[g], sr[g]);
+        }
+    }
+}
+
 /* ================= AOCS encode / decode ================= */
 
 /* Writes headers + datums for one AO block per workgroup (CRCs in a second
@@ -1989,7 +2061,8 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     int64_t nglobal = (which == GX_TPCH_CUSTOMER) ? ncust : nord;
     bool li_numeric = ((int) which == 3);   /* GX_TPCH_LINEITEM_NUMERIC */
     bool li_rlekey = ((int) which == 4);    /* GX_TPCH_LINEITEM_RLEKEY */
-    if (li_numeric || li_rlekey) which = GX_TPCH_LINEITEM;
+    bool li_q1 = ((int) which == 5);        /* GX_TPCH_LINEITEM_Q1 */
+    if (li_numeric || li_rlekey || li_q1) which = GX_TPCH_LINEITEM;
     int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
     int64_t blocks = (nthreads + TPB - 1) / TPB;
     /* the count kernels write counts[t] for EVERY launched thread */
@@ -2052,6 +2125,34 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         uint64_t *doffs; int64_t n;
         st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
         if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        if (li_q1)
+        {
+            int8_t *dfl, *dst2;
+            double *dpr1, *ddi1;
+            int32_t *dsh1;
+            HIP_CHK(ctx, hipMalloc(&dfl, n));
+            HIP_CHK(ctx, hipMalloc(&dst2, n));
+            HIP_CHK(ctx, hipMalloc(&dpr1, n * 8));
+            HIP_CHK(ctx, hipMalloc(&ddi1, n * 8));
+            HIP_CHK(ctx, hipMalloc(&dsh1, n * 4));
+            hipLaunchKernelGGL(k_emit_li_q1, dim3(blocks), dim3(TPB), 0, ctx->stream,
+                               seed, nglobal, ctx->seg, ctx->nsegs, doffs,
+                               dfl, dst2, dpr1, ddi1, dsh1);
+            gx_col q0, q1c, q2, q3c, q4;
+            st = encode_column_device(ctx, dfl, 1, n, &q0);
+            if (st == GX_OK) st = encode_column_device(ctx, dst2, 1, n, &q1c);
+            if (st == GX_OK) st = encode_column_device(ctx, dpr1, 8, n, &q2);
+            if (st == GX_OK) st = encode_column_device(ctx, ddi1, 8, n, &q3c);
+            if (st == GX_OK) st = encode_column_device(ctx, dsh1, 4, n, &q4);
+            HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+            hipFree(dfl); hipFree(dst2); hipFree(dpr1); hipFree(ddi1);
+            hipFree(dsh1); hipFree(doffs); hipFree(dcounts);
+            if (st != GX_OK) { gx_table_free(t); return st; }
+            t->cols = {q0, q1c, q2, q3c, q4};
+            t->nrows = n;
+            *out = t;
+            return GX_OK;
+        }
         int64_t *dlk; double *dpr, *ddi; int32_t *dsh;
         HIP_CHK(ctx, hipMalloc(&dlk, n * 8));
         HIP_CHK(ctx, hipMalloc(&dpr, n * 8));
@@ -2167,6 +2268,46 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
     if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    return GX_OK;
+}
+
+/* TPC-H Q1 core aggregation over a GX_TPCH_LINEITEM_Q1 table: 6 fixed
+ * groups (returnflag×linestatus); out arrays of 6: counts, sum_price,
+ * sum_revenue.  AVG = sum/count on the caller side (float8_avg = Sx/N). */
+extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
+                           int64_t *counts6, double *sum_price6,
+                           double *sum_rev6, double *ms_out)
+{
+    if (!ctx || !t || t->cols.size() != 5) return GX_ERR_INVALID;
+    hipStream_t s = ctx->stream;
+    devbuf cnt, spr, srv;
+    HIP_CHK(ctx, cnt.alloc(6 * 8));
+    HIP_CHK(ctx, spr.alloc(6 * 8));
+    HIP_CHK(ctx, srv.alloc(6 * 8));
+    HIP_CHK(ctx, hipMemsetAsync(cnt.p, 0, 48, s));
+    HIP_CHK(ctx, hipMemsetAsync(spr.p, 0, 48, s));
+    HIP_CHK(ctx, hipMemsetAsync(srv.p, 0, 48, s));
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    HIP_CHK(ctx, hipEventRecord(e0, s));
+    hipLaunchKernelGGL(k_q1_agg, dim3(GRID), dim3(TPB), 0, s,
+                       t->cols[0].dstream, t->cols[0].m,
+                       t->cols[1].dstream, t->cols[1].m,
+                       t->cols[2].dstream, t->cols[2].m,
+                       t->cols[3].dstream, t->cols[3].m,
+                       t->cols[4].dstream, t->cols[4].m,
+                       cutoff, cnt.as<unsigned long long>(),
+                       spr.as<double>(), srv.as<double>());
+    HIP_CHK(ctx, hipEventRecord(e1, s));
+    HIP_CHK(ctx, hipMemcpyAsync(counts6, cnt.p, 48, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(sum_price6, spr.p, 48, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(sum_rev6, srv.p, 48, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    float ms = 0;
+    hipEventElapsedTime(&ms, e0, e1);
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    if (ms_out) *ms_out = (double) ms;
     return GX_OK;
 }
 

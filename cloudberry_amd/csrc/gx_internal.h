@@ -92,7 +92,8 @@ GX_HD uint64_t gx_mix(uint64_t seed, uint64_t stream, uint64_t idx)
 enum {
     GX_ST_CUST_SEG = 1, GX_ST_ORD_CUST = 2, GX_ST_ORD_DATE = 3,
     GX_ST_ORD_PRIO = 4, GX_ST_LI_COUNT = 5, GX_ST_LI_SHIP = 6,
-    GX_ST_LI_PRICE = 7, GX_ST_LI_DISC = 8,
+    GX_ST_LI_PRICE = 7, GX_ST_LI_DISC = 8, GX_ST_LI_FLAG = 9,
+    GX_ST_LI_STATUS = 10,
 };
 
 /* DateADT constants (validated against oracle orc_date_adt in tests):
@@ -121,6 +122,10 @@ GX_HD int32_t gx_gen_shipdate(uint64_t seed, int64_t okey, int32_t line)
 { return GX_LI_DATE_LO + (int32_t) (gx_mix(seed, GX_ST_LI_SHIP, (uint64_t) okey * 8 + line) % (GX_LI_DATE_SPAN + 1)); }
 GX_HD double gx_gen_price(uint64_t seed, int64_t okey, int32_t line)
 { return (double) (90000 + gx_mix(seed, GX_ST_LI_PRICE, (uint64_t) okey * 8 + line) % 10410001ULL) / 100.0; }
+GX_HD uint8_t gx_gen_returnflag(uint64_t seed, int64_t okey, int32_t line)
+{ return (uint8_t) (gx_mix(seed, GX_ST_LI_FLAG, (uint64_t) okey * 8 + line) % 3); }
+GX_HD uint8_t gx_gen_linestatus(uint64_t seed, int64_t okey, int32_t line)
+{ return (uint8_t) (gx_mix(seed, GX_ST_LI_STATUS, (uint64_t) okey * 8 + line) % 2); }
 GX_HD double gx_gen_discount(uint64_t seed, int64_t okey, int32_t line)
 { return (double) (gx_mix(seed, GX_ST_LI_DISC, (uint64_t) okey * 8 + line) % 11) / 100.0; }
 

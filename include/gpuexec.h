@@ -104,7 +104,8 @@ gx_status gx_table_logical_bytes(const gx_table *t, double *out);
  * (deterministic; identical formulas to oracle/oracle.c datagen) */
 typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2,
                GX_TPCH_LINEITEM_NUMERIC = 3, /* measures as scaled int64 */
-               GX_TPCH_LINEITEM_RLEKEY = 4   /* l_orderkey RLE-compressed */ } gx_tpch_table;
+               GX_TPCH_LINEITEM_RLEKEY = 4,  /* l_orderkey RLE-compressed */
+               GX_TPCH_LINEITEM_Q1 = 5       /* Q1 cols: flag,status,price,disc,ship */ } gx_tpch_table;
 gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                       uint64_t seed, gx_table **out);
 
@@ -113,6 +114,12 @@ gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
 gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int col,
                            void *host_out, int64_t cap_rows,
                            int verify_checksums);
+
+/* TPC-H Q1 core (BASELINE config 4): GROUP BY returnflag,linestatus with
+ * COUNT/SUM over a GX_TPCH_LINEITEM_Q1 table; AVG = sum/count (float8_avg) */
+gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
+                int64_t *counts6, double *sum_price6, double *sum_rev6,
+                double *ms_out);
 
 /* standalone columnar scan + filter count (the SeqScan+qual slice;
  * BASELINE config 2): op 0 '<', 1 '>', 2 '=', 3 '!=' vs an integer/date

@@ -188,6 +188,8 @@ enum {
     ST_LI_SHIP  = 6,
     ST_LI_PRICE = 7,
     ST_LI_DISC  = 8,
+    ST_LI_FLAG  = 9,
+    ST_LI_STATUS = 10,
 };
 
 int64_t orc_ncustomer(double sf) { return (int64_t) (150000.0 * sf + 0.5); }
@@ -220,6 +222,10 @@ static inline int32_t gen_shipdate(uint64_t seed, int64_t okey, int32_t line)
 { return LI_DATE_LO() + (int32_t) (orc_mix(seed, ST_LI_SHIP, (uint64_t) okey * 8 + line) % (LI_DATE_SPAN + 1)); }
 static inline double gen_price(uint64_t seed, int64_t okey, int32_t line)
 { return (double) (90000 + orc_mix(seed, ST_LI_PRICE, (uint64_t) okey * 8 + line) % 10410001ULL) / 100.0; }
+static inline uint8_t gen_returnflag(uint64_t seed, int64_t okey, int32_t line)
+{ return (uint8_t) (orc_mix(seed, ST_LI_FLAG, (uint64_t) okey * 8 + line) % 3); }
+static inline uint8_t gen_linestatus(uint64_t seed, int64_t okey, int32_t line)
+{ return (uint8_t) (orc_mix(seed, ST_LI_STATUS, (uint64_t) okey * 8 + line) % 2); }
 static inline double gen_discount(uint64_t seed, int64_t okey, int32_t line)
 { return (double) (orc_mix(seed, ST_LI_DISC, (uint64_t) okey * 8 + line) % 11) / 100.0; }
 
@@ -1151,6 +1157,42 @@ static void tab_init(q3tab *t, int64_t want)
     t->cnt = calloc(sz, sizeof(int64_t));
     t->mask = sz - 1;
     for (uint64_t i = 0; i < sz; i++) t->key[i] = EMPTY_KEY;
+}
+
+
+/* ======================================================================
+ * TPC-H Q1 core (BASELINE config 4): GROUP BY l_returnflag, l_linestatus
+ * over WHERE l_shipdate <= cutoff — 6 fixed groups with COUNT(*),
+ * SUM(l_extendedprice), SUM(l_extendedprice*(1-l_discount)) and
+ * AVG = SUM/COUNT (PG float8_avg = Sx/N from the same running sum,
+ * float.c:3045 transition, utils/adt/float.c float8_avg final).
+ * Streaming-order accumulation → GPU parity at 1e-6 relative.
+ * ====================================================================== */
+int orc_q1(double sf, uint64_t seed, int32_t cutoff, orc_q1_group *out6)
+{
+    int64_t nord = orc_norders(sf);
+    for (int g = 0; g < 6; g++)
+    {
+        out6[g].returnflag = (int8_t) (g / 2);
+        out6[g].linestatus = (int8_t) (g % 2);
+        out6[g].count = 0;
+        out6[g].sum_price = 0.0;
+        out6[g].sum_revenue = 0.0;
+    }
+    for (int64_t o = 1; o <= nord; o++)
+    {
+        int32_t nl = gen_nlines(seed, o);
+        for (int32_t j = 0; j < nl; j++)
+        {
+            if (!(gen_shipdate(seed, o, j) <= cutoff)) continue;
+            int g = gen_returnflag(seed, o, j) * 2 + gen_linestatus(seed, o, j);
+            double p = gen_price(seed, o, j);
+            out6[g].count++;
+            out6[g].sum_price += p;
+            out6[g].sum_revenue += p * (1.0 - gen_discount(seed, o, j));
+        }
+    }
+    return 0;
 }
 
 static int q3n_group_cmp(const void *a, const void *b)

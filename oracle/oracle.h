@@ -146,6 +146,16 @@ int64_t orc_q3_numeric(const orc_customer *c, const orc_orders *o,
                        orc_q3n_group **out);
 void orc_free(void *p);
 
+/* TPC-H Q1 core (BASELINE config 4): 6 fixed groups */
+typedef struct {
+    int8_t  returnflag;          /* 0..2 */
+    int8_t  linestatus;          /* 0..1 */
+    int64_t count;
+    double  sum_price;
+    double  sum_revenue;
+} orc_q1_group;
+int orc_q1(double sf, uint64_t seed, int32_t cutoff, orc_q1_group *out6);
+
 #ifdef __cplusplus
 }
 #endif

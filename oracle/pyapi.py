@@ -131,6 +131,23 @@ lib = _load()
 CUTOFF_19950315 = lib.orc_date_adt(1995, 3, 15)
 
 
+class _Q1Group(ctypes.Structure):
+    _fields_ = [("returnflag", ctypes.c_int8), ("linestatus", ctypes.c_int8),
+                ("count", ctypes.c_int64), ("sum_price", ctypes.c_double),
+                ("sum_revenue", ctypes.c_double)]
+
+
+def q1(sf, cutoff, seed=42):
+    lib.orc_q1.restype = ctypes.c_int
+    lib.orc_q1.argtypes = [ctypes.c_double, ctypes.c_uint64, ctypes.c_int32,
+                           ctypes.POINTER(_Q1Group)]
+    out = (_Q1Group * 6)()
+    assert lib.orc_q1(sf, seed, cutoff, out) == 0
+    return {"count": np.array([g.count for g in out], np.int64),
+            "sum_price": np.array([g.sum_price for g in out]),
+            "sum_revenue": np.array([g.sum_revenue for g in out])}
+
+
 def set_threads(n):
     return lib.orc_set_threads(n)
 

@@ -445,3 +445,18 @@ def test_scan_filter_counts(ctx, orc):
     n, _ = cust.scan_filter(1, "==", 0)           # c_mktsegment = BUILDING
     assert n == int((wc["c_mktsegment"] == 0).sum())
     cust.free()
+
+
+def test_q1_core_parity(ctx, orc):
+    """BASELINE config 4: GROUP BY l_returnflag,l_linestatus SUM/AVG/COUNT —
+    counts bit-exact, f64 sums within 1e-6 (config tolerance)."""
+    sf = 0.3
+    cut = orc.lib.orc_date_adt(1998, 9, 2)   # Q1's shipdate <= date cutoff
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM_Q1, sf)
+    got = ctx.q1(li, cut)
+    want = orc.q1(sf, cut)
+    np.testing.assert_array_equal(got["count"], want["count"])
+    np.testing.assert_allclose(got["sum_price"], want["sum_price"], rtol=1e-6)
+    np.testing.assert_allclose(got["sum_revenue"], want["sum_revenue"], rtol=1e-6)
+    assert got["count"].sum() > 1000
+    li.free()

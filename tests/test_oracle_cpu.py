@@ -439,3 +439,30 @@ def test_zstd_roundtrip():
                                         out.ctypes.data, len(vals), 1, 2)
         assert got == len(vals)
         np.testing.assert_array_equal(out, vals)
+
+
+def test_q1_oracle_vs_numpy():
+    sf, cut = 0.02, orc.lib.orc_date_adt(1998, 9, 2)
+    want = orc.q1(sf, cut)
+    li = orc.gen_lineitem(sf)
+    # recompute flag/status via the shared generator streams
+    o = li["l_orderkey"]
+    # line index within order: position among equal keys
+    line = np.zeros(len(o), np.int64)
+    _, starts = np.unique(o, return_index=True)
+    for s in starts:
+        k = o[s]
+        j = 0
+        while s + j < len(o) and o[s + j] == k:
+            line[s + j] = j
+            j += 1
+    flag = np.array([orc.lib.orc_mix(42, 9, int(k) * 8 + int(j)) % 3
+                     for k, j in zip(o, line)])
+    stat = np.array([orc.lib.orc_mix(42, 10, int(k) * 8 + int(j)) % 2
+                     for k, j in zip(o, line)])
+    ok = li["l_shipdate"] <= cut
+    for g in range(6):
+        m = ok & (flag == g // 2) & (stat == g % 2)
+        assert want["count"][g] == int(m.sum())
+        np.testing.assert_allclose(want["sum_price"][g],
+                                   li["l_extendedprice"][m].sum(), rtol=1e-9)
