@@ -3259,6 +3259,34 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
     return GX_OK;
 }
 
+/* RCCL linkage self-check: init a 1-rank communicator and run one
+ * allgather — proves the collective stack works on this box without
+ * needing multiple GPUs (the multi-rank path differs only in peer count). */
+extern "C" int gx_selftest_rccl(int device)
+{
+    if (hipSetDevice(device) != hipSuccess) return 1;
+    ncclUniqueId id;
+    if (ncclGetUniqueId(&id) != ncclSuccess) return 2;
+    ncclComm_t comm;
+    if (ncclCommInitRank(&comm, 1, id, 0) != ncclSuccess) return 3;
+    devbuf a, b;
+    if (a.alloc(8) != hipSuccess || b.alloc(8) != hipSuccess) return 4;
+    unsigned long long v = 0xC0FFEE;
+    hipMemcpy(a.p, &v, 8, hipMemcpyHostToDevice);
+    hipStream_t s;
+    hipStreamCreate(&s);
+    int rc = 0;
+    if (ncclAllGather(a.p, b.p, 1, ncclUint64, comm, s) != ncclSuccess)
+        rc = 5;
+    hipStreamSynchronize(s);
+    unsigned long long w = 0;
+    hipMemcpy(&w, b.p, 8, hipMemcpyDeviceToHost);
+    if (rc == 0 && w != v) rc = 6;
+    hipStreamDestroy(s);
+    ncclCommDestroy(comm);
+    return rc;
+}
+
 /* host-side self-test of the division-free row→block addressing (callable
  * without a GPU; exercised by tests/test_abi_cpu.py) */
 extern "C" int gx_selftest_addressing(void)

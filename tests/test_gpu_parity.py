@@ -460,3 +460,13 @@ def test_q1_core_parity(ctx, orc):
     np.testing.assert_allclose(got["sum_revenue"], want["sum_revenue"], rtol=1e-6)
     assert got["count"].sum() > 1000
     li.free()
+
+
+def test_rccl_collective_selfcheck(ctx):
+    """RCCL init + one collective on this box (1-rank communicator): the
+    multi-rank exchange differs only in peer count."""
+    import ctypes
+    lib = gx.lib()
+    lib.gx_selftest_rccl.restype = ctypes.c_int
+    lib.gx_selftest_rccl.argtypes = [ctypes.c_int]
+    assert lib.gx_selftest_rccl(0) == 0
