@@ -52,6 +52,22 @@ class _Stats(ctypes.Structure):
                 ("bytes_scanned", ctypes.c_double)]
 
 
+class _Filter(ctypes.Structure):
+    _fields_ = [("col", ctypes.c_int32), ("op", ctypes.c_int32),
+                ("literal", ctypes.c_int64)]
+
+
+class _Q3Desc(ctypes.Structure):
+    _fields_ = [("dim", ctypes.c_void_p), ("dim_key_col", ctypes.c_int32),
+                ("dim_filter", _Filter),
+                ("mid", ctypes.c_void_p), ("mid_key_col", ctypes.c_int32),
+                ("mid_fk_col", ctypes.c_int32), ("mid_attr1_col", ctypes.c_int32),
+                ("mid_attr2_col", ctypes.c_int32), ("mid_filter", _Filter),
+                ("fact", ctypes.c_void_p), ("fact_key_col", ctypes.c_int32),
+                ("fact_a_col", ctypes.c_int32), ("fact_b_col", ctypes.c_int32),
+                ("fact_filter", _Filter)]
+
+
 class _ColDesc(ctypes.Structure):
     _fields_ = [("host_stream", ctypes.c_void_p),
                 ("nbytes", ctypes.c_int64),
@@ -97,6 +113,8 @@ def _load():
                                  ctypes.c_int32, ctypes.c_void_p]
     lib.gx_q3_prepare.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int32,
                                   ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_q3_prepare_desc.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Q3Desc),
+                                       ctypes.POINTER(ctypes.c_void_p)]
     lib.gx_q3_set_numeric.argtypes = [ctypes.c_void_p, ctypes.c_int]
     lib.gx_q3_run.argtypes = [ctypes.c_void_p]
     lib.gx_q3_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
@@ -243,6 +261,29 @@ class Context:
                 "avg_price": np.divide(sp, counts, out=np.zeros(6),
                                        where=counts > 0),
                 "ms": ms.value}
+
+    def q3_desc(self, desc_dict):
+        """Plan-descriptor Q3 slice: column roles + filter
+        {col, op, literal} triples chosen by the caller."""
+        ops = {"<": 0, ">": 1, "==": 2, "!=": 3, "<=": 4, ">=": 5}
+        d = _Q3Desc()
+        for role in ("dim", "mid", "fact"):
+            setattr(d, role, desc_dict[role]._t)
+        d.dim_key_col = desc_dict["dim_key_col"]
+        d.mid_key_col = desc_dict["mid_key_col"]
+        d.mid_fk_col = desc_dict["mid_fk_col"]
+        d.mid_attr1_col = desc_dict["mid_attr1_col"]
+        d.mid_attr2_col = desc_dict["mid_attr2_col"]
+        d.fact_key_col = desc_dict["fact_key_col"]
+        d.fact_a_col = desc_dict["fact_a_col"]
+        d.fact_b_col = desc_dict["fact_b_col"]
+        for role in ("dim_filter", "mid_filter", "fact_filter"):
+            col, op, lit = desc_dict[role]
+            setattr(d, role, _Filter(col, ops[op], lit))
+        q = ctypes.c_void_p()
+        self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
+                                               ctypes.byref(q)))
+        return Q3(self, q)
 
     def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315, numeric=False):
         q = ctypes.c_void_p()

@@ -672,6 +672,7 @@ __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
 /* customer: count BUILDING rows (for set sizing) */
 __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
+                             int cop, int8_t clit,
                              unsigned long long *count,
                              unsigned long long *maxkey)
 {
@@ -679,7 +680,7 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0, kmax = 0;
     for (; i < mkt_m.nrows; i += stride)
-        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) == 0)
+        if (gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit))
         {
             local++;
             unsigned long long k = (unsigned long long) gx_col_get<int64_t>(key_s, key_m, i);
@@ -700,6 +701,7 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
 template <typename KS>
 __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
+                             int cop, int8_t clit,
                              KS *set, uint64_t mask,
                              unsigned long long *bloom, uint64_t bwmask)
 {
@@ -707,7 +709,7 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < key_m.nrows; i += stride)
     {
-        if (gx_col_get<uint8_t>(mkt_s, mkt_m, i) != 0) continue;
+        if (!gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
         d_bloom_set(bloom, bwmask, k);
         uint64_t slot = gx_hmix64(k) & mask;
@@ -739,7 +741,7 @@ template <typename KS>
 __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
-                               int32_t cutoff,
+                               int oop, int32_t olit,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                unsigned long long *count,
@@ -751,7 +753,7 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
     unsigned long long local = 0, kmax = 0, kmin = ~0ULL;
     for (; i < od_m.nrows; i += stride)
     {
-        if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
+        if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
         if (!d_bloom_test(bloom, bwmask, ck)) continue;
         if (!d_set_contains(cset, cmask, ck)) continue;
@@ -787,7 +789,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
-                               int32_t cutoff,
+                               int oop, int32_t olit,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                KT *tkey,
@@ -811,7 +813,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
     for (; i < iend; i += stride)
     {
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
-        if (!(od < cutoff)) continue;
+        if (!gx_cmp(oop, od, olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
         if (!d_bloom_test(bloom, bwmask, ck)) continue;
         if (!d_set_contains(cset, cmask, ck)) continue;
@@ -847,7 +849,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
                                  const uint8_t *di_s, gx_colmeta di_m,
                                  const uint8_t *sh_s, gx_colmeta sh_m,
-                                 int32_t cutoff,
+                                 int fop, int32_t flit,
                                  const KT *tkey,
                                  double *trev, unsigned long long *tcnt,
                                  gx_slotmap smap,
@@ -878,7 +880,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         int64_t stride = gridDim.x * (int64_t) blockDim.x;
         for (; i < lk_m.nrows; i += stride)
         {
-            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t slot = smap.slot0(k);
             uint64_t r = resolve(k, slot, tkey[slot]);
@@ -892,7 +894,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         int64_t stride = gridDim.x * (int64_t) blockDim.x;
         for (; i < lk_m.nrows; i += stride)
         {
-            if (!(gx_col_get_nt<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            if (!gx_cmp(fop, gx_col_get_nt<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get_nt<int64_t>(lk_s, lk_m, i);
             uint64_t slot = smap.slot0(k);
             uint64_t r = resolve(k, slot, tkey[slot]);
@@ -934,7 +936,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
 #pragma unroll
             for (int b = 0; b < WB; b++)
             {
-                pass[b] = ship[b] > cutoff;
+                pass[b] = gx_cmp(fop, ship[b], flit);
                 slot[b] = pass[b] ? smap.slot0((uint64_t) key[b]) : 0;
             }
 #pragma unroll
@@ -953,7 +955,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         for (int64_t i = done + blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
              i < lk_m.nrows; i += gridDim.x * (int64_t) blockDim.x)
         {
-            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t s0 = smap.slot0(k);
             uint64_t res = resolve(k, s0, tkey[s0]);
@@ -970,7 +972,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         int64_t hi = min(lo + chunk, lk_m.nrows);
         for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
         {
-            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t slot = smap.slot0(k);
             uint64_t r = resolve(k, slot, tkey[slot]);
@@ -997,7 +999,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
 #pragma unroll
             for (int b = 0; b < B; b++)
             {
-                pass[b] = ship[b] > cutoff;
+                pass[b] = gx_cmp(fop, ship[b], flit);
                 slot[b] = pass[b] ? smap.slot0((uint64_t) key[b]) : 0;
             }
 #pragma unroll
@@ -1014,7 +1016,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         /* tail rows (at most B-1 per thread, only near nrows) */
         for (int64_t i = base; i < lk_m.nrows && i < base + B; i++)
         {
-            if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t slot0 = smap.slot0(k);
             uint64_t r = resolve(k, slot0, tkey[slot0]);
@@ -1041,7 +1043,7 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
                                    const uint8_t *pr_s, gx_colmeta pr_m,
                                    const uint8_t *di_s, gx_colmeta di_m,
                                    const uint8_t *sh_s, gx_colmeta sh_m,
-                                   int32_t cutoff,
+                                   int fop, int32_t flit,
                                    const KT *tkey,
                                    double *trev, unsigned long long *tcnt,
                                    gx_slotmap smap,
@@ -1217,7 +1219,7 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
             for (uint32_t r = rs; r < re; r++)
             {
                 int64_t g = first + r;
-                if (!(gx_col_get<int32_t>(sh_s, sh_m, g) > cutoff)) continue;
+                if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, g), flit)) continue;
                 double price = gx_col_get<double>(pr_s, pr_m, g);
                 double disc = gx_col_get<double>(di_s, di_m, g);
                 atomicAdd(&trev[slot], price * (1.0 - disc));
@@ -1241,7 +1243,7 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
                                    const uint8_t *pr_s, gx_colmeta pr_m,
                                    const uint8_t *di_s, gx_colmeta di_m,
                                    const uint8_t *sh_s, gx_colmeta sh_m,
-                                   int32_t cutoff,
+                                   int fop, int32_t flit,
                                    const KT *tkey,
                                    unsigned long long *tnum,
                                    unsigned long long *tcnt,
@@ -1254,7 +1256,7 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < lk_m.nrows; i += stride)
     {
-        if (!(gx_col_get<int32_t>(sh_s, sh_m, i) > cutoff)) continue;
+        if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
         uint64_t slot = smap.slot0(k);
         bool found = false;
@@ -1333,14 +1335,14 @@ __global__ void k_extract(const KT *tkey, const int32_t *tdate,
 /* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
 __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
-                              int32_t cutoff, int nsegs,
+                              int oop, int32_t olit, int nsegs,
                               unsigned long long *hist)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < od_m.nrows; i += stride)
     {
-        if (!(gx_col_get<int32_t>(od_s, od_m, i) < cutoff)) continue;
+        if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         int32_t d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
         atomicAdd(&hist[d], 1ULL);
     }
@@ -1350,7 +1352,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
                               const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *op_s, gx_colmeta op_m,
-                              int32_t cutoff, int nsegs,
+                              int oop, int32_t olit, int nsegs,
                               unsigned long long *cursors, /* pre-set to region starts */
                               gx_ord_row *out)
 {
@@ -1359,7 +1361,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
     for (; i < od_m.nrows; i += stride)
     {
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
-        if (!(od < cutoff)) continue;
+        if (!gx_cmp(oop, od, olit)) continue;
         int64_t oc = gx_col_get<int64_t>(oc_s, oc_m, i);
         int32_t d = gx_route_i64(oc, nsegs);
         unsigned long long w = atomicAdd(&cursors[d], 1ULL);
@@ -1572,7 +1574,7 @@ struct gx_table {
 struct gx_q3 {
     gx_ctx *ctx = nullptr;
     gx_table *cust = nullptr, *ord = nullptr, *li = nullptr;
-    int32_t cutoff = 0;
+    gx_q3_desc desc{};          /* column roles + filters (plan descriptor) */
     /* run state (device) — allocated on first run, reused across steps
      * (a re-run rebuilds every table; only the ALLOCATIONS persist) */
     bool sized = false;
@@ -2370,18 +2372,40 @@ extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t
 
 /* ================= Q3 ================= */
 
-extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
-                                   gx_table *lineitem, int32_t cutoff, gx_q3 **out)
+extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
+                                        gx_q3 **out)
 {
-    if (!ctx || !customer || !orders || !lineitem) return GX_ERR_INVALID;
-    if (customer->cols.size() != 2 || orders->cols.size() != 4 ||
-        lineitem->cols.size() != 4) return GX_ERR_INVALID;
-    /* fused-RLE scan is supported for lineitem's l_orderkey column only */
+    if (!ctx || !desc || !desc->dim || !desc->mid || !desc->fact)
+        return GX_ERR_INVALID;
+    gx_table *customer = desc->dim, *orders = desc->mid, *lineitem = desc->fact;
+    auto colw = [](gx_table *t, int c) {
+        return (c >= 0 && c < (int) t->cols.size()) ? t->cols[c].m.width : -1;
+    };
+    if (colw(customer, desc->dim_key_col) != 8 ||
+        colw(customer, desc->dim_filter.col) != 1 ||
+        colw(orders, desc->mid_key_col) != 8 ||
+        colw(orders, desc->mid_fk_col) != 8 ||
+        colw(orders, desc->mid_attr1_col) != 4 ||
+        colw(orders, desc->mid_attr2_col) != 4 ||
+        colw(orders, desc->mid_filter.col) != 4 ||
+        colw(lineitem, desc->fact_key_col) != 8 ||
+        colw(lineitem, desc->fact_a_col) != 8 ||
+        colw(lineitem, desc->fact_b_col) != 8 ||
+        colw(lineitem, desc->fact_filter.col) != 4)
+    {
+        set_err(ctx, "gx_q3_desc: column role/width mismatch%s", "");
+        return GX_ERR_INVALID;
+    }
+    if (desc->mid_filter.col != desc->mid_attr1_col &&
+        desc->mid_filter.col == desc->mid_key_col)
+        return GX_ERR_INVALID;
+    /* fused-RLE scan is supported for the fact key column only */
     for (auto *t : {customer, orders, lineitem})
         for (size_t ci = 0; ci < t->cols.size(); ci++)
-            if (t->cols[ci].format != 0 && !(t == lineitem && ci == 0))
+            if (t->cols[ci].format != 0 &&
+                !(t == lineitem && (int) ci == desc->fact_key_col))
             {
-                set_err(ctx, "Q3 pipeline supports RLE only on l_orderkey; "
+                set_err(ctx, "Q3 pipeline supports RLE only on the fact key; "
                              "decode other RLE columns first%s", "");
                 return GX_ERR_INVALID;
             }
@@ -2390,9 +2414,34 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
     q->cust = customer;
     q->ord = orders;
     q->li = lineitem;
-    q->cutoff = cutoff;
+    q->desc = *desc;
     *out = q;
     return GX_OK;
+}
+
+/* classic entry: the standard Q3 column layout and filters */
+extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
+                                   gx_table *lineitem, int32_t cutoff, gx_q3 **out)
+{
+    if (!customer || !orders || !lineitem) return GX_ERR_INVALID;
+    if (customer->cols.size() != 2 || orders->cols.size() != 4 ||
+        lineitem->cols.size() != 4) return GX_ERR_INVALID;
+    gx_q3_desc d{};
+    d.dim = customer;
+    d.dim_key_col = 0;
+    d.dim_filter = {1, 2, 0};              /* c_mktsegment = BUILDING */
+    d.mid = orders;
+    d.mid_key_col = 0;
+    d.mid_fk_col = 1;
+    d.mid_attr1_col = 2;
+    d.mid_attr2_col = 3;
+    d.mid_filter = {2, 0, cutoff};         /* o_orderdate < cutoff */
+    d.fact = lineitem;
+    d.fact_key_col = 0;
+    d.fact_a_col = 1;
+    d.fact_b_col = 2;
+    d.fact_filter = {3, 1, cutoff};        /* l_shipdate > cutoff */
+    return gx_q3_prepare_desc(ctx, &d, out);
 }
 
 static void q3_free_runstate(gx_q3 *q)
@@ -2410,8 +2459,9 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
 {
     gx_ctx *ctx = q->ctx;
     hipStream_t s = ctx->stream;
-    const gx_col &cm = q->cust->cols[1];
-    const gx_col &oc = q->ord->cols[1], &od = q->ord->cols[2];
+    const gx_q3_desc &D = q->desc;
+    const gx_col &cm = q->cust->cols[D.dim_filter.col];
+    const gx_col &oc = q->ord->cols[D.mid_fk_col], &od = q->ord->cols[D.mid_filter.col];
 
     HIP_CHK(ctx, hipMalloc(&q->dcount, 8));
     HIP_CHK(ctx, hipMalloc(&q->dhits, 8));
@@ -2420,8 +2470,10 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
     HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for max custkey */
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                       q->cust->cols[0].dstream, q->cust->cols[0].m,
-                       cm.dstream, cm.m, q->dcount, q->dhits);
+                       q->cust->cols[D.dim_key_col].dstream,
+                       q->cust->cols[D.dim_key_col].m,
+                       cm.dstream, cm.m, D.dim_filter.op,
+                       (int8_t) D.dim_filter.literal, q->dcount, q->dhits);
     unsigned long long n_building = 0, cmax = 0;
     HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
@@ -2438,13 +2490,19 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     q->bwmask = bwords - 1;
     if (q->cset_width == 4)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[0].dstream, q->cust->cols[0].m,
-                           cm.dstream, cm.m, (unsigned int *) q->cset, q->cmask,
+                           q->cust->cols[D.dim_key_col].dstream,
+                           q->cust->cols[D.dim_key_col].m,
+                           cm.dstream, cm.m, D.dim_filter.op,
+                           (int8_t) D.dim_filter.literal,
+                           (unsigned int *) q->cset, q->cmask,
                            q->bloom, q->bwmask);
     else
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[0].dstream, q->cust->cols[0].m,
-                           cm.dstream, cm.m, (unsigned long long *) q->cset, q->cmask,
+                           q->cust->cols[D.dim_key_col].dstream,
+                           q->cust->cols[D.dim_key_col].m,
+                           cm.dstream, cm.m, D.dim_filter.op,
+                           (int8_t) D.dim_filter.literal,
+                           (unsigned long long *) q->cset, q->cmask,
                            q->bloom, q->bwmask);
 
     /* local qualifying-order count bounds the table for BOTH paths: at
@@ -2460,15 +2518,19 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s)); /* minkey = ~0 */
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_orders_count<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               q->ord->cols[0].dstream, q->ord->cols[0].m,
-                               od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
+                               q->ord->cols[D.mid_key_col].dstream,
+                               q->ord->cols[D.mid_key_col].m,
+                               od.dstream, od.m, oc.dstream, oc.m,
+                               D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned int *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
                                q->dhits, q->dmin);
         else
             hipLaunchKernelGGL(k_orders_count<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               q->ord->cols[0].dstream, q->ord->cols[0].m,
-                               od.dstream, od.m, oc.dstream, oc.m, q->cutoff,
+                               q->ord->cols[D.mid_key_col].dstream,
+                               q->ord->cols[D.mid_key_col].m,
+                               od.dstream, od.m, oc.dstream, oc.m,
+                               D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned long long *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
                                q->dhits, q->dmin);
@@ -2533,11 +2595,19 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     hipEvent_t ev[8];
     for (auto &e : ev) HIP_CHK(ctx, hipEventCreate(&e));
 
-    const gx_col &ck = q->cust->cols[0], &cm = q->cust->cols[1];
-    const gx_col &ok = q->ord->cols[0], &oc = q->ord->cols[1],
-                 &od = q->ord->cols[2], &op = q->ord->cols[3];
-    const gx_col &lk = q->li->cols[0], &lp = q->li->cols[1],
-                 &ld = q->li->cols[2], &ls = q->li->cols[3];
+    const gx_q3_desc &D = q->desc;
+    const gx_col &ck = q->cust->cols[D.dim_key_col],
+                 &cm = q->cust->cols[D.dim_filter.col];
+    const gx_col &ok = q->ord->cols[D.mid_key_col],
+                 &oc = q->ord->cols[D.mid_fk_col],
+                 &od = q->ord->cols[D.mid_filter.col],
+                 &op = q->ord->cols[D.mid_attr2_col];
+    const gx_col &oa1 = q->ord->cols[D.mid_attr1_col];
+    (void) oa1;  /* attr1 == the date filter column in the standard plan */
+    const gx_col &lk = q->li->cols[D.fact_key_col],
+                 &lp = q->li->cols[D.fact_a_col],
+                 &ld = q->li->cols[D.fact_b_col],
+                 &ls = q->li->cols[D.fact_filter.col];
     unsigned long long *dcount = q->dcount;
 
     /* ---- stage 1: customer BUILDING set (rebuilt every run) ---- */
@@ -2547,10 +2617,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     if (q->cset_width == 4)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                            ck.dstream, ck.m, cm.dstream, cm.m,
+                           D.dim_filter.op, (int8_t) D.dim_filter.literal,
                            (unsigned int *) q->cset, q->cmask, q->bloom, q->bwmask);
     else
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                            ck.dstream, ck.m, cm.dstream, cm.m,
+                           D.dim_filter.op, (int8_t) D.dim_filter.literal,
                            (unsigned long long *) q->cset, q->cmask, q->bloom, q->bwmask);
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
 
@@ -2571,7 +2643,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                                    std::decay_t<decltype(*cs)>, true>),
                                    dim3(ogrid), dim3(TPB), 0, s,
                                    ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                                   op.dstream, op.m, q->cutoff, cs, q->cmask,
+                                   op.dstream, op.m, D.mid_filter.op,
+                                   (int32_t) D.mid_filter.literal, cs, q->cmask,
                                    q->bloom, q->bwmask,
                                    tk, q->tdate, q->tprio, q->smap);
             else
@@ -2579,7 +2652,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                                    std::decay_t<decltype(*cs)>, false>),
                                    dim3(ogrid), dim3(TPB), 0, s,
                                    ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                                   op.dstream, op.m, q->cutoff, cs, q->cmask,
+                                   op.dstream, op.m, D.mid_filter.op,
+                                   (int32_t) D.mid_filter.literal, cs, q->cmask,
                                    q->bloom, q->bwmask,
                                    tk, q->tdate, q->tprio, q->smap);
         };
@@ -2608,7 +2682,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         unsigned long long *dhist = dhist_b.as<unsigned long long>();
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
-                           od.dstream, od.m, oc.dstream, oc.m, q->cutoff, n, dhist);
+                           od.dstream, od.m, oc.dstream, oc.m,
+                           D.mid_filter.op, (int32_t) D.mid_filter.literal, n, dhist);
         std::vector<unsigned long long> h1(n);
         HIP_CHK(ctx, hipMemcpyAsync(h1.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -2623,7 +2698,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                           op.dstream, op.m, q->cutoff, n, dcur, send1);
+                           op.dstream, op.m, D.mid_filter.op,
+                           (int32_t) D.mid_filter.literal, n, dcur, send1);
 
         /* exchange counts (all-gather of per-dest counts) */
         HIP_CHK(ctx, cnts_mine_b.alloc(n * 8));
@@ -2778,13 +2854,15 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->key_width == 4)
             hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned int>, dim3(rgrid), dim3(TPB), 0, s,
                                lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
-                               ld.dstream, ld.m, ls.dstream, ls.m, q->cutoff,
+                               ld.dstream, ld.m, ls.dstream, ls.m,
+                               D.fact_filter.op, (int32_t) D.fact_filter.literal,
                                (const unsigned int *) q->tkey,
                                q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
         else
             hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned long long>, dim3(rgrid), dim3(TPB), 0, s,
                                lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
-                               ld.dstream, ld.m, ls.dstream, ls.m, q->cutoff,
+                               ld.dstream, ld.m, ls.dstream, ls.m,
+                               D.fact_filter.op, (int32_t) D.fact_filter.literal,
                                (const unsigned long long *) q->tkey,
                                q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
         int herr = 0;
@@ -2798,14 +2876,16 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->key_width == 4)
             hipLaunchKernelGGL(k_li_probe_agg_num<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, q->cutoff,
+                               ls.dstream, ls.m, D.fact_filter.op,
+                               (int32_t) D.fact_filter.literal,
                                (const unsigned int *) q->tkey,
                                (unsigned long long *) q->trev, q->tcnt, q->smap,
                                dhits, (int *) q->dmin);
         else
             hipLaunchKernelGGL(k_li_probe_agg_num<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, q->cutoff,
+                               ls.dstream, ls.m, D.fact_filter.op,
+                               (int32_t) D.fact_filter.literal,
                                (const unsigned long long *) q->tkey,
                                (unsigned long long *) q->trev, q->tcnt, q->smap,
                                dhits, (int *) q->dmin);
@@ -2825,7 +2905,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         auto launch = [&](auto kern, auto *keys) {
             hipLaunchKernelGGL(kern, dim3(pgrid), dim3(ptpb), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, q->cutoff, keys,
+                               ls.dstream, ls.m, D.fact_filter.op,
+                               (int32_t) D.fact_filter.literal, keys,
                                q->trev, q->tcnt, q->smap, dhits);
         };
         if (q->key_width == 4)
@@ -3057,7 +3138,7 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     unsigned long long *dcur = dcur_b.as<unsigned long long>();
     HIP_CHK(ctx, hipMemsetAsync(dhist, 0, nsegs * 8, s));
     hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
-                       od.dstream, od.m, oc.dstream, oc.m, cutoff, nsegs, dhist);
+                       od.dstream, od.m, oc.dstream, oc.m, 0, cutoff, nsegs, dhist);
     std::vector<unsigned long long> h(nsegs);
     HIP_CHK(ctx, hipMemcpyAsync(h.data(), dhist, nsegs * 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3070,7 +3151,7 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     HIP_CHK(ctx, hipMemcpyAsync(dcur, off.data(), nsegs * 8, hipMemcpyHostToDevice, s));
     hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                        ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                       op.dstream, op.m, cutoff, nsegs, dcur, dsend);
+                       op.dstream, op.m, 0, cutoff, nsegs, dcur, dsend);
     HIP_CHK(ctx, hipMemcpyAsync(out_rows, dsend, total * sizeof(gx_ord_row),
                                 hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3099,7 +3180,7 @@ extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
     HIP_CHK(ctx, cnt_b.alloc(16));
     HIP_CHK(ctx, hipMemsetAsync(cnt_b.p, 0, 16, s));
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       ck.dstream, ck.m, cm.dstream, cm.m, 2, (int8_t) 0,
                        cnt_b.as<unsigned long long>(),
                        cnt_b.as<unsigned long long>() + 1);
     unsigned long long nb[2];
@@ -3113,7 +3194,7 @@ extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
     HIP_CHK(ctx, hipMemsetAsync(set_b.p, 0, cslots * 8, s));
     HIP_CHK(ctx, hipMemsetAsync(bloom_b.p, 0, bwords * 8, s));
     hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       ck.dstream, ck.m, cm.dstream, cm.m, 2, (int8_t) 0,
                        set_b.as<unsigned long long>(), cslots - 1,
                        bloom_b.as<unsigned long long>(), bwords - 1);
     HIP_CHK(ctx, rows_b.alloc(std::max<int64_t>(n, 1) * sizeof(gx_ord_row)));
@@ -3205,7 +3286,7 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
     HIP_CHK(ctx, hipMemsetAsync(hit_b.p, 0, 8, s));
     hipLaunchKernelGGL((k_li_probe_agg_t<1, unsigned long long>), dim3(GRID), dim3(TPB), 0, s,
                        lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                       ls.dstream, ls.m, cutoff,
+                       ls.dstream, ls.m, 1, cutoff,
                        key_b.as<unsigned long long>(),
                        rev_b.as<double>(), cnt_b2.as<unsigned long long>(),
                        smap, hit_b.as<unsigned long long>());

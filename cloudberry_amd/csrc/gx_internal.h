@@ -200,6 +200,21 @@ GX_HD T gx_col_get(const uint8_t *stream, const gx_colmeta m, int64_t row)
                          r * (int64_t) sizeof(T));
 }
 
+/* filter compare (gx_filter.op): 0 '<',1 '>',2 '=',3 '!=',4 '<=',5 '>=' */
+template <typename T>
+GX_HD bool gx_cmp(int op, T v, T lit)
+{
+    switch (op)
+    {
+        case 0: return v < lit;
+        case 1: return v > lit;
+        case 2: return v == lit;
+        case 3: return v != lit;
+        case 4: return v <= lit;
+        default: return v >= lit;
+    }
+}
+
 /* per-block directory entry for variable-geometry (RLE/Dense) streams */
 struct gx_blockref {
     int64_t offset;              /* byte offset of the AO block */

@@ -163,6 +163,37 @@ typedef struct gx_q3_stats {
 gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *orders,
                         gx_table *lineitem, int32_t cutoff_dateadt,
                         gx_q3 **out);
+
+/* ---- plan-descriptor form (what PlanCustomPath lowering
+ * fills; tagged structs, no expression trees yet).  Filter ops:
+ * 0 '<', 1 '>', 2 '=', 3 '!=', 4 '<=', 5 '>='. ---- */
+typedef struct gx_filter {
+    int32_t col;               /* column index in the role's table */
+    int32_t op;
+    int64_t literal;           /* integer/date literal (DateADT for dates) */
+} gx_filter;
+
+typedef struct gx_q3_desc {
+    /* build side of the semijoin (reference: customer) */
+    gx_table *dim;
+    int32_t dim_key_col;       /* i64 join key */
+    gx_filter dim_filter;      /* e.g. mktsegment = literal */
+    /* middle table redistributed/joined on both keys (reference: orders) */
+    gx_table *mid;
+    int32_t mid_key_col;       /* i64 key joined to fact (o_orderkey) */
+    int32_t mid_fk_col;        /* i64 key joined to dim  (o_custkey) */
+    int32_t mid_attr1_col;     /* i32 carried into the group (o_orderdate) */
+    int32_t mid_attr2_col;     /* i32 carried into the group (o_shippriority) */
+    gx_filter mid_filter;      /* e.g. o_orderdate < literal */
+    /* fact side (reference: lineitem); agg = SUM(a*(1-b)), COUNT(*) */
+    gx_table *fact;
+    int32_t fact_key_col;      /* i64 probe key (l_orderkey) */
+    int32_t fact_a_col;        /* f64 (l_extendedprice) */
+    int32_t fact_b_col;        /* f64 (l_discount) */
+    gx_filter fact_filter;     /* e.g. l_shipdate > literal */
+} gx_q3_desc;
+
+gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc, gx_q3 **out);
 /* numeric(15,2) mode: lineitem measures are scaled int64
  * (price cents, discount hundredths — GX_TPCH_LINEITEM_NUMERIC tables);
  * aggregation is integer → results BIT-EXACT vs the oracle. */

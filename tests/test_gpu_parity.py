@@ -470,3 +470,51 @@ def test_rccl_collective_selfcheck(ctx):
     lib.gx_selftest_rccl.restype = ctypes.c_int
     lib.gx_selftest_rccl.argtypes = [ctypes.c_int]
     assert lib.gx_selftest_rccl(0) == 0
+
+
+def test_q3_descriptor_api(ctx, orc):
+    """The plan-descriptor entry (gx_q3_prepare_desc) with the standard Q3
+    roles equals the classic entry; with FLIPPED filter ops it matches a
+    numpy brute force of the altered plan — the descriptor really drives
+    the kernels."""
+    sf = 0.05
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut)}
+    classic = ctx.q3(cust, ordr, li).run().result()
+    via_desc = ctx.q3_desc(base).run().result()
+    np.testing.assert_array_equal(via_desc["l_orderkey"], classic["l_orderkey"])
+    np.testing.assert_array_equal(via_desc["nitems"], classic["nitems"])
+    np.testing.assert_allclose(via_desc["revenue"], classic["revenue"], rtol=0)
+
+    # altered plan: segment != 0, orderdate >= cut, shipdate <= cut
+    alt = dict(base)
+    alt["dim_filter"] = (1, "!=", 0)
+    alt["mid_filter"] = (2, ">=", cut)
+    alt["fact_filter"] = (3, "<=", cut)
+    got = ctx.q3_desc(alt).run().result()
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    w = orc.gen_lineitem(sf)
+    segok = c["c_custkey"][c["c_mktsegment"] != 0]
+    om = (o["o_orderdate"] >= cut) & np.isin(o["o_custkey"], segok)
+    okeys = set(o["o_orderkey"][om].tolist())
+    lm = (w["l_shipdate"] <= cut) & np.isin(w["l_orderkey"],
+                                            o["o_orderkey"][om])
+    keys, counts = np.unique(w["l_orderkey"][lm], return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    rev = {k: 0.0 for k in keys.tolist()}
+    for k, p, dsc in zip(w["l_orderkey"][lm].tolist(),
+                         w["l_extendedprice"][lm], w["l_discount"][lm]):
+        rev[k] += p * (1.0 - dsc)
+    np.testing.assert_allclose(got["revenue"],
+                               np.array([rev[k] for k in keys.tolist()]),
+                               rtol=1e-6)
+    li.free(); ordr.free(); cust.free()
