@@ -491,7 +491,8 @@ def test_q3_descriptor_api(ctx, orc):
     via_desc = ctx.q3_desc(base).run().result()
     np.testing.assert_array_equal(via_desc["l_orderkey"], classic["l_orderkey"])
     np.testing.assert_array_equal(via_desc["nitems"], classic["nitems"])
-    np.testing.assert_allclose(via_desc["revenue"], classic["revenue"], rtol=0)
+    # f64 atomic ordering varies run-to-run (documented tolerance)
+    np.testing.assert_allclose(via_desc["revenue"], classic["revenue"], rtol=1e-12)
 
     # altered plan: segment != 0, orderdate >= cut, shipdate <= cut
     alt = dict(base)
