@@ -371,9 +371,8 @@ def test_motion_full_pipeline_one_gpu(ctx, orc):
     nsegs, sf = 3, 0.05
     glob = orc.q3(orc.gen_customer(sf), orc.gen_orders(sf), orc.gen_lineitem(sf))
 
-    per_rank_ctx = [gx.Context(device=0, seg=r, nsegs=1) for r in range(nsegs)]
-    # NOTE: tables are sharded by generating with seg/nsegs through a throw-
-    # away context; the kernels themselves are seg-agnostic.
+    # tables are sharded by generating with seg/nsegs through a throw-away
+    # context; the kernels themselves are seg-agnostic
     m1_out = {r: [] for r in range(nsegs)}      # rows received by rank r
     for r in range(nsegs):
         # rank r's orders shard lives where o_orderkey routes; partition its
@@ -404,8 +403,6 @@ def test_motion_full_pipeline_one_gpu(ctx, orc):
         qual = np.concatenate(qual_by_dest[r])
         results.append(cshard.test_q3_from_qual(qual, li))
         li.free(); cshard.close()
-    for c in per_rank_ctx:
-        c.close()
     keys = np.concatenate([x["l_orderkey"] for x in results])
     rev = np.concatenate([x["revenue"] for x in results])
     cnt = np.concatenate([x["nitems"] for x in results])

@@ -3,7 +3,6 @@ reference's own hashfn.c compiled standalone — oracle/gen_golden.py) plus an
 independent numpy brute-force of Q3 semantics, plus AOCS codec round-trips."""
 import json
 import os
-import subprocess
 
 import numpy as np
 import pytest
