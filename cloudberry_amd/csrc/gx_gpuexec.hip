@@ -1591,6 +1591,12 @@ struct gx_q3 {
     unsigned long long *tcnt = nullptr;
     uint64_t tmask = 0;
     unsigned long long *dcount = nullptr, *dhits = nullptr, *dmin = nullptr;
+    /* motion-path exchange state (nsegs>1), cached across steps */
+    unsigned long long *m_hist = nullptr, *m_cur = nullptr;
+    unsigned long long *m_cnts_mine = nullptr, *m_cnts_all = nullptr;
+    gx_ord_row *m_send1 = nullptr, *m_recv1 = nullptr;
+    gx_qual_row *m_send2 = nullptr, *m_recv2 = nullptr;
+    uint64_t m_send1_cap = 0, m_recv1_cap = 0, m_send2_cap = 0, m_recv2_cap = 0;
     /* result (device SoA) */
     int64_t *r_okey = nullptr;
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
@@ -2450,6 +2456,9 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
+    fr(q->m_hist); fr(q->m_cur); fr(q->m_cnts_mine); fr(q->m_cnts_all);
+    fr(q->m_send1); fr(q->m_recv1); fr(q->m_send2); fr(q->m_recv2);
+    q->m_send1_cap = q->m_recv1_cap = q->m_send2_cap = q->m_recv2_cap = 0;
     q->sized = false;
 }
 
@@ -2675,11 +2684,28 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         hipEventCreate(&mev0); hipEventCreate(&mev1);
         HIP_CHK(ctx, hipEventRecord(mev0, s));
 
-        /* Motion 1: filtered orders by route(o_custkey) */
-        devbuf dhist_b, dcur_b, send1_b, recv1_b, send2_b, recv2_b,
-               cnts_mine_b, cnts_all_b;
-        HIP_CHK(ctx, dhist_b.alloc(n * 8));
-        unsigned long long *dhist = dhist_b.as<unsigned long long>();
+        /* Motion 1: filtered orders by route(o_custkey).  Exchange buffers
+         * are cached in gx_q3 (grow-only) — per-step hipMalloc would
+         * dominate at high rank counts. */
+        auto grow = [&](auto *&p, uint64_t &cap, uint64_t want) -> gx_status {
+            if (want <= cap) return GX_OK;
+            if (p) hipFree(p);
+            p = nullptr;
+            cap = 0;
+            hipError_t e = hipMalloc(&p, std::max<uint64_t>(want, 1) *
+                                          sizeof(**(&p)));
+            if (e != hipSuccess) return GX_ERR_OOM;
+            cap = want;
+            return GX_OK;
+        };
+        if (!q->m_hist)
+        {
+            HIP_CHK(ctx, hipMalloc(&q->m_hist, n * 8));
+            HIP_CHK(ctx, hipMalloc(&q->m_cur, n * 8));
+            HIP_CHK(ctx, hipMalloc(&q->m_cnts_mine, n * 8));
+            HIP_CHK(ctx, hipMalloc(&q->m_cnts_all, (int64_t) n * n * 8));
+        }
+        unsigned long long *dhist = q->m_hist;
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                            od.dstream, od.m, oc.dstream, oc.m,
@@ -2690,11 +2716,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         std::vector<unsigned long long> off1(n + 1, 0);
         for (int i = 0; i < n; i++) off1[i + 1] = off1[i] + h1[i];
         unsigned long long send1_n = off1[n];
-        (void) send1_n;
-        HIP_CHK(ctx, send1_b.alloc(std::max<uint64_t>(send1_n, 1) * sizeof(gx_ord_row)));
-        gx_ord_row *send1 = send1_b.as<gx_ord_row>();
-        HIP_CHK(ctx, dcur_b.alloc(n * 8));
-        unsigned long long *dcur = dcur_b.as<unsigned long long>();
+        if (grow(q->m_send1, q->m_send1_cap, send1_n) != GX_OK) return GX_ERR_OOM;
+        gx_ord_row *send1 = q->m_send1;
+        unsigned long long *dcur = q->m_cur;
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
@@ -2702,10 +2726,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                            (int32_t) D.mid_filter.literal, n, dcur, send1);
 
         /* exchange counts (all-gather of per-dest counts) */
-        HIP_CHK(ctx, cnts_mine_b.alloc(n * 8));
-        HIP_CHK(ctx, cnts_all_b.alloc((int64_t) n * n * 8));
-        unsigned long long *dcnts_mine = cnts_mine_b.as<unsigned long long>();
-        unsigned long long *dcnts_all = cnts_all_b.as<unsigned long long>();
+        unsigned long long *dcnts_mine = q->m_cnts_mine;
+        unsigned long long *dcnts_all = q->m_cnts_all;
         HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h1.data(), n * 8, hipMemcpyHostToDevice, s));
         RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
         std::vector<unsigned long long> cnts_all(n * n);
@@ -2716,8 +2738,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         for (int r = 0; r < n; r++) rcv1[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff1[r + 1] = roff1[r] + rcv1[r];
         unsigned long long recv1_n = roff1[n];
-        HIP_CHK(ctx, recv1_b.alloc(std::max<uint64_t>(recv1_n, 1) * sizeof(gx_ord_row)));
-        gx_ord_row *recv1 = recv1_b.as<gx_ord_row>();
+        if (grow(q->m_recv1, q->m_recv1_cap, recv1_n) != GX_OK) return GX_ERR_OOM;
+        gx_ord_row *recv1 = q->m_recv1;
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
@@ -2745,8 +2767,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipStreamSynchronize(s));
         std::vector<unsigned long long> off2(n + 1, 0);
         for (int i = 0; i < n; i++) off2[i + 1] = off2[i] + h2[i];
-        HIP_CHK(ctx, send2_b.alloc(std::max<uint64_t>(off2[n], 1) * sizeof(gx_qual_row)));
-        gx_qual_row *send2 = send2_b.as<gx_qual_row>();
+        if (grow(q->m_send2, q->m_send2_cap, off2[n]) != GX_OK) return GX_ERR_OOM;
+        gx_qual_row *send2 = q->m_send2;
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off2.data(), n * 8, hipMemcpyHostToDevice, s));
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_qual_emit<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
@@ -2765,8 +2787,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         for (int r = 0; r < n; r++) rcv2[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff2[r + 1] = roff2[r] + rcv2[r];
         unsigned long long recv2_n = roff2[n];
-        HIP_CHK(ctx, recv2_b.alloc(std::max<uint64_t>(recv2_n, 1) * sizeof(gx_qual_row)));
-        gx_qual_row *recv2 = recv2_b.as<gx_qual_row>();
+        if (grow(q->m_recv2, q->m_recv2_cap, recv2_n) != GX_OK) return GX_ERR_OOM;
+        gx_qual_row *recv2 = q->m_recv2;
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
