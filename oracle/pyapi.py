@@ -148,6 +148,73 @@ def q1(sf, cutoff, seed=42):
             "sum_revenue": np.array([g.sum_revenue for g in out])}
 
 
+_refw = None
+
+
+def ref_writer():
+    """The REFERENCE's own datum-stream writer (oracle/_ref/libpgwriter.so,
+    compiled from /root/reference sources) — None when not built."""
+    global _refw
+    if _refw is None:
+        so = os.path.join(_ROOT, "oracle", "_ref", "libpgwriter.so")
+        if not os.path.exists(so):
+            return None
+        _refw = ctypes.CDLL(so)
+        _refw.refw_encode.restype = ctypes.c_int
+        _refw.refw_encode.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+                                      ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                                      ctypes.c_int32, ctypes.c_void_p, ctypes.c_int64,
+                                      ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
+    return _refw
+
+
+def ref_writer_stream(vals, version=2, rle=1, delta=1, blocksize=32768):
+    """Run the reference writer and wrap its content blocks in our AO
+    envelope (SmallContent/NonBulkDense + CRC32C pair + firstRowNum) —
+    the stream a real segment file would hold.  Returns bytes or None."""
+    w = ref_writer()
+    if w is None:
+        return None
+    vals = np.ascontiguousarray(vals)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 21)
+    out = np.zeros(cap, np.uint8)
+    lens = np.zeros(65536, np.int32)
+    rows = np.zeros(65536, np.int32)
+    nb = w.refw_encode(vals.ctypes.data, width, len(vals), version, rle, delta,
+                       blocksize - 32, out.ctypes.data, cap,
+                       lens.ctypes.data, rows.ctypes.data, 65536)
+    assert nb > 0, nb
+    stream = bytearray()
+    off = 0
+    frn = 1
+    for b in range(nb):
+        content = out[off:off + lens[b]].tobytes()
+        off += lens[b]
+        logical = int(rows[b])
+        clen = len(content)
+        if logical <= 16383:
+            b03 = (1 << 28) | (1 << 27) | (1 << 24) | (logical << 10) | (clen >> 11)
+            b47 = (clen & 0x7FF) << 21
+        else:
+            b03 = (3 << 28) | (1 << 27) | (1 << 24) | (clen & 0x1FFFFF)
+            b47 = logical & 0x3FFFFFFF
+        blocklen = (24 + clen + 7) & ~7
+        blk = bytearray(blocklen)
+        blk[0:4] = b03.to_bytes(4, "little")
+        blk[4:8] = b47.to_bytes(4, "little")
+        blk[16:24] = frn.to_bytes(8, "little")
+        blk[24:24 + clen] = content
+        bc = lib.orc_crc32c(0xFFFFFFFF, bytes(blk[16:]), blocklen - 16)
+        hc_in = bytes(blk[0:8]) + bc.to_bytes(4, "little")
+        hc = lib.orc_crc32c(0xFFFFFFFF, hc_in, 12)
+        blk[8:12] = bc.to_bytes(4, "little")
+        blk[12:16] = hc.to_bytes(4, "little")
+        stream += blk
+        frn += logical
+    return bytes(stream)
+
+
 def set_threads(n):
     return lib.orc_set_threads(n)
 

@@ -48,3 +48,4 @@
 #define PG_VERSION_STR "PostgreSQL 16.9 oracle-ref stub"
 #define MEMSET_LOOP_LIMIT 1024
 #define USE_FLOAT8_BYVAL 1
+#define INT64_MODIFIER "l"

@@ -516,3 +516,17 @@ def test_q3_descriptor_api(ctx, orc):
                                np.array([rev[k] for k in keys.tolist()]),
                                rtol=1e-6)
     li.free(); ordr.free(); cust.free()
+
+
+def test_gpu_decodes_reference_writer_blocks(ctx, orc):
+    """The GPU decoder over blocks written by the REFERENCE's own
+    datumstreamblock.c (rle_type + delta_range)."""
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    rng = np.random.default_rng(20260915)
+    keys = np.repeat(np.arange(1, 12000, dtype=np.int64),
+                     rng.integers(1, 8, 11999))
+    s = open(os.path.join(here, "golden", "refwriter_rle_delta_i64.bin"), "rb").read()
+    t = ctx.bind([(s, 8, len(keys), 1)])
+    np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True), keys)
+    t.free()
