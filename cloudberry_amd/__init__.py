@@ -100,6 +100,10 @@ def _load():
     lib.gx_table_logical_bytes.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_double)]
     lib.gx_tpch_gen.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_double,
                                 ctypes.c_uint64, ctypes.POINTER(ctypes.c_void_p)]
+    lib.gx_table_dump_stream.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_int, ctypes.c_void_p,
+                                         ctypes.c_int64,
+                                         ctypes.POINTER(ctypes.c_int64)]
     lib.gx_decode_column.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                                      ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
     lib.gx_q1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
@@ -325,6 +329,15 @@ class Table:
                                                    literal, ctypes.byref(n),
                                                    ctypes.byref(ms)))
         return n.value, ms.value
+
+    def dump_stream(self, col):
+        """Raw AOCS stream bytes of a column (byte-level parity tests)."""
+        cap = self.nrows * 16 + (1 << 22)
+        buf = np.zeros(cap, np.uint8)
+        n = ctypes.c_int64()
+        self.ctx._chk(self.ctx._lib.gx_table_dump_stream(
+            self.ctx._h, self._t, col, buf.ctypes.data, cap, ctypes.byref(n)))
+        return buf[:n.value].tobytes()
 
     def decode_column(self, col, dtype, verify=True):
         n = self.nrows

@@ -2222,6 +2222,22 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     return GX_OK;
 }
 
+/* copy a column's raw device stream back to the host (parity testing:
+ * byte-compare the GPU encoder against the oracle/reference writer) */
+extern "C" gx_status gx_table_dump_stream(gx_ctx *ctx, const gx_table *t,
+                                          int col, void *host_out,
+                                          int64_t cap_bytes, int64_t *nbytes)
+{
+    if (!ctx || !t || col < 0 || col >= (int) t->cols.size()) return GX_ERR_INVALID;
+    const gx_col &c = t->cols[col];
+    if (c.m.nbytes > cap_bytes) return GX_ERR_INVALID;
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, c.dstream, c.m.nbytes,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    *nbytes = c.m.nbytes;
+    return GX_OK;
+}
+
 extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx,
                                       void *host_out, int64_t cap_rows,
                                       int verify_checksums)

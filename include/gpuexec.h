@@ -109,6 +109,11 @@ typedef enum { GX_TPCH_CUSTOMER = 0, GX_TPCH_ORDERS = 1, GX_TPCH_LINEITEM = 2,
 gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                       uint64_t seed, gx_table **out);
 
+/* copy a column's raw AOCS stream bytes back to the host (parity tests) */
+gx_status gx_table_dump_stream(gx_ctx *ctx, const gx_table *t, int col,
+                               void *host_out, int64_t cap_bytes,
+                               int64_t *nbytes);
+
 /* decode one column back to host values (parity testing / config-2 path);
  * verify_checksums runs the CRC32C pair per block on device */
 gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int col,

@@ -181,8 +181,11 @@ def ref_writer_stream(vals, version=2, rle=1, delta=1, blocksize=32768):
     out = np.zeros(cap, np.uint8)
     lens = np.zeros(65536, np.int32)
     rows = np.zeros(65536, np.int32)
+    # header reserve: AoHeader_Size(isLong, checksum=true, firstRowNum=true)
+    # = 24 for Orig (regular) streams, 32 for Dense (datumstream.c:588-605)
+    reserve = 24 if version == 0 else 32
     nb = w.refw_encode(vals.ctypes.data, width, len(vals), version, rle, delta,
-                       blocksize - 32, out.ctypes.data, cap,
+                       blocksize - reserve, out.ctypes.data, cap,
                        lens.ctypes.data, rows.ctypes.data, 65536)
     assert nb > 0, nb
     stream = bytearray()
