@@ -546,9 +546,9 @@ int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
  *                     cdbappendonlystorage.h:76-96, _int.h:253-347)
  * Capacity mirrors the writer's reservation: content < blocksize − 32
  * (maxAoHeaderSize for Dense streams, datumstream.c:597-605).
- * NOTE: conforming-subset writer, not byte-exact with the reference's
- * rle_type writer (which also delta-compresses int/date columns); the
- * decoder is what the GPU path needs — DESIGN.md "next rows".
+ * The encoder below (dense_encode) is a full restatement of the writer's
+ * state machine and is BYTE-EXACT with the compiled reference writer —
+ * see the block comment above dense_encode and tests/test_oracle_cpu.py.
  * ====================================================================== */
 
 static int32_t varint_size(int32_t v)
@@ -578,122 +578,29 @@ static int32_t varint_decode(const uint8_t *b, int32_t *len)
     return v;
 }
 
-int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
-                            int64_t first_rownum, int32_t blocksize,
-                            uint8_t *out, int64_t outcap)
-{
-    const uint8_t *src = (const uint8_t *) vals;
-    int32_t maxdata = blocksize - 32;
-    int64_t off = 0, row = 0;
-    /* per-block scratch */
-    int32_t cap_phys = blocksize;              /* physical ≤ content bytes */
-    uint8_t *pvals = malloc((size_t) cap_phys * width);
-    int32_t *extra = malloc(sizeof(int32_t) * cap_phys);
-
-    while (row < nrows)
-    {
-        int32_t phys = 0, non = 0;             /* physical datums, ON bits */
-        int32_t csize = 0;                     /* repeat-counts bytes */
-        int64_t logical = 0;
-        /* greedy fill */
-        while (row + logical < nrows && logical < 0x3FFFFFFE)
-        {
-            const uint8_t *d = src + (row + logical) * width;
-            int same = phys > 0 &&
-                       memcmp(pvals + (size_t) (phys - 1) * width, d, width) == 0 &&
-                       extra[phys - 1] < 0x3FFFFFFF;
-            int32_t new_phys = phys, new_non = non, new_csize = csize;
-            if (same)
-            {
-                int32_t e = extra[phys - 1];
-                if (e == 0) { new_non++; new_csize += 1; }
-                else new_csize += varint_size(e + 1) - varint_size(e);
-            }
-            else
-                new_phys++;
-            int32_t bm = (new_phys + 7) >> 3;
-            int32_t hdr = 16 + (new_non ? 16 + bm + new_csize : 0);
-            int64_t tot = ((hdr + 7) & ~7) + (int64_t) new_phys * width;
-            if (tot >= maxdata && logical > 0)
-                break;
-            if (same)
-            {
-                extra[phys - 1]++;
-                non = new_non; csize = new_csize;
-            }
-            else
-            {
-                memcpy(pvals + (size_t) phys * width, d, width);
-                extra[phys] = 0;
-                phys = new_phys;
-            }
-            logical++;
-        }
-
-        /* emit block */
-        int has_rle = non > 0;
-        int32_t bm = (phys + 7) >> 3;
-        int32_t hdr = 16 + (has_rle ? 16 + bm + csize : 0);
-        int32_t datum_off = (hdr + 7) & ~7;
-        int32_t content = datum_off + phys * width;
-        int64_t blocklen = (24 + content + 7) & ~7LL;
-        if (off + blocklen > outcap) { free(pvals); free(extra); return -1; }
-        uint8_t *blk = out + off;
-        memset(blk, 0, blocklen);
-
-        uint32_t kind = (logical <= 16383) ? 1u : 3u;   /* Small / NonBulkDense */
-        uint32_t b03 = (kind << 28) | (1u << 27) | (1u << 24);
-        uint32_t b47 = 0;
-        if (kind == 1)
-        {
-            b03 |= (0x00FFFC00u & ((uint32_t) logical << 10)) |
-                   (((uint32_t) content >> 11) & 0x3FFu);
-            b47 = (((uint32_t) content & 0x7FFu) << 21);
-        }
-        else
-        {
-            b03 |= ((uint32_t) content & 0x1FFFFFu);
-            b47 = (uint32_t) logical & 0x3FFFFFFFu;
-        }
-        put_u32le(blk, b03);
-        put_u32le(blk + 4, b47);
-        int64_t frn = first_rownum + row;
-        memcpy(blk + 16, &frn, 8);
-
-        uint8_t *c = blk + 24;
-        int16_t v16 = 2; memcpy(c, &v16, 2);                 /* Dense_Enhanced */
-        v16 = has_rle ? 2 : 0; memcpy(c + 2, &v16, 2);       /* flags */
-        int32_t v32 = (int32_t) logical; memcpy(c + 4, &v32, 4);
-        v32 = phys; memcpy(c + 8, &v32, 4);
-        v32 = phys * width; memcpy(c + 12, &v32, 4);
-        if (has_rle)
-        {
-            v32 = 0; memcpy(c + 16, &v32, 4);                /* null bitmap bits */
-            v32 = phys; memcpy(c + 20, &v32, 4);             /* compress bitmap bits */
-            v32 = non; memcpy(c + 24, &v32, 4);
-            v32 = csize; memcpy(c + 28, &v32, 4);
-            uint8_t *bmp = c + 32;
-            uint8_t *cnts = bmp + bm;
-            int32_t w = 0;
-            for (int32_t p = 0; p < phys; p++)
-                if (extra[p] > 0)
-                {
-                    bmp[p >> 3] |= (uint8_t) (1u << (p & 7));
-                    w += varint_encode(cnts + w, extra[p]);
-                }
-        }
-        memcpy(c + datum_off, pvals, (size_t) phys * width);
-
-        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
-        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
-
-        off += blocklen;
-        row += logical;
-    }
-    free(pvals); free(extra);
-    return off;
-}
-
+/* ======================================================================
+ * Dense_Enhanced RLE_TYPE / DELTA writer — EXACT restatement of the
+ * reference's DatumStreamBlockWrite state machine for fixed-width,
+ * non-null columns (compresstype=rle_type):
+ *   Put flow        datumstreamblock.c:3341-3541 (PutDense fixed-width)
+ *   capacity        :1992-2330 (DenseHasSpaceItem/Repeat/Delta),
+ *                   :1944-1990 (DenseRleSpace, incl. the pending
+ *                   repeat-count reservation)
+ *   lazy finalize   :2435-2470 (RleFinalizeRepeatCountSize)
+ *   repeat marking  :2477-2672 (RleMarkRepeat/RleIncrRepeated, with the
+ *                   compress-bitmap zero-fill to phys-1+deltaOnCount)
+ *   delta           :2989-3091 (PerformDeltaCompression/DeltaAdd;
+ *                   MAX_DELTA 0x1FFFFFFF, widths 4/8 only, unsigned
+ *                   compare, delta==0 becomes a DELTA item)
+ *   block assembly  :3803-4246 (BlockDense: Dense hdr, Rle_Extension,
+ *                   Delta_Extension, bitmaps, varints, MAXALIGN pad)
+ *   AO envelope     SmallContent when rows <= 0x3FFF else NonBulkDense
+ *                   (cdbappendonlystorage.h:23,30), CRC32C pair
+ * maxDataBlockSize = blocksize - 32 (maxAoHeaderSize for Dense streams,
+ * datumstream.c:588-605).  Byte-for-byte equality with the COMPILED
+ * reference writer (oracle/_ref/libpgwriter.so) is asserted in
+ * tests/test_oracle_cpu.py.
+ * ====================================================================== */
 
 /* sign+magnitude delta varint (Reserved3): top 2 bits = len-1, bit 5 =
  * POSITIVE flag — datumstreamblock.h:790-930 */
@@ -724,7 +631,6 @@ static int64_t varint3_decode(const uint8_t *b, int32_t *len, int *positive)
     return v;
 }
 
-/* load/advance helpers for width-generic item compare */
 static uint64_t item_at(const uint8_t *v, int width, int64_t i)
 {
     uint64_t x = 0;
@@ -732,170 +638,426 @@ static uint64_t item_at(const uint8_t *v, int width, int64_t i)
     return x;
 }
 
+/* bit-map writer, LSB-first per byte (datumstreamblock.h:375-424) */
+typedef struct {
+    uint8_t *buf;
+    int32_t bits, on;
+} dbm_t;
+
+static void dbm_reset(dbm_t *b) { b->bits = 0; b->on = 0; }
+static void dbm_add(dbm_t *b, int on)
+{
+    if ((b->bits & 7) == 0)
+        b->buf[b->bits >> 3] = 0;
+    if (on)
+    {
+        b->buf[b->bits >> 3] |= (uint8_t) (1u << (b->bits & 7));
+        b->on++;
+    }
+    b->bits++;
+}
+static void dbm_set_last(dbm_t *b)          /* BitMapWrite_Set on current */
+{
+    b->buf[(b->bits - 1) >> 3] |= (uint8_t) (1u << ((b->bits - 1) & 7));
+    b->on++;
+}
+static void dbm_zerofill(dbm_t *b, int32_t n)   /* :330-368 */
+{
+    memset(b->buf, 0, (size_t) ((n + 7) >> 3));
+    b->bits = n;
+    b->on = 0;
+}
+#define DBM_SIZE(b) (((b)->bits + 7) >> 3)
+
+#define DWR_MAXALIGN(x) (((x) + 7) & ~7)
+#define DWR_MAXDATUM 0x3FFFFFFF          /* MAXDATUM_PER_AOCS_DENSE_BLOCK */
+#define DWR_MAXREPEAT 0x3FFFFFFF         /* MAXREPEAT_COUNT, datumstreamblock.h:991 */
+
+typedef struct {
+    int width, rle_want, delta_want;
+    int32_t maxdata;                     /* maxDataBlockSize */
+    int32_t nth, phys;                   /* logical rows / physical datums */
+    uint8_t *datum_buffer;
+    int64_t datum_used;
+    /* RLE_TYPE state */
+    int rle_has, last_valid, last_repeated;
+    uint64_t last_item;
+    dbm_t cbm;                           /* compress bitmap */
+    int32_t *repeats;
+    int32_t nrepeats, repeats_size;      /* repeats_size = FINALIZED runs only */
+    /* DELTA state */
+    int delta_has, not_first;
+    uint64_t compare_item;
+    dbm_t dbm;                           /* delta bitmap */
+    int64_t *deltas;
+    uint8_t *dsigns;
+    int32_t ndeltas, deltas_size;
+} dwr_t;
+
+static void dwr_getready(dwr_t *w)       /* GetReady, :3588-3669 */
+{
+    w->nth = 0; w->phys = 0; w->datum_used = 0;
+    w->rle_has = 0; w->last_valid = 0; w->last_repeated = 0; w->last_item = 0;
+    dbm_reset(&w->cbm);
+    w->nrepeats = 0; w->repeats_size = 0;
+    w->delta_has = 0; w->not_first = 0; w->compare_item = 0;
+    dbm_reset(&w->dbm);
+    w->ndeltas = 0; w->deltas_size = 0;
+}
+
+/* DenseRleSpace, non-null path (:1944-1990) */
+static void dwr_rle_space(const dwr_t *w, int32_t *hdr, int32_t *rle)
+{
+    if (!w->rle_has)
+        return;
+    *hdr += 16;                                  /* Rle_Extension */
+    *rle += (w->cbm.bits + 1 + 7) >> 3;          /* NextSize */
+    *rle += w->repeats_size;
+    if (w->last_repeated)                        /* pending finalize reservation */
+        *rle += varint_size(w->repeats[w->nrepeats - 1]);
+}
+
+static int dwr_has_space_item(const dwr_t *w, int32_t sz)   /* :2330-2378 */
+{
+    int32_t hdr = 16, rle = 0, delta = 0;
+    if (w->nth + 1 >= DWR_MAXDATUM)
+        return 0;
+    dwr_rle_space(w, &hdr, &rle);
+    if (w->delta_has)
+    {
+        hdr += 12;                               /* Delta_Extension */
+        delta = ((w->dbm.bits + 1 + 7) >> 3) + w->deltas_size;
+    }
+    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used + sz <= w->maxdata;
+}
+
+static int dwr_has_space_repeat(const dwr_t *w, int new_repeat)  /* :2098-2220 */
+{
+    int32_t hdr = 16, delta = 0, rle, total;
+    if (w->nth + 1 >= DWR_MAXDATUM)
+        return 0;
+    total = w->phys;
+    if (w->delta_has)
+    {
+        hdr += 12;
+        delta = ((w->dbm.bits + 1 + 7) >> 3) + w->deltas_size;
+        total += w->dbm.on;
+    }
+    hdr += 16;                                   /* Rle_Extension, unconditional */
+    rle = (total + (new_repeat ? 1 : 0) + 7) >> 3;
+    rle += w->repeats_size + 4;                  /* + Int32Compress_MaxByteLen */
+    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used <= w->maxdata;
+}
+
+static int dwr_has_space_delta(const dwr_t *w)   /* :2224-2328 */
+{
+    int32_t hdr = 16, rle = 0, total, delta;
+    if (w->nth + 1 >= DWR_MAXDATUM)
+        return 0;
+    dwr_rle_space(w, &hdr, &rle);
+    total = w->phys + (w->delta_has ? w->dbm.on : 0);
+    hdr += 12;
+    delta = ((total + 1 + 7) >> 3) + w->deltas_size + 4;  /* + Reserved3_MaxByteLen */
+    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used <= w->maxdata;
+}
+
+static void dwr_finalize_repeat(dwr_t *w)        /* :2435-2470 */
+{
+    w->last_repeated = 0;
+    w->last_valid = 0;
+    w->repeats_size += varint_size(w->repeats[w->nrepeats - 1]);
+}
+
+static void dwr_incr_repeated(dwr_t *w)          /* RleIncrRepeated, :2558-2651 */
+{
+    if (!w->last_repeated)
+    {
+        w->last_repeated = 1;
+        if (!w->rle_has)
+        {
+            /* zero-fill a bit per prior physical+delta item, then mark this
+             * one repeated (:1880-1886) */
+            dbm_zerofill(&w->cbm, w->phys - 1 + (w->delta_has ? w->dbm.on : 0));
+            dbm_add(&w->cbm, 1);
+            w->rle_has = 1;
+        }
+        else
+            dbm_set_last(&w->cbm);
+        w->repeats[w->nrepeats++] = 1;           /* count = EXTRA repeats */
+    }
+    else
+        w->repeats[w->nrepeats - 1]++;
+    w->nth++;
+}
+
+/* PerformDeltaCompression (:2989-3091): 0 = OK, 1 = ERROR, 2 = NOT_APPLIED */
+static int dwr_perform_delta(dwr_t *w, uint64_t v)
+{
+    uint64_t mag;
+    int positive;
+    if (!w->delta_want)
+        return 2;
+    if (!w->not_first)                   /* first datum of block stays physical */
+    {
+        w->not_first = 1;
+        return 2;
+    }
+    if (w->width == 4)
+    {
+        uint32_t c = (uint32_t) w->compare_item, d = (uint32_t) v;
+        if (c <= d) { mag = d - c; positive = 1; }
+        else        { mag = c - d; positive = 0; }
+    }
+    else
+    {
+        if (w->compare_item <= v) { mag = v - w->compare_item; positive = 1; }
+        else                      { mag = w->compare_item - v; positive = 0; }
+        if (mag > 0x7FFFFFFFFFFFFFFFULL)         /* int64 overflow → delta < 0 */
+            return 2;
+    }
+    if (mag > 0x1FFFFFFF)                        /* MAX_DELTA_SUPPORTED... */
+        return 2;
+    if (!dwr_has_space_delta(w))
+        return 1;
+    w->compare_item = v;
+    /* DeltaAdd (:2847-2987) */
+    if (!w->delta_has)
+        dbm_zerofill(&w->dbm, w->phys);
+    w->delta_has = 1;
+    if (w->last_repeated)
+        dwr_finalize_repeat(w);
+    w->last_item = v;
+    w->last_valid = 1;
+    if (w->rle_has)
+        dbm_add(&w->cbm, 0);
+    dbm_add(&w->dbm, 1);
+    w->deltas[w->ndeltas] = (int64_t) mag;
+    w->dsigns[w->ndeltas] = (uint8_t) positive;
+    w->deltas_size += varint3_size((int64_t) mag);
+    w->ndeltas++;
+    w->nth++;
+    return 0;
+}
+
+/* PutDense, fixed-width non-null (:3341-3541); >0 stored, 0 folded, <0 full */
+static int dwr_put(dwr_t *w, uint64_t v)
+{
+    int have_prev = w->rle_want && w->last_valid;
+    if (w->last_repeated && w->repeats[w->nrepeats - 1] >= DWR_MAXREPEAT)
+        dwr_finalize_repeat(w);
+    else if (have_prev)
+    {
+        int eq = (w->width == 8) ? (w->last_item == v)
+               : (w->width == 4) ? ((uint32_t) w->last_item == (uint32_t) v)
+               : (w->width == 2) ? ((uint16_t) w->last_item == (uint16_t) v)
+               : ((uint8_t) w->last_item == (uint8_t) v);
+        if (eq)
+        {
+            if (!dwr_has_space_repeat(w, !w->last_repeated))
+                return -1;
+            dwr_incr_repeated(w);
+            return 0;
+        }
+        if (w->last_repeated)
+            dwr_finalize_repeat(w);
+    }
+    if (w->delta_want)
+    {
+        int st = dwr_perform_delta(w, v);
+        if (st == 0) return 0;
+        if (st == 1) return -1;
+    }
+    if (!dwr_has_space_item(w, w->width))
+        return -w->width;
+    memcpy(w->datum_buffer + w->datum_used, &v, w->width);
+    w->datum_used += w->width;
+    /* DenseIncrItem (:2506-2556) */
+    if (w->last_repeated)
+        dwr_finalize_repeat(w);
+    if (w->rle_want)
+    {
+        w->last_item = v;
+        w->last_valid = 1;
+    }
+    if (w->rle_has)
+        dbm_add(&w->cbm, 0);
+    w->nth++;
+    w->phys++;
+    /* DeltaMaintain (:2800-2845) */
+    if (w->delta_want)
+    {
+        w->compare_item = v;
+        if (w->delta_has)
+            dbm_add(&w->dbm, 0);
+    }
+    return w->width;
+}
+
+/* BlockDense (:3803-4246) + AO envelope; returns whole-block length */
+static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
+                         uint8_t *blk, int64_t cap)
+{
+    int32_t hdr, rle, delta, meta, aligned;
+    int64_t content, blocklen;
+    int16_t v16;
+    int32_t v32;
+    uint8_t *c, *p;
+    uint32_t kind, b03, b47;
+
+    if (w->last_repeated)
+        dwr_finalize_repeat(w);
+    hdr = 16 + (w->rle_has ? 16 : 0) + (w->delta_has ? 12 : 0);
+    rle = w->rle_has ? DBM_SIZE(&w->cbm) + w->repeats_size : 0;
+    delta = w->delta_has ? DBM_SIZE(&w->dbm) + w->deltas_size : 0;
+    meta = hdr + rle + delta;
+    aligned = DWR_MAXALIGN(meta);
+    content = aligned + w->datum_used;
+    blocklen = (24 + content + 7) & ~7LL;
+    if (blocklen > cap)
+        return -1;
+    memset(blk, 0, (size_t) blocklen);
+
+    kind = (w->nth <= 0x3FFF) ? 1u : 3u;         /* Small / NonBulkDense */
+    b03 = (kind << 28) | (1u << 27) | (1u << 24);
+    b47 = 0;
+    if (kind == 1)
+    {
+        b03 |= (0x00FFFC00u & ((uint32_t) w->nth << 10)) |
+               (((uint32_t) content >> 11) & 0x3FFu);
+        b47 = (((uint32_t) content & 0x7FFu) << 21);
+    }
+    else
+    {
+        b03 |= ((uint32_t) content & 0x1FFFFFu);
+        b47 = (uint32_t) w->nth & 0x3FFFFFFFu;
+    }
+    put_u32le(blk, b03);
+    put_u32le(blk + 4, b47);
+    memcpy(blk + 16, &first_rownum, 8);
+
+    c = blk + 24;
+    v16 = 2;            memcpy(c, &v16, 2);      /* Dense_Enhanced */
+    v16 = (int16_t) ((w->rle_has ? 2 : 0) | (w->delta_has ? 4 : 0));
+    memcpy(c + 2, &v16, 2);
+    v32 = w->nth;       memcpy(c + 4, &v32, 4);
+    v32 = w->phys;      memcpy(c + 8, &v32, 4);
+    v32 = (int32_t) w->datum_used; memcpy(c + 12, &v32, 4);
+    p = c + 16;
+    if (w->rle_has)
+    {
+        v32 = 0;               memcpy(p, &v32, 4);      /* no nulls */
+        v32 = w->cbm.bits;     memcpy(p + 4, &v32, 4);
+        v32 = w->nrepeats;     memcpy(p + 8, &v32, 4);
+        v32 = w->repeats_size; memcpy(p + 12, &v32, 4);
+        p += 16;
+    }
+    if (w->delta_has)
+    {
+        v32 = w->dbm.bits;    memcpy(p, &v32, 4);
+        v32 = w->ndeltas;     memcpy(p + 4, &v32, 4);
+        v32 = w->deltas_size; memcpy(p + 8, &v32, 4);
+        p += 12;
+    }
+    if (w->rle_has)
+    {
+        memcpy(p, w->cbm.buf, DBM_SIZE(&w->cbm));
+        p += DBM_SIZE(&w->cbm);
+        for (int32_t i = 0; i < w->nrepeats; i++)
+            p += varint_encode(p, w->repeats[i]);
+    }
+    if (w->delta_has)
+    {
+        memcpy(p, w->dbm.buf, DBM_SIZE(&w->dbm));
+        p += DBM_SIZE(&w->dbm);
+        for (int32_t i = 0; i < w->ndeltas; i++)
+            p += varint3_encode(p, w->deltas[i], w->dsigns[i]);
+    }
+    memcpy(blk + 24 + aligned, w->datum_buffer, (size_t) w->datum_used);
+
+    put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+    put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+    return blocklen;
+}
+
+static int64_t dense_encode(const void *vals, int width, int64_t nrows,
+                            int64_t first_rownum, int32_t blocksize,
+                            int rle_want, int delta_want,
+                            uint8_t *out, int64_t outcap)
+{
+    const uint8_t *src = (const uint8_t *) vals;
+    dwr_t w;
+    int64_t off = 0, emitted = 0;
+    size_t scratch;
+
+    if (width != 1 && width != 2 && width != 4 && width != 8)
+        return -1;
+    if (delta_want && width != 4 && width != 8)
+        return -1;                       /* DeltaMaintain FATALs otherwise */
+    memset(&w, 0, sizeof(w));
+    w.width = width;
+    w.rle_want = rle_want;
+    w.delta_want = delta_want;
+    w.maxdata = blocksize - 32;          /* Dense maxAoHeaderSize reserve */
+    scratch = (size_t) blocksize * 2 + 64;
+    w.datum_buffer = malloc((size_t) w.maxdata + 16);
+    w.cbm.buf = malloc(scratch);
+    w.dbm.buf = malloc(scratch);
+    w.repeats = malloc(scratch * sizeof(int32_t));
+    w.deltas = malloc(scratch * sizeof(int64_t));
+    w.dsigns = malloc(scratch);
+    if (!w.datum_buffer || !w.cbm.buf || !w.dbm.buf ||
+        !w.repeats || !w.deltas || !w.dsigns)
+        goto fail;
+    dwr_getready(&w);
+    for (int64_t i = 0; i < nrows; i++)
+    {
+        uint64_t v = item_at(src, width, i);
+        if (dwr_put(&w, v) < 0)
+        {
+            int64_t bl = dwr_block(&w, first_rownum + emitted,
+                                   out + off, outcap - off);
+            if (bl < 0)
+                goto fail;
+            off += bl;
+            emitted += w.nth;
+            dwr_getready(&w);
+            if (dwr_put(&w, v) < 0)
+                goto fail;
+        }
+    }
+    if (w.nth > 0)
+    {
+        int64_t bl = dwr_block(&w, first_rownum + emitted,
+                               out + off, outcap - off);
+        if (bl < 0)
+            goto fail;
+        off += bl;
+    }
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf);
+    free(w.repeats); free(w.deltas); free(w.dsigns);
+    return off;
+fail:
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf);
+    free(w.repeats); free(w.deltas); free(w.dsigns);
+    return -1;
+}
+
+int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
+                            int64_t first_rownum, int32_t blocksize,
+                            uint8_t *out, int64_t outcap)
+{
+    return dense_encode(vals, width, nrows, first_rownum, blocksize,
+                        1, 0, out, outcap);
+}
+
 int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
                                   int64_t first_rownum, int32_t blocksize,
                                   uint8_t *out, int64_t outcap)
 {
-    const uint8_t *src = (const uint8_t *) vals;
-    int32_t maxdata = blocksize - 32;
-    int64_t off = 0, row = 0;
-    int cap_items = blocksize * 2;
-    uint8_t *pvals = malloc((size_t) cap_items * width);  /* physical datums */
-    int32_t *extra = malloc(sizeof(int32_t) * cap_items); /* repeats per NEW item */
-    uint8_t *isdelta = malloc(cap_items);                 /* per NEW item */
-    uint8_t *dbytes = malloc((size_t) cap_items * 4);     /* delta varints */
-
-    while (row < nrows)
-    {
-        int32_t items = 0, phys = 0, non = 0, ndelta = 0, dsize = 0, csize = 0;
-        int64_t logical = 0;
-        uint64_t prev = 0;
-        while (row + logical < nrows && items < cap_items)
-        {
-            uint64_t v = item_at(src, width, row + logical);
-            int same = items > 0 && v == prev && extra[items - 1] < 0x3FFFFFFF;
-            if (same)
-            {
-                /* capacity: count growth */
-                int32_t grow = (extra[items - 1] == 0) ? 1 :
-                               varint_size(extra[items - 1] + 1) - varint_size(extra[items - 1]);
-                int32_t bm_items = ((items + 7) >> 3) * 2;   /* compress + delta bitmaps */
-                int32_t hdr = 16 + 16 + 12 + bm_items + (csize + grow) + dsize;
-                int64_t tot = ((hdr + 7) & ~7) + (int64_t) phys * width;
-                if (tot >= maxdata && logical > 0) break;
-                if (extra[items - 1] == 0) non++;
-                csize += grow;
-                extra[items - 1]++;
-                logical++;
-                continue;
-            }
-            /* new item: delta or physical */
-            int64_t d = 0;
-            int use_delta = 0;
-            if (items > 0)
-            {
-                if (width == 8)
-                    d = (int64_t) (v - prev);
-                else
-                    d = (int64_t) (int32_t) ((uint32_t) v - (uint32_t) prev);
-                int64_t mag = d < 0 ? -d : d;
-                if (mag <= 0x1FFFFFFF && d != 0) use_delta = 1;
-                if (d == 0) use_delta = 0;   /* handled by RLE above anyway */
-                if (use_delta) d = mag * (d < 0 ? -1 : 1);
-            }
-            int32_t dvn = use_delta ? varint3_size(d < 0 ? -d : d) : 0;
-            int32_t new_phys = phys + (use_delta ? 0 : 1);
-            int32_t bm_items = (((items + 1) + 7) >> 3) * 2;
-            int32_t hdr = 16 + 16 + 12 + bm_items + csize + dsize + dvn;
-            int64_t tot = ((hdr + 7) & ~7) + (int64_t) new_phys * width;
-            if (tot >= maxdata && logical > 0) break;
-            isdelta[items] = (uint8_t) use_delta;
-            extra[items] = 0;
-            if (use_delta)
-            {
-                dsize += varint3_encode(dbytes + dsize, d < 0 ? -d : d, d > 0);
-                ndelta++;
-            }
-            else
-            {
-                memcpy(pvals + (size_t) phys * width, &v, width);
-                phys++;
-            }
-            items++;
-            prev = v;
-            logical++;
-        }
-
-        int has_rle = non > 0;
-        int has_delta = ndelta > 0;
-        int32_t bm = (items + 7) >> 3;
-        int32_t hdr = 16 + (has_rle ? 16 : 0) + (has_delta ? 12 : 0)
-                    + (has_rle ? bm : 0) + csize + (has_delta ? bm : 0) + dsize;
-        /* when only one of rle/delta present the other bitmap is absent;
-         * when neither, plain dense */
-        int32_t datum_off = (hdr + 7) & ~7;
-        int32_t content = datum_off + phys * width;
-        int64_t blocklen = (24 + content + 7) & ~7LL;
-        if (off + blocklen > outcap)
-        { free(pvals); free(extra); free(isdelta); free(dbytes); return -1; }
-        uint8_t *blk = out + off;
-        memset(blk, 0, blocklen);
-
-        uint32_t kind = (logical <= 16383) ? 1u : 3u;
-        uint32_t b03 = (kind << 28) | (1u << 27) | (1u << 24);
-        uint32_t b47 = 0;
-        if (kind == 1)
-        {
-            b03 |= (0x00FFFC00u & ((uint32_t) logical << 10)) |
-                   (((uint32_t) content >> 11) & 0x3FFu);
-            b47 = (((uint32_t) content & 0x7FFu) << 21);
-        }
-        else
-        {
-            b03 |= ((uint32_t) content & 0x1FFFFFu);
-            b47 = (uint32_t) logical & 0x3FFFFFFFu;
-        }
-        put_u32le(blk, b03);
-        put_u32le(blk + 4, b47);
-        int64_t frn = first_rownum + row;
-        memcpy(blk + 16, &frn, 8);
-
-        uint8_t *c = blk + 24;
-        int16_t v16 = 2; memcpy(c, &v16, 2);
-        v16 = (int16_t) ((has_rle ? 2 : 0) | (has_delta ? 4 : 0));
-        memcpy(c + 2, &v16, 2);
-        int32_t v32 = (int32_t) logical; memcpy(c + 4, &v32, 4);
-        v32 = phys; memcpy(c + 8, &v32, 4);
-        v32 = phys * width; memcpy(c + 12, &v32, 4);
-        uint8_t *p = c + 16;
-        if (has_rle)
-        {
-            v32 = 0; memcpy(p, &v32, 4);
-            v32 = items; memcpy(p + 4, &v32, 4);
-            v32 = non; memcpy(p + 8, &v32, 4);
-            v32 = csize; memcpy(p + 12, &v32, 4);
-            p += 16;
-        }
-        if (has_delta)
-        {
-            v32 = items; memcpy(p, &v32, 4);
-            v32 = ndelta; memcpy(p + 4, &v32, 4);
-            v32 = dsize; memcpy(p + 8, &v32, 4);
-            p += 12;
-        }
-        if (has_rle)
-        {
-            uint8_t *bmp = p;
-            uint8_t *cnts = bmp + bm;
-            int32_t w = 0;
-            for (int32_t it = 0; it < items; it++)
-                if (extra[it] > 0)
-                {
-                    bmp[it >> 3] |= (uint8_t) (1u << (it & 7));
-                    w += varint_encode(cnts + w, extra[it]);
-                }
-            p = cnts + csize;
-        }
-        if (has_delta)
-        {
-            uint8_t *dbm = p;
-            for (int32_t it = 0; it < items; it++)
-                if (isdelta[it])
-                    dbm[it >> 3] |= (uint8_t) (1u << (it & 7));
-            memcpy(dbm + bm, dbytes, dsize);
-            p = dbm + bm + dsize;
-        }
-        memcpy(c + datum_off, pvals, (size_t) phys * width);
-
-        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
-        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
-
-        off += blocklen;
-        row += logical;
-    }
-    free(pvals); free(extra); free(isdelta); free(dbytes);
-    return off;
+    return dense_encode(vals, width, nrows, first_rownum, blocksize,
+                        1, 1, out, outcap);
 }
+
 
 
 static int64_t encode_bulk(const void *vals, int width, int64_t nrows,
