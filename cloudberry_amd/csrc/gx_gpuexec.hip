@@ -2830,8 +2830,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipStreamSynchronize(s));
         int want_kw = (kmax < (1ULL << 32)) ? 4 : 8;
         uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
-        /* motion path sizes from the exchanged counts each run */
-        if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width)
+        /* motion path sizes from the exchanged counts each run; qual can
+         * grow within the same pow2 table size, so rescap is checked too */
+        if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width ||
+            qual > q->rescap)
         {
             auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
             fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
