@@ -106,6 +106,10 @@ def _load():
                                          ctypes.POINTER(ctypes.c_int64)]
     lib.gx_decode_column.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                                      ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.gx_decode_column_nullable.restype = ctypes.c_int
+    lib.gx_decode_column_nullable.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
     lib.gx_q1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                           ctypes.POINTER(ctypes.c_double)]
@@ -345,6 +349,17 @@ class Table:
         self.ctx._chk(self.ctx._lib.gx_decode_column(
             self.ctx._h, self._t, col, out.ctypes.data, n, 1 if verify else 0))
         return out
+
+    def decode_column_nullable(self, col, dtype, verify=True):
+        """Decode a (possibly NULL-bearing) block-directory column.
+        Returns (values, validity) — null datums decode as zero."""
+        n = self.nrows
+        out = np.zeros(n, dtype)
+        validity = np.zeros(n, np.uint8)
+        self.ctx._chk(self.ctx._lib.gx_decode_column_nullable(
+            self.ctx._h, self._t, col, out.ctypes.data, validity.ctypes.data,
+            n, 1 if verify else 0))
+        return out, validity
 
     def free(self):
         if self._t:

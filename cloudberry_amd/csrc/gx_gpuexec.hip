@@ -470,13 +470,16 @@ __global__ void k_verify_crc_dir(const uint8_t *stream, const gx_blockref *dir,
     }
 }
 
-/* Dense(±RLE±DELTA) block decode — one THREAD per AO block (blocks decode
- * in parallel across the grid; the per-block walk is inherently serial).
- * Walker mirrors DatumStreamBlockRead_AdvanceDense/…DenseDelta
- * (datumstreamblock.h:1624-1912) and oracle/oracle.c decode_dense_content. */
+/* Dense(±RLE±DELTA±NULL) and Orig(±NULL) block decode — one THREAD per AO
+ * block (blocks decode in parallel across the grid; the per-block walk is
+ * inherently serial).  Walker mirrors DatumStreamBlockRead_AdvanceDense/
+ * …DenseDelta (datumstreamblock.h:1624-1912) and oracle/oracle.c
+ * decode_dense_content_v.  validity: one byte per row (1 = non-null, null
+ * datums decode as zero); nullptr REFUSES null-bearing blocks. */
 template <typename T>
 __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
-                               int64_t nblocks, int64_t nrows, T *out, int *err)
+                               int64_t nblocks, int64_t nrows, T *out,
+                               uint8_t *validity, int *err)
 {
     int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
@@ -489,33 +492,63 @@ __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
         int32_t phys = ((const int32_t *) c)[2];
         int32_t psize = ((const int32_t *) c)[3];
         T *dst = out + dir[b].first_row;
+        uint8_t *vdst = validity ? validity + dir[b].first_row : nullptr;
         if (dir[b].first_row + logical > nrows || logical != dir[b].rows)
         { atomicOr(err, 1); continue; }
         if (version == 0)
         {
-            /* Orig: ndatum at content+4 (int16), datums at +16 */
+            /* Orig: flags at +2, ndatum at +4 (int16), nullsz at +8,
+             * null bitmap at +16, datums at +16+nullsz */
+            int16_t oflags = ((const int16_t *) c)[1];
             int16_t nd = ((const int16_t *) c)[2];
+            int32_t nullsz = ((const int32_t *) c)[2];
             if (nd != logical) { atomicOr(err, 1); continue; }
-            const T *d = (const T *) (c + 16);
-            for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
+            if (!(oflags & 1))
+            {
+                if (nullsz != 0) { atomicOr(err, 1); continue; }
+                const T *d = (const T *) (c + 16);
+                for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
+                if (vdst)
+                    for (int32_t i = 0; i < logical; i++) vdst[i] = 1;
+                continue;
+            }
+            if (!vdst) { atomicOr(err, 1); continue; }
+            {
+                const uint8_t *nbmp = c + 16;
+                const T *d = (const T *) (c + 16 + nullsz);
+                int32_t vi = 0;
+                for (int32_t i = 0; i < logical; i++)
+                {
+                    if ((nbmp[i >> 3] >> (i & 7)) & 1)
+                    { dst[i] = (T) 0; vdst[i] = 0; }
+                    else
+                    { dst[i] = d[vi++]; vdst[i] = 1; }
+                }
+            }
             continue;
         }
-        if ((version != 1 && version != 2) || (flags & 0x1) ||
+        bool has_null = (flags & 0x1) != 0;
+        if ((version != 1 && version != 2) || (has_null && !vdst) ||
             psize != phys * (int32_t) sizeof(T))
         { atomicOr(err, 1); continue; }
         bool rle = (flags & 0x2) != 0, delta = (flags & 0x4) != 0;
-        if (!rle && !delta)
+        if (!rle && !delta && !has_null)
         {
             if (logical != phys) { atomicOr(err, 1); continue; }
             const T *d = (const T *) (c + 16);
             for (int32_t i = 0; i < logical; i++) dst[i] = d[i];
+            if (vdst)
+                for (int32_t i = 0; i < logical; i++) vdst[i] = 1;
             continue;
         }
         const uint8_t *p = c + 16;
         int32_t bmbits = 0, csize = 0, dbmbits = 0, dsize = 0;
+        int32_t nullbits = has_null ? logical : 0;   /* no-RLE: bit per row */
         if (rle)
         {
-            if (((const int32_t *) p)[0] != 0) { atomicOr(err, 1); continue; }
+            int32_t norepeats = ((const int32_t *) p)[0];
+            if (has_null) nullbits = norepeats;
+            else if (norepeats != 0) { atomicOr(err, 1); continue; }
             bmbits = ((const int32_t *) p)[1];
             csize = ((const int32_t *) p)[3];
             p += 16;
@@ -526,18 +559,33 @@ __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
             dsize = ((const int32_t *) p)[2];
             p += 12;
         }
-        const uint8_t *bmp = nullptr, *cnts = nullptr, *dbm = nullptr, *dbs = nullptr;
+        const uint8_t *nbmp = nullptr, *bmp = nullptr, *cnts = nullptr,
+                      *dbm = nullptr, *dbs = nullptr;
+        if (has_null) { nbmp = p; p += (nullbits + 7) >> 3; }
         if (rle) { bmp = p; p += (bmbits + 7) >> 3; cnts = p; p += csize; }
         if (delta) { dbm = p; p += (dbmbits + 7) >> 3; dbs = p; p += dsize; }
         int32_t hdr = (int32_t) (p - c);
         const T *datum = (const T *) (c + ((hdr + 7) & ~7));
 
         int64_t w = 0;
-        int32_t item = 0, phys_idx = 0, coff = 0, doff = 0;
+        int32_t item = 0, phys_idx = 0, coff = 0, doff = 0, npos = 0;
         T cur = (T) 0;
         bool bad = false;
         while (w < logical)
         {
+            if (has_null)
+            {
+                if (npos >= nullbits) { bad = true; break; }
+                int nbit = (nbmp[npos >> 3] >> (npos & 7)) & 1;
+                npos++;
+                if (nbit)
+                {
+                    dst[w] = (T) 0;
+                    vdst[w] = 0;
+                    w++;
+                    continue;
+                }
+            }
             if ((rle && item >= bmbits) || (delta && item >= dbmbits))
             { bad = true; break; }
             int64_t reps = 1;
@@ -569,11 +617,14 @@ __global__ void k_decode_dense(const uint8_t *stream, const gx_blockref *dir,
             }
             if (w + reps > logical) { bad = true; break; }
             for (int64_t r = 0; r < reps; r++) dst[w + r] = cur;
+            if (vdst)
+                for (int64_t r = 0; r < reps; r++) vdst[w + r] = 1;
             w += reps;
             item++;
         }
         if (bad || w != logical || phys_idx != phys ||
-            (rle && coff != csize) || (delta && doff != dsize))
+            (rle && coff != csize) || (delta && doff != dsize) ||
+            (has_null && npos != nullbits))
             atomicOr(err, 1);
     }
 }
@@ -2257,13 +2308,16 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     {
         if (c.m.width == 8)
             hipLaunchKernelGGL(k_decode_dense<int64_t>, dim3(GRID), dim3(64), 0, ctx->stream,
-                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int64_t *) dout, derr);
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int64_t *) dout,
+                               (uint8_t *) nullptr, derr);
         else if (c.m.width == 4)
             hipLaunchKernelGGL(k_decode_dense<int32_t>, dim3(GRID), dim3(64), 0, ctx->stream,
-                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int32_t *) dout, derr);
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int32_t *) dout,
+                               (uint8_t *) nullptr, derr);
         else
             hipLaunchKernelGGL(k_decode_dense<int8_t>, dim3(GRID), dim3(64), 0, ctx->stream,
-                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int8_t *) dout, derr);
+                               c.dstream, c.ddir, c.nblocks, c.m.nrows, (int8_t *) dout,
+                               (uint8_t *) nullptr, derr);
         if (verify_checksums)
             hipLaunchKernelGGL(k_verify_crc_dir, dim3(GRID), dim3(64), 0, ctx->stream,
                                c.dstream, c.ddir, c.nblocks, derr);
@@ -2286,6 +2340,55 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
                            c.m.width, derr);
     int herr = 0;
     HIP_CHK(ctx, hipMemcpyAsync(host_out, dout, c.m.nrows * (int64_t) c.m.width,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
+    if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    return GX_OK;
+}
+
+/* NULL-bearing column decode (aocs_getnext with a null bitmap,
+ * datumstreamblock.h:1624-1912 null walk): host_validity gets one byte per
+ * row (1 = non-null; null datums decode as zero).  Requires a block-
+ * directory column (format 1 — variable geometry). */
+extern "C" gx_status gx_decode_column_nullable(gx_ctx *ctx, const gx_table *t,
+                                               int colidx, void *host_out,
+                                               uint8_t *host_validity,
+                                               int64_t cap_rows,
+                                               int verify_checksums)
+{
+    if (!ctx || !t || colidx < 0 || colidx >= (int) t->cols.size()) return GX_ERR_INVALID;
+    const gx_col &c = t->cols[colidx];
+    if (c.m.nrows > cap_rows) return GX_ERR_INVALID;
+    if (c.format != 1)
+    { set_err(ctx, "nullable decode requires a block-directory column%s", ""); return GX_ERR_INVALID; }
+    devbuf dout_b, dval_b, derr_b;
+    HIP_CHK(ctx, dout_b.alloc(c.m.nrows * (int64_t) c.m.width));
+    HIP_CHK(ctx, dval_b.alloc(c.m.nrows));
+    HIP_CHK(ctx, derr_b.alloc(4));
+    int *derr = derr_b.as<int>();
+    HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, ctx->stream));
+    if (c.m.width == 8)
+        hipLaunchKernelGGL(k_decode_dense<int64_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, c.ddir, c.nblocks, c.m.nrows,
+                           dout_b.as<int64_t>(), dval_b.as<uint8_t>(), derr);
+    else if (c.m.width == 4)
+        hipLaunchKernelGGL(k_decode_dense<int32_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, c.ddir, c.nblocks, c.m.nrows,
+                           dout_b.as<int32_t>(), dval_b.as<uint8_t>(), derr);
+    else
+        hipLaunchKernelGGL(k_decode_dense<int8_t>, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, c.ddir, c.nblocks, c.m.nrows,
+                           dout_b.as<int8_t>(), dval_b.as<uint8_t>(), derr);
+    if (verify_checksums)
+        hipLaunchKernelGGL(k_verify_crc_dir, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, c.ddir, c.nblocks, derr);
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, dout_b.p, c.m.nrows * (int64_t) c.m.width,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipMemcpyAsync(host_validity, dval_b.p, c.m.nrows,
                                 hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));

@@ -120,6 +120,12 @@ gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int col,
                            void *host_out, int64_t cap_rows,
                            int verify_checksums);
 
+/* NULL-bearing columns (block-directory format only): validity gets one
+ * byte per row, 1 = non-null; null datums decode as zero. */
+gx_status gx_decode_column_nullable(gx_ctx *ctx, const gx_table *t, int col,
+                                    void *host_out, uint8_t *host_validity,
+                                    int64_t cap_rows, int verify_checksums);
+
 /* TPC-H Q1 core (BASELINE config 4): GROUP BY returnflag,linestatus with
  * COUNT/SUM over a GX_TPCH_LINEITEM_Q1 table; AVG = sum/count (float8_avg) */
 gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,

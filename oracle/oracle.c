@@ -350,6 +350,9 @@ void orc_free(void *p) { free(p); }
 
 static int64_t decode_dense_content(const uint8_t *c, int width,
                                     uint8_t *dst, int64_t cap_rows);
+static int64_t decode_dense_content_v(const uint8_t *c, int width,
+                                      uint8_t *dst, uint8_t *validity,
+                                      int64_t cap_rows);
 
 int32_t orc_aocs_rows_per_block(int width, int32_t blocksize)
 {
@@ -441,6 +444,18 @@ int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
                           void *out_vals, int64_t cap, int verify_checksums,
                           int codec)
 {
+    return orc_aocs_decode_nullable(stream, nbytes, width, out_vals, NULL,
+                                    cap, verify_checksums, codec);
+}
+
+/* like orc_aocs_decode_c but handles NULL-bearing blocks: out_validity gets
+ * one byte per row (1 = non-null), null datums decode as zero.  With
+ * out_validity == NULL any null-bearing block fails loudly (-1). */
+int64_t orc_aocs_decode_nullable(const uint8_t *stream, int64_t nbytes,
+                                 int width, void *out_vals,
+                                 uint8_t *out_validity, int64_t cap,
+                                 int verify_checksums, int codec)
+{
     int64_t off = 0, row = 0;
     uint8_t *dst = (uint8_t *) out_vals;
     while (off + 24 <= nbytes)
@@ -500,21 +515,54 @@ int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
         memcpy(&version, content, 2);
         if (version == 0)                       /* Original */
         {
-            int16_t ndatum;
-            int32_t sz;
+            int16_t oflags, ndatum;
+            int32_t nullsz, sz;
+            memcpy(&oflags, content + 2, 2);
             memcpy(&ndatum, content + 4, 2);
+            memcpy(&nullsz, content + 8, 4);
             memcpy(&sz, content + 12, 4);
-            if ((uint32_t) ndatum != rows || sz != (int32_t) rows * width)
-                return -1;
-            if (row + rows > cap) return -1;
-            memcpy(dst + row * (int64_t) width, content + 16, sz);
+            if ((uint32_t) ndatum != rows || row + rows > cap) return -1;
+            if (!(oflags & 1))
+            {
+                if (sz != (int32_t) rows * width || nullsz != 0) return -1;
+                memcpy(dst + row * (int64_t) width, content + 16, sz);
+                if (out_validity)
+                    memset(out_validity + row, 1, rows);
+            }
+            else
+            {
+                /* null bitmap (1 bit/row, ON = null) at 16, datums at
+                 * 16 + MAXALIGNed bitmap (datumstreamblock.c:3694-3733) */
+                const uint8_t *nbmp = content + 16;
+                const uint8_t *vals = content + 16 + nullsz;
+                int32_t vi = 0;
+                if (out_validity == NULL) return -1;
+                if (nullsz != (int32_t) ((((rows + 7) >> 3) + 7) & ~7))
+                    return -1;
+                for (uint32_t r = 0; r < rows; r++)
+                {
+                    if ((nbmp[r >> 3] >> (r & 7)) & 1)
+                    {
+                        memset(dst + (row + r) * (int64_t) width, 0, width);
+                        out_validity[row + r] = 0;
+                    }
+                    else
+                    {
+                        memcpy(dst + (row + r) * (int64_t) width,
+                               vals + (int64_t) vi * width, width);
+                        out_validity[row + r] = 1;
+                        vi++;
+                    }
+                }
+                if (vi * width != sz) return -1;
+            }
             row += rows;
         }
         else if (version == 1 || version == 2)  /* Dense / Dense_Enhanced */
         {
-            int64_t got = decode_dense_content(content, width,
-                                               dst + row * (int64_t) width,
-                                               cap - row);
+            int64_t got = decode_dense_content_v(
+                content, width, dst + row * (int64_t) width,
+                out_validity ? out_validity + row : NULL, cap - row);
             if (got < 0 || (uint32_t) got != rows) return -1;
             row += got;
         }
@@ -679,6 +727,10 @@ typedef struct {
     int32_t nth, phys;                   /* logical rows / physical datums */
     uint8_t *datum_buffer;
     int64_t datum_used;
+    /* NULL state */
+    int has_null;
+    int32_t always;                      /* always_null_bitmap_count */
+    dbm_t nbm;                           /* null bitmap */
     /* RLE_TYPE state */
     int rle_has, last_valid, last_repeated;
     uint64_t last_item;
@@ -697,6 +749,8 @@ typedef struct {
 static void dwr_getready(dwr_t *w)       /* GetReady, :3588-3669 */
 {
     w->nth = 0; w->phys = 0; w->datum_used = 0;
+    w->has_null = 0; w->always = 0;
+    dbm_reset(&w->nbm);
     w->rle_has = 0; w->last_valid = 0; w->last_repeated = 0; w->last_item = 0;
     dbm_reset(&w->cbm);
     w->nrepeats = 0; w->repeats_size = 0;
@@ -705,13 +759,16 @@ static void dwr_getready(dwr_t *w)       /* GetReady, :3588-3669 */
     w->ndeltas = 0; w->deltas_size = 0;
 }
 
-/* DenseRleSpace, non-null path (:1944-1990) */
-static void dwr_rle_space(const dwr_t *w, int32_t *hdr, int32_t *rle)
+/* DenseRleSpace (:1944-1990); for_null uses the CURRENT compress-bitmap
+ * size (nulls add no compress bits) */
+static void dwr_rle_space(const dwr_t *w, int for_null,
+                          int32_t *hdr, int32_t *rle)
 {
     if (!w->rle_has)
         return;
     *hdr += 16;                                  /* Rle_Extension */
-    *rle += (w->cbm.bits + 1 + 7) >> 3;          /* NextSize */
+    *rle += for_null ? (w->cbm.bits + 7) >> 3
+                     : (w->cbm.bits + 1 + 7) >> 3;   /* Size vs NextSize */
     *rle += w->repeats_size;
     if (w->last_repeated)                        /* pending finalize reservation */
         *rle += varint_size(w->repeats[w->nrepeats - 1]);
@@ -720,20 +777,22 @@ static void dwr_rle_space(const dwr_t *w, int32_t *hdr, int32_t *rle)
 static int dwr_has_space_item(const dwr_t *w, int32_t sz)   /* :2330-2378 */
 {
     int32_t hdr = 16, rle = 0, delta = 0;
+    int32_t nul = w->has_null ? (w->always + 1 + 7) >> 3 : 0;
     if (w->nth + 1 >= DWR_MAXDATUM)
         return 0;
-    dwr_rle_space(w, &hdr, &rle);
+    dwr_rle_space(w, 0, &hdr, &rle);
     if (w->delta_has)
     {
         hdr += 12;                               /* Delta_Extension */
         delta = ((w->dbm.bits + 1 + 7) >> 3) + w->deltas_size;
     }
-    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used + sz <= w->maxdata;
+    return DWR_MAXALIGN(hdr + nul + rle + delta) + w->datum_used + sz <= w->maxdata;
 }
 
 static int dwr_has_space_repeat(const dwr_t *w, int new_repeat)  /* :2098-2220 */
 {
     int32_t hdr = 16, delta = 0, rle, total;
+    int32_t nul = w->has_null ? (w->always + 7) >> 3 : 0;   /* no new bit */
     if (w->nth + 1 >= DWR_MAXDATUM)
         return 0;
     total = w->phys;
@@ -746,19 +805,35 @@ static int dwr_has_space_repeat(const dwr_t *w, int new_repeat)  /* :2098-2220 *
     hdr += 16;                                   /* Rle_Extension, unconditional */
     rle = (total + (new_repeat ? 1 : 0) + 7) >> 3;
     rle += w->repeats_size + 4;                  /* + Int32Compress_MaxByteLen */
-    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used <= w->maxdata;
+    return DWR_MAXALIGN(hdr + nul + rle + delta) + w->datum_used <= w->maxdata;
 }
 
 static int dwr_has_space_delta(const dwr_t *w)   /* :2224-2328 */
 {
     int32_t hdr = 16, rle = 0, total, delta;
+    int32_t nul = w->has_null ? (w->always + 1 + 7) >> 3 : 0;
     if (w->nth + 1 >= DWR_MAXDATUM)
         return 0;
-    dwr_rle_space(w, &hdr, &rle);
+    dwr_rle_space(w, 0, &hdr, &rle);
     total = w->phys + (w->delta_has ? w->dbm.on : 0);
     hdr += 12;
     delta = ((total + 1 + 7) >> 3) + w->deltas_size + 4;  /* + Reserved3_MaxByteLen */
-    return DWR_MAXALIGN(hdr + rle + delta) + w->datum_used <= w->maxdata;
+    return DWR_MAXALIGN(hdr + nul + rle + delta) + w->datum_used <= w->maxdata;
+}
+
+static int dwr_has_space_null(const dwr_t *w)    /* :1992-2090 */
+{
+    int32_t hdr = 16, rle = 0, delta = 0;
+    int32_t nul = (w->always + 1 + 7) >> 3;      /* unconditional */
+    if (w->nth + 1 >= DWR_MAXDATUM)
+        return 0;
+    dwr_rle_space(w, 1, &hdr, &rle);
+    if (w->delta_has)
+    {
+        hdr += 12;
+        delta = ((w->dbm.bits + 1 + 7) >> 3) + w->deltas_size;  /* NextSize */
+    }
+    return DWR_MAXALIGN(hdr + nul + rle + delta) + w->datum_used <= w->maxdata;
 }
 
 static void dwr_finalize_repeat(dwr_t *w)        /* :2435-2470 */
@@ -824,6 +899,9 @@ static int dwr_perform_delta(dwr_t *w, uint64_t v)
     if (!w->delta_has)
         dbm_zerofill(&w->dbm, w->phys);
     w->delta_has = 1;
+    if (w->has_null)
+        dbm_add(&w->nbm, 0);
+    w->always++;
     if (w->last_repeated)
         dwr_finalize_repeat(w);
     w->last_item = v;
@@ -839,10 +917,31 @@ static int dwr_perform_delta(dwr_t *w, uint64_t v)
     return 0;
 }
 
-/* PutDense, fixed-width non-null (:3341-3541); >0 stored, 0 folded, <0 full */
-static int dwr_put(dwr_t *w, uint64_t v)
+/* PutDense, fixed-width (:3094-3541); >0 stored, 0 folded/null, <0 full */
+static int dwr_put(dwr_t *w, uint64_t v, int isnull)
 {
-    int have_prev = w->rle_want && w->last_valid;
+    int have_prev;
+    if (isnull)
+    {
+        if (!dwr_has_space_null(w))
+            return -1;
+        if (!w->has_null)                /* MakeNullBitMapSpace first-null */
+        {
+            w->has_null = 1;
+            dbm_zerofill(&w->nbm, w->always);
+        }
+        dbm_add(&w->nbm, 1);             /* DenseIncrNull */
+        w->always++;
+        if (w->rle_want)
+        {
+            if (w->last_repeated)
+                dwr_finalize_repeat(w);
+            w->last_valid = 0;
+        }
+        w->nth++;
+        return 0;
+    }
+    have_prev = w->rle_want && w->last_valid;
     if (w->last_repeated && w->repeats[w->nrepeats - 1] >= DWR_MAXREPEAT)
         dwr_finalize_repeat(w);
     else if (have_prev)
@@ -872,6 +971,9 @@ static int dwr_put(dwr_t *w, uint64_t v)
     memcpy(w->datum_buffer + w->datum_used, &v, w->width);
     w->datum_used += w->width;
     /* DenseIncrItem (:2506-2556) */
+    if (w->has_null)
+        dbm_add(&w->nbm, 0);
+    w->always++;
     if (w->last_repeated)
         dwr_finalize_repeat(w);
     if (w->rle_want)
@@ -909,7 +1011,7 @@ static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
     hdr = 16 + (w->rle_has ? 16 : 0) + (w->delta_has ? 12 : 0);
     rle = w->rle_has ? DBM_SIZE(&w->cbm) + w->repeats_size : 0;
     delta = w->delta_has ? DBM_SIZE(&w->dbm) + w->deltas_size : 0;
-    meta = hdr + rle + delta;
+    meta = hdr + (w->has_null ? DBM_SIZE(&w->nbm) : 0) + rle + delta;
     aligned = DWR_MAXALIGN(meta);
     content = aligned + w->datum_used;
     blocklen = (24 + content + 7) & ~7LL;
@@ -937,7 +1039,8 @@ static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
 
     c = blk + 24;
     v16 = 2;            memcpy(c, &v16, 2);      /* Dense_Enhanced */
-    v16 = (int16_t) ((w->rle_has ? 2 : 0) | (w->delta_has ? 4 : 0));
+    v16 = (int16_t) ((w->has_null ? 1 : 0) | (w->rle_has ? 2 : 0) |
+                     (w->delta_has ? 4 : 0));
     memcpy(c + 2, &v16, 2);
     v32 = w->nth;       memcpy(c + 4, &v32, 4);
     v32 = w->phys;      memcpy(c + 8, &v32, 4);
@@ -945,7 +1048,8 @@ static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
     p = c + 16;
     if (w->rle_has)
     {
-        v32 = 0;               memcpy(p, &v32, 4);      /* no nulls */
+        v32 = w->has_null ? w->nbm.bits : 0;            /* norepeats count */
+        memcpy(p, &v32, 4);
         v32 = w->cbm.bits;     memcpy(p + 4, &v32, 4);
         v32 = w->nrepeats;     memcpy(p + 8, &v32, 4);
         v32 = w->repeats_size; memcpy(p + 12, &v32, 4);
@@ -957,6 +1061,11 @@ static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
         v32 = w->ndeltas;     memcpy(p + 4, &v32, 4);
         v32 = w->deltas_size; memcpy(p + 8, &v32, 4);
         p += 12;
+    }
+    if (w->has_null)
+    {
+        memcpy(p, w->nbm.buf, DBM_SIZE(&w->nbm));
+        p += DBM_SIZE(&w->nbm);
     }
     if (w->rle_has)
     {
@@ -979,7 +1088,8 @@ static int64_t dwr_block(dwr_t *w, int64_t first_rownum,
     return blocklen;
 }
 
-static int64_t dense_encode(const void *vals, int width, int64_t nrows,
+static int64_t dense_encode(const void *vals, const uint8_t *nulls,
+                            int width, int64_t nrows,
                             int64_t first_rownum, int32_t blocksize,
                             int rle_want, int delta_want,
                             uint8_t *out, int64_t outcap)
@@ -1002,17 +1112,19 @@ static int64_t dense_encode(const void *vals, int width, int64_t nrows,
     w.datum_buffer = malloc((size_t) w.maxdata + 16);
     w.cbm.buf = malloc(scratch);
     w.dbm.buf = malloc(scratch);
+    w.nbm.buf = malloc(scratch);
     w.repeats = malloc(scratch * sizeof(int32_t));
     w.deltas = malloc(scratch * sizeof(int64_t));
     w.dsigns = malloc(scratch);
-    if (!w.datum_buffer || !w.cbm.buf || !w.dbm.buf ||
+    if (!w.datum_buffer || !w.cbm.buf || !w.dbm.buf || !w.nbm.buf ||
         !w.repeats || !w.deltas || !w.dsigns)
         goto fail;
     dwr_getready(&w);
     for (int64_t i = 0; i < nrows; i++)
     {
-        uint64_t v = item_at(src, width, i);
-        if (dwr_put(&w, v) < 0)
+        int isnull = nulls != NULL && nulls[i] != 0;
+        uint64_t v = isnull ? 0 : item_at(src, width, i);
+        if (dwr_put(&w, v, isnull) < 0)
         {
             int64_t bl = dwr_block(&w, first_rownum + emitted,
                                    out + off, outcap - off);
@@ -1021,7 +1133,7 @@ static int64_t dense_encode(const void *vals, int width, int64_t nrows,
             off += bl;
             emitted += w.nth;
             dwr_getready(&w);
-            if (dwr_put(&w, v) < 0)
+            if (dwr_put(&w, v, isnull) < 0)
                 goto fail;
         }
     }
@@ -1033,11 +1145,11 @@ static int64_t dense_encode(const void *vals, int width, int64_t nrows,
             goto fail;
         off += bl;
     }
-    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf);
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf); free(w.nbm.buf);
     free(w.repeats); free(w.deltas); free(w.dsigns);
     return off;
 fail:
-    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf);
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf); free(w.nbm.buf);
     free(w.repeats); free(w.deltas); free(w.dsigns);
     return -1;
 }
@@ -1046,7 +1158,7 @@ int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
                             int64_t first_rownum, int32_t blocksize,
                             uint8_t *out, int64_t outcap)
 {
-    return dense_encode(vals, width, nrows, first_rownum, blocksize,
+    return dense_encode(vals, NULL, width, nrows, first_rownum, blocksize,
                         1, 0, out, outcap);
 }
 
@@ -1054,8 +1166,101 @@ int64_t orc_aocs_encode_rle_delta(const void *vals, int width, int64_t nrows,
                                   int64_t first_rownum, int32_t blocksize,
                                   uint8_t *out, int64_t outcap)
 {
-    return dense_encode(vals, width, nrows, first_rownum, blocksize,
+    return dense_encode(vals, NULL, width, nrows, first_rownum, blocksize,
                         1, 1, out, outcap);
+}
+
+int64_t orc_aocs_encode_rle_delta_nulls(const void *vals, const uint8_t *nulls,
+                                        int width, int64_t nrows,
+                                        int64_t first_rownum, int32_t blocksize,
+                                        int delta,
+                                        uint8_t *out, int64_t outcap)
+{
+    return dense_encode(vals, nulls, width, nrows, first_rownum, blocksize,
+                        1, delta, out, outcap);
+}
+
+/* Original-version writer with NULL support — PutOrig (:1569-1770) +
+ * BlockOrig (:3669-3801): header {version,flags,ndatum,enc,nullsz,sz},
+ * null bitmap (1 bit/row, ON = null) MAXALIGNed, then packed datums.
+ * Capacity: OrigHasSpace (:1508-1566), STRICT '<' and
+ * nullSize = MAXALIGN(Size(always+1)) once any null exists. */
+int64_t orc_aocs_encode_orig_nulls(const void *vals, const uint8_t *nulls,
+                                   int width, int64_t nrows,
+                                   int64_t first_rownum, int32_t blocksize,
+                                   uint8_t *out, int64_t outcap)
+{
+    const uint8_t *src = (const uint8_t *) vals;
+    int32_t maxdata = blocksize - 24;
+    int64_t off = 0, row = 0;
+    int32_t cap_rows = blocksize;                /* nth < 16383 anyway */
+    uint8_t *dvals = malloc((size_t) maxdata + 16);
+    uint8_t *nbm = malloc((size_t) ((cap_rows + 7) >> 3) + 8);
+    if (!dvals || !nbm) { free(dvals); free(nbm); return -1; }
+
+    while (row < nrows)
+    {
+        int32_t nth = 0, phys = 0, always = 0;
+        int has_null = 0;
+        int64_t used = 0;
+        memset(nbm, 0, (size_t) ((cap_rows + 7) >> 3) + 8);
+        while (row + nth < nrows && nth + 1 < 16383)
+        {
+            int isnull = nulls != NULL && nulls[row + nth] != 0;
+            int32_t nullsize =
+                (isnull || has_null)
+                    ? (int32_t) ((((always + 1 + 7) >> 3) + 7) & ~7) : 0;
+            int32_t sz = isnull ? 0 : width;
+            if (!(16 + nullsize + used + sz < maxdata))
+                break;
+            if (isnull)
+            {
+                has_null = 1;
+                nbm[always >> 3] |= (uint8_t) (1u << (always & 7));
+            }
+            else
+            {
+                memcpy(dvals + used, src + (row + nth) * width, width);
+                used += width;
+                phys++;
+            }
+            always++;
+            nth++;
+        }
+        if (nth == 0) { free(dvals); free(nbm); return -1; }
+
+        int32_t nullsz = has_null
+            ? (int32_t) ((((nth + 7) >> 3) + 7) & ~7) : 0;
+        int32_t content = 16 + nullsz + (int32_t) used;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        if (off + blocklen > outcap) { free(dvals); free(nbm); return -1; }
+        uint8_t *blk = out + off;
+        memset(blk, 0, (size_t) blocklen);
+        uint32_t b03 = (1u << 28) | (1u << 27) | (1u << 24) |
+                       (0x00FFFC00u & ((uint32_t) nth << 10)) |
+                       (((uint32_t) content >> 11) & 0x3FFu);
+        uint32_t b47 = (((uint32_t) content & 0x7FFu) << 21);
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+        uint8_t *c = blk + 24;
+        int16_t v16 = 0;  memcpy(c, &v16, 2);
+        v16 = has_null ? 1 : 0; memcpy(c + 2, &v16, 2);
+        v16 = (int16_t) nth; memcpy(c + 4, &v16, 2);
+        v16 = 0; memcpy(c + 6, &v16, 2);
+        int32_t v32 = nullsz; memcpy(c + 8, &v32, 4);
+        v32 = (int32_t) used; memcpy(c + 12, &v32, 4);
+        if (has_null)
+            memcpy(c + 16, nbm, (size_t) ((nth + 7) >> 3));
+        memcpy(c + 16 + nullsz, dvals, (size_t) used);
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+        off += blocklen;
+        row += nth;
+    }
+    free(dvals); free(nbm);
+    return off;
 }
 
 
@@ -1145,8 +1350,14 @@ int64_t orc_aocs_encode_zstd(const void *vals, int width, int64_t nrows,
  * (datumstreamblock.h:1624-1912): per NEW item advance the compress bitmap
  * (repeat count varint when ON) and the delta bitmap (signed-magnitude
  * varint applied to the running value when ON; physical datum otherwise). */
-static int64_t decode_dense_content(const uint8_t *c, int width,
-                                    uint8_t *dst, int64_t cap_rows)
+/* Dense/Dense_Enhanced content walker, full feature set: RLE_TYPE,
+ * DELTA_RANGE and the NULL bitmap (one bit per NON-REPEAT logical slot,
+ * ON = null; reader walk datumstreamblock.h:1754-1912, section order
+ * datumstreamblock.c:3963-4040).  validity gets one byte per row
+ * (1 = non-null); NULL validity refuses null-bearing blocks. */
+static int64_t decode_dense_content_v(const uint8_t *c, int width,
+                                      uint8_t *dst, uint8_t *validity,
+                                      int64_t cap_rows)
 {
     int16_t version, flags;
     int32_t logical, phys, psize;
@@ -1156,25 +1367,31 @@ static int64_t decode_dense_content(const uint8_t *c, int width,
     memcpy(&phys, c + 8, 4);
     memcpy(&psize, c + 12, 4);
     if (psize != phys * width || logical > cap_rows) return -1;
-    if (flags & 0x1) return -1;                /* null bitmap unsupported here */
+    int has_null = (flags & 0x1) != 0;
     int rle = (flags & 0x2) != 0, delta = (flags & 0x4) != 0;
-    if (!rle && !delta)
+    if (has_null && validity == NULL) return -1;
+    if (!rle && !delta && !has_null)
     {
         if (logical != phys) return -1;
         memcpy(dst, c + 16, (size_t) psize);
+        if (validity) memset(validity, 1, (size_t) logical);
         return logical;
     }
     const uint8_t *p = c + 16;
     int32_t bmbits = 0, ncnt = 0, csize = 0;
     int32_t dbmbits = 0, ndelta = 0, dsize = 0;
+    int32_t nullbits = has_null ? logical : 0;   /* no-RLE: 1 bit per row */
     if (rle)
     {
-        int32_t nullbits;
-        memcpy(&nullbits, p, 4);
+        int32_t norepeats;
+        memcpy(&norepeats, p, 4);
         memcpy(&bmbits, p + 4, 4);
         memcpy(&ncnt, p + 8, 4);
         memcpy(&csize, p + 12, 4);
-        if (nullbits != 0) return -1;
+        if (has_null)
+            nullbits = norepeats;
+        else if (norepeats != 0)
+            return -1;
         p += 16;
     }
     if (delta)
@@ -1184,7 +1401,12 @@ static int64_t decode_dense_content(const uint8_t *c, int width,
         memcpy(&dsize, p + 8, 4);
         p += 12;
     }
-    const uint8_t *bmp = NULL, *cnts = NULL, *dbm = NULL, *dbs = NULL;
+    const uint8_t *nbmp = NULL, *bmp = NULL, *cnts = NULL, *dbm = NULL, *dbs = NULL;
+    if (has_null)
+    {
+        nbmp = p;
+        p += (nullbits + 7) >> 3;
+    }
     if (rle)
     {
         bmp = p; p += (bmbits + 7) >> 3;
@@ -1199,10 +1421,23 @@ static int64_t decode_dense_content(const uint8_t *c, int width,
     const uint8_t *datum = c + ((hdr + 7) & ~7);
 
     int64_t w = 0;
-    int32_t item = 0, phys_idx = 0, coff = 0, doff = 0, dseen = 0;
+    int32_t item = 0, phys_idx = 0, coff = 0, doff = 0, dseen = 0, npos = 0;
     uint64_t cur = 0;
     while (w < logical)
     {
+        if (has_null)
+        {
+            if (npos >= nullbits) return -1;
+            int nbit = (nbmp[npos >> 3] >> (npos & 7)) & 1;
+            npos++;
+            if (nbit)
+            {
+                memset(dst + w * width, 0, (size_t) width);
+                validity[w] = 0;
+                w++;
+                continue;
+            }
+        }
         if ((rle && item >= bmbits) || (delta && item >= dbmbits)) return -1;
         int64_t reps = 1;
         if (rle && (bmp[item >> 3] & (1u << (item & 7))))
@@ -1232,13 +1467,22 @@ static int64_t decode_dense_content(const uint8_t *c, int width,
         if (w + reps > logical) return -1;
         for (int64_t r = 0; r < reps; r++)
             memcpy(dst + (w + r) * width, &cur, width);
+        if (validity)
+            memset(validity + w, 1, (size_t) reps);
         w += reps;
         item++;
     }
     if (phys_idx != phys || (rle && (coff != csize || item != bmbits)) ||
-        (delta && (doff != dsize || dseen != ndelta || item != dbmbits)))
+        (delta && (doff != dsize || dseen != ndelta || item != dbmbits)) ||
+        (has_null && npos != nullbits))
         return -1;
     return w;
+}
+
+static int64_t decode_dense_content(const uint8_t *c, int width,
+                                    uint8_t *dst, int64_t cap_rows)
+{
+    return decode_dense_content_v(c, width, dst, NULL, cap_rows);
 }
 
 /* ======================================================================

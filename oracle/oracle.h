@@ -84,9 +84,9 @@ int64_t orc_aocs_decode(const uint8_t *stream, int64_t nbytes, int width,
 /* rows per full block for a fixed-width NOT NULL column */
 int32_t orc_aocs_rows_per_block(int width, int32_t blocksize);
 
-/* RLE_TYPE (Dense_Enhanced, no-null no-delta subset): encoder emits
- * conforming blocks (run-length compress bitmap + varint repeat counts);
- * orc_aocs_decode handles Orig AND Dense(±RLE) blocks transparently. */
+/* RLE_TYPE (Dense_Enhanced): exact restatement of the reference writer's
+ * state machine, byte-exact with the compiled writer itself;
+ * orc_aocs_decode handles Orig AND Dense(±RLE±DELTA) blocks transparently. */
 int64_t orc_aocs_encode_rle(const void *vals, int width, int64_t nrows,
                             int64_t first_rownum, int32_t blocksize,
                             uint8_t *out, int64_t outcap);
@@ -114,6 +114,23 @@ int64_t orc_aocs_encode_zstd(const void *vals, int width, int64_t nrows,
 int64_t orc_aocs_decode_c(const uint8_t *stream, int64_t nbytes, int width,
                           void *out_vals, int64_t cap, int verify_checksums,
                           int codec);
+/* NULL-bearing columns: encoders take a per-row null flag array (NULL =
+ * no nulls; both byte-exact vs the reference writer), the decoder fills
+ * one validity byte per row (1 = non-null; null datums decode as zero).
+ * orc_aocs_decode/_c refuse null-bearing blocks (-1). */
+int64_t orc_aocs_encode_orig_nulls(const void *vals, const uint8_t *nulls,
+                                   int width, int64_t nrows,
+                                   int64_t first_rownum, int32_t blocksize,
+                                   uint8_t *out, int64_t outcap);
+int64_t orc_aocs_encode_rle_delta_nulls(const void *vals, const uint8_t *nulls,
+                                        int width, int64_t nrows,
+                                        int64_t first_rownum, int32_t blocksize,
+                                        int delta,
+                                        uint8_t *out, int64_t outcap);
+int64_t orc_aocs_decode_nullable(const uint8_t *stream, int64_t nbytes,
+                                 int width, void *out_vals,
+                                 uint8_t *out_validity, int64_t cap,
+                                 int verify_checksums, int codec);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

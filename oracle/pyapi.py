@@ -94,6 +94,19 @@ def _load():
                                         ctypes.c_void_p, ctypes.c_int64]
     lib.orc_aocs_encode_rle_delta.restype = ctypes.c_int64
     lib.orc_aocs_encode_rle_delta.argtypes = lib.orc_aocs_encode_rle.argtypes
+    lib.orc_aocs_encode_rle_delta_nulls.restype = ctypes.c_int64
+    lib.orc_aocs_encode_rle_delta_nulls.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+        ctypes.c_int64, ctypes.c_int32, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_encode_orig_nulls.restype = ctypes.c_int64
+    lib.orc_aocs_encode_orig_nulls.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+        ctypes.c_int64, ctypes.c_int32, ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_decode_nullable.restype = ctypes.c_int64
+    lib.orc_aocs_decode_nullable.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_int]
     lib.orc_aocs_encode_zstd.restype = ctypes.c_int64
     lib.orc_aocs_encode_zstd.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                          ctypes.c_int64, ctypes.c_int64,
@@ -165,13 +178,21 @@ def ref_writer():
                                       ctypes.c_int, ctypes.c_int, ctypes.c_int,
                                       ctypes.c_int32, ctypes.c_void_p, ctypes.c_int64,
                                       ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
+        _refw.refw_encode_nulls.restype = ctypes.c_int
+        _refw.refw_encode_nulls.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int64,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int32, ctypes.c_void_p, ctypes.c_int64,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
     return _refw
 
 
-def ref_writer_stream(vals, version=2, rle=1, delta=1, blocksize=32768):
+def ref_writer_stream(vals, version=2, rle=1, delta=1, blocksize=32768,
+                      nulls=None):
     """Run the reference writer and wrap its content blocks in our AO
     envelope (SmallContent/NonBulkDense + CRC32C pair + firstRowNum) —
-    the stream a real segment file would hold.  Returns bytes or None."""
+    the stream a real segment file would hold.  Returns bytes or None.
+    nulls: optional per-row bool/uint8 null flags."""
     w = ref_writer()
     if w is None:
         return None
@@ -184,9 +205,15 @@ def ref_writer_stream(vals, version=2, rle=1, delta=1, blocksize=32768):
     # header reserve: AoHeader_Size(isLong, checksum=true, firstRowNum=true)
     # = 24 for Orig (regular) streams, 32 for Dense (datumstream.c:588-605)
     reserve = 24 if version == 0 else 32
-    nb = w.refw_encode(vals.ctypes.data, width, len(vals), version, rle, delta,
-                       blocksize - reserve, out.ctypes.data, cap,
-                       lens.ctypes.data, rows.ctypes.data, 65536)
+    nptr = 0
+    if nulls is not None:
+        nulls = np.ascontiguousarray(nulls, np.uint8)
+        assert len(nulls) == len(vals)
+        nptr = nulls.ctypes.data
+    nb = w.refw_encode_nulls(vals.ctypes.data, nptr, width, len(vals),
+                             version, rle, delta,
+                             blocksize - reserve, out.ctypes.data, cap,
+                             lens.ctypes.data, rows.ctypes.data, 65536)
     assert nb > 0, nb
     stream = bytearray()
     off = 0
@@ -365,6 +392,46 @@ def aocs_encode_rle_delta(vals):
                                         32768, buf.ctypes.data, cap)
     assert got > 0
     return buf[:got].tobytes()
+
+
+def aocs_encode_rle_delta_nulls(vals, nulls, delta=1):
+    """Dense_Enhanced encode with a NULL bitmap (byte-exact vs reference)."""
+    vals = np.ascontiguousarray(vals)
+    nulls = np.ascontiguousarray(nulls, np.uint8)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_rle_delta_nulls(
+        vals.ctypes.data, nulls.ctypes.data, width, len(vals), 1, 32768,
+        delta, buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
+
+
+def aocs_encode_orig_nulls(vals, nulls):
+    """Original-version encode with a NULL bitmap (byte-exact vs reference)."""
+    vals = np.ascontiguousarray(vals)
+    nulls = np.ascontiguousarray(nulls, np.uint8)
+    width = vals.itemsize
+    cap = len(vals) * width + (1 << 21)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_orig_nulls(
+        vals.ctypes.data, nulls.ctypes.data, width, len(vals), 1, 32768,
+        buf.ctypes.data, cap)
+    assert got > 0
+    return buf[:got].tobytes()
+
+
+def aocs_decode_nullable(stream, width, nrows, dtype, verify=1, codec=1):
+    """Decode any AOCS stream incl. NULL-bearing blocks.
+    Returns (values, validity) — null datums decode as zero."""
+    out = np.zeros(nrows, dtype)
+    validity = np.zeros(nrows, np.uint8)
+    got = lib.orc_aocs_decode_nullable(stream, len(stream), width,
+                                       out.ctypes.data, validity.ctypes.data,
+                                       nrows, verify, codec)
+    assert got == nrows, got
+    return out, validity
 
 
 def aocs_encode_zlib(vals, level=6):

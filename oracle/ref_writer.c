@@ -164,12 +164,13 @@ cb_zero(void *arg)
  * CONTENT of each datum-stream block back-to-back into out.  Returns the
  * number of blocks, or -1.  block_lens/block_rows must hold max_blocks.
  * version: 0 = Original, 2 = Dense_Enhanced.  rle/delta toggle compression.
+ * nulls: optional per-row null flags (NULL = no nulls).
  */
 int
-refw_encode(const void *vals, int width, int64 nrows,
-            int version, int rle, int delta, int32 maxDataBlockSize,
-            uint8 *out, int64 outcap,
-            int32 *block_lens, int32 *block_rows, int max_blocks)
+refw_encode_nulls(const void *vals, const uint8 *nulls, int width, int64 nrows,
+                  int version, int rle, int delta, int32 maxDataBlockSize,
+                  uint8 *out, int64 outcap,
+                  int32 *block_lens, int32 *block_rows, int max_blocks)
 {
     DatumStreamTypeInfo ti;
     DatumStreamBlockWrite dsw;
@@ -203,10 +204,12 @@ refw_encode(const void *vals, int width, int64 nrows,
     for (int64 i = 0; i < nrows; i++)
     {
         Datum d = 0;
+        bool isnull = (nulls != NULL && nulls[i] != 0);
         void *tofree = NULL;
 
-        memcpy(&d, src + i * width, width);
-        if (DatumStreamBlockWrite_Put(&dsw, d, false, &tofree) < 0)
+        if (!isnull)
+            memcpy(&d, src + i * width, width);
+        if (DatumStreamBlockWrite_Put(&dsw, d, isnull, &tofree) < 0)
         {
             int64 len;
 
@@ -219,7 +222,7 @@ refw_encode(const void *vals, int width, int64 nrows,
             off += len;
             rows_in_block = 0;
             DatumStreamBlockWrite_GetReady(&dsw);
-            if (DatumStreamBlockWrite_Put(&dsw, d, false, &tofree) < 0)
+            if (DatumStreamBlockWrite_Put(&dsw, d, isnull, &tofree) < 0)
                 return -1;
         }
         rows_in_block++;
@@ -237,4 +240,16 @@ refw_encode(const void *vals, int width, int64 nrows,
         off += len;
     }
     return nblocks;
+}
+
+/* back-compat entry point: no nulls */
+int
+refw_encode(const void *vals, int width, int64 nrows,
+            int version, int rle, int delta, int32 maxDataBlockSize,
+            uint8 *out, int64 outcap,
+            int32 *block_lens, int32 *block_rows, int max_blocks)
+{
+    return refw_encode_nulls(vals, NULL, width, nrows,
+                             version, rle, delta, maxDataBlockSize,
+                             out, outcap, block_lens, block_rows, max_blocks);
 }

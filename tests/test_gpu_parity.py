@@ -543,3 +543,47 @@ def test_gpu_encoder_byte_exact_vs_oracle(ctx, orc):
                      (2, o["o_orderdate"]), (3, o["o_shippriority"])]:
         assert t.dump_stream(col) == orc.aocs_encode(arr), f"col {col}"
     t.free()
+
+
+@pytest.mark.gpu
+def test_gpu_nullable_decode_parity(ctx, orc):
+    """GPU decode of NULL-bearing blocks (Orig and Dense RLE+DELTA) matches
+    the oracle's nullable decoder bit-for-bit on values AND validity; the
+    oracle streams are themselves byte-exact with the reference writer
+    (test_oracle_cpu.py)."""
+    rng = np.random.default_rng(23)
+    vals = np.repeat(np.arange(1, 5000, dtype=np.int64),
+                     rng.integers(1, 9, 4999))[:30000]
+    nulls = (rng.random(len(vals)) < 0.15).astype(np.uint8)
+    v32 = rng.integers(-3000, 3000, 40000).astype(np.int32)
+    n32 = (rng.random(40000) < 0.5).astype(np.uint8)
+    cases = [
+        (orc.aocs_encode_rle_delta_nulls(vals, nulls), 8, vals, nulls, np.int64),
+        (orc.aocs_encode_orig_nulls(vals, nulls), 8, vals, nulls, np.int64),
+        (orc.aocs_encode_rle_delta_nulls(v32, n32), 4, v32, n32, np.int32),
+    ]
+    for stream, width, v, nl, dt in cases:
+        t = ctx.bind([(stream, width, len(v), 1)])
+        out, valid = t.decode_column_nullable(0, dt, verify=True)
+        np.testing.assert_array_equal(valid, (nl == 0).astype(np.uint8))
+        m = nl == 0
+        np.testing.assert_array_equal(out[m], v[m])
+        assert (out[~m] == 0).all()
+        # oracle cross-check on the same bytes
+        oo, ov = orc.aocs_decode_nullable(stream, width, len(v), dt)
+        np.testing.assert_array_equal(out, oo)
+        np.testing.assert_array_equal(valid, ov)
+        t.free()
+
+
+@pytest.mark.gpu
+def test_gpu_plain_decode_refuses_null_blocks(ctx, orc):
+    """gx_decode_column (no validity) must fail loudly on a NULL-bearing
+    block, never silently return garbage."""
+    vals = np.arange(2000, dtype=np.int64)
+    nulls = np.zeros(2000, np.uint8)
+    nulls[7] = 1
+    t = ctx.bind([(orc.aocs_encode_rle_delta_nulls(vals, nulls), 8, 2000, 1)])
+    with pytest.raises(gx.GxError):
+        t.decode_column(0, np.int64)
+    t.free()
