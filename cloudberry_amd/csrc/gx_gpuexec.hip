@@ -2754,10 +2754,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                            (unsigned long long *) q->cset, q->cmask, q->bloom, q->bwmask);
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
 
-    /* ---- stage 2: orders build (local or via Motions) ---- */
+    /* ---- stage 2: orders build (local or via Motions) ----
+     * GX_FORCE_MOTION=1 routes a single-segment run through the FULL
+     * RCCL exchange branch (self send/recv) — the exact code the
+     * multi-GPU scale bench executes, testable on one GPU. */
     int64_t qual = 0;
     double ms_motion = 0.0;
-    if (ctx->nsegs == 1)
+    if (ctx->nsegs == 1 && env_int("GX_FORCE_MOTION", 0) == 0)
     {
         uint64_t tslots = q->tmask + 1;
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));

@@ -587,3 +587,29 @@ def test_gpu_plain_decode_refuses_null_blocks(ctx, orc):
     with pytest.raises(gx.GxError):
         t.decode_column(0, np.int64)
     t.free()
+
+
+@pytest.mark.gpu
+def test_gpu_forced_motion_path_matches_local(ctx, orc):
+    """GX_FORCE_MOTION=1 routes a 1-segment run through the FULL RCCL
+    exchange branch (count AllGather + grouped self send/recv + table
+    rebuild from received rows) — the exact code the 8-GPU scale bench
+    executes.  Results must equal the local-path run and the oracle."""
+    import os
+    sf = 0.5
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    want = ctx.q3(cust, ordr, li).run().result()
+
+    ctx.comm_init(ctx.comm_unique_id())
+    os.environ["GX_FORCE_MOTION"] = "1"
+    try:
+        got = ctx.q3(cust, ordr, li).run().result()
+    finally:
+        del os.environ["GX_FORCE_MOTION"]
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["o_orderdate"], want["o_orderdate"])
+    np.testing.assert_array_equal(got["o_shippriority"], want["o_shippriority"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
