@@ -938,6 +938,82 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -5)
+    {
+        /* software-pipelined B=1: next iteration's ship+key loads issue
+         * BEFORE the current iteration's table round-trip is consumed —
+         * same perfect coalescing as B=1, one row per lane per step. */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        if (i < lk_m.nrows)
+        {
+            int32_t ship = gx_col_get<int32_t>(sh_s, sh_m, i);
+            uint64_t key = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            while (true)
+            {
+                int64_t nx = i + stride;
+                int32_t ship_n = 0;
+                uint64_t key_n = 0;
+                if (nx < lk_m.nrows)
+                {
+                    ship_n = gx_col_get<int32_t>(sh_s, sh_m, nx);
+                    key_n = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, nx);
+                }
+                if (gx_cmp(fop, ship, flit))
+                {
+                    uint64_t slot = smap.slot0(key);
+                    uint64_t r = resolve(key, slot, tkey[slot]);
+                    if (r != ~0ULL) hit(r, i);
+                }
+                if (nx >= lk_m.nrows) break;
+                i = nx;
+                ship = ship_n;
+                key = key_n;
+            }
+        }
+    }
+    else if constexpr (B == -6)
+    {
+        /* 2-deep pipeline: ship/key AND the first table word for the NEXT
+         * row are all in flight while the CURRENT row is resolved, so the
+         * serial chain ship->key->table never stalls back-to-back.
+         * Filtered lanes prefetch slot 0 (stays L1-resident). */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        if (i < lk_m.nrows)
+        {
+            int32_t ship = gx_col_get<int32_t>(sh_s, sh_m, i);
+            uint64_t key = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            bool pass = gx_cmp(fop, ship, flit);
+            uint64_t slot = pass ? smap.slot0(key) : 0;
+            KT v = tkey[slot];
+            while (true)
+            {
+                int64_t nx = i + stride;
+                int32_t ship_n = 0;
+                uint64_t key_n = 0;
+                if (nx < lk_m.nrows)
+                {
+                    ship_n = gx_col_get<int32_t>(sh_s, sh_m, nx);
+                    key_n = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, nx);
+                }
+                bool pass_n = gx_cmp(fop, ship_n, flit) && nx < lk_m.nrows;
+                uint64_t slot_n = pass_n ? smap.slot0(key_n) : 0;
+                KT v_n = tkey[slot_n];
+                if (pass)
+                {
+                    uint64_t r = resolve(key, slot, v);
+                    if (r != ~0ULL) hit(r, i);
+                }
+                if (nx >= lk_m.nrows) break;
+                i = nx;
+                key = key_n;
+                pass = pass_n;
+                slot = slot_n;
+                v = v_n;
+            }
+        }
+    }
     else if constexpr (B == -2)
     {
         /* non-temporal stream loads: keep L2/L3 for the table */
@@ -1515,6 +1591,66 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
             if (prev == (KT) k) break;
             slot = (slot + 1) & tmask;
         }
+    }
+}
+
+/* two-pass orders build, pass 1 (GX_ORDERS_TWOPASS experiment): the
+ * filter scan is kept HOMOGENEOUS (no table CAS in flight) and emits
+ * qualifying orders compactly with per-BLOCK two-pass compaction (count,
+ * LDS scan, ONE cursor atomic per block, then re-filter + write — the
+ * block's range stays L2-resident between passes).  Pass 2 is the
+ * existing k_build_from_rows insert. */
+template <typename KS>
+__global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
+                               const uint8_t *oc_s, gx_colmeta oc_m,
+                               const uint8_t *od_s, gx_colmeta od_m,
+                               const uint8_t *op_s, gx_colmeta op_m,
+                               int oop, int32_t olit,
+                               const KS *cset, uint64_t cmask,
+                               const unsigned long long *bloom, uint64_t bwmask,
+                               gx_qual_row *outq, unsigned long long *cursor)
+{
+    __shared__ unsigned int scan[256];
+    __shared__ unsigned long long sbase;
+    int64_t range = (ok_m.nrows + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * range;
+    int64_t hi = min(lo + range, ok_m.nrows);
+    if (lo >= hi) return;
+
+    auto keep_row = [&](int64_t i) -> bool {
+        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
+        if (!gx_cmp(oop, od, olit)) return false;
+        uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
+        return d_bloom_test(bloom, bwmask, ck) &&
+               d_set_contains(cset, cmask, ck);
+    };
+
+    unsigned int mine = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        if (keep_row(i)) mine++;
+    scan[threadIdx.x] = mine;
+    __syncthreads();
+    for (int o = 1; o < 256; o <<= 1)
+    {
+        unsigned int v = (threadIdx.x >= (unsigned) o) ? scan[threadIdx.x - o] : 0;
+        __syncthreads();
+        scan[threadIdx.x] += v;
+        __syncthreads();
+    }
+    if (threadIdx.x == blockDim.x - 1)
+        sbase = atomicAdd(cursor, (unsigned long long) scan[255]);
+    __syncthreads();
+    unsigned long long w = sbase + scan[threadIdx.x] - mine;
+
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+    {
+        if (!keep_row(i)) continue;
+        gx_qual_row r;
+        r.okey = gx_col_get<int64_t>(ok_s, ok_m, i);
+        r.odate = gx_col_get<int32_t>(od_s, od_m, i);
+        r.oprio = gx_col_get<int32_t>(op_s, op_m, i);
+        outq[w] = r;
+        w++;
     }
 }
 
@@ -2788,7 +2924,46 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                    q->bloom, q->bwmask,
                                    tk, q->tdate, q->tprio, q->smap);
         };
-        if (q->key_width == 4 && q->cset_width == 4)
+        if (env_int("GX_ORDERS_TWOPASS", 0))
+        {
+            /* experiment: homogeneous filter/emit scan + separate insert */
+            if (!q->m_send2 || q->m_send2_cap < (uint64_t) q->rescap)
+            {
+                if (q->m_send2) hipFree(q->m_send2);
+                q->m_send2 = nullptr; q->m_send2_cap = 0;
+                HIP_CHK(ctx, hipMalloc(&q->m_send2,
+                                       std::max<int64_t>(q->rescap, 1) *
+                                           sizeof(gx_qual_row)));
+                q->m_send2_cap = q->rescap;
+            }
+            HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
+            auto launch_emit = [&](auto *cs) {
+                hipLaunchKernelGGL((k_orders_emitq<std::decay_t<decltype(*cs)>>),
+                                   dim3(ogrid), dim3(TPB), 0, s,
+                                   ok.dstream, ok.m, oc.dstream, oc.m,
+                                   od.dstream, od.m, op.dstream, op.m,
+                                   D.mid_filter.op, (int32_t) D.mid_filter.literal,
+                                   cs, q->cmask, q->bloom, q->bwmask,
+                                   q->m_send2, q->dcount);
+            };
+            if (q->cset_width == 4)
+                launch_emit((const unsigned int *) q->cset);
+            else
+                launch_emit((const unsigned long long *) q->cset);
+            if (q->key_width == 4)
+                hipLaunchKernelGGL(k_build_from_rows<unsigned int>,
+                                   dim3(GRID), dim3(TPB), 0, s,
+                                   q->m_send2, q->rescap,
+                                   (unsigned int *) q->tkey,
+                                   q->tdate, q->tprio, q->smap);
+            else
+                hipLaunchKernelGGL(k_build_from_rows<unsigned long long>,
+                                   dim3(GRID), dim3(TPB), 0, s,
+                                   q->m_send2, q->rescap,
+                                   (unsigned long long *) q->tkey,
+                                   q->tdate, q->tprio, q->smap);
+        }
+        else if (q->key_width == 4 && q->cset_width == 4)
             launch_build((unsigned int *) q->tkey, (const unsigned int *) q->cset);
         else if (q->key_width == 4)
             launch_build((unsigned int *) q->tkey, (const unsigned long long *) q->cset);
@@ -3068,6 +3243,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 6: launch(k_li_probe_agg_t<-1, unsigned int>, keys); break;
                 case 7: launch(k_li_probe_agg_t<-4, unsigned int>, keys); break;
                 case 8: launch(k_li_probe_agg_t<-2, unsigned int>, keys); break;
+                case 9: launch(k_li_probe_agg_t<-5, unsigned int>, keys); break;
+                case 10: launch(k_li_probe_agg_t<-6, unsigned int>, keys); break;
             }
         }
         else
@@ -3083,6 +3260,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 6: launch(k_li_probe_agg_t<-1, unsigned long long>, keys); break;
                 case 7: launch(k_li_probe_agg_t<-4, unsigned long long>, keys); break;
                 case 8: launch(k_li_probe_agg_t<-2, unsigned long long>, keys); break;
+                case 9: launch(k_li_probe_agg_t<-5, unsigned long long>, keys); break;
+                case 10: launch(k_li_probe_agg_t<-6, unsigned long long>, keys); break;
             }
         }
     }
