@@ -366,6 +366,12 @@ class Table:
             self.ctx._lib.gx_table_free(self._t)
             self._t = None
 
+    def __del__(self):
+        try:
+            self.free()
+        except Exception:
+            pass            # interpreter teardown / context already gone
+
 
 class Q3:
     def __init__(self, ctx, q):
@@ -412,3 +418,10 @@ class Q3:
         if self._q:
             self.ctx._lib.gx_q3_free(self._q)
             self._q = None
+
+    def __del__(self):
+        try:
+            self.free()
+        except Exception:
+            pass
+
