@@ -291,7 +291,8 @@ class Context:
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
                                                ctypes.byref(q)))
-        return Q3(self, q)
+        return Q3(self, q,
+                  tables=(desc_dict["dim"], desc_dict["mid"], desc_dict["fact"]))
 
     def q3(self, cust, orders, lineitem, cutoff=CUTOFF_19950315, numeric=False):
         q = ctypes.c_void_p()
@@ -299,7 +300,7 @@ class Context:
                                           cutoff, ctypes.byref(q)))
         if numeric:
             self._chk(self._lib.gx_q3_set_numeric(q, 1))
-        return Q3(self, q)
+        return Q3(self, q, tables=(cust, orders, lineitem))
 
     def close(self):
         if self._h:
@@ -374,9 +375,11 @@ class Table:
 
 
 class Q3:
-    def __init__(self, ctx, q):
+    def __init__(self, ctx, q, tables=()):
         self.ctx = ctx
         self._q = q
+        self._tables = tables    # keep the Tables alive: the native Q3
+                                 # holds raw device pointers into them
 
     def run(self):
         self.ctx._chk(self.ctx._lib.gx_q3_run(self._q))
