@@ -110,6 +110,9 @@ def _load():
     lib.gx_decode_column_nullable.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+    lib.gx_table_set_visimap.restype = ctypes.c_int
+    lib.gx_table_set_visimap.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_void_p, ctypes.c_int64]
     lib.gx_q1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                           ctypes.POINTER(ctypes.c_double)]
@@ -366,6 +369,21 @@ class Table:
         if self._t:
             self.ctx._lib.gx_table_free(self._t)
             self._t = None
+
+    def set_visimap(self, deleted):
+        """Attach an AO visimap: `deleted` is a per-row bool array
+        (True = tuple hidden/deleted); None clears.  Set before ctx.q3()."""
+        if deleted is None:
+            self.ctx._chk(self.ctx._lib.gx_table_set_visimap(
+                self.ctx._h, self._t, None, 0))
+            self._vmap = None
+            return
+        deleted = np.ascontiguousarray(deleted, np.uint8)
+        assert len(deleted) == self.nrows
+        bits = np.packbits(deleted, bitorder="little")
+        self.ctx._chk(self.ctx._lib.gx_table_set_visimap(
+            self.ctx._h, self._t, bits.ctypes.data, len(deleted)))
+        self._vmap = bits        # keep the host copy alive until replaced
 
     def __del__(self):
         try:

@@ -280,6 +280,7 @@ __global__ void k_q1_agg(const uint8_t *fl_s, gx_colmeta fl_m,
     double sr[6] = {0, 0, 0, 0, 0, 0};
     for (; i < fl_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(fl_m.vmap, i)) continue;
         if (!(gx_col_get<int32_t>(sh_s, sh_m, i) <= cutoff)) continue;
         int g = gx_col_get<int8_t>(fl_s, fl_m, i) * 2 +
                 gx_col_get<int8_t>(st_s, st_m, i);
@@ -712,6 +713,7 @@ __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
     unsigned long long local = 0;
     for (; i < m.nrows; i += stride)
     {
+        if (gx_vm_hidden(m.vmap, i)) continue;
         T v = gx_col_get<T>(col_s, m, i);
         bool pass = (op == 0) ? (v < lit) : (op == 1) ? (v > lit)
                   : (op == 2) ? (v == lit) : (v != lit);
@@ -731,7 +733,8 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0, kmax = 0;
     for (; i < mkt_m.nrows; i += stride)
-        if (gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit))
+        if (!gx_vm_hidden(mkt_m.vmap, i) &&
+            gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit))
         {
             local++;
             unsigned long long k = (unsigned long long) gx_col_get<int64_t>(key_s, key_m, i);
@@ -760,6 +763,7 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < key_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(mkt_m.vmap, i)) continue;
         if (!gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
         d_bloom_set(bloom, bwmask, k);
@@ -804,6 +808,7 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
     unsigned long long local = 0, kmax = 0, kmin = ~0ULL;
     for (; i < od_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(od_m.vmap, i)) continue;
         if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
         if (!d_bloom_test(bloom, bwmask, ck)) continue;
@@ -863,6 +868,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
     uint64_t tmask = smap.mask;
     for (; i < iend; i += stride)
     {
+        if (gx_vm_hidden(od_m.vmap, i)) continue;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -931,6 +937,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         int64_t stride = gridDim.x * (int64_t) blockDim.x;
         for (; i < lk_m.nrows; i += stride)
         {
+            if (gx_vm_hidden(sh_m.vmap, i)) continue;
             if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t slot = smap.slot0(k);
@@ -1346,6 +1353,7 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
             for (uint32_t r = rs; r < re; r++)
             {
                 int64_t g = first + r;
+                if (gx_vm_hidden(sh_m.vmap, g)) continue;
                 if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, g), flit)) continue;
                 double price = gx_col_get<double>(pr_s, pr_m, g);
                 double disc = gx_col_get<double>(di_s, di_m, g);
@@ -1383,6 +1391,7 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < lk_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(sh_m.vmap, i)) continue;
         if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
         uint64_t slot = smap.slot0(k);
@@ -1469,6 +1478,7 @@ __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < od_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(od_m.vmap, i)) continue;
         if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         int32_t d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
         atomicAdd(&hist[d], 1ULL);
@@ -1487,6 +1497,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < od_m.nrows; i += stride)
     {
+        if (gx_vm_hidden(od_m.vmap, i)) continue;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) continue;
         int64_t oc = gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -1618,6 +1629,7 @@ __global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
     if (lo >= hi) return;
 
     auto keep_row = [&](int64_t i) -> bool {
+        if (gx_vm_hidden(od_m.vmap, i)) return false;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) return false;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -1756,6 +1768,7 @@ struct gx_table {
     gx_ctx *ctx = nullptr;
     std::vector<gx_col> cols;
     int64_t nrows = 0;
+    uint8_t *dvmap = nullptr;    /* device visimap (1 bit/row, ON = hidden) */
 };
 
 struct gx_q3 {
@@ -2208,7 +2221,38 @@ extern "C" gx_status gx_table_free(gx_table *t)
         if (c.dstream) hipFree(c.dstream);
         if (c.ddir) hipFree(c.ddir);
     }
+    if (t->dvmap) hipFree(t->dvmap);
     delete t;
+    return GX_OK;
+}
+
+/* Attach / clear a scan-time visibility map: one bit per logical row,
+ * ON = tuple hidden (deleted) — the executor-visible semantics of
+ * AppendOnlyVisimap_IsVisible (cdbappendonlyvisimap.c:140-210), which
+ * every AOCS scan consults per tuple (aocsam.c:1205-1230).  Set it
+ * BEFORE gx_q3_prepare so table sizing sees the same row set. */
+extern "C" gx_status gx_table_set_visimap(gx_ctx *ctx, gx_table *t,
+                                          const uint8_t *bitmap, int64_t nbits)
+{
+    if (!ctx || !t) return GX_ERR_INVALID;
+    if (t->dvmap)
+    {
+        hipFree(t->dvmap);
+        t->dvmap = nullptr;
+    }
+    if (bitmap)
+    {
+        if (nbits != t->nrows)
+        {
+            set_err(ctx, "visimap bits != table rows%s", "");
+            return GX_ERR_INVALID;
+        }
+        int64_t bytes = (nbits + 7) >> 3;
+        HIP_CHK(ctx, hipMalloc(&t->dvmap, std::max<int64_t>(bytes, 1)));
+        HIP_CHK(ctx, hipMemcpy(t->dvmap, bitmap, bytes, hipMemcpyHostToDevice));
+    }
+    for (auto &c : t->cols)
+        c.m.vmap = t->dvmap;
     return GX_OK;
 }
 
@@ -3219,6 +3263,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
+        if (variant != 0 && q->li && q->li->dvmap)
+        {
+            set_err(ctx, "visimap requires the default probe variant%s", "");
+            return GX_ERR_INVALID;
+        }
         const char *pg = getenv("GX_PROBE_GRID");
         int pgrid = pg ? atoi(pg) : GRID;
         const char *pt = getenv("GX_PROBE_TPB");

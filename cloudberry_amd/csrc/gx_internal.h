@@ -159,7 +159,16 @@ struct gx_colmeta {
     int64_t full_block_len;    /* gx_aocs_block_len(width, rpb) */
     int64_t nbytes;            /* whole stream */
     uint64_t magic;            /* floor(2^64/rpb)+1 — division-free row→block */
+    const uint8_t *vmap;       /* table visimap (1 bit/row, ON = hidden) or
+                                  null — AppendOnlyVisimap scan visibility,
+                                  cdbappendonlyvisimap.c:140-210 */
 };
+
+/* scan-time tuple visibility (aocs_getnext → AppendOnlyVisimap_IsVisible) */
+GX_HD bool gx_vm_hidden(const uint8_t *vm, int64_t row)
+{
+    return vm != nullptr && ((vm[row >> 3] >> (row & 7)) & 1);
+}
 
 /* fill the magic multiplier: q = mulhi64(row, magic) == row / rpb, exact for
  * row < 2^64/rpb (rpb ≤ 16382 → exact beyond 10^15 rows; verified in tests

@@ -126,6 +126,13 @@ gx_status gx_decode_column_nullable(gx_ctx *ctx, const gx_table *t, int col,
                                     void *host_out, uint8_t *host_validity,
                                     int64_t cap_rows, int verify_checksums);
 
+/* Scan-time visibility map (AO visimap executor semantics,
+ * cdbappendonlyvisimap.c:140-210): bitmap has one bit per logical row,
+ * ON = tuple hidden (deleted); every scan/build/probe kernel skips
+ * hidden rows.  bitmap=NULL clears.  Set before gx_q3_prepare. */
+gx_status gx_table_set_visimap(gx_ctx *ctx, gx_table *t,
+                               const uint8_t *bitmap, int64_t nbits);
+
 /* TPC-H Q1 core (BASELINE config 4): GROUP BY returnflag,linestatus with
  * COUNT/SUM over a GX_TPCH_LINEITEM_Q1 table; AVG = sum/count (float8_avg) */
 gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,

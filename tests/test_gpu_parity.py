@@ -613,3 +613,74 @@ def test_gpu_forced_motion_path_matches_local(ctx, orc):
     np.testing.assert_array_equal(got["o_shippriority"], want["o_shippriority"])
     np.testing.assert_array_equal(got["nitems"], want["nitems"])
     np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
+
+
+@pytest.mark.gpu
+def test_gpu_visimap_q3_parity(ctx, orc):
+    """AO visimap executor semantics (cdbappendonlyvisimap.c): scans skip
+    hidden tuples on every table.  GPU Q3 with random deletes on customer,
+    orders AND lineitem must equal the oracle run on the kept rows only."""
+    sf = 0.2
+    rng = np.random.default_rng(31)
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    dc = rng.random(cust.nrows) < 0.1
+    do = rng.random(ordr.nrows) < 0.2
+    dl = rng.random(li.nrows) < 0.15
+    cust.set_visimap(dc)
+    ordr.set_visimap(do)
+    li.set_visimap(dl)
+    got = ctx.q3(cust, ordr, li).run().result()
+
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    l = orc.gen_lineitem(sf)
+    c = {k: v[~dc] for k, v in c.items()}
+    o = {k: v[~do] for k, v in o.items()}
+    l = {k: v[~dl] for k, v in l.items()}
+    want = orc.q3(c, o, l)
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["o_orderdate"], want["o_orderdate"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-9)
+
+    # clearing restores the full result
+    cust.set_visimap(None)
+    ordr.set_visimap(None)
+    li.set_visimap(None)
+    full = ctx.q3(cust, ordr, li).run().result()
+    want_full = orc.q3(orc.gen_customer(sf), orc.gen_orders(sf),
+                       orc.gen_lineitem(sf))
+    np.testing.assert_array_equal(full["l_orderkey"], want_full["l_orderkey"])
+
+
+@pytest.mark.gpu
+def test_gpu_visimap_rle_and_scan(ctx, orc):
+    """Visimap on the RLE fused-scan path and on gx_scan_filter."""
+    sf = 0.1
+    rng = np.random.default_rng(37)
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM_RLEKEY, sf)
+    dl = rng.random(li.nrows) < 0.25
+    li.set_visimap(dl)
+    got = ctx.q3(cust, ordr, li).run().result()
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    l = orc.gen_lineitem(sf)
+    l = {k: v[~dl] for k, v in l.items()}
+    want = orc.q3(c, o, l)
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-9)
+
+    # scan_filter skips hidden rows
+    t = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    dd = rng.random(t.nrows) < 0.5
+    odate = orc.gen_orders(sf)["o_orderdate"]
+    n_all, _ = t.scan_filter(2, "<", gx.CUTOFF_19950315)
+    t.set_visimap(dd)
+    n_vis, _ = t.scan_filter(2, "<", gx.CUTOFF_19950315)
+    assert n_all == int((odate < gx.CUTOFF_19950315).sum())
+    assert n_vis == int((odate[~dd] < gx.CUTOFF_19950315).sum())
