@@ -684,3 +684,20 @@ def test_gpu_visimap_rle_and_scan(ctx, orc):
     n_vis, _ = t.scan_filter(2, "<", gx.CUTOFF_19950315)
     assert n_all == int((odate < gx.CUTOFF_19950315).sum())
     assert n_vis == int((odate[~dd] < gx.CUTOFF_19950315).sum())
+
+
+@pytest.mark.gpu
+def test_gpu_multi_segfile_concat_decode(ctx, orc):
+    """GPU dir-based decode of a column concatenated from three segment
+    files' streams (aocsam.c open_next_scan_seg semantics)."""
+    rng = np.random.default_rng(43)
+    vals = np.repeat(np.arange(1, 7000, dtype=np.int64),
+                     rng.integers(1, 9, 6999))
+    cut = len(vals) // 3
+    s = (orc.aocs_encode_rle_delta(vals[:cut])
+         + orc.aocs_encode_rle_delta(vals[cut:2 * cut])
+         + orc.aocs_encode_rle_delta(vals[2 * cut:]))
+    t = ctx.bind([(s, 8, len(vals), 1)])
+    np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True),
+                                  vals)
+    t.free()

@@ -616,3 +616,23 @@ def test_plain_decode_refuses_null_blocks():
     s0 = orc.aocs_encode_orig_nulls(vals, nulls)
     with pytest.raises(AssertionError):
         orc.aocs_decode(s0, 8, 1000, np.int64)
+
+
+def test_multi_segfile_stream_concat_decodes():
+    """A column scanned across MULTIPLE AO segment files is the
+    concatenation of their streams (aocsam.c open_next_scan_seg advances
+    through segfiles); the decoder walks block-by-block with a running row
+    counter, so concatenated streams decode as one logical column."""
+    rng = np.random.default_rng(41)
+    vals = np.repeat(np.arange(1, 7000, dtype=np.int64),
+                     rng.integers(1, 9, 6999))
+    cut = len(vals) // 3
+    s = (orc.aocs_encode_rle_delta(vals[:cut])
+         + orc.aocs_encode_rle_delta(vals[cut:2 * cut])
+         + orc.aocs_encode_rle_delta(vals[2 * cut:]))
+    out = orc.aocs_decode(s, 8, len(vals), np.int64)
+    np.testing.assert_array_equal(out, vals)
+    # Orig segfiles concatenate the same way
+    s0 = orc.aocs_encode(vals[:cut]) + orc.aocs_encode(vals[cut:])
+    np.testing.assert_array_equal(orc.aocs_decode(s0, 8, len(vals), np.int64),
+                                  vals)
