@@ -269,7 +269,7 @@ __global__ void k_q1_agg(const uint8_t *fl_s, gx_colmeta fl_m,
                          const uint8_t *pr_s, gx_colmeta pr_m,
                          const uint8_t *di_s, gx_colmeta di_m,
                          const uint8_t *sh_s, gx_colmeta sh_m,
-                         int32_t cutoff,
+                         const uint8_t *vmap, int32_t cutoff,
                          unsigned long long *g_count, double *g_price,
                          double *g_rev)
 {
@@ -280,7 +280,7 @@ __global__ void k_q1_agg(const uint8_t *fl_s, gx_colmeta fl_m,
     double sr[6] = {0, 0, 0, 0, 0, 0};
     for (; i < fl_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(fl_m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         if (!(gx_col_get<int32_t>(sh_s, sh_m, i) <= cutoff)) continue;
         int g = gx_col_get<int8_t>(fl_s, fl_m, i) * 2 +
                 gx_col_get<int8_t>(st_s, st_m, i);
@@ -706,6 +706,7 @@ __device__ __forceinline__ void gx_wave_count_add(unsigned long long *dst,
  * two-valued on NOT NULL columns (execScan.c:241). */
 template <typename T>
 __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
+                              const uint8_t *vmap,
                               int op, T lit, unsigned long long *count)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
@@ -713,7 +714,7 @@ __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
     unsigned long long local = 0;
     for (; i < m.nrows; i += stride)
     {
-        if (gx_vm_hidden(m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         T v = gx_col_get<T>(col_s, m, i);
         bool pass = (op == 0) ? (v < lit) : (op == 1) ? (v > lit)
                   : (op == 2) ? (v == lit) : (v != lit);
@@ -725,7 +726,7 @@ __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
 /* customer: count BUILDING rows (for set sizing) */
 __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
-                             int cop, int8_t clit,
+                             const uint8_t *vmap, int cop, int8_t clit,
                              unsigned long long *count,
                              unsigned long long *maxkey)
 {
@@ -733,7 +734,7 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0, kmax = 0;
     for (; i < mkt_m.nrows; i += stride)
-        if (!gx_vm_hidden(mkt_m.vmap, i) &&
+        if (!gx_vm_hidden(vmap, i) &&
             gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit))
         {
             local++;
@@ -755,7 +756,7 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
 template <typename KS>
 __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
-                             int cop, int8_t clit,
+                             const uint8_t *vmap, int cop, int8_t clit,
                              KS *set, uint64_t mask,
                              unsigned long long *bloom, uint64_t bwmask)
 {
@@ -763,7 +764,7 @@ __global__ void k_cust_build(const uint8_t *key_s, gx_colmeta key_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < key_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(mkt_m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
         d_bloom_set(bloom, bwmask, k);
@@ -796,7 +797,7 @@ template <typename KS>
 __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
-                               int oop, int32_t olit,
+                               const uint8_t *vmap, int oop, int32_t olit,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                unsigned long long *count,
@@ -808,7 +809,7 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
     unsigned long long local = 0, kmax = 0, kmin = ~0ULL;
     for (; i < od_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(od_m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
         if (!d_bloom_test(bloom, bwmask, ck)) continue;
@@ -840,12 +841,12 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
  * key array then sits comfortably in the 256 MiB Infinity Cache), u64
  * otherwise.  Key compares stay exact either way (PG narrow-int hashing
  * spirit; sentinel 0 is safe — orderkeys start at 1). */
-template <typename KT, typename KS, bool CHUNKED = false>
+template <typename KT, typename KS, bool CHUNKED = false, bool VM = false>
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
-                               int oop, int32_t olit,
+                               const uint8_t *vmap, int oop, int32_t olit,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                KT *tkey,
@@ -868,7 +869,8 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
     uint64_t tmask = smap.mask;
     for (; i < iend; i += stride)
     {
-        if (gx_vm_hidden(od_m.vmap, i)) continue;
+        if constexpr (VM)
+            if (gx_vm_hidden(vmap, i)) continue;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -901,12 +903,12 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * ~300M L3 probe round-trips).  Guard-free main region — the tail is a
  * separate scalar loop — and filtered lanes probe slot 0 (L1-resident)
  * instead of branching, so each load batch stays in one basic block. */
-template <int B, typename KT>
+template <int B, typename KT, bool VM = false>
 __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
                                  const uint8_t *di_s, gx_colmeta di_m,
                                  const uint8_t *sh_s, gx_colmeta sh_m,
-                                 int fop, int32_t flit,
+                                 const uint8_t *vmap, int fop, int32_t flit,
                                  const KT *tkey,
                                  double *trev, unsigned long long *tcnt,
                                  gx_slotmap smap,
@@ -935,9 +937,13 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
     {
         int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
         int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        /* visimap via a compile-time template split (VM): the default
+         * instantiation carries no per-row check at all — a runtime guard
+         * here cost 57% (measured, profiles/bw_probe_r01.txt) */
         for (; i < lk_m.nrows; i += stride)
         {
-            if (gx_vm_hidden(sh_m.vmap, i)) continue;
+            if constexpr (VM)
+                if (gx_vm_hidden(vmap, i)) continue;
             if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
             uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
             uint64_t slot = smap.slot0(k);
@@ -1171,13 +1177,13 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
  * serial per-block walk. */
 static constexpr int RLE_MAX_PHYS = 4096;
 
-template <typename KT>
+template <typename KT, bool VM = false>
 __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
                                    int64_t nblocks,
                                    const uint8_t *pr_s, gx_colmeta pr_m,
                                    const uint8_t *di_s, gx_colmeta di_m,
                                    const uint8_t *sh_s, gx_colmeta sh_m,
-                                   int fop, int32_t flit,
+                                   const uint8_t *vmap, int fop, int32_t flit,
                                    const KT *tkey,
                                    double *trev, unsigned long long *tcnt,
                                    gx_slotmap smap,
@@ -1353,7 +1359,8 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
             for (uint32_t r = rs; r < re; r++)
             {
                 int64_t g = first + r;
-                if (gx_vm_hidden(sh_m.vmap, g)) continue;
+                if constexpr (VM)
+                    if (gx_vm_hidden(vmap, g)) continue;
                 if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, g), flit)) continue;
                 double price = gx_col_get<double>(pr_s, pr_m, g);
                 double disc = gx_col_get<double>(di_s, di_m, g);
@@ -1373,12 +1380,12 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
  * the PG numeric SUM for these ranges.  The numerator lives in the trev
  * buffer (reinterpreted u64).  Overflow guard: numerators past 2^62 set the
  * error flag (impossible for sane groups; detects corrupt input). */
-template <typename KT>
+template <typename KT, bool VM = false>
 __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
                                    const uint8_t *pr_s, gx_colmeta pr_m,
                                    const uint8_t *di_s, gx_colmeta di_m,
                                    const uint8_t *sh_s, gx_colmeta sh_m,
-                                   int fop, int32_t flit,
+                                   const uint8_t *vmap, int fop, int32_t flit,
                                    const KT *tkey,
                                    unsigned long long *tnum,
                                    unsigned long long *tcnt,
@@ -1391,7 +1398,8 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < lk_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(sh_m.vmap, i)) continue;
+        if constexpr (VM)
+            if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
         uint64_t slot = smap.slot0(k);
@@ -1471,6 +1479,7 @@ __global__ void k_extract(const KT *tkey, const int32_t *tdate,
 /* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
 __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
+                              const uint8_t *vmap,
                               int oop, int32_t olit, int nsegs,
                               unsigned long long *hist)
 {
@@ -1478,7 +1487,7 @@ __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < od_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(od_m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         int32_t d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
         atomicAdd(&hist[d], 1ULL);
@@ -1489,6 +1498,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
                               const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *op_s, gx_colmeta op_m,
+                              const uint8_t *vmap,
                               int oop, int32_t olit, int nsegs,
                               unsigned long long *cursors, /* pre-set to region starts */
                               gx_ord_row *out)
@@ -1497,7 +1507,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     for (; i < od_m.nrows; i += stride)
     {
-        if (gx_vm_hidden(od_m.vmap, i)) continue;
+        if (gx_vm_hidden(vmap, i)) continue;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) continue;
         int64_t oc = gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -1616,7 +1626,7 @@ __global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *op_s, gx_colmeta op_m,
-                               int oop, int32_t olit,
+                               const uint8_t *vmap, int oop, int32_t olit,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                gx_qual_row *outq, unsigned long long *cursor)
@@ -1629,7 +1639,7 @@ __global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
     if (lo >= hi) return;
 
     auto keep_row = [&](int64_t i) -> bool {
-        if (gx_vm_hidden(od_m.vmap, i)) return false;
+        if (gx_vm_hidden(vmap, i)) return false;
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) return false;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
@@ -2251,8 +2261,6 @@ extern "C" gx_status gx_table_set_visimap(gx_ctx *ctx, gx_table *t,
         HIP_CHK(ctx, hipMalloc(&t->dvmap, std::max<int64_t>(bytes, 1)));
         HIP_CHK(ctx, hipMemcpy(t->dvmap, bitmap, bytes, hipMemcpyHostToDevice));
     }
-    for (auto &c : t->cols)
-        c.m.vmap = t->dvmap;
     return GX_OK;
 }
 
@@ -2603,7 +2611,7 @@ extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
                        t->cols[2].dstream, t->cols[2].m,
                        t->cols[3].dstream, t->cols[3].m,
                        t->cols[4].dstream, t->cols[4].m,
-                       cutoff, cnt.as<unsigned long long>(),
+                       t->dvmap, cutoff, cnt.as<unsigned long long>(),
                        spr.as<double>(), srv.as<double>());
     HIP_CHK(ctx, hipEventRecord(e1, s));
     HIP_CHK(ctx, hipMemcpyAsync(counts6, cnt.p, 48, hipMemcpyDeviceToHost, s));
@@ -2636,15 +2644,15 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     HIP_CHK(ctx, hipEventRecord(e0, s));
     if (c.m.width == 8)
         hipLaunchKernelGGL(k_scan_filter<int64_t>, dim3(GRID), dim3(TPB), 0, s,
-                           c.dstream, c.m, op, (int64_t) literal,
+                           c.dstream, c.m, t->dvmap, op, (int64_t) literal,
                            cnt.as<unsigned long long>());
     else if (c.m.width == 4)
         hipLaunchKernelGGL(k_scan_filter<int32_t>, dim3(GRID), dim3(TPB), 0, s,
-                           c.dstream, c.m, op, (int32_t) literal,
+                           c.dstream, c.m, t->dvmap, op, (int32_t) literal,
                            cnt.as<unsigned long long>());
     else
         hipLaunchKernelGGL(k_scan_filter<int8_t>, dim3(GRID), dim3(TPB), 0, s,
-                           c.dstream, c.m, op, (int8_t) literal,
+                           c.dstream, c.m, t->dvmap, op, (int8_t) literal,
                            cnt.as<unsigned long long>());
     HIP_CHK(ctx, hipEventRecord(e1, s));
     unsigned long long n = 0;
@@ -2780,7 +2788,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
                        q->cust->cols[D.dim_key_col].dstream,
                        q->cust->cols[D.dim_key_col].m,
-                       cm.dstream, cm.m, D.dim_filter.op,
+                       cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
                        (int8_t) D.dim_filter.literal, q->dcount, q->dhits);
     unsigned long long n_building = 0, cmax = 0;
     HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
@@ -2800,7 +2808,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.m, D.dim_filter.op,
+                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
                            (int8_t) D.dim_filter.literal,
                            (unsigned int *) q->cset, q->cmask,
                            q->bloom, q->bwmask);
@@ -2808,7 +2816,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.m, D.dim_filter.op,
+                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
                            (int8_t) D.dim_filter.literal,
                            (unsigned long long *) q->cset, q->cmask,
                            q->bloom, q->bwmask);
@@ -2829,6 +2837,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                q->ord->cols[D.mid_key_col].dstream,
                                q->ord->cols[D.mid_key_col].m,
                                od.dstream, od.m, oc.dstream, oc.m,
+                               q->ord->dvmap,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned int *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
@@ -2838,6 +2847,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                q->ord->cols[D.mid_key_col].dstream,
                                q->ord->cols[D.mid_key_col].m,
                                od.dstream, od.m, oc.dstream, oc.m,
+                               q->ord->dvmap,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned long long *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
@@ -2924,12 +2934,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, (q->bwmask + 1) * 8, s));
     if (q->cset_width == 4)
         hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                           ck.dstream, ck.m, cm.dstream, cm.m,
+                           ck.dstream, ck.m, cm.dstream, cm.m, q->cust->dvmap,
                            D.dim_filter.op, (int8_t) D.dim_filter.literal,
                            (unsigned int *) q->cset, q->cmask, q->bloom, q->bwmask);
     else
         hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                           ck.dstream, ck.m, cm.dstream, cm.m,
+                           ck.dstream, ck.m, cm.dstream, cm.m, q->cust->dvmap,
                            D.dim_filter.op, (int8_t) D.dim_filter.literal,
                            (unsigned long long *) q->cset, q->cmask, q->bloom, q->bwmask);
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
@@ -2948,25 +2958,30 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
         int ogrid = env_int("GX_ORDERS_GRID", 32768);  /* measured optimum */
         bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
+        const uint8_t *ovm = q->ord->dvmap;
         auto launch_build = [&](auto *tk, auto *cs) {
+            auto go = [&](auto ch, auto vm) {
+                hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
+                                                   std::decay_t<decltype(*cs)>,
+                                                   decltype(ch)::value,
+                                                   decltype(vm)::value>),
+                                   dim3(ogrid), dim3(TPB), 0, s,
+                                   ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
+                                   op.dstream, op.m, ovm, D.mid_filter.op,
+                                   (int32_t) D.mid_filter.literal, cs, q->cmask,
+                                   q->bloom, q->bwmask,
+                                   tk, q->tdate, q->tprio, q->smap);
+            };
             if (ochunk)
-                hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
-                                                   std::decay_t<decltype(*cs)>, true>),
-                                   dim3(ogrid), dim3(TPB), 0, s,
-                                   ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                                   op.dstream, op.m, D.mid_filter.op,
-                                   (int32_t) D.mid_filter.literal, cs, q->cmask,
-                                   q->bloom, q->bwmask,
-                                   tk, q->tdate, q->tprio, q->smap);
+            {
+                if (ovm) go(std::true_type{}, std::true_type{});
+                else go(std::true_type{}, std::false_type{});
+            }
             else
-                hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
-                                                   std::decay_t<decltype(*cs)>, false>),
-                                   dim3(ogrid), dim3(TPB), 0, s,
-                                   ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                                   op.dstream, op.m, D.mid_filter.op,
-                                   (int32_t) D.mid_filter.literal, cs, q->cmask,
-                                   q->bloom, q->bwmask,
-                                   tk, q->tdate, q->tprio, q->smap);
+            {
+                if (ovm) go(std::false_type{}, std::true_type{});
+                else go(std::false_type{}, std::false_type{});
+            }
         };
         if (env_int("GX_ORDERS_TWOPASS", 0))
         {
@@ -2986,6 +3001,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                    dim3(ogrid), dim3(TPB), 0, s,
                                    ok.dstream, ok.m, oc.dstream, oc.m,
                                    od.dstream, od.m, op.dstream, op.m,
+                                   q->ord->dvmap,
                                    D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                    cs, q->cmask, q->bloom, q->bwmask,
                                    q->m_send2, q->dcount);
@@ -3049,7 +3065,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         unsigned long long *dhist = q->m_hist;
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
-                           od.dstream, od.m, oc.dstream, oc.m,
+                           od.dstream, od.m, oc.dstream, oc.m, q->ord->dvmap,
                            D.mid_filter.op, (int32_t) D.mid_filter.literal, n, dhist);
         std::vector<unsigned long long> h1(n);
         HIP_CHK(ctx, hipMemcpyAsync(h1.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
@@ -3063,7 +3079,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                           op.dstream, op.m, D.mid_filter.op,
+                           op.dstream, op.m, q->ord->dvmap, D.mid_filter.op,
                            (int32_t) D.mid_filter.literal, n, dcur, send1);
 
         /* exchange counts (all-gather of per-dest counts) */
@@ -3216,20 +3232,27 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         int64_t nb = lk.nblocks;
         int rgrid = (int) std::min<int64_t>(nb, 16384);
+        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
+        auto launch_rle = [&](auto *keys, auto vm) {
+            hipLaunchKernelGGL((k_li_probe_agg_rle<std::decay_t<decltype(*keys)>,
+                                                   decltype(vm)::value>),
+                               dim3(rgrid), dim3(TPB), 0, s,
+                               lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
+                               ld.dstream, ld.m, ls.dstream, ls.m, lvm,
+                               D.fact_filter.op, (int32_t) D.fact_filter.literal,
+                               keys, q->trev, q->tcnt, q->smap, dhits,
+                               (int *) q->dmin);
+        };
         if (q->key_width == 4)
-            hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned int>, dim3(rgrid), dim3(TPB), 0, s,
-                               lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
-                               ld.dstream, ld.m, ls.dstream, ls.m,
-                               D.fact_filter.op, (int32_t) D.fact_filter.literal,
-                               (const unsigned int *) q->tkey,
-                               q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
+        {
+            if (lvm) launch_rle((const unsigned int *) q->tkey, std::true_type{});
+            else launch_rle((const unsigned int *) q->tkey, std::false_type{});
+        }
         else
-            hipLaunchKernelGGL(k_li_probe_agg_rle<unsigned long long>, dim3(rgrid), dim3(TPB), 0, s,
-                               lk.dstream, lk.ddir, nb, lp.dstream, lp.m,
-                               ld.dstream, ld.m, ls.dstream, ls.m,
-                               D.fact_filter.op, (int32_t) D.fact_filter.literal,
-                               (const unsigned long long *) q->tkey,
-                               q->trev, q->tcnt, q->smap, dhits, (int *) q->dmin);
+        {
+            if (lvm) launch_rle((const unsigned long long *) q->tkey, std::true_type{});
+            else launch_rle((const unsigned long long *) q->tkey, std::false_type{});
+        }
         int herr = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&herr, q->dmin, 4, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3238,22 +3261,27 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     else if (q->numeric)
     {
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
+        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
+        auto launch_num = [&](auto *keys, auto vm) {
+            hipLaunchKernelGGL((k_li_probe_agg_num<std::decay_t<decltype(*keys)>,
+                                                   decltype(vm)::value>),
+                               dim3(GRID), dim3(TPB), 0, s,
+                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
+                               ls.dstream, ls.m, lvm, D.fact_filter.op,
+                               (int32_t) D.fact_filter.literal, keys,
+                               (unsigned long long *) q->trev, q->tcnt, q->smap,
+                               dhits, (int *) q->dmin);
+        };
         if (q->key_width == 4)
-            hipLaunchKernelGGL(k_li_probe_agg_num<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, D.fact_filter.op,
-                               (int32_t) D.fact_filter.literal,
-                               (const unsigned int *) q->tkey,
-                               (unsigned long long *) q->trev, q->tcnt, q->smap,
-                               dhits, (int *) q->dmin);
+        {
+            if (lvm) launch_num((const unsigned int *) q->tkey, std::true_type{});
+            else launch_num((const unsigned int *) q->tkey, std::false_type{});
+        }
         else
-            hipLaunchKernelGGL(k_li_probe_agg_num<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, D.fact_filter.op,
-                               (int32_t) D.fact_filter.literal,
-                               (const unsigned long long *) q->tkey,
-                               (unsigned long long *) q->trev, q->tcnt, q->smap,
-                               dhits, (int *) q->dmin);
+        {
+            if (lvm) launch_num((const unsigned long long *) q->tkey, std::true_type{});
+            else launch_num((const unsigned long long *) q->tkey, std::false_type{});
+        }
         int herr = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&herr, q->dmin, 4, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3272,10 +3300,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         int pgrid = pg ? atoi(pg) : GRID;
         const char *pt = getenv("GX_PROBE_TPB");
         int ptpb = pt ? atoi(pt) : TPB;
+        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
         auto launch = [&](auto kern, auto *keys) {
             hipLaunchKernelGGL(kern, dim3(pgrid), dim3(ptpb), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                               ls.dstream, ls.m, D.fact_filter.op,
+                               ls.dstream, ls.m, lvm, D.fact_filter.op,
                                (int32_t) D.fact_filter.literal, keys,
                                q->trev, q->tcnt, q->smap, dhits);
         };
@@ -3285,7 +3314,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             switch (variant)
             {
                 default:
-                case 0: launch(k_li_probe_agg_t<1, unsigned int>, keys); break;
+                case 0:
+                    if (lvm) launch((k_li_probe_agg_t<1, unsigned int, true>), keys);
+                    else launch(k_li_probe_agg_t<1, unsigned int>, keys);
+                    break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned int>, keys); break;
@@ -3302,7 +3334,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             switch (variant)
             {
                 default:
-                case 0: launch(k_li_probe_agg_t<1, unsigned long long>, keys); break;
+                case 0:
+                    if (lvm) launch((k_li_probe_agg_t<1, unsigned long long, true>), keys);
+                    else launch(k_li_probe_agg_t<1, unsigned long long>, keys);
+                    break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
                 case 5: launch(k_li_probe_agg_t<8, unsigned long long>, keys); break;
@@ -3512,7 +3547,8 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     unsigned long long *dcur = dcur_b.as<unsigned long long>();
     HIP_CHK(ctx, hipMemsetAsync(dhist, 0, nsegs * 8, s));
     hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
-                       od.dstream, od.m, oc.dstream, oc.m, 0, cutoff, nsegs, dhist);
+                       od.dstream, od.m, oc.dstream, oc.m,
+                       (const uint8_t *) nullptr, 0, cutoff, nsegs, dhist);
     std::vector<unsigned long long> h(nsegs);
     HIP_CHK(ctx, hipMemcpyAsync(h.data(), dhist, nsegs * 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3525,7 +3561,8 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     HIP_CHK(ctx, hipMemcpyAsync(dcur, off.data(), nsegs * 8, hipMemcpyHostToDevice, s));
     hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                        ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                       op.dstream, op.m, 0, cutoff, nsegs, dcur, dsend);
+                       op.dstream, op.m, (const uint8_t *) nullptr,
+                       0, cutoff, nsegs, dcur, dsend);
     HIP_CHK(ctx, hipMemcpyAsync(out_rows, dsend, total * sizeof(gx_ord_row),
                                 hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -3554,7 +3591,8 @@ extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
     HIP_CHK(ctx, cnt_b.alloc(16));
     HIP_CHK(ctx, hipMemsetAsync(cnt_b.p, 0, 16, s));
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                       ck.dstream, ck.m, cm.dstream, cm.m, 2, (int8_t) 0,
+                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       (const uint8_t *) nullptr, 2, (int8_t) 0,
                        cnt_b.as<unsigned long long>(),
                        cnt_b.as<unsigned long long>() + 1);
     unsigned long long nb[2];
@@ -3568,7 +3606,8 @@ extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
     HIP_CHK(ctx, hipMemsetAsync(set_b.p, 0, cslots * 8, s));
     HIP_CHK(ctx, hipMemsetAsync(bloom_b.p, 0, bwords * 8, s));
     hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                       ck.dstream, ck.m, cm.dstream, cm.m, 2, (int8_t) 0,
+                       ck.dstream, ck.m, cm.dstream, cm.m,
+                       (const uint8_t *) nullptr, 2, (int8_t) 0,
                        set_b.as<unsigned long long>(), cslots - 1,
                        bloom_b.as<unsigned long long>(), bwords - 1);
     HIP_CHK(ctx, rows_b.alloc(std::max<int64_t>(n, 1) * sizeof(gx_ord_row)));
@@ -3660,7 +3699,7 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
     HIP_CHK(ctx, hipMemsetAsync(hit_b.p, 0, 8, s));
     hipLaunchKernelGGL((k_li_probe_agg_t<1, unsigned long long>), dim3(GRID), dim3(TPB), 0, s,
                        lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
-                       ls.dstream, ls.m, 1, cutoff,
+                       ls.dstream, ls.m, (const uint8_t *) nullptr, 1, cutoff,
                        key_b.as<unsigned long long>(),
                        rev_b.as<double>(), cnt_b2.as<unsigned long long>(),
                        smap, hit_b.as<unsigned long long>());

@@ -159,9 +159,6 @@ struct gx_colmeta {
     int64_t full_block_len;    /* gx_aocs_block_len(width, rpb) */
     int64_t nbytes;            /* whole stream */
     uint64_t magic;            /* floor(2^64/rpb)+1 — division-free row→block */
-    const uint8_t *vmap;       /* table visimap (1 bit/row, ON = hidden) or
-                                  null — AppendOnlyVisimap scan visibility,
-                                  cdbappendonlyvisimap.c:140-210 */
 };
 
 /* scan-time tuple visibility (aocs_getnext → AppendOnlyVisimap_IsVisible) */
