@@ -1885,7 +1885,7 @@ static gx_status encode_column_device(gx_ctx *ctx, const void *dvals, int width,
     uint8_t *stream = nullptr;
     HIP_CHK(ctx, hipMalloc(&stream, bytes));
     HIP_CHK(ctx, hipMemsetAsync(stream, 0, bytes, ctx->stream));
-    int grid = (int) std::min<int64_t>(nblocks, 65535);
+    int grid = (int) std::min<int64_t>(std::max<int64_t>(nblocks, 1), 65535);
     if (width == 8)
         hipLaunchKernelGGL(k_encode<int64_t>, dim3(grid), dim3(TPB), 0, ctx->stream,
                            (const int64_t *) dvals, nrows, rpb, full_len, nblocks, stream);
@@ -2311,7 +2311,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
     bool li_q1 = ((int) which == 5);        /* GX_TPCH_LINEITEM_Q1 */
     if (li_numeric || li_rlekey || li_q1) which = GX_TPCH_LINEITEM;
     int64_t nthreads = (nglobal + GEN_CHUNK - 1) / GEN_CHUNK;
-    int64_t blocks = (nthreads + TPB - 1) / TPB;
+    int64_t blocks = std::max<int64_t>((nthreads + TPB - 1) / TPB, 1);
     /* the count kernels write counts[t] for EVERY launched thread */
     int64_t nthreads_alloc = blocks * TPB;
     uint32_t *dcounts = nullptr;
@@ -2485,6 +2485,11 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     const gx_col &c = t->cols[colidx];
     if (c.m.nrows > cap_rows) return GX_ERR_INVALID;
     int64_t nblocks = (c.m.nrows + c.m.rpb - 1) / c.m.rpb;
+    if (c.m.nrows == 0)
+    {
+        /* empty stream: nothing to decode (reference: zero AO blocks) */
+        return GX_OK;
+    }
     devbuf dout_b, derr_b;
     HIP_CHK(ctx, dout_b.alloc(c.m.nrows * (int64_t) c.m.width));
     HIP_CHK(ctx, derr_b.alloc(4));
@@ -3231,7 +3236,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->numeric) { set_err(ctx, "numeric+RLE combo not supported%s", ""); return GX_ERR_INVALID; }
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         int64_t nb = lk.nblocks;
-        int rgrid = (int) std::min<int64_t>(nb, 16384);
+        int rgrid = (int) std::min<int64_t>(std::max<int64_t>(nb, 1), 16384);
         const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
         auto launch_rle = [&](auto *keys, auto vm) {
             hipLaunchKernelGGL((k_li_probe_agg_rle<std::decay_t<decltype(*keys)>,

@@ -701,3 +701,30 @@ def test_gpu_multi_segfile_concat_decode(ctx, orc):
     np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True),
                                   vals)
     t.free()
+
+
+@pytest.mark.gpu
+def test_gpu_empty_tables(ctx, orc):
+    """Empty tables (zero AO blocks) through gen, bind, decode and the
+    full Q3 pipeline — the uao_* regress edge the reference tests."""
+    c = ctx.tpch_gen(gx.TPCH_CUSTOMER, 0.0)
+    o = ctx.tpch_gen(gx.TPCH_ORDERS, 0.0)
+    l = ctx.tpch_gen(gx.TPCH_LINEITEM, 0.0)
+    assert (c.nrows, o.nrows, l.nrows) == (0, 0, 0)
+    assert len(ctx.q3(c, o, l).run().result()["l_orderkey"]) == 0
+    # one empty side only
+    c2 = ctx.tpch_gen(gx.TPCH_CUSTOMER, 0.05)
+    o2 = ctx.tpch_gen(gx.TPCH_ORDERS, 0.05)
+    assert len(ctx.q3(c2, o2, l).run().result()["l_orderkey"]) == 0
+    # empty bind + decode, both formats
+    for fmt in (0, 1):
+        t = ctx.bind([(b"", 8, 0, fmt)])
+        assert t.nrows == 0
+        assert t.decode_column(0, np.int64).shape == (0,)
+        t.free()
+    # single-row stream end-to-end
+    s = orc.aocs_encode_rle_delta(np.array([7], np.int64))
+    t = ctx.bind([(s, 8, 1, 1)])
+    np.testing.assert_array_equal(t.decode_column(0, np.int64, verify=True),
+                                  [7])
+    t.free()

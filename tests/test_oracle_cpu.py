@@ -636,3 +636,26 @@ def test_multi_segfile_stream_concat_decodes():
     s0 = orc.aocs_encode(vals[:cut]) + orc.aocs_encode(vals[cut:])
     np.testing.assert_array_equal(orc.aocs_decode(s0, 8, len(vals), np.int64),
                                   vals)
+
+
+def test_empty_and_single_row_tables():
+    """Empty AOCS streams (0 blocks) and single-row blocks — the edge cases
+    the reference's AO regress tests exercise (uao_* schedules)."""
+    empty = np.array([], np.int64)
+    for enc in (orc.aocs_encode, orc.aocs_encode_rle, orc.aocs_encode_rle_delta):
+        s = enc(empty)
+        assert s == b""
+        np.testing.assert_array_equal(orc.aocs_decode(s, 8, 0, np.int64), empty)
+    one = np.array([42], np.int64)
+    for enc in (orc.aocs_encode, orc.aocs_encode_rle, orc.aocs_encode_rle_delta):
+        np.testing.assert_array_equal(orc.aocs_decode(enc(one), 8, 1, np.int64),
+                                      one)
+    # empty inputs through the whole pipeline
+    c = {"c_custkey": empty, "c_mktsegment": np.array([], np.uint8)}
+    o = {"o_orderkey": empty, "o_custkey": empty,
+         "o_orderdate": np.array([], np.int32),
+         "o_shippriority": np.array([], np.int32)}
+    l = {"l_orderkey": empty, "l_extendedprice": np.array([], np.float64),
+         "l_discount": np.array([], np.float64),
+         "l_shipdate": np.array([], np.int32)}
+    assert len(orc.q3(c, o, l)["l_orderkey"]) == 0
