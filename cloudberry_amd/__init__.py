@@ -113,6 +113,10 @@ def _load():
     lib.gx_table_set_visimap.restype = ctypes.c_int
     lib.gx_table_set_visimap.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                          ctypes.c_void_p, ctypes.c_int64]
+    lib.gx_decode_column_varlena.restype = ctypes.c_int
+    lib.gx_decode_column_varlena.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_int]
     lib.gx_q1.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                           ctypes.POINTER(ctypes.c_double)]
@@ -205,7 +209,9 @@ class Context:
             descs[i].codec = spec[4] if len(spec) > 4 else 0
         t = ctypes.c_void_p()
         self._chk(self._lib.gx_table_bind(self._h, descs, len(streams), ctypes.byref(t)))
-        return Table(self, t)
+        tb = Table(self, t)
+        tb._col_nbytes = [len(s[0]) for s in streams]
+        return tb
 
     def partition(self, keys, nsegs):
         keys = np.ascontiguousarray(keys, np.int64)
@@ -353,6 +359,24 @@ class Table:
         self.ctx._chk(self.ctx._lib.gx_decode_column(
             self.ctx._h, self._t, col, out.ctypes.data, n, 1 if verify else 0))
         return out
+
+    def _coldesc_nbytes(self, col):
+        return self._col_nbytes[col] if hasattr(self, "_col_nbytes") else 1 << 22
+
+    def decode_column_varlena(self, col, verify=True):
+        """Decode a varlena (text) directory column -> list of bytes|None."""
+        n = self.nrows
+        offsets = np.zeros(n + 1, np.int64)
+        cap = self._coldesc_nbytes(col) + 16
+        payload = np.zeros(cap, np.uint8)
+        validity = np.zeros(max(n, 1), np.uint8)
+        self.ctx._chk(self.ctx._lib.gx_decode_column_varlena(
+            self.ctx._h, self._t, col, offsets.ctypes.data,
+            payload.ctypes.data, cap, validity.ctypes.data,
+            1 if verify else 0))
+        return [None if not validity[i]
+                else payload[offsets[i]:offsets[i + 1]].tobytes()
+                for i in range(n)]
 
     def decode_column_nullable(self, col, dtype, verify=True):
         """Decode a (possibly NULL-bearing) block-directory column.

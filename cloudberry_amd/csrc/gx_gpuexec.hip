@@ -2185,8 +2185,15 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
             stream_src = inflated.data();
             stream_len = (int64_t) inflated.size();
         }
+        if (cols[c].width < 0 && cols[c].format != 1)
+        {
+            set_err(ctx, "varlena columns need a block directory (format 1)%s", "");
+            delete t;
+            return GX_ERR_INVALID;
+        }
         col.m.width = cols[c].width;
-        col.m.rpb = gx_aocs_rows_per_block(cols[c].width, cols[c].blocksize);
+        col.m.rpb = gx_aocs_rows_per_block(std::max(cols[c].width, 1),
+                                           cols[c].blocksize);
         col.m.nrows = cols[c].nrows;
         col.m.full_block_len = gx_aocs_block_len(cols[c].width, col.m.rpb);
         col.m.nbytes = stream_len;
@@ -2542,6 +2549,82 @@ extern "C" gx_status gx_decode_column(gx_ctx *ctx, const gx_table *t, int colidx
     return GX_OK;
 }
 
+/* Varlena (text) Orig block decode — one THREAD per AO block; the walk
+ * mirrors the reader's VARSIZE_ANY advance + zero-pad skip
+ * (datumstreamblock.h:1509-1545).  pass 0 counts payload bytes per block;
+ * pass 1 (with per-block bases from a host scan) writes payload bytes,
+ * exclusive offsets and optional validity. */
+__global__ void k_decode_varlena(const uint8_t *stream, const gx_blockref *dir,
+                                 int64_t nblocks, int64_t nrows,
+                                 const int64_t *base,
+                                 uint8_t *out_payload, int64_t *out_offsets,
+                                 uint8_t *validity, int64_t *blk_bytes,
+                                 int pass, int *err)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        const uint8_t *c = stream + dir[b].offset + 24;
+        int16_t version = ((const int16_t *) c)[0];
+        int16_t flags = ((const int16_t *) c)[1];
+        int16_t nd = ((const int16_t *) c)[2];
+        int32_t nullsz = ((const int32_t *) c)[2];
+        int32_t sz = ((const int32_t *) c)[3];
+        if (version != 0 || nd != dir[b].rows ||
+            dir[b].first_row + nd > nrows ||
+            ((flags & 1) && validity == nullptr))
+        { atomicOr(err, 1); continue; }
+        const uint8_t *nbmp = c + 16;
+        const uint8_t *p0 = c + 16 + nullsz;
+        const uint8_t *p = p0, *pend = p0 + sz;
+        int64_t w = pass ? base[b] : 0;
+        bool bad = false;
+        for (int32_t r = 0; r < nd; r++)
+        {
+            int64_t row = dir[b].first_row + r;
+            if ((flags & 1) && ((nbmp[r >> 3] >> (r & 7)) & 1))
+            {
+                if (pass)
+                {
+                    validity[row] = 0;
+                    out_offsets[row + 1] = w;
+                }
+                continue;
+            }
+            if (pass && validity) validity[row] = 1;
+            if (p < pend && *p == 0)
+                p = p0 + (((p - p0) + 3) & ~(int64_t) 3);
+            if (p >= pend) { bad = true; break; }
+            int64_t len;
+            const uint8_t *data;
+            if (*p & 1)
+            {
+                len = (int64_t) (*p >> 1) - 1;
+                data = p + 1;
+                p += 1 + len;
+            }
+            else
+            {
+                uint32_t hdr;
+                memcpy(&hdr, p, 4);
+                len = (int64_t) (hdr >> 2) - 4;
+                data = p + 4;
+                p += 4 + len;
+            }
+            if (len < 0 || p > pend) { bad = true; break; }
+            if (pass)
+            {
+                for (int64_t i = 0; i < len; i++)
+                    out_payload[w + i] = data[i];
+                out_offsets[row + 1] = w + len;
+            }
+            w += len;
+        }
+        if (bad) { atomicOr(err, 1); continue; }
+        if (!pass) blk_bytes[b] = w;
+    }
+}
+
 /* NULL-bearing column decode (aocs_getnext with a null bitmap,
  * datumstreamblock.h:1624-1912 null walk): host_validity gets one byte per
  * row (1 = non-null; null datums decode as zero).  Requires a block-
@@ -2587,6 +2670,78 @@ extern "C" gx_status gx_decode_column_nullable(gx_ctx *ctx, const gx_table *t,
     HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 1) { set_err(ctx, "decode: malformed block header%s", ""); return GX_ERR_INVALID; }
+    if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    return GX_OK;
+}
+
+/* Varlena (text) column decode: out_offsets[nrows+1] exclusive offsets into
+ * out_payload; optional validity byte per row (required when the stream has
+ * a NULL bitmap).  Requires a block-directory column with width -1. */
+extern "C" gx_status gx_decode_column_varlena(gx_ctx *ctx, const gx_table *t,
+                                              int colidx,
+                                              int64_t *host_offsets,
+                                              void *host_payload,
+                                              int64_t payload_cap,
+                                              uint8_t *host_validity,
+                                              int verify_checksums)
+{
+    if (!ctx || !t || colidx < 0 || colidx >= (int) t->cols.size()) return GX_ERR_INVALID;
+    const gx_col &c = t->cols[colidx];
+    if (c.format != 1 || c.m.width >= 0)
+    { set_err(ctx, "varlena decode needs a width=-1 directory column%s", ""); return GX_ERR_INVALID; }
+    int64_t n = c.m.nrows;
+    devbuf dbase_b, doff_b, dval_b, dpay_b, derr_b;
+    HIP_CHK(ctx, dbase_b.alloc(std::max<int64_t>(c.nblocks, 1) * 8));
+    HIP_CHK(ctx, doff_b.alloc((n + 1) * 8));
+    HIP_CHK(ctx, dval_b.alloc(std::max<int64_t>(n, 1)));
+    HIP_CHK(ctx, derr_b.alloc(4));
+    int *derr = derr_b.as<int>();
+    HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, ctx->stream));
+    /* pass 0: per-block payload byte counts */
+    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, ctx->stream,
+                       c.dstream, c.ddir, c.nblocks, n,
+                       (const int64_t *) nullptr, (uint8_t *) nullptr,
+                       (int64_t *) nullptr,
+                       host_validity ? dval_b.as<uint8_t>() : nullptr,
+                       dbase_b.as<int64_t>(), 0, derr);
+    std::vector<int64_t> bases(std::max<int64_t>(c.nblocks, 1));
+    HIP_CHK(ctx, hipMemcpyAsync(bases.data(), dbase_b.p, c.nblocks * 8,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    int64_t total = 0;
+    for (int64_t b = 0; b < c.nblocks; b++)
+    {
+        int64_t v = bases[b];
+        bases[b] = total;
+        total += v;
+    }
+    if (total > payload_cap)
+    { set_err(ctx, "varlena payload buffer too small%s", ""); return GX_ERR_INVALID; }
+    HIP_CHK(ctx, dpay_b.alloc(std::max<int64_t>(total, 1)));
+    HIP_CHK(ctx, hipMemcpyAsync(dbase_b.p, bases.data(), c.nblocks * 8,
+                                hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHK(ctx, hipMemsetAsync(doff_b.p, 0, 8, ctx->stream));   /* offsets[0] */
+    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, ctx->stream,
+                       c.dstream, c.ddir, c.nblocks, n,
+                       dbase_b.as<int64_t>(), dpay_b.as<uint8_t>(),
+                       doff_b.as<int64_t>(), dval_b.as<uint8_t>(),
+                       (int64_t *) nullptr, 1, derr);
+    if (verify_checksums)
+        hipLaunchKernelGGL(k_verify_crc_dir, dim3(GRID), dim3(64), 0, ctx->stream,
+                           c.dstream, c.ddir, c.nblocks, derr);
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(host_offsets, doff_b.p, (n + 1) * 8,
+                                hipMemcpyDeviceToHost, ctx->stream));
+    if (total > 0)
+        HIP_CHK(ctx, hipMemcpyAsync(host_payload, dpay_b.p, total,
+                                    hipMemcpyDeviceToHost, ctx->stream));
+    if (host_validity && n > 0)
+        HIP_CHK(ctx, hipMemcpyAsync(host_validity, dval_b.p, n,
+                                    hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr & 1) { set_err(ctx, "varlena decode: malformed block%s", ""); return GX_ERR_INVALID; }
     if (herr & 2) { set_err(ctx, "decode: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
     return GX_OK;
 }

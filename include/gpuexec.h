@@ -133,6 +133,15 @@ gx_status gx_decode_column_nullable(gx_ctx *ctx, const gx_table *t, int col,
 gx_status gx_table_set_visimap(gx_ctx *ctx, gx_table *t,
                                const uint8_t *bitmap, int64_t nbits);
 
+/* Varlena (text) Orig columns (bind with width = -1, format = 1):
+ * offsets[nrows+1] exclusive into payload; validity required when the
+ * stream carries a NULL bitmap. */
+gx_status gx_decode_column_varlena(gx_ctx *ctx, const gx_table *t, int col,
+                                   int64_t *host_offsets, void *host_payload,
+                                   int64_t payload_cap,
+                                   uint8_t *host_validity,
+                                   int verify_checksums);
+
 /* TPC-H Q1 core (BASELINE config 4): GROUP BY returnflag,linestatus with
  * COUNT/SUM over a GX_TPCH_LINEITEM_Q1 table; AVG = sum/count (float8_avg) */
 gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,

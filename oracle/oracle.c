@@ -1796,3 +1796,214 @@ int64_t orc_q3(const orc_customer *c, const orc_orders *o,
     *out = g;
     return ng;
 }
+
+
+/* ======================================================================
+ * Varlena (text-like) Orig codec — restatement of the writer's
+ * variable-length path (datumstreamblock.c:1620-1720) and the reader's
+ * VARSIZE_ANY walk (datumstreamblock.h:1509-1545):
+ *   payload <= 126 B  -> 1-byte short header ((len+1)<<1 | 1), UNALIGNED
+ *   larger            -> zero-pad to 4-byte alignment (att_align_zero),
+ *                        4-byte header ((len+4)<<2) + payload
+ *   the zero padding of a value that then FAILS the capacity check stays
+ *   in the emitted block (the writer pads datump before OrigHasSpace)
+ * Nulls as in fixed-width Orig.  Byte-exact vs the compiled reference
+ * writer (refw_encode_varlena) in tests/test_oracle_cpu.py.
+ * ====================================================================== */
+
+int64_t orc_aocs_encode_varlena(const uint8_t *payload, const int64_t *offsets,
+                                const uint8_t *nulls, int64_t nrows,
+                                int64_t first_rownum, int32_t blocksize,
+                                uint8_t *out, int64_t outcap)
+{
+    int32_t maxdata = blocksize - 24;
+    int64_t off = 0, row = 0;
+    uint8_t *dvals = malloc((size_t) maxdata + 16);
+    uint8_t *nbm = malloc((size_t) ((16383 + 7) >> 3) + 8);
+    if (!dvals || !nbm) { free(dvals); free(nbm); return -1; }
+
+    while (row < nrows)
+    {
+        int32_t nth = 0, always = 0;
+        int has_null = 0;
+        int64_t used = 0;
+        memset(nbm, 0, (size_t) ((16383 + 7) >> 3) + 8);
+        while (row + nth < nrows && nth + 1 < 16383)
+        {
+            int isnull = nulls != NULL && nulls[row + nth] != 0;
+            int32_t nullsize =
+                (isnull || has_null)
+                    ? (int32_t) ((((always + 1 + 7) >> 3) + 7) & ~7) : 0;
+            if (isnull)
+            {
+                if (!(16 + nullsize + used + 0 < maxdata))
+                    break;
+                has_null = 1;
+                nbm[always >> 3] |= (uint8_t) (1u << (always & 7));
+            }
+            else
+            {
+                int64_t len = offsets[row + nth + 1] - offsets[row + nth];
+                int64_t sz, pad = 0;
+                if (len + 1 <= 0x7F)
+                    sz = len + 1;                 /* short form */
+                else
+                {
+                    pad = (-used) & 3;            /* att_align_zero to 4 */
+                    sz = len + 4;
+                }
+                if (!(16 + nullsize + (used + pad) + sz < maxdata))
+                {
+                    /* the writer pads BEFORE the failed check; the pad
+                     * stays in this block (only the 4B-header case pads) */
+                    if (pad && used > 0)
+                    {
+                        memset(dvals + used, 0, (size_t) pad);
+                        used += pad;
+                    }
+                    break;
+                }
+                if (pad)
+                {
+                    memset(dvals + used, 0, (size_t) pad);
+                    used += pad;
+                }
+                if (len + 1 <= 0x7F)
+                {
+                    dvals[used] = (uint8_t) (((len + 1) << 1) | 1);
+                    memcpy(dvals + used + 1, payload + offsets[row + nth], len);
+                    used += sz;
+                }
+                else
+                {
+                    uint32_t hdr = (uint32_t) ((len + 4) << 2);
+                    memcpy(dvals + used, &hdr, 4);
+                    memcpy(dvals + used + 4, payload + offsets[row + nth], len);
+                    used += sz;
+                }
+            }
+            always++;
+            nth++;
+        }
+        if (nth == 0) { free(dvals); free(nbm); return -1; }
+
+        int32_t nullsz = has_null
+            ? (int32_t) ((((nth + 7) >> 3) + 7) & ~7) : 0;
+        int32_t content = 16 + nullsz + (int32_t) used;
+        int64_t blocklen = (24 + content + 7) & ~7LL;
+        if (off + blocklen > outcap) { free(dvals); free(nbm); return -1; }
+        uint8_t *blk = out + off;
+        memset(blk, 0, (size_t) blocklen);
+        uint32_t b03 = (1u << 28) | (1u << 27) | (1u << 24) |
+                       (0x00FFFC00u & ((uint32_t) nth << 10)) |
+                       (((uint32_t) content >> 11) & 0x3FFu);
+        uint32_t b47 = (((uint32_t) content & 0x7FFu) << 21);
+        put_u32le(blk, b03);
+        put_u32le(blk + 4, b47);
+        int64_t frn = first_rownum + row;
+        memcpy(blk + 16, &frn, 8);
+        uint8_t *c = blk + 24;
+        int16_t v16 = 0;  memcpy(c, &v16, 2);
+        v16 = has_null ? 1 : 0; memcpy(c + 2, &v16, 2);
+        v16 = (int16_t) nth; memcpy(c + 4, &v16, 2);
+        v16 = 0; memcpy(c + 6, &v16, 2);
+        int32_t v32 = nullsz; memcpy(c + 8, &v32, 4);
+        v32 = (int32_t) used; memcpy(c + 12, &v32, 4);
+        if (has_null)
+            memcpy(c + 16, nbm, (size_t) ((nth + 7) >> 3));
+        memcpy(c + 16 + nullsz, dvals, (size_t) used);
+        put_u32le(blk + 8, orc_crc32c(0xFFFFFFFFu, blk + 16, blocklen - 16));
+        put_u32le(blk + 12, orc_crc32c(0xFFFFFFFFu, blk, 12));
+        off += blocklen;
+        row += nth;
+    }
+    free(dvals); free(nbm);
+    return off;
+}
+
+/* decode: fills out_payload/out_offsets (exclusive, [nrows+1]) and
+ * optionally out_validity; null rows have length 0.  Returns rows or
+ * -1 / -2 (checksum). */
+int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
+                                int64_t nrows,
+                                uint8_t *out_payload, int64_t payload_cap,
+                                int64_t *out_offsets, uint8_t *out_validity,
+                                int verify_checksums)
+{
+    int64_t off = 0, row = 0, w = 0;
+    out_offsets[0] = 0;
+    while (off + 24 <= nbytes)
+    {
+        uint32_t b03, b47;
+        memcpy(&b03, stream + off, 4);
+        memcpy(&b47, stream + off + 4, 4);
+        if (b03 == 0 && b47 == 0) break;
+        if (((b03 >> 28) & 7) != 1 || !((b03 >> 27) & 1)) return -1;
+        uint32_t rows = (b03 & 0x00FFFC00u) >> 10;
+        uint32_t datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+        if ((b47 & 0x1FFFFFu) != 0) return -1;     /* no bulk compression */
+        int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
+        if (off + blocklen > nbytes) return -1;
+        if (verify_checksums)
+        {
+            uint32_t bc, hc;
+            memcpy(&bc, stream + off + 8, 4);
+            memcpy(&hc, stream + off + 12, 4);
+            if (hc != orc_crc32c(0xFFFFFFFFu, stream + off, 12)) return -2;
+            if (bc != orc_crc32c(0xFFFFFFFFu, stream + off + 16, blocklen - 16)) return -2;
+        }
+        const uint8_t *c = stream + off + 24;
+        int16_t version, flags, nd;
+        int32_t nullsz, sz;
+        memcpy(&version, c, 2);
+        memcpy(&flags, c + 2, 2);
+        memcpy(&nd, c + 4, 2);
+        memcpy(&nullsz, c + 8, 4);
+        memcpy(&sz, c + 12, 4);
+        if (version != 0) return -1;
+        if ((flags & 1) && out_validity == NULL) return -1;
+        if (row + nd > nrows) return -1;
+        const uint8_t *nbmp = c + 16;
+        const uint8_t *p = c + 16 + nullsz;
+        const uint8_t *pend = p + sz;
+        for (int32_t r = 0; r < nd; r++)
+        {
+            if ((flags & 1) && ((nbmp[r >> 3] >> (r & 7)) & 1))
+            {
+                if (out_validity) out_validity[row + r] = 0;
+                out_offsets[row + r + 1] = w;
+                continue;
+            }
+            if (out_validity) out_validity[row + r] = 1;
+            /* skip zero padding (reader: *p == 0 -> align_nominal) */
+            if (p < pend && *p == 0)
+                p = c + 16 + nullsz +
+                    (((p - (c + 16 + nullsz)) + 3) & ~(int64_t) 3);
+            if (p >= pend) return -1;
+            int64_t len;
+            const uint8_t *data;
+            if (*p & 1)
+            {
+                len = (int64_t) (*p >> 1) - 1;
+                data = p + 1;
+                p += 1 + len;
+            }
+            else
+            {
+                uint32_t hdr;
+                if (p + 4 > pend) return -1;
+                memcpy(&hdr, p, 4);
+                len = (int64_t) (hdr >> 2) - 4;
+                data = p + 4;
+                p += 4 + len;
+            }
+            if (len < 0 || p > pend || w + len > payload_cap) return -1;
+            memcpy(out_payload + w, data, (size_t) len);
+            w += len;
+            out_offsets[row + r + 1] = w;
+        }
+        row += nd;
+        off += blocklen;
+    }
+    return row;
+}

@@ -131,6 +131,18 @@ int64_t orc_aocs_decode_nullable(const uint8_t *stream, int64_t nbytes,
                                  int width, void *out_vals,
                                  uint8_t *out_validity, int64_t cap,
                                  int verify_checksums, int codec);
+/* varlena (text-like) Orig columns: payload = concatenated value bytes,
+ * offsets[nrows+1] exclusive; short-form conversion and alignment follow
+ * the reference writer exactly (byte-exact, see tests). */
+int64_t orc_aocs_encode_varlena(const uint8_t *payload, const int64_t *offsets,
+                                const uint8_t *nulls, int64_t nrows,
+                                int64_t first_rownum, int32_t blocksize,
+                                uint8_t *out, int64_t outcap);
+int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
+                                int64_t nrows,
+                                uint8_t *out_payload, int64_t payload_cap,
+                                int64_t *out_offsets, uint8_t *out_validity,
+                                int verify_checksums);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

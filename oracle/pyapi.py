@@ -107,6 +107,14 @@ def _load():
     lib.orc_aocs_decode_nullable.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int, ctypes.c_int]
+    lib.orc_aocs_encode_varlena.restype = ctypes.c_int64
+    lib.orc_aocs_encode_varlena.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+        ctypes.c_int64, ctypes.c_int32, ctypes.c_void_p, ctypes.c_int64]
+    lib.orc_aocs_decode_varlena.restype = ctypes.c_int64
+    lib.orc_aocs_decode_varlena.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
     lib.orc_aocs_encode_zstd.restype = ctypes.c_int64
     lib.orc_aocs_encode_zstd.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                          ctypes.c_int64, ctypes.c_int64,
@@ -432,6 +440,104 @@ def aocs_decode_nullable(stream, width, nrows, dtype, verify=1, codec=1):
                                        nrows, verify, codec)
     assert got == nrows, got
     return out, validity
+
+
+def _varlena_args(strings, nulls):
+    payload = b"".join(b"" if (nulls is not None and nulls[i]) else s
+                       for i, s in enumerate(strings))
+    offsets = np.zeros(len(strings) + 1, np.int64)
+    w = 0
+    for i, s in enumerate(strings):
+        if nulls is None or not nulls[i]:
+            w += len(s)
+        offsets[i + 1] = w
+    pay = np.frombuffer(payload, np.uint8).copy() if payload else np.zeros(1, np.uint8)
+    nl = None
+    nptr = 0
+    if nulls is not None:
+        nl = np.ascontiguousarray(nulls, np.uint8)
+        nptr = nl.ctypes.data
+    return pay, offsets, nl, nptr
+
+
+def aocs_encode_varlena(strings, nulls=None, blocksize=32768):
+    """Orig-format varlena (text) stream from a list of bytes values."""
+    pay, offsets, nl, nptr = _varlena_args(strings, nulls)
+    cap = int(offsets[-1]) + 16 * len(strings) + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_varlena(pay.ctypes.data, offsets.ctypes.data,
+                                      nptr, len(strings), 1, blocksize,
+                                      buf.ctypes.data, cap)
+    assert got >= 0, got
+    return buf[:got].tobytes()
+
+
+def aocs_decode_varlena(stream, nrows, verify=1):
+    """Decode an Orig varlena stream -> (list of bytes|None)."""
+    cap = len(stream) + 16
+    payload = np.zeros(cap, np.uint8)
+    offsets = np.zeros(nrows + 1, np.int64)
+    validity = np.zeros(nrows, np.uint8)
+    got = lib.orc_aocs_decode_varlena(stream, len(stream), nrows,
+                                      payload.ctypes.data, cap,
+                                      offsets.ctypes.data,
+                                      validity.ctypes.data, verify)
+    assert got == nrows, got
+    out = []
+    for i in range(nrows):
+        if not validity[i]:
+            out.append(None)
+        else:
+            out.append(payload[offsets[i]:offsets[i + 1]].tobytes())
+    return out
+
+
+def ref_writer_varlena_stream(strings, nulls=None, blocksize=32768):
+    """REAL reference-writer varlena stream wrapped in our AO envelope."""
+    w = ref_writer()
+    if w is None:
+        return None
+    if not hasattr(w, "_varlena_decl"):
+        w.refw_encode_varlena.restype = ctypes.c_int
+        w.refw_encode_varlena.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int32,
+            ctypes.c_void_p, ctypes.c_int64,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
+        w._varlena_decl = True
+    pay, offsets, nl, nptr = _varlena_args(strings, nulls)
+    cap = int(offsets[-1]) + 16 * len(strings) + (1 << 21)
+    out = np.zeros(cap, np.uint8)
+    lens = np.zeros(65536, np.int32)
+    rows = np.zeros(65536, np.int32)
+    nb = w.refw_encode_varlena(pay.ctypes.data, offsets.ctypes.data, nptr,
+                               len(strings), 0, 0, blocksize - 24,
+                               out.ctypes.data, cap,
+                               lens.ctypes.data, rows.ctypes.data, 65536)
+    assert nb > 0, nb
+    stream = bytearray()
+    off = 0
+    frn = 1
+    for b in range(nb):
+        content = out[off:off + lens[b]].tobytes()
+        off += lens[b]
+        logical = int(rows[b])
+        clen = len(content)
+        b03 = (1 << 28) | (1 << 27) | (1 << 24) | (logical << 10) | (clen >> 11)
+        b47 = (clen & 0x7FF) << 21
+        blocklen = (24 + clen + 7) & ~7
+        blk = bytearray(blocklen)
+        blk[0:4] = b03.to_bytes(4, "little")
+        blk[4:8] = b47.to_bytes(4, "little")
+        blk[16:24] = frn.to_bytes(8, "little")
+        blk[24:24 + clen] = content
+        bc = lib.orc_crc32c(0xFFFFFFFF, bytes(blk[16:]), blocklen - 16)
+        hc = lib.orc_crc32c(0xFFFFFFFF, bytes(blk[0:8]) + bc.to_bytes(4, "little"), 12)
+        blk[8:12] = bc.to_bytes(4, "little")
+        blk[12:16] = hc.to_bytes(4, "little")
+        stream += blk
+        frn += logical
+    return bytes(stream)
 
 
 def aocs_encode_zlib(vals, level=6):

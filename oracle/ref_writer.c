@@ -128,9 +128,23 @@ pg_sprintf(char *str, const char *fmt,...)
 void
 varattrib_untoast_ptr_len(Datum d, char **datastart, int *len, void **tofree)
 {
-    (void) d; (void) datastart; (void) len; (void) tofree;
-    fprintf(stderr, "STUB varattrib_untoast_ptr_len called\n");
-    abort();                    /* varlena path unused (fixed-width only) */
+    /* restatement of detoast.c:652-705 for PLAIN (non-TOASTed) datums —
+     * the only kind this harness feeds */
+    struct varlena *va = (struct varlena *) DatumGetPointer(d);
+
+    *tofree = NULL;
+    if (va == NULL)
+        abort();
+    if (VARATT_IS_SHORT(va))
+    {
+        *len = VARSIZE_SHORT(va) - VARHDRSZ_SHORT;
+        *datastart = VARDATA_SHORT(va);
+        return;
+    }
+    if (VARATT_IS_EXTENDED(va))
+        abort();                /* TOAST pointers never reach this harness */
+    *datastart = VARDATA(va);
+    *len = VARSIZE(va) - VARHDRSZ;
 }
 
 void
@@ -252,4 +266,110 @@ refw_encode(const void *vals, int width, int64 nrows,
     return refw_encode_nulls(vals, NULL, width, nrows,
                              version, rle, delta, maxDataBlockSize,
                              out, outcap, block_lens, block_rows, max_blocks);
+}
+
+/*
+ * Varlena (text-like) columns: feed 4-byte-header varlena datums through
+ * the reference writer (typid 25, typstorage 'x', align 'i'); the writer
+ * itself converts to short form / aligns (datumstreamblock.c:1620-1720).
+ * payload = concatenated value bytes, offsets[nrows+1] exclusive.
+ */
+int
+refw_encode_varlena(const uint8 *payload, const int64 *offsets,
+                    const uint8 *nulls, int64 nrows,
+                    int version, int rle, int32 maxDataBlockSize,
+                    uint8 *out, int64 outcap,
+                    int32 *block_lens, int32 *block_rows, int max_blocks)
+{
+    DatumStreamTypeInfo ti;
+    DatumStreamBlockWrite dsw;
+    RelFileLocator loc;
+    int64 off = 0;
+    int nblocks = 0;
+    int32 rows_in_block = 0;
+    uint8 *scratch = malloc(16 * 1024 * 1024);
+
+    if (!scratch)
+        return -1;
+    memset(&ti, 0, sizeof(ti));
+    ti.datumlen = -1;
+    ti.typid = 25;              /* text */
+    ti.typstorage = 'x';
+    ti.align = 'i';
+    ti.byval = false;
+
+    memset(&loc, 0, sizeof(loc));
+    memset(&dsw, 0, sizeof(dsw));
+    DatumStreamBlockWrite_Init(&dsw, &ti,
+                               version == 0 ? DatumStreamVersion_Original
+                                            : DatumStreamVersion_Dense_Enhanced,
+                               rle != 0, /* delta */ false,
+                               version == 0 ? MAXDATUM_PER_AOCS_ORIG_BLOCK
+                                            : INITIALDATUM_PER_AOCS_DENSE_BLOCK,
+                               version == 0 ? MAXDATUM_PER_AOCS_ORIG_BLOCK
+                                            : MAXDATUM_PER_AOCS_DENSE_BLOCK,
+                               maxDataBlockSize,
+                               cb_zero, NULL, cb_zero, NULL, &loc);
+    DatumStreamBlockWrite_GetReady(&dsw);
+
+    for (int64 i = 0; i < nrows; i++)
+    {
+        Datum d = 0;
+        bool isnull = (nulls != NULL && nulls[i] != 0);
+        void *tofree = NULL;
+
+        if (!isnull)
+        {
+            int64 len = offsets[i + 1] - offsets[i];
+
+            if (len + 4 > 16 * 1024 * 1024)
+            {
+                free(scratch);
+                return -1;
+            }
+            SET_VARSIZE(scratch, len + 4);
+            memcpy(scratch + 4, payload + offsets[i], len);
+            d = PointerGetDatum(scratch);
+        }
+        if (DatumStreamBlockWrite_Put(&dsw, d, isnull, &tofree) < 0)
+        {
+            int64 len;
+
+            if (nblocks >= max_blocks || off + maxDataBlockSize > outcap)
+            {
+                free(scratch);
+                return -1;
+            }
+            len = DatumStreamBlockWrite_Block(&dsw, out + off, &loc);
+            block_lens[nblocks] = (int32) len;
+            block_rows[nblocks] = rows_in_block;
+            nblocks++;
+            off += len;
+            rows_in_block = 0;
+            DatumStreamBlockWrite_GetReady(&dsw);
+            if (DatumStreamBlockWrite_Put(&dsw, d, isnull, &tofree) < 0)
+            {
+                free(scratch);
+                return -1;
+            }
+        }
+        rows_in_block++;
+    }
+    if (rows_in_block > 0)
+    {
+        int64 len;
+
+        if (nblocks >= max_blocks || off + maxDataBlockSize > outcap)
+        {
+            free(scratch);
+            return -1;
+        }
+        len = DatumStreamBlockWrite_Block(&dsw, out + off, &loc);
+        block_lens[nblocks] = (int32) len;
+        block_rows[nblocks] = rows_in_block;
+        nblocks++;
+        off += len;
+    }
+    free(scratch);
+    return nblocks;
 }
