@@ -364,16 +364,25 @@ class Table:
         return self._col_nbytes[col] if hasattr(self, "_col_nbytes") else 1 << 22
 
     def decode_column_varlena(self, col, verify=True):
-        """Decode a varlena (text) directory column -> list of bytes|None."""
+        """Decode a varlena (text) directory column -> list of bytes|None.
+        RLE streams expand past the stream size; grow the buffer on demand."""
         n = self.nrows
-        offsets = np.zeros(n + 1, np.int64)
         cap = self._coldesc_nbytes(col) + 16
-        payload = np.zeros(cap, np.uint8)
-        validity = np.zeros(max(n, 1), np.uint8)
-        self.ctx._chk(self.ctx._lib.gx_decode_column_varlena(
-            self.ctx._h, self._t, col, offsets.ctypes.data,
-            payload.ctypes.data, cap, validity.ctypes.data,
-            1 if verify else 0))
+        while True:
+            offsets = np.zeros(n + 1, np.int64)
+            payload = np.zeros(cap, np.uint8)
+            validity = np.zeros(max(n, 1), np.uint8)
+            try:
+                self.ctx._chk(self.ctx._lib.gx_decode_column_varlena(
+                    self.ctx._h, self._t, col, offsets.ctypes.data,
+                    payload.ctypes.data, cap, validity.ctypes.data,
+                    1 if verify else 0))
+            except GxError as e:
+                if "too small" in str(e):
+                    cap *= 4
+                    continue
+                raise
+            break
         return [None if not validity[i]
                 else payload[offsets[i]:offsets[i + 1]].tobytes()
                 for i in range(n)]

@@ -2567,31 +2567,73 @@ __global__ void k_decode_varlena(const uint8_t *stream, const gx_blockref *dir,
         const uint8_t *c = stream + dir[b].offset + 24;
         int16_t version = ((const int16_t *) c)[0];
         int16_t flags = ((const int16_t *) c)[1];
-        int16_t nd = ((const int16_t *) c)[2];
-        int32_t nullsz = ((const int32_t *) c)[2];
-        int32_t sz = ((const int32_t *) c)[3];
-        if (version != 0 || nd != dir[b].rows ||
-            dir[b].first_row + nd > nrows ||
-            ((flags & 1) && validity == nullptr))
+        int32_t logical = dir[b].rows;
+        if ((version != 0 && version != 1 && version != 2) ||
+            dir[b].first_row + logical > nrows ||
+            ((flags & 1) && validity == nullptr) || (flags & 4))
         { atomicOr(err, 1); continue; }
-        const uint8_t *nbmp = c + 16;
-        const uint8_t *p0 = c + 16 + nullsz;
-        const uint8_t *p = p0, *pend = p0 + sz;
-        int64_t w = pass ? base[b] : 0;
-        bool bad = false;
-        for (int32_t r = 0; r < nd; r++)
+        bool rle = false;
+        const uint8_t *nbmp = nullptr, *bmp = nullptr, *cnts = nullptr;
+        int32_t bmbits = 0, csize = 0, nullbits = 0, psize = 0;
+        const uint8_t *p0;
+        if (version == 0)
         {
-            int64_t row = dir[b].first_row + r;
-            if ((flags & 1) && ((nbmp[r >> 3] >> (r & 7)) & 1))
+            int16_t nd = ((const int16_t *) c)[2];
+            int32_t nullsz = ((const int32_t *) c)[2];
+            psize = ((const int32_t *) c)[3];
+            if (nd != logical) { atomicOr(err, 1); continue; }
+            nullbits = (flags & 1) ? logical : 0;
+            nbmp = c + 16;
+            p0 = c + 16 + nullsz;
+        }
+        else
+        {
+            /* Dense(±RLE) varlena: same section order as fixed-width */
+            int32_t hlogical = ((const int32_t *) c)[1];
+            psize = ((const int32_t *) c)[3];
+            if (hlogical != logical) { atomicOr(err, 1); continue; }
+            rle = (flags & 2) != 0;
+            const uint8_t *q = c + 16;
+            nullbits = (flags & 1) ? logical : 0;
+            if (rle)
             {
-                if (pass)
-                {
-                    validity[row] = 0;
-                    out_offsets[row + 1] = w;
-                }
-                continue;
+                int32_t norepeats = ((const int32_t *) q)[0];
+                bmbits = ((const int32_t *) q)[1];
+                csize = ((const int32_t *) q)[3];
+                if (flags & 1) nullbits = norepeats;
+                else if (norepeats != 0) { atomicOr(err, 1); continue; }
+                q += 16;
             }
-            if (pass && validity) validity[row] = 1;
+            if (flags & 1) { nbmp = q; q += (nullbits + 7) >> 3; }
+            if (rle) { bmp = q; q += (bmbits + 7) >> 3; cnts = q; q += csize; }
+            int32_t hdr = (int32_t) (q - c);
+            p0 = c + ((hdr + 7) & ~7);
+        }
+        const uint8_t *p = p0, *pend = p0 + psize;
+        int64_t w = pass ? base[b] : 0;
+        int64_t out = 0;
+        int32_t item = 0, coff = 0, npos = 0;
+        bool bad = false;
+        while (out < logical)
+        {
+            int64_t row = dir[b].first_row + out;
+            if (nbmp != nullptr && nullbits > 0)
+            {
+                if (npos >= nullbits) { bad = true; break; }
+                int nb = (nbmp[npos >> 3] >> (npos & 7)) & 1;
+                npos++;
+                if (nb)
+                {
+                    if (pass)
+                    {
+                        validity[row] = 0;
+                        out_offsets[row + 1] = w;
+                    }
+                    out++;
+                    continue;
+                }
+            }
+            if (rle && item >= bmbits) { bad = true; break; }
             if (p < pend && *p == 0)
                 p = p0 + (((p - p0) + 3) & ~(int64_t) 3);
             if (p >= pend) { bad = true; break; }
@@ -2612,15 +2654,32 @@ __global__ void k_decode_varlena(const uint8_t *stream, const gx_blockref *dir,
                 p += 4 + len;
             }
             if (len < 0 || p > pend) { bad = true; break; }
-            if (pass)
+            int64_t reps = 1;
+            if (rle && (bmp[item >> 3] & (1u << (item & 7))))
             {
-                for (int64_t i = 0; i < len; i++)
-                    out_payload[w + i] = data[i];
-                out_offsets[row + 1] = w + len;
+                int32_t n = (cnts[coff] >> 6) + 1;
+                uint32_t v = cnts[coff] & 0x3F;
+                for (int32_t i = 1; i < n; i++) v = (v << 8) | cnts[coff + i];
+                coff += n;
+                reps += v;
             }
-            w += len;
+            if (out + reps > logical) { bad = true; break; }
+            for (int64_t rr = 0; rr < reps; rr++)
+            {
+                if (pass)
+                {
+                    for (int64_t i = 0; i < len; i++)
+                        out_payload[w + i] = data[i];
+                    if (validity) validity[dir[b].first_row + out] = 1;
+                    out_offsets[dir[b].first_row + out + 1] = w + len;
+                }
+                w += len;
+                out++;
+            }
+            item++;
         }
-        if (bad) { atomicOr(err, 1); continue; }
+        if (bad || (rle && (coff != csize || item != bmbits)))
+        { atomicOr(err, 1); continue; }
         if (!pass) blk_bytes[b] = w;
     }
 }

@@ -737,6 +737,10 @@ typedef struct {
     dbm_t cbm;                           /* compress bitmap */
     int32_t *repeats;
     int32_t nrepeats, repeats_size;      /* repeats_size = FINALIZED runs only */
+    /* varlena RLE: stored payload of the last item (offset into
+     * datum_buffer) — equality is on PAYLOAD bytes (:3240-3267) */
+    int64_t vl_last_off;
+    int32_t vl_last_len;
     /* DELTA state */
     int delta_has, not_first;
     uint64_t compare_item;
@@ -1178,6 +1182,163 @@ int64_t orc_aocs_encode_rle_delta_nulls(const void *vals, const uint8_t *nulls,
 {
     return dense_encode(vals, nulls, width, nrows, first_rownum, blocksize,
                         1, delta, out, outcap);
+}
+
+/* Dense_Enhanced varlena put (:3160-3380): RLE equality on payload bytes,
+ * short-form / aligned 4-byte storage as in Orig, same capacity machinery
+ * as the fixed-width dense writer; no delta for varlena. */
+static int dwr_put_varlena(dwr_t *w, const uint8_t *payload, int64_t len,
+                           int isnull)
+{
+    int have_prev;
+    if (isnull)
+    {
+        if (!dwr_has_space_null(w))
+            return -1;
+        if (!w->has_null)
+        {
+            w->has_null = 1;
+            dbm_zerofill(&w->nbm, w->always);
+        }
+        dbm_add(&w->nbm, 1);
+        w->always++;
+        if (w->rle_want)
+        {
+            if (w->last_repeated)
+                dwr_finalize_repeat(w);
+            w->last_valid = 0;
+        }
+        w->nth++;
+        return 0;
+    }
+    have_prev = w->rle_want && w->last_valid;
+    if (w->last_repeated && w->repeats[w->nrepeats - 1] >= DWR_MAXREPEAT)
+        dwr_finalize_repeat(w);
+    else if (have_prev)
+    {
+        int eq = (int64_t) w->vl_last_len == len &&
+                 memcmp(w->datum_buffer + w->vl_last_off, payload,
+                        (size_t) len) == 0;
+        if (eq)
+        {
+            if (!dwr_has_space_repeat(w, !w->last_repeated))
+                return -1;
+            dwr_incr_repeated(w);
+            return 0;
+        }
+        if (w->last_repeated)
+            dwr_finalize_repeat(w);
+    }
+    int64_t sz, pad = 0;
+    if (len + 1 <= 0x7F)
+        sz = len + 1;
+    else
+    {
+        pad = (-w->datum_used) & 3;          /* att_align_zero, pad stays
+                                                even if the check fails */
+        sz = len + 4;
+    }
+    if (pad)
+    {
+        memset(w->datum_buffer + w->datum_used, 0, (size_t) pad);
+        w->datum_used += pad;
+    }
+    if (!dwr_has_space_item(w, (int32_t) sz))
+        return (int) -sz;
+    int64_t data_off;
+    if (len + 1 <= 0x7F)
+    {
+        w->datum_buffer[w->datum_used] = (uint8_t) (((len + 1) << 1) | 1);
+        memcpy(w->datum_buffer + w->datum_used + 1, payload, (size_t) len);
+        data_off = w->datum_used + 1;
+    }
+    else
+    {
+        uint32_t hdr = (uint32_t) ((len + 4) << 2);
+        memcpy(w->datum_buffer + w->datum_used, &hdr, 4);
+        memcpy(w->datum_buffer + w->datum_used + 4, payload, (size_t) len);
+        data_off = w->datum_used + 4;
+    }
+    w->datum_used += sz;
+    /* DenseIncrItem */
+    if (w->has_null)
+        dbm_add(&w->nbm, 0);
+    w->always++;
+    if (w->last_repeated)
+        dwr_finalize_repeat(w);
+    if (w->rle_want)
+    {
+        w->vl_last_off = data_off;
+        w->vl_last_len = (int32_t) len;
+        w->last_valid = 1;
+    }
+    if (w->rle_has)
+        dbm_add(&w->cbm, 0);
+    w->nth++;
+    w->phys++;
+    return (int) sz;
+}
+
+int64_t orc_aocs_encode_varlena_rle(const uint8_t *payload,
+                                    const int64_t *offsets,
+                                    const uint8_t *nulls, int64_t nrows,
+                                    int64_t first_rownum, int32_t blocksize,
+                                    uint8_t *out, int64_t outcap)
+{
+    dwr_t w;
+    int64_t off = 0, emitted = 0;
+    size_t scratch;
+
+    memset(&w, 0, sizeof(w));
+    w.width = -1;
+    w.rle_want = 1;
+    w.delta_want = 0;
+    w.maxdata = blocksize - 32;
+    scratch = (size_t) blocksize * 2 + 64;
+    w.datum_buffer = malloc((size_t) w.maxdata + 16);
+    w.cbm.buf = malloc(scratch);
+    w.dbm.buf = malloc(scratch);
+    w.nbm.buf = malloc(scratch);
+    w.repeats = malloc(scratch * sizeof(int32_t));
+    w.deltas = malloc(scratch * sizeof(int64_t));
+    w.dsigns = malloc(scratch);
+    if (!w.datum_buffer || !w.cbm.buf || !w.dbm.buf || !w.nbm.buf ||
+        !w.repeats || !w.deltas || !w.dsigns)
+        goto fail;
+    dwr_getready(&w);
+    for (int64_t i = 0; i < nrows; i++)
+    {
+        int isnull = nulls != NULL && nulls[i] != 0;
+        const uint8_t *p = payload + offsets[i];
+        int64_t len = offsets[i + 1] - offsets[i];
+        if (dwr_put_varlena(&w, p, isnull ? 0 : len, isnull) < 0)
+        {
+            int64_t bl = dwr_block(&w, first_rownum + emitted,
+                                   out + off, outcap - off);
+            if (bl < 0)
+                goto fail;
+            off += bl;
+            emitted += w.nth;
+            dwr_getready(&w);
+            if (dwr_put_varlena(&w, p, isnull ? 0 : len, isnull) < 0)
+                goto fail;
+        }
+    }
+    if (w.nth > 0)
+    {
+        int64_t bl = dwr_block(&w, first_rownum + emitted,
+                               out + off, outcap - off);
+        if (bl < 0)
+            goto fail;
+        off += bl;
+    }
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf); free(w.nbm.buf);
+    free(w.repeats); free(w.deltas); free(w.dsigns);
+    return off;
+fail:
+    free(w.datum_buffer); free(w.cbm.buf); free(w.dbm.buf); free(w.nbm.buf);
+    free(w.repeats); free(w.deltas); free(w.dsigns);
+    return -1;
 }
 
 /* Original-version writer with NULL support — PutOrig (:1569-1770) +
@@ -1811,6 +1972,111 @@ int64_t orc_q3(const orc_customer *c, const orc_orders *o,
  * writer (refw_encode_varlena) in tests/test_oracle_cpu.py.
  * ====================================================================== */
 
+/* Dense_Enhanced varlena content walk: null bits per non-repeat slot,
+ * compress bit per physical item, varlena datums with zero-pad alignment.
+ * Writes into the caller's running payload cursor *w_io. */
+static int64_t decode_dense_varlena_content(const uint8_t *c, int64_t logical,
+                                            uint8_t *out_payload,
+                                            int64_t payload_cap,
+                                            int64_t *out_offsets,
+                                            uint8_t *out_validity,
+                                            int64_t row0, int64_t *w_io)
+{
+    int16_t flags;
+    int32_t hlogical, phys, psize;
+    memcpy(&flags, c + 2, 2);
+    memcpy(&hlogical, c + 4, 4);
+    memcpy(&phys, c + 8, 4);
+    memcpy(&psize, c + 12, 4);
+    if (hlogical != logical) return -1;
+    int has_null = (flags & 0x1) != 0;
+    int rle = (flags & 0x2) != 0;
+    if (flags & 0x4) return -1;              /* no delta for varlena */
+    if (has_null && out_validity == NULL) return -1;
+    const uint8_t *p = c + 16;
+    int32_t bmbits = 0, csize = 0;
+    int32_t nullbits = has_null ? (int32_t) logical : 0;
+    if (rle)
+    {
+        int32_t norepeats;
+        memcpy(&norepeats, p, 4);
+        memcpy(&bmbits, p + 4, 4);
+        memcpy(&csize, p + 12, 4);
+        if (has_null) nullbits = norepeats;
+        else if (norepeats != 0) return -1;
+        p += 16;
+    }
+    const uint8_t *nbmp = NULL, *bmp = NULL, *cnts = NULL;
+    if (has_null) { nbmp = p; p += (nullbits + 7) >> 3; }
+    if (rle) { bmp = p; p += (bmbits + 7) >> 3; cnts = p; p += csize; }
+    int32_t hdr = (int32_t) (p - c);
+    const uint8_t *d0 = c + ((hdr + 7) & ~7);
+    const uint8_t *dp = d0, *dend = d0 + psize;
+
+    int64_t lw = *w_io, out = 0;
+    int32_t item = 0, coff = 0, npos = 0;
+    while (out < logical)
+    {
+        if (has_null)
+        {
+            if (npos >= nullbits) return -1;
+            int nb = (nbmp[npos >> 3] >> (npos & 7)) & 1;
+            npos++;
+            if (nb)
+            {
+                out_validity[row0 + out] = 0;
+                out_offsets[row0 + out + 1] = lw;
+                out++;
+                continue;
+            }
+        }
+        if (rle && item >= bmbits) return -1;
+        if (dp < dend && *dp == 0)
+            dp = d0 + (((dp - d0) + 3) & ~(int64_t) 3);
+        if (dp >= dend) return -1;
+        int64_t len;
+        const uint8_t *data;
+        if (*dp & 1)
+        {
+            len = (int64_t) (*dp >> 1) - 1;
+            data = dp + 1;
+            dp += 1 + len;
+        }
+        else
+        {
+            uint32_t hdr4;
+            if (dp + 4 > dend) return -1;
+            memcpy(&hdr4, dp, 4);
+            len = (int64_t) (hdr4 >> 2) - 4;
+            data = dp + 4;
+            dp += 4 + len;
+        }
+        if (len < 0 || dp > dend) return -1;
+        int64_t reps = 1;
+        if (rle && (bmp[item >> 3] & (1u << (item & 7))))
+        {
+            int32_t vlen, v = varint_decode(cnts + coff, &vlen);
+            coff += vlen;
+            reps += v;
+        }
+        if (out + reps > logical) return -1;
+        if (lw + reps * len > payload_cap) return -3;
+        for (int64_t r = 0; r < reps; r++)
+        {
+            memcpy(out_payload + lw, data, (size_t) len);
+            lw += len;
+            if (out_validity) out_validity[row0 + out] = 1;
+            out_offsets[row0 + out + 1] = lw;
+            out++;
+        }
+        item++;
+    }
+    if (rle && (coff != csize || item != bmbits)) return -1;
+    if (has_null && npos != nullbits) return -1;
+    *w_io = lw;
+    return out;
+}
+
 int64_t orc_aocs_encode_varlena(const uint8_t *payload, const int64_t *offsets,
                                 const uint8_t *nulls, int64_t nrows,
                                 int64_t first_rownum, int32_t blocksize,
@@ -1938,10 +2204,22 @@ int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
         memcpy(&b03, stream + off, 4);
         memcpy(&b47, stream + off + 4, 4);
         if (b03 == 0 && b47 == 0) break;
-        if (((b03 >> 28) & 7) != 1 || !((b03 >> 27) & 1)) return -1;
-        uint32_t rows = (b03 & 0x00FFFC00u) >> 10;
-        uint32_t datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
-        if ((b47 & 0x1FFFFFu) != 0) return -1;     /* no bulk compression */
+        uint32_t kind = (b03 >> 28) & 7;
+        uint32_t rows, datalen;
+        if (kind == 1)
+        {
+            rows = (b03 & 0x00FFFC00u) >> 10;
+            datalen = ((b03 & 0x3FFu) << 11) | ((b47 & 0xFFE00000u) >> 21);
+            if ((b47 & 0x1FFFFFu) != 0) return -1; /* no bulk compression */
+        }
+        else if (kind == 3)                        /* NonBulkDense (RLE) */
+        {
+            rows = b47 & 0x3FFFFFFFu;
+            datalen = b03 & 0x1FFFFFu;
+        }
+        else
+            return -1;
+        if (!((b03 >> 27) & 1)) return -1;
         int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
         if (off + blocklen > nbytes) return -1;
         if (verify_checksums)
@@ -1960,6 +2238,17 @@ int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
         memcpy(&nd, c + 4, 2);
         memcpy(&nullsz, c + 8, 4);
         memcpy(&sz, c + 12, 4);
+        if (version == 1 || version == 2)
+        {
+            int64_t got = decode_dense_varlena_content(
+                c, (int64_t) rows, out_payload, payload_cap,
+                out_offsets, out_validity, row, &w);
+            if (got == -3) return -3;
+            if (got < 0 || got != (int64_t) rows) return -1;
+            row += got;
+            off += blocklen;
+            continue;
+        }
         if (version != 0) return -1;
         if ((flags & 1) && out_validity == NULL) return -1;
         if (row + nd > nrows) return -1;
@@ -1997,7 +2286,8 @@ int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
                 data = p + 4;
                 p += 4 + len;
             }
-            if (len < 0 || p > pend || w + len > payload_cap) return -1;
+            if (len < 0 || p > pend) return -1;
+            if (w + len > payload_cap) return -3;
             memcpy(out_payload + w, data, (size_t) len);
             w += len;
             out_offsets[row + r + 1] = w;

@@ -143,6 +143,12 @@ int64_t orc_aocs_decode_varlena(const uint8_t *stream, int64_t nbytes,
                                 uint8_t *out_payload, int64_t payload_cap,
                                 int64_t *out_offsets, uint8_t *out_validity,
                                 int verify_checksums);
+/* Dense_Enhanced rle_type varlena (RLE on repeated payloads) */
+int64_t orc_aocs_encode_varlena_rle(const uint8_t *payload,
+                                    const int64_t *offsets,
+                                    const uint8_t *nulls, int64_t nrows,
+                                    int64_t first_rownum, int32_t blocksize,
+                                    uint8_t *out, int64_t outcap);
 
 /* ---- Q3 pipeline (reference executor semantics) ---- */
 typedef struct {

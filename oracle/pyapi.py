@@ -115,6 +115,8 @@ def _load():
     lib.orc_aocs_decode_varlena.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
         ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
+    lib.orc_aocs_encode_varlena_rle.restype = ctypes.c_int64
+    lib.orc_aocs_encode_varlena_rle.argtypes = lib.orc_aocs_encode_varlena.argtypes
     lib.orc_aocs_encode_zstd.restype = ctypes.c_int64
     lib.orc_aocs_encode_zstd.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                          ctypes.c_int64, ctypes.c_int64,
@@ -472,16 +474,34 @@ def aocs_encode_varlena(strings, nulls=None, blocksize=32768):
     return buf[:got].tobytes()
 
 
+def aocs_encode_varlena_rle(strings, nulls=None, blocksize=32768):
+    """Dense_Enhanced rle_type varlena stream (RLE on repeated payloads)."""
+    pay, offsets, nl, nptr = _varlena_args(strings, nulls)
+    cap = int(offsets[-1]) + 16 * len(strings) + (1 << 20)
+    buf = np.zeros(cap, np.uint8)
+    got = lib.orc_aocs_encode_varlena_rle(pay.ctypes.data, offsets.ctypes.data,
+                                          nptr, len(strings), 1, blocksize,
+                                          buf.ctypes.data, cap)
+    assert got >= 0, got
+    return buf[:got].tobytes()
+
+
 def aocs_decode_varlena(stream, nrows, verify=1):
-    """Decode an Orig varlena stream -> (list of bytes|None)."""
+    """Decode an Orig/Dense varlena stream -> (list of bytes|None).
+    RLE streams can expand well past the stream size; grow on -3."""
     cap = len(stream) + 16
-    payload = np.zeros(cap, np.uint8)
-    offsets = np.zeros(nrows + 1, np.int64)
-    validity = np.zeros(nrows, np.uint8)
-    got = lib.orc_aocs_decode_varlena(stream, len(stream), nrows,
-                                      payload.ctypes.data, cap,
-                                      offsets.ctypes.data,
-                                      validity.ctypes.data, verify)
+    while True:
+        payload = np.zeros(cap, np.uint8)
+        offsets = np.zeros(nrows + 1, np.int64)
+        validity = np.zeros(nrows, np.uint8)
+        got = lib.orc_aocs_decode_varlena(stream, len(stream), nrows,
+                                          payload.ctypes.data, cap,
+                                          offsets.ctypes.data,
+                                          validity.ctypes.data, verify)
+        if got == -3:
+            cap *= 4
+            continue
+        break
     assert got == nrows, got
     out = []
     for i in range(nrows):
@@ -492,7 +512,8 @@ def aocs_decode_varlena(stream, nrows, verify=1):
     return out
 
 
-def ref_writer_varlena_stream(strings, nulls=None, blocksize=32768):
+def ref_writer_varlena_stream(strings, nulls=None, blocksize=32768,
+                              version=0, rle=0):
     """REAL reference-writer varlena stream wrapped in our AO envelope."""
     w = ref_writer()
     if w is None:
@@ -511,7 +532,8 @@ def ref_writer_varlena_stream(strings, nulls=None, blocksize=32768):
     lens = np.zeros(65536, np.int32)
     rows = np.zeros(65536, np.int32)
     nb = w.refw_encode_varlena(pay.ctypes.data, offsets.ctypes.data, nptr,
-                               len(strings), 0, 0, blocksize - 24,
+                               len(strings), version, rle,
+                               blocksize - (24 if version == 0 else 32),
                                out.ctypes.data, cap,
                                lens.ctypes.data, rows.ctypes.data, 65536)
     assert nb > 0, nb
@@ -523,8 +545,12 @@ def ref_writer_varlena_stream(strings, nulls=None, blocksize=32768):
         off += lens[b]
         logical = int(rows[b])
         clen = len(content)
-        b03 = (1 << 28) | (1 << 27) | (1 << 24) | (logical << 10) | (clen >> 11)
-        b47 = (clen & 0x7FF) << 21
+        if logical <= 16383:
+            b03 = (1 << 28) | (1 << 27) | (1 << 24) | (logical << 10) | (clen >> 11)
+            b47 = (clen & 0x7FF) << 21
+        else:
+            b03 = (3 << 28) | (1 << 27) | (1 << 24) | (clen & 0x1FFFFF)
+            b47 = logical & 0x3FFFFFFF
         blocklen = (24 + clen + 7) & ~7
         blk = bytearray(blocklen)
         blk[0:4] = b03.to_bytes(4, "little")

@@ -751,3 +751,13 @@ def test_gpu_varlena_decode_parity(ctx, orc):
     got = t2.decode_column_varlena(0, verify=True)
     assert got == [None if nulls[i] else strings[i] for i in range(30000)]
     t2.free()
+    # Dense rle_type varlena (incl. RLE expansion past the stream size)
+    reps = []
+    for i, r in enumerate(rng.integers(1, 60, 1500)):
+        reps += [[b"BUILDING", b"MACHINERY", b"HOUSEHOLD"][i % 3]] * int(r)
+    s3 = orc.aocs_encode_varlena_rle(reps, nulls=(rng.random(len(reps)) < 0.1
+                                                  ).astype(np.uint8))
+    want3 = orc.aocs_decode_varlena(s3, len(reps))
+    t3 = ctx.bind([(s3, -1, len(reps), 1)])
+    assert t3.decode_column_varlena(0, verify=True) == want3
+    t3.free()

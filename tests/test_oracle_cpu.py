@@ -686,3 +686,23 @@ def test_varlena_encoder_byte_exact_vs_reference_writer():
     out2 = orc.aocs_decode_varlena(mine2, len(strings))
     assert out2 == [None if nulls[i] else strings[i]
                     for i in range(len(strings))]
+    # Dense_Enhanced rle_type varlena (RLE on repeated payloads)
+    reps = []
+    for i, r in enumerate(rng.integers(1, 60, 2000)):
+        reps += [words[i % 5]] * int(r)
+    for i in range(0, len(reps), 97):
+        reps[i] = bytes(rng.integers(65, 90,
+                                     int(rng.integers(127, 300)))
+                        .astype(np.uint8))
+    mrle = orc.aocs_encode_varlena_rle(reps)
+    assert mrle == orc.ref_writer_varlena_stream(reps, version=2, rle=1)
+    assert orc.aocs_decode_varlena(mrle, len(reps)) == reps
+    nl = (rng.random(len(reps)) < 0.1).astype(np.uint8)
+    mrle2 = orc.aocs_encode_varlena_rle(reps, nl)
+    assert mrle2 == orc.ref_writer_varlena_stream(reps, nulls=nl,
+                                                  version=2, rle=1)
+    # one giant run crosses into the NonBulkDense envelope
+    big = [b"BUILDING"] * 100000
+    mbig = orc.aocs_encode_varlena_rle(big)
+    assert mbig == orc.ref_writer_varlena_stream(big, version=2, rle=1)
+    assert orc.aocs_decode_varlena(mbig, len(big)) == big
