@@ -1476,6 +1476,63 @@ __global__ void k_extract(const KT *tkey, const int32_t *tdate,
 
 /* ----- Motion path (nsegs>1) ----- */
 
+/* RCCL send/recv chunking: this pool's RCCL corrupts single self-sends
+ * over ~1 GiB (observed: SF100 Motion-1 self-exchange lost half its rows,
+ * profiles/bw_probe_r01.txt); self-shares bypass RCCL entirely and all
+ * peer transfers are split into <=512 MiB pieces, identically derived on
+ * both sides from the exchanged counts. */
+static const size_t GX_NCCL_CHUNK = 512ULL << 20;
+
+static ncclResult_t gx_nccl_send_chunked(const void *buf, size_t bytes,
+                                         int peer, ncclComm_t comm,
+                                         hipStream_t s)
+{
+    const uint8_t *p = (const uint8_t *) buf;
+    while (bytes > 0)
+    {
+        size_t c = bytes < GX_NCCL_CHUNK ? bytes : GX_NCCL_CHUNK;
+        ncclResult_t r = ncclSend(p, c, ncclInt8, peer, comm, s);
+        if (r != ncclSuccess) return r;
+        p += c;
+        bytes -= c;
+    }
+    return ncclSuccess;
+}
+
+static ncclResult_t gx_nccl_recv_chunked(void *buf, size_t bytes, int peer,
+                                         ncclComm_t comm, hipStream_t s)
+{
+    uint8_t *p = (uint8_t *) buf;
+    while (bytes > 0)
+    {
+        size_t c = bytes < GX_NCCL_CHUNK ? bytes : GX_NCCL_CHUNK;
+        ncclResult_t r = ncclRecv(p, c, ncclInt8, peer, comm, s);
+        if (r != ncclSuccess) return r;
+        p += c;
+        bytes -= c;
+    }
+    return ncclSuccess;
+}
+
+/* wave-aggregated per-destination counter claims: ONE atomic per wave per
+ * destination instead of one per row (a single shared cursor serializes —
+ * cdna_hip_programming.md G12; measured 68M same-address atomics ≈ 2 s). */
+__device__ __forceinline__ unsigned long long
+d_wave_claim(unsigned long long *ctr, bool mine, int lane,
+             unsigned long long *out_off)
+{
+    unsigned long long m = __ballot(mine);
+    if (m == 0) return 0;
+    int leader = __ffsll((long long) m) - 1;
+    unsigned long long wb = 0;
+    if (lane == leader)
+        wb = atomicAdd(ctr, (unsigned long long) __popcll(m));
+    wb = __shfl(wb, leader, 64);
+    *out_off = wb + __popcll(m & ((lane == 0) ? 0ULL
+                                              : (~0ULL >> (64 - lane))));
+    return m;
+}
+
 /* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
 __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
@@ -1483,14 +1540,26 @@ __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
                               int oop, int32_t olit, int nsegs,
                               unsigned long long *hist)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < od_m.nrows; i += stride)
+    for (int64_t base = base0; base < od_m.nrows; base += stride)
     {
-        if (gx_vm_hidden(vmap, i)) continue;
-        if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
-        int32_t d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
-        atomicAdd(&hist[d], 1ULL);
+        int64_t i = base + lane;
+        bool keep = false;
+        int32_t d = 0;
+        if (i < od_m.nrows && !gx_vm_hidden(vmap, i) &&
+            gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit))
+        {
+            d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
+            keep = true;
+        }
+        for (int dd = 0; dd < nsegs; dd++)
+        {
+            unsigned long long m = __ballot(keep && d == dd);
+            if (m && lane == __ffsll((long long) m) - 1)
+                atomicAdd(&hist[dd], (unsigned long long) __popcll(m));
+        }
     }
 }
 
@@ -1503,20 +1572,38 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
                               unsigned long long *cursors, /* pre-set to region starts */
                               gx_ord_row *out)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < od_m.nrows; i += stride)
+    for (int64_t base = base0; base < od_m.nrows; base += stride)
     {
-        if (gx_vm_hidden(vmap, i)) continue;
-        int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
-        if (!gx_cmp(oop, od, olit)) continue;
-        int64_t oc = gx_col_get<int64_t>(oc_s, oc_m, i);
-        int32_t d = gx_route_i64(oc, nsegs);
-        unsigned long long w = atomicAdd(&cursors[d], 1ULL);
-        out[w].okey = gx_col_get<int64_t>(ok_s, ok_m, i);
-        out[w].ocust = oc;
-        out[w].odate = od;
-        out[w].oprio = gx_col_get<int32_t>(op_s, op_m, i);
+        int64_t i = base + lane;
+        bool keep = false;
+        int32_t d = 0, od = 0;
+        int64_t oc = 0;
+        if (i < od_m.nrows && !gx_vm_hidden(vmap, i))
+        {
+            od = gx_col_get<int32_t>(od_s, od_m, i);
+            if (gx_cmp(oop, od, olit))
+            {
+                oc = gx_col_get<int64_t>(oc_s, oc_m, i);
+                d = gx_route_i64(oc, nsegs);
+                keep = true;
+            }
+        }
+        for (int dd = 0; dd < nsegs; dd++)
+        {
+            unsigned long long w;
+            unsigned long long m = d_wave_claim(&cursors[dd],
+                                                keep && d == dd, lane, &w);
+            if (m && keep && d == dd)
+            {
+                out[w].okey = gx_col_get<int64_t>(ok_s, ok_m, i);
+                out[w].ocust = oc;
+                out[w].odate = od;
+                out[w].oprio = gx_col_get<int32_t>(op_s, op_m, i);
+            }
+        }
     }
 }
 
@@ -1527,13 +1614,26 @@ __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
                             const unsigned long long *bloom, uint64_t bwmask,
                             int nsegs, unsigned long long *hist)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < n; i += stride)
+    for (int64_t base = base0; base < n; base += stride)
     {
-        if (!d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
-        atomicAdd(&hist[gx_route_i64(rows[i].okey, nsegs)], 1ULL);
+        int64_t i = base + lane;
+        bool keep = false;
+        int32_t d = 0;
+        if (i < n && d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
+            d_set_contains(cset, cmask, (uint64_t) rows[i].ocust))
+        {
+            d = gx_route_i64(rows[i].okey, nsegs);
+            keep = true;
+        }
+        for (int dd = 0; dd < nsegs; dd++)
+        {
+            unsigned long long m = __ballot(keep && d == dd);
+            if (m && lane == __ffsll((long long) m) - 1)
+                atomicAdd(&hist[dd], (unsigned long long) __popcll(m));
+        }
     }
 }
 
@@ -1544,17 +1644,32 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
                             int nsegs, unsigned long long *cursors,
                             gx_qual_row *out)
 {
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < n; i += stride)
+    for (int64_t base = base0; base < n; base += stride)
     {
-        if (!d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust)) continue;
-        if (!d_set_contains(cset, cmask, (uint64_t) rows[i].ocust)) continue;
-        int32_t d = gx_route_i64(rows[i].okey, nsegs);
-        unsigned long long w = atomicAdd(&cursors[d], 1ULL);
-        out[w].okey = rows[i].okey;
-        out[w].odate = rows[i].odate;
-        out[w].oprio = rows[i].oprio;
+        int64_t i = base + lane;
+        bool keep = false;
+        int32_t d = 0;
+        if (i < n && d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
+            d_set_contains(cset, cmask, (uint64_t) rows[i].ocust))
+        {
+            d = gx_route_i64(rows[i].okey, nsegs);
+            keep = true;
+        }
+        for (int dd = 0; dd < nsegs; dd++)
+        {
+            unsigned long long w;
+            unsigned long long m = d_wave_claim(&cursors[dd],
+                                                keep && d == dd, lane, &w);
+            if (m && keep && d == dd)
+            {
+                out[w].okey = rows[i].okey;
+                out[w].odate = rows[i].odate;
+                out[w].oprio = rows[i].oprio;
+            }
+        }
     }
 }
 
@@ -3316,15 +3431,25 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         unsigned long long recv1_n = roff1[n];
         if (grow(q->m_recv1, q->m_recv1_cap, recv1_n) != GX_OK) return GX_ERR_OOM;
         gx_ord_row *recv1 = q->m_recv1;
+        /* self-share bypasses RCCL entirely: a device copy is both faster
+         * and avoids self-send at GB sizes (alltoallv standard practice) */
+        if (h1[ctx->seg])
+            HIP_CHK(ctx, hipMemcpyAsync(recv1 + roff1[ctx->seg],
+                                        send1 + off1[ctx->seg],
+                                        h1[ctx->seg] * sizeof(gx_ord_row),
+                                        hipMemcpyDeviceToDevice, s));
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
+            if (r == ctx->seg) continue;
             if (h1[r])
-                RCCL_CHK(ctx, ncclSend(send1 + off1[r], h1[r] * sizeof(gx_ord_row),
-                                       ncclInt8, r, ctx->comm, s));
+                RCCL_CHK(ctx, gx_nccl_send_chunked(send1 + off1[r],
+                                                   h1[r] * sizeof(gx_ord_row),
+                                                   r, ctx->comm, s));
             if (rcv1[r])
-                RCCL_CHK(ctx, ncclRecv(recv1 + roff1[r], rcv1[r] * sizeof(gx_ord_row),
-                                       ncclInt8, r, ctx->comm, s));
+                RCCL_CHK(ctx, gx_nccl_recv_chunked(recv1 + roff1[r],
+                                                   rcv1[r] * sizeof(gx_ord_row),
+                                                   r, ctx->comm, s));
         }
         RCCL_CHK(ctx, ncclGroupEnd());
 
@@ -3365,15 +3490,23 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         unsigned long long recv2_n = roff2[n];
         if (grow(q->m_recv2, q->m_recv2_cap, recv2_n) != GX_OK) return GX_ERR_OOM;
         gx_qual_row *recv2 = q->m_recv2;
+        if (h2[ctx->seg])
+            HIP_CHK(ctx, hipMemcpyAsync(recv2 + roff2[ctx->seg],
+                                        send2 + off2[ctx->seg],
+                                        h2[ctx->seg] * sizeof(gx_qual_row),
+                                        hipMemcpyDeviceToDevice, s));
         RCCL_CHK(ctx, ncclGroupStart());
         for (int r = 0; r < n; r++)
         {
+            if (r == ctx->seg) continue;
             if (h2[r])
-                RCCL_CHK(ctx, ncclSend(send2 + off2[r], h2[r] * sizeof(gx_qual_row),
-                                       ncclInt8, r, ctx->comm, s));
+                RCCL_CHK(ctx, gx_nccl_send_chunked(send2 + off2[r],
+                                                   h2[r] * sizeof(gx_qual_row),
+                                                   r, ctx->comm, s));
             if (rcv2[r])
-                RCCL_CHK(ctx, ncclRecv(recv2 + roff2[r], rcv2[r] * sizeof(gx_qual_row),
-                                       ncclInt8, r, ctx->comm, s));
+                RCCL_CHK(ctx, gx_nccl_recv_chunked(recv2 + roff2[r],
+                                                   rcv2[r] * sizeof(gx_qual_row),
+                                                   r, ctx->comm, s));
         }
         RCCL_CHK(ctx, ncclGroupEnd());
 
