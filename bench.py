@@ -89,7 +89,20 @@ def main():
                     help="store l_orderkey RLE-compressed (rle_type); the fused "
                          "probe kernel then scans runs, not rows (extra mode — "
                          "the judged default stays compresstype=none)")
+    ap.add_argument("--force-motion", action="store_true",
+                    help="route the run through the FULL RCCL exchange branch "
+                         "even at 1 rank (self send/recv) — the multi-GPU code "
+                         "path, testable on one GPU (extra mode)")
     args = ap.parse_args()
+    if args.force_motion:
+        os.environ["GX_FORCE_MOTION"] = "1"
+
+    # the contract is ONE JSON line on stdout from rank 0; libraries (RCCL
+    # prints a version banner at communicator init) must not pollute it —
+    # route C-level stdout to stderr for the whole run and restore the real
+    # stdout only for the final JSON line.
+    real_stdout = os.dup(1)
+    os.dup2(2, 1)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -123,6 +136,8 @@ def main():
             t = torch.zeros(128, dtype=torch.uint8)
         dist.broadcast(t, src=0)
         ctx.comm_init(bytes(t.tolist()))
+    elif args.force_motion:
+        ctx.comm_init(ctx.comm_unique_id())   # 1-rank communicator
 
     # ---- setup (untimed): generate + AOCS-encode this segment's shard ----
     t0 = time.time()
@@ -222,7 +237,10 @@ def main():
             "roofline": roofline,
             "cpu_baseline": cpu,
         }
+        sys.stdout.flush()
+        os.dup2(real_stdout, 1)
         print(json.dumps(out), flush=True)
+        os.dup2(2, 1)          # teardown prints (RCCL banner) go to stderr
 
     q.free(); li.free(); ordr.free(); cust.free(); ctx.close()
     if dist:
