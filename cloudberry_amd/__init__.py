@@ -65,7 +65,9 @@ class _Q3Desc(ctypes.Structure):
                 ("mid_attr2_col", ctypes.c_int32), ("mid_filter", _Filter),
                 ("fact", ctypes.c_void_p), ("fact_key_col", ctypes.c_int32),
                 ("fact_a_col", ctypes.c_int32), ("fact_b_col", ctypes.c_int32),
-                ("fact_filter", _Filter)]
+                ("fact_filter", _Filter),
+                ("dim_text", ctypes.c_char * 64),
+                ("dim_text_len", ctypes.c_int32)]
 
 
 class _ColDesc(ctypes.Structure):
@@ -296,7 +298,15 @@ class Context:
         d.fact_b_col = desc_dict["fact_b_col"]
         for role in ("dim_filter", "mid_filter", "fact_filter"):
             col, op, lit = desc_dict[role]
-            setattr(d, role, _Filter(col, ops[op], lit))
+            if isinstance(lit, (str, bytes)):
+                # TEXT predicate (texteq on a varlena column, dim only)
+                assert role == "dim_filter" and op == "=="
+                blit = lit.encode() if isinstance(lit, str) else lit
+                d.dim_text = blit
+                d.dim_text_len = len(blit)
+                setattr(d, role, _Filter(col, ops[op], 0))
+            else:
+                setattr(d, role, _Filter(col, ops[op], int(lit)))
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
                                                ctypes.byref(q)))

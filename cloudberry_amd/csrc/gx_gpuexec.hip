@@ -723,6 +723,170 @@ __global__ void k_scan_filter(const uint8_t *col_s, gx_colmeta m,
     gx_wave_count_add(count, local);
 }
 
+/* TEXT dim predicate (texteq on a varlena RLE column — the reference's
+ * actual Q3 qual c_mktsegment = 'BUILDING', execQual over text).  One
+ * THREAD per varlena AO block; with rle_type segments the comparison runs
+ * ONCE PER RUN, the run's rows then stream through.  NOT NULL scope. */
+__device__ __forceinline__ bool d_texteq(const uint8_t *a, int64_t alen,
+                                         const uint8_t *b, int64_t blen)
+{
+    if (alen != blen) return false;
+    for (int64_t i = 0; i < alen; i++)
+        if (a[i] != b[i]) return false;
+    return true;
+}
+
+/* shared varlena-block walk calling CB(row, reps) for each MATCHING item */
+template <typename CB>
+__device__ __forceinline__ bool d_walk_text_matches(
+    const uint8_t *blk_content, const gx_blockref &ref,
+    const uint8_t *lit, int32_t lit_len, CB &&cb)
+{
+    const uint8_t *c = blk_content;
+    int16_t version = ((const int16_t *) c)[0];
+    int16_t flags = ((const int16_t *) c)[1];
+    int32_t logical = ref.rows;
+    bool rle = false;
+    const uint8_t *bmp = nullptr, *cnts = nullptr;
+    int32_t bmbits = 0, csize = 0, psize = 0;
+    const uint8_t *p0;
+    if (flags & 0x5)                 /* nulls/delta out of scope here */
+        return false;
+    if (version == 0)
+    {
+        if (((const int16_t *) c)[2] != logical) return false;
+        psize = ((const int32_t *) c)[3];
+        p0 = c + 16;
+    }
+    else if (version == 1 || version == 2)
+    {
+        if (((const int32_t *) c)[1] != logical) return false;
+        psize = ((const int32_t *) c)[3];
+        rle = (flags & 2) != 0;
+        const uint8_t *q = c + 16;
+        if (rle)
+        {
+            if (((const int32_t *) q)[0] != 0) return false;
+            bmbits = ((const int32_t *) q)[1];
+            csize = ((const int32_t *) q)[3];
+            q += 16;
+            bmp = q; q += (bmbits + 7) >> 3;
+            cnts = q; q += csize;
+        }
+        p0 = c + ((((int32_t) (q - c)) + 7) & ~7);
+    }
+    else
+        return false;
+    const uint8_t *p = p0, *pend = p0 + psize;
+    int64_t out = 0;
+    int32_t item = 0, coff = 0;
+    while (out < logical)
+    {
+        if (rle && item >= bmbits) return false;
+        if (p < pend && *p == 0)
+            p = p0 + (((p - p0) + 3) & ~(int64_t) 3);
+        if (p >= pend) return false;
+        int64_t len;
+        const uint8_t *data;
+        if (*p & 1)
+        {
+            len = (int64_t) (*p >> 1) - 1;
+            data = p + 1;
+            p += 1 + len;
+        }
+        else
+        {
+            uint32_t hdr;
+            memcpy(&hdr, p, 4);
+            len = (int64_t) (hdr >> 2) - 4;
+            data = p + 4;
+            p += 4 + len;
+        }
+        if (len < 0 || p > pend) return false;
+        int64_t reps = 1;
+        if (rle && (bmp[item >> 3] & (1u << (item & 7))))
+        {
+            int32_t nb = (cnts[coff] >> 6) + 1;
+            uint32_t v = cnts[coff] & 0x3F;
+            for (int32_t i = 1; i < nb; i++) v = (v << 8) | cnts[coff + i];
+            coff += nb;
+            reps += v;
+        }
+        if (out + reps > logical) return false;
+        if (d_texteq(data, len, lit, lit_len))
+            cb(ref.first_row + out, reps);
+        out += reps;
+        item++;
+    }
+    return true;
+}
+
+__global__ void k_cust_count_text(const uint8_t *key_s, gx_colmeta key_m,
+                                  const uint8_t *mkt_s, const gx_blockref *dir,
+                                  int64_t nblocks,
+                                  const uint8_t *lit, int32_t lit_len,
+                                  const uint8_t *vmap,
+                                  unsigned long long *count,
+                                  unsigned long long *maxkey, int *err)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    unsigned long long local = 0, kmax = 0;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        auto cb = [&](int64_t row0, int64_t reps) {
+            for (int64_t r = 0; r < reps; r++)
+            {
+                int64_t row = row0 + r;
+                if (gx_vm_hidden(vmap, row)) continue;
+                local++;
+                unsigned long long k = (unsigned long long)
+                    gx_col_get<int64_t>(key_s, key_m, row);
+                if (k > kmax) kmax = k;
+            }
+        };
+        if (!d_walk_text_matches(mkt_s + dir[b].offset + 24, dir[b],
+                                 lit, lit_len, cb))
+            atomicOr(err, 1);
+    }
+    if (local) atomicAdd(count, local);
+    if (kmax) atomicMax(maxkey, kmax);
+}
+
+template <typename KS>
+__global__ void k_cust_build_text(const uint8_t *key_s, gx_colmeta key_m,
+                                  const uint8_t *mkt_s, const gx_blockref *dir,
+                                  int64_t nblocks,
+                                  const uint8_t *lit, int32_t lit_len,
+                                  const uint8_t *vmap,
+                                  KS *set, uint64_t mask,
+                                  unsigned long long *bloom, uint64_t bwmask,
+                                  int *err)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        auto cb = [&](int64_t row0, int64_t reps) {
+            for (int64_t r = 0; r < reps; r++)
+            {
+                int64_t row = row0 + r;
+                if (gx_vm_hidden(vmap, row)) continue;
+                uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, row);
+                d_bloom_set(bloom, bwmask, k);
+                uint64_t slot = gx_hmix64(k) & mask;
+                while (true)
+                {
+                    KS prev = atomicCAS(&set[slot], (KS) 0, (KS) k);
+                    if (prev == (KS) 0 || prev == (KS) k) break;
+                    slot = (slot + 1) & mask;
+                }
+            }
+        };
+        if (!d_walk_text_matches(mkt_s + dir[b].offset + 24, dir[b],
+                                 lit, lit_len, cb))
+            atomicOr(err, 1);
+    }
+}
+
 /* customer: count BUILDING rows (for set sizing) */
 __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
@@ -1922,6 +2086,7 @@ struct gx_q3 {
     gx_ord_row *m_send1 = nullptr, *m_recv1 = nullptr;
     gx_qual_row *m_send2 = nullptr, *m_recv2 = nullptr;
     uint64_t m_send1_cap = 0, m_recv1_cap = 0, m_send2_cap = 0, m_recv2_cap = 0;
+    uint8_t *dtext = nullptr;          /* device dim TEXT literal (texteq) */
     /* result (device SoA) */
     int64_t *r_okey = nullptr;
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
@@ -3028,8 +3193,18 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
     auto colw = [](gx_table *t, int c) {
         return (c >= 0 && c < (int) t->cols.size()) ? t->cols[c].m.width : -1;
     };
+    bool dim_text = desc->dim_text_len > 0;
+    if (dim_text &&
+        (desc->dim_text_len > (int32_t) sizeof(desc->dim_text) ||
+         desc->dim_filter.op != 2 ||
+         colw(customer, desc->dim_filter.col) != -1 ||
+         customer->cols[desc->dim_filter.col].format != 1))
+    {
+        set_err(ctx, "TEXT dim filter needs op '==' on a varlena dir column%s", "");
+        return GX_ERR_INVALID;
+    }
     if (colw(customer, desc->dim_key_col) != 8 ||
-        colw(customer, desc->dim_filter.col) != 1 ||
+        (!dim_text && colw(customer, desc->dim_filter.col) != 1) ||
         colw(orders, desc->mid_key_col) != 8 ||
         colw(orders, desc->mid_fk_col) != 8 ||
         colw(orders, desc->mid_attr1_col) != 4 ||
@@ -3050,7 +3225,8 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
     for (auto *t : {customer, orders, lineitem})
         for (size_t ci = 0; ci < t->cols.size(); ci++)
             if (t->cols[ci].format != 0 &&
-                !(t == lineitem && (int) ci == desc->fact_key_col))
+                !(t == lineitem && (int) ci == desc->fact_key_col) &&
+                !(dim_text && t == customer && (int) ci == desc->dim_filter.col))
             {
                 set_err(ctx, "Q3 pipeline supports RLE only on the fact key; "
                              "decode other RLE columns first%s", "");
@@ -3062,6 +3238,15 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
     q->ord = orders;
     q->li = lineitem;
     q->desc = *desc;
+    if (dim_text)
+    {
+        hipError_t e = hipMalloc(&q->dtext,
+                                 std::max(desc->dim_text_len, 1));
+        if (e != hipSuccess) { delete q; return GX_ERR_OOM; }
+        e = hipMemcpy(q->dtext, desc->dim_text, desc->dim_text_len,
+                      hipMemcpyHostToDevice);
+        if (e != hipSuccess) { hipFree(q->dtext); delete q; return GX_ERR_OOM; }
+    }
     *out = q;
     return GX_OK;
 }
@@ -3099,6 +3284,7 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
     fr(q->m_hist); fr(q->m_cur); fr(q->m_cnts_mine); fr(q->m_cnts_all);
     fr(q->m_send1); fr(q->m_recv1); fr(q->m_send2); fr(q->m_recv2);
+    fr(q->dtext);
     q->m_send1_cap = q->m_recv1_cap = q->m_send2_cap = q->m_recv2_cap = 0;
     q->sized = false;
 }
@@ -3119,15 +3305,32 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
 
     HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
     HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for max custkey */
-    hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                       q->cust->cols[D.dim_key_col].dstream,
-                       q->cust->cols[D.dim_key_col].m,
-                       cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
-                       (int8_t) D.dim_filter.literal, q->dcount, q->dhits);
+    if (D.dim_text_len > 0)
+    {
+        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 4, s));   /* err flag */
+        hipLaunchKernelGGL(k_cust_count_text, dim3(GRID), dim3(64), 0, s,
+                           q->cust->cols[D.dim_key_col].dstream,
+                           q->cust->cols[D.dim_key_col].m,
+                           cm.dstream, cm.ddir, cm.nblocks,
+                           q->dtext, D.dim_text_len, q->cust->dvmap,
+                           q->dcount, q->dhits, (int *) q->dmin);
+    }
+    else
+        hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
+                           q->cust->cols[D.dim_key_col].dstream,
+                           q->cust->cols[D.dim_key_col].m,
+                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
+                           (int8_t) D.dim_filter.literal, q->dcount, q->dhits);
     unsigned long long n_building = 0, cmax = 0;
     HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
+    if (D.dim_text_len > 0)
+    {
+        int terr = 0;
+        HIP_CHK(ctx, hipMemcpy(&terr, q->dmin, 4, hipMemcpyDeviceToHost));
+        if (terr) { set_err(ctx, "malformed varlena dim column%s", ""); return GX_ERR_INVALID; }
+    }
     q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
     uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
     HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
@@ -3138,22 +3341,29 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMalloc(&q->bloom, bwords * 8));
     HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, bwords * 8, s));
     q->bwmask = bwords - 1;
+    auto launch_cbuild = [&](auto *cs) {
+        if (D.dim_text_len > 0)
+            hipLaunchKernelGGL((k_cust_build_text<std::decay_t<decltype(*cs)>>),
+                               dim3(GRID), dim3(64), 0, s,
+                               q->cust->cols[D.dim_key_col].dstream,
+                               q->cust->cols[D.dim_key_col].m,
+                               cm.dstream, cm.ddir, cm.nblocks,
+                               q->dtext, D.dim_text_len, q->cust->dvmap,
+                               cs, q->cmask, q->bloom, q->bwmask,
+                               (int *) q->dmin);
+        else
+            hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
+                               dim3(GRID), dim3(TPB), 0, s,
+                               q->cust->cols[D.dim_key_col].dstream,
+                               q->cust->cols[D.dim_key_col].m,
+                               cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
+                               (int8_t) D.dim_filter.literal,
+                               cs, q->cmask, q->bloom, q->bwmask);
+    };
     if (q->cset_width == 4)
-        hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[D.dim_key_col].dstream,
-                           q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
-                           (int8_t) D.dim_filter.literal,
-                           (unsigned int *) q->cset, q->cmask,
-                           q->bloom, q->bwmask);
+        launch_cbuild((unsigned int *) q->cset);
     else
-        hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[D.dim_key_col].dstream,
-                           q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
-                           (int8_t) D.dim_filter.literal,
-                           (unsigned long long *) q->cset, q->cmask,
-                           q->bloom, q->bwmask);
+        launch_cbuild((unsigned long long *) q->cset);
 
     /* local qualifying-order count bounds the table for BOTH paths: at
      * nsegs>1 the table holds rows received for THIS segment; the global
@@ -3266,16 +3476,28 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     HIP_CHK(ctx, hipEventRecord(ev[0], s));
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, (q->cmask + 1) * q->cset_width, s));
     HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, (q->bwmask + 1) * 8, s));
-    if (q->cset_width == 4)
-        hipLaunchKernelGGL(k_cust_build<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                           ck.dstream, ck.m, cm.dstream, cm.m, q->cust->dvmap,
-                           D.dim_filter.op, (int8_t) D.dim_filter.literal,
-                           (unsigned int *) q->cset, q->cmask, q->bloom, q->bwmask);
-    else
-        hipLaunchKernelGGL(k_cust_build<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                           ck.dstream, ck.m, cm.dstream, cm.m, q->cust->dvmap,
-                           D.dim_filter.op, (int8_t) D.dim_filter.literal,
-                           (unsigned long long *) q->cset, q->cmask, q->bloom, q->bwmask);
+    {
+        auto launch_cb = [&](auto *cs) {
+            if (D.dim_text_len > 0)
+                hipLaunchKernelGGL((k_cust_build_text<std::decay_t<decltype(*cs)>>),
+                                   dim3(GRID), dim3(64), 0, s,
+                                   ck.dstream, ck.m, cm.dstream, cm.ddir,
+                                   cm.nblocks, q->dtext, D.dim_text_len,
+                                   q->cust->dvmap, cs, q->cmask,
+                                   q->bloom, q->bwmask, (int *) q->dmin);
+            else
+                hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
+                                   dim3(GRID), dim3(TPB), 0, s,
+                                   ck.dstream, ck.m, cm.dstream, cm.m,
+                                   q->cust->dvmap,
+                                   D.dim_filter.op, (int8_t) D.dim_filter.literal,
+                                   cs, q->cmask, q->bloom, q->bwmask);
+        };
+        if (q->cset_width == 4)
+            launch_cb((unsigned int *) q->cset);
+        else
+            launch_cb((unsigned long long *) q->cset);
+    }
     HIP_CHK(ctx, hipEventRecord(ev[1], s));
 
     /* ---- stage 2: orders build (local or via Motions) ----

@@ -218,6 +218,13 @@ typedef struct gx_q3_desc {
     int32_t fact_a_col;        /* f64 (l_extendedprice) */
     int32_t fact_b_col;        /* f64 (l_discount) */
     gx_filter fact_filter;     /* e.g. l_shipdate > literal */
+    /* optional TEXT dim predicate (texteq — the reference's actual Q3 qual
+     * c_mktsegment = 'BUILDING'): when dim_text_len > 0, dim_filter.col
+     * names a VARLENA directory column (width -1, format 1) and the
+     * predicate is payload == dim_text with op '=='.  With rle_type
+     * segments the comparison runs once per RUN. */
+    char dim_text[64];
+    int32_t dim_text_len;
 } gx_q3_desc;
 
 gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc, gx_q3 **out);

@@ -761,3 +761,46 @@ def test_gpu_varlena_decode_parity(ctx, orc):
     t3 = ctx.bind([(s3, -1, len(reps), 1)])
     assert t3.decode_column_varlena(0, verify=True) == want3
     t3.free()
+
+
+@pytest.mark.gpu
+def test_gpu_text_dim_predicate_q3(ctx, orc):
+    """The reference's ACTUAL Q3 dim qual — c_mktsegment = 'BUILDING' as a
+    TEXT predicate (texteq) evaluated on a varlena rle_type column inside
+    the pipeline (one comparison per run) — matches the oracle run on the
+    equivalent code-filtered tables."""
+    sf = 0.5
+    segs = [b"BUILDING", b"AUTOMOBILE", b"MACHINERY", b"HOUSEHOLD",
+            b"FURNITURE"]
+    c = orc.gen_customer(sf)
+    strings = [segs[int(x)] for x in c["c_mktsegment"]]
+    cust = ctx.bind([
+        (orc.aocs_encode(c["c_custkey"]), 8, len(strings), 0),
+        (orc.aocs_encode_varlena_rle(strings), -1, len(strings), 1),
+    ])
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    q = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", "BUILDING"),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3,
+        "mid_filter": (2, "<", float(gx.CUTOFF_19950315)),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", float(gx.CUTOFF_19950315)),
+    }).run()
+    got = q.result()
+    want = orc.q3(c, orc.gen_orders(sf), orc.gen_lineitem(sf))
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["o_orderdate"], want["o_orderdate"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-9)
+    # a non-matching literal selects nothing
+    q2 = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", "NOSEGMENT"),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3,
+        "mid_filter": (2, "<", float(gx.CUTOFF_19950315)),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", float(gx.CUTOFF_19950315)),
+    }).run()
+    assert len(q2.result()["l_orderkey"]) == 0
