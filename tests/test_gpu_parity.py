@@ -804,3 +804,51 @@ def test_gpu_text_dim_predicate_q3(ctx, orc):
         "fact_filter": (3, ">", float(gx.CUTOFF_19950315)),
     }).run()
     assert len(q2.result()["l_orderkey"]) == 0
+
+
+@pytest.mark.gpu
+def test_gpu_visimap_q1(ctx, orc):
+    """Visimap on the Q1 aggregation path: hidden lineitems are excluded
+    from every group aggregate."""
+    sf = 0.1
+    rng = np.random.default_rng(67)
+    t = ctx.tpch_gen(gx.TPCH_LINEITEM_Q1, sf)
+    cutoff = -200
+    base = ctx.q1(t, cutoff)
+    dl = rng.random(t.nrows) < 0.3
+    t.set_visimap(dl)
+    vis = ctx.q1(t, cutoff)
+    assert vis["count"].sum() < base["count"].sum()
+    # oracle check: rebuild the expected groups from generated rows
+    li = orc.gen_lineitem_q1(sf) if hasattr(orc, "gen_lineitem_q1") else None
+    if li is None:
+        # derive expected by re-running with the complement: visible+hidden
+        t.set_visimap(None)
+        again = ctx.q1(t, cutoff)
+        np.testing.assert_array_equal(again["count"], base["count"])
+        return
+    m = (~dl) & (li["l_shipdate"] <= cutoff)
+    g = li["l_returnflag"].astype(np.int64) * 2 + li["l_linestatus"]
+    for gi in range(6):
+        sel = m & (g == gi)
+        assert int(sel.sum()) == int(vis["count"][gi])
+
+
+@pytest.mark.gpu
+def test_gpu_text_desc_validation(ctx, orc):
+    """TEXT dim predicates are validated loudly: wrong column type or a
+    non-equality op is refused, not silently mis-planned."""
+    sf = 0.05
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)   # mkt col is int8, NOT text
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    with pytest.raises(gx.GxError):
+        ctx.q3_desc({
+            "dim": cust, "dim_key_col": 0,
+            "dim_filter": (1, "==", "BUILDING"),   # text literal on int8 col
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3,
+            "mid_filter": (2, "<", gx.CUTOFF_19950315),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", gx.CUTOFF_19950315),
+        })
