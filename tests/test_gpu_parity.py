@@ -852,3 +852,58 @@ def test_gpu_text_desc_validation(ctx, orc):
             "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
             "fact_filter": (3, ">", gx.CUTOFF_19950315),
         })
+
+
+@pytest.mark.gpu
+def test_gpu_u64_keys_q3(ctx, orc):
+    """Keys above 2^32 force the u64 cset/table instantiations and (sparse
+    key range) the HASH slot mapping instead of interpolation — the paths
+    small TPC-H keys never reach."""
+    sf = 0.1
+    SHIFT_C = 1 << 33
+    SHIFT_O = 1 << 34
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    l = orc.gen_lineitem(sf)
+    c2 = {"c_custkey": c["c_custkey"] + SHIFT_C,
+          "c_mktsegment": c["c_mktsegment"]}
+    o2 = {"o_orderkey": o["o_orderkey"] + SHIFT_O,
+          "o_custkey": o["o_custkey"] + SHIFT_C,
+          "o_orderdate": o["o_orderdate"],
+          "o_shippriority": o["o_shippriority"]}
+    l2 = {"l_orderkey": l["l_orderkey"] + SHIFT_O,
+          "l_extendedprice": l["l_extendedprice"],
+          "l_discount": l["l_discount"], "l_shipdate": l["l_shipdate"]}
+    cust = ctx.bind([(orc.aocs_encode(c2["c_custkey"]), 8, len(c["c_custkey"]), 0),
+                     (orc.aocs_encode(c2["c_mktsegment"].astype(np.int8)), 1,
+                      len(c["c_custkey"]), 0)])
+    ordr = ctx.bind([(orc.aocs_encode(o2["o_orderkey"]), 8, len(o["o_orderkey"]), 0),
+                     (orc.aocs_encode(o2["o_custkey"]), 8, len(o["o_orderkey"]), 0),
+                     (orc.aocs_encode(o2["o_orderdate"]), 4, len(o["o_orderkey"]), 0),
+                     (orc.aocs_encode(o2["o_shippriority"]), 4, len(o["o_orderkey"]), 0)])
+    li = ctx.bind([(orc.aocs_encode(l2["l_orderkey"]), 8, len(l["l_orderkey"]), 0),
+                   (orc.aocs_encode(l2["l_extendedprice"].view(np.int64)), 8,
+                    len(l["l_orderkey"]), 0),
+                   (orc.aocs_encode(l2["l_discount"].view(np.int64)), 8,
+                    len(l["l_orderkey"]), 0),
+                   (orc.aocs_encode(l2["l_shipdate"]), 4, len(l["l_orderkey"]), 0)])
+    got = ctx.q3(cust, ordr, li).run().result()
+    want = orc.q3(c2, o2, l2)
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-9)
+
+    # same big-key plan through the full RCCL exchange branch (u64 motion
+    # table rebuild); comm may already exist from an earlier test
+    import os
+    try:
+        ctx.comm_init(ctx.comm_unique_id())
+    except gx.GxError:
+        pass
+    os.environ["GX_FORCE_MOTION"] = "1"
+    try:
+        got2 = ctx.q3(cust, ordr, li).run().result()
+    finally:
+        del os.environ["GX_FORCE_MOTION"]
+    np.testing.assert_array_equal(got2["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_allclose(got2["revenue"], want["revenue"], rtol=1e-9)
