@@ -1115,6 +1115,51 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -9)
+    {
+        /* DIAGNOSTIC ONLY (wrong results): B=1 without the ship filter —
+         * isolates how much of the probe's critical path is the
+         * ship→key→table dependent chain vs the key→table chain alone. */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        for (; i < lk_m.nrows; i += stride)
+        {
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = smap.slot0(k);
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r != ~0ULL) hit(r, i);
+        }
+    }
+    else if constexpr (B == -10)
+    {
+        /* DIAGNOSTIC ONLY: ship+key loads and filter, NO table probe —
+         * the pure two-stream scan cost. */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        unsigned long long acc = 0;
+        for (; i < lk_m.nrows; i += stride)
+        {
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
+            acc += (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+        }
+        local_hits += acc & 1;       /* keep the loads */
+    }
+    else if constexpr (B == -11)
+    {
+        /* DIAGNOSTIC ONLY: filter + key + table probe, NO aggregation
+         * (no price/disc loads, no atomics) — isolates the table-lookup
+         * chain from the hit processing. */
+        int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+        int64_t stride = gridDim.x * (int64_t) blockDim.x;
+        for (; i < lk_m.nrows; i += stride)
+        {
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = smap.slot0(k);
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r != ~0ULL) local_hits++;
+        }
+    }
     else if constexpr (B == -5)
     {
         /* software-pipelined B=1: next iteration's ship+key loads issue
@@ -3900,6 +3945,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 8: launch(k_li_probe_agg_t<-2, unsigned int>, keys); break;
                 case 9: launch(k_li_probe_agg_t<-5, unsigned int>, keys); break;
                 case 10: launch(k_li_probe_agg_t<-6, unsigned int>, keys); break;
+                case 11: launch(k_li_probe_agg_t<-9, unsigned int>, keys); break;
+                case 12: launch(k_li_probe_agg_t<-10, unsigned int>, keys); break;
+                case 13: launch(k_li_probe_agg_t<-11, unsigned int>, keys); break;
             }
         }
         else
@@ -3920,6 +3968,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 8: launch(k_li_probe_agg_t<-2, unsigned long long>, keys); break;
                 case 9: launch(k_li_probe_agg_t<-5, unsigned long long>, keys); break;
                 case 10: launch(k_li_probe_agg_t<-6, unsigned long long>, keys); break;
+                case 11: launch(k_li_probe_agg_t<-9, unsigned long long>, keys); break;
+                case 12: launch(k_li_probe_agg_t<-10, unsigned long long>, keys); break;
+                case 13: launch(k_li_probe_agg_t<-11, unsigned long long>, keys); break;
             }
         }
     }
