@@ -1115,6 +1115,62 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             if (r != ~0ULL) hit(r, i);
         }
     }
+    else if constexpr (B == -12 || B == -13)
+    {
+        /* TILE-COMPACT-THEN-PROBE: the per-thread batching variants lost
+         * because batching the SCAN breaks its coalescing; here each
+         * workgroup alternates phases over a tile — a coalesced scan
+         * compacts survivors (key,row) into LDS, then the probe phase
+         * works off LDS with fully independent iterations the compiler
+         * can overlap (the decomposition in profiles/bw_probe_r01.txt
+         * showed the three chain stages are exactly additive in the
+         * fused form). */
+        constexpr int TILE = (B == -13) ? 2048 : 1024;
+        __shared__ unsigned int s_cnt;
+        __shared__ uint64_t s_key[TILE];
+        __shared__ uint32_t s_row[TILE];
+        int64_t chunk = (lk_m.nrows + gridDim.x - 1) / gridDim.x;
+        int64_t lo = blockIdx.x * chunk;
+        int64_t hi = min(lo + chunk, lk_m.nrows);
+        int lane = threadIdx.x & 63;
+        for (int64_t t0 = lo; t0 < hi; t0 += TILE)
+        {
+            int64_t tend = min(t0 + TILE, hi);
+            if (threadIdx.x == 0) s_cnt = 0;
+            __syncthreads();
+            for (int64_t i = t0 + threadIdx.x; i < tend; i += blockDim.x)
+            {
+                bool keep = gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit);
+                uint64_t k = keep ? (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i) : 0;
+                unsigned long long m = __ballot(keep);
+                if (m)
+                {
+                    unsigned wb = 0;
+                    int leader = __ffsll((long long) m) - 1;
+                    if (lane == leader)
+                        wb = atomicAdd(&s_cnt, (unsigned) __popcll(m));
+                    wb = __shfl(wb, leader, 64);
+                    if (keep)
+                    {
+                        unsigned o = __popcll(m & ((lane == 0) ? 0ULL
+                                                  : (~0ULL >> (64 - lane))));
+                        s_key[wb + o] = k;
+                        s_row[wb + o] = (uint32_t) (i - lo);
+                    }
+                }
+            }
+            __syncthreads();
+            int nsv = (int) s_cnt;
+            for (int j = threadIdx.x; j < nsv; j += blockDim.x)
+            {
+                uint64_t k = s_key[j];
+                uint64_t slot = smap.slot0(k);
+                uint64_t r = resolve(k, slot, tkey[slot]);
+                if (r != ~0ULL) hit(r, lo + (int64_t) s_row[j]);
+            }
+            __syncthreads();
+        }
+    }
     else if constexpr (B == -9)
     {
         /* DIAGNOSTIC ONLY (wrong results): B=1 without the ship filter —
@@ -3948,6 +4004,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 11: launch(k_li_probe_agg_t<-9, unsigned int>, keys); break;
                 case 12: launch(k_li_probe_agg_t<-10, unsigned int>, keys); break;
                 case 13: launch(k_li_probe_agg_t<-11, unsigned int>, keys); break;
+                case 14: launch(k_li_probe_agg_t<-12, unsigned int>, keys); break;
+                case 15: launch(k_li_probe_agg_t<-13, unsigned int>, keys); break;
             }
         }
         else
@@ -3971,6 +4029,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 11: launch(k_li_probe_agg_t<-9, unsigned long long>, keys); break;
                 case 12: launch(k_li_probe_agg_t<-10, unsigned long long>, keys); break;
                 case 13: launch(k_li_probe_agg_t<-11, unsigned long long>, keys); break;
+                case 14: launch(k_li_probe_agg_t<-12, unsigned long long>, keys); break;
+                case 15: launch(k_li_probe_agg_t<-13, unsigned long long>, keys); break;
             }
         }
     }
