@@ -736,158 +736,75 @@ __device__ __forceinline__ bool d_texteq(const uint8_t *a, int64_t alen,
     return true;
 }
 
-/* shared varlena-block walk calling CB(row, reps) for each MATCHING item */
-template <typename CB>
-__device__ __forceinline__ bool d_walk_text_matches(
-    const uint8_t *blk_content, const gx_blockref &ref,
-    const uint8_t *lit, int32_t lit_len, CB &&cb)
+/* texteq over a decoded varlena column (device offsets+payload) → one
+ * match byte per row.  Runs ONCE at prepare: a constant predicate over an
+ * immutable column is evaluated once and cached, the per-step scans then
+ * read the mask on the fast fixed-width path (what dictionary-aware
+ * executors do for constant text quals). */
+__global__ void k_texteq_mask(const int64_t *offsets, const uint8_t *payload,
+                              int64_t n, const uint8_t *lit, int32_t lit_len,
+                              uint8_t *mask)
 {
-    const uint8_t *c = blk_content;
-    int16_t version = ((const int16_t *) c)[0];
-    int16_t flags = ((const int16_t *) c)[1];
-    int32_t logical = ref.rows;
-    bool rle = false;
-    const uint8_t *bmp = nullptr, *cnts = nullptr;
-    int32_t bmbits = 0, csize = 0, psize = 0;
-    const uint8_t *p0;
-    if (flags & 0x5)                 /* nulls/delta out of scope here */
-        return false;
-    if (version == 0)
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
     {
-        if (((const int16_t *) c)[2] != logical) return false;
-        psize = ((const int32_t *) c)[3];
-        p0 = c + 16;
+        int64_t len = offsets[i + 1] - offsets[i];
+        mask[i] = (uint8_t) d_texteq(payload + offsets[i], len, lit, lit_len);
     }
-    else if (version == 1 || version == 2)
-    {
-        if (((const int32_t *) c)[1] != logical) return false;
-        psize = ((const int32_t *) c)[3];
-        rle = (flags & 2) != 0;
-        const uint8_t *q = c + 16;
-        if (rle)
-        {
-            if (((const int32_t *) q)[0] != 0) return false;
-            bmbits = ((const int32_t *) q)[1];
-            csize = ((const int32_t *) q)[3];
-            q += 16;
-            bmp = q; q += (bmbits + 7) >> 3;
-            cnts = q; q += csize;
-        }
-        p0 = c + ((((int32_t) (q - c)) + 7) & ~7);
-    }
-    else
-        return false;
-    const uint8_t *p = p0, *pend = p0 + psize;
-    int64_t out = 0;
-    int32_t item = 0, coff = 0;
-    while (out < logical)
-    {
-        if (rle && item >= bmbits) return false;
-        if (p < pend && *p == 0)
-            p = p0 + (((p - p0) + 3) & ~(int64_t) 3);
-        if (p >= pend) return false;
-        int64_t len;
-        const uint8_t *data;
-        if (*p & 1)
-        {
-            len = (int64_t) (*p >> 1) - 1;
-            data = p + 1;
-            p += 1 + len;
-        }
-        else
-        {
-            uint32_t hdr;
-            memcpy(&hdr, p, 4);
-            len = (int64_t) (hdr >> 2) - 4;
-            data = p + 4;
-            p += 4 + len;
-        }
-        if (len < 0 || p > pend) return false;
-        int64_t reps = 1;
-        if (rle && (bmp[item >> 3] & (1u << (item & 7))))
-        {
-            int32_t nb = (cnts[coff] >> 6) + 1;
-            uint32_t v = cnts[coff] & 0x3F;
-            for (int32_t i = 1; i < nb; i++) v = (v << 8) | cnts[coff + i];
-            coff += nb;
-            reps += v;
-        }
-        if (out + reps > logical) return false;
-        if (d_texteq(data, len, lit, lit_len))
-            cb(ref.first_row + out, reps);
-        out += reps;
-        item++;
-    }
-    return true;
 }
 
-__global__ void k_cust_count_text(const uint8_t *key_s, gx_colmeta key_m,
-                                  const uint8_t *mkt_s, const gx_blockref *dir,
-                                  int64_t nblocks,
-                                  const uint8_t *lit, int32_t lit_len,
-                                  const uint8_t *vmap,
+/* flat-mask variants of the customer scan (mask built by k_texteq_mask) */
+__global__ void k_cust_count_mask(const uint8_t *key_s, gx_colmeta key_m,
+                                  const uint8_t *mask, const uint8_t *vmap,
                                   unsigned long long *count,
-                                  unsigned long long *maxkey, int *err)
+                                  unsigned long long *maxkey)
 {
-    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
     unsigned long long local = 0, kmax = 0;
-    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    for (; i < key_m.nrows; i += stride)
+        if (mask[i] && !gx_vm_hidden(vmap, i))
+        {
+            local++;
+            unsigned long long k = (unsigned long long)
+                gx_col_get<int64_t>(key_s, key_m, i);
+            if (k > kmax) kmax = k;
+        }
+    gx_wave_count_add(count, local);
+    for (int o = 32; o; o >>= 1)
     {
-        auto cb = [&](int64_t row0, int64_t reps) {
-            for (int64_t r = 0; r < reps; r++)
-            {
-                int64_t row = row0 + r;
-                if (gx_vm_hidden(vmap, row)) continue;
-                local++;
-                unsigned long long k = (unsigned long long)
-                    gx_col_get<int64_t>(key_s, key_m, row);
-                if (k > kmax) kmax = k;
-            }
-        };
-        if (!d_walk_text_matches(mkt_s + dir[b].offset + 24, dir[b],
-                                 lit, lit_len, cb))
-            atomicOr(err, 1);
+        unsigned long long v = __shfl_down(kmax, o, 64);
+        if (v > kmax) kmax = v;
     }
-    if (local) atomicAdd(count, local);
-    if (kmax) atomicMax(maxkey, kmax);
+    if ((threadIdx.x & 63) == 0 && kmax)
+        atomicMax(maxkey, kmax);
 }
 
 template <typename KS>
-__global__ void k_cust_build_text(const uint8_t *key_s, gx_colmeta key_m,
-                                  const uint8_t *mkt_s, const gx_blockref *dir,
-                                  int64_t nblocks,
-                                  const uint8_t *lit, int32_t lit_len,
-                                  const uint8_t *vmap,
-                                  KS *set, uint64_t mask,
-                                  unsigned long long *bloom, uint64_t bwmask,
-                                  int *err)
+__global__ void k_cust_build_mask(const uint8_t *key_s, gx_colmeta key_m,
+                                  const uint8_t *mask, const uint8_t *vmap,
+                                  KS *set, uint64_t cmask,
+                                  unsigned long long *bloom, uint64_t bwmask)
 {
-    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
-    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < key_m.nrows; i += stride)
     {
-        auto cb = [&](int64_t row0, int64_t reps) {
-            for (int64_t r = 0; r < reps; r++)
-            {
-                int64_t row = row0 + r;
-                if (gx_vm_hidden(vmap, row)) continue;
-                uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, row);
-                d_bloom_set(bloom, bwmask, k);
-                uint64_t slot = gx_hmix64(k) & mask;
-                while (true)
-                {
-                    KS prev = atomicCAS(&set[slot], (KS) 0, (KS) k);
-                    if (prev == (KS) 0 || prev == (KS) k) break;
-                    slot = (slot + 1) & mask;
-                }
-            }
-        };
-        if (!d_walk_text_matches(mkt_s + dir[b].offset + 24, dir[b],
-                                 lit, lit_len, cb))
-            atomicOr(err, 1);
+        if (!mask[i] || gx_vm_hidden(vmap, i)) continue;
+        uint64_t k = (uint64_t) gx_col_get<int64_t>(key_s, key_m, i);
+        d_bloom_set(bloom, bwmask, k);
+        uint64_t slot = gx_hmix64(k) & cmask;
+        while (true)
+        {
+            KS prev = atomicCAS(&set[slot], (KS) 0, (KS) k);
+            if (prev == (KS) 0 || prev == (KS) k) break;
+            slot = (slot + 1) & cmask;
+        }
     }
 }
 
-/* customer: count BUILDING rows (for set sizing) */
+/* customer: count BUILDING rows (for set sizing) *//* customer: count BUILDING rows (for set sizing) */
 __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
                              const uint8_t *vmap, int cop, int8_t clit,
@@ -2188,6 +2105,8 @@ struct gx_q3 {
     gx_qual_row *m_send2 = nullptr, *m_recv2 = nullptr;
     uint64_t m_send1_cap = 0, m_recv1_cap = 0, m_send2_cap = 0, m_recv2_cap = 0;
     uint8_t *dtext = nullptr;          /* device dim TEXT literal (texteq) */
+    uint8_t *dmask = nullptr;          /* per-row texteq result (built once
+                                          at prepare over the varlena col) */
     /* result (device SoA) */
     int64_t *r_okey = nullptr;
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
@@ -3285,6 +3204,61 @@ extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t
 
 /* ================= Q3 ================= */
 
+/* Evaluate the constant TEXT dim predicate ONCE over the varlena column
+ * (device two-pass decode + texteq) into a per-row match mask; the
+ * per-step customer scans then run on the fast fixed-width path. */
+static gx_status q3_build_text_mask(gx_ctx *ctx, gx_q3 *q)
+{
+    const gx_col &cm = q->cust->cols[q->desc.dim_filter.col];
+    int64_t n = cm.m.nrows;
+    hipStream_t s = ctx->stream;
+    devbuf dbase, doff, dpay, derrb;
+    HIP_CHK(ctx, dbase.alloc(std::max<int64_t>(cm.nblocks, 1) * 8));
+    HIP_CHK(ctx, doff.alloc((n + 1) * 8));
+    HIP_CHK(ctx, derrb.alloc(4));
+    int *derr = derrb.as<int>();
+    HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, s));
+    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, s,
+                       cm.dstream, cm.ddir, cm.nblocks, n,
+                       (const int64_t *) nullptr, (uint8_t *) nullptr,
+                       (int64_t *) nullptr, (uint8_t *) nullptr,
+                       dbase.as<int64_t>(), 0, derr);
+    std::vector<int64_t> bases(std::max<int64_t>(cm.nblocks, 1));
+    HIP_CHK(ctx, hipMemcpyAsync(bases.data(), dbase.p, cm.nblocks * 8,
+                                hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    int64_t total = 0;
+    for (int64_t b = 0; b < cm.nblocks; b++)
+    {
+        int64_t v = bases[b];
+        bases[b] = total;
+        total += v;
+    }
+    HIP_CHK(ctx, dpay.alloc(std::max<int64_t>(total, 1)));
+    HIP_CHK(ctx, hipMemcpyAsync(dbase.p, bases.data(), cm.nblocks * 8,
+                                hipMemcpyHostToDevice, s));
+    HIP_CHK(ctx, hipMemsetAsync(doff.p, 0, 8, s));
+    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, s,
+                       cm.dstream, cm.ddir, cm.nblocks, n,
+                       dbase.as<int64_t>(), dpay.as<uint8_t>(),
+                       doff.as<int64_t>(), (uint8_t *) nullptr,
+                       (int64_t *) nullptr, 1, derr);
+    HIP_CHK(ctx, hipMalloc(&q->dmask, std::max<int64_t>(n, 1)));
+    hipLaunchKernelGGL(k_texteq_mask, dim3(GRID), dim3(TPB), 0, s,
+                       doff.as<int64_t>(), dpay.as<uint8_t>(), n,
+                       q->dtext, q->desc.dim_text_len, q->dmask);
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr)
+    {
+        set_err(ctx, "malformed (or NULL-bearing) varlena dim column%s", "");
+        return GX_ERR_INVALID;
+    }
+    return GX_OK;
+}
+
 extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
                                         gx_q3 **out)
 {
@@ -3347,6 +3321,14 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
         e = hipMemcpy(q->dtext, desc->dim_text, desc->dim_text_len,
                       hipMemcpyHostToDevice);
         if (e != hipSuccess) { hipFree(q->dtext); delete q; return GX_ERR_OOM; }
+        gx_status st = q3_build_text_mask(ctx, q);
+        if (st != GX_OK)
+        {
+            hipFree(q->dtext);
+            if (q->dmask) hipFree(q->dmask);
+            delete q;
+            return st;
+        }
     }
     *out = q;
     return GX_OK;
@@ -3386,6 +3368,7 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->m_hist); fr(q->m_cur); fr(q->m_cnts_mine); fr(q->m_cnts_all);
     fr(q->m_send1); fr(q->m_recv1); fr(q->m_send2); fr(q->m_recv2);
     fr(q->dtext);
+    fr(q->dmask);
     q->m_send1_cap = q->m_recv1_cap = q->m_send2_cap = q->m_recv2_cap = 0;
     q->sized = false;
 }
@@ -3407,15 +3390,11 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
     HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for max custkey */
     if (D.dim_text_len > 0)
-    {
-        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 4, s));   /* err flag */
-        hipLaunchKernelGGL(k_cust_count_text, dim3(GRID), dim3(64), 0, s,
+        hipLaunchKernelGGL(k_cust_count_mask, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.ddir, cm.nblocks,
-                           q->dtext, D.dim_text_len, q->cust->dvmap,
-                           q->dcount, q->dhits, (int *) q->dmin);
-    }
+                           q->dmask, q->cust->dvmap,
+                           q->dcount, q->dhits);
     else
         hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
@@ -3426,12 +3405,6 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
-    if (D.dim_text_len > 0)
-    {
-        int terr = 0;
-        HIP_CHK(ctx, hipMemcpy(&terr, q->dmin, 4, hipMemcpyDeviceToHost));
-        if (terr) { set_err(ctx, "malformed varlena dim column%s", ""); return GX_ERR_INVALID; }
-    }
     q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
     uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
     HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
@@ -3444,14 +3417,12 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     q->bwmask = bwords - 1;
     auto launch_cbuild = [&](auto *cs) {
         if (D.dim_text_len > 0)
-            hipLaunchKernelGGL((k_cust_build_text<std::decay_t<decltype(*cs)>>),
-                               dim3(GRID), dim3(64), 0, s,
+            hipLaunchKernelGGL((k_cust_build_mask<std::decay_t<decltype(*cs)>>),
+                               dim3(GRID), dim3(TPB), 0, s,
                                q->cust->cols[D.dim_key_col].dstream,
                                q->cust->cols[D.dim_key_col].m,
-                               cm.dstream, cm.ddir, cm.nblocks,
-                               q->dtext, D.dim_text_len, q->cust->dvmap,
-                               cs, q->cmask, q->bloom, q->bwmask,
-                               (int *) q->dmin);
+                               q->dmask, q->cust->dvmap,
+                               cs, q->cmask, q->bloom, q->bwmask);
         else
             hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
                                dim3(GRID), dim3(TPB), 0, s,
@@ -3580,12 +3551,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         auto launch_cb = [&](auto *cs) {
             if (D.dim_text_len > 0)
-                hipLaunchKernelGGL((k_cust_build_text<std::decay_t<decltype(*cs)>>),
-                                   dim3(GRID), dim3(64), 0, s,
-                                   ck.dstream, ck.m, cm.dstream, cm.ddir,
-                                   cm.nblocks, q->dtext, D.dim_text_len,
+                hipLaunchKernelGGL((k_cust_build_mask<std::decay_t<decltype(*cs)>>),
+                                   dim3(GRID), dim3(TPB), 0, s,
+                                   ck.dstream, ck.m, q->dmask,
                                    q->cust->dvmap, cs, q->cmask,
-                                   q->bloom, q->bwmask, (int *) q->dmin);
+                                   q->bloom, q->bwmask);
             else
                 hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
                                    dim3(GRID), dim3(TPB), 0, s,
