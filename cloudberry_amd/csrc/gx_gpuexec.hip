@@ -349,43 +349,62 @@ __global__ void k_encode(const T *vals, int64_t nrows, int32_t rpb,
 }
 
 /* CRC32C (pg COMP_CRC32C state, no final xor — cdbappendonlystorageformat.c:
- * 41-47).  One thread per AO block; table built in LDS per workgroup. */
-__device__ __forceinline__ uint32_t d_crc32c(const uint32_t *tab, uint32_t crc,
+ * 41-47), SLICE-BY-8: 8 derived tables collapse the byte-serial chain to
+ * one XOR tree per 8 input bytes (the standard Intel slicing scheme —
+ * same polynomial, bit-identical result).  Tables built in LDS per
+ * workgroup (8 KB). */
+__device__ __forceinline__ uint32_t d_crc32c(const uint32_t (*tab)[256],
+                                             uint32_t crc,
                                              const uint8_t *p, int64_t len)
 {
-    /* bulk 8 bytes per iteration from a register (per-thread stride-1 stays
-     * in L1 across iterations) */
-    while (len >= 8 && ((uintptr_t) p & 7)) { crc = tab[(crc ^ *p++) & 0xFF] ^ (crc >> 8); len--; }
+    while (len > 0 && ((uintptr_t) p & 7))
+    {
+        crc = tab[0][(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+        len--;
+    }
     while (len >= 8)
     {
         uint64_t v = *(const uint64_t *) p;
-        for (int k = 0; k < 8; k++)
-        {
-            crc = tab[(crc ^ (uint32_t) v) & 0xFF] ^ (crc >> 8);
-            v >>= 8;
-        }
-        p += 8; len -= 8;
+        uint32_t lo = (uint32_t) v ^ crc;
+        uint32_t hi = (uint32_t) (v >> 32);
+        crc = tab[7][lo & 0xFF] ^ tab[6][(lo >> 8) & 0xFF] ^
+              tab[5][(lo >> 16) & 0xFF] ^ tab[4][lo >> 24] ^
+              tab[3][hi & 0xFF] ^ tab[2][(hi >> 8) & 0xFF] ^
+              tab[1][(hi >> 16) & 0xFF] ^ tab[0][hi >> 24];
+        p += 8;
+        len -= 8;
     }
-    while (len--) crc = tab[(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+    while (len--)
+        crc = tab[0][(crc ^ *p++) & 0xFF] ^ (crc >> 8);
     return crc;
 }
 
-__device__ void d_crc_table_init(uint32_t *tab)
+__device__ void d_crc_table_init(uint32_t (*tab)[256])
 {
     for (int i = threadIdx.x; i < 256; i += blockDim.x)
     {
         uint32_t c = i;
         for (int k = 0; k < 8; k++)
             c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
-        tab[i] = c;
+        tab[0][i] = c;
     }
     __syncthreads();
+    /* tab[k][i] = crc of byte i followed by k zero bytes */
+    for (int k = 1; k < 8; k++)
+    {
+        for (int i = threadIdx.x; i < 256; i += blockDim.x)
+        {
+            uint32_t c = tab[k - 1][i];
+            tab[k][i] = tab[0][c & 0xFF] ^ (c >> 8);
+        }
+        __syncthreads();
+    }
 }
 
 __global__ void k_crc_fill(uint8_t *stream, int64_t nblocks, int64_t full_len,
                            int64_t nrows, int32_t rpb, int32_t width)
 {
-    __shared__ uint32_t tab[256];
+    __shared__ uint32_t tab[8][256];
     d_crc_table_init(tab);
     int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
@@ -434,7 +453,7 @@ __global__ void k_decode(const uint8_t *stream, int64_t nblocks, int64_t full_le
 __global__ void k_verify_crc(const uint8_t *stream, int64_t nblocks, int64_t full_len,
                              int64_t nrows, int32_t rpb, int32_t width, int *err)
 {
-    __shared__ uint32_t tab[256];
+    __shared__ uint32_t tab[8][256];
     d_crc_table_init(tab);
     int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
@@ -452,7 +471,7 @@ __global__ void k_verify_crc(const uint8_t *stream, int64_t nblocks, int64_t ful
 __global__ void k_verify_crc_dir(const uint8_t *stream, const gx_blockref *dir,
                                  int64_t nblocks, int *err)
 {
-    __shared__ uint32_t tab[256];
+    __shared__ uint32_t tab[8][256];
     d_crc_table_init(tab);
     int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
