@@ -706,3 +706,23 @@ def test_varlena_encoder_byte_exact_vs_reference_writer():
     mbig = orc.aocs_encode_varlena_rle(big)
     assert mbig == orc.ref_writer_varlena_stream(big, version=2, rle=1)
     assert orc.aocs_decode_varlena(mbig, len(big)) == big
+
+
+def test_varlena_short_form_boundary():
+    """The short-form threshold: payload <= 126 B stores as 1-byte-header
+    short varlena, 127 B+ stores aligned with a 4-byte header
+    (VARATT_CAN_MAKE_SHORT, varatt.h:261-264).  Byte-exact vs the
+    reference writer right at the boundary, both formats."""
+    if orc.ref_writer() is None:
+        pytest.skip("reference writer not built")
+    vals = []
+    for ln in (0, 1, 125, 126, 127, 128, 300):
+        vals += [bytes([65 + (ln % 26)]) * ln] * 3    # runs of 3 each
+    for ver, rle in ((0, 0), (2, 1)):
+        if ver == 0:
+            ours = orc.aocs_encode_varlena(vals)
+        else:
+            ours = orc.aocs_encode_varlena_rle(vals)
+        ref = orc.ref_writer_varlena_stream(vals, version=ver, rle=rle)
+        assert ours == ref, f"v{ver} boundary diverges"
+        assert orc.aocs_decode_varlena(ours, len(vals)) == vals
