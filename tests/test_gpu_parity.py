@@ -938,3 +938,20 @@ def test_gpu_sf100_full_size_properties(ctx, orc):
     # numeric revenue is an EXACT integer with implied scale 1e-4
     np.testing.assert_allclose(rn["revenue_num"] * 1e-4, r["revenue"],
                                rtol=1e-9)
+
+
+@pytest.mark.gpu
+def test_gpu_sf100_rle_mode_cross_check(ctx, orc):
+    """Full-size (SF100) cross-validation of the FUSED RLE probe: Q3 over
+    rle_type l_orderkey must equal Q3 over the plain stream — same 600M
+    rows through two different scan kernels."""
+    sf = 100.0
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    r = ctx.q3(cust, ordr, li).run().result()
+    li_rle = ctx.tpch_gen(gx.TPCH_LINEITEM_RLEKEY, sf)
+    rr = ctx.q3(cust, ordr, li_rle).run().result()
+    np.testing.assert_array_equal(rr["l_orderkey"], r["l_orderkey"])
+    np.testing.assert_array_equal(rr["nitems"], r["nitems"])
+    np.testing.assert_allclose(rr["revenue"], r["revenue"], rtol=1e-12)
