@@ -91,7 +91,7 @@ static constexpr int GRID = 2048;        /* grid-stride grid (fills 256 CUs) */
  * them automatically.  Long-lived state stays owned by gx_table / gx_q3. */
 struct devbuf {
     void *p = nullptr;
-    ~devbuf() { if (p) (void) hipFree(p); }
+    ~devbuf() { if (p) (void) (void) hipFree(p); }
     hipError_t alloc(size_t bytes) { return hipMalloc(&p, bytes ? bytes : 1); }
     template <typename T> T *as() const { return (T *) p; }
 };
@@ -2168,7 +2168,7 @@ extern "C" gx_status gx_shutdown(gx_ctx *ctx)
 {
     if (!ctx) return GX_OK;
     if (ctx->comm) ncclCommDestroy(ctx->comm);
-    if (ctx->stream) hipStreamDestroy(ctx->stream);
+    if (ctx->stream) (void) hipStreamDestroy(ctx->stream);
     delete ctx;
     return GX_OK;
 }
@@ -2532,19 +2532,28 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
             col.nblocks = (int64_t) dir.size();
             hipError_t e = hipMalloc(&col.ddir, dir.size() * sizeof(gx_blockref));
             if (e != hipSuccess) { delete t; return GX_ERR_OOM; }
-            hipMemcpyAsync(col.ddir, dir.data(), dir.size() * sizeof(gx_blockref),
-                           hipMemcpyHostToDevice, ctx->stream);
-            hipStreamSynchronize(ctx->stream);
+            e = hipMemcpyAsync(col.ddir, dir.data(),
+                               dir.size() * sizeof(gx_blockref),
+                               hipMemcpyHostToDevice, ctx->stream);
+            if (e == hipSuccess)
+                e = hipStreamSynchronize(ctx->stream);
+            if (e != hipSuccess)
+            {
+                set_err(ctx, "dir upload: %s", hipGetErrorString(e));
+                delete t;
+                return GX_ERR_HIP;
+            }
         }
         hipError_t e = hipMalloc(&col.dstream, stream_len);
         if (e != hipSuccess) { set_err(ctx, "hipMalloc: %s", hipGetErrorString(e)); delete t; return GX_ERR_OOM; }
         e = hipMemcpyAsync(col.dstream, stream_src, stream_len,
                            hipMemcpyHostToDevice, ctx->stream);
         if (e != hipSuccess) { set_err(ctx, "hipMemcpy: %s", hipGetErrorString(e)); delete t; return GX_ERR_HIP; }
-        hipStreamSynchronize(ctx->stream);      /* inflated buffer is stack-local */
+        e = hipStreamSynchronize(ctx->stream);  /* inflated buffer is stack-local */
+        if (e != hipSuccess) { set_err(ctx, "bind sync: %s", hipGetErrorString(e)); delete t; return GX_ERR_HIP; }
         t->cols.push_back(col);
     }
-    hipStreamSynchronize(ctx->stream);
+    (void) hipStreamSynchronize(ctx->stream);
     *out = t;
     return GX_OK;
 }
@@ -2554,10 +2563,10 @@ extern "C" gx_status gx_table_free(gx_table *t)
     if (!t) return GX_OK;
     for (auto &c : t->cols)
     {
-        if (c.dstream) hipFree(c.dstream);
-        if (c.ddir) hipFree(c.ddir);
+        if (c.dstream) (void) hipFree(c.dstream);
+        if (c.ddir) (void) hipFree(c.ddir);
     }
-    if (t->dvmap) hipFree(t->dvmap);
+    if (t->dvmap) (void) hipFree(t->dvmap);
     delete t;
     return GX_OK;
 }
@@ -2573,7 +2582,7 @@ extern "C" gx_status gx_table_set_visimap(gx_ctx *ctx, gx_table *t,
     if (!ctx || !t) return GX_ERR_INVALID;
     if (t->dvmap)
     {
-        hipFree(t->dvmap);
+        (void) hipFree(t->dvmap);
         t->dvmap = nullptr;
     }
     if (bitmap)
@@ -2653,7 +2662,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                            seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
         uint64_t *doffs; int64_t n;
         st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
-        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        if (st != GX_OK) { delete t; (void) hipFree(dcounts); return st; }
         int64_t *dkey; uint8_t *dmkt;
         HIP_CHK(ctx, hipMalloc(&dkey, n * 8));
         HIP_CHK(ctx, hipMalloc(&dmkt, std::max<int64_t>(n, 1)));
@@ -2663,7 +2672,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         st = encode_column_device(ctx, dkey, 8, n, &c0);
         if (st == GX_OK) st = encode_column_device(ctx, dmkt, 1, n, &c1);
         HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-        hipFree(dkey); hipFree(dmkt); hipFree(doffs);
+        (void) hipFree(dkey); (void) hipFree(dmkt); (void) hipFree(doffs);
         t->cols = {c0, c1};
         t->nrows = n;
     }
@@ -2673,7 +2682,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                            seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
         uint64_t *doffs; int64_t n;
         st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
-        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        if (st != GX_OK) { delete t; (void) hipFree(dcounts); return st; }
         int64_t *dok, *doc; int32_t *dod, *dop;
         HIP_CHK(ctx, hipMalloc(&dok, n * 8));
         HIP_CHK(ctx, hipMalloc(&doc, n * 8));
@@ -2687,7 +2696,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         if (st == GX_OK) st = encode_column_device(ctx, dod, 4, n, &c2);
         if (st == GX_OK) st = encode_column_device(ctx, dop, 4, n, &c3);
         HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-        hipFree(dok); hipFree(doc); hipFree(dod); hipFree(dop); hipFree(doffs);
+        (void) hipFree(dok); (void) hipFree(doc); (void) hipFree(dod); (void) hipFree(dop); (void) hipFree(doffs);
         t->cols = {c0, c1, c2, c3};
         t->nrows = n;
     }
@@ -2697,7 +2706,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                            seed, nglobal, ctx->seg, ctx->nsegs, dcounts);
         uint64_t *doffs; int64_t n;
         st = scan_counts(ctx, dcounts, nthreads, &doffs, &n);
-        if (st != GX_OK) { delete t; hipFree(dcounts); return st; }
+        if (st != GX_OK) { delete t; (void) hipFree(dcounts); return st; }
         if (li_q1)
         {
             int8_t *dfl, *dst2;
@@ -2718,8 +2727,8 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
             if (st == GX_OK) st = encode_column_device(ctx, ddi1, 8, n, &q3c);
             if (st == GX_OK) st = encode_column_device(ctx, dsh1, 4, n, &q4);
             HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-            hipFree(dfl); hipFree(dst2); hipFree(dpr1); hipFree(ddi1);
-            hipFree(dsh1); hipFree(doffs); hipFree(dcounts);
+            (void) hipFree(dfl); (void) hipFree(dst2); (void) hipFree(dpr1); (void) hipFree(ddi1);
+            (void) hipFree(dsh1); (void) hipFree(doffs); (void) hipFree(dcounts);
             if (st != GX_OK) { gx_table_free(t); return st; }
             t->cols = {q0, q1c, q2, q3c, q4};
             t->nrows = n;
@@ -2752,7 +2761,7 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
             int64_t rows = 0;
             if (parse_block_dir(stream_bytes.data(), (int64_t) stream_bytes.size(),
                                 dir, &rows) != GX_OK || rows != n)
-            { set_err(ctx, "rle self-encode mismatch%s", ""); delete t; hipFree(dcounts); return GX_ERR_INVALID; }
+            { set_err(ctx, "rle self-encode mismatch%s", ""); delete t; (void) hipFree(dcounts); return GX_ERR_INVALID; }
             c0.format = 1;
             c0.nblocks = (int64_t) dir.size();
             c0.m.width = 8;
@@ -2776,11 +2785,11 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         if (st == GX_OK) st = encode_column_device(ctx, ddi, 8, n, &c2);
         if (st == GX_OK) st = encode_column_device(ctx, dsh, 4, n, &c3);
         HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-        hipFree(dlk); hipFree(dpr); hipFree(ddi); hipFree(dsh); hipFree(doffs);
+        (void) hipFree(dlk); (void) hipFree(dpr); (void) hipFree(ddi); (void) hipFree(dsh); (void) hipFree(doffs);
         t->cols = {c0, c1, c2, c3};
         t->nrows = n;
     }
-    hipFree(dcounts);
+    (void) hipFree(dcounts);
     HIP_CHK(ctx, hipGetLastError());
     if (st != GX_OK) { gx_table_free(t); return st; }
     *out = t;
@@ -3141,7 +3150,7 @@ extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
     HIP_CHK(ctx, hipMemsetAsync(spr.p, 0, 48, s));
     HIP_CHK(ctx, hipMemsetAsync(srv.p, 0, 48, s));
     hipEvent_t e0, e1;
-    hipEventCreate(&e0); hipEventCreate(&e1);
+    (void) hipEventCreate(&e0); (void) hipEventCreate(&e1);
     HIP_CHK(ctx, hipEventRecord(e0, s));
     hipLaunchKernelGGL(k_q1_agg, dim3(GRID), dim3(TPB), 0, s,
                        t->cols[0].dstream, t->cols[0].m,
@@ -3158,8 +3167,8 @@ extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
     HIP_CHK(ctx, hipStreamSynchronize(s));
     HIP_CHK(ctx, hipGetLastError());
     float ms = 0;
-    hipEventElapsedTime(&ms, e0, e1);
-    hipEventDestroy(e0); hipEventDestroy(e1);
+    (void) hipEventElapsedTime(&ms, e0, e1);
+    (void) hipEventDestroy(e0); (void) hipEventDestroy(e1);
     if (ms_out) *ms_out = (double) ms;
     return GX_OK;
 }
@@ -3178,7 +3187,7 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     HIP_CHK(ctx, cnt.alloc(8));
     HIP_CHK(ctx, hipMemsetAsync(cnt.p, 0, 8, s));
     hipEvent_t e0, e1;
-    hipEventCreate(&e0); hipEventCreate(&e1);
+    (void) hipEventCreate(&e0); (void) hipEventCreate(&e1);
     HIP_CHK(ctx, hipEventRecord(e0, s));
     if (c.m.width == 8)
         hipLaunchKernelGGL(k_scan_filter<int64_t>, dim3(GRID), dim3(TPB), 0, s,
@@ -3198,8 +3207,8 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     HIP_CHK(ctx, hipStreamSynchronize(s));
     HIP_CHK(ctx, hipGetLastError());
     float ms = 0;
-    hipEventElapsedTime(&ms, e0, e1);
-    hipEventDestroy(e0); hipEventDestroy(e1);
+    (void) hipEventElapsedTime(&ms, e0, e1);
+    (void) hipEventDestroy(e0); (void) hipEventDestroy(e1);
     *count_out = (int64_t) n;
     if (ms_out) *ms_out = (double) ms;
     return GX_OK;
@@ -3339,12 +3348,12 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
         if (e != hipSuccess) { delete q; return GX_ERR_OOM; }
         e = hipMemcpy(q->dtext, desc->dim_text, desc->dim_text_len,
                       hipMemcpyHostToDevice);
-        if (e != hipSuccess) { hipFree(q->dtext); delete q; return GX_ERR_OOM; }
+        if (e != hipSuccess) { (void) hipFree(q->dtext); delete q; return GX_ERR_OOM; }
         gx_status st = q3_build_text_mask(ctx, q);
         if (st != GX_OK)
         {
-            hipFree(q->dtext);
-            if (q->dmask) hipFree(q->dmask);
+            (void) hipFree(q->dtext);
+            if (q->dmask) (void) hipFree(q->dmask);
             delete q;
             return st;
         }
@@ -3380,7 +3389,7 @@ extern "C" gx_status gx_q3_prepare(gx_ctx *ctx, gx_table *customer, gx_table *or
 
 static void q3_free_runstate(gx_q3 *q)
 {
-    auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
+    auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
     fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
@@ -3634,7 +3643,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             /* experiment: homogeneous filter/emit scan + separate insert */
             if (!q->m_send2 || q->m_send2_cap < (uint64_t) q->rescap)
             {
-                if (q->m_send2) hipFree(q->m_send2);
+                if (q->m_send2) (void) hipFree(q->m_send2);
                 q->m_send2 = nullptr; q->m_send2_cap = 0;
                 HIP_CHK(ctx, hipMalloc(&q->m_send2,
                                        std::max<int64_t>(q->rescap, 1) *
@@ -3684,7 +3693,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (!ctx->comm) { set_err(ctx, "nsegs>1 but gx_comm_init not called%s", ""); return GX_ERR_STATE; }
         int n = ctx->nsegs;
         hipEvent_t mev0, mev1;
-        hipEventCreate(&mev0); hipEventCreate(&mev1);
+        (void) hipEventCreate(&mev0); (void) hipEventCreate(&mev1);
         HIP_CHK(ctx, hipEventRecord(mev0, s));
 
         /* Motion 1: filtered orders by route(o_custkey).  Exchange buffers
@@ -3692,7 +3701,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
          * dominate at high rank counts. */
         auto grow = [&](auto *&p, uint64_t &cap, uint64_t want) -> gx_status {
             if (want <= cap) return GX_OK;
-            if (p) hipFree(p);
+            if (p) (void) hipFree(p);
             p = nullptr;
             cap = 0;
             hipError_t e = hipMalloc(&p, std::max<uint64_t>(want, 1) *
@@ -3840,7 +3849,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width ||
             qual > q->rescap)
         {
-            auto fr = [](auto *&p) { if (p) { hipFree(p); p = nullptr; } };
+            auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
             fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
             q->key_width = want_kw;
@@ -3880,9 +3889,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0;
-        hipEventElapsedTime(&mms, mev0, mev1);
+        (void) hipEventElapsedTime(&mms, mev0, mev1);
         ms_motion = mms;
-        hipEventDestroy(mev0); hipEventDestroy(mev1);
+        (void) hipEventDestroy(mev0); (void) hipEventDestroy(mev1);
     }
     q->qual_orders = qual;
     HIP_CHK(ctx, hipEventRecord(ev[2], s));
@@ -4048,11 +4057,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     q->ngroups = (int64_t) ngroups;
 
     float t01 = 0, t12 = 0, t23 = 0, t34 = 0, t04 = 0;
-    hipEventElapsedTime(&t01, ev[0], ev[1]);
-    hipEventElapsedTime(&t12, ev[1], ev[2]);
-    hipEventElapsedTime(&t23, ev[2], ev[3]);
-    hipEventElapsedTime(&t34, ev[3], ev[4]);
-    hipEventElapsedTime(&t04, ev[0], ev[4]);
+    (void) hipEventElapsedTime(&t01, ev[0], ev[1]);
+    (void) hipEventElapsedTime(&t12, ev[1], ev[2]);
+    (void) hipEventElapsedTime(&t23, ev[2], ev[3]);
+    (void) hipEventElapsedTime(&t34, ev[3], ev[4]);
+    (void) hipEventElapsedTime(&t04, ev[0], ev[4]);
     q->stats.ms_cust_build = t01;
     q->stats.ms_orders_build = std::max(0.0, (double) t12 - ms_motion);
     q->stats.ms_motion = ms_motion;
@@ -4071,7 +4080,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     q->stats.bytes_scanned = bb;
 
     /* dcount/dhits are cached run-state, freed in gx_q3_free */
-    for (auto &e : ev) hipEventDestroy(e);
+    for (auto &e : ev) (void) hipEventDestroy(e);
     q->ran = true;
     return GX_OK;
 }
@@ -4440,17 +4449,17 @@ extern "C" int gx_selftest_rccl(int device)
     devbuf a, b;
     if (a.alloc(8) != hipSuccess || b.alloc(8) != hipSuccess) return 4;
     unsigned long long v = 0xC0FFEE;
-    hipMemcpy(a.p, &v, 8, hipMemcpyHostToDevice);
+    (void) hipMemcpy(a.p, &v, 8, hipMemcpyHostToDevice);
     hipStream_t s;
-    hipStreamCreate(&s);
+    if (hipStreamCreate(&s) != hipSuccess) return 4;
     int rc = 0;
     if (ncclAllGather(a.p, b.p, 1, ncclUint64, comm, s) != ncclSuccess)
         rc = 5;
-    hipStreamSynchronize(s);
+    (void) hipStreamSynchronize(s);
     unsigned long long w = 0;
-    hipMemcpy(&w, b.p, 8, hipMemcpyDeviceToHost);
+    (void) hipMemcpy(&w, b.p, 8, hipMemcpyDeviceToHost);
     if (rc == 0 && w != v) rc = 6;
-    hipStreamDestroy(s);
+    (void) hipStreamDestroy(s);
     ncclCommDestroy(comm);
     return rc;
 }
