@@ -1,4 +1,5 @@
 /* bw_probe — measures achievable HBM read bandwidth on gfx950 for the
+ * (build: hipcc --offload-arch=gfx950 -O3 tools/bw_probe.hip -o tools/bw_probe)
  * access shapes our scan kernels use: plain 8B grid-stride, 16B (uint4-
  * style) grid-stride, nontemporal 8B, and block-chunked 8B.  Prints
  * GB/s per (shape, grid) so kernel targets are set from MEASURED
