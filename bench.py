@@ -107,9 +107,11 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    n = max(world, 1)
-    assert n == args.gpus or world == 1, \
-        f"WORLD_SIZE={world} but --gpus={args.gpus}"
+    # --gpus N must match the actual process count: a 1-process run may not
+    # stamp n_gpus>1 on the record
+    assert max(world, 1) == args.gpus, \
+        f"WORLD_SIZE={world} but --gpus={args.gpus}: launch one rank per GPU " \
+        f"(torch.distributed.run --nproc-per-node {args.gpus})"
     n = args.gpus
 
     import numpy as np

@@ -96,6 +96,15 @@ struct devbuf {
     template <typename T> T *as() const { return (T *) p; }
 };
 
+/* RAII for HIP events: early returns (HIP_CHK/RCCL_CHK/OOM) destroy them
+ * automatically instead of accumulating across repeated failing steps */
+struct evholder {
+    hipEvent_t e = nullptr;
+    ~evholder() { if (e) (void) hipEventDestroy(e); }
+    hipError_t create() { return hipEventCreate(&e); }
+    operator hipEvent_t() const { return e; }
+};
+
 static inline int env_int(const char *name, int dflt)
 {
     const char *v = getenv(name);
@@ -777,11 +786,12 @@ __global__ void k_texteq_mask(const int64_t *offsets, const uint8_t *payload,
 __global__ void k_cust_count_mask(const uint8_t *key_s, gx_colmeta key_m,
                                   const uint8_t *mask, const uint8_t *vmap,
                                   unsigned long long *count,
-                                  unsigned long long *maxkey)
+                                  unsigned long long *maxkey,
+                                  unsigned long long *minkey)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    unsigned long long local = 0, kmax = 0;
+    unsigned long long local = 0, kmax = 0, kmin = ~0ULL;
     for (; i < key_m.nrows; i += stride)
         if (mask[i] && !gx_vm_hidden(vmap, i))
         {
@@ -789,15 +799,21 @@ __global__ void k_cust_count_mask(const uint8_t *key_s, gx_colmeta key_m,
             unsigned long long k = (unsigned long long)
                 gx_col_get<int64_t>(key_s, key_m, i);
             if (k > kmax) kmax = k;
+            if (k < kmin) kmin = k;
         }
     gx_wave_count_add(count, local);
     for (int o = 32; o; o >>= 1)
     {
         unsigned long long v = __shfl_down(kmax, o, 64);
         if (v > kmax) kmax = v;
+        unsigned long long w = __shfl_down(kmin, o, 64);
+        if (w < kmin) kmin = w;
     }
-    if ((threadIdx.x & 63) == 0 && kmax)
-        atomicMax(maxkey, kmax);
+    if ((threadIdx.x & 63) == 0)
+    {
+        if (kmax) atomicMax(maxkey, kmax);
+        if (kmin != ~0ULL) atomicMin(minkey, kmin);
+    }
 }
 
 template <typename KS>
@@ -828,11 +844,12 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
                              const uint8_t *mkt_s, gx_colmeta mkt_m,
                              const uint8_t *vmap, int cop, int8_t clit,
                              unsigned long long *count,
-                             unsigned long long *maxkey)
+                             unsigned long long *maxkey,
+                             unsigned long long *minkey)
 {
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    unsigned long long local = 0, kmax = 0;
+    unsigned long long local = 0, kmax = 0, kmin = ~0ULL;
     for (; i < mkt_m.nrows; i += stride)
         if (!gx_vm_hidden(vmap, i) &&
             gx_cmp(cop, gx_col_get<int8_t>(mkt_s, mkt_m, i), clit))
@@ -840,15 +857,21 @@ __global__ void k_cust_count(const uint8_t *key_s, gx_colmeta key_m,
             local++;
             unsigned long long k = (unsigned long long) gx_col_get<int64_t>(key_s, key_m, i);
             if (k > kmax) kmax = k;
+            if (k < kmin) kmin = k;
         }
     gx_wave_count_add(count, local);
     for (int o = 32; o; o >>= 1)
     {
         unsigned long long v = __shfl_down(kmax, o, 64);
         if (v > kmax) kmax = v;
+        unsigned long long w = __shfl_down(kmin, o, 64);
+        if (w < kmin) kmin = w;
     }
-    if ((threadIdx.x & 63) == 0 && kmax)
-        atomicMax(maxkey, kmax);
+    if ((threadIdx.x & 63) == 0)
+    {
+        if (kmax) atomicMax(maxkey, kmax);
+        if (kmin != ~0ULL) atomicMin(minkey, kmin);
+    }
 }
 
 /* customer: filter mktsegment=BUILDING, insert c_custkey into open set.
@@ -882,6 +905,10 @@ template <typename KS>
 __device__ __forceinline__ bool d_set_contains(const KS *set,
                                                uint64_t mask, uint64_t k)
 {
+    /* narrow-set guard: u32 slots mean every RESIDENT key < 2^32 (the sizing
+     * pass's max is over build-side keys only); a probe key ≥ 2^32 can never
+     * match and must not be compared truncated */
+    if (sizeof(KS) == 4 && (k >> 32)) return false;
     uint64_t slot = gx_hmix64(k) & mask;
     while (true)
     {
@@ -1017,6 +1044,9 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
     uint64_t tmask = smap.mask;
     unsigned long long local_hits = 0;
     auto resolve = [&](uint64_t k, uint64_t slot, KT v) -> uint64_t {
+        /* narrow-table guard: u32 slots hold only build keys < 2^32; a probe
+         * key ≥ 2^32 must miss, not match its truncated low half */
+        if (sizeof(KT) == 4 && (k >> 32)) return ~0ULL;
         /* first slot already loaded as v; walk on collision */
         while (true)
         {
@@ -1546,6 +1576,7 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
         for (int p = threadIdx.x; p < phys; p += blockDim.x)
         {
             uint64_t k = (uint64_t) datum[p];
+            if (sizeof(KT) == 4 && (k >> 32)) continue;  /* narrow-table guard */
             uint64_t slot = smap.slot0(k);
             bool found = false;
             while (true)
@@ -1603,6 +1634,7 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
             if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+        if (sizeof(KT) == 4 && (k >> 32)) continue;  /* narrow-table guard */
         uint64_t slot = smap.slot0(k);
         bool found = false;
         while (true)
@@ -1904,11 +1936,17 @@ __global__ void k_rows_minmax(const gx_qual_row *rows, int64_t n,
 }
 
 /* received qualifying orders → build the join/agg table */
+/* dn (optional, device): the count actually emitted by a producer kernel —
+ * authoritative over n when the two could diverge (e.g. GX_ORDERS_TWOPASS,
+ * where n is the prepare-time capacity; inserting up to capacity would read
+ * uninitialized rows if the emit ever fell short) */
 template <typename KT>
 __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
+                                  const unsigned long long *dn,
                                   KT *tkey,
                                   int32_t *tdate, int32_t *tprio, gx_slotmap smap)
 {
+    if (dn) n = min(n, (int64_t) *dn);
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
     int64_t stride = gridDim.x * (int64_t) blockDim.x;
     uint64_t tmask = smap.mask;
@@ -3149,8 +3187,8 @@ extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
     HIP_CHK(ctx, hipMemsetAsync(cnt.p, 0, 48, s));
     HIP_CHK(ctx, hipMemsetAsync(spr.p, 0, 48, s));
     HIP_CHK(ctx, hipMemsetAsync(srv.p, 0, 48, s));
-    hipEvent_t e0, e1;
-    (void) hipEventCreate(&e0); (void) hipEventCreate(&e1);
+    evholder e0, e1;
+    HIP_CHK(ctx, e0.create()); HIP_CHK(ctx, e1.create());
     HIP_CHK(ctx, hipEventRecord(e0, s));
     hipLaunchKernelGGL(k_q1_agg, dim3(GRID), dim3(TPB), 0, s,
                        t->cols[0].dstream, t->cols[0].m,
@@ -3168,7 +3206,6 @@ extern "C" gx_status gx_q1(gx_ctx *ctx, const gx_table *t, int32_t cutoff,
     HIP_CHK(ctx, hipGetLastError());
     float ms = 0;
     (void) hipEventElapsedTime(&ms, e0, e1);
-    (void) hipEventDestroy(e0); (void) hipEventDestroy(e1);
     if (ms_out) *ms_out = (double) ms;
     return GX_OK;
 }
@@ -3186,8 +3223,8 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     devbuf cnt;
     HIP_CHK(ctx, cnt.alloc(8));
     HIP_CHK(ctx, hipMemsetAsync(cnt.p, 0, 8, s));
-    hipEvent_t e0, e1;
-    (void) hipEventCreate(&e0); (void) hipEventCreate(&e1);
+    evholder e0, e1;
+    HIP_CHK(ctx, e0.create()); HIP_CHK(ctx, e1.create());
     HIP_CHK(ctx, hipEventRecord(e0, s));
     if (c.m.width == 8)
         hipLaunchKernelGGL(k_scan_filter<int64_t>, dim3(GRID), dim3(TPB), 0, s,
@@ -3208,7 +3245,6 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     HIP_CHK(ctx, hipGetLastError());
     float ms = 0;
     (void) hipEventElapsedTime(&ms, e0, e1);
-    (void) hipEventDestroy(e0); (void) hipEventDestroy(e1);
     *count_out = (int64_t) n;
     if (ms_out) *ms_out = (double) ms;
     return GX_OK;
@@ -3401,6 +3437,33 @@ static void q3_free_runstate(gx_q3 *q)
     q->sized = false;
 }
 
+/* HBM-budget guard: the reference batches/spills when the
+ * build side exceeds its memory budget (ExecChooseHashTableSize nodeHash.c:812,
+ * batch split nodeHashjoin.c:1355); the GPU path instead sizes up front and
+ * REJECTS with the numbers in the message before allocating — no silent
+ * fallback (GX_ERR_NOGPU semantics: callers must fail, not fall back).
+ * GX_HBM_BUDGET_MB overrides the free-memory query (tests). */
+static gx_status hbm_budget_check(gx_ctx *ctx, uint64_t want, const char *what)
+{
+    size_t free_b = 0, total_b = 0;
+    HIP_CHK(ctx, hipMemGetInfo(&free_b, &total_b));
+    long mb = env_int("GX_HBM_BUDGET_MB", 0);
+    uint64_t budget = mb > 0 ? (uint64_t) mb << 20
+                             : (uint64_t) free_b - ((uint64_t) free_b >> 4);
+    if (want > budget)
+    {
+        char buf[200];
+        snprintf(buf, sizeof buf,
+                 "%s needs %.2f GB but HBM budget is %.2f GB (free %.2f of "
+                 "%.2f GB); reduce the build side or add segments",
+                 what, want / 1073741824.0, budget / 1073741824.0,
+                 free_b / 1073741824.0, total_b / 1073741824.0);
+        set_err(ctx, "%s", buf);
+        return GX_ERR_OOM;
+    }
+    return GX_OK;
+}
+
 /* first run only: size the customer set and join/agg table from counting
  * passes, allocate everything once (reused across bench steps) */
 static gx_status q3_size_and_alloc(gx_q3 *q)
@@ -3417,24 +3480,42 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
 
     HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
     HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for max custkey */
+    HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s)); /* borrowed for min custkey */
     if (D.dim_text_len > 0)
         hipLaunchKernelGGL(k_cust_count_mask, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
                            q->dmask, q->cust->dvmap,
-                           q->dcount, q->dhits);
+                           q->dcount, q->dhits, q->dmin);
     else
         hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
                            cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
-                           (int8_t) D.dim_filter.literal, q->dcount, q->dhits);
-    unsigned long long n_building = 0, cmax = 0;
+                           (int8_t) D.dim_filter.literal, q->dcount, q->dhits,
+                           q->dmin);
+    unsigned long long n_building = 0, cmax = 0, cmin = 0;
     HIP_CHK(ctx, hipMemcpyAsync(&n_building, q->dcount, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&cmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
+    if (n_building > 0 && cmin == 0)
+    {
+        /* slot value 0 is the empty-slot sentinel (keys start at 1 in every
+         * PG sequence-keyed table); a real key 0 would be silently dropped */
+        set_err(ctx, "dim join key 0 found: 0 is reserved as the empty-slot "
+                     "sentinel (gpuexec.h gx_q3_prepare_desc)%s", "");
+        return GX_ERR_INVALID;
+    }
     q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
     uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
+    {
+        uint64_t bw = (uint64_t) pow2_at_least(
+            std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
+        gx_status st = hbm_budget_check(ctx, cslots * q->cset_width + bw * 8,
+                                        "dim semijoin set");
+        if (st != GX_OK) return st;
+    }
     HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * q->cset_width, s));
     q->cmask = cslots - 1;
@@ -3502,9 +3583,24 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMemcpyAsync(&kmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         qual = (int64_t) nq;
+        if (qual > 0 && kmin == 0)
+        {
+            set_err(ctx, "mid join key 0 found among qualifying rows: 0 is "
+                         "the empty-slot sentinel (gpuexec.h "
+                         "gx_q3_prepare_desc)%s", "");
+            return GX_ERR_INVALID;
+        }
         q->key_width = (kmax < (1ULL << 32)) ? 4 : 8;
         int tf = env_int("GX_TABLE_FACTOR_PCT", 200);
         uint64_t tslots = (uint64_t) pow2_at_least(qual * tf / 100 + 1);
+        {
+            gx_status st = hbm_budget_check(
+                ctx,
+                tslots * (q->key_width + 4 + 4 + 8 + 8) +
+                    (uint64_t) std::max<int64_t>(qual, 1) * (8 + 4 + 4 + 8 + 8),
+                "join/agg table");
+            if (st != GX_OK) return st;
+        }
         q->tmask = tslots - 1;
         /* order-preserving interpolation layout when the qualifying keys'
          * density over [kmin,kmax] is high enough that runs stay short;
@@ -3554,8 +3650,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (st != GX_OK) return st;
     }
 
-    hipEvent_t ev[8];
-    for (auto &e : ev) HIP_CHK(ctx, hipEventCreate(&e));
+    evholder ev[8];
+    for (auto &e : ev) HIP_CHK(ctx, e.create());
 
     const gx_q3_desc &D = q->desc;
     const gx_col &ck = q->cust->cols[D.dim_key_col],
@@ -3668,13 +3764,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             if (q->key_width == 4)
                 hipLaunchKernelGGL(k_build_from_rows<unsigned int>,
                                    dim3(GRID), dim3(TPB), 0, s,
-                                   q->m_send2, q->rescap,
+                                   q->m_send2, q->rescap, q->dcount,
                                    (unsigned int *) q->tkey,
                                    q->tdate, q->tprio, q->smap);
             else
                 hipLaunchKernelGGL(k_build_from_rows<unsigned long long>,
                                    dim3(GRID), dim3(TPB), 0, s,
-                                   q->m_send2, q->rescap,
+                                   q->m_send2, q->rescap, q->dcount,
                                    (unsigned long long *) q->tkey,
                                    q->tdate, q->tprio, q->smap);
         }
@@ -3692,8 +3788,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         if (!ctx->comm) { set_err(ctx, "nsegs>1 but gx_comm_init not called%s", ""); return GX_ERR_STATE; }
         int n = ctx->nsegs;
-        hipEvent_t mev0, mev1;
-        (void) hipEventCreate(&mev0); (void) hipEventCreate(&mev1);
+        evholder mev0, mev1;
+        HIP_CHK(ctx, mev0.create()); HIP_CHK(ctx, mev1.create());
         HIP_CHK(ctx, hipEventRecord(mev0, s));
 
         /* Motion 1: filtered orders by route(o_custkey).  Exchange buffers
@@ -3842,6 +3938,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemcpyAsync(&kmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipMemcpyAsync(&kmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
+        if (qual > 0 && kmin == 0)
+        {
+            set_err(ctx, "mid join key 0 received: 0 is the empty-slot "
+                         "sentinel (gpuexec.h gx_q3_prepare_desc)%s", "");
+            return GX_ERR_INVALID;
+        }
         int want_kw = (kmax < (1ULL << 32)) ? 4 : 8;
         uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
         /* motion path sizes from the exchanged counts each run; qual can
@@ -3880,18 +3982,19 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         }
         if (q->key_width == 4)
             hipLaunchKernelGGL(k_build_from_rows<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               recv2, qual, (unsigned int *) q->tkey,
+                               recv2, qual, (const unsigned long long *) nullptr,
+                               (unsigned int *) q->tkey,
                                q->tdate, q->tprio, q->smap);
         else
             hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               recv2, qual, (unsigned long long *) q->tkey,
+                               recv2, qual, (const unsigned long long *) nullptr,
+                               (unsigned long long *) q->tkey,
                                q->tdate, q->tprio, q->smap);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0;
         (void) hipEventElapsedTime(&mms, mev0, mev1);
         ms_motion = mms;
-        (void) hipEventDestroy(mev0); (void) hipEventDestroy(mev1);
     }
     q->qual_orders = qual;
     HIP_CHK(ctx, hipEventRecord(ev[2], s));
@@ -4079,8 +4182,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     gx_table_logical_bytes(q->li, &b); bb += b;
     q->stats.bytes_scanned = bb;
 
-    /* dcount/dhits are cached run-state, freed in gx_q3_free */
-    for (auto &e : ev) (void) hipEventDestroy(e);
+    /* dcount/dhits are cached run-state, freed in gx_q3_free; events are
+     * RAII (evholder) */
     q->ran = true;
     return GX_OK;
 }
@@ -4271,13 +4374,15 @@ extern "C" gx_status gx_test_qual(gx_ctx *ctx, gx_table *customer,
     hipStream_t s = ctx->stream;
     const gx_col &ck = customer->cols[0], &cm = customer->cols[1];
     devbuf cnt_b, set_b, bloom_b, rows_b, hist_b, cur_b, send_b;
-    HIP_CHK(ctx, cnt_b.alloc(16));
+    HIP_CHK(ctx, cnt_b.alloc(24));
     HIP_CHK(ctx, hipMemsetAsync(cnt_b.p, 0, 16, s));
+    HIP_CHK(ctx, hipMemsetAsync((char *) cnt_b.p + 16, 0xFF, 8, s));
     hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
                        ck.dstream, ck.m, cm.dstream, cm.m,
                        (const uint8_t *) nullptr, 2, (int8_t) 0,
                        cnt_b.as<unsigned long long>(),
-                       cnt_b.as<unsigned long long>() + 1);
+                       cnt_b.as<unsigned long long>() + 1,
+                       cnt_b.as<unsigned long long>() + 2);
     unsigned long long nb[2];
     HIP_CHK(ctx, hipMemcpyAsync(nb, cnt_b.p, 16, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -4374,6 +4479,7 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
     HIP_CHK(ctx, hipMemsetAsync(cnt_b2.p, 0, tslots * 8, s));
     hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                        rows_b.as<gx_qual_row>(), n,
+                       (const unsigned long long *) nullptr,
                        key_b.as<unsigned long long>(), date_b.as<int32_t>(),
                        prio_b.as<int32_t>(), smap);
     const gx_col &lk = lineitem->cols[0], &lp = lineitem->cols[1],

@@ -227,6 +227,15 @@ typedef struct gx_q3_desc {
     int32_t dim_text_len;
 } gx_q3_desc;
 
+/* Restrictions checked at sizing (first gx_q3_run):
+ * - join keys must be >= 1: slot value 0 is the empty-slot sentinel (PG
+ *   sequence-keyed tables start at 1); a key 0 on the build side returns
+ *   GX_ERR_INVALID instead of silently dropping the row.
+ * - HBM budget: if the semijoin set / join table would exceed free HBM
+ *   (override: GX_HBM_BUDGET_MB), sizing fails with GX_ERR_OOM and the
+ *   required vs available GB in gx_last_error BEFORE any allocation — the
+ *   reference spills to batches (nodeHashjoin.c:1355); this executor
+ *   rejects, callers keep the reference CPU path for such plans. */
 gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc, gx_q3 **out);
 /* numeric(15,2) mode: lineitem measures are scaled int64
  * (price cents, discount hundredths — GX_TPCH_LINEITEM_NUMERIC tables);
