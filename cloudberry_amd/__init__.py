@@ -67,7 +67,12 @@ class _Q3Desc(ctypes.Structure):
                 ("fact_a_col", ctypes.c_int32), ("fact_b_col", ctypes.c_int32),
                 ("fact_filter", _Filter),
                 ("dim_text", ctypes.c_char * 64),
-                ("dim_text_len", ctypes.c_int32)]
+                ("dim_text_len", ctypes.c_int32),
+                ("dim_extra", _Filter * 4), ("mid_extra", _Filter * 4),
+                ("fact_extra", _Filter * 4),
+                ("n_dim_extra", ctypes.c_int32),
+                ("n_mid_extra", ctypes.c_int32),
+                ("n_fact_extra", ctypes.c_int32)]
 
 
 class _ColDesc(ctypes.Structure):
@@ -128,6 +133,10 @@ def _load():
                                    ctypes.POINTER(ctypes.c_double)]
     lib.gx_partition.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                                  ctypes.c_int32, ctypes.c_void_p]
+    lib.gx_partition_multi.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_int32, ctypes.c_int64,
+                                       ctypes.c_int32, ctypes.c_void_p]
     lib.gx_q3_prepare.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int32,
                                   ctypes.POINTER(ctypes.c_void_p)]
     lib.gx_q3_prepare_desc.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Q3Desc),
@@ -222,6 +231,23 @@ class Context:
                                          nsegs, out.ctypes.data))
         return out
 
+    def partition_multi(self, vals, types, nsegs, isnull=None):
+        """Multi-column distribution-key routing (cdbhash rotate-combine);
+        vals (n, nkeys) int64; types[k] 0 = int8, 1 = int4/date."""
+        vals = np.ascontiguousarray(vals, np.int64)
+        n, nkeys = vals.shape
+        types = np.ascontiguousarray(types, np.int32)
+        nul_ptr = None
+        if isnull is not None:
+            nul = np.ascontiguousarray(isnull, np.uint8)
+            assert nul.shape == vals.shape
+            nul_ptr = nul.ctypes.data
+        out = np.zeros(n, np.int32)
+        self._chk(self._lib.gx_partition_multi(
+            self._h, vals.ctypes.data, nul_ptr, types.ctypes.data,
+            nkeys, n, nsegs, out.ctypes.data))
+        return out
+
     def test_motion1(self, orders, nsegs, cutoff=CUTOFF_19950315):
         """Run the Motion-1 partition kernels; returns (counts, rows dict)."""
         cap = orders.nrows
@@ -307,6 +333,15 @@ class Context:
                 setattr(d, role, _Filter(col, ops[op], 0))
             else:
                 setattr(d, role, _Filter(col, ops[op], int(lit)))
+        # AND-ed extra qual lists (execScan.c:241 semantics)
+        for role, arr, cnt in (("dim_extra", d.dim_extra, "n_dim_extra"),
+                               ("mid_extra", d.mid_extra, "n_mid_extra"),
+                               ("fact_extra", d.fact_extra, "n_fact_extra")):
+            quals = desc_dict.get(role, [])
+            assert len(quals) <= 4
+            for i, (col, op, lit) in enumerate(quals):
+                arr[i] = _Filter(col, ops[op], int(lit))
+            setattr(d, cnt, len(quals))
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
                                                ctypes.byref(q)))

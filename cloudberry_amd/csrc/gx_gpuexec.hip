@@ -668,6 +668,22 @@ __global__ void k_route(const int64_t *keys, int64_t n, int32_t nsegs, int32_t *
         out[i] = gx_route_i64(keys[i], nsegs);
 }
 
+/* multi-column distribution keys (cdbhash.c:189-247 rotate-combine loop):
+ * row-major vals/isnull, per-key type tags in constant-ish arg memory */
+__global__ void k_route_multi(const int64_t *vals, const uint8_t *isnull,
+                              const int32_t *types, int32_t nkeys, int64_t n,
+                              int32_t nsegs, int32_t *out)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < n; i += stride)
+        out[i] = gx_jump_consistent_hash(
+            (uint64_t) gx_cdbhash_multi(vals + i * nkeys,
+                                        isnull ? isnull + i * nkeys : nullptr,
+                                        types, nkeys),
+            nsegs);
+}
+
 /* ================= Q3 kernels ================= */
 
 /* Join-table slot mapping.  When the build keys' [min,max] stats admit it
@@ -779,6 +795,51 @@ __global__ void k_texteq_mask(const int64_t *offsets, const uint8_t *payload,
     {
         int64_t len = offsets[i + 1] - offsets[i];
         mask[i] = (uint8_t) d_texteq(payload + offsets[i], len, lit, lit_len);
+    }
+}
+
+/* AND-ed extra qual lists (execScan.c:241: every qual in the list must
+ * pass; three-valued logic degenerates to two-valued on NOT NULL columns).
+ * Folded ONCE at prepare into a per-table HIDDEN bitmap (same encoding as
+ * the AO visimap, OR-combined with it), which every scan/build/probe
+ * kernel already honors through its visibility parameter. */
+struct gx_qualargs {
+    int32_t n;
+    const uint8_t *s[GX_MAX_EXTRA_QUALS];
+    gx_colmeta m[GX_MAX_EXTRA_QUALS];
+    int32_t op[GX_MAX_EXTRA_QUALS];
+    int64_t lit[GX_MAX_EXTRA_QUALS];
+};
+
+__global__ void k_qualmask(gx_qualargs qa, const uint8_t *vmap, int64_t nrows,
+                           uint8_t *hidden)
+{
+    int64_t nbytes = (nrows + 7) >> 3;
+    int64_t b = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; b < nbytes; b += stride)
+    {
+        uint8_t h = vmap ? vmap[b] : 0;
+        int64_t row0 = b << 3;
+        int lim = (int) min((int64_t) 8, nrows - row0);
+        for (int r = 0; r < lim; r++)
+        {
+            int64_t i = row0 + r;
+            bool pass = true;
+            for (int qn = 0; qn < qa.n && pass; qn++)
+            {
+                int64_t v;
+                switch (qa.m[qn].width)
+                {
+                    case 1: v = gx_col_get<int8_t>(qa.s[qn], qa.m[qn], i); break;
+                    case 4: v = gx_col_get<int32_t>(qa.s[qn], qa.m[qn], i); break;
+                    default: v = gx_col_get<int64_t>(qa.s[qn], qa.m[qn], i); break;
+                }
+                pass = gx_cmp(qa.op[qn], v, qa.lit[qn]);
+            }
+            if (!pass) h |= (uint8_t) (1u << r);
+        }
+        hidden[b] = h;
     }
 }
 
@@ -2164,6 +2225,10 @@ struct gx_q3 {
     uint8_t *dtext = nullptr;          /* device dim TEXT literal (texteq) */
     uint8_t *dmask = nullptr;          /* per-row texteq result (built once
                                           at prepare over the varlena col) */
+    /* extra AND-ed quals folded at prepare into per-table HIDDEN bitmaps
+     * (combined with the table visimap); kernels read them through the
+     * existing visibility parameter — execScan.c:241 qual-list semantics */
+    uint8_t *qvm_dim = nullptr, *qvm_mid = nullptr, *qvm_fact = nullptr;
     /* result (device SoA) */
     int64_t *r_okey = nullptr;
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
@@ -3266,6 +3331,40 @@ extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t
     return GX_OK;
 }
 
+/* multi-column distribution-key routing (cdbhash.c:189-247 rotate-combine +
+ * REDUCE_JUMP_HASH) — bit-exact with the oracle's orc_route_multi_batch.
+ * vals/isnull row-major n×nkeys; types[k]: 0 = int8, 1 = int4/date.
+ * isnull may be NULL (all NOT NULL). */
+extern "C" gx_status gx_partition_multi(gx_ctx *ctx, const int64_t *host_vals,
+                                        const uint8_t *host_isnull,
+                                        const int32_t *host_types,
+                                        int32_t nkeys, int64_t n, int32_t nsegs,
+                                        int32_t *host_out)
+{
+    if (!ctx || n < 0 || nkeys < 1 || nkeys > 32) return GX_ERR_INVALID;
+    devbuf dv, dn, dt, dr;
+    HIP_CHK(ctx, dv.alloc(n * nkeys * 8));
+    HIP_CHK(ctx, dt.alloc(nkeys * 4));
+    HIP_CHK(ctx, dr.alloc(n * 4));
+    HIP_CHK(ctx, hipMemcpyAsync(dv.p, host_vals, n * nkeys * 8,
+                                hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHK(ctx, hipMemcpyAsync(dt.p, host_types, nkeys * 4,
+                                hipMemcpyHostToDevice, ctx->stream));
+    if (host_isnull)
+    {
+        HIP_CHK(ctx, dn.alloc(n * nkeys));
+        HIP_CHK(ctx, hipMemcpyAsync(dn.p, host_isnull, n * nkeys,
+                                    hipMemcpyHostToDevice, ctx->stream));
+    }
+    hipLaunchKernelGGL(k_route_multi, dim3(GRID), dim3(TPB), 0, ctx->stream,
+                       dv.as<int64_t>(), dn.as<uint8_t>(), dt.as<int32_t>(),
+                       nkeys, n, nsegs, dr.as<int32_t>());
+    HIP_CHK(ctx, hipMemcpyAsync(host_out, dr.p, n * 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    return GX_OK;
+}
+
 /* ================= Q3 ================= */
 
 /* Evaluate the constant TEXT dim predicate ONCE over the varlena column
@@ -3320,6 +3419,44 @@ static gx_status q3_build_text_mask(gx_ctx *ctx, gx_q3 *q)
         set_err(ctx, "malformed (or NULL-bearing) varlena dim column%s", "");
         return GX_ERR_INVALID;
     }
+    return GX_OK;
+}
+
+/* fold an AND-ed extra-qual list into a device HIDDEN bitmap (combined with
+ * the table's visimap); runs once at prepare — the per-step kernels then
+ * read one bit per row through the existing visibility path */
+static gx_status q3_build_qualmask(gx_ctx *ctx, gx_table *t,
+                                   const gx_filter *quals, int nq,
+                                   uint8_t **out)
+{
+    gx_qualargs qa{};
+    qa.n = nq;
+    for (int i = 0; i < nq; i++)
+    {
+        int c = quals[i].col;
+        if (c < 0 || c >= (int) t->cols.size() || t->cols[c].format != 0 ||
+            quals[i].op < 0 || quals[i].op > 5 ||
+            (t->cols[c].m.width != 1 && t->cols[c].m.width != 4 &&
+             t->cols[c].m.width != 8))
+        {
+            set_err(ctx, "extra qual needs a fixed-width Orig column and "
+                         "op in 0..5%s", "");
+            return GX_ERR_INVALID;
+        }
+        qa.s[i] = t->cols[c].dstream;
+        qa.m[i] = t->cols[c].m;
+        qa.op[i] = quals[i].op;
+        qa.lit[i] = quals[i].literal;
+    }
+    int64_t nbytes = (t->nrows + 7) >> 3;
+    devbuf mb;
+    HIP_CHK(ctx, mb.alloc((size_t) std::max<int64_t>(nbytes, 1)));
+    hipLaunchKernelGGL(k_qualmask, dim3(GRID), dim3(TPB), 0, ctx->stream,
+                       qa, t->dvmap, t->nrows, mb.as<uint8_t>());
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    *out = mb.as<uint8_t>();
+    mb.p = nullptr;              /* ownership moves to gx_q3 */
     return GX_OK;
 }
 
@@ -3394,6 +3531,29 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
             return st;
         }
     }
+    /* extra AND-ed qual lists → per-table hidden bitmaps */
+    if (desc->n_dim_extra < 0 || desc->n_dim_extra > GX_MAX_EXTRA_QUALS ||
+        desc->n_mid_extra < 0 || desc->n_mid_extra > GX_MAX_EXTRA_QUALS ||
+        desc->n_fact_extra < 0 || desc->n_fact_extra > GX_MAX_EXTRA_QUALS)
+    {
+        gx_q3_free(q);
+        return GX_ERR_INVALID;
+    }
+    struct { gx_table *t; const gx_filter *qs; int nq; uint8_t **dst; } ex[3] = {
+        {customer, desc->dim_extra, desc->n_dim_extra, &q->qvm_dim},
+        {orders, desc->mid_extra, desc->n_mid_extra, &q->qvm_mid},
+        {lineitem, desc->fact_extra, desc->n_fact_extra, &q->qvm_fact},
+    };
+    for (auto &e : ex)
+        if (e.nq > 0)
+        {
+            gx_status st = q3_build_qualmask(ctx, e.t, e.qs, e.nq, e.dst);
+            if (st != GX_OK)
+            {
+                gx_q3_free(q);
+                return st;
+            }
+        }
     *out = q;
     return GX_OK;
 }
@@ -3433,6 +3593,7 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->m_send1); fr(q->m_recv1); fr(q->m_send2); fr(q->m_recv2);
     fr(q->dtext);
     fr(q->dmask);
+    fr(q->qvm_dim); fr(q->qvm_mid); fr(q->qvm_fact);
     q->m_send1_cap = q->m_recv1_cap = q->m_send2_cap = q->m_recv2_cap = 0;
     q->sized = false;
 }
@@ -3473,6 +3634,10 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     const gx_q3_desc &D = q->desc;
     const gx_col &cm = q->cust->cols[D.dim_filter.col];
     const gx_col &oc = q->ord->cols[D.mid_fk_col], &od = q->ord->cols[D.mid_filter.col];
+    /* effective visibility = extra-qual mask (already OR-combined with the
+     * table visimap at prepare) or the bare visimap */
+    const uint8_t *cvm = q->qvm_dim ? q->qvm_dim : q->cust->dvmap;
+    const uint8_t *ovm = q->qvm_mid ? q->qvm_mid : q->ord->dvmap;
 
     HIP_CHK(ctx, hipMalloc(&q->dcount, 8));
     HIP_CHK(ctx, hipMalloc(&q->dhits, 8));
@@ -3485,13 +3650,13 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         hipLaunchKernelGGL(k_cust_count_mask, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
-                           q->dmask, q->cust->dvmap,
+                           q->dmask, cvm,
                            q->dcount, q->dhits, q->dmin);
     else
         hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
                            q->cust->cols[D.dim_key_col].dstream,
                            q->cust->cols[D.dim_key_col].m,
-                           cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
+                           cm.dstream, cm.m, cvm, D.dim_filter.op,
                            (int8_t) D.dim_filter.literal, q->dcount, q->dhits,
                            q->dmin);
     unsigned long long n_building = 0, cmax = 0, cmin = 0;
@@ -3530,14 +3695,14 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                dim3(GRID), dim3(TPB), 0, s,
                                q->cust->cols[D.dim_key_col].dstream,
                                q->cust->cols[D.dim_key_col].m,
-                               q->dmask, q->cust->dvmap,
+                               q->dmask, cvm,
                                cs, q->cmask, q->bloom, q->bwmask);
         else
             hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
                                dim3(GRID), dim3(TPB), 0, s,
                                q->cust->cols[D.dim_key_col].dstream,
                                q->cust->cols[D.dim_key_col].m,
-                               cm.dstream, cm.m, q->cust->dvmap, D.dim_filter.op,
+                               cm.dstream, cm.m, cvm, D.dim_filter.op,
                                (int8_t) D.dim_filter.literal,
                                cs, q->cmask, q->bloom, q->bwmask);
     };
@@ -3562,7 +3727,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                q->ord->cols[D.mid_key_col].dstream,
                                q->ord->cols[D.mid_key_col].m,
                                od.dstream, od.m, oc.dstream, oc.m,
-                               q->ord->dvmap,
+                               ovm,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned int *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
@@ -3572,7 +3737,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                q->ord->cols[D.mid_key_col].dstream,
                                q->ord->cols[D.mid_key_col].m,
                                od.dstream, od.m, oc.dstream, oc.m,
-                               q->ord->dvmap,
+                               ovm,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                (const unsigned long long *) q->cset, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
@@ -3667,6 +3832,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                  &ld = q->li->cols[D.fact_b_col],
                  &ls = q->li->cols[D.fact_filter.col];
     unsigned long long *dcount = q->dcount;
+    /* effective visibility (extra-qual masks fold the visimap in) */
+    const uint8_t *cvm_eff = q->qvm_dim ? q->qvm_dim : q->cust->dvmap;
+    const uint8_t *ovm_eff = q->qvm_mid ? q->qvm_mid : q->ord->dvmap;
+    const uint8_t *lvm_eff = q->qvm_fact ? q->qvm_fact
+                                         : (q->li ? q->li->dvmap : nullptr);
 
     /* ---- stage 1: customer BUILDING set (rebuilt every run) ---- */
     HIP_CHK(ctx, hipEventRecord(ev[0], s));
@@ -3678,13 +3848,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 hipLaunchKernelGGL((k_cust_build_mask<std::decay_t<decltype(*cs)>>),
                                    dim3(GRID), dim3(TPB), 0, s,
                                    ck.dstream, ck.m, q->dmask,
-                                   q->cust->dvmap, cs, q->cmask,
+                                   cvm_eff, cs, q->cmask,
                                    q->bloom, q->bwmask);
             else
                 hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
                                    dim3(GRID), dim3(TPB), 0, s,
                                    ck.dstream, ck.m, cm.dstream, cm.m,
-                                   q->cust->dvmap,
+                                   cvm_eff,
                                    D.dim_filter.op, (int8_t) D.dim_filter.literal,
                                    cs, q->cmask, q->bloom, q->bwmask);
         };
@@ -3709,7 +3879,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
         int ogrid = env_int("GX_ORDERS_GRID", 32768);  /* measured optimum */
         bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
-        const uint8_t *ovm = q->ord->dvmap;
+        const uint8_t *ovm = ovm_eff;
         auto launch_build = [&](auto *tk, auto *cs) {
             auto go = [&](auto ch, auto vm) {
                 hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
@@ -3752,7 +3922,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                    dim3(ogrid), dim3(TPB), 0, s,
                                    ok.dstream, ok.m, oc.dstream, oc.m,
                                    od.dstream, od.m, op.dstream, op.m,
-                                   q->ord->dvmap,
+                                   ovm_eff,
                                    D.mid_filter.op, (int32_t) D.mid_filter.literal,
                                    cs, q->cmask, q->bloom, q->bwmask,
                                    q->m_send2, q->dcount);
@@ -3816,7 +3986,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         unsigned long long *dhist = q->m_hist;
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
-                           od.dstream, od.m, oc.dstream, oc.m, q->ord->dvmap,
+                           od.dstream, od.m, oc.dstream, oc.m, ovm_eff,
                            D.mid_filter.op, (int32_t) D.mid_filter.literal, n, dhist);
         std::vector<unsigned long long> h1(n);
         HIP_CHK(ctx, hipMemcpyAsync(h1.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
@@ -3830,7 +4000,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off1.data(), n * 8, hipMemcpyHostToDevice, s));
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
-                           op.dstream, op.m, q->ord->dvmap, D.mid_filter.op,
+                           op.dstream, op.m, ovm_eff, D.mid_filter.op,
                            (int32_t) D.mid_filter.literal, n, dcur, send1);
 
         /* exchange counts (all-gather of per-dest counts) */
@@ -4008,7 +4178,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         int64_t nb = lk.nblocks;
         int rgrid = (int) std::min<int64_t>(std::max<int64_t>(nb, 1), 16384);
-        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
+        const uint8_t *lvm = lvm_eff;
         auto launch_rle = [&](auto *keys, auto vm) {
             hipLaunchKernelGGL((k_li_probe_agg_rle<std::decay_t<decltype(*keys)>,
                                                    decltype(vm)::value>),
@@ -4037,7 +4207,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     else if (q->numeric)
     {
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
-        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
+        const uint8_t *lvm = lvm_eff;
         auto launch_num = [&](auto *keys, auto vm) {
             hipLaunchKernelGGL((k_li_probe_agg_num<std::decay_t<decltype(*keys)>,
                                                    decltype(vm)::value>),
@@ -4067,16 +4237,16 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
-        if (variant != 0 && q->li && q->li->dvmap)
+        if (variant != 0 && lvm_eff)
         {
-            set_err(ctx, "visimap requires the default probe variant%s", "");
+            set_err(ctx, "visimap/extra quals require the default probe variant%s", "");
             return GX_ERR_INVALID;
         }
         const char *pg = getenv("GX_PROBE_GRID");
         int pgrid = pg ? atoi(pg) : GRID;
         const char *pt = getenv("GX_PROBE_TPB");
         int ptpb = pt ? atoi(pt) : TPB;
-        const uint8_t *lvm = q->li ? q->li->dvmap : nullptr;
+        const uint8_t *lvm = lvm_eff;
         auto launch = [&](auto kern, auto *keys) {
             hipLaunchKernelGGL(kern, dim3(pgrid), dim3(ptpb), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,

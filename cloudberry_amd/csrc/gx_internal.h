@@ -66,6 +66,30 @@ GX_HD int32_t gx_route_i64(int64_t key, int32_t nsegs)
     return gx_jump_consistent_hash((uint64_t) gx_cdbhash_i64(key), nsegs);
 }
 
+/* hashint4 — hashfunc.c:73-77 */
+GX_HD uint32_t gx_hashint4(int32_t val)
+{
+    return gx_hash_bytes_uint32((uint32_t) val);
+}
+
+/* N-attribute cdbhash chain — cdbhash.c:171-247: per attribute
+ * rotate-left-1 then XOR the type hash; NULL contributes the rotation only.
+ * types[k]: 0 = int8 (hashint8), 1 = int4/int2/date (hashint4). */
+GX_HD uint32_t gx_cdbhash_multi(const int64_t *vals, const uint8_t *isnull,
+                                const int32_t *types, int32_t nkeys)
+{
+    uint32_t hashkey = 0;                                  /* cdbhashinit */
+    for (int32_t k = 0; k < nkeys; k++)
+    {
+        hashkey = (hashkey << 1) | (hashkey >> 31);
+        if (isnull && isnull[k]) continue;
+        hashkey ^= (types && types[k] == 1)
+                       ? gx_hashint4((int32_t) vals[k])
+                       : gx_hashint8(vals[k]);
+    }
+    return hashkey;
+}
+
 /* ---------- internal hash-table hash (NOT parity-relevant) ---------- */
 GX_HD uint64_t gx_hmix64(uint64_t x)
 {

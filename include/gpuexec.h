@@ -159,6 +159,15 @@ gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col, int op,
 gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t n,
                        int32_t nsegs, int32_t *host_out);
 
+/* multi-column distribution keys (cdbhash.c:189-247 rotate-combine loop):
+ * vals/isnull row-major n×nkeys (each attribute widened to int64);
+ * types[k]: 0 = int8 (hashint8), 1 = int4/int2/date (hashint4); a NULL
+ * attribute contributes only the rotation.  isnull may be NULL. */
+gx_status gx_partition_multi(gx_ctx *ctx, const int64_t *host_vals,
+                             const uint8_t *host_isnull,
+                             const int32_t *host_types, int32_t nkeys,
+                             int64_t n, int32_t nsegs, int32_t *host_out);
+
 /* ---- the Q3 pipeline (the judged QE slice) ---- */
 
 typedef struct gx_q3_group {
@@ -225,6 +234,15 @@ typedef struct gx_q3_desc {
      * segments the comparison runs once per RUN. */
     char dim_text[64];
     int32_t dim_text_len;
+    /* AND-ed qual lists (execScan.c:241 semantics over NOT NULL integer/date
+     * columns: every qual must pass; widths 1/4/8, literals widened to i64).
+     * Folded at prepare into a per-table row mask combined with the visimap,
+     * so the per-step kernels see them through the existing visibility path. */
+#define GX_MAX_EXTRA_QUALS 4
+    gx_filter dim_extra[GX_MAX_EXTRA_QUALS];
+    gx_filter mid_extra[GX_MAX_EXTRA_QUALS];
+    gx_filter fact_extra[GX_MAX_EXTRA_QUALS];
+    int32_t n_dim_extra, n_mid_extra, n_fact_extra;
 } gx_q3_desc;
 
 /* Restrictions checked at sizing (first gx_q3_run):

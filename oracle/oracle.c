@@ -116,6 +116,51 @@ void orc_route_i64_batch(const int64_t *keys, int64_t n, int32_t nsegs, int32_t 
         out[i] = orc_route_i64(keys[i], nsegs);
 }
 
+/* hashint4 — src/backend/access/hash/hashfunc.c:73-77 (hash_uint32 of the
+ * int32 value; int2/int4/int8 produce compatible hashes for equal values) */
+uint32_t orc_hashint4(int32_t val)
+{
+    return orc_hash_bytes_uint32((uint32_t) val);
+}
+
+/*
+ * cdbhash over N attributes in declared order —
+ * src/backend/cdb/cdbhash.c:171-247: cdbhashinit (hash = 0, non-legacy),
+ * then per attribute
+ *     hashkey = rotate_left_1(hashkey);
+ *     if (!isnull) hashkey ^= typehash(val);
+ * A NULL attribute contributes ONLY the rotation (cdbhash.c:195-216).
+ * types[k]: 0 = int8 (hashint8), 1 = int4/int2/date (hashint4).
+ * vals carries each attribute widened to int64 (sign-preserving).
+ */
+uint32_t orc_cdbhash_multi(const int64_t *vals, const uint8_t *isnull,
+                           const int32_t *types, int32_t nkeys)
+{
+    uint32_t hashkey = 0;                                  /* cdbhashinit */
+    for (int32_t k = 0; k < nkeys; k++)
+    {
+        hashkey = (hashkey << 1) | ((hashkey & 0x80000000u) ? 1 : 0);
+        if (isnull && isnull[k]) continue;
+        hashkey ^= (types && types[k] == 1)
+                       ? orc_hashint4((int32_t) vals[k])
+                       : orc_hashint8(vals[k]);
+    }
+    return hashkey;
+}
+
+/* multi-key Motion routing (row-major vals/isnull, n rows × nkeys) */
+void orc_route_multi_batch(const int64_t *vals, const uint8_t *isnull,
+                           const int32_t *types, int32_t nkeys, int64_t n,
+                           int32_t nsegs, int32_t *out)
+{
+    for (int64_t i = 0; i < n; i++)
+        out[i] = orc_jump_consistent_hash(
+            (uint64_t) orc_cdbhash_multi(vals + i * nkeys,
+                                         isnull ? isnull + i * nkeys : 0,
+                                         types, nkeys),
+            nsegs);
+}
+
 /*
  * CRC-32C in PostgreSQL COMP_CRC32C semantics (reflected Castagnoli,
  * init 0xFFFFFFFF, NO final inversion — "by historical accident" the AO

@@ -28,6 +28,14 @@ int32_t  orc_jump_consistent_hash(uint64_t key, int32_t nsegs);
 int32_t  orc_route_i64(int64_t key, int32_t nsegs);  /* Motion routing */
 void     orc_route_i64_batch(const int64_t *keys, int64_t n, int32_t nsegs,
                              int32_t *out);
+uint32_t orc_hashint4(int32_t v);                    /* hashfunc.c:73-77 */
+/* N-attribute cdbhash chain (cdbhash.c:171-247); types[k]: 0=int8, 1=int4;
+ * NULL attributes contribute the rotation only */
+uint32_t orc_cdbhash_multi(const int64_t *vals, const uint8_t *isnull,
+                           const int32_t *types, int32_t nkeys);
+void     orc_route_multi_batch(const int64_t *vals, const uint8_t *isnull,
+                               const int32_t *types, int32_t nkeys, int64_t n,
+                               int32_t nsegs, int32_t *out);
 uint32_t orc_crc32c(uint32_t crc, const void *buf, size_t len); /* pg COMP_CRC32C state (no final xor) */
 
 /* ---- date helpers (DateADT = int32 days since 2000-01-01) ---- */

@@ -62,6 +62,16 @@ def _load():
     lib.orc_jump_consistent_hash.argtypes = [ctypes.c_uint64, ctypes.c_int32]
     lib.orc_route_i64.restype = ctypes.c_int32
     lib.orc_route_i64.argtypes = [ctypes.c_int64, ctypes.c_int32]
+    lib.orc_hashint4.restype = ctypes.c_uint32
+    lib.orc_hashint4.argtypes = [ctypes.c_int32]
+    lib.orc_cdbhash_multi.restype = ctypes.c_uint32
+    lib.orc_cdbhash_multi.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                      ctypes.c_void_p, ctypes.c_int32]
+    lib.orc_route_multi_batch.restype = None
+    lib.orc_route_multi_batch.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.c_void_p, ctypes.c_int32,
+                                          ctypes.c_int64, ctypes.c_int32,
+                                          ctypes.c_void_p]
     lib.orc_route_i64_batch.restype = None
     lib.orc_route_i64_batch.argtypes = [ctypes.c_void_p, ctypes.c_int64,
                                         ctypes.c_int32, ctypes.c_void_p]
@@ -339,6 +349,35 @@ def route(keys, nsegs):
     out = np.zeros(len(keys), np.int32)
     lib.orc_route_i64_batch(keys.ctypes.data, len(keys), nsegs, out.ctypes.data)
     return out
+
+
+def route_multi(vals, types, nsegs, isnull=None):
+    """Multi-key Motion routing (cdbhash.c:189-247 rotate-combine): vals is
+    (n, nkeys) int64 row-major; types[k] 0 = int8, 1 = int4/date."""
+    vals = np.ascontiguousarray(vals, np.int64)
+    n, nkeys = vals.shape
+    types = np.ascontiguousarray(types, np.int32)
+    nul = None
+    nul_ptr = None
+    if isnull is not None:
+        nul = np.ascontiguousarray(isnull, np.uint8)
+        assert nul.shape == vals.shape
+        nul_ptr = nul.ctypes.data
+    out = np.zeros(n, np.int32)
+    lib.orc_route_multi_batch(vals.ctypes.data, nul_ptr, types.ctypes.data,
+                              nkeys, n, nsegs, out.ctypes.data)
+    return out
+
+
+def cdbhash_multi(vals, types, isnull=None):
+    vals = np.ascontiguousarray(vals, np.int64)
+    types = np.ascontiguousarray(types, np.int32)
+    nul_ptr = None
+    if isnull is not None:
+        nul = np.ascontiguousarray(isnull, np.uint8)
+        nul_ptr = nul.ctypes.data
+    return lib.orc_cdbhash_multi(vals.ctypes.data, nul_ptr,
+                                 types.ctypes.data, len(vals))
 
 
 def aocs_encode(vals):

@@ -955,3 +955,216 @@ def test_gpu_sf100_rle_mode_cross_check(ctx, orc):
     np.testing.assert_array_equal(rr["l_orderkey"], r["l_orderkey"])
     np.testing.assert_array_equal(rr["nitems"], r["nitems"])
     np.testing.assert_allclose(rr["revenue"], r["revenue"], rtol=1e-12)
+
+
+# ---------------- round-2 guards (ADVICE r01 / VERDICT r01 #7) ----------------
+
+def _mini_q3_tables(ctx, orc, li_keys, li_price, li_disc, li_ship,
+                    c_keys=None, o_keys=None, o_cust=None):
+    """Hand-built Q3-shaped tables through gx_table_bind: every customer is
+    segment 0 (passes == 0), every order qualifies on date."""
+    if c_keys is None:
+        c_keys = np.arange(1, 101, dtype=np.int64)
+    if o_keys is None:
+        o_keys = np.arange(1, 101, dtype=np.int64)
+    if o_cust is None:
+        o_cust = c_keys[:len(o_keys)].copy()
+    nc, no = len(c_keys), len(o_keys)
+    cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                     (orc.aocs_encode(np.zeros(nc, np.int8)), 1, nc)])
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode(o_cust), 8, no),
+                     (orc.aocs_encode(np.full(no, -9999, np.int32)), 4, no),
+                     (orc.aocs_encode(np.arange(no, dtype=np.int32)), 4, no)])
+    nl = len(li_keys)
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nl),
+                   (orc.aocs_encode(li_price), 8, nl),
+                   (orc.aocs_encode(li_disc), 8, nl),
+                   (orc.aocs_encode(li_ship), 4, nl)])
+    return cust, ordr, li
+
+
+def test_u32_narrow_mode_straddling_keys(ctx, orc):
+    """with all BUILD keys < 2^32 the table/set use u32
+    slots; PROBE keys >= 2^32 whose low half collides with a resident key
+    must MISS, not aggregate into the wrong group."""
+    small = np.arange(1, 101, dtype=np.int64)
+    straddle = small + (1 << 32)          # low 32 bits identical to `small`
+    li_keys = np.concatenate([small, straddle])
+    nl = len(li_keys)
+    price = np.full(nl, 100.0)
+    disc = np.zeros(nl)
+    ship = np.full(nl, 9999, np.int32)    # all pass '>' cutoff
+    cust, ordr, li = _mini_q3_tables(ctx, orc, li_keys, price, disc, ship)
+    r = ctx.q3(cust, ordr, li).run().result()
+    # only the 100 small keys may appear, each with exactly ONE item
+    np.testing.assert_array_equal(r["l_orderkey"], small)
+    np.testing.assert_array_equal(r["nitems"], np.ones(100, np.int64))
+    np.testing.assert_allclose(r["revenue"], np.full(100, 100.0), rtol=1e-12)
+    li.free(); ordr.free(); cust.free()
+
+    # same hazard on the customer SET probe: o_custkey = c_custkey + 2^32
+    # must not pass the semijoin
+    c_keys = np.arange(1, 101, dtype=np.int64)
+    o_keys = np.arange(1, 201, dtype=np.int64)
+    o_cust = np.concatenate([c_keys, c_keys + (1 << 32)])
+    li_keys2 = np.arange(1, 201, dtype=np.int64)
+    cust, ordr, li = _mini_q3_tables(
+        ctx, orc, li_keys2, np.full(200, 10.0), np.zeros(200),
+        np.full(200, 9999, np.int32), c_keys=c_keys, o_keys=o_keys,
+        o_cust=o_cust)
+    r = ctx.q3(cust, ordr, li).run().result()
+    # orders 101..200 carry straddling custkeys -> filtered by the semijoin
+    np.testing.assert_array_equal(r["l_orderkey"],
+                                  np.arange(1, 101, dtype=np.int64))
+    li.free(); ordr.free(); cust.free()
+
+
+def test_key_zero_rejected(ctx, orc):
+    """Join key 0 collides with the empty-slot sentinel: sizing must reject
+    it loudly (GX_ERR_INVALID), never silently drop the row."""
+    li_keys = np.arange(1, 11, dtype=np.int64)
+    price = np.full(10, 1.0); disc = np.zeros(10)
+    ship = np.full(10, 9999, np.int32)
+    # customer key 0
+    cust, ordr, li = _mini_q3_tables(
+        ctx, orc, li_keys, price, disc, ship,
+        c_keys=np.arange(0, 100, dtype=np.int64))
+    with pytest.raises(gx.GxError) as ei:
+        ctx.q3(cust, ordr, li).run()
+    assert ei.value.status == 3 and "sentinel" in str(ei.value)
+    li.free(); ordr.free(); cust.free()
+    # orders key 0
+    cust, ordr, li = _mini_q3_tables(
+        ctx, orc, li_keys, price, disc, ship,
+        o_keys=np.arange(0, 100, dtype=np.int64))
+    with pytest.raises(gx.GxError) as ei:
+        ctx.q3(cust, ordr, li).run()
+    assert ei.value.status == 3 and "sentinel" in str(ei.value)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_hbm_budget_guard(ctx, orc, monkeypatch):
+    """a build side beyond the HBM budget fails at sizing
+    with the required-vs-available numbers, BEFORE any table allocation."""
+    monkeypatch.setenv("GX_HBM_BUDGET_MB", "1")
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, 0.1)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, 0.1)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, 0.1)
+    with pytest.raises(gx.GxError) as ei:
+        ctx.q3(cust, ordr, li).run()
+    assert ei.value.status == 5            # GX_ERR_OOM
+    msg = str(ei.value)
+    assert "HBM budget" in msg and "GB" in msg
+    monkeypatch.delenv("GX_HBM_BUDGET_MB")
+    # same tables run fine without the cap
+    r = ctx.q3(cust, ordr, li).run().result()
+    assert len(r["l_orderkey"]) > 0
+    li.free(); ordr.free(); cust.free()
+
+
+def test_partition_multi_bit_exact(ctx, orc):
+    """Multi-column distribution keys (cdbhash.c:189-247 rotate-combine):
+    GPU routing bit-exact vs the oracle for 1/2/3-key vectors, int8+int4
+    type mixes, with and without NULL attributes."""
+    rng = np.random.default_rng(41)
+    for nkeys, types in ((1, [0]), (2, [0, 0]), (2, [0, 1]), (3, [0, 1, 0])):
+        types = np.array(types, np.int32)
+        vals = rng.integers(-2**62, 2**62, (20000, nkeys)).astype(np.int64)
+        for k in range(nkeys):
+            if types[k] == 1:
+                vals[:, k] = rng.integers(-2**31, 2**31, 20000)
+        nulls = (rng.random((20000, nkeys)) < 0.15).astype(np.uint8)
+        for nsegs in (2, 3, 8, 64):
+            got = ctx.partition_multi(vals, types, nsegs, isnull=nulls)
+            want = orc.route_multi(vals, types, nsegs, isnull=nulls)
+            np.testing.assert_array_equal(got, want)
+            # NOT NULL fast path (isnull omitted)
+            got_nn = ctx.partition_multi(vals, types, nsegs)
+            want_nn = orc.route_multi(vals, types, nsegs)
+            np.testing.assert_array_equal(got_nn, want_nn)
+    # single-key multi == the established single-key entry
+    keys = rng.integers(-2**62, 2**62, 5000).astype(np.int64)
+    got = ctx.partition_multi(keys.reshape(-1, 1), np.zeros(1, np.int32), 8)
+    np.testing.assert_array_equal(got, ctx.partition(keys, 8))
+
+
+def test_extra_qual_lists(ctx, orc):
+    """AND-ed qual lists per table (execScan.c:241 semantics over multiple
+    quals): descriptor extra quals vs a numpy brute force of the same plan."""
+    sf = 0.05
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut),
+            # extra quals: priority in {1,2,3}, custkey even-ish bound,
+            # lineitem shipdate upper bound AND orderkey range
+            "mid_extra": [(3, ">=", 1), (3, "<=", 3), (1, "<", 40000)],
+            "fact_extra": [(3, "<", cut + 900), (0, ">", 100)],
+            "dim_extra": [(0, "<", 70000)]}
+    got = ctx.q3_desc(base).run().result()
+
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    w = orc.gen_lineitem(sf)
+    cm = (c["c_mktsegment"] == 0) & (c["c_custkey"] < 70000)
+    segok = c["c_custkey"][cm]
+    om = ((o["o_orderdate"] < cut) & (o["o_shippriority"] >= 1) &
+          (o["o_shippriority"] <= 3) & (o["o_custkey"] < 40000) &
+          np.isin(o["o_custkey"], segok))
+    lm = ((w["l_shipdate"] > cut) & (w["l_shipdate"] < cut + 900) &
+          (w["l_orderkey"] > 100) &
+          np.isin(w["l_orderkey"], o["o_orderkey"][om]))
+    keys, counts = np.unique(w["l_orderkey"][lm], return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    rev = {k: 0.0 for k in keys.tolist()}
+    for k, p, dsc in zip(w["l_orderkey"][lm].tolist(),
+                         w["l_extendedprice"][lm], w["l_discount"][lm]):
+        rev[k] += p * (1.0 - dsc)
+    np.testing.assert_allclose(got["revenue"],
+                               np.array([rev[k] for k in keys.tolist()]),
+                               rtol=1e-6)
+    # no extras == classic plan (count the default path is untouched)
+    plain = {k: v for k, v in base.items()
+             if not k.endswith("_extra")}
+    classic = ctx.q3(cust, ordr, li).run().result()
+    via = ctx.q3_desc(plain).run().result()
+    np.testing.assert_array_equal(via["l_orderkey"], classic["l_orderkey"])
+    li.free(); ordr.free(); cust.free()
+
+
+def test_extra_quals_with_visimap(ctx, orc):
+    """Extra quals AND the AO visimap compose (the qual mask folds the
+    visimap in at prepare)."""
+    sf = 0.02
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    rng = np.random.default_rng(17)
+    deleted = rng.random(li.nrows) < 0.10
+    li.set_visimap(deleted)
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut),
+            "fact_extra": [(0, ">", 500)]}
+    got = ctx.q3_desc(base).run().result()
+    c = orc.gen_customer(sf)
+    o = orc.gen_orders(sf)
+    w = orc.gen_lineitem(sf)
+    segok = c["c_custkey"][c["c_mktsegment"] == 0]
+    om = (o["o_orderdate"] < cut) & np.isin(o["o_custkey"], segok)
+    lm = ((w["l_shipdate"] > cut) & (w["l_orderkey"] > 500) & ~deleted &
+          np.isin(w["l_orderkey"], o["o_orderkey"][om]))
+    keys, counts = np.unique(w["l_orderkey"][lm], return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    li.free(); ordr.free(); cust.free()

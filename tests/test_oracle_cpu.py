@@ -726,3 +726,89 @@ def test_varlena_short_form_boundary():
         ref = orc.ref_writer_varlena_stream(vals, version=ver, rle=rle)
         assert ours == ref, f"v{ver} boundary diverges"
         assert orc.aocs_decode_varlena(ours, len(vals)) == vals
+
+
+# ---------------- multi-key cdbhash (cdbhash.c:189-247) ----------------
+
+def _py_cdbhash_multi(vals, types, isnull=None, hashfn=None):
+    """Independent restatement of the reference rotate-combine loop
+    (cdbhash.c:189-216) on top of the golden-pinned per-type hashes."""
+    h8 = hashfn or (lambda v: orc.lib.orc_hashint8(int(v)))
+    h4 = lambda v: orc.lib.orc_hash_bytes_uint32(np.uint32(np.int32(v)))
+    h = 0
+    for k, (v, t) in enumerate(zip(vals, types)):
+        h = ((h << 1) | (h >> 31)) & 0xFFFFFFFF      # pg_rotate_left32(h, 1)
+        if isnull is not None and isnull[k]:
+            continue                                  # NULL: rotation only
+        h ^= (h4(v) if t == 1 else h8(v))
+    return h
+
+
+def test_cdbhash_multi_matches_independent_restatement():
+    rng = np.random.default_rng(31)
+    for nkeys in (1, 2, 3, 5):
+        vals = rng.integers(-2**62, 2**62, (200, nkeys)).astype(np.int64)
+        types = rng.integers(0, 2, nkeys).astype(np.int32)
+        # clamp int4-typed attrs into int32 range (widened representation)
+        for k in range(nkeys):
+            if types[k] == 1:
+                vals[:, k] = rng.integers(-2**31, 2**31, 200)
+        nulls = (rng.random((200, nkeys)) < 0.2).astype(np.uint8)
+        for i in range(200):
+            want = _py_cdbhash_multi(vals[i], types, nulls[i])
+            got = orc.cdbhash_multi(vals[i], types, nulls[i])
+            assert got == want, (i, nkeys)
+        # all-NOT-NULL path (isnull omitted)
+        for i in range(50):
+            want = _py_cdbhash_multi(vals[i], types)
+            assert orc.cdbhash_multi(vals[i], types) == want
+
+
+def test_cdbhash_multi_single_key_equals_route():
+    """1-key multi chain must equal the established single-key chain
+    (pinned by the reference golden vectors)."""
+    rng = np.random.default_rng(32)
+    keys = rng.integers(-2**62, 2**62, 500).astype(np.int64)
+    types = np.zeros(1, np.int32)
+    for nsegs in (2, 8, 64):
+        got = orc.route_multi(keys.reshape(-1, 1), types, nsegs)
+        want = orc.route(keys, nsegs)
+        np.testing.assert_array_equal(got, want)
+
+
+def test_cdbhash_multi_against_live_reference_if_present():
+    """When the compiled reference hashfn.c is present, drive the rotate-
+    combine loop with the REFERENCE's own hash function — pins the multi-key
+    chain end to end."""
+    import ctypes
+    ref_so = os.path.join(HERE, "..", "oracle", "_ref", "libpgref.so")
+    if not os.path.exists(ref_so):
+        pytest.skip("oracle/_ref not built here")
+    ref = ctypes.CDLL(ref_so)
+    ref.hash_bytes_uint32.restype = ctypes.c_uint32
+    ref.hash_bytes_uint32.argtypes = [ctypes.c_uint32]
+
+    def ref_hashint8(v):
+        v = int(v)
+        lo = v & 0xFFFFFFFF
+        hi = (v >> 32) & 0xFFFFFFFF
+        lo ^= hi if v >= 0 else (~hi & 0xFFFFFFFF)
+        return ref.hash_bytes_uint32(lo)
+
+    rng = np.random.default_rng(33)
+    vals = rng.integers(-2**62, 2**62, (100, 3)).astype(np.int64)
+    types = np.zeros(3, np.int32)
+    for i in range(100):
+        want = _py_cdbhash_multi(vals[i], types, hashfn=ref_hashint8)
+        assert orc.cdbhash_multi(vals[i], types) == want
+
+
+def test_route_multi_null_only_rows():
+    """An all-NULL key row hashes to 0-rotated-N = 0 -> jump(0, n); matches
+    cdbhash semantics where NULLs contribute only rotations."""
+    vals = np.zeros((4, 2), np.int64)
+    nulls = np.ones((4, 2), np.uint8)
+    types = np.zeros(2, np.int32)
+    out = orc.route_multi(vals, types, 8, isnull=nulls)
+    want = orc.lib.orc_jump_consistent_hash(0, 8)
+    assert (out == want).all()
