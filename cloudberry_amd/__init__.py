@@ -52,6 +52,12 @@ class _Stats(ctypes.Structure):
                 ("bytes_scanned", ctypes.c_double)]
 
 
+class _KvGroup(ctypes.Structure):
+    _fields_ = [("key", ctypes.c_int64), ("key_is_null", ctypes.c_uint8),
+                ("_pad", ctypes.c_uint8 * 7), ("sum", ctypes.c_double),
+                ("count", ctypes.c_int64)]
+
+
 class _Filter(ctypes.Structure):
     _fields_ = [("col", ctypes.c_int32), ("op", ctypes.c_int32),
                 ("literal", ctypes.c_int64)]
@@ -133,6 +139,10 @@ def _load():
                                    ctypes.POINTER(ctypes.c_double)]
     lib.gx_partition.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                                  ctypes.c_int32, ctypes.c_void_p]
+    lib.gx_groupby.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                               ctypes.c_int,
+                               ctypes.POINTER(ctypes.POINTER(_KvGroup)),
+                               ctypes.POINTER(ctypes.c_int64)]
     lib.gx_partition_multi.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.c_int32, ctypes.c_int64,
@@ -247,6 +257,22 @@ class Context:
             self._h, vals.ctypes.data, nul_ptr, types.ctypes.data,
             nkeys, n, nsegs, out.ctypes.data))
         return out
+
+    def groupby(self, table, key_col, val_col):
+        """GROUP BY key_col with COUNT(*)/SUM(val_col); NULL keys form one
+        group (returned last, key_is_null True)."""
+        gp = ctypes.POINTER(_KvGroup)()
+        n = ctypes.c_int64()
+        self._chk(self._lib.gx_groupby(self._h, table._t, key_col, val_col,
+                                       ctypes.byref(gp), ctypes.byref(n)))
+        n = n.value
+        res = {"key": np.array([gp[i].key for i in range(n)], np.int64),
+               "key_is_null": np.array([gp[i].key_is_null for i in range(n)],
+                                       np.bool_),
+               "sum": np.array([gp[i].sum for i in range(n)], np.float64),
+               "count": np.array([gp[i].count for i in range(n)], np.int64)}
+        self._lib.gx_free(gp)
+        return res
 
     def test_motion1(self, orders, nsegs, cutoff=CUTOFF_19950315):
         """Run the Motion-1 partition kernels; returns (counts, rows dict)."""

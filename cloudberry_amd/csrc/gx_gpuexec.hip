@@ -37,6 +37,8 @@ extern "C" unsigned ZSTD_isError(size_t code);
 #include <cstring>
 #include <string>
 #include <vector>
+#include <map>
+#include <utility>
 
 /* ================= error plumbing ================= */
 
@@ -684,6 +686,51 @@ __global__ void k_route_multi(const int64_t *vals, const uint8_t *isnull,
             nsegs);
 }
 
+/* Hash GROUP BY with the reference's grouping semantics (nodeAgg.c:2288 +
+ * execGrouping.c:436-495): group keys compare NOT DISTINCT, so ALL NULL
+ * keys land in ONE group; SUM's transition fn is strict (skips NULL
+ * inputs, float.c:769) while COUNT(*) counts every row.  Keys are stored
+ * biased (k ^ 2^63) so slot 0 stays the empty sentinel for any int64 key
+ * except INT64_MIN (rejected with an error flag). */
+__global__ void k_groupby(const uint8_t *k_s, gx_colmeta k_m,
+                          const uint8_t *k_val,
+                          const uint8_t *v_s, gx_colmeta v_m,
+                          const uint8_t *v_val,
+                          const uint8_t *vmap,
+                          unsigned long long *tkey, double *tsum,
+                          unsigned long long *tcnt, uint64_t tmask,
+                          double *null_sum, unsigned long long *null_cnt,
+                          int *err)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; i < k_m.nrows; i += stride)
+    {
+        if (gx_vm_hidden(vmap, i)) continue;
+        bool vnull = v_val && !v_val[i];
+        double v = vnull ? 0.0 : gx_col_get<double>(v_s, v_m, i);
+        if (k_val && !k_val[i])
+        {
+            /* NULLs-equal grouping: one shared group */
+            if (!vnull) atomicAdd(null_sum, v);
+            atomicAdd(null_cnt, 1ULL);
+            continue;
+        }
+        uint64_t bk = (uint64_t) gx_col_get<int64_t>(k_s, k_m, i) ^
+                      (1ULL << 63);
+        if (bk == 0) { atomicOr(err, 4); continue; }
+        uint64_t slot = gx_hmix64(bk) & tmask;
+        while (true)
+        {
+            unsigned long long prev = atomicCAS(&tkey[slot], 0ULL, bk);
+            if (prev == 0ULL || prev == bk) break;
+            slot = (slot + 1) & tmask;
+        }
+        if (!vnull) atomicAdd(&tsum[slot], v);
+        atomicAdd(&tcnt[slot], 1ULL);
+    }
+}
+
 /* ================= Q3 kernels ================= */
 
 /* Join-table slot mapping.  When the build keys' [min,max] stats admit it
@@ -840,6 +887,28 @@ __global__ void k_qualmask(gx_qualargs qa, const uint8_t *vmap, int64_t nrows,
             if (!pass) h |= (uint8_t) (1u << r);
         }
         hidden[b] = h;
+    }
+}
+
+/* OR rows with validity==0 into a hidden bitmap: strict-NULL reject for
+ * join keys (nodeHash.c:2168-2181 — a NULL key cannot pass a strict hash
+ * operator on either side of an inner join) and three-valued filter
+ * semantics for qual columns (execScan.c:241 — a NULL qual result filters
+ * the row) */
+__global__ void k_validity_or_hidden(const uint8_t *validity, int64_t nrows,
+                                     uint8_t *hidden)
+{
+    int64_t nbytes = (nrows + 7) >> 3;
+    int64_t b = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (; b < nbytes; b += stride)
+    {
+        uint8_t h = 0;
+        int64_t row0 = b << 3;
+        int lim = (int) min((int64_t) 8, nrows - row0);
+        for (int r = 0; r < lim; r++)
+            if (!validity[row0 + r]) h |= (uint8_t) (1u << r);
+        if (h) hidden[b] |= h;
     }
 }
 
@@ -2187,6 +2256,7 @@ struct gx_col {
                                     1 = Dense/RLE (directory-based) */
     gx_blockref *ddir = nullptr; /* device block directory (format 1) */
     int64_t nblocks = 0;
+    bool has_null = false;       /* any block carries a NULL bitmap (flags&1) */
 };
 
 struct gx_table {
@@ -2229,6 +2299,10 @@ struct gx_q3 {
      * (combined with the table visimap); kernels read them through the
      * existing visibility parameter — execScan.c:241 qual-list semantics */
     uint8_t *qvm_dim = nullptr, *qvm_mid = nullptr, *qvm_fact = nullptr;
+    /* flat materializations of null-bearing / RLE role columns (built at
+     * prepare; strict-NULL reject folds their validity into the qvm masks) */
+    std::map<std::pair<const gx_table *, int>, gx_col> mat;
+    std::vector<void *> mat_mem;
     /* result (device SoA) */
     int64_t *r_okey = nullptr;
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
@@ -2553,9 +2627,11 @@ static bool host_decompress_stream(const uint8_t *s, int64_t nbytes,
 /* walk a stream's AO envelope headers on the HOST, building the per-block
  * directory a variable-geometry (Dense/RLE) stream needs */
 static gx_status parse_block_dir(const uint8_t *s, int64_t nbytes,
-                                 std::vector<gx_blockref> &dir, int64_t *rows_out)
+                                 std::vector<gx_blockref> &dir, int64_t *rows_out,
+                                 bool *has_null_out = nullptr)
 {
     int64_t off = 0, row = 0;
+    if (has_null_out) *has_null_out = false;
     while (off + 24 <= nbytes)
     {
         uint32_t b03, b47;
@@ -2579,6 +2655,12 @@ static gx_status parse_block_dir(const uint8_t *s, int64_t nbytes,
             return GX_ERR_INVALID;
         int64_t blocklen = (24 + (int64_t) datalen + 7) & ~7LL;
         if (off + blocklen > nbytes) return GX_ERR_INVALID;
+        if (has_null_out && blocklen >= 28)
+        {
+            int16_t flags;                       /* DatumStreamBlock flags */
+            memcpy(&flags, s + off + 26, 2);
+            if (flags & 1) *has_null_out = true; /* HAS_NULLBITMAP */
+        }
         dir.push_back({off, row, (int32_t) rows, 0});
         row += rows;
         off += blocklen;
@@ -2625,7 +2707,8 @@ extern "C" gx_status gx_table_bind(gx_ctx *ctx, const gx_coldesc *cols, int ncol
             std::vector<gx_blockref> dir;
             int64_t rows = 0;
             gx_status st = parse_block_dir((const uint8_t *) stream_src,
-                                           stream_len, dir, &rows);
+                                           stream_len, dir, &rows,
+                                           &col.has_null);
             if (st != GX_OK || rows != cols[c].nrows)
             {
                 set_err(ctx, "bad Dense/RLE stream%s", "");
@@ -3315,6 +3398,152 @@ extern "C" gx_status gx_scan_filter(gx_ctx *ctx, const gx_table *t, int col,
     return GX_OK;
 }
 
+/* Standalone hash GROUP BY (nodeAgg.c hash strategy + execGrouping.c
+ * NOT-DISTINCT grouping): GROUP BY key_col with COUNT(*) and SUM(val_col).
+ * Key col: i64, Orig or Dense/RLE (nullable — all NULL keys form ONE
+ * group, returned last with key_is_null=1); val col: f64, same formats
+ * (NULL inputs are skipped by SUM, counted by COUNT(*)).  Groups are
+ * returned sorted by key; caller frees with gx_free. */
+static gx_status hbm_budget_check(gx_ctx *ctx, uint64_t want, const char *what);
+
+static gx_status gb_flat(gx_ctx *ctx, const gx_col &c, devbuf &flat,
+                         devbuf &val, const uint8_t **s_out, gx_colmeta *m_out,
+                         const uint8_t **val_out)
+{
+    if (c.format == 0)
+    {
+        if (c.has_null)
+        { set_err(ctx, "null-bearing Orig stream needs format 1 bind%s", ""); return GX_ERR_INVALID; }
+        *s_out = c.dstream;
+        *m_out = c.m;
+        *val_out = nullptr;
+        return GX_OK;
+    }
+    int64_t n = c.m.nrows;
+    if (n >= (int64_t) INT32_MAX) return GX_ERR_INVALID;
+    devbuf errb;
+    HIP_CHK(ctx, flat.alloc((size_t) (GX_AOCS_DATUM_OFF +
+                                      std::max<int64_t>(n, 1) * c.m.width)));
+    HIP_CHK(ctx, val.alloc((size_t) std::max<int64_t>(n, 1)));
+    HIP_CHK(ctx, errb.alloc(4));
+    HIP_CHK(ctx, hipMemsetAsync(errb.p, 0, 4, ctx->stream));
+    uint8_t *dv = flat.as<uint8_t>() + GX_AOCS_DATUM_OFF;
+    if (c.m.width == 8)
+        hipLaunchKernelGGL(k_decode_dense<int64_t>, dim3(GRID), dim3(64), 0,
+                           ctx->stream, c.dstream, c.ddir, c.nblocks, n,
+                           (int64_t *) dv, val.as<uint8_t>(), errb.as<int>());
+    else if (c.m.width == 4)
+        hipLaunchKernelGGL(k_decode_dense<int32_t>, dim3(GRID), dim3(64), 0,
+                           ctx->stream, c.dstream, c.ddir, c.nblocks, n,
+                           (int32_t *) dv, val.as<uint8_t>(), errb.as<int>());
+    else
+        hipLaunchKernelGGL(k_decode_dense<int8_t>, dim3(GRID), dim3(64), 0,
+                           ctx->stream, c.dstream, c.ddir, c.nblocks, n,
+                           (int8_t *) dv, val.as<uint8_t>(), errb.as<int>());
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, errb.p, 4, hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr) { set_err(ctx, "groupby: malformed block%s", ""); return GX_ERR_INVALID; }
+    gx_colmeta m{};
+    m.width = c.m.width;
+    m.rpb = (int32_t) std::max<int64_t>(n, 1);
+    m.nrows = n;
+    m.full_block_len = GX_AOCS_DATUM_OFF + n * c.m.width;
+    m.nbytes = m.full_block_len;
+    gx_colmeta_finish(&m);
+    *s_out = flat.as<uint8_t>();
+    *m_out = m;
+    *val_out = c.has_null ? val.as<uint8_t>() : nullptr;
+    return GX_OK;
+}
+
+extern "C" gx_status gx_groupby(gx_ctx *ctx, const gx_table *t, int key_col,
+                                int val_col, gx_kv_group **out,
+                                int64_t *ngroups)
+{
+    if (!ctx || !t || key_col < 0 || key_col >= (int) t->cols.size() ||
+        val_col < 0 || val_col >= (int) t->cols.size() || !out || !ngroups)
+        return GX_ERR_INVALID;
+    const gx_col &kc = t->cols[key_col], &vc = t->cols[val_col];
+    if (kc.m.width != 8 || vc.m.width != 8) return GX_ERR_INVALID;
+    hipStream_t s = ctx->stream;
+    devbuf kflat, kval, vflat, vval;
+    const uint8_t *k_s, *v_s, *k_val, *v_val;
+    gx_colmeta k_m, v_m;
+    gx_status st = gb_flat(ctx, kc, kflat, kval, &k_s, &k_m, &k_val);
+    if (st != GX_OK) return st;
+    st = gb_flat(ctx, vc, vflat, vval, &v_s, &v_m, &v_val);
+    if (st != GX_OK) return st;
+
+    int64_t n = t->nrows;
+    uint64_t tslots = (uint64_t) pow2_at_least(n * 2);
+    {
+        gx_status bs = hbm_budget_check(ctx, tslots * 24 + 64, "groupby table");
+        if (bs != GX_OK) return bs;
+    }
+    devbuf tk, ts, tc, nacc, errb;
+    HIP_CHK(ctx, tk.alloc(tslots * 8));
+    HIP_CHK(ctx, ts.alloc(tslots * 8));
+    HIP_CHK(ctx, tc.alloc(tslots * 8));
+    HIP_CHK(ctx, nacc.alloc(16));
+    HIP_CHK(ctx, errb.alloc(4));
+    HIP_CHK(ctx, hipMemsetAsync(tk.p, 0, tslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(ts.p, 0, tslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(tc.p, 0, tslots * 8, s));
+    HIP_CHK(ctx, hipMemsetAsync(nacc.p, 0, 16, s));
+    HIP_CHK(ctx, hipMemsetAsync(errb.p, 0, 4, s));
+    hipLaunchKernelGGL(k_groupby, dim3(GRID), dim3(TPB), 0, s,
+                       k_s, k_m, k_val, v_s, v_m, v_val, t->dvmap,
+                       tk.as<unsigned long long>(), ts.as<double>(),
+                       tc.as<unsigned long long>(), tslots - 1,
+                       nacc.as<double>(),
+                       nacc.as<unsigned long long>() + 1, errb.as<int>());
+    int herr = 0;
+    double nsum = 0;
+    unsigned long long ncnt = 0;
+    std::vector<unsigned long long> hk(tslots), hc(tslots);
+    std::vector<double> hs(tslots);
+    HIP_CHK(ctx, hipMemcpyAsync(hk.data(), tk.p, tslots * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hs.data(), ts.p, tslots * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(hc.data(), tc.p, tslots * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&nsum, nacc.p, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&ncnt, (char *) nacc.p + 8, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, errb.p, 4, hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipStreamSynchronize(s));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr & 4)
+    { set_err(ctx, "groupby: key INT64_MIN unsupported (biased sentinel)%s", ""); return GX_ERR_INVALID; }
+
+    std::vector<gx_kv_group> groups;
+    for (uint64_t i = 0; i < tslots; i++)
+        if (hk[i])
+        {
+            gx_kv_group g{};
+            g.key = (int64_t) (hk[i] ^ (1ULL << 63));
+            g.sum = hs[i];
+            g.count = (int64_t) hc[i];
+            groups.push_back(g);
+        }
+    std::sort(groups.begin(), groups.end(),
+              [](const gx_kv_group &a, const gx_kv_group &b)
+              { return a.key < b.key; });
+    if (ncnt)
+    {
+        gx_kv_group g{};
+        g.key_is_null = 1;
+        g.sum = nsum;
+        g.count = (int64_t) ncnt;
+        groups.push_back(g);
+    }
+    *ngroups = (int64_t) groups.size();
+    *out = (gx_kv_group *) malloc(std::max<size_t>(groups.size(), 1) *
+                                  sizeof(gx_kv_group));
+    if (!*out) return GX_ERR_OOM;
+    memcpy(*out, groups.data(), groups.size() * sizeof(gx_kv_group));
+    return GX_OK;
+}
+
 extern "C" gx_status gx_partition(gx_ctx *ctx, const int64_t *host_keys, int64_t n,
                                   int32_t nsegs, int32_t *host_out)
 {
@@ -3422,6 +3651,85 @@ static gx_status q3_build_text_mask(gx_ctx *ctx, gx_q3 *q)
     return GX_OK;
 }
 
+/* role-column accessor: materialized flat override or the bound column */
+static inline const gx_col &q3_col(const gx_q3 *q, const gx_table *t, int c)
+{
+    auto it = q->mat.find(std::make_pair((const gx_table *) t, c));
+    return it != q->mat.end() ? it->second : t->cols[c];
+}
+
+/* ensure a per-table hidden bitmap exists (seeded from the visimap) */
+static gx_status q3_ensure_hidden(gx_ctx *ctx, gx_table *t, uint8_t **hidden);
+
+/* Decode a Dense/RLE (possibly null-bearing) ROLE column once at prepare
+ * into a flat device array readable through the standard gx_col_get
+ * addressing (rpb >= nrows -> always block 0, datums at +GX_AOCS_DATUM_OFF),
+ * and OR its NULL rows into the table's hidden mask: strict-NULL reject for
+ * join keys (nodeHash.c:2168-2181), NULL-qual filtering for filter columns
+ * (execScan.c:241).  The per-step kernels run unchanged. */
+static gx_status q3_materialize_col(gx_ctx *ctx, gx_q3 *q, gx_table *t,
+                                    int cidx, uint8_t **hidden)
+{
+    const gx_col &c = t->cols[cidx];
+    int64_t n = c.m.nrows;
+    int w = c.m.width;
+    if (n >= (int64_t) INT32_MAX)
+    {
+        set_err(ctx, "materialized column too large (rpb is int32)%s", "");
+        return GX_ERR_INVALID;
+    }
+    hipStream_t st = ctx->stream;
+    devbuf flat, val, errb;
+    HIP_CHK(ctx, flat.alloc((size_t) (GX_AOCS_DATUM_OFF +
+                                      std::max<int64_t>(n, 1) * w)));
+    HIP_CHK(ctx, val.alloc((size_t) std::max<int64_t>(n, 1)));
+    HIP_CHK(ctx, errb.alloc(4));
+    HIP_CHK(ctx, hipMemsetAsync(errb.p, 0, 4, st));
+    uint8_t *dvals = flat.as<uint8_t>() + GX_AOCS_DATUM_OFF;
+    if (w == 8)
+        hipLaunchKernelGGL(k_decode_dense<int64_t>, dim3(GRID), dim3(64), 0, st,
+                           c.dstream, c.ddir, c.nblocks, n,
+                           (int64_t *) dvals, val.as<uint8_t>(), errb.as<int>());
+    else if (w == 4)
+        hipLaunchKernelGGL(k_decode_dense<int32_t>, dim3(GRID), dim3(64), 0, st,
+                           c.dstream, c.ddir, c.nblocks, n,
+                           (int32_t *) dvals, val.as<uint8_t>(), errb.as<int>());
+    else
+        hipLaunchKernelGGL(k_decode_dense<int8_t>, dim3(GRID), dim3(64), 0, st,
+                           c.dstream, c.ddir, c.nblocks, n,
+                           (int8_t *) dvals, val.as<uint8_t>(), errb.as<int>());
+    hipLaunchKernelGGL(k_verify_crc_dir, dim3(GRID), dim3(64), 0, st,
+                       c.dstream, c.ddir, c.nblocks, errb.as<int>());
+    int herr = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&herr, errb.p, 4, hipMemcpyDeviceToHost, st));
+    HIP_CHK(ctx, hipStreamSynchronize(st));
+    HIP_CHK(ctx, hipGetLastError());
+    if (herr & 1) { set_err(ctx, "materialize: malformed block%s", ""); return GX_ERR_INVALID; }
+    if (herr & 2) { set_err(ctx, "materialize: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
+    if (c.has_null)
+    {
+        gx_status rs = q3_ensure_hidden(ctx, t, hidden);
+        if (rs != GX_OK) return rs;
+        hipLaunchKernelGGL(k_validity_or_hidden, dim3(GRID), dim3(TPB), 0, st,
+                           val.as<uint8_t>(), n, *hidden);
+        HIP_CHK(ctx, hipStreamSynchronize(st));
+        HIP_CHK(ctx, hipGetLastError());
+    }
+    gx_col oc{};
+    oc.dstream = flat.as<uint8_t>();
+    oc.m.width = w;
+    oc.m.rpb = (int32_t) std::max<int64_t>(n, 1);
+    oc.m.nrows = n;
+    oc.m.full_block_len = GX_AOCS_DATUM_OFF + n * w;
+    oc.m.nbytes = GX_AOCS_DATUM_OFF + n * w;
+    gx_colmeta_finish(&oc.m);
+    oc.format = 0;
+    q->mat[std::make_pair((const gx_table *) t, cidx)] = oc;
+    q->mat_mem.push_back(flat.p);
+    flat.p = nullptr;            /* ownership moves to gx_q3 */
+    return GX_OK;
+}
+
 /* fold an AND-ed extra-qual list into a device HIDDEN bitmap (combined with
  * the table's visimap); runs once at prepare — the per-step kernels then
  * read one bit per row through the existing visibility path */
@@ -3457,6 +3765,22 @@ static gx_status q3_build_qualmask(gx_ctx *ctx, gx_table *t,
     HIP_CHK(ctx, hipGetLastError());
     *out = mb.as<uint8_t>();
     mb.p = nullptr;              /* ownership moves to gx_q3 */
+    return GX_OK;
+}
+
+static gx_status q3_ensure_hidden(gx_ctx *ctx, gx_table *t, uint8_t **hidden)
+{
+    if (*hidden) return GX_OK;
+    int64_t nbytes = (t->nrows + 7) >> 3;
+    devbuf mb;
+    HIP_CHK(ctx, mb.alloc((size_t) std::max<int64_t>(nbytes, 1)));
+    gx_qualargs qa{};            /* n = 0: copies the visimap (or zeros) */
+    hipLaunchKernelGGL(k_qualmask, dim3(GRID), dim3(TPB), 0, ctx->stream,
+                       qa, t->dvmap, t->nrows, mb.as<uint8_t>());
+    HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHK(ctx, hipGetLastError());
+    *hidden = mb.as<uint8_t>();
+    mb.p = nullptr;
     return GX_OK;
 }
 
@@ -3497,17 +3821,47 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
     if (desc->mid_filter.col != desc->mid_attr1_col &&
         desc->mid_filter.col == desc->mid_key_col)
         return GX_ERR_INVALID;
-    /* fused-RLE scan is supported for the fact key column only */
+    /* format-1 (Dense/RLE/null-bearing) columns: the fused probe scans RLE
+     * on the fact key directly (no-null); other KEY/FILTER role columns are
+     * materialized to flat arrays at prepare with strict-NULL reject;
+     * measure/attr roles must be plain fixed-width */
+    auto is_matable = [&](gx_table *t, int ci) {
+        if (t == customer) return ci == desc->dim_key_col ||
+                                  (!dim_text && ci == desc->dim_filter.col);
+        if (t == orders) return ci == desc->mid_key_col ||
+                                ci == desc->mid_fk_col ||
+                                ci == desc->mid_filter.col;
+        return ci == desc->fact_key_col || ci == desc->fact_filter.col;
+    };
+    auto is_used = [&](gx_table *t, int ci) {
+        if (t == customer) return ci == desc->dim_key_col ||
+                                  ci == desc->dim_filter.col;
+        if (t == orders) return ci == desc->mid_key_col ||
+                                ci == desc->mid_fk_col ||
+                                ci == desc->mid_attr1_col ||
+                                ci == desc->mid_attr2_col ||
+                                ci == desc->mid_filter.col;
+        return ci == desc->fact_key_col || ci == desc->fact_a_col ||
+               ci == desc->fact_b_col || ci == desc->fact_filter.col;
+    };
     for (auto *t : {customer, orders, lineitem})
         for (size_t ci = 0; ci < t->cols.size(); ci++)
-            if (t->cols[ci].format != 0 &&
-                !(t == lineitem && (int) ci == desc->fact_key_col) &&
-                !(dim_text && t == customer && (int) ci == desc->dim_filter.col))
+        {
+            const gx_col &c = t->cols[ci];
+            if (c.format == 0 || !is_used(t, (int) ci)) continue;
+            if (dim_text && t == customer && (int) ci == desc->dim_filter.col)
+                continue;                              /* varlena texteq path */
+            if (t == lineitem && (int) ci == desc->fact_key_col && !c.has_null)
+                continue;                              /* fused RLE scan */
+            if (!is_matable(t, (int) ci))
             {
-                set_err(ctx, "Q3 pipeline supports RLE only on the fact key; "
-                             "decode other RLE columns first%s", "");
+                set_err(ctx, "Dense/RLE/null-bearing streams are supported on "
+                             "key and filter roles (materialized at prepare) "
+                             "and the fact key (fused RLE); decode "
+                             "measure/attr columns first%s", "");
                 return GX_ERR_INVALID;
             }
+        }
     gx_q3 *q = new gx_q3();
     q->ctx = ctx;
     q->cust = customer;
@@ -3554,6 +3908,24 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
                 return st;
             }
         }
+    /* materialize format-1 key/filter role columns (strict-NULL reject) */
+    for (auto &e : ex)                  /* same table/mask trio */
+        for (size_t ci = 0; ci < e.t->cols.size(); ci++)
+        {
+            const gx_col &c = e.t->cols[ci];
+            if (c.format == 0 || !is_used(e.t, (int) ci) ||
+                !is_matable(e.t, (int) ci))
+                continue;
+            if (e.t == lineitem && (int) ci == desc->fact_key_col &&
+                !c.has_null)
+                continue;               /* fused RLE path */
+            gx_status st = q3_materialize_col(ctx, q, e.t, (int) ci, e.dst);
+            if (st != GX_OK)
+            {
+                gx_q3_free(q);
+                return st;
+            }
+        }
     *out = q;
     return GX_OK;
 }
@@ -3594,6 +3966,9 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->dtext);
     fr(q->dmask);
     fr(q->qvm_dim); fr(q->qvm_mid); fr(q->qvm_fact);
+    for (void *pm : q->mat_mem) (void) hipFree(pm);
+    q->mat_mem.clear();
+    q->mat.clear();
     q->m_send1_cap = q->m_recv1_cap = q->m_send2_cap = q->m_recv2_cap = 0;
     q->sized = false;
 }
@@ -3632,8 +4007,10 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     gx_ctx *ctx = q->ctx;
     hipStream_t s = ctx->stream;
     const gx_q3_desc &D = q->desc;
-    const gx_col &cm = q->cust->cols[D.dim_filter.col];
-    const gx_col &oc = q->ord->cols[D.mid_fk_col], &od = q->ord->cols[D.mid_filter.col];
+    const gx_col &ckk = q3_col(q, q->cust, D.dim_key_col);
+    const gx_col &cm = q3_col(q, q->cust, D.dim_filter.col);
+    const gx_col &oc = q3_col(q, q->ord, D.mid_fk_col),
+                 &od = q3_col(q, q->ord, D.mid_filter.col);
     /* effective visibility = extra-qual mask (already OR-combined with the
      * table visimap at prepare) or the bare visimap */
     const uint8_t *cvm = q->qvm_dim ? q->qvm_dim : q->cust->dvmap;
@@ -3648,14 +4025,12 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s)); /* borrowed for min custkey */
     if (D.dim_text_len > 0)
         hipLaunchKernelGGL(k_cust_count_mask, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[D.dim_key_col].dstream,
-                           q->cust->cols[D.dim_key_col].m,
+                           ckk.dstream, ckk.m,
                            q->dmask, cvm,
                            q->dcount, q->dhits, q->dmin);
     else
         hipLaunchKernelGGL(k_cust_count, dim3(GRID), dim3(TPB), 0, s,
-                           q->cust->cols[D.dim_key_col].dstream,
-                           q->cust->cols[D.dim_key_col].m,
+                           ckk.dstream, ckk.m,
                            cm.dstream, cm.m, cvm, D.dim_filter.op,
                            (int8_t) D.dim_filter.literal, q->dcount, q->dhits,
                            q->dmin);
@@ -3693,15 +4068,13 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         if (D.dim_text_len > 0)
             hipLaunchKernelGGL((k_cust_build_mask<std::decay_t<decltype(*cs)>>),
                                dim3(GRID), dim3(TPB), 0, s,
-                               q->cust->cols[D.dim_key_col].dstream,
-                               q->cust->cols[D.dim_key_col].m,
+                               ckk.dstream, ckk.m,
                                q->dmask, cvm,
                                cs, q->cmask, q->bloom, q->bwmask);
         else
             hipLaunchKernelGGL((k_cust_build<std::decay_t<decltype(*cs)>>),
                                dim3(GRID), dim3(TPB), 0, s,
-                               q->cust->cols[D.dim_key_col].dstream,
-                               q->cust->cols[D.dim_key_col].m,
+                               ckk.dstream, ckk.m,
                                cm.dstream, cm.m, cvm, D.dim_filter.op,
                                (int8_t) D.dim_filter.literal,
                                cs, q->cmask, q->bloom, q->bwmask);
@@ -3724,8 +4097,8 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s)); /* minkey = ~0 */
         if (q->cset_width == 4)
             hipLaunchKernelGGL(k_orders_count<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               q->ord->cols[D.mid_key_col].dstream,
-                               q->ord->cols[D.mid_key_col].m,
+                               q3_col(q, q->ord, D.mid_key_col).dstream,
+                               q3_col(q, q->ord, D.mid_key_col).m,
                                od.dstream, od.m, oc.dstream, oc.m,
                                ovm,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
@@ -3734,8 +4107,8 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
                                q->dhits, q->dmin);
         else
             hipLaunchKernelGGL(k_orders_count<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               q->ord->cols[D.mid_key_col].dstream,
-                               q->ord->cols[D.mid_key_col].m,
+                               q3_col(q, q->ord, D.mid_key_col).dstream,
+                               q3_col(q, q->ord, D.mid_key_col).m,
                                od.dstream, od.m, oc.dstream, oc.m,
                                ovm,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
@@ -3819,18 +4192,16 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     for (auto &e : ev) HIP_CHK(ctx, e.create());
 
     const gx_q3_desc &D = q->desc;
-    const gx_col &ck = q->cust->cols[D.dim_key_col],
-                 &cm = q->cust->cols[D.dim_filter.col];
-    const gx_col &ok = q->ord->cols[D.mid_key_col],
-                 &oc = q->ord->cols[D.mid_fk_col],
-                 &od = q->ord->cols[D.mid_filter.col],
-                 &op = q->ord->cols[D.mid_attr2_col];
-    const gx_col &oa1 = q->ord->cols[D.mid_attr1_col];
-    (void) oa1;  /* attr1 == the date filter column in the standard plan */
-    const gx_col &lk = q->li->cols[D.fact_key_col],
-                 &lp = q->li->cols[D.fact_a_col],
-                 &ld = q->li->cols[D.fact_b_col],
-                 &ls = q->li->cols[D.fact_filter.col];
+    const gx_col &ck = q3_col(q, q->cust, D.dim_key_col),
+                 &cm = q3_col(q, q->cust, D.dim_filter.col);
+    const gx_col &ok = q3_col(q, q->ord, D.mid_key_col),
+                 &oc = q3_col(q, q->ord, D.mid_fk_col),
+                 &od = q3_col(q, q->ord, D.mid_filter.col),
+                 &op = q3_col(q, q->ord, D.mid_attr2_col);
+    const gx_col &lk = q3_col(q, q->li, D.fact_key_col),
+                 &lp = q3_col(q, q->li, D.fact_a_col),
+                 &ld = q3_col(q, q->li, D.fact_b_col),
+                 &ls = q3_col(q, q->li, D.fact_filter.col);
     unsigned long long *dcount = q->dcount;
     /* effective visibility (extra-qual masks fold the visimap in) */
     const uint8_t *cvm_eff = q->qvm_dim ? q->qvm_dim : q->cust->dvmap;

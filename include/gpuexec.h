@@ -270,6 +270,23 @@ gx_status gx_q3_free(gx_q3 *q);
 
 void gx_free(void *p);
 
+/* ---- standalone hash GROUP BY (nodeAgg.c:2288 hash strategy over one key;
+ * grouping equality is NOT DISTINCT, execGrouping.c:436-495: all NULL keys
+ * form ONE group, returned LAST with key_is_null=1).  COUNT(*) counts every
+ * row; SUM(float8)'s transition is strict (float.c:769) and skips NULL
+ * inputs.  key/val cols are i64/f64, Orig or Dense/RLE (nullable via
+ * format-1 streams).  Key INT64_MIN is rejected (biased sentinel).  Groups
+ * sorted by key asc; caller frees with gx_free. ---- */
+typedef struct gx_kv_group {
+    int64_t key;
+    uint8_t key_is_null;
+    uint8_t _pad[7];
+    double  sum;
+    int64_t count;
+} gx_kv_group;
+gx_status gx_groupby(gx_ctx *ctx, const gx_table *t, int key_col, int val_col,
+                     gx_kv_group **out, int64_t *ngroups);
+
 /* ---- test-only entry points (parity harness; not part of the drop-in) ---- */
 typedef struct gx_ord_row { int64_t okey, ocust; int32_t odate, oprio; } gx_ord_row;
 /* Motion-1 partition kernels on one GPU: filter orders by cutoff, route by

@@ -1168,3 +1168,150 @@ def test_extra_quals_with_visimap(ctx, orc):
     np.testing.assert_array_equal(got["l_orderkey"], keys)
     np.testing.assert_array_equal(got["nitems"], counts)
     li.free(); ordr.free(); cust.free()
+
+
+def test_null_join_keys_q3(ctx, orc):
+    """NULL join keys through the pipeline.  Null-bearing
+    key/filter columns (format-1 streams) are materialized at prepare with
+    strict-NULL reject (nodeHash.c:2168-2181: a NULL key cannot pass the
+    strict hash operator on either side of the inner join) and NULL-qual
+    filtering (execScan.c:241).  Parity vs numpy brute force at ~10-15%
+    NULLs on l_orderkey, o_custkey and o_orderdate."""
+    rng = np.random.default_rng(55)
+    nc, no, nl = 400, 1200, 4000
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    o_cust_null = rng.random(no) < 0.12
+    o_date = rng.integers(-3000, -1000, no).astype(np.int32)
+    o_date_null = rng.random(no) < 0.08
+    o_prio = rng.integers(0, 5, no).astype(np.int32)
+    li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+    li_null = rng.random(nl) < 0.15
+    price = rng.uniform(1, 100, nl)
+    disc = rng.integers(0, 11, nl) / 100.0
+    ship = rng.integers(-2500, -500, nl).astype(np.int32)
+    cut = -1753
+
+    cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                     (orc.aocs_encode(np.zeros(nc, np.int8)), 1, nc)])
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode_orig_nulls(o_cust, o_cust_null), 8, no, 1),
+                     (orc.aocs_encode_orig_nulls(o_date, o_date_null), 4, no, 1),
+                     (orc.aocs_encode(o_prio), 4, no)])
+    li = ctx.bind([(orc.aocs_encode_rle_delta_nulls(li_keys, li_null), 8, nl, 1),
+                   (orc.aocs_encode(price), 8, nl),
+                   (orc.aocs_encode(disc), 8, nl),
+                   (orc.aocs_encode(ship), 4, nl)])
+    got = ctx.q3(cust, ordr, li).run().result()
+
+    om = (~o_cust_null & ~o_date_null & (o_date < cut) &
+          np.isin(o_cust, c_keys))
+    lm = (~li_null & (ship > cut) & np.isin(li_keys, o_keys[om]))
+    keys, counts = np.unique(li_keys[lm], return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    rev = {k: 0.0 for k in keys.tolist()}
+    for k, p, dsc in zip(li_keys[lm].tolist(), price[lm], disc[lm]):
+        rev[k] += p * (1.0 - dsc)
+    np.testing.assert_allclose(got["revenue"],
+                               np.array([rev[k] for k in keys.tolist()]),
+                               rtol=1e-6)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_null_dim_key_strict_reject(ctx, orc):
+    """NULL dim (customer) keys never enter the semijoin set."""
+    rng = np.random.default_rng(56)
+    nc = 300
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    c_null = rng.random(nc) < 0.2
+    o_keys = np.arange(1, nc + 1, dtype=np.int64)
+    o_cust = np.arange(1, nc + 1, dtype=np.int64)   # 1:1 with customers
+    li_keys = np.arange(1, nc + 1, dtype=np.int64)
+    cust = ctx.bind([(orc.aocs_encode_orig_nulls(c_keys, c_null), 8, nc, 1),
+                     (orc.aocs_encode(np.zeros(nc, np.int8)), 1, nc)])
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, nc),
+                     (orc.aocs_encode(o_cust), 8, nc),
+                     (orc.aocs_encode(np.full(nc, -9999, np.int32)), 4, nc),
+                     (orc.aocs_encode(np.zeros(nc, np.int32)), 4, nc)])
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nc),
+                   (orc.aocs_encode(np.full(nc, 10.0)), 8, nc),
+                   (orc.aocs_encode(np.zeros(nc)), 8, nc),
+                   (orc.aocs_encode(np.full(nc, 9999, np.int32)), 4, nc)])
+    got = ctx.q3(cust, ordr, li).run().result()
+    np.testing.assert_array_equal(got["l_orderkey"], c_keys[~c_null])
+    li.free(); ordr.free(); cust.free()
+
+
+def test_groupby_null_keys(ctx, orc):
+    """NULLs-equal grouping (execGrouping.c:436-495): all NULL group keys
+    form ONE group; SUM skips NULL inputs (strict transfn), COUNT(*) counts
+    them.  GPU vs numpy on 15% NULL keys / 10% NULL values."""
+    rng = np.random.default_rng(57)
+    n = 50000
+    keys = rng.integers(-50, 2000, n).astype(np.int64)
+    knull = rng.random(n) < 0.15
+    vals = rng.uniform(-5, 5, n)
+    vnull = rng.random(n) < 0.10
+    t = ctx.bind([(orc.aocs_encode_orig_nulls(keys, knull), 8, n, 1),
+                  (orc.aocs_encode_orig_nulls(vals, vnull), 8, n, 1)])
+    got = ctx.groupby(t, 0, 1)
+
+    want_keys = np.unique(keys[~knull])
+    body = got["key"][~got["key_is_null"]]
+    np.testing.assert_array_equal(body, want_keys)
+    for i, k in enumerate(want_keys):
+        m = (~knull) & (keys == k)
+        assert got["count"][i] == int(m.sum())
+        np.testing.assert_allclose(got["sum"][i],
+                                   vals[m & ~vnull].sum(), rtol=1e-9, atol=1e-9)
+    # the NULL group is last
+    assert got["key_is_null"][-1]
+    assert got["count"][-1] == int(knull.sum())
+    np.testing.assert_allclose(got["sum"][-1], vals[knull & ~vnull].sum(),
+                               rtol=1e-9, atol=1e-9)
+    t.free()
+
+
+def test_groupby_not_null_columns(ctx, orc):
+    """groupby on plain NOT NULL Orig columns (no NULL group at all)."""
+    rng = np.random.default_rng(58)
+    n = 10000
+    keys = rng.integers(0, 500, n).astype(np.int64)
+    vals = rng.uniform(0, 10, n)
+    t = ctx.bind([(orc.aocs_encode(keys), 8, n),
+                  (orc.aocs_encode(vals), 8, n)])
+    got = ctx.groupby(t, 0, 1)
+    want_keys = np.unique(keys)
+    np.testing.assert_array_equal(got["key"], want_keys)
+    assert not got["key_is_null"].any()
+    for i, k in enumerate(want_keys):
+        m = keys == k
+        assert got["count"][i] == int(m.sum())
+        np.testing.assert_allclose(got["sum"][i], vals[m].sum(), rtol=1e-9)
+    t.free()
+
+
+def test_rle_key_roles_materialized(ctx, orc):
+    """RLE streams on NON-fact key roles (o_orderkey, o_custkey) now work
+    via prepare-time materialization — results equal the plain-format run."""
+    sf = 0.02
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    want = ctx.q3(cust, ordr, li).run().result()
+    okeys = ordr.decode_column(0, np.int64)
+    ocust = ordr.decode_column(1, np.int64)
+    odate = ordr.decode_column(2, np.int32)
+    oprio = ordr.decode_column(3, np.int32)
+    no = len(okeys)
+    ordr2 = ctx.bind([(orc.aocs_encode_rle_delta(okeys), 8, no, 1),
+                      (orc.aocs_encode_rle(ocust), 8, no, 1),
+                      (orc.aocs_encode(odate), 4, no),
+                      (orc.aocs_encode(oprio), 4, no)])
+    got = ctx.q3(cust, ordr2, li).run().result()
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
+    ordr2.free(); li.free(); ordr.free(); cust.free()
