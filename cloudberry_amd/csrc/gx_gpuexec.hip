@@ -1267,6 +1267,83 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             __syncthreads();
         }
     }
+    else if constexpr (B == -14)
+    {
+        /* glds double-buffered tile scan: the ship+key STREAM loads move
+         * to async global->LDS DMA (fire-and-forget on the VM counter, no
+         * VGPR destinations, no wave stall at issue), so each wave's
+         * latency budget is spent exclusively on the probe gathers and
+         * hit processing while the NEXT tile's stream bytes are already
+         * in flight.  The r1 decomposition showed the three chain stages
+         * exactly additive at max occupancy — this is the one formulation
+         * that adds outstanding-request capacity instead of re-scheduling
+         * the same per-wave budget (cdna_hip_programming.md §5 glds).
+         * Requires: fixed-format streams, even rpb for the 8-B key column
+         * (16-B glds covers two consecutive rows; rpb 4090 at the default
+         * 32 KB blocksize), no visimap.  Env-gated experiment. */
+        constexpr int TILE = 1024;
+        __shared__ int32_t s_ship[2][TILE];
+        __shared__ int64_t s_key[2][TILE];
+        const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+        const int nwaves = blockDim.x >> 6;
+        const int wrows = TILE / nwaves;
+        int64_t chunk = (((lk_m.nrows + gridDim.x - 1) / gridDim.x) + 1) & ~1LL;
+        int64_t lo = blockIdx.x * chunk;
+        int64_t hi = min(lo + chunk, lk_m.nrows);
+        if (lo >= hi) { gx_wave_count_add(hits, local_hits); return; }
+        int64_t full_end = lo + ((hi - lo) / TILE) * TILE;
+        auto goff = [](const gx_colmeta &m, int64_t row) -> int64_t {
+            int64_t b = (int64_t) gx_mulhi64((uint64_t) row, m.magic);
+            return b * m.full_block_len + GX_AOCS_DATUM_OFF +
+                   (row - b * m.rpb) * m.width;
+        };
+        auto issue_tile = [&](int64_t t0, int buf) {
+            int64_t wbase = t0 + wave * wrows;
+            for (int g = 0; g < wrows / 64; g++)
+            {
+                const uint8_t *ga = sh_s + goff(sh_m, wbase + g * 64 + lane);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void *) ga,
+                    (__attribute__((address_space(3))) void *)
+                        &s_ship[buf][wave * wrows + g * 64], 4, 0, 0);
+            }
+            for (int g = 0; g < wrows / 128; g++)
+            {
+                const uint8_t *ga = lk_s + goff(lk_m, wbase + g * 128 + lane * 2);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void *) ga,
+                    (__attribute__((address_space(3))) void *)
+                        &s_key[buf][wave * wrows + g * 128], 16, 0, 0);
+            }
+        };
+        if (lo < full_end)
+            issue_tile(lo, 0);
+        for (int64_t t0 = lo; t0 < full_end; t0 += TILE)
+        {
+            int cur = (int) (((t0 - lo) / TILE) & 1);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();      /* tile cur visible to all */
+            if (t0 + TILE < full_end)
+                issue_tile(t0 + TILE, cur ^ 1);
+            for (int i = threadIdx.x; i < TILE; i += blockDim.x)
+            {
+                if (!gx_cmp(fop, s_ship[cur][i], flit)) continue;
+                uint64_t k = (uint64_t) s_key[cur][i];
+                uint64_t slot = smap.slot0(k);
+                uint64_t r = resolve(k, slot, tkey[slot]);
+                if (r != ~0ULL) hit(r, t0 + i);
+            }
+            __builtin_amdgcn_s_barrier();      /* done reading buf cur */
+        }
+        for (int64_t i = full_end + threadIdx.x; i < hi; i += blockDim.x)
+        {
+            if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
+            uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
+            uint64_t slot = smap.slot0(k);
+            uint64_t r = resolve(k, slot, tkey[slot]);
+            if (r != ~0ULL) hit(r, i);
+        }
+    }
     else if constexpr (B == -9)
     {
         /* DIAGNOSTIC ONLY (wrong results): B=1 without the ship filter —
@@ -4648,6 +4725,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 13: launch(k_li_probe_agg_t<-11, unsigned int>, keys); break;
                 case 14: launch(k_li_probe_agg_t<-12, unsigned int>, keys); break;
                 case 15: launch(k_li_probe_agg_t<-13, unsigned int>, keys); break;
+                case 16: launch(k_li_probe_agg_t<-14, unsigned int>, keys); break;
             }
         }
         else
@@ -4673,6 +4751,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                 case 13: launch(k_li_probe_agg_t<-11, unsigned long long>, keys); break;
                 case 14: launch(k_li_probe_agg_t<-12, unsigned long long>, keys); break;
                 case 15: launch(k_li_probe_agg_t<-13, unsigned long long>, keys); break;
+                case 16: launch(k_li_probe_agg_t<-14, unsigned long long>, keys); break;
             }
         }
     }
