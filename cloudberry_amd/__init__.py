@@ -49,7 +49,9 @@ class _Stats(ctypes.Structure):
                 ("li_rows", ctypes.c_int64),
                 ("probe_hits", ctypes.c_int64),
                 ("groups", ctypes.c_int64),
-                ("bytes_scanned", ctypes.c_double)]
+                ("bytes_scanned", ctypes.c_double),
+                ("ms_motion_counts", ctypes.c_double),
+                ("ms_motion_payload", ctypes.c_double)]
 
 
 class _KvGroup(ctypes.Structure):

@@ -32,6 +32,7 @@ extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap, const void *src,
 extern "C" unsigned ZSTD_isError(size_t code);
 
 #include <algorithm>
+#include <chrono>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -1035,16 +1036,14 @@ template <typename KS>
 __device__ __forceinline__ bool d_set_contains(const KS *set,
                                                uint64_t mask, uint64_t k)
 {
-    /* narrow-set guard: u32 slots mean every RESIDENT key < 2^32 (the sizing
-     * pass's max is over build-side keys only); a probe key ≥ 2^32 can never
-     * match and must not be compared truncated */
-    if (sizeof(KS) == 4 && (k >> 32)) return false;
+    /* u32 slots hold only resident keys < 2^32 (sizing max); the compare
+     * zero-extends the stored key so a wider probe key never matches */
     uint64_t slot = gx_hmix64(k) & mask;
     while (true)
     {
         KS v = set[slot];
         if (v == (KS) 0) return false;
-        if (v == (KS) k) return true;
+        if ((uint64_t) v == k) return true;   /* zext compare: width-safe */
         slot = (slot + 1) & mask;
     }
 }
@@ -1174,14 +1173,14 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
     uint64_t tmask = smap.mask;
     unsigned long long local_hits = 0;
     auto resolve = [&](uint64_t k, uint64_t slot, KT v) -> uint64_t {
-        /* narrow-table guard: u32 slots hold only build keys < 2^32; a probe
-         * key ≥ 2^32 must miss, not match its truncated low half */
-        if (sizeof(KT) == 4 && (k >> 32)) return ~0ULL;
-        /* first slot already loaded as v; walk on collision */
+        /* first slot already loaded as v; walk on collision; the compare
+         * ZERO-EXTENDS the stored key, so in u32 mode a probe key >= 2^32
+         * never matches — width-safe without a hot-loop branch (a per-row
+         * guard here measured +57%, the r1 visimap lesson) */
         while (true)
         {
             if (v == (KT) 0) return ~0ULL;
-            if (v == (KT) k) return slot;
+            if ((uint64_t) v == k) return slot;
             slot = (slot + 1) & tmask;
             v = tkey[slot];
         }
@@ -1783,14 +1782,13 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
         for (int p = threadIdx.x; p < phys; p += blockDim.x)
         {
             uint64_t k = (uint64_t) datum[p];
-            if (sizeof(KT) == 4 && (k >> 32)) continue;  /* narrow-table guard */
             uint64_t slot = smap.slot0(k);
             bool found = false;
             while (true)
             {
                 KT v = tkey[slot];
                 if (v == (KT) 0) break;
-                if (v == (KT) k) { found = true; break; }
+                if ((uint64_t) v == k) { found = true; break; }  /* zext */
                 slot = (slot + 1) & tmask;
             }
             if (!found) continue;
@@ -1841,14 +1839,13 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
             if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(fop, gx_col_get<int32_t>(sh_s, sh_m, i), flit)) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(lk_s, lk_m, i);
-        if (sizeof(KT) == 4 && (k >> 32)) continue;  /* narrow-table guard */
         uint64_t slot = smap.slot0(k);
         bool found = false;
         while (true)
         {
             KT v = tkey[slot];
             if (v == (KT) 0) break;
-            if (v == (KT) k) { found = true; break; }
+            if ((uint64_t) v == k) { found = true; break; }  /* zext */
             slot = (slot + 1) & tmask;
         }
         if (!found) continue;
@@ -4431,14 +4428,31 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             HIP_CHK(ctx, hipMalloc(&q->m_cnts_mine, n * 8));
             HIP_CHK(ctx, hipMalloc(&q->m_cnts_all, (int64_t) n * n * 8));
         }
+        /* Motion-1 count exchange: the per-dest histogram goes straight
+         * into the all-gather, so ONE host sync yields both this rank's
+         * send layout (its own row) and every peer's counts (r2: the two
+         * separate count syncs were ~half the fixed motion latency) */
         unsigned long long *dhist = q->m_hist;
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                            od.dstream, od.m, oc.dstream, oc.m, ovm_eff,
                            D.mid_filter.op, (int32_t) D.mid_filter.literal, n, dhist);
-        std::vector<unsigned long long> h1(n);
-        HIP_CHK(ctx, hipMemcpyAsync(h1.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
+        unsigned long long *dcnts_all = q->m_cnts_all;
+        double t_counts = 0;
+        auto now_ms = []() {
+            return (double) std::chrono::duration_cast<std::chrono::nanoseconds>(
+                       std::chrono::steady_clock::now().time_since_epoch())
+                       .count() / 1e6;
+        };
+        double tc0 = now_ms();
+        RCCL_CHK(ctx, ncclAllGather(dhist, dcnts_all, n, ncclUint64, ctx->comm, s));
+        std::vector<unsigned long long> cnts_all(n * n);
+        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
+                                    hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
+        t_counts += now_ms() - tc0;
+        std::vector<unsigned long long> h1(n);
+        for (int i = 0; i < n; i++) h1[i] = cnts_all[(int64_t) ctx->seg * n + i];
         std::vector<unsigned long long> off1(n + 1, 0);
         for (int i = 0; i < n; i++) off1[i + 1] = off1[i] + h1[i];
         unsigned long long send1_n = off1[n];
@@ -4450,16 +4464,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                            op.dstream, op.m, ovm_eff, D.mid_filter.op,
                            (int32_t) D.mid_filter.literal, n, dcur, send1);
-
-        /* exchange counts (all-gather of per-dest counts) */
-        unsigned long long *dcnts_mine = q->m_cnts_mine;
-        unsigned long long *dcnts_all = q->m_cnts_all;
-        HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h1.data(), n * 8, hipMemcpyHostToDevice, s));
-        RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
-        std::vector<unsigned long long> cnts_all(n * n);
-        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
-                                    hipMemcpyDeviceToHost, s));
-        HIP_CHK(ctx, hipStreamSynchronize(s));
         std::vector<unsigned long long> rcv1(n), roff1(n + 1, 0);
         for (int r = 0; r < n; r++) rcv1[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff1[r + 1] = roff1[r] + rcv1[r];
@@ -4468,6 +4472,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         gx_ord_row *recv1 = q->m_recv1;
         /* self-share bypasses RCCL entirely: a device copy is both faster
          * and avoids self-send at GB sizes (alltoallv standard practice) */
+        evholder pev0, pev1, pev2, pev3;
+        HIP_CHK(ctx, pev0.create()); HIP_CHK(ctx, pev1.create());
+        HIP_CHK(ctx, pev2.create()); HIP_CHK(ctx, pev3.create());
+        HIP_CHK(ctx, hipEventRecord(pev0, s));
         if (h1[ctx->seg])
             HIP_CHK(ctx, hipMemcpyAsync(recv1 + roff1[ctx->seg],
                                         send1 + off1[ctx->seg],
@@ -4487,6 +4495,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                                    r, ctx->comm, s));
         }
         RCCL_CHK(ctx, ncclGroupEnd());
+        HIP_CHK(ctx, hipEventRecord(pev1, s));
 
         /* Motion 2: probe local customer set, route qualifying by o_orderkey */
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
@@ -4498,9 +4507,15 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             hipLaunchKernelGGL(k_qual_hist<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
                                q->cmask, q->bloom, q->bwmask, n, dhist);
-        std::vector<unsigned long long> h2(n);
-        HIP_CHK(ctx, hipMemcpyAsync(h2.data(), dhist, n * 8, hipMemcpyDeviceToHost, s));
+        /* Motion-2 count exchange: same single-sync shape */
+        tc0 = now_ms();
+        RCCL_CHK(ctx, ncclAllGather(dhist, dcnts_all, n, ncclUint64, ctx->comm, s));
+        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
+                                    hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
+        t_counts += now_ms() - tc0;
+        std::vector<unsigned long long> h2(n);
+        for (int i = 0; i < n; i++) h2[i] = cnts_all[(int64_t) ctx->seg * n + i];
         std::vector<unsigned long long> off2(n + 1, 0);
         for (int i = 0; i < n; i++) off2[i + 1] = off2[i] + h2[i];
         if (grow(q->m_send2, q->m_send2_cap, off2[n]) != GX_OK) return GX_ERR_OOM;
@@ -4514,17 +4529,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             hipLaunchKernelGGL(k_qual_emit<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
                                q->cmask, q->bloom, q->bwmask, n, dcur, send2);
-        HIP_CHK(ctx, hipMemcpyAsync(dcnts_mine, h2.data(), n * 8, hipMemcpyHostToDevice, s));
-        RCCL_CHK(ctx, ncclAllGather(dcnts_mine, dcnts_all, n, ncclUint64, ctx->comm, s));
-        HIP_CHK(ctx, hipMemcpyAsync(cnts_all.data(), dcnts_all, (int64_t) n * n * 8,
-                                    hipMemcpyDeviceToHost, s));
-        HIP_CHK(ctx, hipStreamSynchronize(s));
         std::vector<unsigned long long> rcv2(n), roff2(n + 1, 0);
         for (int r = 0; r < n; r++) rcv2[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff2[r + 1] = roff2[r] + rcv2[r];
         unsigned long long recv2_n = roff2[n];
         if (grow(q->m_recv2, q->m_recv2_cap, recv2_n) != GX_OK) return GX_ERR_OOM;
         gx_qual_row *recv2 = q->m_recv2;
+        HIP_CHK(ctx, hipEventRecord(pev2, s));
         if (h2[ctx->seg])
             HIP_CHK(ctx, hipMemcpyAsync(recv2 + roff2[ctx->seg],
                                         send2 + off2[ctx->seg],
@@ -4544,6 +4555,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                                    r, ctx->comm, s));
         }
         RCCL_CHK(ctx, ncclGroupEnd());
+        HIP_CHK(ctx, hipEventRecord(pev3, s));
 
         qual = (int64_t) recv2_n;
         /* key stats over the received rows — the motion path then gets the
@@ -4610,9 +4622,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                q->tdate, q->tprio, q->smap);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
-        float mms = 0;
+        float mms = 0, p01 = 0, p23 = 0;
         (void) hipEventElapsedTime(&mms, mev0, mev1);
+        (void) hipEventElapsedTime(&p01, pev0, pev1);
+        (void) hipEventElapsedTime(&p23, pev2, pev3);
         ms_motion = mms;
+        q->stats.ms_motion_counts = t_counts;
+        q->stats.ms_motion_payload = (double) p01 + (double) p23;
     }
     q->qual_orders = qual;
     HIP_CHK(ctx, hipEventRecord(ev[2], s));

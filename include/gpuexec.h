@@ -188,6 +188,8 @@ typedef struct gx_q3_stats {
     double ms_extract;
     double ms_motion;          /* partition+exchange (0 at nsegs==1) */
     double ms_total;           /* event span over the whole pipeline */
+    double ms_motion_counts;   /* count all-gathers incl. their host syncs */
+    double ms_motion_payload;  /* row payload send/recv (device events) */
     int64_t cust_rows, ord_rows, li_rows;
     int64_t probe_hits, groups;
     double bytes_scanned;      /* logical uncompressed bytes (cdbaocsam.h:283) */

@@ -78,16 +78,23 @@ def test_rle_stream_gpu_decode(ctx, orc):
 
 
 def test_rle_q3_input_rejected(ctx, orc):
-    """The fused Q3 kernels require fixed-format streams (for now) — an RLE
-    table must be rejected loudly, never silently mis-scanned."""
-    keys = np.repeat(np.arange(1, 1000, dtype=np.int64), 3)
-    t = ctx.bind([(orc.aocs_encode_rle(keys), 8, len(keys), 1),
-                  (orc.aocs_encode(np.zeros(len(keys), np.int8)), 1, len(keys))])
+    """RLE on key/filter roles is materialized at prepare (r2), but a
+    Dense/RLE MEASURE column (fact price) must still be rejected loudly,
+    never silently mis-scanned."""
+    c1 = ctx.tpch_gen(gx.TPCH_CUSTOMER, 0.01)
     c2 = ctx.tpch_gen(gx.TPCH_ORDERS, 0.01)
-    c3 = ctx.tpch_gen(gx.TPCH_LINEITEM, 0.01)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, 0.01)
+    n = li.nrows
+    keys = li.decode_column(0, np.int64)
+    ship = li.decode_column(3, np.int32)
+    price_rle = np.repeat(np.float64(7.0), n)   # compressible measure
+    li_bad = ctx.bind([(orc.aocs_encode(keys), 8, n),
+                       (orc.aocs_encode_rle(price_rle.view(np.int64)), 8, n, 1),
+                       (orc.aocs_encode(np.zeros(n)), 8, n),
+                       (orc.aocs_encode(ship), 4, n)])
     with pytest.raises(gx.GxError):
-        ctx.q3(t, c2, c3)
-    t.free(); c2.free(); c3.free()
+        ctx.q3(c1, c2, li_bad)
+    li_bad.free(); li.free(); c2.free(); c1.free()
 
 
 def test_corrupted_stream_rejected(ctx, orc):

@@ -13,7 +13,6 @@ import torch.multiprocessing as mp
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SF = 0.02
-NSEGS = 2
 
 
 def _exchange(dist, arrays, dest, nsegs, rank):
@@ -55,7 +54,7 @@ def _exchange(dist, arrays, dest, nsegs, rank):
 def _worker(rank, world, result_q):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = "29517"
+    os.environ["MASTER_PORT"] = str(29500 + 17 * world)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     sys.path.insert(0, ROOT)
     from oracle import pyapi as orc
@@ -85,14 +84,14 @@ def _worker(rank, world, result_q):
     dist.destroy_process_group()
 
 
-def test_two_rank_gloo_motion_equals_global():
+def _run_motion_pipeline(nsegs):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, NSEGS, q)) for r in range(NSEGS)]
+    procs = [ctx.Process(target=_worker, args=(r, nsegs, q)) for r in range(nsegs)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(NSEGS):
+    for _ in range(nsegs):
         rank, res = q.get(timeout=300)
         results[rank] = res
     for p in procs:
@@ -101,14 +100,24 @@ def test_two_rank_gloo_motion_equals_global():
 
     from oracle import pyapi as orc
     glob = orc.q3(orc.gen_customer(SF), orc.gen_orders(SF), orc.gen_lineitem(SF))
-    keys = np.concatenate([results[r]["l_orderkey"] for r in range(NSEGS)])
-    rev = np.concatenate([results[r]["revenue"] for r in range(NSEGS)])
-    cnt = np.concatenate([results[r]["nitems"] for r in range(NSEGS)])
+    keys = np.concatenate([results[r]["l_orderkey"] for r in range(nsegs)])
+    rev = np.concatenate([results[r]["revenue"] for r in range(nsegs)])
+    cnt = np.concatenate([results[r]["nitems"] for r in range(nsegs)])
     order = np.argsort(keys)
     assert len(keys) == len(glob["l_orderkey"])
     assert (keys[order] == glob["l_orderkey"]).all()
     assert (cnt[order] == glob["nitems"]).all()
     np.testing.assert_allclose(rev[order], glob["revenue"], rtol=1e-9)
     # and each rank only produced groups that route to it
-    for r in range(NSEGS):
-        assert (orc.route(results[r]["l_orderkey"], NSEGS) == r).all()
+    for r in range(nsegs):
+        assert (orc.route(results[r]["l_orderkey"], nsegs) == r).all()
+
+
+def test_two_rank_gloo_motion_equals_global():
+    _run_motion_pipeline(2)
+
+
+def test_four_rank_gloo_motion_equals_global():
+    """the exact count layout the 8-rank scale bench uses,
+    replayed at ws=4 on CPU (pairwise exchanges, per-rank shard gen)."""
+    _run_motion_pipeline(4)
