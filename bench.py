@@ -41,15 +41,27 @@ def _usable_cores():
     return len(os.sched_getaffinity(0))
 
 
-def cpu_baseline_leg(sample_sf=8.0):
+def cpu_baseline_leg(sample_sf=8.0, full=False):
     """Oracle (CPU restatement, 'port', OpenMP) timed on this host: the
     reported CPU baseline (BASELINE.md — the reference publishes no
-    numbers).  Bounded sample; pipeline only (inputs pre-materialized)."""
+    numbers).  Bounded sample by default; --full-cpu-baseline times the
+    COMPLETE workload SF instead (one pass, no extrapolation — VERDICT r01
+    weak #4)."""
     cores = int(os.environ.get("GX_CPU_THREADS", _usable_cores()))
     from oracle import pyapi as orc   # checker/baseline use only
     c = orc.gen_customer(sample_sf)
     o = orc.gen_orders(sample_sf)
     li = orc.gen_lineitem(sample_sf)
+    if full:
+        orc.set_threads(cores)
+        t0 = time.perf_counter()
+        orc.q3(c, o, li)
+        dt = time.perf_counter() - t0
+        rows = len(li["l_orderkey"])
+        return {"value": rows / dt, "unit": "rows/s", "cores": cores,
+                "kind": "port",
+                "sample": f"tpch_q3_sf{sample_sf:g} oracle pipeline FULL x1 "
+                          f"({rows} lineitem rows, {cores} OpenMP threads)"}
     # calibrate: cgroup quotas can make "nproc" threads slower than fewer —
     # pick the faster of {1, cores/2, cores} on one pass and report that count
     best, cores_eff = None, 1
@@ -85,6 +97,13 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--sf", type=float, default=100.0)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--full-cpu-baseline", action="store_true",
+                    help="time the oracle on the FULL workload SF (one pass, "
+                         "no extrapolation) instead of the bounded sample")
+    ap.add_argument("--no-traffic", action="store_true",
+                    help="skip the rocprofv3 --pmc FETCH_SIZE side-run that "
+                         "fills roofline.traffic (used by the side-run "
+                         "itself to avoid recursion)")
     ap.add_argument("--rle-keys", action="store_true",
                     help="store l_orderkey RLE-compressed (rle_type); the fused "
                          "probe kernel then scans runs, not rows (extra mode — "
@@ -194,9 +213,25 @@ def main():
     probe_ms_avg = probe_ms / args.steps
     li_bytes_local = li_rows_local * 28.0
     achieved = li_bytes_local / 1e9 / (probe_ms_avg / 1000.0)
+    # traffic: DRAM-side read bytes of the dominant kernel from a separate
+    # rocprofv3 --pmc FETCH_SIZE pass (counters only), with the documented
+    # gfx950 x2 correction — tools/pmc_traffic.py.  Fail-soft: null.
+    traffic = None
+    if rank == 0 and n == 1 and not args.no_traffic:
+        try:
+            from tools.pmc_traffic import probe_kernel_fetch_bytes
+            log(rank, "PMC side-run (rocprofv3 --pmc FETCH_SIZE)...")
+            r = probe_kernel_fetch_bytes(args.sf)
+            if r:
+                traffic = round(r[0], 0)
+                log(rank, f"  FETCH_SIZE {r[1]/1048576.0:.2f} GB/launch raw "
+                          f"x2 -> {traffic/1e9:.2f} GB over {r[2]} dispatches")
+        except Exception as e:
+            log(rank, f"  traffic probe failed: {e}")
     roofline = {"bound": "hbm", "achieved": round(achieved, 1),
                 "peak": HBM_PEAK_GBPS, "unit": "GB/s",
-                "frac": round(achieved / HBM_PEAK_GBPS, 4), "traffic": None}
+                "frac": round(achieved / HBM_PEAK_GBPS, 4),
+                "traffic": traffic}
 
     # config-2 evidence: the pure scan+filter kernel's GB/s on the widest
     # filter column (untimed extra; printed into config below)
@@ -208,8 +243,12 @@ def main():
 
     cpu = None
     if rank == 0 and n == 1 and not args.no_cpu_baseline:
-        log(rank, "timing CPU baseline (oracle, bounded sample)...")
-        cpu = cpu_baseline_leg()
+        if args.full_cpu_baseline:
+            log(rank, f"timing CPU baseline (oracle, FULL sf={args.sf:g})...")
+            cpu = cpu_baseline_leg(args.sf, full=True)
+        else:
+            log(rank, "timing CPU baseline (oracle, bounded sample)...")
+            cpu = cpu_baseline_leg()
 
     if rank == 0:
         out = {

@@ -4947,6 +4947,24 @@ extern "C" gx_status gx_q3_free(gx_q3 *q)
 
 extern "C" void gx_free(void *p) { free(p); }
 
+/* ABI self-description: struct sizes for cross-checking foreign-language
+ * mirrors (ctypes tests; the PG extension's abi-check compiles against the
+ * real header instead).  kind: 0 stats, 1 group, 2 kv_group, 3 desc,
+ * 4 coldesc, 5 filter. */
+extern "C" int64_t gx_abi_sizeof(int kind)
+{
+    switch (kind)
+    {
+        case 0: return (int64_t) sizeof(gx_q3_stats);
+        case 1: return (int64_t) sizeof(gx_q3_group);
+        case 2: return (int64_t) sizeof(gx_kv_group);
+        case 3: return (int64_t) sizeof(gx_q3_desc);
+        case 4: return (int64_t) sizeof(gx_coldesc);
+        case 5: return (int64_t) sizeof(gx_filter);
+        default: return -1;
+    }
+}
+
 /* Test ABI: run the Motion-1 partition kernels (filter + route + emit) on a
  * bound/generated orders table for a given nsegs, returning the packed rows
  * and per-destination counts to the host.  Lets the Motion path's kernels be

@@ -188,11 +188,12 @@ typedef struct gx_q3_stats {
     double ms_extract;
     double ms_motion;          /* partition+exchange (0 at nsegs==1) */
     double ms_total;           /* event span over the whole pipeline */
-    double ms_motion_counts;   /* count all-gathers incl. their host syncs */
-    double ms_motion_payload;  /* row payload send/recv (device events) */
     int64_t cust_rows, ord_rows, li_rows;
     int64_t probe_hits, groups;
     double bytes_scanned;      /* logical uncompressed bytes (cdbaocsam.h:283) */
+    /* Motion sub-phases (appended r2; 0 at nsegs==1) */
+    double ms_motion_counts;   /* count all-gathers incl. their host syncs */
+    double ms_motion_payload;  /* row payload send/recv (device events) */
 } gx_q3_stats;
 
 /* customer cols: [c_custkey i64, c_mktsegment i8]
@@ -271,6 +272,10 @@ gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *nout);
 gx_status gx_q3_free(gx_q3 *q);
 
 void gx_free(void *p);
+
+/* ABI self-description for foreign mirrors (0 stats, 1 q3_group,
+ * 2 kv_group, 3 q3_desc, 4 coldesc, 5 filter); -1 for unknown kinds */
+int64_t gx_abi_sizeof(int kind);
 
 /* ---- standalone hash GROUP BY (nodeAgg.c:2288 hash strategy over one key;
  * grouping equality is NOT DISTINCT, execGrouping.c:436-495: all NULL keys

@@ -92,3 +92,19 @@ def test_customscan_extension_compiles():
     r = subprocess.run(["make", "-C", os.path.join(root, "integration"),
                         "abi-check"], capture_output=True, text=True)
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_ctypes_struct_sizes_match_c():
+    """Pin the ctypes mirrors to the C structs (an r2 mid-struct insertion
+    once shifted every int64 stat field — this test makes that impossible
+    to repeat silently)."""
+    import cloudberry_amd as gx_pkg
+    lib = gx_pkg.lib()
+    lib.gx_abi_sizeof.restype = ctypes.c_int64
+    lib.gx_abi_sizeof.argtypes = [ctypes.c_int]
+    assert lib.gx_abi_sizeof(0) == ctypes.sizeof(gx_pkg._Stats)
+    assert lib.gx_abi_sizeof(1) == ctypes.sizeof(gx_pkg._Group)
+    assert lib.gx_abi_sizeof(2) == ctypes.sizeof(gx_pkg._KvGroup)
+    assert lib.gx_abi_sizeof(3) == ctypes.sizeof(gx_pkg._Q3Desc)
+    assert lib.gx_abi_sizeof(4) == ctypes.sizeof(gx_pkg._ColDesc)
+    assert lib.gx_abi_sizeof(5) == ctypes.sizeof(gx_pkg._Filter)

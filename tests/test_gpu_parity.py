@@ -1322,3 +1322,14 @@ def test_rle_key_roles_materialized(ctx, orc):
     np.testing.assert_array_equal(got["nitems"], want["nitems"])
     np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
     ordr2.free(); li.free(); ordr.free(); cust.free()
+
+
+def test_q3_full_parity_sf25(ctx, orc):
+    """the SF25 full-result comparison promoted from a
+    DESIGN.md prose spot-check to a committed test — every group bit-exact
+    on keys/dates/priority/counts, revenue within the stated tolerance,
+    at 150M lineitem rows."""
+    sf = 25.0
+    orc.set_threads(0)            # all cores
+    ng = _q3_parity_at(ctx, orc, sf)
+    assert ng > 3_000_000
