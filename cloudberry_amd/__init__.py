@@ -80,7 +80,8 @@ class _Q3Desc(ctypes.Structure):
                 ("fact_extra", _Filter * 4),
                 ("n_dim_extra", ctypes.c_int32),
                 ("n_mid_extra", ctypes.c_int32),
-                ("n_fact_extra", ctypes.c_int32)]
+                ("n_fact_extra", ctypes.c_int32),
+                ("dim_join", ctypes.c_int32)]
 
 
 class _ColDesc(ctypes.Structure):
@@ -370,6 +371,8 @@ class Context:
             for i, (col, op, lit) in enumerate(quals):
                 arr[i] = _Filter(col, ops[op], int(lit))
             setattr(d, cnt, len(quals))
+        d.dim_join = {"semi": 0, "anti": 1, "anti_notin": 2}[
+            desc_dict.get("dim_join", "semi")]
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
                                                ctypes.byref(q)))

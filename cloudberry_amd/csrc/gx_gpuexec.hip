@@ -1048,8 +1048,10 @@ __device__ __forceinline__ bool d_set_contains(const KS *set,
     }
 }
 
-/* orders local path: count qualifying rows (date filter + customer semijoin) */
-template <typename KS>
+/* orders local path: count qualifying rows (date filter + dim semi/anti
+ * join; ANTI = JOIN_LASJ/LASJ_NOTIN — keep rows whose fk is NOT in the
+ * set, nodeHashjoin.c:652-659) */
+template <typename KS, bool ANTI = false>
 __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *od_s, gx_colmeta od_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
@@ -1068,8 +1070,9 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
         if (gx_vm_hidden(vmap, i)) continue;
         if (!gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
-        if (!d_bloom_test(bloom, bwmask, ck)) continue;
-        if (!d_set_contains(cset, cmask, ck)) continue;
+        bool in_set = d_bloom_test(bloom, bwmask, ck) &&
+                      d_set_contains(cset, cmask, ck);
+        if (in_set == ANTI) continue;
         local++;
         unsigned long long k = (unsigned long long) gx_col_get<int64_t>(ok_s, ok_m, i);
         if (k > kmax) kmax = k;
@@ -1097,7 +1100,8 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
  * key array then sits comfortably in the 256 MiB Infinity Cache), u64
  * otherwise.  Key compares stay exact either way (PG narrow-int hashing
  * spirit; sentinel 0 is safe — orderkeys start at 1). */
-template <typename KT, typename KS, bool CHUNKED = false, bool VM = false>
+template <typename KT, typename KS, bool CHUNKED = false, bool VM = false,
+          bool ANTI = false>
 __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
@@ -1130,8 +1134,9 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) continue;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
-        if (!d_bloom_test(bloom, bwmask, ck)) continue;
-        if (!d_set_contains(cset, cmask, ck)) continue;
+        bool in_set = d_bloom_test(bloom, bwmask, ck) &&
+                      d_set_contains(cset, cmask, ck);
+        if (in_set == ANTI) continue;
         uint64_t k = (uint64_t) gx_col_get<int64_t>(ok_s, ok_m, i);
         uint64_t slot = smap.slot0(k);
         while (true)
@@ -2045,7 +2050,7 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
 }
 
 /* received orders rows: probe local customer set, histogram by route(okey) */
-template <typename KS>
+template <typename KS, bool ANTI = false>
 __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
                             const KS *cset, uint64_t cmask,
                             const unsigned long long *bloom, uint64_t bwmask,
@@ -2059,11 +2064,16 @@ __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
         int64_t i = base + lane;
         bool keep = false;
         int32_t d = 0;
-        if (i < n && d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
-            d_set_contains(cset, cmask, (uint64_t) rows[i].ocust))
+        if (i < n)
         {
-            d = gx_route_i64(rows[i].okey, nsegs);
-            keep = true;
+            bool in_set =
+                d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
+                d_set_contains(cset, cmask, (uint64_t) rows[i].ocust);
+            if (in_set != ANTI)
+            {
+                d = gx_route_i64(rows[i].okey, nsegs);
+                keep = true;
+            }
         }
         for (int dd = 0; dd < nsegs; dd++)
         {
@@ -2074,7 +2084,7 @@ __global__ void k_qual_hist(const gx_ord_row *rows, int64_t n,
     }
 }
 
-template <typename KS>
+template <typename KS, bool ANTI = false>
 __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
                             const KS *cset, uint64_t cmask,
                             const unsigned long long *bloom, uint64_t bwmask,
@@ -2089,11 +2099,16 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
         int64_t i = base + lane;
         bool keep = false;
         int32_t d = 0;
-        if (i < n && d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
-            d_set_contains(cset, cmask, (uint64_t) rows[i].ocust))
+        if (i < n)
         {
-            d = gx_route_i64(rows[i].okey, nsegs);
-            keep = true;
+            bool in_set =
+                d_bloom_test(bloom, bwmask, (uint64_t) rows[i].ocust) &&
+                d_set_contains(cset, cmask, (uint64_t) rows[i].ocust);
+            if (in_set != ANTI)
+            {
+                d = gx_route_i64(rows[i].okey, nsegs);
+                keep = true;
+            }
         }
         for (int dd = 0; dd < nsegs; dd++)
         {
@@ -2179,7 +2194,7 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
  * LDS scan, ONE cursor atomic per block, then re-filter + write — the
  * block's range stays L2-resident between passes).  Pass 2 is the
  * existing k_build_from_rows insert. */
-template <typename KS>
+template <typename KS, bool ANTI = false>
 __global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
                                const uint8_t *oc_s, gx_colmeta oc_m,
                                const uint8_t *od_s, gx_colmeta od_m,
@@ -2201,8 +2216,9 @@ __global__ void k_orders_emitq(const uint8_t *ok_s, gx_colmeta ok_m,
         int32_t od = gx_col_get<int32_t>(od_s, od_m, i);
         if (!gx_cmp(oop, od, olit)) return false;
         uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
-        return d_bloom_test(bloom, bwmask, ck) &&
-               d_set_contains(cset, cmask, ck);
+        bool in_set = d_bloom_test(bloom, bwmask, ck) &&
+                      d_set_contains(cset, cmask, ck);
+        return in_set != ANTI;
     };
 
     unsigned int mine = 0;
@@ -2383,6 +2399,9 @@ struct gx_q3 {
     double *r_rev = nullptr;
     int64_t *r_cnt = nullptr;
     int numeric = 0;                 /* numeric(15,2) scaled-i64 measures */
+    int empty = 0;                   /* LASJ_NOTIN with a NULL dim key:
+                                        the whole result is empty
+                                        (nodeHashjoin.c:442) */
     int64_t rescap = 0;
     int64_t ngroups = 0;
     int64_t qual_orders = 0;
@@ -3742,7 +3761,8 @@ static gx_status q3_ensure_hidden(gx_ctx *ctx, gx_table *t, uint8_t **hidden);
  * join keys (nodeHash.c:2168-2181), NULL-qual filtering for filter columns
  * (execScan.c:241).  The per-step kernels run unchanged. */
 static gx_status q3_materialize_col(gx_ctx *ctx, gx_q3 *q, gx_table *t,
-                                    int cidx, uint8_t **hidden)
+                                    int cidx, uint8_t **hidden,
+                                    bool fold_nulls = true)
 {
     const gx_col &c = t->cols[cidx];
     int64_t n = c.m.nrows;
@@ -3780,7 +3800,7 @@ static gx_status q3_materialize_col(gx_ctx *ctx, gx_q3 *q, gx_table *t,
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 1) { set_err(ctx, "materialize: malformed block%s", ""); return GX_ERR_INVALID; }
     if (herr & 2) { set_err(ctx, "materialize: CRC32C mismatch%s", ""); return GX_ERR_CHECKSUM; }
-    if (c.has_null)
+    if (c.has_null && fold_nulls)
     {
         gx_status rs = q3_ensure_hidden(ctx, t, hidden);
         if (rs != GX_OK) return rs;
@@ -3993,7 +4013,22 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
             if (e.t == lineitem && (int) ci == desc->fact_key_col &&
                 !c.has_null)
                 continue;               /* fused RLE path */
-            gx_status st = q3_materialize_col(ctx, q, e.t, (int) ci, e.dst);
+            /* join-variety NULL rules (nodeHashjoin.c:425,442,652-659):
+             * - LASJ (anti): a NULL mid fk never matches the dim set, so
+             *   the row PASSES — do NOT hide it; its decoded 0 misses the
+             *   set (0 is the rejected sentinel) and the anti test keeps it
+             * - LASJ_NOTIN: a NULL DIM key empties the result — do not
+             *   hide such rows either; sizing sees the decoded 0 on a
+             *   has_null column and flags q->empty */
+            bool fold = true;
+            if (desc->dim_join == 1 && e.t == orders &&
+                (int) ci == desc->mid_fk_col)
+                fold = false;
+            if (desc->dim_join == 2 && e.t == customer &&
+                (int) ci == desc->dim_key_col)
+                fold = false;
+            gx_status st = q3_materialize_col(ctx, q, e.t, (int) ci, e.dst,
+                                              fold);
             if (st != GX_OK)
             {
                 gx_q3_free(q);
@@ -4115,6 +4150,21 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipStreamSynchronize(s));
     if (n_building > 0 && cmin == 0)
     {
+        if (D.dim_join == 2 &&
+            q3_col(q, q->cust, D.dim_key_col).has_null)
+        {
+            /* NOT IN with a NULL key on the inner side: the whole join
+             * yields nothing (nodeHashjoin.c:442 hs_hashkeys_null) */
+            q->empty = 1;
+            q->rescap = 1;
+            HIP_CHK(ctx, hipMalloc(&q->r_okey, 8));
+            HIP_CHK(ctx, hipMalloc(&q->r_odate, 4));
+            HIP_CHK(ctx, hipMalloc(&q->r_oprio, 4));
+            HIP_CHK(ctx, hipMalloc(&q->r_rev, 8));
+            HIP_CHK(ctx, hipMalloc(&q->r_cnt, 8));
+            q->sized = true;
+            return GX_OK;
+        }
         /* slot value 0 is the empty-slot sentinel (keys start at 1 in every
          * PG sequence-keyed table); a real key 0 would be silently dropped */
         set_err(ctx, "dim join key 0 found: 0 is reserved as the empty-slot "

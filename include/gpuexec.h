@@ -246,6 +246,15 @@ typedef struct gx_q3_desc {
     gx_filter mid_extra[GX_MAX_EXTRA_QUALS];
     gx_filter fact_extra[GX_MAX_EXTRA_QUALS];
     int32_t n_dim_extra, n_mid_extra, n_fact_extra;
+    /* dim-join type (nodeHashjoin.c join variety on the dim semijoin):
+     * 0 = JOIN_SEMI (IN / EXISTS — the Q3 shape)
+     * 1 = JOIN_LASJ (anti, NOT EXISTS): mid rows pass when the fk is NOT
+     *     in the dim set; a NULL fk never matches, so it PASSES
+     *     (nodeHashjoin.c:652-659 left-anti semantics)
+     * 2 = JOIN_LASJ_NOTIN (NOT IN): a NULL fk is rejected, and ANY NULL
+     *     dim key passing the dim filter empties the whole result
+     *     (nodeHashjoin.c:425,442 hs_hashkeys_null) */
+    int32_t dim_join;
 } gx_q3_desc;
 
 /* Restrictions checked at sizing (first gx_q3_run):

@@ -119,6 +119,7 @@ typedef struct gx_q3_desc
 	int32		n_dim_extra;
 	int32		n_mid_extra;
 	int32		n_fact_extra;
+	int32		dim_join;
 }			gx_q3_desc;
 
 typedef struct gx_q3_group
