@@ -3818,6 +3818,7 @@ static gx_status q3_materialize_col(gx_ctx *ctx, gx_q3 *q, gx_table *t,
     oc.m.nbytes = GX_AOCS_DATUM_OFF + n * w;
     gx_colmeta_finish(&oc.m);
     oc.format = 0;
+    oc.has_null = c.has_null;    /* sizing consults this (NOTIN emptiness) */
     q->mat[std::make_pair((const gx_table *) t, cidx)] = oc;
     q->mat_mem.push_back(flat.p);
     flat.p = nullptr;            /* ownership moves to gx_q3 */
@@ -3980,7 +3981,8 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
         }
     }
     /* extra AND-ed qual lists → per-table hidden bitmaps */
-    if (desc->n_dim_extra < 0 || desc->n_dim_extra > GX_MAX_EXTRA_QUALS ||
+    if (desc->dim_join < 0 || desc->dim_join > 2 ||
+        desc->n_dim_extra < 0 || desc->n_dim_extra > GX_MAX_EXTRA_QUALS ||
         desc->n_mid_extra < 0 || desc->n_mid_extra > GX_MAX_EXTRA_QUALS ||
         desc->n_fact_extra < 0 || desc->n_fact_extra > GX_MAX_EXTRA_QUALS)
     {
@@ -4219,26 +4221,28 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));   /* borrowed for maxkey */
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s)); /* minkey = ~0 */
-        if (q->cset_width == 4)
-            hipLaunchKernelGGL(k_orders_count<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
+        auto launch_ocount = [&](auto *cs, auto anti) {
+            hipLaunchKernelGGL((k_orders_count<std::decay_t<decltype(*cs)>,
+                                               decltype(anti)::value>),
+                               dim3(GRID), dim3(TPB), 0, s,
                                q3_col(q, q->ord, D.mid_key_col).dstream,
                                q3_col(q, q->ord, D.mid_key_col).m,
                                od.dstream, od.m, oc.dstream, oc.m,
                                ovm,
                                D.mid_filter.op, (int32_t) D.mid_filter.literal,
-                               (const unsigned int *) q->cset, q->cmask,
+                               cs, q->cmask,
                                q->bloom, q->bwmask, q->dcount,
                                q->dhits, q->dmin);
+        };
+        bool anti = D.dim_join != 0;
+        if (q->cset_width == 4 && anti)
+            launch_ocount((const unsigned int *) q->cset, std::true_type{});
+        else if (q->cset_width == 4)
+            launch_ocount((const unsigned int *) q->cset, std::false_type{});
+        else if (anti)
+            launch_ocount((const unsigned long long *) q->cset, std::true_type{});
         else
-            hipLaunchKernelGGL(k_orders_count<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               q3_col(q, q->ord, D.mid_key_col).dstream,
-                               q3_col(q, q->ord, D.mid_key_col).m,
-                               od.dstream, od.m, oc.dstream, oc.m,
-                               ovm,
-                               D.mid_filter.op, (int32_t) D.mid_filter.literal,
-                               (const unsigned long long *) q->cset, q->cmask,
-                               q->bloom, q->bwmask, q->dcount,
-                               q->dhits, q->dmin);
+            launch_ocount((const unsigned long long *) q->cset, std::false_type{});
         unsigned long long nq = 0, kmax = 0, kmin = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&nq, q->dcount, 8, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipMemcpyAsync(&kmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
@@ -4311,6 +4315,16 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         gx_status st = q3_size_and_alloc(q);
         if (st != GX_OK) return st;
     }
+    if (q->empty)
+    {
+        /* LASJ_NOTIN with a NULL inner key: nothing qualifies */
+        q->ngroups = 0;
+        q->stats.cust_rows = q->cust->nrows;
+        q->stats.ord_rows = q->ord->nrows;
+        q->stats.li_rows = q->li->nrows;
+        q->ran = true;
+        return GX_OK;
+    }
 
     evholder ev[8];
     for (auto &e : ev) HIP_CHK(ctx, e.create());
@@ -4375,18 +4389,24 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         int ogrid = env_int("GX_ORDERS_GRID", 32768);  /* measured optimum */
         bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
         const uint8_t *ovm = ovm_eff;
+        bool anti_join = D.dim_join != 0;
         auto launch_build = [&](auto *tk, auto *cs) {
-            auto go = [&](auto ch, auto vm) {
+            auto go2 = [&](auto ch, auto vm, auto anti) {
                 hipLaunchKernelGGL((k_orders_build<std::decay_t<decltype(*tk)>,
                                                    std::decay_t<decltype(*cs)>,
                                                    decltype(ch)::value,
-                                                   decltype(vm)::value>),
+                                                   decltype(vm)::value,
+                                                   decltype(anti)::value>),
                                    dim3(ogrid), dim3(TPB), 0, s,
                                    ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                                    op.dstream, op.m, ovm, D.mid_filter.op,
                                    (int32_t) D.mid_filter.literal, cs, q->cmask,
                                    q->bloom, q->bwmask,
                                    tk, q->tdate, q->tprio, q->smap);
+            };
+            auto go = [&](auto ch, auto vm) {
+                if (anti_join) go2(ch, vm, std::true_type{});
+                else go2(ch, vm, std::false_type{});
             };
             if (ochunk)
             {
@@ -4413,14 +4433,19 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             }
             HIP_CHK(ctx, hipMemsetAsync(q->dcount, 0, 8, s));
             auto launch_emit = [&](auto *cs) {
-                hipLaunchKernelGGL((k_orders_emitq<std::decay_t<decltype(*cs)>>),
-                                   dim3(ogrid), dim3(TPB), 0, s,
-                                   ok.dstream, ok.m, oc.dstream, oc.m,
-                                   od.dstream, od.m, op.dstream, op.m,
-                                   ovm_eff,
-                                   D.mid_filter.op, (int32_t) D.mid_filter.literal,
-                                   cs, q->cmask, q->bloom, q->bwmask,
-                                   q->m_send2, q->dcount);
+                auto goe = [&](auto anti) {
+                    hipLaunchKernelGGL((k_orders_emitq<std::decay_t<decltype(*cs)>,
+                                                       decltype(anti)::value>),
+                                       dim3(ogrid), dim3(TPB), 0, s,
+                                       ok.dstream, ok.m, oc.dstream, oc.m,
+                                       od.dstream, od.m, op.dstream, op.m,
+                                       ovm_eff,
+                                       D.mid_filter.op, (int32_t) D.mid_filter.literal,
+                                       cs, q->cmask, q->bloom, q->bwmask,
+                                       q->m_send2, q->dcount);
+                };
+                if (anti_join) goe(std::true_type{});
+                else goe(std::false_type{});
             };
             if (q->cset_width == 4)
                 launch_emit((const unsigned int *) q->cset);
@@ -4549,14 +4574,22 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
         /* Motion 2: probe local customer set, route qualifying by o_orderkey */
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
-        if (q->cset_width == 4)
-            hipLaunchKernelGGL(k_qual_hist<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
+        bool anti_join = D.dim_join != 0;
+        auto launch_qhist = [&](auto *cs, auto anti) {
+            hipLaunchKernelGGL((k_qual_hist<std::decay_t<decltype(*cs)>,
+                                            decltype(anti)::value>),
+                               dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, cs,
                                q->cmask, q->bloom, q->bwmask, n, dhist);
+        };
+        if (q->cset_width == 4 && anti_join)
+            launch_qhist((const unsigned int *) q->cset, std::true_type{});
+        else if (q->cset_width == 4)
+            launch_qhist((const unsigned int *) q->cset, std::false_type{});
+        else if (anti_join)
+            launch_qhist((const unsigned long long *) q->cset, std::true_type{});
         else
-            hipLaunchKernelGGL(k_qual_hist<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
-                               q->cmask, q->bloom, q->bwmask, n, dhist);
+            launch_qhist((const unsigned long long *) q->cset, std::false_type{});
         /* Motion-2 count exchange: same single-sync shape */
         tc0 = now_ms();
         RCCL_CHK(ctx, ncclAllGather(dhist, dcnts_all, n, ncclUint64, ctx->comm, s));
@@ -4571,14 +4604,21 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         if (grow(q->m_send2, q->m_send2_cap, off2[n]) != GX_OK) return GX_ERR_OOM;
         gx_qual_row *send2 = q->m_send2;
         HIP_CHK(ctx, hipMemcpyAsync(dcur, off2.data(), n * 8, hipMemcpyHostToDevice, s));
-        if (q->cset_width == 4)
-            hipLaunchKernelGGL(k_qual_emit<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
-                               recv1, (int64_t) recv1_n, (const unsigned int *) q->cset,
+        auto launch_qemit = [&](auto *cs, auto anti) {
+            hipLaunchKernelGGL((k_qual_emit<std::decay_t<decltype(*cs)>,
+                                            decltype(anti)::value>),
+                               dim3(GRID), dim3(TPB), 0, s,
+                               recv1, (int64_t) recv1_n, cs,
                                q->cmask, q->bloom, q->bwmask, n, dcur, send2);
+        };
+        if (q->cset_width == 4 && anti_join)
+            launch_qemit((const unsigned int *) q->cset, std::true_type{});
+        else if (q->cset_width == 4)
+            launch_qemit((const unsigned int *) q->cset, std::false_type{});
+        else if (anti_join)
+            launch_qemit((const unsigned long long *) q->cset, std::true_type{});
         else
-            hipLaunchKernelGGL(k_qual_emit<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
-                               recv1, (int64_t) recv1_n, (const unsigned long long *) q->cset,
-                               q->cmask, q->bloom, q->bwmask, n, dcur, send2);
+            launch_qemit((const unsigned long long *) q->cset, std::false_type{});
         std::vector<unsigned long long> rcv2(n), roff2(n + 1, 0);
         for (int r = 0; r < n; r++) rcv2[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff2[r + 1] = roff2[r] + rcv2[r];

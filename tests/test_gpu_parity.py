@@ -1333,3 +1333,118 @@ def test_q3_full_parity_sf25(ctx, orc):
     orc.set_threads(0)            # all cores
     ng = _q3_parity_at(ctx, orc, sf)
     assert ng > 3_000_000
+
+
+def _join_variety_tables(ctx, orc, rng, with_fk_nulls=False,
+                         with_dim_nulls=False):
+    nc, no, nl = 300, 1500, 5000
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    c_seg = (np.arange(nc) % 3).astype(np.int8)    # segment 0 = "BUILDING"
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    o_date = rng.integers(-3000, -1000, no).astype(np.int32)
+    o_prio = rng.integers(0, 5, no).astype(np.int32)
+    li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+    price = rng.uniform(1, 100, nl)
+    disc = rng.integers(0, 11, nl) / 100.0
+    ship = rng.integers(-2500, -500, nl).astype(np.int32)
+    fk_null = (rng.random(no) < 0.12) if with_fk_nulls else np.zeros(no, bool)
+    dim_null = (rng.random(nc) < 0.1) if with_dim_nulls else np.zeros(nc, bool)
+    if with_dim_nulls:
+        cust = ctx.bind([(orc.aocs_encode_orig_nulls(c_keys, dim_null), 8, nc, 1),
+                         (orc.aocs_encode(c_seg), 1, nc)])
+    else:
+        cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                         (orc.aocs_encode(c_seg), 1, nc)])
+    if with_fk_nulls:
+        ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                         (orc.aocs_encode_orig_nulls(o_cust, fk_null), 8, no, 1),
+                         (orc.aocs_encode(o_date), 4, no),
+                         (orc.aocs_encode(o_prio), 4, no)])
+    else:
+        ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                         (orc.aocs_encode(o_cust), 8, no),
+                         (orc.aocs_encode(o_date), 4, no),
+                         (orc.aocs_encode(o_prio), 4, no)])
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nl),
+                   (orc.aocs_encode(price), 8, nl),
+                   (orc.aocs_encode(disc), 8, nl),
+                   (orc.aocs_encode(ship), 4, nl)])
+    data = dict(c_keys=c_keys, c_seg=c_seg, o_keys=o_keys, o_cust=o_cust,
+                o_date=o_date, li_keys=li_keys, price=price, disc=disc,
+                ship=ship, fk_null=fk_null, dim_null=dim_null)
+    return cust, ordr, li, data
+
+
+def _join_variety_desc(cust, ordr, li, dim_join):
+    cut = -1753
+    return {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut), "dim_join": dim_join}
+
+
+def _expect_groups(d, keep_orders_mask):
+    cut = -1753
+    lm = (d["ship"] > cut) & np.isin(d["li_keys"], d["o_keys"][keep_orders_mask])
+    keys, counts = np.unique(d["li_keys"][lm], return_counts=True)
+    return keys, counts
+
+
+def test_anti_join_lasj(ctx, orc):
+    """JOIN_LASJ (NOT EXISTS / anti) on the dim semijoin
+    (nodeHashjoin.c:652-659): mid rows pass when the fk is NOT in the dim
+    set; a NULL fk never matches, so it PASSES."""
+    rng = np.random.default_rng(71)
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng, with_fk_nulls=True)
+    got = ctx.q3_desc(_join_variety_desc(cust, ordr, li, "anti")).run().result()
+    cut = -1753
+    segok = d["c_keys"][d["c_seg"] == 0]
+    in_set = np.isin(d["o_cust"], segok) & ~d["fk_null"]
+    keep = (d["o_date"] < cut) & ~in_set     # NULL fk rows pass the anti join
+    keys, counts = _expect_groups(d, keep)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_anti_join_lasj_notin(ctx, orc):
+    """JOIN_LASJ_NOTIN (NOT IN): a NULL fk is REJECTED (NULL NOT IN (...)
+    is unknown), and any NULL dim key passing the dim filter empties the
+    whole result (nodeHashjoin.c:425,442)."""
+    rng = np.random.default_rng(72)
+    # case 1: NULL fks rejected, no dim NULLs
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng, with_fk_nulls=True)
+    got = ctx.q3_desc(
+        _join_variety_desc(cust, ordr, li, "anti_notin")).run().result()
+    cut = -1753
+    segok = d["c_keys"][d["c_seg"] == 0]
+    in_set = np.isin(d["o_cust"], segok)
+    keep = (d["o_date"] < cut) & ~in_set & ~d["fk_null"]
+    keys, counts = _expect_groups(d, keep)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    li.free(); ordr.free(); cust.free()
+
+    # case 2: a NULL dim key passing the filter -> EMPTY result
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng,
+                                             with_dim_nulls=True)
+    q = ctx.q3_desc(_join_variety_desc(cust, ordr, li, "anti_notin")).run()
+    got = q.result()
+    # segment-0 rows include NULL keys with probability ~1 at this size
+    has_null_in_filter = (d["dim_null"] & (d["c_seg"] == 0)).any()
+    assert has_null_in_filter, "fixture must include a filtered NULL dim key"
+    assert len(got["l_orderkey"]) == 0
+    li.free(); ordr.free(); cust.free()
+
+
+def test_semi_join_unchanged_by_dim_join_field(ctx, orc):
+    """dim_join='semi' equals the classic default plan."""
+    rng = np.random.default_rng(73)
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng)
+    classic = ctx.q3(cust, ordr, li).run().result()
+    semi = ctx.q3_desc(_join_variety_desc(cust, ordr, li, "semi")).run().result()
+    np.testing.assert_array_equal(semi["l_orderkey"], classic["l_orderkey"])
+    np.testing.assert_array_equal(semi["nitems"], classic["nitems"])
+    li.free(); ordr.free(); cust.free()
