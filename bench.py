@@ -274,7 +274,9 @@ def main():
                        "groups": st["groups"],
                        "stage_ms": {k: round(st[k], 3) for k in
                                     ("ms_cust_build", "ms_orders_build",
-                                     "ms_motion", "ms_probe_agg", "ms_extract")}},
+                                     "ms_motion", "ms_motion_counts",
+                                     "ms_motion_payload",
+                                     "ms_probe_agg", "ms_extract")}},
             "roofline": roofline,
             "cpu_baseline": cpu,
         }
