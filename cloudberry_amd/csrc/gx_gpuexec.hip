@@ -50,6 +50,9 @@ struct gx_ctx {
     int seg = 0;
     int nsegs = 1;
     hipStream_t stream = nullptr;
+    hipStream_t stream2 = nullptr;   /* Motion overlap (r2): the exchange
+                                        runs here concurrently with the
+                                        dim build on `stream` */
     ncclComm_t comm = nullptr;
     char err[512] = "";
 };
@@ -2430,6 +2433,8 @@ extern "C" gx_status gx_init(int device_id, int seg_id, int nsegs, gx_ctx **out)
     if (e2 != hipSuccess) { set_err(nullptr, "hipSetDevice: %s", hipGetErrorString(e2)); delete ctx; return GX_ERR_HIP; }
     e2 = hipStreamCreate(&ctx->stream);
     if (e2 != hipSuccess) { set_err(nullptr, "hipStreamCreate: %s", hipGetErrorString(e2)); delete ctx; return GX_ERR_HIP; }
+    e2 = hipStreamCreate(&ctx->stream2);
+    if (e2 != hipSuccess) { set_err(nullptr, "hipStreamCreate2: %s", hipGetErrorString(e2)); (void) hipStreamDestroy(ctx->stream); delete ctx; return GX_ERR_HIP; }
     *out = ctx;
     return GX_OK;
 }
@@ -2438,6 +2443,7 @@ extern "C" gx_status gx_shutdown(gx_ctx *ctx)
 {
     if (!ctx) return GX_OK;
     if (ctx->comm) ncclCommDestroy(ctx->comm);
+    if (ctx->stream2) (void) hipStreamDestroy(ctx->stream2);
     if (ctx->stream) (void) hipStreamDestroy(ctx->stream);
     delete ctx;
     return GX_OK;
@@ -4478,6 +4484,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         if (!ctx->comm) { set_err(ctx, "nsegs>1 but gx_comm_init not called%s", ""); return GX_ERR_STATE; }
         int n = ctx->nsegs;
+        /* the exchange runs on stream2, concurrent with the dim-set build
+         * still in flight on the primary stream (they only meet at the
+         * local semijoin below, via a cross-stream event wait); the branch
+         * ends in a host sync, so stage 3 on the primary stream is ordered */
+        hipStream_t s = ctx->stream2;
         evholder mev0, mev1;
         HIP_CHK(ctx, mev0.create()); HIP_CHK(ctx, mev1.create());
         HIP_CHK(ctx, hipEventRecord(mev0, s));
@@ -4572,7 +4583,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         RCCL_CHK(ctx, ncclGroupEnd());
         HIP_CHK(ctx, hipEventRecord(pev1, s));
 
-        /* Motion 2: probe local customer set, route qualifying by o_orderkey */
+        /* Motion 2: probe local customer set, route qualifying by o_orderkey
+         * — first wait (on-stream) for the dim-set build on the primary
+         * stream to finish */
+        HIP_CHK(ctx, hipStreamWaitEvent(s, ev[1], 0));
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         bool anti_join = D.dim_join != 0;
         auto launch_qhist = [&](auto *cs, auto anti) {
