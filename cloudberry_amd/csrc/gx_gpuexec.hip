@@ -916,6 +916,127 @@ __global__ void k_validity_or_hidden(const uint8_t *validity, int64_t nrows,
     }
 }
 
+/* Constant texteq evaluated DIRECTLY over varlena AO blocks, one compare
+ * per PHYSICAL datum (i.e. once per RLE run — what a dictionary-aware
+ * executor does), mask bytes fanned out over the run's rows.  Replaces
+ * the decode-whole-column-then-compare prepare path (r2: the SF100
+ * prepare cost was dominated by the payload materialization).  NULL
+ * datums fail the qual (three-valued texteq). */
+__global__ void k_texteq_mask_blocks(const uint8_t *stream,
+                                     const gx_blockref *dir, int64_t nblocks,
+                                     int64_t nrows, const uint8_t *lit,
+                                     int32_t lit_len, uint8_t *mask, int *err)
+{
+    int64_t t = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    for (int64_t b = t; b < nblocks; b += gridDim.x * (int64_t) blockDim.x)
+    {
+        const uint8_t *c = stream + dir[b].offset + 24;
+        int16_t version = ((const int16_t *) c)[0];
+        int16_t flags = ((const int16_t *) c)[1];
+        int32_t logical = dir[b].rows;
+        if ((version != 0 && version != 1 && version != 2) ||
+            dir[b].first_row + logical > nrows || (flags & 4))
+        { atomicOr(err, 1); continue; }
+        bool rle = false;
+        const uint8_t *nbmp = nullptr, *bmp = nullptr, *cnts = nullptr;
+        int32_t bmbits = 0, csize = 0, nullbits = 0, psize = 0;
+        const uint8_t *p0;
+        if (version == 0)
+        {
+            int16_t nd = ((const int16_t *) c)[2];
+            int32_t nullsz = ((const int32_t *) c)[2];
+            psize = ((const int32_t *) c)[3];
+            if (nd != logical) { atomicOr(err, 1); continue; }
+            nullbits = (flags & 1) ? logical : 0;
+            nbmp = c + 16;
+            p0 = c + 16 + nullsz;
+        }
+        else
+        {
+            int32_t hlogical = ((const int32_t *) c)[1];
+            psize = ((const int32_t *) c)[3];
+            if (hlogical != logical) { atomicOr(err, 1); continue; }
+            rle = (flags & 2) != 0;
+            const uint8_t *q = c + 16;
+            nullbits = (flags & 1) ? logical : 0;
+            if (rle)
+            {
+                int32_t norepeats = ((const int32_t *) q)[0];
+                bmbits = ((const int32_t *) q)[1];
+                csize = ((const int32_t *) q)[3];
+                if (flags & 1) nullbits = norepeats;
+                else if (norepeats != 0) { atomicOr(err, 1); continue; }
+                q += 16;
+            }
+            if (flags & 1) { nbmp = q; q += (nullbits + 7) >> 3; }
+            if (rle) { bmp = q; q += (bmbits + 7) >> 3; cnts = q; q += csize; }
+            int32_t hdr = (int32_t) (q - c);
+            p0 = c + ((hdr + 7) & ~7);
+        }
+        const uint8_t *p = p0, *pend = p0 + psize;
+        int64_t out = 0;
+        int32_t item = 0, coff = 0, npos = 0;
+        bool bad = false;
+        while (out < logical)
+        {
+            int64_t row = dir[b].first_row + out;
+            if (nbmp != nullptr && nullbits > 0)
+            {
+                if (npos >= nullbits) { bad = true; break; }
+                int nb = (nbmp[npos >> 3] >> (npos & 7)) & 1;
+                npos++;
+                if (nb)
+                {
+                    mask[row] = 0;          /* NULL fails the qual */
+                    out++;
+                    continue;
+                }
+            }
+            if (rle && item >= bmbits) { bad = true; break; }
+            if (p < pend && *p == 0)
+                p = p0 + (((p - p0) + 3) & ~(int64_t) 3);
+            if (p >= pend) { bad = true; break; }
+            int64_t len;
+            const uint8_t *data;
+            if (*p & 1)
+            {
+                len = (int64_t) (*p >> 1) - 1;
+                data = p + 1;
+                p += 1 + len;
+            }
+            else
+            {
+                uint32_t hdr;
+                memcpy(&hdr, p, 4);
+                len = (int64_t) (hdr >> 2) - 4;
+                data = p + 4;
+                p += 4 + len;
+            }
+            if (len < 0 || p > pend) { bad = true; break; }
+            /* ONE compare per physical datum (per run) */
+            uint8_t m = (uint8_t) d_texteq(data, len, lit, lit_len);
+            int64_t reps = 1;
+            if (rle && (bmp[item >> 3] & (1u << (item & 7))))
+            {
+                int32_t nn = (cnts[coff] >> 6) + 1;
+                uint32_t v = cnts[coff] & 0x3F;
+                for (int32_t i = 1; i < nn; i++) v = (v << 8) | cnts[coff + i];
+                coff += nn;
+                reps += v;
+            }
+            if (out + reps > logical) { bad = true; break; }
+            for (int64_t rr = 0; rr < reps; rr++)
+            {
+                mask[dir[b].first_row + out] = m;
+                out++;
+            }
+            item++;
+        }
+        if (bad || (rle && (coff != csize || item != bmbits)))
+            atomicOr(err, 1);
+    }
+}
+
 /* flat-mask variants of the customer scan (mask built by k_texteq_mask) */
 __global__ void k_cust_count_mask(const uint8_t *key_s, gx_colmeta key_m,
                                   const uint8_t *mask, const uint8_t *vmap,
@@ -3703,48 +3824,23 @@ static gx_status q3_build_text_mask(gx_ctx *ctx, gx_q3 *q)
     const gx_col &cm = q->cust->cols[q->desc.dim_filter.col];
     int64_t n = cm.m.nrows;
     hipStream_t s = ctx->stream;
-    devbuf dbase, doff, dpay, derrb;
-    HIP_CHK(ctx, dbase.alloc(std::max<int64_t>(cm.nblocks, 1) * 8));
-    HIP_CHK(ctx, doff.alloc((n + 1) * 8));
+    devbuf derrb;
     HIP_CHK(ctx, derrb.alloc(4));
     int *derr = derrb.as<int>();
     HIP_CHK(ctx, hipMemsetAsync(derr, 0, 4, s));
-    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, s,
-                       cm.dstream, cm.ddir, cm.nblocks, n,
-                       (const int64_t *) nullptr, (uint8_t *) nullptr,
-                       (int64_t *) nullptr, (uint8_t *) nullptr,
-                       dbase.as<int64_t>(), 0, derr);
-    std::vector<int64_t> bases(std::max<int64_t>(cm.nblocks, 1));
-    HIP_CHK(ctx, hipMemcpyAsync(bases.data(), dbase.p, cm.nblocks * 8,
-                                hipMemcpyDeviceToHost, s));
-    HIP_CHK(ctx, hipStreamSynchronize(s));
-    int64_t total = 0;
-    for (int64_t b = 0; b < cm.nblocks; b++)
-    {
-        int64_t v = bases[b];
-        bases[b] = total;
-        total += v;
-    }
-    HIP_CHK(ctx, dpay.alloc(std::max<int64_t>(total, 1)));
-    HIP_CHK(ctx, hipMemcpyAsync(dbase.p, bases.data(), cm.nblocks * 8,
-                                hipMemcpyHostToDevice, s));
-    HIP_CHK(ctx, hipMemsetAsync(doff.p, 0, 8, s));
-    hipLaunchKernelGGL(k_decode_varlena, dim3(GRID), dim3(64), 0, s,
-                       cm.dstream, cm.ddir, cm.nblocks, n,
-                       dbase.as<int64_t>(), dpay.as<uint8_t>(),
-                       doff.as<int64_t>(), (uint8_t *) nullptr,
-                       (int64_t *) nullptr, 1, derr);
     HIP_CHK(ctx, hipMalloc(&q->dmask, std::max<int64_t>(n, 1)));
-    hipLaunchKernelGGL(k_texteq_mask, dim3(GRID), dim3(TPB), 0, s,
-                       doff.as<int64_t>(), dpay.as<uint8_t>(), n,
-                       q->dtext, q->desc.dim_text_len, q->dmask);
+    /* single pass, one texteq per RLE run — no payload materialization
+     * (the r1 path decoded the whole column first: 510 ms at SF100) */
+    hipLaunchKernelGGL(k_texteq_mask_blocks, dim3(GRID), dim3(64), 0, s,
+                       cm.dstream, cm.ddir, cm.nblocks, n,
+                       q->dtext, q->desc.dim_text_len, q->dmask, derr);
     int herr = 0;
     HIP_CHK(ctx, hipMemcpyAsync(&herr, derr, 4, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
     HIP_CHK(ctx, hipGetLastError());
     if (herr)
     {
-        set_err(ctx, "malformed (or NULL-bearing) varlena dim column%s", "");
+        set_err(ctx, "malformed varlena dim column%s", "");
         return GX_ERR_INVALID;
     }
     return GX_OK;

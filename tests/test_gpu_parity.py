@@ -1448,3 +1448,46 @@ def test_semi_join_unchanged_by_dim_join_field(ctx, orc):
     np.testing.assert_array_equal(semi["l_orderkey"], classic["l_orderkey"])
     np.testing.assert_array_equal(semi["nitems"], classic["nitems"])
     li.free(); ordr.free(); cust.free()
+
+
+def test_texteq_null_bearing_varlena(ctx, orc):
+    """r2: the per-run texteq mask handles NULL-bearing varlena dim columns
+    — a NULL mktsegment fails the qual (three-valued texteq), it no longer
+    errors out."""
+    rng = np.random.default_rng(81)
+    nc = 500
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    segs = [b"BUILDING", b"AUTOMOBILE", b"MACHINERY"]
+    seg_idx = rng.integers(0, 3, nc)
+    nulls = rng.random(nc) < 0.2
+    strings = [segs[int(i)] for i in seg_idx]
+    cust = ctx.bind([
+        (orc.aocs_encode(c_keys), 8, nc, 0),
+        (orc.aocs_encode_varlena(strings, nulls=nulls), -1, nc, 1)])
+    no = 1000
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode(o_cust), 8, no),
+                     (orc.aocs_encode(np.full(no, -9999, np.int32)), 4, no),
+                     (orc.aocs_encode(np.zeros(no, np.int32)), 4, no)])
+    nl = 3000
+    li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 5.0)), 8, nl),
+                   (orc.aocs_encode(np.zeros(nl)), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 9999, np.int32)), 4, nl)])
+    got = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", "BUILDING"),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3,
+        "mid_filter": (2, "<", -1753),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", -1753)}).run().result()
+    segok = c_keys[(seg_idx == 0) & ~nulls]
+    om = np.isin(o_cust, segok)
+    keys, counts = np.unique(li_keys[np.isin(li_keys, o_keys[om])],
+                             return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    li.free(); ordr.free(); cust.free()
