@@ -53,14 +53,17 @@
 #include "optimizer/cost.h"
 #include "optimizer/pathnode.h"
 #include "optimizer/paths.h"
+#include "optimizer/optimizer.h"
 #include "optimizer/planner.h"
 #include "optimizer/tlist.h"
+#include "parser/parsetree.h"
 #include "utils/builtins.h"
 #include "utils/date.h"
 #include "utils/guc.h"
 #include "utils/lsyscache.h"
 #include "utils/rel.h"
 #include "utils/snapmgr.h"
+#include "utils/syscache.h"
 
 PG_MODULE_MAGIC;
 
@@ -361,8 +364,19 @@ gpuexec_match_q3(PlannerInfo *root, RelOptInfo *input_rel, GpuQ3PlanInfo *pi)
 
 		if (rte == NULL || rte->rtekind != RTE_RELATION)
 			return false;
-		if (get_rel_relam(rte->relid) != AO_COLUMN_TABLE_AM_OID)
-			return false;		/* AOCS only (pg_am.dat:43 "ao_column") */
+		{
+			/* AOCS only (pg_am.dat:43 "ao_column") */
+			HeapTuple	tp = SearchSysCache1(RELOID,
+											 ObjectIdGetDatum(rte->relid));
+			Oid			relam;
+
+			if (!HeapTupleIsValid(tp))
+				return false;
+			relam = ((Form_pg_class) GETSTRUCT(tp))->relam;
+			ReleaseSysCache(tp);
+			if (relam != AO_COLUMN_TABLE_AM_OID)
+				return false;
+		}
 		nrels++;
 	}
 	if (nrels != 3)
