@@ -2100,10 +2100,20 @@ d_wave_claim(unsigned long long *ctr, bool mine, int lane,
 }
 
 /* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
+/* Motion-1 histogram/emit carry an optional DESTINATION-AWARE bloom
+ * prefilter (bloom_all = all ranks' dim blooms, all-gathered): an order
+ * only ships when its custkey passes the bloom of the rank it routes to —
+ * the reference's runtime-filter pushdown (nodeRuntimeFilter.c) applied
+ * ACROSS the interconnect.  False positives only ship extra rows (the
+ * destination's exact set probe filters them); results stay bit-exact.
+ * Disabled (nullptr) for anti joins, where non-membership is the keep
+ * condition. */
 __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
                               const uint8_t *oc_s, gx_colmeta oc_m,
                               const uint8_t *vmap,
                               int oop, int32_t olit, int nsegs,
+                              const unsigned long long *bloom_all,
+                              uint64_t bwmask,
                               unsigned long long *hist)
 {
     int lane = threadIdx.x & 63;
@@ -2117,8 +2127,11 @@ __global__ void k_ord_m1_hist(const uint8_t *od_s, gx_colmeta od_m,
         if (i < od_m.nrows && !gx_vm_hidden(vmap, i) &&
             gx_cmp(oop, gx_col_get<int32_t>(od_s, od_m, i), olit))
         {
-            d = gx_route_i64(gx_col_get<int64_t>(oc_s, oc_m, i), nsegs);
-            keep = true;
+            uint64_t ck = (uint64_t) gx_col_get<int64_t>(oc_s, oc_m, i);
+            d = gx_route_i64((int64_t) ck, nsegs);
+            keep = bloom_all == nullptr ||
+                   d_bloom_test(bloom_all + (int64_t) d * (bwmask + 1),
+                                bwmask, ck);
         }
         for (int dd = 0; dd < nsegs; dd++)
         {
@@ -2135,6 +2148,8 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
                               const uint8_t *op_s, gx_colmeta op_m,
                               const uint8_t *vmap,
                               int oop, int32_t olit, int nsegs,
+                              const unsigned long long *bloom_all,
+                              uint64_t bwmask,
                               unsigned long long *cursors, /* pre-set to region starts */
                               gx_ord_row *out)
 {
@@ -2154,7 +2169,9 @@ __global__ void k_ord_m1_emit(const uint8_t *ok_s, gx_colmeta ok_m,
             {
                 oc = gx_col_get<int64_t>(oc_s, oc_m, i);
                 d = gx_route_i64(oc, nsegs);
-                keep = true;
+                keep = bloom_all == nullptr ||
+                       d_bloom_test(bloom_all + (int64_t) d * (bwmask + 1),
+                                    bwmask, (uint64_t) oc);
             }
         }
         for (int dd = 0; dd < nsegs; dd++)
@@ -2502,6 +2519,7 @@ struct gx_q3 {
     unsigned long long *dcount = nullptr, *dhits = nullptr, *dmin = nullptr;
     /* motion-path exchange state (nsegs>1), cached across steps */
     unsigned long long *m_hist = nullptr, *m_cur = nullptr;
+    unsigned long long *m_bloom_all = nullptr;   /* all ranks' dim blooms */
     unsigned long long *m_cnts_mine = nullptr, *m_cnts_all = nullptr;
     gx_ord_row *m_send1 = nullptr, *m_recv1 = nullptr;
     gx_qual_row *m_send2 = nullptr, *m_recv2 = nullptr;
@@ -4175,6 +4193,7 @@ static void q3_free_runstate(gx_q3 *q)
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
     fr(q->m_hist); fr(q->m_cur); fr(q->m_cnts_mine); fr(q->m_cnts_all);
+    fr(q->m_bloom_all);
     fr(q->m_send1); fr(q->m_recv1); fr(q->m_send2); fr(q->m_recv2);
     fr(q->dtext);
     fr(q->dmask);
@@ -4289,6 +4308,20 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     q->cmask = cslots - 1;
     uint64_t bwords = (uint64_t) pow2_at_least(
         std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
+    if (ctx->nsegs > 1 && ctx->comm)
+    {
+        /* the Motion-1 destination-bloom prefilter all-gathers every
+         * rank's bloom — sizes must agree: take the max over ranks */
+        devbuf bw;
+        HIP_CHK(ctx, bw.alloc(8));
+        unsigned long long hv = bwords;
+        HIP_CHK(ctx, hipMemcpyAsync(bw.p, &hv, 8, hipMemcpyHostToDevice, s));
+        RCCL_CHK(ctx, ncclAllReduce(bw.p, bw.p, 1, ncclUint64, ncclMax,
+                                    ctx->comm, s));
+        HIP_CHK(ctx, hipMemcpyAsync(&hv, bw.p, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        bwords = hv;
+    }
     HIP_CHK(ctx, hipMalloc(&q->bloom, bwords * 8));
     HIP_CHK(ctx, hipMemsetAsync(q->bloom, 0, bwords * 8, s));
     q->bwmask = bwords - 1;
@@ -4615,10 +4648,27 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
          * send layout (its own row) and every peer's counts (r2: the two
          * separate count syncs were ~half the fixed motion latency) */
         unsigned long long *dhist = q->m_hist;
+        /* destination-aware bloom prefilter (semi join only): all-gather
+         * every rank's dim bloom, then only ship orders whose fk passes
+         * the DESTINATION's bloom — the runtime filter pushed across the
+         * interconnect.  Requires the local dim build (primary stream). */
+        const unsigned long long *bloom_all = nullptr;
+        if (D.dim_join == 0)
+        {
+            uint64_t bwords = q->bwmask + 1;
+            if (!q->m_bloom_all)
+                HIP_CHK(ctx, hipMalloc(&q->m_bloom_all,
+                                       (uint64_t) n * bwords * 8));
+            HIP_CHK(ctx, hipStreamWaitEvent(s, ev[1], 0));
+            RCCL_CHK(ctx, ncclAllGather(q->bloom, q->m_bloom_all, bwords,
+                                        ncclUint64, ctx->comm, s));
+            bloom_all = q->m_bloom_all;
+        }
         HIP_CHK(ctx, hipMemsetAsync(dhist, 0, n * 8, s));
         hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                            od.dstream, od.m, oc.dstream, oc.m, ovm_eff,
-                           D.mid_filter.op, (int32_t) D.mid_filter.literal, n, dhist);
+                           D.mid_filter.op, (int32_t) D.mid_filter.literal, n,
+                           bloom_all, q->bwmask, dhist);
         unsigned long long *dcnts_all = q->m_cnts_all;
         double t_counts = 0;
         auto now_ms = []() {
@@ -4645,7 +4695,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                            ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                            op.dstream, op.m, ovm_eff, D.mid_filter.op,
-                           (int32_t) D.mid_filter.literal, n, dcur, send1);
+                           (int32_t) D.mid_filter.literal, n,
+                           bloom_all, q->bwmask, dcur, send1);
         std::vector<unsigned long long> rcv1(n), roff1(n + 1, 0);
         for (int r = 0; r < n; r++) rcv1[r] = cnts_all[(int64_t) r * n + ctx->seg];
         for (int r = 0; r < n; r++) roff1[r + 1] = roff1[r] + rcv1[r];
@@ -5188,7 +5239,8 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     HIP_CHK(ctx, hipMemsetAsync(dhist, 0, nsegs * 8, s));
     hipLaunchKernelGGL(k_ord_m1_hist, dim3(GRID), dim3(TPB), 0, s,
                        od.dstream, od.m, oc.dstream, oc.m,
-                       (const uint8_t *) nullptr, 0, cutoff, nsegs, dhist);
+                       (const uint8_t *) nullptr, 0, cutoff, nsegs,
+                       (const unsigned long long *) nullptr, 0, dhist);
     std::vector<unsigned long long> h(nsegs);
     HIP_CHK(ctx, hipMemcpyAsync(h.data(), dhist, nsegs * 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
@@ -5202,7 +5254,8 @@ extern "C" gx_status gx_test_motion1(gx_ctx *ctx, gx_table *orders,
     hipLaunchKernelGGL(k_ord_m1_emit, dim3(GRID), dim3(TPB), 0, s,
                        ok.dstream, ok.m, oc.dstream, oc.m, od.dstream, od.m,
                        op.dstream, op.m, (const uint8_t *) nullptr,
-                       0, cutoff, nsegs, dcur, dsend);
+                       0, cutoff, nsegs,
+                       (const unsigned long long *) nullptr, 0, dcur, dsend);
     HIP_CHK(ctx, hipMemcpyAsync(out_rows, dsend, total * sizeof(gx_ord_row),
                                 hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
