@@ -2600,6 +2600,8 @@ extern "C" gx_status gx_comm_unique_id(unsigned char uid[GX_UNIQUE_ID_BYTES])
 extern "C" gx_status gx_comm_init(gx_ctx *ctx, const unsigned char uid[GX_UNIQUE_ID_BYTES])
 {
     if (!ctx) return GX_ERR_INVALID;
+    if (ctx->comm) return GX_OK;   /* one communicator per process lifetime
+                                    * (mirrors gang reuse; gpuexec.h) */
     ncclUniqueId id;
     memcpy(&id, uid, GX_UNIQUE_ID_BYTES);
     RCCL_CHK(ctx, ncclCommInitRank(&ctx->comm, ctx->nsegs, id, ctx->seg));

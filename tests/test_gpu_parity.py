@@ -1491,3 +1491,78 @@ def test_texteq_null_bearing_varlena(ctx, orc):
     np.testing.assert_array_equal(got["l_orderkey"], keys)
     np.testing.assert_array_equal(got["nitems"], counts)
     li.free(); ordr.free(); cust.free()
+
+
+def test_anti_join_forced_motion(ctx, orc):
+    """Anti join through the FULL RCCL exchange branch (GX_FORCE_MOTION):
+    the destination-bloom prefilter must stay OFF for anti joins — results
+    equal the local anti run."""
+    import os
+    rng = np.random.default_rng(91)
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng, with_fk_nulls=True)
+    want = ctx.q3_desc(_join_variety_desc(cust, ordr, li, "anti")).run().result()
+    try:
+        ctx.comm_init(ctx.comm_unique_id())
+    except gx.GxError:
+        pass    # communicator may already exist from an earlier test
+    os.environ["GX_FORCE_MOTION"] = "1"
+    try:
+        got = ctx.q3_desc(
+            _join_variety_desc(cust, ordr, li, "anti")).run().result()
+    finally:
+        del os.environ["GX_FORCE_MOTION"]
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_groupby_rle_key_column(ctx, orc):
+    """gx_groupby over a Dense/RLE (no-null) key column — decoded at entry,
+    same groups as the plain-format bind."""
+    rng = np.random.default_rng(92)
+    n = 30000
+    keys = np.repeat(rng.integers(1, 900, 600).astype(np.int64), 50)
+    vals = rng.uniform(0, 4, n)
+    t_plain = ctx.bind([(orc.aocs_encode(keys), 8, n),
+                        (orc.aocs_encode(vals), 8, n)])
+    t_rle = ctx.bind([(orc.aocs_encode_rle(keys), 8, n, 1),
+                      (orc.aocs_encode(vals), 8, n)])
+    a = ctx.groupby(t_plain, 0, 1)
+    b = ctx.groupby(t_rle, 0, 1)
+    np.testing.assert_array_equal(a["key"], b["key"])
+    np.testing.assert_array_equal(a["count"], b["count"])
+    np.testing.assert_allclose(a["sum"], b["sum"], rtol=1e-9)
+    t_rle.free(); t_plain.free()
+
+
+def test_extra_quals_on_materialized_table(ctx, orc):
+    """Extra quals (on plain columns) compose with a NULL-bearing
+    materialized key column on the same table: the hidden mask carries
+    both the qual failures and the strict-NULL rejects."""
+    rng = np.random.default_rng(93)
+    nl = 4000
+    li_keys = rng.integers(1, 101, nl).astype(np.int64)
+    li_null = rng.random(nl) < 0.2
+    price = np.full(nl, 2.0); disc = np.zeros(nl)
+    ship = rng.integers(-3000, 3000, nl).astype(np.int32)
+    cust, ordr, li = _mini_q3_tables(
+        ctx, orc, li_keys, price, disc, ship)
+    li.free()
+    li = ctx.bind([(orc.aocs_encode_orig_nulls(li_keys, li_null), 8, nl, 1),
+                   (orc.aocs_encode(price), 8, nl),
+                   (orc.aocs_encode(disc), 8, nl),
+                   (orc.aocs_encode(ship), 4, nl)])
+    cut = 0
+    got = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", -2000),
+        "fact_extra": [(3, "<", 2000), (0, ">", 10)]}).run().result()
+    lm = (~li_null & (ship > -2000) & (ship < 2000) & (li_keys > 10))
+    keys, counts = np.unique(li_keys[lm], return_counts=True)
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    np.testing.assert_array_equal(got["nitems"], counts)
+    li.free(); ordr.free(); cust.free()
