@@ -4273,43 +4273,72 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     HIP_CHK(ctx, hipMemcpyAsync(&cmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&cmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
+    /* key-0 / NOTIN-NULL state.  At nsegs>1 the verdict must be GLOBAL and
+     * agreed by every rank BEFORE any early return — a lone rank bailing
+     * out of sizing while its peers continue into the collectives below
+     * would hang the job.  state: 0 ok, 1 empty (NOTIN NULL inner key,
+     * nodeHashjoin.c:442), 2 invalid (real key 0 vs the slot sentinel). */
+    int dim_state = 0;
     if (n_building > 0 && cmin == 0)
+        dim_state = (D.dim_join == 2 &&
+                     q3_col(q, q->cust, D.dim_key_col).has_null) ? 1 : 2;
+    q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
+    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
+    uint64_t bwords0 = (uint64_t) pow2_at_least(
+        std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
+    if (dim_state == 0 &&
+        hbm_budget_check(ctx, cslots * q->cset_width + bwords0 * 8,
+                         "dim semijoin set") != GX_OK)
+        dim_state = 3;                 /* over budget — agree collectively */
+    if (ctx->nsegs > 1 && ctx->comm)
     {
-        if (D.dim_join == 2 &&
-            q3_col(q, q->cust, D.dim_key_col).has_null)
-        {
-            /* NOT IN with a NULL key on the inner side: the whole join
-             * yields nothing (nodeHashjoin.c:442 hs_hashkeys_null) */
-            q->empty = 1;
-            q->rescap = 1;
-            HIP_CHK(ctx, hipMalloc(&q->r_okey, 8));
-            HIP_CHK(ctx, hipMalloc(&q->r_odate, 4));
-            HIP_CHK(ctx, hipMalloc(&q->r_oprio, 4));
-            HIP_CHK(ctx, hipMalloc(&q->r_rev, 8));
-            HIP_CHK(ctx, hipMalloc(&q->r_cnt, 8));
-            q->sized = true;
-            return GX_OK;
-        }
+        devbuf stb;
+        HIP_CHK(ctx, stb.alloc(8));
+        unsigned long long hv = (unsigned long long) dim_state;
+        HIP_CHK(ctx, hipMemcpyAsync(stb.p, &hv, 8, hipMemcpyHostToDevice, s));
+        RCCL_CHK(ctx, ncclAllReduce(stb.p, stb.p, 1, ncclUint64, ncclMax,
+                                    ctx->comm, s));
+        HIP_CHK(ctx, hipMemcpyAsync(&hv, stb.p, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        dim_state = (int) hv;
+    }
+    if (dim_state == 1)
+    {
+        /* NOT IN with a NULL key on the inner side anywhere: the whole
+         * join yields nothing (nodeHashjoin.c:442 hs_hashkeys_null) */
+        q->empty = 1;
+        q->rescap = 1;
+        HIP_CHK(ctx, hipMalloc(&q->r_okey, 8));
+        HIP_CHK(ctx, hipMalloc(&q->r_odate, 4));
+        HIP_CHK(ctx, hipMalloc(&q->r_oprio, 4));
+        HIP_CHK(ctx, hipMalloc(&q->r_rev, 8));
+        HIP_CHK(ctx, hipMalloc(&q->r_cnt, 8));
+        q->sized = true;
+        return GX_OK;
+    }
+    if (dim_state == 2)
+    {
         /* slot value 0 is the empty-slot sentinel (keys start at 1 in every
-         * PG sequence-keyed table); a real key 0 would be silently dropped */
+         * PG sequence-keyed table); a real key 0 would be silently dropped.
+         * All ranks agree on this verdict, so everyone errors together. */
         set_err(ctx, "dim join key 0 found: 0 is reserved as the empty-slot "
                      "sentinel (gpuexec.h gx_q3_prepare_desc)%s", "");
         return GX_ERR_INVALID;
     }
-    q->cset_width = (cmax < (1ULL << 32)) ? 4 : 8;
-    uint64_t cslots = (uint64_t) pow2_at_least((int64_t) n_building * 2);
+    if (dim_state == 3)
     {
-        uint64_t bw = (uint64_t) pow2_at_least(
-            std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
-        gx_status st = hbm_budget_check(ctx, cslots * q->cset_width + bw * 8,
+        /* re-derive the message on every rank (the verdict is global) */
+        gx_status st = hbm_budget_check(ctx,
+                                        cslots * q->cset_width + bwords0 * 8,
                                         "dim semijoin set");
         if (st != GX_OK) return st;
+        set_err(ctx, "dim semijoin set over HBM budget on a peer rank%s", "");
+        return GX_ERR_OOM;
     }
     HIP_CHK(ctx, hipMalloc(&q->cset, cslots * q->cset_width));
     HIP_CHK(ctx, hipMemsetAsync(q->cset, 0, cslots * q->cset_width, s));
     q->cmask = cslots - 1;
-    uint64_t bwords = (uint64_t) pow2_at_least(
-        std::max<int64_t>((int64_t) n_building * 8 / 64, 4096));
+    uint64_t bwords = bwords0;
     if (ctx->nsegs > 1 && ctx->comm)
     {
         /* the Motion-1 destination-bloom prefilter all-gathers every
