@@ -2099,6 +2099,33 @@ d_wave_claim(unsigned long long *ctr, bool mine, int lane,
     return m;
 }
 
+/* compact non-empty groupby slots into SoA outputs (wave-aggregated
+ * cursor claims — a per-slot shared cursor serializes, G12) */
+__global__ void k_kv_extract(const unsigned long long *tkey,
+                             const double *tsum,
+                             const unsigned long long *tcnt, uint64_t tslots,
+                             unsigned long long *okey, double *osum,
+                             unsigned long long *ocnt,
+                             unsigned long long *cursor)
+{
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (int64_t base = base0; base < (int64_t) tslots; base += stride)
+    {
+        int64_t i = base + lane;
+        bool mine = i < (int64_t) tslots && tkey[i] != 0ULL;
+        unsigned long long w;
+        unsigned long long m = d_wave_claim(cursor, mine, lane, &w);
+        if (m && mine)
+        {
+            okey[w] = tkey[i];
+            osum[w] = tsum[i];
+            ocnt[w] = tcnt[i];
+        }
+    }
+}
+
 /* filtered orders → per-destination histogram by route(o_custkey) (Motion 1) */
 /* Motion-1 histogram/emit carry an optional DESTINATION-AWARE bloom
  * prefilter (bloom_all = all ranks' dim blooms, all-gathered): an order
@@ -3739,14 +3766,25 @@ extern "C" gx_status gx_groupby(gx_ctx *ctx, const gx_table *t, int key_col,
                        tc.as<unsigned long long>(), tslots - 1,
                        nacc.as<double>(),
                        nacc.as<unsigned long long>() + 1, errb.as<int>());
+    /* device-side compaction: only the actual groups travel to the host
+     * (the slot table itself can be tens of GB at SF-scale inputs) */
+    int64_t cap = std::min<int64_t>((int64_t) tslots, std::max<int64_t>(n, 1));
+    devbuf ok_b, os_b, oc_b, cur_b;
+    HIP_CHK(ctx, ok_b.alloc((size_t) cap * 8));
+    HIP_CHK(ctx, os_b.alloc((size_t) cap * 8));
+    HIP_CHK(ctx, oc_b.alloc((size_t) cap * 8));
+    HIP_CHK(ctx, cur_b.alloc(8));
+    HIP_CHK(ctx, hipMemsetAsync(cur_b.p, 0, 8, s));
+    hipLaunchKernelGGL(k_kv_extract, dim3(GRID), dim3(TPB), 0, s,
+                       tk.as<unsigned long long>(), ts.as<double>(),
+                       tc.as<unsigned long long>(), tslots,
+                       ok_b.as<unsigned long long>(), os_b.as<double>(),
+                       oc_b.as<unsigned long long>(),
+                       cur_b.as<unsigned long long>());
     int herr = 0;
     double nsum = 0;
-    unsigned long long ncnt = 0;
-    std::vector<unsigned long long> hk(tslots), hc(tslots);
-    std::vector<double> hs(tslots);
-    HIP_CHK(ctx, hipMemcpyAsync(hk.data(), tk.p, tslots * 8, hipMemcpyDeviceToHost, s));
-    HIP_CHK(ctx, hipMemcpyAsync(hs.data(), ts.p, tslots * 8, hipMemcpyDeviceToHost, s));
-    HIP_CHK(ctx, hipMemcpyAsync(hc.data(), tc.p, tslots * 8, hipMemcpyDeviceToHost, s));
+    unsigned long long ncnt = 0, ng = 0;
+    HIP_CHK(ctx, hipMemcpyAsync(&ng, cur_b.p, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&nsum, nacc.p, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&ncnt, (char *) nacc.p + 8, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&herr, errb.p, 4, hipMemcpyDeviceToHost, s));
@@ -3754,17 +3792,28 @@ extern "C" gx_status gx_groupby(gx_ctx *ctx, const gx_table *t, int key_col,
     HIP_CHK(ctx, hipGetLastError());
     if (herr & 4)
     { set_err(ctx, "groupby: key INT64_MIN unsupported (biased sentinel)%s", ""); return GX_ERR_INVALID; }
+    if ((int64_t) ng > cap)
+    { set_err(ctx, "groupby: extract overflow%s", ""); return GX_ERR_INVALID; }
+    std::vector<unsigned long long> hk(ng), hc(ng);
+    std::vector<double> hs(ng);
+    if (ng)
+    {
+        HIP_CHK(ctx, hipMemcpyAsync(hk.data(), ok_b.p, ng * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipMemcpyAsync(hs.data(), os_b.p, ng * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipMemcpyAsync(hc.data(), oc_b.p, ng * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+    }
 
     std::vector<gx_kv_group> groups;
-    for (uint64_t i = 0; i < tslots; i++)
-        if (hk[i])
-        {
-            gx_kv_group g{};
-            g.key = (int64_t) (hk[i] ^ (1ULL << 63));
-            g.sum = hs[i];
-            g.count = (int64_t) hc[i];
-            groups.push_back(g);
-        }
+    groups.reserve((size_t) ng);
+    for (uint64_t i = 0; i < ng; i++)
+    {
+        gx_kv_group g{};
+        g.key = (int64_t) (hk[i] ^ (1ULL << 63));
+        g.sum = hs[i];
+        g.count = (int64_t) hc[i];
+        groups.push_back(g);
+    }
     std::sort(groups.begin(), groups.end(),
               [](const gx_kv_group &a, const gx_kv_group &b)
               { return a.key < b.key; });
