@@ -3746,7 +3746,10 @@ extern "C" gx_status gx_groupby(gx_ctx *ctx, const gx_table *t, int key_col,
     int64_t n = t->nrows;
     uint64_t tslots = (uint64_t) pow2_at_least(n * 2);
     {
-        gx_status bs = hbm_budget_check(ctx, tslots * 24 + 64, "groupby table");
+        uint64_t out_cap = (uint64_t) std::min<int64_t>((int64_t) tslots,
+                                                        std::max<int64_t>(n, 1));
+        gx_status bs = hbm_budget_check(ctx, tslots * 24 + out_cap * 24 + 64,
+                                        "groupby table");
         if (bs != GX_OK) return bs;
     }
     devbuf tk, ts, tc, nacc, errb;
