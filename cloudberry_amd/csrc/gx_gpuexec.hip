@@ -4002,7 +4002,7 @@ static gx_status q3_materialize_col(gx_ctx *ctx, gx_q3 *q, gx_table *t,
 /* fold an AND-ed extra-qual list into a device HIDDEN bitmap (combined with
  * the table's visimap); runs once at prepare — the per-step kernels then
  * read one bit per row through the existing visibility path */
-static gx_status q3_build_qualmask(gx_ctx *ctx, gx_table *t,
+static gx_status q3_build_qualmask(gx_ctx *ctx, gx_q3 *q, gx_table *t,
                                    const gx_filter *quals, int nq,
                                    uint8_t **out)
 {
@@ -4011,19 +4011,35 @@ static gx_status q3_build_qualmask(gx_ctx *ctx, gx_table *t,
     for (int i = 0; i < nq; i++)
     {
         int c = quals[i].col;
-        if (c < 0 || c >= (int) t->cols.size() || t->cols[c].format != 0 ||
-            quals[i].op < 0 || quals[i].op > 5 ||
-            (t->cols[c].m.width != 1 && t->cols[c].m.width != 4 &&
-             t->cols[c].m.width != 8))
+        if (c < 0 || c >= (int) t->cols.size())
+        {
+            set_err(ctx, "extra qual column out of range%s", "");
+            return GX_ERR_INVALID;
+        }
+        /* materialized (formerly format-1) role columns qualify too —
+         * materialization runs BEFORE the qual masks */
+        const gx_col &col = q3_col(q, t, c);
+        if (col.format != 0 || quals[i].op < 0 || quals[i].op > 5 ||
+            (col.m.width != 1 && col.m.width != 4 && col.m.width != 8))
         {
             set_err(ctx, "extra qual needs a fixed-width Orig column and "
                          "op in 0..5%s", "");
             return GX_ERR_INVALID;
         }
-        qa.s[i] = t->cols[c].dstream;
-        qa.m[i] = t->cols[c].m;
+        qa.s[i] = col.dstream;
+        qa.m[i] = col.m;
         qa.op[i] = quals[i].op;
         qa.lit[i] = quals[i].literal;
+    }
+    if (*out != nullptr)
+    {
+        /* a hidden mask already exists (NULL folds from materialization):
+         * AND the quals in, IN PLACE (seed == output is element-aliased) */
+        hipLaunchKernelGGL(k_qualmask, dim3(GRID), dim3(TPB), 0, ctx->stream,
+                           qa, *out, t->nrows, *out);
+        HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
+        HIP_CHK(ctx, hipGetLastError());
+        return GX_OK;
     }
     int64_t nbytes = (t->nrows + 7) >> 3;
     devbuf mb;
@@ -4168,18 +4184,10 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
         {orders, desc->mid_extra, desc->n_mid_extra, &q->qvm_mid},
         {lineitem, desc->fact_extra, desc->n_fact_extra, &q->qvm_fact},
     };
-    for (auto &e : ex)
-        if (e.nq > 0)
-        {
-            gx_status st = q3_build_qualmask(ctx, e.t, e.qs, e.nq, e.dst);
-            if (st != GX_OK)
-            {
-                gx_q3_free(q);
-                return st;
-            }
-        }
-    /* materialize format-1 key/filter role columns (strict-NULL reject) */
-    for (auto &e : ex)                  /* same table/mask trio */
+    /* materialize format-1 key/filter role columns FIRST (strict-NULL
+     * reject seeds the hidden masks; the qual masks then AND into them
+     * and may reference the materialized columns) */
+    for (auto &e : ex)                  /* table/mask trio */
         for (size_t ci = 0; ci < e.t->cols.size(); ci++)
         {
             const gx_col &c = e.t->cols[ci];
@@ -4205,6 +4213,17 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
                 fold = false;
             gx_status st = q3_materialize_col(ctx, q, e.t, (int) ci, e.dst,
                                               fold);
+            if (st != GX_OK)
+            {
+                gx_q3_free(q);
+                return st;
+            }
+        }
+    /* AND-ed extra qual lists → per-table hidden bitmaps */
+    for (auto &e : ex)
+        if (e.nq > 0)
+        {
+            gx_status st = q3_build_qualmask(ctx, q, e.t, e.qs, e.nq, e.dst);
             if (st != GX_OK)
             {
                 gx_q3_free(q);
