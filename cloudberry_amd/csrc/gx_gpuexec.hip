@@ -3144,12 +3144,18 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
         hipLaunchKernelGGL(k_emit_ord, dim3(blocks), dim3(TPB), 0, ctx->stream,
                            seed, nglobal, ncust, ctx->seg, ctx->nsegs, doffs, dok, doc, dod, dop);
         gx_col c0, c1, c2, c3;
+        /* free each flat right after its encode: halves the generator's
+         * transient footprint (hipFree synchronizes prior stream work) */
         st = encode_column_device(ctx, dok, 8, n, &c0);
+        (void) hipFree(dok);
         if (st == GX_OK) st = encode_column_device(ctx, doc, 8, n, &c1);
+        (void) hipFree(doc);
         if (st == GX_OK) st = encode_column_device(ctx, dod, 4, n, &c2);
+        (void) hipFree(dod);
         if (st == GX_OK) st = encode_column_device(ctx, dop, 4, n, &c3);
+        (void) hipFree(dop);
         HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-        (void) hipFree(dok); (void) hipFree(doc); (void) hipFree(dod); (void) hipFree(dop); (void) hipFree(doffs);
+        (void) hipFree(doffs);
         t->cols = {c0, c1, c2, c3};
         t->nrows = n;
     }
@@ -3175,13 +3181,17 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
                                dfl, dst2, dpr1, ddi1, dsh1);
             gx_col q0, q1c, q2, q3c, q4;
             st = encode_column_device(ctx, dfl, 1, n, &q0);
+            (void) hipFree(dfl);
             if (st == GX_OK) st = encode_column_device(ctx, dst2, 1, n, &q1c);
+            (void) hipFree(dst2);
             if (st == GX_OK) st = encode_column_device(ctx, dpr1, 8, n, &q2);
+            (void) hipFree(dpr1);
             if (st == GX_OK) st = encode_column_device(ctx, ddi1, 8, n, &q3c);
+            (void) hipFree(ddi1);
             if (st == GX_OK) st = encode_column_device(ctx, dsh1, 4, n, &q4);
+            (void) hipFree(dsh1);
             HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-            (void) hipFree(dfl); (void) hipFree(dst2); (void) hipFree(dpr1); (void) hipFree(ddi1);
-            (void) hipFree(dsh1); (void) hipFree(doffs); (void) hipFree(dcounts);
+            (void) hipFree(doffs); (void) hipFree(dcounts);
             if (st != GX_OK) { gx_table_free(t); return st; }
             t->cols = {q0, q1c, q2, q3c, q4};
             t->nrows = n;
@@ -3231,14 +3241,24 @@ extern "C" gx_status gx_tpch_gen(gx_ctx *ctx, gx_tpch_table which, double sf,
             HIP_CHK(ctx, hipMemcpyAsync(c0.ddir, dir.data(),
                                         dir.size() * sizeof(gx_blockref),
                                         hipMemcpyHostToDevice, ctx->stream));
+            /* stream_bytes/dir are block-local: drain the copies before
+             * they go out of scope */
+            HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
         }
         else
             st = encode_column_device(ctx, dlk, 8, n, &c0);
+        /* free each flat right after its encode: halves the generator's
+         * transient footprint — SF1000 (206 GB of tables) now fits one
+         * GPU's 288 GiB (hipFree synchronizes prior stream work) */
+        (void) hipFree(dlk);
         if (st == GX_OK) st = encode_column_device(ctx, dpr, 8, n, &c1);
+        (void) hipFree(dpr);
         if (st == GX_OK) st = encode_column_device(ctx, ddi, 8, n, &c2);
+        (void) hipFree(ddi);
         if (st == GX_OK) st = encode_column_device(ctx, dsh, 4, n, &c3);
+        (void) hipFree(dsh);
         HIP_CHK(ctx, hipStreamSynchronize(ctx->stream));
-        (void) hipFree(dlk); (void) hipFree(dpr); (void) hipFree(ddi); (void) hipFree(dsh); (void) hipFree(doffs);
+        (void) hipFree(doffs);
         t->cols = {c0, c1, c2, c3};
         t->nrows = n;
     }
