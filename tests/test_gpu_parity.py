@@ -1566,3 +1566,106 @@ def test_extra_quals_on_materialized_table(ctx, orc):
     np.testing.assert_array_equal(got["l_orderkey"], keys)
     np.testing.assert_array_equal(got["nitems"], counts)
     li.free(); ordr.free(); cust.free()
+
+
+def test_q3_desc_fuzz(ctx, orc):
+    """Randomized plan fuzz over the widened descriptor surface: random
+    comparison ops/literals, 0-3 extra quals per table, all three dim-join
+    types, optional NULL-bearing key columns and visimaps — every plan
+    checked against an independent numpy evaluation.  Seeded: failures
+    reproduce."""
+    rng = np.random.default_rng(1234)
+    OPS = ["<", ">", "==", "!=", "<=", ">="]
+
+    def np_cmp(v, op, lit):
+        return {"<": v < lit, ">": v > lit, "==": v == lit, "!=": v != lit,
+                "<=": v <= lit, ">=": v >= lit}[op]
+
+    for trial in range(12):
+        nc = int(rng.integers(50, 400))
+        no = int(rng.integers(200, 1500))
+        nl = int(rng.integers(500, 5000))
+        c_keys = np.arange(1, nc + 1, dtype=np.int64)
+        c_seg = rng.integers(0, 4, nc).astype(np.int8)
+        o_keys = np.arange(1, no + 1, dtype=np.int64)
+        o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+        o_date = rng.integers(-400, 400, no).astype(np.int32)
+        o_prio = rng.integers(0, 5, no).astype(np.int32)
+        li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+        price = rng.uniform(1, 50, nl)
+        disc = rng.integers(0, 11, nl) / 100.0
+        ship = rng.integers(-400, 400, nl).astype(np.int32)
+
+        with_li_nulls = bool(rng.random() < 0.4)
+        li_null = (rng.random(nl) < 0.15) if with_li_nulls else np.zeros(nl, bool)
+        with_vmap = bool(rng.random() < 0.3)
+        deleted = (rng.random(nl) < 0.1) if with_vmap else np.zeros(nl, bool)
+
+        dim_join = ["semi", "anti", "anti_notin"][int(rng.integers(0, 3))]
+
+        cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                         (orc.aocs_encode(c_seg), 1, nc)])
+        ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                         (orc.aocs_encode(o_cust), 8, no),
+                         (orc.aocs_encode(o_date), 4, no),
+                         (orc.aocs_encode(o_prio), 4, no)])
+        if with_li_nulls:
+            li_key_stream = (orc.aocs_encode_orig_nulls(li_keys, li_null), 8, nl, 1)
+        else:
+            li_key_stream = (orc.aocs_encode(li_keys), 8, nl)
+        li = ctx.bind([li_key_stream,
+                       (orc.aocs_encode(price), 8, nl),
+                       (orc.aocs_encode(disc), 8, nl),
+                       (orc.aocs_encode(ship), 4, nl)])
+        if with_vmap:
+            li.set_visimap(deleted)
+
+        dop = OPS[int(rng.integers(0, 6))]
+        dlit = int(rng.integers(0, 4))
+        mop = OPS[int(rng.integers(0, 6))]
+        mlit = int(rng.integers(-300, 300))
+        fop = OPS[int(rng.integers(0, 6))]
+        flit = int(rng.integers(-300, 300))
+        n_mid_x = int(rng.integers(0, 3))
+        n_fact_x = int(rng.integers(0, 3))
+        mid_x = [(3, OPS[int(rng.integers(0, 6))], int(rng.integers(0, 5)))
+                 for _ in range(n_mid_x)]
+        fact_x = [(3, OPS[int(rng.integers(0, 6))], int(rng.integers(-300, 300)))
+                  for _ in range(n_fact_x)]
+
+        got = ctx.q3_desc({
+            "dim": cust, "dim_key_col": 0, "dim_filter": (1, dop, dlit),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3,
+            "mid_filter": (2, mop, mlit),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, fop, flit),
+            "mid_extra": mid_x, "fact_extra": fact_x,
+            "dim_join": dim_join}).run().result()
+
+        # independent evaluation
+        segok = c_keys[np_cmp(c_seg, dop, dlit)]
+        in_set = np.isin(o_cust, segok)
+        om = np_cmp(o_date, mop, mlit)
+        for col, op, lit in mid_x:
+            om &= np_cmp(o_prio, op, lit)
+        if dim_join == "semi":
+            om &= in_set
+        else:           # anti / anti_notin: NOT-NULL fks, plain complement
+            om &= ~in_set
+        lm = np_cmp(ship, fop, flit) & ~li_null & ~deleted
+        for col, op, lit in fact_x:
+            lm &= np_cmp(ship, op, lit)
+        lm &= np.isin(li_keys, o_keys[om])
+        keys, counts = np.unique(li_keys[lm], return_counts=True)
+        np.testing.assert_array_equal(got["l_orderkey"], keys,
+                                      err_msg=f"trial {trial}")
+        np.testing.assert_array_equal(got["nitems"], counts,
+                                      err_msg=f"trial {trial}")
+        rev = {k: 0.0 for k in keys.tolist()}
+        for k, p, dsc in zip(li_keys[lm].tolist(), price[lm], disc[lm]):
+            rev[k] += p * (1.0 - dsc)
+        np.testing.assert_allclose(
+            got["revenue"], np.array([rev[k] for k in keys.tolist()]),
+            rtol=1e-6, err_msg=f"trial {trial}")
+        li.free(); ordr.free(); cust.free()
