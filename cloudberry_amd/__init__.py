@@ -34,7 +34,10 @@ class _Group(ctypes.Structure):
                 ("o_shippriority", ctypes.c_int32),
                 ("revenue", ctypes.c_double),
                 ("revenue_num", ctypes.c_int64),
-                ("nitems", ctypes.c_int64)]
+                ("nitems", ctypes.c_int64),
+                ("key_is_null", ctypes.c_uint8),
+                ("attrs_null", ctypes.c_uint8),
+                ("_pad", ctypes.c_uint8 * 6)]
 
 
 class _Stats(ctypes.Structure):
@@ -81,7 +84,8 @@ class _Q3Desc(ctypes.Structure):
                 ("n_dim_extra", ctypes.c_int32),
                 ("n_mid_extra", ctypes.c_int32),
                 ("n_fact_extra", ctypes.c_int32),
-                ("dim_join", ctypes.c_int32)]
+                ("dim_join", ctypes.c_int32),
+                ("fact_join", ctypes.c_int32)]
 
 
 class _ColDesc(ctypes.Structure):
@@ -318,7 +322,11 @@ class Context:
                "o_orderdate": np.array([gp[i].o_orderdate for i in range(n)], np.int32),
                "o_shippriority": np.array([gp[i].o_shippriority for i in range(n)], np.int32),
                "revenue": np.array([gp[i].revenue for i in range(n)], np.float64),
-               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
+               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64),
+               "key_is_null": np.array([gp[i].key_is_null for i in range(n)],
+                                       np.bool_),
+               "attrs_null": np.array([gp[i].attrs_null for i in range(n)],
+                                      np.bool_)}
         self._lib.gx_free(gp)
         return res
 
@@ -373,6 +381,8 @@ class Context:
             setattr(d, cnt, len(quals))
         d.dim_join = {"semi": 0, "anti": 1, "anti_notin": 2}[
             desc_dict.get("dim_join", "semi")]
+        d.fact_join = {"inner": 0, "left_outer": 1}[
+            desc_dict.get("fact_join", "inner")]
         q = ctypes.c_void_p()
         self._chk(self._lib.gx_q3_prepare_desc(self._h, ctypes.byref(d),
                                                ctypes.byref(q)))
@@ -528,7 +538,11 @@ class Q3:
                "o_shippriority": np.array([gp[i].o_shippriority for i in range(n)], np.int32),
                "revenue": np.array([gp[i].revenue for i in range(n)], np.float64),
                "revenue_num": np.array([gp[i].revenue_num for i in range(n)], np.int64),
-               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64)}
+               "nitems": np.array([gp[i].nitems for i in range(n)], np.int64),
+               "key_is_null": np.array([gp[i].key_is_null for i in range(n)],
+                                       np.bool_),
+               "attrs_null": np.array([gp[i].attrs_null for i in range(n)],
+                                      np.bool_)}
         self.ctx._lib.gx_free(gp)
         return res
 

@@ -1288,7 +1288,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * ~300M L3 probe round-trips).  Guard-free main region — the tail is a
  * separate scalar loop — and filtered lanes probe slot 0 (L1-resident)
  * instead of branching, so each load batch stays in one basic block. */
-template <int B, typename KT, bool VM = false>
+template <int B, typename KT, bool VM = false, bool OUTER = false>
 __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
                                  const uint8_t *di_s, gx_colmeta di_m,
@@ -1297,10 +1297,39 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const KT *tkey,
                                  double *trev, unsigned long long *tcnt,
                                  gx_slotmap smap,
-                                 unsigned long long *hits)
+                                 unsigned long long *hits,
+                                 /* LEFT OUTER (HJ_FILL_OUTER): unmatched
+                                  * fact rows aggregate into a second
+                                  * biased-key table; NULL fact keys
+                                  * (decoded 0) coalesce into one group */
+                                 unsigned long long *ukey = nullptr,
+                                 double *urev = nullptr,
+                                 unsigned long long *ucnt = nullptr,
+                                 uint64_t umask = 0)
 {
     uint64_t tmask = smap.mask;
     unsigned long long local_hits = 0;
+    auto umiss = [&](uint64_t k, int64_t i) {
+        if constexpr (OUTER)
+        {
+            uint64_t bk = k ^ (1ULL << 63);
+            uint64_t slot = gx_hmix64(bk) & umask;
+            while (true)
+            {
+                unsigned long long prev = atomicCAS(&ukey[slot], 0ULL, bk);
+                if (prev == 0ULL || prev == bk) break;
+                slot = (slot + 1) & umask;
+            }
+            double price = gx_col_get<double>(pr_s, pr_m, i);
+            double disc = gx_col_get<double>(di_s, di_m, i);
+            atomicAdd(&urev[slot], price * (1.0 - disc));
+            atomicAdd(&ucnt[slot], 1ULL);
+        }
+        else
+        {
+            (void) k; (void) i;
+        }
+    };
     auto resolve = [&](uint64_t k, uint64_t slot, KT v) -> uint64_t {
         /* first slot already loaded as v; walk on collision; the compare
          * ZERO-EXTENDS the stored key, so in u32 mode a probe key >= 2^32
@@ -1337,6 +1366,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
             uint64_t slot = smap.slot0(k);
             uint64_t r = resolve(k, slot, tkey[slot]);
             if (r != ~0ULL) hit(r, i);
+            else umiss(k, i);
         }
     }
     else if constexpr (B == -12 || B == -13)
@@ -1994,6 +2024,53 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
  * range, counts its keeps, claims an output region with ONE atomic, then
  * writes (cdna_hip_programming.md G12 — a single shared cursor serializes;
  * the first version lost 6 ms to ~500k same-address atomics). */
+__device__ __forceinline__ unsigned long long
+d_wave_claim(unsigned long long *ctr, bool mine, int lane,
+             unsigned long long *out_off)
+{
+    unsigned long long m = __ballot(mine);
+    if (m == 0) return 0;
+    int leader = __ffsll((long long) m) - 1;
+    unsigned long long wb = 0;
+    if (lane == leader)
+        wb = atomicAdd(ctr, (unsigned long long) __popcll(m));
+    wb = __shfl(wb, leader, 64);
+    *out_off = wb + __popcll(m & ((lane == 0) ? 0ULL
+                                              : (~0ULL >> (64 - lane))));
+    return m;
+}
+
+/* append LEFT-OUTER unmatched groups after the matched extract: same
+ * cursor, NULL mid attrs (attrs_null), biased key 2^63 = the NULL-key
+ * group (key_is_null).  flags bit0 = key_is_null, bit1 = attrs_null. */
+__global__ void k_extract_u(const unsigned long long *ukey, const double *urev,
+                            const unsigned long long *ucnt, uint64_t uslots,
+                            int64_t *okey, int32_t *odate, int32_t *oprio,
+                            double *rev, int64_t *cnt, uint8_t *flags,
+                            unsigned long long *cursor)
+{
+    int lane = threadIdx.x & 63;
+    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    for (int64_t base = base0; base < (int64_t) uslots; base += stride)
+    {
+        int64_t i = base + lane;
+        bool mine = i < (int64_t) uslots && ukey[i] != 0ULL;
+        unsigned long long w;
+        unsigned long long m = d_wave_claim(cursor, mine, lane, &w);
+        if (m && mine)
+        {
+            int64_t k = (int64_t) (ukey[i] ^ (1ULL << 63));
+            okey[w] = k;
+            odate[w] = 0;
+            oprio[w] = 0;
+            rev[w] = urev[i];
+            cnt[w] = (int64_t) ucnt[i];
+            flags[w] = (uint8_t) (k == 0 ? 3 : 2);
+        }
+    }
+}
+
 template <typename KT>
 __global__ void k_extract(const KT *tkey, const int32_t *tdate,
                           const int32_t *tprio, const double *trev,
@@ -2083,21 +2160,6 @@ static ncclResult_t gx_nccl_recv_chunked(void *buf, size_t bytes, int peer,
 /* wave-aggregated per-destination counter claims: ONE atomic per wave per
  * destination instead of one per row (a single shared cursor serializes —
  * cdna_hip_programming.md G12; measured 68M same-address atomics ≈ 2 s). */
-__device__ __forceinline__ unsigned long long
-d_wave_claim(unsigned long long *ctr, bool mine, int lane,
-             unsigned long long *out_off)
-{
-    unsigned long long m = __ballot(mine);
-    if (m == 0) return 0;
-    int leader = __ffsll((long long) m) - 1;
-    unsigned long long wb = 0;
-    if (lane == leader)
-        wb = atomicAdd(ctr, (unsigned long long) __popcll(m));
-    wb = __shfl(wb, leader, 64);
-    *out_off = wb + __popcll(m & ((lane == 0) ? 0ULL
-                                              : (~0ULL >> (64 - lane))));
-    return m;
-}
 
 /* compact non-empty groupby slots into SoA outputs (wave-aggregated
  * cursor claims — a per-slot shared cursor serializes, G12) */
@@ -2295,6 +2357,35 @@ __global__ void k_qual_emit(const gx_ord_row *rows, int64_t n,
 
 /* min/max key stats over received qualifying orders (sizes the motion
  * path's table layout exactly like the local path's counting pass) */
+/* min/max over an i64 column (sizing bound for the LEFT-OUTER unmatched
+ * table: distinct keys <= min(nrows, range)) */
+__global__ void k_col_minmax(const uint8_t *col_s, gx_colmeta m,
+                             unsigned long long *maxkey,
+                             unsigned long long *minkey)
+{
+    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
+    int64_t stride = gridDim.x * (int64_t) blockDim.x;
+    unsigned long long kmax = 0, kmin = ~0ULL;
+    for (; i < m.nrows; i += stride)
+    {
+        unsigned long long k = (unsigned long long) gx_col_get<int64_t>(col_s, m, i);
+        if (k > kmax) kmax = k;
+        if (k < kmin) kmin = k;
+    }
+    for (int o = 32; o; o >>= 1)
+    {
+        unsigned long long v = __shfl_down(kmax, o, 64);
+        if (v > kmax) kmax = v;
+        unsigned long long w = __shfl_down(kmin, o, 64);
+        if (w < kmin) kmin = w;
+    }
+    if ((threadIdx.x & 63) == 0)
+    {
+        atomicMax(maxkey, kmax);
+        atomicMin(minkey, kmin);
+    }
+}
+
 __global__ void k_rows_minmax(const gx_qual_row *rows, int64_t n,
                               unsigned long long *maxkey,
                               unsigned long long *minkey)
@@ -2567,6 +2658,11 @@ struct gx_q3 {
     int32_t *r_odate = nullptr, *r_oprio = nullptr;
     double *r_rev = nullptr;
     int64_t *r_cnt = nullptr;
+    uint8_t *r_flags = nullptr;      /* per-group key/attr NULL flags */
+    /* LEFT OUTER: unmatched-row aggregation table (biased keys) */
+    unsigned long long *ukey = nullptr, *ucnt_u = nullptr;
+    double *urev = nullptr;
+    uint64_t umask = 0;
     int numeric = 0;                 /* numeric(15,2) scaled-i64 measures */
     int empty = 0;                   /* LASJ_NOTIN with a NULL dim key:
                                         the whole result is empty
@@ -4156,7 +4252,8 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
             if (c.format == 0 || !is_used(t, (int) ci)) continue;
             if (dim_text && t == customer && (int) ci == desc->dim_filter.col)
                 continue;                              /* varlena texteq path */
-            if (t == lineitem && (int) ci == desc->fact_key_col && !c.has_null)
+            if (t == lineitem && (int) ci == desc->fact_key_col &&
+                !c.has_null && desc->fact_join == 0)
                 continue;                              /* fused RLE scan */
             if (!is_matable(t, (int) ci))
             {
@@ -4191,7 +4288,8 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
         }
     }
     /* extra AND-ed qual lists → per-table hidden bitmaps */
-    if (desc->dim_join < 0 || desc->dim_join > 2 ||
+    if (desc->fact_join < 0 || desc->fact_join > 1 ||
+        desc->dim_join < 0 || desc->dim_join > 2 ||
         desc->n_dim_extra < 0 || desc->n_dim_extra > GX_MAX_EXTRA_QUALS ||
         desc->n_mid_extra < 0 || desc->n_mid_extra > GX_MAX_EXTRA_QUALS ||
         desc->n_fact_extra < 0 || desc->n_fact_extra > GX_MAX_EXTRA_QUALS)
@@ -4215,7 +4313,7 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
                 !is_matable(e.t, (int) ci))
                 continue;
             if (e.t == lineitem && (int) ci == desc->fact_key_col &&
-                !c.has_null)
+                !c.has_null && desc->fact_join == 0)
                 continue;               /* fused RLE path */
             /* join-variety NULL rules (nodeHashjoin.c:425,442,652-659):
              * - LASJ (anti): a NULL mid fk never matches the dim set, so
@@ -4230,6 +4328,12 @@ extern "C" gx_status gx_q3_prepare_desc(gx_ctx *ctx, const gx_q3_desc *desc,
                 fold = false;
             if (desc->dim_join == 2 && e.t == customer &&
                 (int) ci == desc->dim_key_col)
+                fold = false;
+            /* LEFT OUTER keeps fact rows that fail the ON clause: a NULL
+             * fact key never matches, so the row EMITS with NULL attrs
+             * (decoded 0 -> the NULL-key group) — do not hide it */
+            if (desc->fact_join == 1 && e.t == lineitem &&
+                (int) ci == desc->fact_key_col)
                 fold = false;
             gx_status st = q3_materialize_col(ctx, q, e.t, (int) ci, e.dst,
                                               fold);
@@ -4284,6 +4388,8 @@ static void q3_free_runstate(gx_q3 *q)
     auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
     fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
+    fr(q->r_flags);
+    fr(q->ukey); fr(q->ucnt_u); fr(q->urev);
     fr(q->dcount); fr(q->dhits); fr(q->dmin);
     fr(q->m_hist); fr(q->m_cur); fr(q->m_cnts_mine); fr(q->m_cnts_all);
     fr(q->m_bloom_all);
@@ -4542,12 +4648,43 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
         HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
-        q->rescap = std::max<int64_t>(qual, 1);
+        int64_t ucap = 0;
+        if (D.fact_join == 1)
+        {
+            /* LEFT OUTER: bound the unmatched table by the fact key stats
+             * (distinct keys <= min(nrows, range); NULL-decoded zeros only
+             * widen the range, which stays a valid bound) */
+            const gx_col &lkc = q3_col(q, q->li, D.fact_key_col);
+            HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));
+            HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s));
+            hipLaunchKernelGGL(k_col_minmax, dim3(GRID), dim3(TPB), 0, s,
+                               lkc.dstream, lkc.m, q->dhits, q->dmin);
+            unsigned long long lmax = 0, lmin = 0;
+            HIP_CHK(ctx, hipMemcpyAsync(&lmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
+            HIP_CHK(ctx, hipMemcpyAsync(&lmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
+            HIP_CHK(ctx, hipStreamSynchronize(s));
+            uint64_t range = (q->li->nrows > 0 && lmax >= lmin)
+                                 ? lmax - lmin + 1 : 1;
+            uint64_t bound = std::min<uint64_t>((uint64_t) q->li->nrows,
+                                                range) + 1;
+            uint64_t uslots = (uint64_t) pow2_at_least((int64_t) bound * 2);
+            gx_status bs = hbm_budget_check(
+                ctx, uslots * 24 + ((uint64_t) qual + bound + 1) * 33,
+                "left-outer unmatched table");
+            if (bs != GX_OK) return bs;
+            HIP_CHK(ctx, hipMalloc(&q->ukey, uslots * 8));
+            HIP_CHK(ctx, hipMalloc(&q->urev, uslots * 8));
+            HIP_CHK(ctx, hipMalloc(&q->ucnt_u, uslots * 8));
+            q->umask = uslots - 1;
+            ucap = (int64_t) bound;
+        }
+        q->rescap = std::max<int64_t>(qual + ucap, 1);
         HIP_CHK(ctx, hipMalloc(&q->r_okey, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_odate, q->rescap * 4));
         HIP_CHK(ctx, hipMalloc(&q->r_oprio, q->rescap * 4));
         HIP_CHK(ctx, hipMalloc(&q->r_rev, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_cnt, q->rescap * 8));
+        HIP_CHK(ctx, hipMalloc(&q->r_flags, q->rescap));
     }
     HIP_CHK(ctx, hipStreamSynchronize(s));
     q->sized = true;
@@ -4571,6 +4708,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         gx_status st = q3_size_and_alloc(q);
         if (st != GX_OK) return st;
+    }
+    if (q->desc.fact_join == 1 &&
+        !(ctx->nsegs == 1 && env_int("GX_FORCE_MOTION", 0) == 0))
+    {
+        set_err(ctx, "left-outer fact join runs on the single-segment local "
+                     "path only (Motion-branch support is a next-round "
+                     "widening)%s", "");
+        return GX_ERR_INVALID;
     }
     if (q->empty)
     {
@@ -4643,6 +4788,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
+        if (D.fact_join == 1)
+        {
+            HIP_CHK(ctx, hipMemsetAsync(q->ukey, 0, (q->umask + 1) * 8, s));
+            HIP_CHK(ctx, hipMemsetAsync(q->urev, 0, (q->umask + 1) * 8, s));
+            HIP_CHK(ctx, hipMemsetAsync(q->ucnt_u, 0, (q->umask + 1) * 8, s));
+        }
         int ogrid = env_int("GX_ORDERS_GRID", 32768);  /* measured optimum */
         bool ochunk = env_int("GX_ORDERS_CHUNKED", 0) != 0;
         const uint8_t *ovm = ovm_eff;
@@ -5043,6 +5194,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     }
     else if (q->numeric)
     {
+        if (D.fact_join == 1)
+        { set_err(ctx, "numeric + left-outer combo not supported%s", ""); return GX_ERR_INVALID; }
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         const uint8_t *lvm = lvm_eff;
         auto launch_num = [&](auto *keys, auto vm) {
@@ -5074,9 +5227,10 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         const char *pv = getenv("GX_PROBE_VARIANT");
         int variant = pv ? atoi(pv) : 0;
-        if (variant != 0 && lvm_eff)
+        bool outer = D.fact_join == 1;
+        if (variant != 0 && (lvm_eff || outer))
         {
-            set_err(ctx, "visimap/extra quals require the default probe variant%s", "");
+            set_err(ctx, "visimap/extra quals/outer require the default probe variant%s", "");
             return GX_ERR_INVALID;
         }
         const char *pg = getenv("GX_PROBE_GRID");
@@ -5089,7 +5243,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
                                ls.dstream, ls.m, lvm, D.fact_filter.op,
                                (int32_t) D.fact_filter.literal, keys,
-                               q->trev, q->tcnt, q->smap, dhits);
+                               q->trev, q->tcnt, q->smap, dhits,
+                               q->ukey, q->urev, q->ucnt_u, q->umask);
         };
         if (q->key_width == 4)
         {
@@ -5098,8 +5253,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             {
                 default:
                 case 0:
-                    if (lvm) launch((k_li_probe_agg_t<1, unsigned int, true>), keys);
-                    else launch(k_li_probe_agg_t<1, unsigned int>, keys);
+                    if (outer && lvm)
+                        launch((k_li_probe_agg_t<1, unsigned int, true, true>), keys);
+                    else if (outer)
+                        launch((k_li_probe_agg_t<1, unsigned int, false, true>), keys);
+                    else if (lvm)
+                        launch((k_li_probe_agg_t<1, unsigned int, true>), keys);
+                    else
+                        launch(k_li_probe_agg_t<1, unsigned int>, keys);
                     break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
@@ -5124,8 +5285,14 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             {
                 default:
                 case 0:
-                    if (lvm) launch((k_li_probe_agg_t<1, unsigned long long, true>), keys);
-                    else launch(k_li_probe_agg_t<1, unsigned long long>, keys);
+                    if (outer && lvm)
+                        launch((k_li_probe_agg_t<1, unsigned long long, true, true>), keys);
+                    else if (outer)
+                        launch((k_li_probe_agg_t<1, unsigned long long, false, true>), keys);
+                    else if (lvm)
+                        launch((k_li_probe_agg_t<1, unsigned long long, true>), keys);
+                    else
+                        launch(k_li_probe_agg_t<1, unsigned long long>, keys);
                     break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
@@ -5148,6 +5315,8 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
 
     /* ---- stage 4: extract ---- */
     HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
+    if (q->r_flags)
+        HIP_CHK(ctx, hipMemsetAsync(q->r_flags, 0, q->rescap, s));
     int egrid = env_int("GX_EXTRACT_GRID", 32768);  /* measured optimum */
     if (q->key_width == 4)
         hipLaunchKernelGGL(k_extract<unsigned int>, dim3(egrid), dim3(TPB), 0, s,
@@ -5159,6 +5328,11 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                            (const unsigned long long *) q->tkey, q->tdate, q->tprio,
                            q->trev, q->tcnt, q->tmask + 1,
                            q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
+    if (q->desc.fact_join == 1)
+        hipLaunchKernelGGL(k_extract_u, dim3(egrid), dim3(TPB), 0, s,
+                           q->ukey, q->urev, q->ucnt_u, q->umask + 1,
+                           q->r_okey, q->r_odate, q->r_oprio, q->r_rev,
+                           q->r_cnt, q->r_flags, dcount);
     HIP_CHK(ctx, hipEventRecord(ev[4], s));
 
     unsigned long long ngroups = 0, hits = 0;
@@ -5212,6 +5386,7 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
     std::vector<int64_t> okey(n), cnt(n);
     std::vector<int32_t> odate(n), oprio(n);
     std::vector<double> rev(n);
+    std::vector<uint8_t> flags(n, 0);
     if (n)
     {
         HIP_CHK(ctx, hipMemcpy(okey.data(), q->r_okey, n * 8, hipMemcpyDeviceToHost));
@@ -5219,11 +5394,18 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
         HIP_CHK(ctx, hipMemcpy(oprio.data(), q->r_oprio, n * 4, hipMemcpyDeviceToHost));
         HIP_CHK(ctx, hipMemcpy(rev.data(), q->r_rev, n * 8, hipMemcpyDeviceToHost));
         HIP_CHK(ctx, hipMemcpy(cnt.data(), q->r_cnt, n * 8, hipMemcpyDeviceToHost));
+        if (q->r_flags)
+            HIP_CHK(ctx, hipMemcpy(flags.data(), q->r_flags, n,
+                                   hipMemcpyDeviceToHost));
     }
     std::vector<int64_t> idx(n);
     for (int64_t i = 0; i < n; i++) idx[i] = i;
     std::sort(idx.begin(), idx.end(),
-              [&](int64_t a, int64_t b) { return okey[a] < okey[b]; });
+              [&](int64_t a, int64_t b) {
+                  int an = flags[a] & 1, bn = flags[b] & 1;
+                  if (an != bn) return an < bn;   /* NULL-key group LAST */
+                  return okey[a] < okey[b];
+              });
     gx_q3_group *g = (gx_q3_group *) malloc(sizeof(gx_q3_group) * std::max<int64_t>(n, 1));
     for (int64_t i = 0; i < n; i++)
     {
@@ -5243,6 +5425,9 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
             g[i].revenue_num = 0;
         }
         g[i].nitems = cnt[idx[i]];
+        g[i].key_is_null = (uint8_t) (flags[idx[i]] & 1);
+        g[i].attrs_null = (uint8_t) ((flags[idx[i]] >> 1) & 1);
+        memset(g[i]._pad, 0, sizeof(g[i]._pad));
     }
     *out = g;
     *ngroups = n;

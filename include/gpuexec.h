@@ -178,6 +178,13 @@ typedef struct gx_q3_group {
     int64_t revenue_num;       /* numeric(15,2) mode: exact Σ price_c·(100−d),
                                   implied scale 1e-4 (0 in f64 mode) */
     int64_t nitems;
+    /* appended r2 for LEFT OUTER fact joins (fact_join=1): unmatched fact
+     * rows form groups with NULL mid attributes (attrs_null=1, date/prio
+     * 0); NULL fact keys form ONE group (key_is_null=1, NULLs-equal
+     * grouping) returned LAST.  Both 0 for inner joins. */
+    uint8_t key_is_null;
+    uint8_t attrs_null;
+    uint8_t _pad[6];
 } gx_q3_group;
 
 typedef struct gx_q3_stats {
@@ -255,6 +262,13 @@ typedef struct gx_q3_desc {
      *     dim key passing the dim filter empties the whole result
      *     (nodeHashjoin.c:425,442 hs_hashkeys_null) */
     int32_t dim_join;
+    /* fact-join type (the fact⋈mid join): 0 = inner (the Q3 shape);
+     * 1 = LEFT OUTER (nodeHashjoin.c HJ_FILL_OUTER): fact rows passing
+     *     their WHERE quals but matching no mid row form groups with NULL
+     *     mid attrs; NULL fact keys (never equal under the strict op)
+     *     also emit, all in ONE group (NULLs-equal grouping).  Requires a
+     *     plain or materialized fact key (no fused-RLE) and f64 measures. */
+    int32_t fact_join;
 } gx_q3_desc;
 
 /* Restrictions checked at sizing (first gx_q3_run):

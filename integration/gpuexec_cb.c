@@ -123,6 +123,7 @@ typedef struct gx_q3_desc
 	int32		n_mid_extra;
 	int32		n_fact_extra;
 	int32		dim_join;
+	int32		fact_join;
 }			gx_q3_desc;
 
 typedef struct gx_q3_group
@@ -133,6 +134,9 @@ typedef struct gx_q3_group
 	double		revenue;
 	int64		revenue_num;
 	int64		nitems;
+	uint8		key_is_null;
+	uint8		attrs_null;
+	uint8		pad_[6];
 }			gx_q3_group;
 
 typedef struct gx_q3_stats

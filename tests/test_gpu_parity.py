@@ -1669,3 +1669,106 @@ def test_q3_desc_fuzz(ctx, orc):
             got["revenue"], np.array([rev[k] for k in keys.tolist()]),
             rtol=1e-6, err_msg=f"trial {trial}")
         li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_fact_join(ctx, orc):
+    """LEFT OUTER fact⋈mid (HJ_FILL_OUTER): fact rows passing their WHERE
+    quals but matching no qualifying order form groups with NULL mid attrs;
+    NULL fact keys all land in ONE NULL-key group, returned last."""
+    rng = np.random.default_rng(101)
+    nc, no, nl = 200, 600, 4000
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    c_seg = (np.arange(nc) % 3).astype(np.int8)
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    o_date = rng.integers(-400, 400, no).astype(np.int32)
+    o_prio = rng.integers(0, 5, no).astype(np.int32)
+    # lineitem keys: some matching, some beyond the orders domain, some NULL
+    li_keys = rng.integers(1, no + 300, nl).astype(np.int64)
+    li_null = rng.random(nl) < 0.1
+    price = rng.uniform(1, 20, nl)
+    disc = rng.integers(0, 11, nl) / 100.0
+    ship = rng.integers(-400, 400, nl).astype(np.int32)
+    cut = 0
+    cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                     (orc.aocs_encode(c_seg), 1, nc)])
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode(o_cust), 8, no),
+                     (orc.aocs_encode(o_date), 4, no),
+                     (orc.aocs_encode(o_prio), 4, no)])
+    li = ctx.bind([(orc.aocs_encode_orig_nulls(li_keys, li_null), 8, nl, 1),
+                   (orc.aocs_encode(price), 8, nl),
+                   (orc.aocs_encode(disc), 8, nl),
+                   (orc.aocs_encode(ship), 4, nl)])
+    got = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", cut),
+        "fact_join": "left_outer"}).run().result()
+
+    segok = c_keys[c_seg == 0]
+    om = (o_date < cut) & np.isin(o_cust, segok)
+    qual_keys = o_keys[om]
+    lm = (ship > cut)                      # WHERE on the left table
+    matched = lm & ~li_null & np.isin(li_keys, qual_keys)
+    unmatched = lm & ~li_null & ~np.isin(li_keys, qual_keys)
+    nullkey = lm & li_null
+
+    mk, mc = np.unique(li_keys[matched], return_counts=True)
+    uk, uc = np.unique(li_keys[unmatched], return_counts=True)
+
+    m_got = got["l_orderkey"][~got["attrs_null"]]
+    np.testing.assert_array_equal(np.sort(m_got), mk)
+    u_mask = got["attrs_null"] & ~got["key_is_null"]
+    np.testing.assert_array_equal(np.sort(got["l_orderkey"][u_mask]), uk)
+    # counts per group
+    order = np.argsort(got["l_orderkey"][u_mask])
+    np.testing.assert_array_equal(got["nitems"][u_mask][order], uc)
+    assert (got["o_orderdate"][got["attrs_null"]] == 0).all()
+    # NULL-key group: one, last, counts all null-key rows
+    nk = got["key_is_null"]
+    if nullkey.any():
+        assert nk.sum() == 1 and nk[-1]
+        assert got["nitems"][nk][0] == int(nullkey.sum())
+        np.testing.assert_allclose(
+            got["revenue"][nk][0],
+            (price[nullkey] * (1 - disc[nullkey])).sum(), rtol=1e-9)
+    else:
+        assert nk.sum() == 0
+    # revenue parity on unmatched groups
+    rev = {k: 0.0 for k in uk.tolist()}
+    for k, p, dsc in zip(li_keys[unmatched].tolist(),
+                         price[unmatched], disc[unmatched]):
+        rev[k] += p * (1.0 - dsc)
+    np.testing.assert_allclose(
+        got["revenue"][u_mask][order],
+        np.array([rev[k] for k in uk.tolist()]), rtol=1e-6)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_inner_equivalence(ctx, orc):
+    """With every fact key matched and no NULLs, left-outer equals inner."""
+    sf = 0.02
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut)}
+    inner = ctx.q3_desc(base).run().result()
+    outer = ctx.q3_desc(dict(base, fact_join="left_outer")).run().result()
+    # inner groups = outer groups with attrs present
+    m = ~outer["attrs_null"]
+    np.testing.assert_array_equal(outer["l_orderkey"][m], inner["l_orderkey"])
+    np.testing.assert_array_equal(outer["o_orderdate"][m], inner["o_orderdate"])
+    np.testing.assert_array_equal(outer["nitems"][m], inner["nitems"])
+    np.testing.assert_allclose(outer["revenue"][m], inner["revenue"], rtol=1e-9)
+    # every unmatched group comes from a real lineitem key outside the
+    # qualifying set and carries NULL attrs
+    assert (outer["o_orderdate"][~m] == 0).all()
+    li.free(); ordr.free(); cust.free()
