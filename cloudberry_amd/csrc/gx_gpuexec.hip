@@ -2663,6 +2663,7 @@ struct gx_q3 {
     unsigned long long *ukey = nullptr, *ucnt_u = nullptr;
     double *urev = nullptr;
     uint64_t umask = 0;
+    int64_t u_cap = 0;               /* bound on unmatched groups */
     int numeric = 0;                 /* numeric(15,2) scaled-i64 measures */
     int empty = 0;                   /* LASJ_NOTIN with a NULL dim key:
                                         the whole result is empty
@@ -4573,6 +4574,37 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
     else
         launch_cbuild((unsigned long long *) q->cset);
 
+    if (D.fact_join == 1)
+    {
+        /* LEFT OUTER: bound the unmatched table by the fact key stats
+         * (distinct keys <= min(nrows, range); NULL-decoded zeros only
+         * widen the range, which stays a valid bound).  Sized here for
+         * BOTH the local and Motion paths — the unmatched side is always
+         * this rank's local fact shard. */
+        const gx_q3_desc &DD = q->desc;
+        const gx_col &lkc = q3_col(q, q->li, DD.fact_key_col);
+        HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));
+        HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s));
+        hipLaunchKernelGGL(k_col_minmax, dim3(GRID), dim3(TPB), 0, s,
+                           lkc.dstream, lkc.m, q->dhits, q->dmin);
+        unsigned long long lmax = 0, lmin = 0;
+        HIP_CHK(ctx, hipMemcpyAsync(&lmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipMemcpyAsync(&lmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHK(ctx, hipStreamSynchronize(s));
+        uint64_t range = (q->li->nrows > 0 && lmax >= lmin)
+                             ? lmax - lmin + 1 : 1;
+        uint64_t bound = std::min<uint64_t>((uint64_t) q->li->nrows, range) + 1;
+        uint64_t uslots = (uint64_t) pow2_at_least((int64_t) bound * 2);
+        gx_status bs = hbm_budget_check(
+            ctx, uslots * 24 + (bound + 1) * 33, "left-outer unmatched table");
+        if (bs != GX_OK) return bs;
+        HIP_CHK(ctx, hipMalloc(&q->ukey, uslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->urev, uslots * 8));
+        HIP_CHK(ctx, hipMalloc(&q->ucnt_u, uslots * 8));
+        q->umask = uslots - 1;
+        q->u_cap = (int64_t) bound;
+    }
+
     /* local qualifying-order count bounds the table for BOTH paths: at
      * nsegs>1 the table holds rows received for THIS segment; the global
      * qualifying count is conserved by the Motions, and each rank's received
@@ -4648,37 +4680,7 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
         HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
-        int64_t ucap = 0;
-        if (D.fact_join == 1)
-        {
-            /* LEFT OUTER: bound the unmatched table by the fact key stats
-             * (distinct keys <= min(nrows, range); NULL-decoded zeros only
-             * widen the range, which stays a valid bound) */
-            const gx_col &lkc = q3_col(q, q->li, D.fact_key_col);
-            HIP_CHK(ctx, hipMemsetAsync(q->dhits, 0, 8, s));
-            HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0xFF, 8, s));
-            hipLaunchKernelGGL(k_col_minmax, dim3(GRID), dim3(TPB), 0, s,
-                               lkc.dstream, lkc.m, q->dhits, q->dmin);
-            unsigned long long lmax = 0, lmin = 0;
-            HIP_CHK(ctx, hipMemcpyAsync(&lmax, q->dhits, 8, hipMemcpyDeviceToHost, s));
-            HIP_CHK(ctx, hipMemcpyAsync(&lmin, q->dmin, 8, hipMemcpyDeviceToHost, s));
-            HIP_CHK(ctx, hipStreamSynchronize(s));
-            uint64_t range = (q->li->nrows > 0 && lmax >= lmin)
-                                 ? lmax - lmin + 1 : 1;
-            uint64_t bound = std::min<uint64_t>((uint64_t) q->li->nrows,
-                                                range) + 1;
-            uint64_t uslots = (uint64_t) pow2_at_least((int64_t) bound * 2);
-            gx_status bs = hbm_budget_check(
-                ctx, uslots * 24 + ((uint64_t) qual + bound + 1) * 33,
-                "left-outer unmatched table");
-            if (bs != GX_OK) return bs;
-            HIP_CHK(ctx, hipMalloc(&q->ukey, uslots * 8));
-            HIP_CHK(ctx, hipMalloc(&q->urev, uslots * 8));
-            HIP_CHK(ctx, hipMalloc(&q->ucnt_u, uslots * 8));
-            q->umask = uslots - 1;
-            ucap = (int64_t) bound;
-        }
-        q->rescap = std::max<int64_t>(qual + ucap, 1);
+        q->rescap = std::max<int64_t>(qual + q->u_cap, 1);
         HIP_CHK(ctx, hipMalloc(&q->r_okey, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_odate, q->rescap * 4));
         HIP_CHK(ctx, hipMalloc(&q->r_oprio, q->rescap * 4));
@@ -4708,14 +4710,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     {
         gx_status st = q3_size_and_alloc(q);
         if (st != GX_OK) return st;
-    }
-    if (q->desc.fact_join == 1 &&
-        !(ctx->nsegs == 1 && env_int("GX_FORCE_MOTION", 0) == 0))
-    {
-        set_err(ctx, "left-outer fact join runs on the single-segment local "
-                     "path only (Motion-branch support is a next-round "
-                     "widening)%s", "");
-        return GX_ERR_INVALID;
     }
     if (q->empty)
     {
@@ -5103,28 +5097,36 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         /* motion path sizes from the exchanged counts each run; qual can
          * grow within the same pow2 table size, so rescap is checked too */
         if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width ||
-            qual > q->rescap)
+            qual + q->u_cap > q->rescap)
         {
             auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
             fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
+            fr(q->r_flags);
             q->key_width = want_kw;
             HIP_CHK(ctx, hipMalloc(&q->tkey, tslots * q->key_width));
             HIP_CHK(ctx, hipMalloc(&q->tdate, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
             HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
-            q->rescap = std::max<int64_t>(qual, 1);
+            q->rescap = std::max<int64_t>(qual + q->u_cap, 1);
             HIP_CHK(ctx, hipMalloc(&q->r_okey, q->rescap * 8));
             HIP_CHK(ctx, hipMalloc(&q->r_odate, q->rescap * 4));
             HIP_CHK(ctx, hipMalloc(&q->r_oprio, q->rescap * 4));
             HIP_CHK(ctx, hipMalloc(&q->r_rev, q->rescap * 8));
             HIP_CHK(ctx, hipMalloc(&q->r_cnt, q->rescap * 8));
+            HIP_CHK(ctx, hipMalloc(&q->r_flags, q->rescap));
             q->tmask = tslots - 1;
         }
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, (q->tmask + 1) * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, (q->tmask + 1) * 8, s));
+        if (D.fact_join == 1)
+        {
+            HIP_CHK(ctx, hipMemsetAsync(q->ukey, 0, (q->umask + 1) * 8, s));
+            HIP_CHK(ctx, hipMemsetAsync(q->urev, 0, (q->umask + 1) * 8, s));
+            HIP_CHK(ctx, hipMemsetAsync(q->ucnt_u, 0, (q->umask + 1) * 8, s));
+        }
         q->smap.mask = q->tmask;
         q->smap.kmin = (int64_t) kmin;
         q->smap.scale = -1.0;

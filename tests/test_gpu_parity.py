@@ -1772,3 +1772,34 @@ def test_left_outer_inner_equivalence(ctx, orc):
     # qualifying set and carries NULL attrs
     assert (outer["o_orderdate"][~m] == 0).all()
     li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_forced_motion(ctx, orc):
+    """LEFT OUTER through the FULL RCCL exchange branch equals the local
+    outer run (unmatched side is always the local fact shard)."""
+    import os
+    sf = 0.05
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut), "fact_join": "left_outer"}
+    want = ctx.q3_desc(base).run().result()
+    try:
+        ctx.comm_init(ctx.comm_unique_id())
+    except gx.GxError:
+        pass
+    os.environ["GX_FORCE_MOTION"] = "1"
+    try:
+        got = ctx.q3_desc(base).run().result()
+    finally:
+        del os.environ["GX_FORCE_MOTION"]
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_array_equal(got["attrs_null"], want["attrs_null"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
+    li.free(); ordr.free(); cust.free()
