@@ -1976,7 +1976,7 @@ __global__ void k_li_probe_agg_rle(const uint8_t *lk_s, const gx_blockref *dir,
  * the PG numeric SUM for these ranges.  The numerator lives in the trev
  * buffer (reinterpreted u64).  Overflow guard: numerators past 2^62 set the
  * error flag (impossible for sane groups; detects corrupt input). */
-template <typename KT, bool VM = false>
+template <typename KT, bool VM = false, bool OUTER = false>
 __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
                                    const uint8_t *pr_s, gx_colmeta pr_m,
                                    const uint8_t *di_s, gx_colmeta di_m,
@@ -1986,7 +1986,11 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
                                    unsigned long long *tnum,
                                    unsigned long long *tcnt,
                                    gx_slotmap smap,
-                                   unsigned long long *hits, int *err)
+                                   unsigned long long *hits, int *err,
+                                   unsigned long long *ukey = nullptr,
+                                   unsigned long long *unum = nullptr,
+                                   unsigned long long *ucnt = nullptr,
+                                   uint64_t umask = 0)
 {
     uint64_t tmask = smap.mask;
     unsigned long long local_hits = 0;
@@ -2007,7 +2011,28 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
             if ((uint64_t) v == k) { found = true; break; }  /* zext */
             slot = (slot + 1) & tmask;
         }
-        if (!found) continue;
+        if (!found)
+        {
+            if constexpr (OUTER)
+            {
+                uint64_t bk = k ^ (1ULL << 63);
+                uint64_t uslot = gx_hmix64(bk) & umask;
+                while (true)
+                {
+                    unsigned long long prev = atomicCAS(&ukey[uslot], 0ULL, bk);
+                    if (prev == 0ULL || prev == bk) break;
+                    uslot = (uslot + 1) & umask;
+                }
+                int64_t pc = gx_col_get<int64_t>(pr_s, pr_m, i);
+                int64_t dc = gx_col_get<int64_t>(di_s, di_m, i);
+                unsigned long long uadd =
+                    (unsigned long long) (pc * (100 - dc));
+                unsigned long long uold = atomicAdd(&unum[uslot], uadd);
+                if (uold + uadd > (1ULL << 62)) atomicOr(err, 4);
+                atomicAdd(&ucnt[uslot], 1ULL);
+            }
+            continue;
+        }
         int64_t price_c = gx_col_get<int64_t>(pr_s, pr_m, i);
         int64_t disc_c = gx_col_get<int64_t>(di_s, di_m, i);
         unsigned long long add = (unsigned long long) (price_c * (100 - disc_c));
@@ -5196,30 +5221,32 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     }
     else if (q->numeric)
     {
-        if (D.fact_join == 1)
-        { set_err(ctx, "numeric + left-outer combo not supported%s", ""); return GX_ERR_INVALID; }
         HIP_CHK(ctx, hipMemsetAsync(q->dmin, 0, 8, s));   /* borrowed err flag */
         const uint8_t *lvm = lvm_eff;
-        auto launch_num = [&](auto *keys, auto vm) {
+        auto launch_num = [&](auto *keys, auto vm, auto outer) {
             hipLaunchKernelGGL((k_li_probe_agg_num<std::decay_t<decltype(*keys)>,
-                                                   decltype(vm)::value>),
+                                                   decltype(vm)::value,
+                                                   decltype(outer)::value>),
                                dim3(GRID), dim3(TPB), 0, s,
                                lk.dstream, lk.m, lp.dstream, lp.m, ld.dstream, ld.m,
                                ls.dstream, ls.m, lvm, D.fact_filter.op,
                                (int32_t) D.fact_filter.literal, keys,
                                (unsigned long long *) q->trev, q->tcnt, q->smap,
-                               dhits, (int *) q->dmin);
+                               dhits, (int *) q->dmin,
+                               q->ukey, (unsigned long long *) q->urev,
+                               q->ucnt_u, q->umask);
+        };
+        auto dis_num = [&](auto *keys) {
+            bool outer = D.fact_join == 1;
+            if (lvm && outer) launch_num(keys, std::true_type{}, std::true_type{});
+            else if (lvm) launch_num(keys, std::true_type{}, std::false_type{});
+            else if (outer) launch_num(keys, std::false_type{}, std::true_type{});
+            else launch_num(keys, std::false_type{}, std::false_type{});
         };
         if (q->key_width == 4)
-        {
-            if (lvm) launch_num((const unsigned int *) q->tkey, std::true_type{});
-            else launch_num((const unsigned int *) q->tkey, std::false_type{});
-        }
+            dis_num((const unsigned int *) q->tkey);
         else
-        {
-            if (lvm) launch_num((const unsigned long long *) q->tkey, std::true_type{});
-            else launch_num((const unsigned long long *) q->tkey, std::false_type{});
-        }
+            dis_num((const unsigned long long *) q->tkey);
         int herr = 0;
         HIP_CHK(ctx, hipMemcpyAsync(&herr, q->dmin, 4, hipMemcpyDeviceToHost, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));

@@ -1803,3 +1803,30 @@ def test_left_outer_forced_motion(ctx, orc):
     np.testing.assert_array_equal(got["attrs_null"], want["attrs_null"])
     np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
     li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_numeric(ctx, orc):
+    """numeric(15,2) + LEFT OUTER: unmatched groups' revenue numerators are
+    exact integers too."""
+    sf = 0.02
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li_f = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    li_n = ctx.tpch_gen(gx.TPCH_LINEITEM_NUMERIC, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li_f, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut), "fact_join": "left_outer"}
+    f64 = ctx.q3_desc(base).run().result()
+    qn = ctx.q3_desc(dict(base, fact=li_n))
+    ctx._chk(ctx._lib.gx_q3_set_numeric(qn._q, 1))
+    num = qn.run().result()
+    np.testing.assert_array_equal(num["l_orderkey"], f64["l_orderkey"])
+    np.testing.assert_array_equal(num["nitems"], f64["nitems"])
+    np.testing.assert_array_equal(num["attrs_null"], f64["attrs_null"])
+    # exact integer numerators, scale 1e-4
+    np.testing.assert_allclose(num["revenue_num"] * 1e-4, f64["revenue"],
+                               rtol=1e-9)
+    li_n.free(); li_f.free(); ordr.free(); cust.free()
