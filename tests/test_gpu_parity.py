@@ -1830,3 +1830,81 @@ def test_left_outer_numeric(ctx, orc):
     np.testing.assert_allclose(num["revenue_num"] * 1e-4, f64["revenue"],
                                rtol=1e-9)
     li_n.free(); li_f.free(); ordr.free(); cust.free()
+
+
+def test_anti_notin_nulls_failing_filter_not_empty(ctx, orc):
+    """LASJ_NOTIN: a NULL dim key that FAILS the dim filter never reaches
+    the hash build, so it must NOT empty the result (nodeHashjoin.c:442
+    counts nulls seen during the build, i.e., after the scan quals)."""
+    rng = np.random.default_rng(111)
+    nc = 300
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    c_seg = (np.arange(nc) % 3).astype(np.int8)
+    # NULL keys only on rows whose segment != 0 (they fail the filter)
+    dim_null = (rng.random(nc) < 0.3) & (c_seg != 0)
+    assert dim_null.any()
+    cust = ctx.bind([(orc.aocs_encode_orig_nulls(c_keys, dim_null), 8, nc, 1),
+                     (orc.aocs_encode(c_seg), 1, nc)])
+    no, nl = 900, 2500
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode(o_cust), 8, no),
+                     (orc.aocs_encode(np.full(no, -9999, np.int32)), 4, no),
+                     (orc.aocs_encode(np.zeros(no, np.int32)), 4, no)])
+    li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 3.0)), 8, nl),
+                   (orc.aocs_encode(np.zeros(nl)), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 9999, np.int32)), 4, nl)])
+    got = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", 0),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", 0),
+        "dim_join": "anti_notin"}).run().result()
+    # NOT IN over the non-null segment-0 keys (none of which are NULL here)
+    segok = c_keys[(c_seg == 0) & ~dim_null]
+    om = ~np.isin(o_cust, segok)
+    keys = np.unique(li_keys[np.isin(li_keys, o_keys[om])])
+    assert len(got["l_orderkey"]) > 0          # NOT emptied
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    li.free(); ordr.free(); cust.free()
+
+
+def test_anti_lasj_with_null_dim_keys(ctx, orc):
+    """Plain LASJ with NULL dim keys: strict build-side reject means NULL
+    keys never enter the set; the anti join is simply the complement of
+    the non-null filtered keys (no emptiness rule for LASJ)."""
+    rng = np.random.default_rng(112)
+    nc = 250
+    c_keys = np.arange(1, nc + 1, dtype=np.int64)
+    c_seg = (np.arange(nc) % 2).astype(np.int8)
+    dim_null = rng.random(nc) < 0.25
+    cust = ctx.bind([(orc.aocs_encode_orig_nulls(c_keys, dim_null), 8, nc, 1),
+                     (orc.aocs_encode(c_seg), 1, nc)])
+    no, nl = 800, 2000
+    o_keys = np.arange(1, no + 1, dtype=np.int64)
+    o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+    ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                     (orc.aocs_encode(o_cust), 8, no),
+                     (orc.aocs_encode(np.full(no, -9999, np.int32)), 4, no),
+                     (orc.aocs_encode(np.zeros(no, np.int32)), 4, no)])
+    li_keys = rng.integers(1, no + 1, nl).astype(np.int64)
+    li = ctx.bind([(orc.aocs_encode(li_keys), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 3.0)), 8, nl),
+                   (orc.aocs_encode(np.zeros(nl)), 8, nl),
+                   (orc.aocs_encode(np.full(nl, 9999, np.int32)), 4, nl)])
+    got = ctx.q3_desc({
+        "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+        "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+        "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", 0),
+        "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+        "fact_filter": (3, ">", 0),
+        "dim_join": "anti"}).run().result()
+    segok = c_keys[(c_seg == 0) & ~dim_null]   # strict: nulls never build
+    om = ~np.isin(o_cust, segok)
+    keys = np.unique(li_keys[np.isin(li_keys, o_keys[om])])
+    np.testing.assert_array_equal(got["l_orderkey"], keys)
+    li.free(); ordr.free(); cust.free()
