@@ -5469,6 +5469,15 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
  * of nsegs×N rows is the reference's final Gather/Limit (trivial). */
 extern "C" gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *nout)
 {
+    if (q && q->desc.fact_join == 1)
+    {
+        /* ORDER BY revenue DESC, o_orderdate over groups with NULL dates
+         * needs NULLS-LAST tie handling the device kernel doesn't model;
+         * fetch gx_q3_result and sort host-side for outer plans */
+        set_err(q->ctx, "gx_q3_topn is inner-join only (outer groups carry "
+                        "NULL dates)%s", "");
+        return GX_ERR_INVALID;
+    }
     if (!q || !q->ran || topn <= 0 || topn > 10) return GX_ERR_STATE;
     gx_ctx *ctx = q->ctx;
     hipStream_t s = ctx->stream;
