@@ -121,3 +121,10 @@ def test_four_rank_gloo_motion_equals_global():
     """the exact count layout the 8-rank scale bench uses,
     replayed at ws=4 on CPU (pairwise exchanges, per-rank shard gen)."""
     _run_motion_pipeline(4)
+
+
+def test_eight_rank_gloo_motion_equals_global():
+    """The exact 8-rank topology of the round-end scale bench, replayed on
+    CPU: 8 hash-distributed shards, two Motions, unioned result equals the
+    global single-segment Q3."""
+    _run_motion_pipeline(8)
