@@ -22,7 +22,7 @@ def main():
     li = ctx.tpch_gen(gx.TPCH_LINEITEM, SF)
     results = {}
     base = None
-    for tf in (200, 100, 125, 150, 300, 400):
+    for tf in tuple(int(x) for x in os.environ.get("TF_LIST", "200,100,125,150,300,400").split(",")):
         os.environ["GX_TABLE_FACTOR_PCT"] = str(tf)
         q = ctx.q3(cust, ordr, li)     # fresh q: sizing rereads the env
         best = None
