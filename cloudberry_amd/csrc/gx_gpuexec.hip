@@ -2096,11 +2096,6 @@ __global__ void k_extract_u(const unsigned long long *ukey, const double *urev,
     }
 }
 
-/* Single-pass compaction with wave-aggregated cursor claims (one atomic
- * per 64 slots — r2: the r1 two-pass per-workgroup form read the table
- * twice, which dominates once the fill-factor sweep grows the table; the
- * r1 fear of cursor serialization is answered by d_wave_claim, not by a
- * second pass).  tcnt is only touched for occupied slots. */
 template <typename KT>
 __global__ void k_extract(const KT *tkey, const int32_t *tdate,
                           const int32_t *tprio, const double *trev,
@@ -2108,25 +2103,42 @@ __global__ void k_extract(const KT *tkey, const int32_t *tdate,
                           int64_t *okey, int32_t *odate, int32_t *oprio,
                           double *rev, int64_t *cnt, unsigned long long *cursor)
 {
-    int lane = threadIdx.x & 63;
-    int64_t base0 = blockIdx.x * (int64_t) blockDim.x + threadIdx.x - lane;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (int64_t base = base0; base < (int64_t) tslots; base += stride)
+    __shared__ unsigned int scan[256];
+    __shared__ unsigned long long sbase;
+    int64_t range = ((int64_t) tslots + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * range;
+    int64_t hi = min(lo + range, (int64_t) tslots);
+    if (lo >= hi) return;
+
+    /* pass 1: count my keeps (thread-strided over the block's range) */
+    unsigned int mine = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        if (tkey[i] != (KT) 0 && tcnt[i] != 0ULL) mine++;
+    scan[threadIdx.x] = mine;
+    __syncthreads();
+    /* exclusive scan of 256 per-thread counts (Hillis-Steele in LDS) */
+    for (int o = 1; o < 256; o <<= 1)
     {
-        int64_t i = base + lane;
-        unsigned long long c = 0;
-        if (i < (int64_t) tslots && tkey[i] != (KT) 0)
-            c = tcnt[i];
-        unsigned long long w;
-        unsigned long long m = d_wave_claim(cursor, c != 0ULL, lane, &w);
-        if (m && c != 0ULL)
-        {
-            okey[w] = (int64_t) tkey[i];
-            odate[w] = tdate[i];
-            oprio[w] = tprio[i];
-            rev[w] = trev[i];
-            cnt[w] = (int64_t) c;
-        }
+        unsigned int v = (threadIdx.x >= (unsigned) o) ? scan[threadIdx.x - o] : 0;
+        __syncthreads();
+        scan[threadIdx.x] += v;
+        __syncthreads();
+    }
+    if (threadIdx.x == blockDim.x - 1)
+        sbase = atomicAdd(cursor, (unsigned long long) scan[255]);
+    __syncthreads();
+    unsigned long long w = sbase + scan[threadIdx.x] - mine;
+
+    /* pass 2: write at my claimed positions */
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+    {
+        if (tkey[i] == (KT) 0 || tcnt[i] == 0ULL) continue;
+        okey[w] = (int64_t) tkey[i];
+        odate[w] = tdate[i];
+        oprio[w] = tprio[i];
+        rev[w] = trev[i];
+        cnt[w] = (int64_t) tcnt[i];
+        w++;
     }
 }
 
