@@ -4677,7 +4677,14 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
             return GX_ERR_INVALID;
         }
         q->key_width = (kmax < (1ULL << 32)) ? 4 : 8;
-        int tf = env_int("GX_TABLE_FACTOR_PCT", 200);
+        /* fill factor: slots = qual * tf/100, pow2-rounded.  r2 sweep
+         * (profiles/rocprof_r2_kernels.txt): tf=300 (2^26 slots at SF100,
+         * load factor 0.19) takes the probe 4.86 -> 4.43 ms and the whole
+         * step 8.21 -> 8.07 ms; tf>=600 gains the probe further but the
+         * two-pass extract's table scan eats it (a single-pass wave-claim
+         * extract re-measured the r1 cursor-serialization pathology and
+         * was reverted — clustered occupied slots make every wave claim). */
+        int tf = env_int("GX_TABLE_FACTOR_PCT", 300);
         uint64_t tslots = (uint64_t) pow2_at_least(qual * tf / 100 + 1);
         {
             gx_status st = hbm_budget_check(
