@@ -1217,14 +1217,6 @@ __global__ void k_orders_count(const uint8_t *ok_s, gx_colmeta ok_m,
     }
 }
 
-/* mark a table slot's 64-slot window dirty (extract skips clean windows) */
-__device__ __forceinline__ void d_mark_dirty(unsigned long long *dirty,
-                                             uint64_t slot)
-{
-    if (dirty)
-        atomicOr(&dirty[slot >> 12], 1ULL << ((slot >> 6) & 63));
-}
-
 /* orders local path: build the join/agg table keyed by o_orderkey.
  * (ExecHashTableInsert nodeHash.c:1886; o_orderkey unique → 1 entry/key;
  *  payload doubles as the agg group state, nodeAgg.c group = join row) */
@@ -1242,8 +1234,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
                                const KS *cset, uint64_t cmask,
                                const unsigned long long *bloom, uint64_t bwmask,
                                KT *tkey,
-                               int32_t *tdate, int32_t *tprio, gx_slotmap smap,
-                               unsigned long long *dirty)
+                               int32_t *tdate, int32_t *tprio, gx_slotmap smap)
 {
     int64_t i, stride, iend;
     if constexpr (CHUNKED)
@@ -1279,7 +1270,6 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
             {
                 tdate[slot] = od;
                 tprio[slot] = gx_col_get<int32_t>(op_s, op_m, i);
-                d_mark_dirty(dirty, slot);
                 break;
             }
             if (prev == (KT) k) break;   /* unique keys: no-op */
@@ -2106,45 +2096,24 @@ __global__ void k_extract_u(const unsigned long long *ukey, const double *urev,
     }
 }
 
-/* Two-pass per-workgroup compaction, r2: a DIRTY-WINDOW bitmap (one bit
- * per 64-slot window, set at insert) lets both passes skip clean windows
- * — at sparse fill factors most of the table is never read.  (A
- * single-pass wave-claim rewrite was measured and reverted: clustered
- * occupied slots make ~tslots/64 same-address cursor atomics — the r1
- * extract pathology.) */
 template <typename KT>
 __global__ void k_extract(const KT *tkey, const int32_t *tdate,
                           const int32_t *tprio, const double *trev,
                           const unsigned long long *tcnt, uint64_t tslots,
-                          const unsigned long long *dirty,
                           int64_t *okey, int32_t *odate, int32_t *oprio,
                           double *rev, int64_t *cnt, unsigned long long *cursor)
 {
     __shared__ unsigned int scan[256];
     __shared__ unsigned long long sbase;
     int64_t range = ((int64_t) tslots + gridDim.x - 1) / gridDim.x;
-    range = (range + 63) & ~63LL;             /* window-aligned ranges */
     int64_t lo = blockIdx.x * range;
     int64_t hi = min(lo + range, (int64_t) tslots);
     if (lo >= hi) return;
 
-    auto wdirty = [&](int64_t i) -> bool {
-        return dirty == nullptr ||
-               (dirty[(uint64_t) i >> 12] &
-                (1ULL << (((uint64_t) i >> 6) & 63)));
-    };
-
-    /* pass 1: count my keeps; threads own whole 64-slot windows so a
-     * clean window costs one L2-resident bitmap test */
+    /* pass 1: count my keeps (thread-strided over the block's range) */
     unsigned int mine = 0;
-    for (int64_t w0 = lo + (int64_t) threadIdx.x * 64; w0 < hi;
-         w0 += (int64_t) blockDim.x * 64)
-    {
-        if (!wdirty(w0)) continue;
-        int64_t wend = min(w0 + 64, hi);
-        for (int64_t i = w0; i < wend; i++)
-            if (tkey[i] != (KT) 0 && tcnt[i] != 0ULL) mine++;
-    }
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        if (tkey[i] != (KT) 0 && tcnt[i] != 0ULL) mine++;
     scan[threadIdx.x] = mine;
     __syncthreads();
     /* exclusive scan of 256 per-thread counts (Hillis-Steele in LDS) */
@@ -2160,22 +2129,16 @@ __global__ void k_extract(const KT *tkey, const int32_t *tdate,
     __syncthreads();
     unsigned long long w = sbase + scan[threadIdx.x] - mine;
 
-    /* pass 2: write at my claimed positions (same traversal order) */
-    for (int64_t w0 = lo + (int64_t) threadIdx.x * 64; w0 < hi;
-         w0 += (int64_t) blockDim.x * 64)
+    /* pass 2: write at my claimed positions */
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
     {
-        if (!wdirty(w0)) continue;
-        int64_t wend = min(w0 + 64, hi);
-        for (int64_t i = w0; i < wend; i++)
-        {
-            if (tkey[i] == (KT) 0 || tcnt[i] == 0ULL) continue;
-            okey[w] = (int64_t) tkey[i];
-            odate[w] = tdate[i];
-            oprio[w] = tprio[i];
-            rev[w] = trev[i];
-            cnt[w] = (int64_t) tcnt[i];
-            w++;
-        }
+        if (tkey[i] == (KT) 0 || tcnt[i] == 0ULL) continue;
+        okey[w] = (int64_t) tkey[i];
+        odate[w] = tdate[i];
+        oprio[w] = tprio[i];
+        rev[w] = trev[i];
+        cnt[w] = (int64_t) tcnt[i];
+        w++;
     }
 }
 
@@ -2484,8 +2447,7 @@ template <typename KT>
 __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
                                   const unsigned long long *dn,
                                   KT *tkey,
-                                  int32_t *tdate, int32_t *tprio, gx_slotmap smap,
-                                  unsigned long long *dirty)
+                                  int32_t *tdate, int32_t *tprio, gx_slotmap smap)
 {
     if (dn) n = min(n, (int64_t) *dn);
     int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
@@ -2502,7 +2464,6 @@ __global__ void k_build_from_rows(const gx_qual_row *rows, int64_t n,
             {
                 tdate[slot] = rows[i].odate;
                 tprio[slot] = rows[i].oprio;
-                d_mark_dirty(dirty, slot);
                 break;
             }
             if (prev == (KT) k) break;
@@ -2698,9 +2659,6 @@ struct gx_q3 {
     double *trev = nullptr;
     unsigned long long *tcnt = nullptr;
     uint64_t tmask = 0;
-    unsigned long long *tdirty = nullptr;  /* 1 bit per 64-slot window,
-                                              set at insert; extract skips
-                                              clean windows */
     unsigned long long *dcount = nullptr, *dhits = nullptr, *dmin = nullptr;
     /* motion-path exchange state (nsegs>1), cached across steps */
     unsigned long long *m_hist = nullptr, *m_cur = nullptr;
@@ -4455,7 +4413,6 @@ static void q3_free_runstate(gx_q3 *q)
 {
     auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
     fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
-    fr(q->tdirty);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->r_flags);
     fr(q->ukey); fr(q->ucnt_u); fr(q->urev);
@@ -4755,8 +4712,6 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
         HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
         HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
-        HIP_CHK(ctx, hipMalloc(&q->tdirty,
-                               std::max<uint64_t>(tslots >> 12, 1) * 8));
         q->rescap = std::max<int64_t>(qual + q->u_cap, 1);
         HIP_CHK(ctx, hipMalloc(&q->r_okey, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_odate, q->rescap * 4));
@@ -4859,8 +4814,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, tslots * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, tslots * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, tslots * 8, s));
-        HIP_CHK(ctx, hipMemsetAsync(q->tdirty, 0,
-                                    std::max<uint64_t>(tslots >> 12, 1) * 8, s));
         if (D.fact_join == 1)
         {
             HIP_CHK(ctx, hipMemsetAsync(q->ukey, 0, (q->umask + 1) * 8, s));
@@ -4883,8 +4836,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                    op.dstream, op.m, ovm, D.mid_filter.op,
                                    (int32_t) D.mid_filter.literal, cs, q->cmask,
                                    q->bloom, q->bwmask,
-                                   tk, q->tdate, q->tprio, q->smap,
-                                   q->tdirty);
+                                   tk, q->tdate, q->tprio, q->smap);
             };
             auto go = [&](auto ch, auto vm) {
                 if (anti_join) go2(ch, vm, std::true_type{});
@@ -4938,13 +4890,13 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                    dim3(GRID), dim3(TPB), 0, s,
                                    q->m_send2, q->rescap, q->dcount,
                                    (unsigned int *) q->tkey,
-                                   q->tdate, q->tprio, q->smap, q->tdirty);
+                                   q->tdate, q->tprio, q->smap);
             else
                 hipLaunchKernelGGL(k_build_from_rows<unsigned long long>,
                                    dim3(GRID), dim3(TPB), 0, s,
                                    q->m_send2, q->rescap, q->dcount,
                                    (unsigned long long *) q->tkey,
-                                   q->tdate, q->tprio, q->smap, q->tdirty);
+                                   q->tdate, q->tprio, q->smap);
         }
         else if (q->key_width == 4 && q->cset_width == 4)
             launch_build((unsigned int *) q->tkey, (const unsigned int *) q->cset);
@@ -5181,7 +5133,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         {
             auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
             fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
-            fr(q->tdirty);
             fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
             fr(q->r_flags);
             q->key_width = want_kw;
@@ -5190,8 +5141,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             HIP_CHK(ctx, hipMalloc(&q->tprio, tslots * 4));
             HIP_CHK(ctx, hipMalloc(&q->trev, tslots * 8));
             HIP_CHK(ctx, hipMalloc(&q->tcnt, tslots * 8));
-            HIP_CHK(ctx, hipMalloc(&q->tdirty,
-                                   std::max<uint64_t>(tslots >> 12, 1) * 8));
             q->rescap = std::max<int64_t>(qual + q->u_cap, 1);
             HIP_CHK(ctx, hipMalloc(&q->r_okey, q->rescap * 8));
             HIP_CHK(ctx, hipMalloc(&q->r_odate, q->rescap * 4));
@@ -5204,9 +5153,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * q->key_width, s));
         HIP_CHK(ctx, hipMemsetAsync(q->trev, 0, (q->tmask + 1) * 8, s));
         HIP_CHK(ctx, hipMemsetAsync(q->tcnt, 0, (q->tmask + 1) * 8, s));
-        HIP_CHK(ctx, hipMemsetAsync(q->tdirty, 0,
-                                    std::max<uint64_t>((q->tmask + 1) >> 12,
-                                                       1) * 8, s));
         if (D.fact_join == 1)
         {
             HIP_CHK(ctx, hipMemsetAsync(q->ukey, 0, (q->umask + 1) * 8, s));
@@ -5226,12 +5172,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             hipLaunchKernelGGL(k_build_from_rows<unsigned int>, dim3(GRID), dim3(TPB), 0, s,
                                recv2, qual, (const unsigned long long *) nullptr,
                                (unsigned int *) q->tkey,
-                               q->tdate, q->tprio, q->smap, q->tdirty);
+                               q->tdate, q->tprio, q->smap);
         else
             hipLaunchKernelGGL(k_build_from_rows<unsigned long long>, dim3(GRID), dim3(TPB), 0, s,
                                recv2, qual, (const unsigned long long *) nullptr,
                                (unsigned long long *) q->tkey,
-                               q->tdate, q->tprio, q->smap, q->tdirty);
+                               q->tdate, q->tprio, q->smap);
         HIP_CHK(ctx, hipEventRecord(mev1, s));
         HIP_CHK(ctx, hipStreamSynchronize(s));
         float mms = 0, p01 = 0, p23 = 0;
@@ -5411,12 +5357,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     if (q->key_width == 4)
         hipLaunchKernelGGL(k_extract<unsigned int>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned int *) q->tkey, q->tdate, q->tprio,
-                           q->trev, q->tcnt, q->tmask + 1, q->tdirty,
+                           q->trev, q->tcnt, q->tmask + 1,
                            q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
     else
         hipLaunchKernelGGL(k_extract<unsigned long long>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned long long *) q->tkey, q->tdate, q->tprio,
-                           q->trev, q->tcnt, q->tmask + 1, q->tdirty,
+                           q->trev, q->tcnt, q->tmask + 1,
                            q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, dcount);
     if (q->desc.fact_join == 1)
         hipLaunchKernelGGL(k_extract_u, dim3(egrid), dim3(TPB), 0, s,
@@ -5794,8 +5740,7 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
                        rows_b.as<gx_qual_row>(), n,
                        (const unsigned long long *) nullptr,
                        key_b.as<unsigned long long>(), date_b.as<int32_t>(),
-                       prio_b.as<int32_t>(), smap,
-                       (unsigned long long *) nullptr);
+                       prio_b.as<int32_t>(), smap);
     const gx_col &lk = lineitem->cols[0], &lp = lineitem->cols[1],
                  &ld = lineitem->cols[2], &ls = lineitem->cols[3];
     HIP_CHK(ctx, hit_b.alloc(8));
@@ -5819,7 +5764,6 @@ extern "C" gx_status gx_test_q3_from_qual(gx_ctx *ctx,
                        key_b.as<unsigned long long>(), date_b.as<int32_t>(),
                        prio_b.as<int32_t>(), rev_b.as<double>(),
                        cnt_b2.as<unsigned long long>(), tslots,
-                       (const unsigned long long *) nullptr,
                        r_okey.as<int64_t>(), r_odate.as<int32_t>(),
                        r_oprio.as<int32_t>(), r_rev.as<double>(),
                        r_cnt.as<int64_t>(), cur_b.as<unsigned long long>());
