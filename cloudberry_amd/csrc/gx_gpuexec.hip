@@ -785,27 +785,6 @@ __device__ __forceinline__ bool d_bloom_test(const unsigned long long *bloom,
     return (bloom[h & wmask] & m) == m;
 }
 
-/* extract groups with ≥1 matched lineitem into SoA result arrays.
- * Two-pass per-workgroup compaction: each workgroup owns a contiguous slot
- * range, counts its keeps, claims an output region with ONE atomic, then
- * writes (cdna_hip_programming.md G12 — a single shared cursor serializes;
- * the first version lost 6 ms to ~500k same-address atomics). */
-__device__ __forceinline__ unsigned long long
-d_wave_claim(unsigned long long *ctr, bool mine, int lane,
-             unsigned long long *out_off)
-{
-    unsigned long long m = __ballot(mine);
-    if (m == 0) return 0;
-    int leader = __ffsll((long long) m) - 1;
-    unsigned long long wb = 0;
-    if (lane == leader)
-        wb = atomicAdd(ctr, (unsigned long long) __popcll(m));
-    wb = __shfl(wb, leader, 64);
-    *out_off = wb + __popcll(m & ((lane == 0) ? 0ULL
-                                              : (~0ULL >> (64 - lane))));
-    return m;
-}
-
 /* wave-aggregated counter add: ONE atomic per 64-lane wave (G12) */
 __device__ __forceinline__ void gx_wave_count_add(unsigned long long *dst,
                                                   unsigned long long v)
@@ -1309,8 +1288,7 @@ __global__ void k_orders_build(const uint8_t *ok_s, gx_colmeta ok_m,
  * ~300M L3 probe round-trips).  Guard-free main region — the tail is a
  * separate scalar loop — and filtered lanes probe slot 0 (L1-resident)
  * instead of branching, so each load batch stays in one basic block. */
-template <int B, typename KT, bool VM = false, bool OUTER = false,
-          bool GLIST = false>
+template <int B, typename KT, bool VM = false, bool OUTER = false>
 __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  const uint8_t *pr_s, gx_colmeta pr_m,
                                  const uint8_t *di_s, gx_colmeta di_m,
@@ -1320,8 +1298,6 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
                                  double *trev, unsigned long long *tcnt,
                                  gx_slotmap smap,
                                  unsigned long long *hits,
-                                 uint32_t *glist = nullptr,
-                                 unsigned long long *gcur = nullptr,
                                  /* LEFT OUTER (HJ_FILL_OUTER): unmatched
                                   * fact rows aggregate into a second
                                   * biased-key table; NULL fact keys
@@ -1371,21 +1347,7 @@ __global__ void k_li_probe_agg_t(const uint8_t *lk_s, gx_colmeta lk_m,
         double price = gx_col_get<double>(pr_s, pr_m, i);
         double disc = gx_col_get<double>(di_s, di_m, i);
         atomicAdd(&trev[slot], price * (1.0 - disc));
-        if constexpr (GLIST)
-        {
-            /* first count transition claims a group-list entry: stage 4
-             * then GATHERS ngroups entries instead of scanning tslots.
-             * d_wave_claim is divergence-safe (ballot over active lanes). */
-            unsigned long long prev = atomicAdd(&tcnt[slot], 1ULL);
-            bool first = prev == 0ULL;
-            unsigned long long w;
-            unsigned long long m = d_wave_claim(gcur, first,
-                                                (int) (threadIdx.x & 63), &w);
-            if (m && first)
-                glist[w] = (uint32_t) slot;
-        }
-        else
-            atomicAdd(&tcnt[slot], 1ULL);
+        atomicAdd(&tcnt[slot], 1ULL);
         local_hits++;
     };
     if constexpr (B == 1)
@@ -2082,6 +2044,26 @@ __global__ void k_li_probe_agg_num(const uint8_t *lk_s, gx_colmeta lk_m,
     gx_wave_count_add(hits, local_hits);
 }
 
+/* extract groups with ≥1 matched lineitem into SoA result arrays.
+ * Two-pass per-workgroup compaction: each workgroup owns a contiguous slot
+ * range, counts its keeps, claims an output region with ONE atomic, then
+ * writes (cdna_hip_programming.md G12 — a single shared cursor serializes;
+ * the first version lost 6 ms to ~500k same-address atomics). */
+__device__ __forceinline__ unsigned long long
+d_wave_claim(unsigned long long *ctr, bool mine, int lane,
+             unsigned long long *out_off)
+{
+    unsigned long long m = __ballot(mine);
+    if (m == 0) return 0;
+    int leader = __ffsll((long long) m) - 1;
+    unsigned long long wb = 0;
+    if (lane == leader)
+        wb = atomicAdd(ctr, (unsigned long long) __popcll(m));
+    wb = __shfl(wb, leader, 64);
+    *out_off = wb + __popcll(m & ((lane == 0) ? 0ULL
+                                              : (~0ULL >> (64 - lane))));
+    return m;
+}
 
 /* append LEFT-OUTER unmatched groups after the matched extract: same
  * cursor, NULL mid attrs (attrs_null), biased key 2^63 = the NULL-key
@@ -2111,31 +2093,6 @@ __global__ void k_extract_u(const unsigned long long *ukey, const double *urev,
             cnt[w] = (int64_t) ucnt[i];
             flags[w] = (uint8_t) (k == 0 ? 3 : 2);
         }
-    }
-}
-
-/* gather the groups named by the probe-built slot list (GLIST path):
- * ngroups entries instead of a tslots-wide table scan */
-template <typename KT>
-__global__ void k_gather_groups(const uint32_t *glist,
-                                const unsigned long long *gn,
-                                const KT *tkey, const int32_t *tdate,
-                                const int32_t *tprio, const double *trev,
-                                const unsigned long long *tcnt,
-                                int64_t *okey, int32_t *odate, int32_t *oprio,
-                                double *rev, int64_t *cnt)
-{
-    int64_t n = (int64_t) *gn;
-    int64_t i = blockIdx.x * (int64_t) blockDim.x + threadIdx.x;
-    int64_t stride = gridDim.x * (int64_t) blockDim.x;
-    for (; i < n; i += stride)
-    {
-        uint32_t slot = glist[i];
-        okey[i] = (int64_t) tkey[slot];
-        odate[i] = tdate[slot];
-        oprio[i] = tprio[slot];
-        rev[i] = trev[slot];
-        cnt[i] = (int64_t) tcnt[slot];
     }
 }
 
@@ -2702,11 +2659,6 @@ struct gx_q3 {
     double *trev = nullptr;
     unsigned long long *tcnt = nullptr;
     uint64_t tmask = 0;
-    /* group-slot list: the probe claims an entry at each slot's FIRST
-     * count transition (atomicAdd(tcnt)==0), so extraction is a gather
-     * over ngroups entries instead of a scan over tslots */
-    uint32_t *glist = nullptr;
-    unsigned long long *gcur = nullptr;
     unsigned long long *dcount = nullptr, *dhits = nullptr, *dmin = nullptr;
     /* motion-path exchange state (nsegs>1), cached across steps */
     unsigned long long *m_hist = nullptr, *m_cur = nullptr;
@@ -4461,7 +4413,6 @@ static void q3_free_runstate(gx_q3 *q)
 {
     auto fr = [](auto *&p) { if (p) { (void) hipFree(p); p = nullptr; } };
     fr(q->cset); fr(q->bloom); fr(q->tkey); fr(q->tdate); fr(q->tprio); fr(q->trev); fr(q->tcnt);
-    fr(q->glist); fr(q->gcur);
     fr(q->r_okey); fr(q->r_odate); fr(q->r_oprio); fr(q->r_rev); fr(q->r_cnt);
     fr(q->r_flags);
     fr(q->ukey); fr(q->ucnt_u); fr(q->urev);
@@ -4768,8 +4719,6 @@ static gx_status q3_size_and_alloc(gx_q3 *q)
         HIP_CHK(ctx, hipMalloc(&q->r_rev, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_cnt, q->rescap * 8));
         HIP_CHK(ctx, hipMalloc(&q->r_flags, q->rescap));
-        HIP_CHK(ctx, hipMalloc(&q->glist, q->rescap * 4));
-        HIP_CHK(ctx, hipMalloc(&q->gcur, 8));
     }
     HIP_CHK(ctx, hipStreamSynchronize(s));
     q->sized = true;
@@ -5199,10 +5148,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             HIP_CHK(ctx, hipMalloc(&q->r_rev, q->rescap * 8));
             HIP_CHK(ctx, hipMalloc(&q->r_cnt, q->rescap * 8));
             HIP_CHK(ctx, hipMalloc(&q->r_flags, q->rescap));
-            if (q->glist) { (void) hipFree(q->glist); q->glist = nullptr; }
-            HIP_CHK(ctx, hipMalloc(&q->glist, q->rescap * 4));
-            if (!q->gcur)
-                HIP_CHK(ctx, hipMalloc(&q->gcur, 8));
             q->tmask = tslots - 1;
         }
         HIP_CHK(ctx, hipMemsetAsync(q->tkey, 0, (q->tmask + 1) * q->key_width, s));
@@ -5247,7 +5192,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     HIP_CHK(ctx, hipEventRecord(ev[2], s));
 
     /* ---- stage 3: lineitem scan+probe+agg (dominant kernel) ---- */
-    bool used_glist = false;
     unsigned long long *dhits = q->dhits;
     HIP_CHK(ctx, hipMemsetAsync(dhits, 0, 8, s));
     if (lk.format == 1)
@@ -5325,13 +5269,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             set_err(ctx, "visimap/extra quals/outer require the default probe variant%s", "");
             return GX_ERR_INVALID;
         }
-        /* group-slot list: stage 4 gathers ngroups entries instead of
-         * scanning the table (eligible on the default f64 inner path) */
-        used_glist = variant == 0 && !outer && q->glist != nullptr;
-        if (used_glist)
-            HIP_CHK(ctx, hipMemsetAsync(q->gcur, 0, 8, s));
-        /* probe variants (env experiments) bypass the group list and fall
-         * back to the table-scan extract below */
         const char *pg = getenv("GX_PROBE_GRID");
         int pgrid = pg ? atoi(pg) : GRID;
         const char *pt = getenv("GX_PROBE_TPB");
@@ -5343,7 +5280,6 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                                ls.dstream, ls.m, lvm, D.fact_filter.op,
                                (int32_t) D.fact_filter.literal, keys,
                                q->trev, q->tcnt, q->smap, dhits,
-                               q->glist, q->gcur,
                                q->ukey, q->urev, q->ucnt_u, q->umask);
         };
         if (q->key_width == 4)
@@ -5358,9 +5294,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                     else if (outer)
                         launch((k_li_probe_agg_t<1, unsigned int, false, true>), keys);
                     else if (lvm)
-                        launch((k_li_probe_agg_t<1, unsigned int, true, false, true>), keys);
+                        launch((k_li_probe_agg_t<1, unsigned int, true>), keys);
                     else
-                        launch((k_li_probe_agg_t<1, unsigned int, false, false, true>), keys);
+                        launch(k_li_probe_agg_t<1, unsigned int>, keys);
                     break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned int>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned int>, keys); break;
@@ -5390,9 +5326,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
                     else if (outer)
                         launch((k_li_probe_agg_t<1, unsigned long long, false, true>), keys);
                     else if (lvm)
-                        launch((k_li_probe_agg_t<1, unsigned long long, true, false, true>), keys);
+                        launch((k_li_probe_agg_t<1, unsigned long long, true>), keys);
                     else
-                        launch((k_li_probe_agg_t<1, unsigned long long, false, false, true>), keys);
+                        launch(k_li_probe_agg_t<1, unsigned long long>, keys);
                     break;
                 case 2: launch(k_li_probe_agg_t<4, unsigned long long>, keys); break;
                 case 4: launch(k_li_probe_agg_t<2, unsigned long long>, keys); break;
@@ -5413,31 +5349,12 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     }
     HIP_CHK(ctx, hipEventRecord(ev[3], s));
 
-    /* ---- stage 4: extract (table scan) or gather (probe-built list) ---- */
+    /* ---- stage 4: extract ---- */
     HIP_CHK(ctx, hipMemsetAsync(dcount, 0, 8, s));
     if (q->r_flags)
         HIP_CHK(ctx, hipMemsetAsync(q->r_flags, 0, q->rescap, s));
     int egrid = env_int("GX_EXTRACT_GRID", 32768);  /* measured optimum */
-    if (used_glist)
-    {
-        if (q->key_width == 4)
-            hipLaunchKernelGGL(k_gather_groups<unsigned int>, dim3(GRID),
-                               dim3(TPB), 0, s,
-                               q->glist, q->gcur,
-                               (const unsigned int *) q->tkey, q->tdate,
-                               q->tprio, q->trev, q->tcnt,
-                               q->r_okey, q->r_odate, q->r_oprio,
-                               q->r_rev, q->r_cnt);
-        else
-            hipLaunchKernelGGL(k_gather_groups<unsigned long long>, dim3(GRID),
-                               dim3(TPB), 0, s,
-                               q->glist, q->gcur,
-                               (const unsigned long long *) q->tkey, q->tdate,
-                               q->tprio, q->trev, q->tcnt,
-                               q->r_okey, q->r_odate, q->r_oprio,
-                               q->r_rev, q->r_cnt);
-    }
-    else if (q->key_width == 4)
+    if (q->key_width == 4)
         hipLaunchKernelGGL(k_extract<unsigned int>, dim3(egrid), dim3(TPB), 0, s,
                            (const unsigned int *) q->tkey, q->tdate, q->tprio,
                            q->trev, q->tcnt, q->tmask + 1,
@@ -5455,8 +5372,7 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
     HIP_CHK(ctx, hipEventRecord(ev[4], s));
 
     unsigned long long ngroups = 0, hits = 0;
-    HIP_CHK(ctx, hipMemcpyAsync(&ngroups, used_glist ? q->gcur : dcount, 8,
-                                hipMemcpyDeviceToHost, s));
+    HIP_CHK(ctx, hipMemcpyAsync(&ngroups, dcount, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(&hits, dhits, 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipStreamSynchronize(s));
     HIP_CHK(ctx, hipGetLastError());
