@@ -1908,3 +1908,32 @@ def test_anti_lasj_with_null_dim_keys(ctx, orc):
     keys = np.unique(li_keys[np.isin(li_keys, o_keys[om])])
     np.testing.assert_array_equal(got["l_orderkey"], keys)
     li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_sf100_conservation(ctx, orc):
+    """Full-size LEFT OUTER: at SF100 every shipdate-passing lineitem row
+    lands in exactly ONE group (matched or NULL-attr unmatched), so
+    sum(nitems) equals the standalone filter count; matched items equal
+    the inner join's probe hits."""
+    sf = 100.0
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    cut = gx.CUTOFF_19950315
+    nfilt, _ = li.scan_filter(3, ">", cut)
+    qi = ctx.q3(cust, ordr, li).run()
+    inner_hits = qi.stats()["probe_hits"]
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut), "fact_join": "left_outer"}
+    q = ctx.q3_desc(base).run()
+    r = q.result()
+    assert int(r["nitems"].sum()) == int(nfilt)
+    m = ~r["attrs_null"]
+    assert int(r["nitems"][m].sum()) == int(inner_hits)
+    assert not r["key_is_null"].any()         # synthetic keys are NOT NULL
+    # group keys unique across matched+unmatched
+    assert len(np.unique(r["l_orderkey"])) == len(r["l_orderkey"])
+    q.free(); li.free(); ordr.free(); cust.free()
