@@ -556,7 +556,11 @@ class Q3:
                 "o_shippriority": np.array([out[i].o_shippriority for i in range(m)], np.int32),
                 "revenue": np.array([out[i].revenue for i in range(m)], np.float64),
                 "revenue_num": np.array([out[i].revenue_num for i in range(m)], np.int64),
-                "nitems": np.array([out[i].nitems for i in range(m)], np.int64)}
+                "nitems": np.array([out[i].nitems for i in range(m)], np.int64),
+                "key_is_null": np.array([out[i].key_is_null for i in range(m)],
+                                        np.bool_),
+                "attrs_null": np.array([out[i].attrs_null for i in range(m)],
+                                       np.bool_)}
 
     def free(self):
         if self._q:

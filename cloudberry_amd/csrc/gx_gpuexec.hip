@@ -2554,9 +2554,9 @@ __device__ __forceinline__ bool topn_less(double ra, int32_t da,
 __global__ void __launch_bounds__(64)
 k_topn(const int64_t *okey, const int32_t *odate,
        const int32_t *oprio, const double *rev,
-       const int64_t *cnt, int64_t n,
+       const int64_t *cnt, const uint8_t *flags, int64_t n,
        int64_t *c_okey, int32_t *c_odate, int32_t *c_oprio,
-       double *c_rev, int64_t *c_cnt)
+       double *c_rev, int64_t *c_cnt, uint8_t *c_flags)
 {
     /* one wave per block: 64 threads × K candidates → 640 in LDS */
     __shared__ double s_rev[64 * TOPK];
@@ -2574,7 +2574,9 @@ k_topn(const int64_t *okey, const int32_t *odate,
     for (int64_t i = start; i < n; i += stride)
     {
         double r = rev[i];
-        int32_t d = odate[i];
+        /* NULL dates (outer-join groups) rank LAST on revenue ties: PG
+         * ORDER BY o_orderdate ASC defaults to NULLS LAST */
+        int32_t d = (flags && (flags[i] & 2)) ? INT32_MAX : odate[i];
         /* fails here for all but ~K·blocks·log(n) elements */
         if (t_idx[TOPK - 1] >= 0 && topn_less(r, d, t_rev[TOPK - 1], t_date[TOPK - 1]))
             continue;
@@ -2616,6 +2618,7 @@ k_topn(const int64_t *okey, const int32_t *odate,
             c_oprio[w] = oprio[ib];
             c_rev[w] = rev[ib];
             c_cnt[w] = cnt[ib];
+            c_flags[w] = flags ? flags[ib] : 0;
             s_idx[best] = -1;
         }
     }
@@ -5476,34 +5479,30 @@ extern "C" gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups)
  * of nsegs×N rows is the reference's final Gather/Limit (trivial). */
 extern "C" gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *nout)
 {
-    if (q && q->desc.fact_join == 1)
-    {
-        /* ORDER BY revenue DESC, o_orderdate over groups with NULL dates
-         * needs NULLS-LAST tie handling the device kernel doesn't model;
-         * fetch gx_q3_result and sort host-side for outer plans */
-        set_err(q->ctx, "gx_q3_topn is inner-join only (outer groups carry "
-                        "NULL dates)%s", "");
-        return GX_ERR_INVALID;
-    }
     if (!q || !q->ran || topn <= 0 || topn > 10) return GX_ERR_STATE;
     gx_ctx *ctx = q->ctx;
     hipStream_t s = ctx->stream;
     int64_t n = q->ngroups;
     int grid = 256;
-    devbuf c_okey, c_odate, c_oprio, c_rev, c_cnt;
+    devbuf c_okey, c_odate, c_oprio, c_rev, c_cnt, c_fl;
     HIP_CHK(ctx, c_okey.alloc(grid * 10 * 8));
     HIP_CHK(ctx, c_odate.alloc(grid * 10 * 4));
     HIP_CHK(ctx, c_oprio.alloc(grid * 10 * 4));
     HIP_CHK(ctx, c_rev.alloc(grid * 10 * 8));
     HIP_CHK(ctx, c_cnt.alloc(grid * 10 * 8));
+    HIP_CHK(ctx, c_fl.alloc(grid * 10));
+    const uint8_t *gflags = q->desc.fact_join == 1 ? q->r_flags : nullptr;
     hipLaunchKernelGGL(k_topn, dim3(grid), dim3(64), 0, s,
-                       q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt, n,
+                       q->r_okey, q->r_odate, q->r_oprio, q->r_rev, q->r_cnt,
+                       gflags, n,
                        c_okey.as<int64_t>(), c_odate.as<int32_t>(),
                        c_oprio.as<int32_t>(), c_rev.as<double>(),
-                       c_cnt.as<int64_t>());
+                       c_cnt.as<int64_t>(), c_fl.as<uint8_t>());
     std::vector<int64_t> hk(grid * 10), hc(grid * 10);
     std::vector<int32_t> hd(grid * 10), hp(grid * 10);
     std::vector<double> hr(grid * 10);
+    std::vector<uint8_t> hf(grid * 10);
+    HIP_CHK(ctx, hipMemcpyAsync(hf.data(), c_fl.p, grid * 10, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(hk.data(), c_okey.p, grid * 10 * 8, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(hd.data(), c_odate.p, grid * 10 * 4, hipMemcpyDeviceToHost, s));
     HIP_CHK(ctx, hipMemcpyAsync(hp.data(), c_oprio.p, grid * 10 * 4, hipMemcpyDeviceToHost, s));
@@ -5516,7 +5515,9 @@ extern "C" gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *n
         if (hk[i] >= 0) idx.push_back(i);
     std::sort(idx.begin(), idx.end(), [&](int a, int b) {
         if (hr[a] != hr[b]) return hr[a] > hr[b];
-        return hd[a] < hd[b];
+        int32_t da = (hf[a] & 2) ? INT32_MAX : hd[a];
+        int32_t db = (hf[b] & 2) ? INT32_MAX : hd[b];
+        return da < db;
     });
     int64_t m = std::min<int64_t>(topn, (int64_t) idx.size());
     for (int64_t i = 0; i < m; i++)
@@ -5535,6 +5536,9 @@ extern "C" gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *n
         else
             out[i].revenue_num = 0;
         out[i].nitems = hc[idx[i]];
+        out[i].key_is_null = (uint8_t) (hf[idx[i]] & 1);
+        out[i].attrs_null = (uint8_t) ((hf[idx[i]] >> 1) & 1);
+        memset(out[i]._pad, 0, sizeof(out[i]._pad));
     }
     *nout = m;
     return GX_OK;

@@ -1937,3 +1937,22 @@ def test_left_outer_sf100_conservation(ctx, orc):
     # group keys unique across matched+unmatched
     assert len(np.unique(r["l_orderkey"])) == len(r["l_orderkey"])
     q.free(); li.free(); ordr.free(); cust.free()
+
+
+def test_topn_left_outer_nulls_last(ctx, orc):
+    """gx_q3_topn over an outer plan: ORDER BY revenue DESC, o_orderdate
+    with NULL dates ranking LAST on revenue ties (PG NULLS LAST)."""
+    rng = np.random.default_rng(121)
+    cust, ordr, li, d = _join_variety_tables(ctx, orc, rng)
+    base = _join_variety_desc(cust, ordr, li, "semi")
+    base["fact_join"] = "left_outer"
+    q = ctx.q3_desc(base).run()
+    r = q.result()
+    top = q.topn(10)
+    # host reference: sort all groups by (rev desc, eff-date asc)
+    eff = np.where(r["attrs_null"], np.iinfo(np.int32).max, r["o_orderdate"])
+    order = np.lexsort((eff, -r["revenue"]))
+    want = order[:len(top["l_orderkey"])]
+    np.testing.assert_allclose(top["revenue"], r["revenue"][want], rtol=1e-12)
+    np.testing.assert_array_equal(top["attrs_null"], r["attrs_null"][want])
+    li.free(); ordr.free(); cust.free()
