@@ -290,7 +290,8 @@ gx_status gx_q3_stats_get(const gx_q3 *q, gx_q3_stats *out);
 /* groups of THIS segment, sorted by l_orderkey asc; caller frees with gx_free */
 gx_status gx_q3_result(gx_q3 *q, gx_q3_group **out, int64_t *ngroups);
 /* top-N of this segment's groups (Q3's ORDER BY revenue DESC, o_orderdate
- * LIMIT N; N ≤ 10) — the nodeSort/nodeLimit stage, device-selected */
+ * LIMIT N; N ≤ 10) — the nodeSort/nodeLimit stage, device-selected.
+ * Outer-join groups sort with NULL dates LAST (PG ASC NULLS LAST). */
 gx_status gx_q3_topn(gx_q3 *q, int topn, gx_q3_group *out, int64_t *nout);
 gx_status gx_q3_free(gx_q3 *q);
 
