@@ -1956,3 +1956,27 @@ def test_topn_left_outer_nulls_last(ctx, orc):
     np.testing.assert_allclose(top["revenue"], r["revenue"][want], rtol=1e-12)
     np.testing.assert_array_equal(top["attrs_null"], r["attrs_null"][want])
     li.free(); ordr.free(); cust.free()
+
+
+def test_left_outer_rle_fact_key(ctx, orc):
+    """Outer + RLE fact key: the fused-RLE scan is inner-only, so the key
+    column materializes at prepare — results equal the plain-format outer
+    run."""
+    sf = 0.02
+    cust = ctx.tpch_gen(gx.TPCH_CUSTOMER, sf)
+    ordr = ctx.tpch_gen(gx.TPCH_ORDERS, sf)
+    li = ctx.tpch_gen(gx.TPCH_LINEITEM, sf)
+    li_rle = ctx.tpch_gen(gx.TPCH_LINEITEM_RLEKEY, sf)
+    cut = gx.CUTOFF_19950315
+    base = {"dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3, "mid_filter": (2, "<", cut),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, ">", cut), "fact_join": "left_outer"}
+    want = ctx.q3_desc(base).run().result()
+    got = ctx.q3_desc(dict(base, fact=li_rle)).run().result()
+    np.testing.assert_array_equal(got["l_orderkey"], want["l_orderkey"])
+    np.testing.assert_array_equal(got["nitems"], want["nitems"])
+    np.testing.assert_array_equal(got["attrs_null"], want["attrs_null"])
+    np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
+    li_rle.free(); li.free(); ordr.free(); cust.free()
