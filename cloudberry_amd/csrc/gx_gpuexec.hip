@@ -5128,7 +5128,9 @@ extern "C" gx_status gx_q3_run(gx_q3 *q)
             return GX_ERR_INVALID;
         }
         int want_kw = (kmax < (1ULL << 32)) ? 4 : 8;
-        uint64_t tslots = (uint64_t) pow2_at_least(qual * 2);
+        /* same tuned fill factor as the local path (r2 sweep: tf=300) */
+        int mtf = env_int("GX_TABLE_FACTOR_PCT", 300);
+        uint64_t tslots = (uint64_t) pow2_at_least(qual * mtf / 100 + 1);
         /* motion path sizes from the exchanged counts each run; qual can
          * grow within the same pow2 table size, so rescap is checked too */
         if (q->tkey == nullptr || tslots > q->tmask + 1 || want_kw != q->key_width ||
