@@ -812,3 +812,28 @@ def test_route_multi_null_only_rows():
     out = orc.route_multi(vals, types, 8, isnull=nulls)
     want = orc.lib.orc_jump_consistent_hash(0, 8)
     assert (out == want).all()
+
+
+def test_cdbhash_multi_boundary_values():
+    """Boundary attributes through the rotate-combine chain: INT32_MIN/MAX
+    as int4, INT64_MIN/MAX as int8, zeros, sign boundaries — vs the
+    independent restatement."""
+    cases = [
+        ([0], [0]), ([0], [1]),
+        ([-1], [0]), ([-1], [1]),
+        ([2**31 - 1], [1]), ([-2**31], [1]),
+        ([2**63 - 1], [0]), ([-2**63], [0]),
+        ([2**31, -2**31], [0, 1]),
+        ([2**63 - 1, -2**63, 0], [0, 0, 1]),
+        ([-2**31, 2**31 - 1, -1, 0, 1], [1, 1, 0, 1, 0]),
+    ]
+    for vals, types in cases:
+        v = np.array(vals, np.int64)
+        t = np.array(types, np.int32)
+        want = _py_cdbhash_multi(v, t)
+        got = orc.cdbhash_multi(v, t)
+        assert got == want, (vals, types)
+        # and the reduce step at several segment counts
+        for nsegs in (2, 3, 8, 17, 64):
+            r = orc.route_multi(v.reshape(1, -1), t, nsegs)[0]
+            assert 0 <= r < nsegs
