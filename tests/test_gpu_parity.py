@@ -1980,3 +1980,19 @@ def test_left_outer_rle_fact_key(ctx, orc):
     np.testing.assert_array_equal(got["attrs_null"], want["attrs_null"])
     np.testing.assert_allclose(got["revenue"], want["revenue"], rtol=1e-12)
     li_rle.free(); li.free(); ordr.free(); cust.free()
+
+
+def test_partition_multi_boundaries(ctx, orc):
+    """Boundary attributes through the GPU rotate-combine chain."""
+    cases = [
+        (np.array([[0, 0]], np.int64), [0, 1]),
+        (np.array([[2**63 - 1, -2**63]], np.int64), [0, 0]),
+        (np.array([[-2**31, 2**31 - 1]], np.int64), [1, 1]),
+        (np.array([[-1, 1, 0]], np.int64), [0, 1, 0]),
+    ]
+    for vals, types in cases:
+        t = np.array(types, np.int32)
+        for nsegs in (2, 8, 64):
+            got = ctx.partition_multi(vals, t, nsegs)
+            want = orc.route_multi(vals, t, nsegs)
+            np.testing.assert_array_equal(got, want)
