@@ -1996,3 +1996,88 @@ def test_partition_multi_boundaries(ctx, orc):
             got = ctx.partition_multi(vals, t, nsegs)
             want = orc.route_multi(vals, t, nsegs)
             np.testing.assert_array_equal(got, want)
+
+
+def test_q3_desc_fuzz_outer(ctx, orc):
+    """Plan fuzz over the fact-join axis: random inner/left_outer plans
+    with NULL-bearing fact keys, checked against a numpy outer model
+    (matched groups + NULL-attr unmatched groups + one NULL-key group)."""
+    rng = np.random.default_rng(4321)
+    OPS = ["<", ">", "==", "!=", "<=", ">="]
+
+    def np_cmp(v, op, lit):
+        return {"<": v < lit, ">": v > lit, "==": v == lit, "!=": v != lit,
+                "<=": v <= lit, ">=": v >= lit}[op]
+
+    for trial in range(8):
+        nc = int(rng.integers(50, 300))
+        no = int(rng.integers(200, 1000))
+        nl = int(rng.integers(500, 4000))
+        c_keys = np.arange(1, nc + 1, dtype=np.int64)
+        c_seg = rng.integers(0, 4, nc).astype(np.int8)
+        o_keys = np.arange(1, no + 1, dtype=np.int64)
+        o_cust = rng.integers(1, nc + 1, no).astype(np.int64)
+        o_date = rng.integers(-400, 400, no).astype(np.int32)
+        o_prio = rng.integers(0, 5, no).astype(np.int32)
+        li_keys = rng.integers(1, no + 200, nl).astype(np.int64)
+        with_nulls = bool(rng.random() < 0.5)
+        li_null = (rng.random(nl) < 0.12) if with_nulls else np.zeros(nl, bool)
+        price = rng.uniform(1, 30, nl)
+        disc = rng.integers(0, 11, nl) / 100.0
+        ship = rng.integers(-400, 400, nl).astype(np.int32)
+        fop = OPS[int(rng.integers(0, 6))]
+        flit = int(rng.integers(-300, 300))
+        mop = OPS[int(rng.integers(0, 6))]
+        mlit = int(rng.integers(-300, 300))
+
+        cust = ctx.bind([(orc.aocs_encode(c_keys), 8, nc),
+                         (orc.aocs_encode(c_seg), 1, nc)])
+        ordr = ctx.bind([(orc.aocs_encode(o_keys), 8, no),
+                         (orc.aocs_encode(o_cust), 8, no),
+                         (orc.aocs_encode(o_date), 4, no),
+                         (orc.aocs_encode(o_prio), 4, no)])
+        if with_nulls:
+            key_stream = (orc.aocs_encode_orig_nulls(li_keys, li_null), 8, nl, 1)
+        else:
+            key_stream = (orc.aocs_encode(li_keys), 8, nl)
+        li = ctx.bind([key_stream,
+                       (orc.aocs_encode(price), 8, nl),
+                       (orc.aocs_encode(disc), 8, nl),
+                       (orc.aocs_encode(ship), 4, nl)])
+        got = ctx.q3_desc({
+            "dim": cust, "dim_key_col": 0, "dim_filter": (1, "==", 0),
+            "mid": ordr, "mid_key_col": 0, "mid_fk_col": 1,
+            "mid_attr1_col": 2, "mid_attr2_col": 3,
+            "mid_filter": (2, mop, mlit),
+            "fact": li, "fact_key_col": 0, "fact_a_col": 1, "fact_b_col": 2,
+            "fact_filter": (3, fop, flit),
+            "fact_join": "left_outer"}).run().result()
+
+        segok = c_keys[c_seg == 0]
+        om = np_cmp(o_date, mop, mlit) & np.isin(o_cust, segok)
+        qual_keys = o_keys[om]
+        lm = np_cmp(ship, fop, flit)
+        matched = lm & ~li_null & np.isin(li_keys, qual_keys)
+        unmatched = lm & ~li_null & ~np.isin(li_keys, qual_keys)
+        nullrows = lm & li_null
+
+        mk, mc2 = np.unique(li_keys[matched], return_counts=True)
+        uk, uc2 = np.unique(li_keys[unmatched], return_counts=True)
+        gm = ~got["attrs_null"]
+        gu = got["attrs_null"] & ~got["key_is_null"]
+        np.testing.assert_array_equal(np.sort(got["l_orderkey"][gm]), mk,
+                                      err_msg=f"trial {trial} matched keys")
+        np.testing.assert_array_equal(np.sort(got["l_orderkey"][gu]), uk,
+                                      err_msg=f"trial {trial} unmatched keys")
+        order = np.argsort(got["l_orderkey"][gu])
+        np.testing.assert_array_equal(got["nitems"][gu][order], uc2,
+                                      err_msg=f"trial {trial} unmatched counts")
+        nk = got["key_is_null"]
+        if nullrows.any():
+            assert nk.sum() == 1 and int(got["nitems"][nk][0]) == int(nullrows.sum()), \
+                f"trial {trial} null group"
+        else:
+            assert nk.sum() == 0, f"trial {trial} spurious null group"
+        assert int(got["nitems"].sum()) == int(matched.sum() + unmatched.sum()
+                                               + nullrows.sum())
+        li.free(); ordr.free(); cust.free()
